@@ -1,0 +1,236 @@
+"""Execution context: runner selection, device ownership, execution config,
+partition cache, subscriber/event bus.
+
+Ref capabilities: daft-context (DaftContext, subscribers, partition cache),
+common/daft-config (DaftExecutionConfig, /root/reference/src/common/
+daft-config/src/lib.rs:120-200).  MI355X-native difference: the context OWNS
+a GPU (one process per GPU; `cuda:LOCAL_RANK` under torch.distributed) and
+data placement defaults to HBM when a GPU is visible.
+"""
+from __future__ import annotations
+
+import contextlib
+import os
+import threading
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+
+@dataclass
+class ExecutionConfig:
+    """Mirrors the reference's DaftExecutionConfig knobs where meaningful."""
+    scan_tasks_min_size_bytes: int = 96 * 1024 * 1024
+    scan_tasks_max_size_bytes: int = 384 * 1024 * 1024
+    broadcast_join_size_bytes_threshold: int = 64 * 1024 * 1024  # sized for HBM3E, vs 10 MiB on CPU
+    morsel_size_rows: int = 4_000_000          # GPU-scale morsels (128 ≪ CUs need work)
+    target_batch_rows: int = 4_000_000
+    pre_shuffle_merge_threshold: int = 1 << 30
+    shuffle_algorithm: str = "rccl_all_to_all"
+    default_num_partitions: Optional[int] = None
+    parquet_target_filesize: int = 512 * 1024 * 1024
+    parquet_target_row_group_size: int = 16 * 1024 * 1024
+    csv_target_filesize: int = 512 * 1024 * 1024
+    maintain_order: bool = True
+    memory_limit_bytes: Optional[int] = None   # DAFT_MEMORY_LIMIT analog
+    enable_gpu: bool = True
+
+
+@dataclass
+class PlanningConfig:
+    default_io_config: Optional[dict] = None
+
+
+class Subscriber:
+    """Event-bus subscriber ABC (ref: daft/subscribers/abc.py:28)."""
+
+    def on_query_start(self, query_id: str, explain: str) -> None: ...
+    def on_query_end(self, query_id: str, seconds: float,
+                     error: Optional[str]) -> None: ...
+    def on_optimization_start(self, query_id: str) -> None: ...
+    def on_optimization_end(self, query_id: str, seconds: float) -> None: ...
+    def on_exec_start(self, query_id: str, node_names: List[str]) -> None: ...
+    def on_operator_start(self, query_id: str, node_id: int,
+                          name: str) -> None: ...
+    def on_operator_end(self, query_id: str, node_id: int, name: str,
+                        rows_in: int, rows_out: int,
+                        seconds: float) -> None: ...
+    def on_exec_end(self, query_id: str) -> None: ...
+    def on_query_heartbeat(self, query_id: str) -> None: ...
+
+
+class PartitionCache:
+    """In-memory partition-set cache keyed by cache_key (ref:
+    common/partitioning PartitionSetCache)."""
+
+    def __init__(self):
+        self._lock = threading.Lock()
+        self._parts: Dict[str, list] = {}
+
+    def put(self, key: str, partitions: list) -> None:
+        with self._lock:
+            self._parts[key] = partitions
+
+    def get(self, key: str) -> list:
+        with self._lock:
+            if key not in self._parts:
+                raise KeyError(f"partition set {key} not in cache")
+            return self._parts[key]
+
+    def contains(self, key: str) -> bool:
+        with self._lock:
+            return key in self._parts
+
+    def drop(self, key: str) -> None:
+        with self._lock:
+            self._parts.pop(key, None)
+
+    def new_key(self) -> str:
+        return uuid.uuid4().hex
+
+
+class Context:
+    def __init__(self):
+        self.execution_config = ExecutionConfig()
+        self.planning_config = PlanningConfig()
+        self.cache = PartitionCache()
+        self.subscribers: List[Subscriber] = []
+        self._runner = None
+        self._lock = threading.Lock()
+        if os.environ.get("DAFT_MEMORY_LIMIT"):
+            self.execution_config.memory_limit_bytes = \
+                int(os.environ["DAFT_MEMORY_LIMIT"])
+
+    # ---- device ownership -------------------------------------------
+    def device(self) -> torch.device:
+        if self.execution_config.enable_gpu and torch.cuda.is_available():
+            import torch.distributed as dist
+            if dist.is_available() and dist.is_initialized():
+                local = int(os.environ.get("LOCAL_RANK", 0))
+                return torch.device(f"cuda:{local}")
+            return torch.device("cuda:0")
+        return torch.device("cpu")
+
+    def default_device_str(self) -> str:
+        return str(self.device())
+
+    # ---- runner ------------------------------------------------------
+    def runner(self):
+        with self._lock:
+            if self._runner is None:
+                mode = os.environ.get("DAFT_RUNNER", "").lower()
+                if mode in ("", "native"):
+                    from .execution.runner import NativeRunner
+                    self._runner = NativeRunner(self)
+                elif mode in ("distributed", "dist", "flotilla"):
+                    from .distributed.runner import DistributedRunner
+                    self._runner = DistributedRunner(self)
+                else:
+                    raise ValueError(f"unknown DAFT_RUNNER={mode}")
+            return self._runner
+
+    def set_runner(self, runner) -> None:
+        with self._lock:
+            self._runner = runner
+
+    # ---- events ------------------------------------------------------
+    def notify(self, method: str, *args) -> None:
+        for s in self.subscribers:
+            try:
+                getattr(s, method)(*args)
+            except Exception:
+                pass
+
+
+_context: Optional[Context] = None
+_ctx_lock = threading.Lock()
+
+
+def get_context() -> Context:
+    global _context
+    with _ctx_lock:
+        if _context is None:
+            _context = Context()
+        return _context
+
+
+def set_runner_native() -> Context:
+    ctx = get_context()
+    from .execution.runner import NativeRunner
+    ctx.set_runner(NativeRunner(ctx))
+    return ctx
+
+
+def set_runner_distributed(world_size: Optional[int] = None,
+                           backend: Optional[str] = None) -> Context:
+    ctx = get_context()
+    from .distributed.runner import DistributedRunner
+    ctx.set_runner(DistributedRunner(ctx, backend=backend))
+    return ctx
+
+
+def set_execution_config(**kwargs) -> Context:
+    ctx = get_context()
+    for k, v in kwargs.items():
+        if not hasattr(ctx.execution_config, k):
+            raise AttributeError(f"no execution config field {k}")
+        setattr(ctx.execution_config, k, v)
+    return ctx
+
+
+@contextlib.contextmanager
+def execution_config_ctx(**kwargs):
+    ctx = get_context()
+    old = {k: getattr(ctx.execution_config, k) for k in kwargs}
+    try:
+        set_execution_config(**kwargs)
+        yield ctx
+    finally:
+        for k, v in old.items():
+            setattr(ctx.execution_config, k, v)
+
+
+def set_planning_config(**kwargs) -> Context:
+    ctx = get_context()
+    for k, v in kwargs.items():
+        setattr(ctx.planning_config, k, v)
+    return ctx
+
+
+def attach_subscriber(sub: Subscriber) -> None:
+    get_context().subscribers.append(sub)
+
+
+def detach_subscriber(sub: Subscriber) -> None:
+    subs = get_context().subscribers
+    if sub in subs:
+        subs.remove(sub)
+
+
+class Heartbeat:
+    """Background heartbeat thread (ref: daft/runners/heartbeat.py)."""
+
+    def __init__(self, ctx: Context, query_id: str, interval: float = 10.0):
+        self.ctx = ctx
+        self.query_id = query_id
+        self.interval = interval
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def __enter__(self):
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+        return self
+
+    def _run(self):
+        while not self._stop.wait(self.interval):
+            self.ctx.notify("on_query_heartbeat", self.query_id)
+
+    def __exit__(self, *exc):
+        self._stop.set()
+        if self._thread:
+            self._thread.join(timeout=1.0)
+        return False

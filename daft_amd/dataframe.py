@@ -1,0 +1,424 @@
+"""User-facing lazy DataFrame (ref: /root/reference/daft/dataframe/
+dataframe.py — 6.5k LoC API surface; this implements the same verbs over the
+MI355X-native engine)."""
+from __future__ import annotations
+
+from typing import Any, Dict, Iterator, List, Optional, Sequence, Union
+
+from .context import get_context
+from .expressions.expressions import (Agg, AggKind, Alias, ColumnRef,
+                                      Expression, resolve_exprs)
+from .logical.builder import LogicalPlanBuilder
+from .recordbatch import RecordBatch
+from .schema import Schema
+
+ColumnInput = Union[str, Expression]
+
+
+class DataFrame:
+    def __init__(self, builder: LogicalPlanBuilder):
+        self._builder = builder
+        self._result: Optional[List[RecordBatch]] = None
+
+    # ------------------------------------------------------------------
+    @property
+    def schema(self) -> Schema:
+        return self._builder.schema
+
+    def column_names(self) -> List[str]:
+        return self.schema.names()
+
+    def __getitem__(self, name: str) -> Expression:
+        if name not in self.schema:
+            raise KeyError(name)
+        return Expression(ColumnRef(name))
+
+    def explain(self, show_all: bool = False) -> str:
+        s = self._builder.explain()
+        if show_all:
+            s += "\n== Optimized ==\n" + self._builder.optimize().explain()
+        print(s)
+        return s
+
+    def _wrap(self, builder: LogicalPlanBuilder) -> "DataFrame":
+        return DataFrame(builder)
+
+    # ---- transforms ---------------------------------------------------
+    def select(self, *exprs: ColumnInput) -> "DataFrame":
+        return self._wrap(self._builder.select(list(exprs)))
+
+    def with_column(self, name: str, expr: Expression) -> "DataFrame":
+        return self._wrap(self._builder.with_columns([expr.alias(name)]))
+
+    def with_columns(self, cols: Dict[str, Expression]) -> "DataFrame":
+        return self._wrap(self._builder.with_columns(
+            [e.alias(n) for n, e in cols.items()]))
+
+    def with_column_renamed(self, old: str, new: str) -> "DataFrame":
+        return self._wrap(self._builder.rename({old: new}))
+
+    def with_columns_renamed(self, mapping: Dict[str, str]) -> "DataFrame":
+        return self._wrap(self._builder.rename(mapping))
+
+    def exclude(self, *names: str) -> "DataFrame":
+        return self._wrap(self._builder.exclude(list(names)))
+
+    drop = exclude
+
+    def where(self, predicate: Union[Expression, str]) -> "DataFrame":
+        if isinstance(predicate, str):
+            from .sql import sql_expr
+            predicate = sql_expr(predicate)
+        return self._wrap(self._builder.filter(predicate))
+
+    filter = where
+
+    def limit(self, n: int) -> "DataFrame":
+        return self._wrap(self._builder.limit(n))
+
+    def offset(self, n: int) -> "DataFrame":
+        return self._wrap(self._builder.limit(1 << 62, offset=n))
+
+    def head(self, n: int = 10) -> "DataFrame":
+        return self.limit(n)
+
+    def sort(self, by, desc: Union[bool, Sequence[bool]] = False,
+             nulls_first=None) -> "DataFrame":
+        if not isinstance(by, (list, tuple)):
+            by = [by]
+        return self._wrap(self._builder.sort(list(by), desc, nulls_first))
+
+    def distinct(self, *subset: ColumnInput) -> "DataFrame":
+        return self._wrap(self._builder.distinct(list(subset) or None))
+
+    unique = distinct
+    drop_duplicates = distinct
+
+    def explode(self, *cols: ColumnInput) -> "DataFrame":
+        return self._wrap(self._builder.explode(list(cols)))
+
+    def unpivot(self, ids: Sequence[ColumnInput],
+                values: Sequence[ColumnInput] = (),
+                variable_name: str = "variable",
+                value_name: str = "value") -> "DataFrame":
+        if not values:
+            idn = {c if isinstance(c, str) else c.name() for c in ids}
+            values = [n for n in self.column_names() if n not in idn]
+        return self._wrap(self._builder.unpivot(
+            list(ids), list(values), variable_name, value_name))
+
+    melt = unpivot
+
+    def pivot(self, group_by, pivot_col, value_col, agg_fn: str,
+              names: Optional[List[str]] = None) -> "DataFrame":
+        if not isinstance(group_by, (list, tuple)):
+            group_by = [group_by]
+        if names is None:
+            pname = pivot_col if isinstance(pivot_col, str) \
+                else pivot_col.name()
+            vals = self.select(pivot_col).distinct().to_pydict()[pname]
+            names = sorted(str(v) for v in vals if v is not None)
+        return self._wrap(self._builder.pivot(
+            list(group_by), pivot_col, value_col, agg_fn, names))
+
+    def sample(self, fraction: float, with_replacement: bool = False,
+               seed: Optional[int] = None) -> "DataFrame":
+        return self._wrap(self._builder.sample(fraction, with_replacement,
+                                               seed))
+
+    def concat(self, other: "DataFrame") -> "DataFrame":
+        return self._wrap(self._builder.concat(other._builder))
+
+    union_all = concat
+
+    def union(self, other: "DataFrame") -> "DataFrame":
+        return self.concat(other).distinct()
+
+    def intersect(self, other: "DataFrame") -> "DataFrame":
+        cols = self.column_names()
+        return self.join(other, on=cols, how="semi").distinct()
+
+    def except_distinct(self, other: "DataFrame") -> "DataFrame":
+        cols = self.column_names()
+        return self.join(other, on=cols, how="anti").distinct()
+
+    def join(self, other: "DataFrame", on=None, left_on=None, right_on=None,
+             how: str = "inner", suffix: str = "_right",
+             prefix: Optional[str] = None) -> "DataFrame":
+        if how == "cross":
+            return self._wrap(self._builder.cross_join(other._builder,
+                                                       suffix))
+        if on is not None:
+            left_on = right_on = on
+        assert left_on is not None and right_on is not None, \
+            "join requires on= or left_on=/right_on="
+        if not isinstance(left_on, (list, tuple)):
+            left_on = [left_on]
+        if not isinstance(right_on, (list, tuple)):
+            right_on = [right_on]
+        return self._wrap(self._builder.join(
+            other._builder, list(left_on), list(right_on), how, suffix,
+            prefix))
+
+    def repartition(self, num_partitions: Optional[int],
+                    *by: ColumnInput) -> "DataFrame":
+        scheme = "hash" if by else "random"
+        return self._wrap(self._builder.repartition(
+            num_partitions, scheme, list(by) or None))
+
+    def into_partitions(self, n: int) -> "DataFrame":
+        return self._wrap(self._builder.into_partitions(n))
+
+    def into_batches(self, batch_size: int) -> "DataFrame":
+        return self._wrap(self._builder.into_batches(batch_size))
+
+    def add_monotonically_increasing_id(self, column_name: str = "id"
+                                        ) -> "DataFrame":
+        return self._wrap(
+            self._builder.add_monotonically_increasing_id(column_name))
+
+    def transform(self, fn, *args, **kwargs) -> "DataFrame":
+        out = fn(self, *args, **kwargs)
+        assert isinstance(out, DataFrame), "transform fn must return DataFrame"
+        return out
+
+    # ---- window --------------------------------------------------------
+    def with_window_columns(self, cols: Dict[str, Expression]) -> "DataFrame":
+        """Add window-function columns (exprs built with .over(window))."""
+        from .physical.window import WindowFn
+        names = list(cols.keys())
+        nodes = []
+        spec = None
+        for e in cols.values():
+            n = e._node
+            base = n.child if isinstance(n, Alias) else n
+            assert isinstance(base, WindowFn), "expected .over(window) exprs"
+            if spec is None:
+                spec = base.spec
+            nodes.append(base)
+        assert spec is not None
+        return self._wrap(self._builder.window(
+            nodes, spec.partition_by_exprs, spec.order_by_exprs,
+            spec.descending, names))
+
+    # ---- aggregation ----------------------------------------------------
+    def groupby(self, *group_by: ColumnInput) -> "GroupedDataFrame":
+        cols = []
+        for g in group_by:
+            if isinstance(g, (list, tuple)):
+                cols.extend(g)
+            else:
+                cols.append(g)
+        return GroupedDataFrame(self, cols)
+
+    group_by = groupby
+
+    def agg(self, *exprs) -> "DataFrame":
+        flat = []
+        for e in exprs:
+            if isinstance(e, (list, tuple)):
+                flat.extend(e)
+            else:
+                flat.append(e)
+        return self._wrap(self._builder.aggregate(flat, []))
+
+    def _agg_all(self, kind: str, cols: Sequence[ColumnInput]) -> "DataFrame":
+        if not cols:
+            cols = [f.name for f in self.schema if f.dtype.is_numeric()]
+        exprs = []
+        for c in cols:
+            e = Expression(ColumnRef(c)) if isinstance(c, str) else c
+            exprs.append(getattr(e, kind)())
+        return self.agg(*exprs)
+
+    def sum(self, *cols): return self._agg_all("sum", cols)
+    def mean(self, *cols): return self._agg_all("mean", cols)
+    def min(self, *cols): return self._agg_all("min", cols)
+    def max(self, *cols): return self._agg_all("max", cols)
+    def stddev(self, *cols): return self._agg_all("stddev", cols)
+    def any_value(self, *cols): return self._agg_all("any_value", cols)
+    def agg_list(self, *cols): return self._agg_all("agg_list", cols)
+
+    def count(self, *cols) -> "DataFrame":
+        if not cols:
+            return self.agg(Expression(Agg(AggKind.COUNT_ALL, None)))
+        return self._agg_all("count", cols)
+
+    def count_rows(self) -> int:
+        df = self.agg(Expression(Alias(Agg(AggKind.COUNT_ALL, None),
+                                       "count")))
+        return int(df.to_pydict()["count"][0])
+
+    def __len__(self) -> int:
+        return self.count_rows()
+
+    # ---- writes ---------------------------------------------------------
+    def write_parquet(self, root_dir: str, write_mode: str = "overwrite",
+                      partition_cols=None, compression: str = "snappy"
+                      ) -> "DataFrame":
+        b = self._builder.write("parquet", root_dir, write_mode,
+                                partition_cols,
+                                {"compression": compression})
+        return self._wrap(b).collect()
+
+    def write_csv(self, root_dir: str, write_mode: str = "overwrite",
+                  partition_cols=None) -> "DataFrame":
+        b = self._builder.write("csv", root_dir, write_mode, partition_cols)
+        return self._wrap(b).collect()
+
+    def write_json(self, root_dir: str, write_mode: str = "overwrite",
+                   partition_cols=None) -> "DataFrame":
+        b = self._builder.write("json", root_dir, write_mode, partition_cols)
+        return self._wrap(b).collect()
+
+    # ---- execution -------------------------------------------------------
+    def collect(self) -> "DataFrame":
+        if self._result is not None:
+            return self
+        ctx = get_context()
+        parts = ctx.runner().run(self._builder)
+        key = ctx.cache.new_key()
+        ctx.cache.put(key, parts)
+        rows = sum(len(p) for p in parts)
+        size = sum(p.size_bytes() for p in parts)
+        from .logical.builder import LogicalPlanBuilder as B
+        out = DataFrame(B.from_in_memory(self.schema, key, rows, size))
+        out._result = parts
+        return out
+
+    def iter_partitions(self) -> Iterator[RecordBatch]:
+        if self._result is not None:
+            yield from self._result
+            return
+        ctx = get_context()
+        yield from ctx.runner().run_iter(self._builder)
+
+    def iter_rows(self) -> Iterator[Dict[str, Any]]:
+        for part in self.iter_partitions():
+            d = part.to_pydict()
+            names = list(d.keys())
+            for i in range(len(part)):
+                yield {n: d[n][i] for n in names}
+
+    def to_pydict(self) -> Dict[str, list]:
+        df = self.collect()
+        parts = df._result
+        if not parts:
+            return {f.name: [] for f in self.schema}
+        out: Dict[str, list] = {f.name: [] for f in parts[0].schema}
+        for p in parts:
+            d = p.to_pydict()
+            for k, v in d.items():
+                out[k].extend(v)
+        return out
+
+    def to_pylist(self) -> List[Dict[str, Any]]:
+        d = self.to_pydict()
+        names = list(d.keys())
+        n = len(d[names[0]]) if names else 0
+        return [{k: d[k][i] for k in names} for i in range(n)]
+
+    def to_arrow(self):
+        import pyarrow as pa
+        df = self.collect()
+        tables = [p.to_arrow() for p in df._result if len(p)]
+        if not tables:
+            return RecordBatch.empty(self.schema).to_arrow()
+        return pa.concat_tables(tables)
+
+    def to_pandas(self):
+        return self.to_arrow().to_pandas()
+
+    def to_torch_map_dataset(self):
+        d = self.to_pydict()
+        from .utils.torch_data import DictDataset
+        return DictDataset(d)
+
+    def to_recordbatch(self) -> RecordBatch:
+        df = self.collect()
+        parts = [p for p in df._result if len(p)]
+        if not parts:
+            return RecordBatch.empty(self.schema)
+        return RecordBatch.concat(parts) if len(parts) > 1 else parts[0]
+
+    def num_partitions(self) -> int:
+        df = self.collect()
+        return len(df._result)
+
+    def show(self, n: int = 8) -> None:
+        d = self.limit(n).to_pydict()
+        print(_format_table(d, self.schema))
+
+    def __repr__(self) -> str:
+        if self._result is not None:
+            d = self.limit(8).to_pydict()
+            return _format_table(d, self.schema)
+        return f"DataFrame(schema={self.schema!r}, lazy)"
+
+
+class GroupedDataFrame:
+    def __init__(self, df: DataFrame, group_by: List[ColumnInput]):
+        self.df = df
+        self.group_by = group_by
+
+    def agg(self, *exprs) -> DataFrame:
+        flat = []
+        for e in exprs:
+            if isinstance(e, (list, tuple)):
+                flat.extend(e)
+            else:
+                flat.append(e)
+        return self.df._wrap(
+            self.df._builder.aggregate(flat, self.group_by))
+
+    def _agg_all(self, kind: str, cols) -> DataFrame:
+        gnames = {c if isinstance(c, str) else c.name() for c in self.group_by}
+        if not cols:
+            cols = [f.name for f in self.df.schema
+                    if f.dtype.is_numeric() and f.name not in gnames]
+        exprs = []
+        for c in cols:
+            e = Expression(ColumnRef(c)) if isinstance(c, str) else c
+            exprs.append(getattr(e, kind)())
+        return self.agg(*exprs)
+
+    def sum(self, *cols): return self._agg_all("sum", cols)
+    def mean(self, *cols): return self._agg_all("mean", cols)
+    def min(self, *cols): return self._agg_all("min", cols)
+    def max(self, *cols): return self._agg_all("max", cols)
+    def any_value(self, *cols): return self._agg_all("any_value", cols)
+    def agg_list(self, *cols): return self._agg_all("agg_list", cols)
+    def count(self, *cols):
+        if not cols:
+            return self.agg(Expression(Alias(Agg(AggKind.COUNT_ALL, None),
+                                             "count")))
+        return self._agg_all("count", cols)
+
+
+def _format_table(d: Dict[str, list], schema: Schema) -> str:
+    names = list(d.keys())
+    if not names:
+        return "(empty)"
+    n = len(d[names[0]])
+    header = [f"{nm} ({schema[nm].dtype!r})" if nm in schema else nm
+              for nm in names]
+    rows = [[_fmt_cell(d[nm][i]) for nm in names] for i in range(n)]
+    widths = [max(len(h), *(len(r[j]) for r in rows)) if rows else len(h)
+              for j, h in enumerate(header)]
+    sep = "+" + "+".join("-" * (w + 2) for w in widths) + "+"
+    out = [sep,
+           "|" + "|".join(f" {h:<{w}} " for h, w in zip(header, widths)) + "|",
+           sep]
+    for r in rows:
+        out.append("|" + "|".join(f" {c:<{w}} "
+                                  for c, w in zip(r, widths)) + "|")
+    out.append(sep)
+    out.append(f"({n} rows shown)")
+    return "\n".join(out)
+
+
+def _fmt_cell(v) -> str:
+    if v is None:
+        return "None"
+    s = str(v)
+    return s if len(s) <= 30 else s[:27] + "..."

@@ -1,0 +1,1356 @@
+"""Expression IR + vectorized evaluation.
+
+The analog of the reference's daft-dsl Expr enum
+(/root/reference/src/daft-dsl/src/expr/mod.rs:222-330): Column, Alias,
+Literal, BinaryOp, Cast, Not/IsNull/NotNull/FillNull, IsIn, Between, IfElse,
+Agg, ScalarFn, Coalesce, plus python UDFs.  Evaluation maps an expression
+tree over a RecordBatch into a Series, launching GPU kernels column-at-a-time
+(ref eval: daft-recordbatch/src/lib.rs:875 eval_expression_list).
+"""
+from __future__ import annotations
+
+import datetime as _dt
+from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
+
+import torch
+
+from ..schema import DataType, Field, Schema, TypeKind, supertype
+from ..series import Series, lit_series
+
+# ---------------------------------------------------------------------------
+# nodes
+# ---------------------------------------------------------------------------
+
+
+class ExprNode:
+    """Base expression node; immutable."""
+
+    def children(self) -> List["ExprNode"]:
+        return []
+
+    def with_children(self, ch: List["ExprNode"]) -> "ExprNode":
+        assert not ch
+        return self
+
+    # output name resolution (daft semantics: leftmost column name)
+    def out_name(self) -> str:
+        ch = self.children()
+        if ch:
+            return ch[0].out_name()
+        return "literal"
+
+    def to_field(self, schema: Schema) -> Field:
+        raise NotImplementedError(type(self))
+
+    def evaluate(self, batch) -> Series:
+        raise NotImplementedError(type(self))
+
+    def is_aggregation(self) -> bool:
+        """Does this subtree contain an Agg node?"""
+        return any(c.is_aggregation() for c in self.children())
+
+    def column_refs(self) -> List[str]:
+        out: List[str] = []
+        seen = set()
+
+        def rec(n: ExprNode):
+            if isinstance(n, ColumnRef):
+                if n.name not in seen:
+                    seen.add(n.name)
+                    out.append(n.name)
+            for c in n.children():
+                rec(c)
+        rec(self)
+        return out
+
+    def semantic_id(self) -> str:
+        """Canonical string for plan fingerprinting / CSE."""
+        return repr(self)
+
+
+class ColumnRef(ExprNode):
+    def __init__(self, name: str):
+        self.name = name
+
+    def out_name(self) -> str:
+        return self.name
+
+    def to_field(self, schema: Schema) -> Field:
+        return schema[self.name]
+
+    def evaluate(self, batch) -> Series:
+        return batch.column(self.name)
+
+    def __repr__(self):
+        return f"col({self.name})"
+
+
+class Literal(ExprNode):
+    def __init__(self, value: Any, dtype: Optional[DataType] = None):
+        self.value = value
+        if dtype is None:
+            from ..series import _infer_dtype
+            dtype = _infer_dtype([value])
+        self.dtype = dtype
+
+    def to_field(self, schema: Schema) -> Field:
+        return Field("literal", self.dtype)
+
+    def evaluate(self, batch) -> Series:
+        s = lit_series("literal", self.value, self.dtype, device=batch.device)
+        return s.broadcast(len(batch)) if len(batch) != 1 else s
+
+    def __repr__(self):
+        return f"lit({self.value!r})"
+
+
+class Alias(ExprNode):
+    def __init__(self, child: ExprNode, name: str):
+        self.child = child
+        self.name = name
+
+    def children(self):
+        return [self.child]
+
+    def with_children(self, ch):
+        return Alias(ch[0], self.name)
+
+    def out_name(self) -> str:
+        return self.name
+
+    def to_field(self, schema: Schema) -> Field:
+        return Field(self.name, self.child.to_field(schema).dtype)
+
+    def evaluate(self, batch) -> Series:
+        return self.child.evaluate(batch).rename(self.name)
+
+    def is_aggregation(self):
+        return self.child.is_aggregation()
+
+    def __repr__(self):
+        return f"{self.child!r}.alias({self.name})"
+
+
+_ARITH = {"add", "sub", "mul", "div", "floordiv", "mod", "pow"}
+_CMP = {"eq", "ne", "lt", "le", "gt", "ge"}
+_LOGIC = {"and", "or", "xor"}
+
+
+class BinaryOp(ExprNode):
+    def __init__(self, op: str, left: ExprNode, right: ExprNode):
+        self.op = op
+        self.left = left
+        self.right = right
+
+    def children(self):
+        return [self.left, self.right]
+
+    def with_children(self, ch):
+        return BinaryOp(self.op, ch[0], ch[1])
+
+    def to_field(self, schema: Schema) -> Field:
+        lf = self.left.to_field(schema)
+        rf = self.right.to_field(schema)
+        if self.op in _CMP or self.op in _LOGIC:
+            return Field(lf.name, DataType.bool())
+        if self.op == "div":
+            return Field(lf.name, DataType.float64())
+        if self.op == "sub" and lf.dtype.kind == rf.dtype.kind == TypeKind.DATE:
+            return Field(lf.name, DataType.duration("us"))
+        if lf.dtype.is_temporal():
+            return Field(lf.name, lf.dtype)
+        if lf.dtype.is_string() and self.op == "add":
+            return Field(lf.name, DataType.string())
+        return Field(lf.name, supertype(lf.dtype, rf.dtype))
+
+    def evaluate(self, batch) -> Series:
+        l = self.left.evaluate(batch)
+        r = self.right.evaluate(batch)
+        if self.op in _CMP:
+            return l.compare(r, self.op)
+        if self.op in _LOGIC:
+            return l.logical(r, self.op)
+        if l.dtype.is_string() and self.op == "add":
+            from ..kernels import strings as strk
+            return strk.concat_str([l, r])
+        from .. import kernels
+        return kernels.binary_op(l, r, self.op)
+
+    def __repr__(self):
+        return f"({self.left!r} {self.op} {self.right!r})"
+
+
+class Not(ExprNode):
+    def __init__(self, child: ExprNode):
+        self.child = child
+
+    def children(self):
+        return [self.child]
+
+    def with_children(self, ch):
+        return Not(ch[0])
+
+    def to_field(self, schema):
+        return Field(self.child.to_field(schema).name, DataType.bool())
+
+    def evaluate(self, batch) -> Series:
+        return self.child.evaluate(batch).logical_not()
+
+    def __repr__(self):
+        return f"~{self.child!r}"
+
+
+class IsNull(ExprNode):
+    def __init__(self, child: ExprNode, negate: bool = False):
+        self.child = child
+        self.negate = negate
+
+    def children(self):
+        return [self.child]
+
+    def with_children(self, ch):
+        return IsNull(ch[0], self.negate)
+
+    def to_field(self, schema):
+        return Field(self.child.to_field(schema).name, DataType.bool())
+
+    def evaluate(self, batch) -> Series:
+        s = self.child.evaluate(batch)
+        return s.not_null() if self.negate else s.is_null()
+
+    def __repr__(self):
+        return f"{self.child!r}.{'not_null' if self.negate else 'is_null'}()"
+
+
+class FillNull(ExprNode):
+    def __init__(self, child: ExprNode, fill: ExprNode):
+        self.child = child
+        self.fill = fill
+
+    def children(self):
+        return [self.child, self.fill]
+
+    def with_children(self, ch):
+        return FillNull(ch[0], ch[1])
+
+    def to_field(self, schema):
+        f = self.child.to_field(schema)
+        return Field(f.name, supertype(f.dtype, self.fill.to_field(schema).dtype))
+
+    def evaluate(self, batch) -> Series:
+        s = self.child.evaluate(batch)
+        f = self.fill.evaluate(batch)
+        if len(f) == 1:
+            f = f.broadcast(len(s))
+        return s.fill_null(f)
+
+    def __repr__(self):
+        return f"{self.child!r}.fill_null({self.fill!r})"
+
+
+class Cast(ExprNode):
+    def __init__(self, child: ExprNode, dtype: DataType):
+        self.child = child
+        self.dtype = dtype
+
+    def children(self):
+        return [self.child]
+
+    def with_children(self, ch):
+        return Cast(ch[0], self.dtype)
+
+    def to_field(self, schema):
+        return Field(self.child.to_field(schema).name, self.dtype)
+
+    def evaluate(self, batch) -> Series:
+        return self.child.evaluate(batch).cast(self.dtype)
+
+    def __repr__(self):
+        return f"{self.child!r}.cast({self.dtype!r})"
+
+
+class IsIn(ExprNode):
+    def __init__(self, child: ExprNode, values: List[Any]):
+        self.child = child
+        self.values = values
+
+    def children(self):
+        return [self.child]
+
+    def with_children(self, ch):
+        return IsIn(ch[0], self.values)
+
+    def to_field(self, schema):
+        return Field(self.child.to_field(schema).name, DataType.bool())
+
+    def evaluate(self, batch) -> Series:
+        s = self.child.evaluate(batch)
+        vals = Series.from_pylist("values", list(self.values),
+                                  device=s.device)
+        return s.is_in(vals)
+
+    def __repr__(self):
+        return f"{self.child!r}.is_in({self.values!r})"
+
+
+class Between(ExprNode):
+    def __init__(self, child: ExprNode, lo: ExprNode, hi: ExprNode):
+        self.child = child
+        self.lo = lo
+        self.hi = hi
+
+    def children(self):
+        return [self.child, self.lo, self.hi]
+
+    def with_children(self, ch):
+        return Between(ch[0], ch[1], ch[2])
+
+    def to_field(self, schema):
+        return Field(self.child.to_field(schema).name, DataType.bool())
+
+    def evaluate(self, batch) -> Series:
+        s = self.child.evaluate(batch)
+        lo = self.lo.evaluate(batch)
+        hi = self.hi.evaluate(batch)
+        return s.between(lo, hi)
+
+    def __repr__(self):
+        return f"{self.child!r}.between({self.lo!r},{self.hi!r})"
+
+
+class IfElse(ExprNode):
+    def __init__(self, pred: ExprNode, truthy: ExprNode, falsy: ExprNode):
+        self.pred = pred
+        self.truthy = truthy
+        self.falsy = falsy
+
+    def children(self):
+        return [self.pred, self.truthy, self.falsy]
+
+    def with_children(self, ch):
+        return IfElse(ch[0], ch[1], ch[2])
+
+    def out_name(self):
+        return self.truthy.out_name()
+
+    def to_field(self, schema):
+        t = self.truthy.to_field(schema)
+        f = self.falsy.to_field(schema)
+        return Field(t.name, supertype(t.dtype, f.dtype))
+
+    def evaluate(self, batch) -> Series:
+        p = self.pred.evaluate(batch)
+        t = self.truthy.evaluate(batch)
+        f = self.falsy.evaluate(batch)
+        return p.if_else(t, f)
+
+    def __repr__(self):
+        return f"if({self.pred!r}, {self.truthy!r}, {self.falsy!r})"
+
+
+class Coalesce(ExprNode):
+    def __init__(self, args: List[ExprNode]):
+        self.args = args
+
+    def children(self):
+        return list(self.args)
+
+    def with_children(self, ch):
+        return Coalesce(ch)
+
+    def to_field(self, schema):
+        f0 = self.args[0].to_field(schema)
+        dt = f0.dtype
+        for a in self.args[1:]:
+            dt = supertype(dt, a.to_field(schema).dtype)
+        return Field(f0.name, dt)
+
+    def evaluate(self, batch) -> Series:
+        out = self.args[0].evaluate(batch)
+        for a in self.args[1:]:
+            nxt = a.evaluate(batch)
+            if len(nxt) == 1:
+                nxt = nxt.broadcast(len(out))
+            out = out.fill_null(nxt)
+        return out
+
+    def __repr__(self):
+        return f"coalesce({', '.join(map(repr, self.args))})"
+
+
+class AggKind:
+    SUM = "sum"
+    MEAN = "mean"
+    MIN = "min"
+    MAX = "max"
+    COUNT = "count"           # non-null count
+    COUNT_ALL = "count_all"   # row count
+    COUNT_DISTINCT = "count_distinct"
+    ANY_VALUE = "any_value"
+    LIST = "list"
+    CONCAT = "concat"
+    STDDEV = "stddev"
+    VARIANCE = "variance"
+    SKEW = "skew"
+    APPROX_COUNT_DISTINCT = "approx_count_distinct"
+    APPROX_PERCENTILE = "approx_percentile"
+    BOOL_AND = "bool_and"
+    BOOL_OR = "bool_or"
+
+
+_NUMERIC_AGGS = {AggKind.SUM, AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE,
+                 AggKind.SKEW}
+
+
+class Agg(ExprNode):
+    """Aggregation expression; evaluated by the aggregate operators."""
+
+    def __init__(self, kind: str, child: Optional[ExprNode],
+                 param: Any = None):
+        self.kind = kind
+        self.child = child
+        self.param = param
+
+    def children(self):
+        return [self.child] if self.child is not None else []
+
+    def with_children(self, ch):
+        return Agg(self.kind, ch[0] if ch else None, self.param)
+
+    def out_name(self):
+        if self.child is None:
+            return "count"
+        return self.child.out_name()
+
+    def is_aggregation(self):
+        return True
+
+    def to_field(self, schema: Schema) -> Field:
+        if self.child is None:
+            return Field("count", DataType.uint64())
+        f = self.child.to_field(schema)
+        k = self.kind
+        if k in (AggKind.COUNT, AggKind.COUNT_ALL, AggKind.COUNT_DISTINCT,
+                 AggKind.APPROX_COUNT_DISTINCT):
+            return Field(f.name, DataType.uint64())
+        if k == AggKind.SUM:
+            if f.dtype.is_integer():
+                return Field(f.name, DataType.int64()
+                             if f.dtype.is_signed_integer()
+                             else DataType.uint64())
+            return Field(f.name, DataType.float64()
+                         if not f.dtype.is_floating() or
+                         f.dtype.kind == TypeKind.FLOAT64
+                         else DataType.float32())
+        if k in (AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE, AggKind.SKEW,
+                 AggKind.APPROX_PERCENTILE):
+            return Field(f.name, DataType.float64())
+        if k in (AggKind.MIN, AggKind.MAX, AggKind.ANY_VALUE):
+            return Field(f.name, f.dtype)
+        if k in (AggKind.LIST, AggKind.CONCAT):
+            return Field(f.name, DataType.list(f.dtype))
+        if k in (AggKind.BOOL_AND, AggKind.BOOL_OR):
+            return Field(f.name, DataType.bool())
+        raise ValueError(f"unknown agg kind {k}")
+
+    def evaluate(self, batch) -> Series:
+        raise RuntimeError(
+            "aggregation expressions must run under an aggregate operator")
+
+    def __repr__(self):
+        return f"{self.child!r}.{self.kind}()" if self.child is not None \
+            else "count(*)"
+
+
+class ScalarFn(ExprNode):
+    """Named scalar function with a vectorized Series implementation."""
+
+    def __init__(self, name: str, fn: Callable[..., Series],
+                 args: List[ExprNode], ret_dtype,
+                 literal_args: Tuple = (), kwargs: Optional[dict] = None):
+        self.name = name
+        self.fn = fn
+        self.args = args
+        self.ret_dtype = ret_dtype  # DataType or callable(list[Field])->DataType
+        self.literal_args = literal_args
+        self.kwargs = kwargs or {}
+
+    def children(self):
+        return list(self.args)
+
+    def with_children(self, ch):
+        return ScalarFn(self.name, self.fn, ch, self.ret_dtype,
+                        self.literal_args, self.kwargs)
+
+    def to_field(self, schema):
+        fields = [a.to_field(schema) for a in self.args]
+        name = fields[0].name if fields else self.name
+        dt = self.ret_dtype(fields) if callable(self.ret_dtype) \
+            else self.ret_dtype
+        return Field(name, dt)
+
+    def evaluate(self, batch) -> Series:
+        series = [a.evaluate(batch) for a in self.args]
+        return self.fn(*series, *self.literal_args, **self.kwargs)
+
+    def __repr__(self):
+        inner = ", ".join(map(repr, self.args))
+        extra = "".join(f", {a!r}" for a in self.literal_args)
+        return f"{self.name}({inner}{extra})"
+
+
+class PyUDF(ExprNode):
+    """Row/batch-wise python UDF (ref: daft-dsl/src/python_udf/)."""
+
+    def __init__(self, name: str, fn: Callable, args: List[ExprNode],
+                 return_dtype: DataType, batched: bool = False,
+                 max_retries: int = 0, on_error: str = "raise",
+                 use_process: bool = False, concurrency: Optional[int] = None,
+                 gpus: int = 0):
+        self.name = name
+        self.fn = fn
+        self.args = args
+        self.return_dtype = return_dtype
+        self.batched = batched
+        self.max_retries = max_retries
+        self.on_error = on_error
+        self.use_process = use_process
+        self.concurrency = concurrency
+        self.gpus = gpus
+
+    def children(self):
+        return list(self.args)
+
+    def with_children(self, ch):
+        return PyUDF(self.name, self.fn, ch, self.return_dtype, self.batched,
+                     self.max_retries, self.on_error, self.use_process,
+                     self.concurrency, self.gpus)
+
+    def out_name(self):
+        return self.name
+
+    def to_field(self, schema):
+        return Field(self.name, self.return_dtype)
+
+    def evaluate(self, batch) -> Series:
+        from ..udf import run_udf_node
+        return run_udf_node(self, batch)
+
+    def __repr__(self):
+        return f"udf:{self.name}({', '.join(map(repr, self.args))})"
+
+
+# ---------------------------------------------------------------------------
+# user-facing Expression wrapper
+# ---------------------------------------------------------------------------
+
+def _to_node(v) -> ExprNode:
+    if isinstance(v, Expression):
+        return v._node
+    if isinstance(v, ExprNode):
+        return v
+    return Literal(v)
+
+
+class Expression:
+    __slots__ = ("_node",)
+
+    def __init__(self, node: ExprNode):
+        self._node = node
+
+    # naming / casting
+    def alias(self, name: str) -> "Expression":
+        return Expression(Alias(self._node, name))
+
+    def cast(self, dtype: DataType) -> "Expression":
+        return Expression(Cast(self._node, dtype))
+
+    def name(self) -> str:
+        return self._node.out_name()
+
+    # arithmetic
+    def __add__(self, o): return Expression(BinaryOp("add", self._node, _to_node(o)))
+    def __radd__(self, o): return Expression(BinaryOp("add", _to_node(o), self._node))
+    def __sub__(self, o): return Expression(BinaryOp("sub", self._node, _to_node(o)))
+    def __rsub__(self, o): return Expression(BinaryOp("sub", _to_node(o), self._node))
+    def __mul__(self, o): return Expression(BinaryOp("mul", self._node, _to_node(o)))
+    def __rmul__(self, o): return Expression(BinaryOp("mul", _to_node(o), self._node))
+    def __truediv__(self, o): return Expression(BinaryOp("div", self._node, _to_node(o)))
+    def __rtruediv__(self, o): return Expression(BinaryOp("div", _to_node(o), self._node))
+    def __floordiv__(self, o): return Expression(BinaryOp("floordiv", self._node, _to_node(o)))
+    def __mod__(self, o): return Expression(BinaryOp("mod", self._node, _to_node(o)))
+    def __pow__(self, o): return Expression(BinaryOp("pow", self._node, _to_node(o)))
+    def __neg__(self): return Expression(BinaryOp("sub", Literal(0), self._node))
+
+    # comparison
+    def __eq__(self, o): return Expression(BinaryOp("eq", self._node, _to_node(o)))  # type: ignore
+    def __ne__(self, o): return Expression(BinaryOp("ne", self._node, _to_node(o)))  # type: ignore
+    def __lt__(self, o): return Expression(BinaryOp("lt", self._node, _to_node(o)))
+    def __le__(self, o): return Expression(BinaryOp("le", self._node, _to_node(o)))
+    def __gt__(self, o): return Expression(BinaryOp("gt", self._node, _to_node(o)))
+    def __ge__(self, o): return Expression(BinaryOp("ge", self._node, _to_node(o)))
+
+    def eq(self, o): return self.__eq__(o)
+    def ne(self, o): return self.__ne__(o)
+
+    def __hash__(self):
+        return hash(repr(self._node))
+
+    # logic
+    def __and__(self, o): return Expression(BinaryOp("and", self._node, _to_node(o)))
+    def __rand__(self, o): return Expression(BinaryOp("and", _to_node(o), self._node))
+    def __or__(self, o): return Expression(BinaryOp("or", self._node, _to_node(o)))
+    def __ror__(self, o): return Expression(BinaryOp("or", _to_node(o), self._node))
+    def __xor__(self, o): return Expression(BinaryOp("xor", self._node, _to_node(o)))
+    def __invert__(self): return Expression(Not(self._node))
+
+    # null handling
+    def is_null(self): return Expression(IsNull(self._node))
+    def not_null(self): return Expression(IsNull(self._node, negate=True))
+    def fill_null(self, fill): return Expression(FillNull(self._node, _to_node(fill)))
+
+    def is_in(self, values) -> "Expression":
+        if isinstance(values, Expression):
+            raise TypeError("is_in expects a python list of literals")
+        return Expression(IsIn(self._node, list(values)))
+
+    def between(self, lo, hi) -> "Expression":
+        return Expression(Between(self._node, _to_node(lo), _to_node(hi)))
+
+    def if_else(self, truthy, falsy) -> "Expression":
+        return Expression(IfElse(self._node, _to_node(truthy), _to_node(falsy)))
+
+    def apply(self, fn: Callable, return_dtype: DataType) -> "Expression":
+        return Expression(PyUDF(getattr(fn, "__name__", "apply"), fn,
+                                [self._node], return_dtype))
+
+    # aggregations
+    def sum(self): return Expression(Agg(AggKind.SUM, self._node))
+    def mean(self): return Expression(Agg(AggKind.MEAN, self._node))
+    def avg(self): return self.mean()
+    def min(self): return Expression(Agg(AggKind.MIN, self._node))
+    def max(self): return Expression(Agg(AggKind.MAX, self._node))
+    def count(self, mode: str = "valid"):
+        kind = AggKind.COUNT_ALL if mode == "all" else AggKind.COUNT
+        return Expression(Agg(kind, self._node))
+    def count_distinct(self): return Expression(Agg(AggKind.COUNT_DISTINCT, self._node))
+    def approx_count_distinct(self):
+        return Expression(Agg(AggKind.APPROX_COUNT_DISTINCT, self._node))
+    def approx_percentile(self, q: float):
+        return Expression(Agg(AggKind.APPROX_PERCENTILE, self._node, q))
+    def any_value(self): return Expression(Agg(AggKind.ANY_VALUE, self._node))
+    def agg_list(self): return Expression(Agg(AggKind.LIST, self._node))
+    def agg_concat(self): return Expression(Agg(AggKind.CONCAT, self._node))
+    def stddev(self): return Expression(Agg(AggKind.STDDEV, self._node))
+    def variance(self): return Expression(Agg(AggKind.VARIANCE, self._node))
+    def skew(self): return Expression(Agg(AggKind.SKEW, self._node))
+    def bool_and(self): return Expression(Agg(AggKind.BOOL_AND, self._node))
+    def bool_or(self): return Expression(Agg(AggKind.BOOL_OR, self._node))
+
+    # window
+    def over(self, window) -> "Expression":
+        from ..physical.window import WindowFn
+        node = self._node
+        base = node.child if isinstance(node, Alias) else node
+        name = node.out_name()
+        if isinstance(base, WindowFn):
+            wf = WindowFn(base.kind, base.inner, window, base.offset,
+                          base.default)
+        elif isinstance(base, Agg):
+            wf = WindowFn("agg", base, window)
+        else:
+            raise TypeError(
+                "over() expects an aggregation expression (or use "
+                "daft_amd.functions.row_number/rank/lag/lead)")
+        return Expression(Alias(wf, name) if isinstance(node, Alias) else wf)
+
+    def lag(self, offset: int = 1, default=None) -> "Expression":
+        """Use together with .over(window)."""
+        from ..physical.window import WindowFn
+        return Expression(WindowFn("lag", self._node, None, offset, default))
+
+    def lead(self, offset: int = 1, default=None) -> "Expression":
+        from ..physical.window import WindowFn
+        return Expression(WindowFn("lead", self._node, None, offset, default))
+
+    # namespaces
+    @property
+    def str(self) -> "StringNamespace":
+        return StringNamespace(self._node)
+
+    @property
+    def dt(self) -> "TemporalNamespace":
+        return TemporalNamespace(self._node)
+
+    @property
+    def list(self) -> "ListNamespace":
+        return ListNamespace(self._node)
+
+    @property
+    def struct(self) -> "StructNamespace":
+        return StructNamespace(self._node)
+
+    @property
+    def float(self) -> "FloatNamespace":
+        return FloatNamespace(self._node)
+
+    @property
+    def embedding(self) -> "EmbeddingNamespace":
+        return EmbeddingNamespace(self._node)
+
+    @property
+    def image(self) -> "ImageNamespace":
+        from ..functions.image import ImageNamespace
+        return ImageNamespace(self._node)
+
+    def hash(self, seed: int = 0) -> "Expression":
+        from .. import kernels
+
+        def _h(s: Series, sd) -> Series:
+            h = kernels.hash_columns([s], sd)
+            return Series(s.name, DataType.uint64(), data=h.view(torch.uint64))
+        return Expression(ScalarFn("hash", _h, [self._node],
+                                   DataType.uint64(), (seed,)))
+
+    def minhash(self, num_hashes: int, ngram_size: int = 1,
+                seed: int = 1) -> "Expression":
+        from ..functions.minhash import minhash_series
+        return Expression(ScalarFn(
+            "minhash", minhash_series, [self._node],
+            DataType.fixed_size_list(DataType.uint32(), num_hashes),
+            (num_hashes, ngram_size, seed)))
+
+    # misc
+    def abs(self):
+        return Expression(ScalarFn(
+            "abs", _series_abs, [self._node],
+            lambda f: f[0].dtype))
+
+    def round(self, decimals: int = 0):
+        return Expression(ScalarFn(
+            "round", _series_round, [self._node],
+            lambda f: f[0].dtype, (decimals,)))
+
+    def floor(self):
+        return Expression(ScalarFn("floor", _series_floor, [self._node],
+                                   lambda f: f[0].dtype))
+
+    def ceil(self):
+        return Expression(ScalarFn("ceil", _series_ceil, [self._node],
+                                   lambda f: f[0].dtype))
+
+    def sqrt(self):
+        return Expression(ScalarFn("sqrt", _series_sqrt, [self._node],
+                                   DataType.float64()))
+
+    def exp(self):
+        return Expression(ScalarFn("exp", _series_exp, [self._node],
+                                   DataType.float64()))
+
+    def log(self, base: Optional[float] = None):
+        return Expression(ScalarFn("log", _series_log, [self._node],
+                                   DataType.float64(), (base,)))
+
+    def clip(self, lo=None, hi=None):
+        return Expression(ScalarFn("clip", _series_clip, [self._node],
+                                   lambda f: f[0].dtype, (lo, hi)))
+
+    def __repr__(self):
+        return repr(self._node)
+
+
+def _series_unary(fn):
+    def impl(s: Series, *args) -> Series:
+        out = fn(s.data, *args)
+        return Series(s.name, s.dtype if not out.dtype.is_floating_point or
+                      s.dtype.is_floating() else DataType.float64(),
+                      data=out, validity=s.validity)
+    return impl
+
+
+def _series_abs(s: Series) -> Series:
+    return Series(s.name, s.dtype, data=torch.abs(s.data),
+                  validity=s.validity)
+
+
+def _series_round(s: Series, decimals: int) -> Series:
+    if s.dtype.is_integer():
+        return s
+    return Series(s.name, s.dtype, data=torch.round(s.data, decimals=decimals),
+                  validity=s.validity)
+
+
+def _series_floor(s: Series) -> Series:
+    if s.dtype.is_integer():
+        return s
+    return Series(s.name, s.dtype, data=torch.floor(s.data),
+                  validity=s.validity)
+
+
+def _series_ceil(s: Series) -> Series:
+    if s.dtype.is_integer():
+        return s
+    return Series(s.name, s.dtype, data=torch.ceil(s.data),
+                  validity=s.validity)
+
+
+def _series_sqrt(s: Series) -> Series:
+    return Series(s.name, DataType.float64(),
+                  data=torch.sqrt(s.data.to(torch.float64)),
+                  validity=s.validity)
+
+
+def _series_exp(s: Series) -> Series:
+    return Series(s.name, DataType.float64(),
+                  data=torch.exp(s.data.to(torch.float64)),
+                  validity=s.validity)
+
+
+def _series_log(s: Series, base) -> Series:
+    d = torch.log(s.data.to(torch.float64))
+    if base is not None:
+        d = d / torch.log(torch.tensor(float(base)))
+    return Series(s.name, DataType.float64(), data=d, validity=s.validity)
+
+
+def _series_clip(s: Series, lo, hi) -> Series:
+    return Series(s.name, s.dtype, data=torch.clamp(s.data, lo, hi),
+                  validity=s.validity)
+
+
+# ---------------------------------------------------------------------------
+# namespaces
+# ---------------------------------------------------------------------------
+
+class _Namespace:
+    def __init__(self, node: ExprNode):
+        self._node = node
+
+    def _fn(self, name, fn, ret, *literal_args, **kwargs) -> Expression:
+        return Expression(ScalarFn(name, fn, [self._node], ret,
+                                   tuple(literal_args), kwargs))
+
+
+class StringNamespace(_Namespace):
+    def contains(self, pat: str):
+        from ..kernels import strings as k
+        return self._fn("contains", k.contains, DataType.bool(), pat)
+
+    def startswith(self, pat: str):
+        from ..kernels import strings as k
+        return self._fn("startswith", k.startswith, DataType.bool(), pat)
+
+    def endswith(self, pat: str):
+        from ..kernels import strings as k
+        return self._fn("endswith", k.endswith, DataType.bool(), pat)
+
+    def like(self, pattern: str):
+        from ..kernels import strings as k
+        return self._fn("like", k.like, DataType.bool(), pattern)
+
+    def ilike(self, pattern: str):
+        from ..kernels import strings as k
+        return self._fn("ilike", k.like, DataType.bool(), pattern, True)
+
+    def match(self, pattern: str):
+        from ..kernels import strings as k
+        return self._fn("regexp_match", k.regexp_match, DataType.bool(), pattern)
+
+    def length(self):
+        from ..kernels import strings as k
+        return self._fn("length", k.length, DataType.uint64())
+
+    def length_bytes(self):
+        from ..kernels import strings as k
+        return self._fn("length_bytes", k.length_bytes, DataType.uint64())
+
+    def lower(self):
+        from ..kernels import strings as k
+        return self._fn("lower", k.lower, DataType.string())
+
+    def upper(self):
+        from ..kernels import strings as k
+        return self._fn("upper", k.upper, DataType.string())
+
+    def lstrip(self):
+        from ..kernels import strings as k
+        return self._fn("lstrip", k.lstrip, DataType.string())
+
+    def rstrip(self):
+        from ..kernels import strings as k
+        return self._fn("rstrip", k.rstrip, DataType.string())
+
+    def strip(self):
+        from ..kernels import strings as k
+        return self._fn("strip", k.strip, DataType.string())
+
+    def reverse(self):
+        from ..kernels import strings as k
+        return self._fn("reverse", k.reverse, DataType.string())
+
+    def capitalize(self):
+        from ..kernels import strings as k
+        return self._fn("capitalize", k.capitalize, DataType.string())
+
+    def substr(self, start: int, length: Optional[int] = None):
+        from ..kernels import strings as k
+        return self._fn("substr", k.substr, DataType.string(), start, length)
+
+    def left(self, n: int):
+        from ..kernels import strings as k
+        return self._fn("left", k.left, DataType.string(), n)
+
+    def right(self, n: int):
+        from ..kernels import strings as k
+        return self._fn("right", k.right, DataType.string(), n)
+
+    def find(self, pat: str):
+        from ..kernels import strings as k
+        return self._fn("find", k.find, DataType.int64(), pat)
+
+    def split(self, sep: str):
+        from ..kernels import strings as k
+        return self._fn("split", k.split, DataType.list(DataType.string()), sep)
+
+    def concat(self, other):
+        from ..kernels import strings as k
+        return Expression(ScalarFn(
+            "concat", lambda a, b: k.concat_str([a, b]),
+            [self._node, _to_node(other)], DataType.string()))
+
+    def repeat(self, n: int):
+        from ..kernels import strings as k
+        return self._fn("repeat", k.repeat, DataType.string(), n)
+
+    def lpad(self, width: int, fillchar: str = " "):
+        from ..kernels import strings as k
+        return self._fn("lpad", k.lpad, DataType.string(), width, fillchar)
+
+    def rpad(self, width: int, fillchar: str = " "):
+        from ..kernels import strings as k
+        return self._fn("rpad", k.rpad, DataType.string(), width, fillchar)
+
+    def to_date(self, fmt: str = "%Y-%m-%d"):
+        return self._fn("to_date", _str_to_date, DataType.date(), fmt)
+
+    def tokenize_encode(self, tokenizer: str = "simple"):
+        from ..functions.tokenize import tokenize_encode_series
+        return self._fn("tokenize_encode", tokenize_encode_series,
+                        DataType.list(DataType.int32()), tokenizer)
+
+
+def _str_to_date(s: Series, fmt: str) -> Series:
+    vals = s.cpu().to_pylist()
+    out = [None if v is None else _dt.datetime.strptime(v, fmt).date()
+           for v in vals]
+    res = Series.from_pylist(s.name, out, DataType.date())
+    return res.to(s.device) if s.is_gpu() else res
+
+
+class TemporalNamespace(_Namespace):
+    def _civil(self, part: str) -> Expression:
+        return self._fn(part, _dt_extract, DataType.int32(), part)
+
+    def year(self): return self._civil("year")
+    def month(self): return self._civil("month")
+    def day(self): return self._civil("day")
+    def quarter(self): return self._civil("quarter")
+    def day_of_week(self): return self._civil("day_of_week")
+    def day_of_year(self): return self._civil("day_of_year")
+    def week_of_year(self): return self._civil("week_of_year")
+    def hour(self): return self._civil("hour")
+    def minute(self): return self._civil("minute")
+    def second(self): return self._civil("second")
+
+    def date(self):
+        return self._fn("date", _dt_to_date, DataType.date())
+
+    def truncate(self, interval: str):
+        return self._fn("truncate", _dt_truncate, lambda f: f[0].dtype,
+                        interval)
+
+    def total_seconds(self):
+        return self._fn("total_seconds", _dur_total, DataType.int64(),
+                        1_000_000)
+
+    def total_days(self):
+        return self._fn("total_days", _dur_total, DataType.int64(),
+                        86_400_000_000)
+
+
+def _days_to_civil(days: torch.Tensor):
+    """Howard Hinnant's civil_from_days, vectorized on torch int ops."""
+    z = days.to(torch.int64) + 719468
+    era = torch.div(torch.where(z >= 0, z, z - 146096), 146097,
+                    rounding_mode="floor")
+    doe = z - era * 146097
+    yoe = torch.div(doe - torch.div(doe, 1460, rounding_mode="floor")
+                    + torch.div(doe, 36524, rounding_mode="floor")
+                    - torch.div(doe, 146096, rounding_mode="floor"),
+                    365, rounding_mode="floor")
+    y = yoe + era * 400
+    doy = doe - (365 * yoe + torch.div(yoe, 4, rounding_mode="floor")
+                 - torch.div(yoe, 100, rounding_mode="floor"))
+    mp = torch.div(5 * doy + 2, 153, rounding_mode="floor")
+    d = doy - torch.div(153 * mp + 2, 5, rounding_mode="floor") + 1
+    m = mp + torch.where(mp < 10, torch.full_like(mp, 3),
+                         torch.full_like(mp, -9))
+    y = y + (m <= 2).to(torch.int64)
+    return y, m, d, doy
+
+
+def _ts_to_days_and_us(s: Series):
+    if s.dtype.kind == TypeKind.DATE:
+        return s.data.to(torch.int64), None
+    unit = s.dtype.timeunit
+    mult = {"s": 1_000_000, "ms": 1_000, "us": 1, "ns": 1}[unit]
+    us = s.data if unit != "ns" else torch.div(s.data, 1000,
+                                               rounding_mode="floor")
+    us = us * mult if mult != 1 else us
+    days = torch.div(us, 86_400_000_000, rounding_mode="floor")
+    tod = us - days * 86_400_000_000
+    return days, tod
+
+
+def _dt_extract(s: Series, part: str) -> Series:
+    days, tod = _ts_to_days_and_us(s)
+    if part in ("hour", "minute", "second"):
+        assert tod is not None, f"{part} requires a timestamp"
+        sec = torch.div(tod, 1_000_000, rounding_mode="floor")
+        if part == "hour":
+            out = torch.div(sec, 3600, rounding_mode="floor")
+        elif part == "minute":
+            out = torch.div(sec, 60, rounding_mode="floor") % 60
+        else:
+            out = sec % 60
+        return Series(s.name, DataType.int32(), data=out.to(torch.int32),
+                      validity=s.validity)
+    y, m, d, _doy = _days_to_civil(days)
+    if part == "year":
+        out = y
+    elif part == "month":
+        out = m
+    elif part == "day":
+        out = d
+    elif part == "quarter":
+        out = torch.div(m - 1, 3, rounding_mode="floor") + 1
+    elif part == "day_of_week":
+        out = (days + 3) % 7  # 1970-01-01 was a Thursday; 0 = Monday
+    elif part == "day_of_year":
+        jan1 = _civil_to_days(y, torch.ones_like(m), torch.ones_like(d))
+        out = days - jan1 + 1
+    elif part == "week_of_year":
+        jan1 = _civil_to_days(y, torch.ones_like(m), torch.ones_like(d))
+        out = torch.div(days - jan1, 7, rounding_mode="floor") + 1
+    else:
+        raise ValueError(part)
+    return Series(s.name, DataType.int32(), data=out.to(torch.int32),
+                  validity=s.validity)
+
+
+def _civil_to_days(y: torch.Tensor, m: torch.Tensor,
+                   d: torch.Tensor) -> torch.Tensor:
+    """days_from_civil, vectorized."""
+    y = y - (m <= 2).to(torch.int64)
+    era = torch.div(torch.where(y >= 0, y, y - 399), 400,
+                    rounding_mode="floor")
+    yoe = y - era * 400
+    mp = torch.where(m > 2, m - 3, m + 9)
+    doy = torch.div(153 * mp + 2, 5, rounding_mode="floor") + d - 1
+    doe = yoe * 365 + torch.div(yoe, 4, rounding_mode="floor") \
+        - torch.div(yoe, 100, rounding_mode="floor") + doy
+    return era * 146097 + doe - 719468
+
+
+def _dt_to_date(s: Series) -> Series:
+    days, _ = _ts_to_days_and_us(s)
+    return Series(s.name, DataType.date(), data=days.to(torch.int32),
+                  validity=s.validity)
+
+
+def _dt_truncate(s: Series, interval: str) -> Series:
+    days, tod = _ts_to_days_and_us(s)
+    y, m, d, _ = _days_to_civil(days)
+    if interval in ("year", "1 year"):
+        new_days = _civil_to_days(y, torch.ones_like(m), torch.ones_like(d))
+    elif interval in ("month", "1 month"):
+        new_days = _civil_to_days(y, m, torch.ones_like(d))
+    elif interval in ("week", "1 week"):
+        new_days = days - ((days + 3) % 7)
+    elif interval in ("day", "1 day"):
+        new_days = days
+    else:
+        raise ValueError(f"unsupported truncate interval {interval}")
+    if s.dtype.kind == TypeKind.DATE:
+        return Series(s.name, s.dtype, data=new_days.to(torch.int32),
+                      validity=s.validity)
+    unit = s.dtype.timeunit
+    mult = {"s": 1, "ms": 10**3, "us": 10**6, "ns": 10**9}[unit]
+    out = new_days * 86_400 * mult
+    return Series(s.name, s.dtype, data=out, validity=s.validity)
+
+
+def _dur_total(s: Series, div_us: int) -> Series:
+    assert s.dtype.kind == TypeKind.DURATION
+    mult = {"s": 1_000_000, "ms": 1_000, "us": 1, "ns": 1}[s.dtype.timeunit]
+    us = s.data * mult if s.dtype.timeunit != "ns" else \
+        torch.div(s.data, 1000, rounding_mode="floor")
+    out = torch.div(us, div_us, rounding_mode="floor")
+    return Series(s.name, DataType.int64(), data=out, validity=s.validity)
+
+
+class ListNamespace(_Namespace):
+    def length(self):
+        return self._fn("list_length", _list_length, DataType.uint64())
+
+    def get(self, idx: int, default=None):
+        return self._fn("list_get", _list_get,
+                        lambda f: f[0].dtype.inner, idx, default)
+
+    def sum(self):
+        return self._fn("list_sum", _list_agg,
+                        lambda f: f[0].dtype.inner, "sum")
+
+    def mean(self):
+        return self._fn("list_mean", _list_agg, DataType.float64(), "mean")
+
+    def min(self):
+        return self._fn("list_min", _list_agg,
+                        lambda f: f[0].dtype.inner, "min")
+
+    def max(self):
+        return self._fn("list_max", _list_agg,
+                        lambda f: f[0].dtype.inner, "max")
+
+    def join(self, sep: str):
+        return self._fn("list_join", _list_join, DataType.string(), sep)
+
+
+def _list_length(s: Series) -> Series:
+    if s.dtype.kind == TypeKind.LIST:
+        lens = s.offsets[1:] - s.offsets[:-1]
+    else:
+        lens = torch.full((len(s),), s.dtype.size, dtype=torch.int64,
+                          device=s.device)
+    return Series(s.name, DataType.uint64(), data=lens.view(torch.uint64),
+                  validity=s.validity)
+
+
+def _list_get(s: Series, idx: int, default) -> Series:
+    if s.dtype.kind == TypeKind.LIST:
+        lens = s.offsets[1:] - s.offsets[:-1]
+        child_idx = torch.where(
+            lens > idx, s.offsets[:-1] + idx,
+            torch.full_like(lens, -1))
+    else:
+        n = s.dtype.size
+        base = torch.arange(len(s), device=s.device, dtype=torch.int64) * n
+        child_idx = base + idx if idx < n else torch.full_like(base, -1)
+    out = s.children[0].take(child_idx)
+    return out.rename(s.name)
+
+
+def _list_agg(s: Series, op: str) -> Series:
+    from . import expressions as _  # noqa
+    from ..kernels import rowops
+    if s.dtype.kind == TypeKind.LIST:
+        n = len(s)
+        # group child elements by parent row
+        lens = (s.offsets[1:] - s.offsets[:-1])
+        gid = torch.repeat_interleave(
+            torch.arange(n, device=s.device, dtype=torch.int64), lens)
+        kind = "sum" if op in ("sum", "mean") else op
+        child = s.children[0]
+        data, cnt = rowops.grouped_agg(gid, n, child, kind)
+        if op == "mean":
+            out = data.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
+            validity = cnt > 0
+            return Series(s.name, DataType.float64(), data=out,
+                          validity=validity)
+        validity = cnt > 0
+        dt = s.dtype.inner if op != "sum" else (
+            DataType.int64() if s.dtype.inner.is_integer()
+            else DataType.float64())
+        return Series(s.name, dt, data=data.to(dt.to_torch()),
+                      validity=validity)
+    # fixed size list: tensor reshape reduction
+    n = len(s)
+    sz = s.dtype.size
+    mat = s.children[0].data.reshape(n, sz)
+    if op == "sum":
+        out = mat.sum(dim=1)
+    elif op == "mean":
+        out = mat.to(torch.float64).mean(dim=1)
+    elif op == "min":
+        out = mat.min(dim=1).values
+    else:
+        out = mat.max(dim=1).values
+    from ..schema import from_torch_dtype
+    return Series(s.name, from_torch_dtype(out.dtype), data=out,
+                  validity=s.validity)
+
+
+def _list_join(s: Series, sep: str) -> Series:
+    vals = s.cpu().to_pylist()
+    out = [None if v is None else sep.join("" if x is None else str(x)
+                                           for x in v) for v in vals]
+    res = Series.from_pylist(s.name, out, DataType.string())
+    return res.to(s.device) if s.is_gpu() else res
+
+
+class StructNamespace(_Namespace):
+    def get(self, field_name: str):
+        return self._fn("struct_get", _struct_get,
+                        lambda f: next(x.dtype for x in f[0].dtype.fields
+                                       if x.name == field_name), field_name)
+
+
+def _struct_get(s: Series, field_name: str) -> Series:
+    for i, f in enumerate(s.dtype.fields):
+        if f.name == field_name:
+            child = s.children[i]
+            if s.validity is not None:
+                v = child.validity & s.validity if child.validity is not None \
+                    else s.validity.clone()
+                child = child.with_validity(v)
+            return child.rename(field_name)
+    raise KeyError(field_name)
+
+
+class FloatNamespace(_Namespace):
+    def is_nan(self):
+        return self._fn("is_nan", _float_is_nan, DataType.bool())
+
+    def is_inf(self):
+        return self._fn("is_inf", _float_is_inf, DataType.bool())
+
+    def fill_nan(self, value: float):
+        return self._fn("fill_nan", _float_fill_nan,
+                        lambda f: f[0].dtype, value)
+
+
+def _float_is_nan(s: Series) -> Series:
+    return Series(s.name, DataType.bool(), data=torch.isnan(s.data),
+                  validity=s.validity)
+
+
+def _float_is_inf(s: Series) -> Series:
+    return Series(s.name, DataType.bool(), data=torch.isinf(s.data),
+                  validity=s.validity)
+
+
+def _float_fill_nan(s: Series, value: float) -> Series:
+    return Series(s.name, s.dtype, data=torch.nan_to_num(s.data, nan=value),
+                  validity=s.validity)
+
+
+class EmbeddingNamespace(_Namespace):
+    def cosine_distance(self, other) -> Expression:
+        from ..functions.distance import cosine_distance_series
+        return Expression(ScalarFn(
+            "cosine_distance", cosine_distance_series,
+            [self._node, _to_node(other)], DataType.float64()))
+
+    def dot(self, other) -> Expression:
+        from ..functions.distance import dot_series
+        return Expression(ScalarFn(
+            "dot", dot_series, [self._node, _to_node(other)],
+            DataType.float64()))
+
+    def l2_norm(self) -> Expression:
+        from ..functions.distance import l2_norm_series
+        return Expression(ScalarFn("l2_norm", l2_norm_series, [self._node],
+                                   DataType.float64()))
+
+
+# ---------------------------------------------------------------------------
+# public constructors
+# ---------------------------------------------------------------------------
+
+def col(name: str) -> Expression:
+    return Expression(ColumnRef(name))
+
+
+def lit(value: Any, dtype: Optional[DataType] = None) -> Expression:
+    return Expression(Literal(value, dtype))
+
+
+def element() -> Expression:
+    """Placeholder for list.map-style element references."""
+    return Expression(ColumnRef("__element__"))
+
+
+def interval(days: int = 0, months: int = 0, years: int = 0,
+             hours: int = 0, minutes: int = 0, seconds: int = 0) -> Expression:
+    """Date interval literal — round 1 supports day-granularity offsets
+    (month/year arithmetic handled at the binary-op layer for dates)."""
+    total_days = days + 30 * months + 365 * years  # calendar-approx; see dt.truncate for exact
+    if months or years:
+        # exact calendar intervals handled via _dt_add_interval at eval
+        return Expression(Literal({"days": days, "months": months,
+                                   "years": years}, DataType.python()))
+    us = ((hours * 60 + minutes) * 60 + seconds) * 1_000_000
+    if us:
+        return Expression(Literal(total_days * 86_400_000_000 + us,
+                                  DataType.duration("us")))
+    return Expression(Literal(total_days, DataType.int32()))
+
+
+def list_(*exprs) -> Expression:
+    nodes = [_to_node(e) for e in exprs]
+
+    def make(*series: Series) -> Series:
+        n = max(len(s) for s in series)
+        series = [s.broadcast(n) if len(s) == 1 else s for s in series]
+        from ..schema import supertype as st
+        dt = series[0].dtype
+        for s in series[1:]:
+            dt = st(dt, s.dtype)
+        series = [s.cast(dt) for s in series]
+        k = len(series)
+        # interleave rows: out child = row-major [n, k]
+        idx = torch.arange(n * k, device=series[0].device)
+        from ..series import Series as S
+        from .. import kernels
+        stacked = kernels.concat(series)
+        # stacked is column-major (all of s0, then s1...) -> gather interleave
+        gather = (idx % k) * n + torch.div(idx, k, rounding_mode="floor")
+        child = stacked.take(gather)
+        offs = torch.arange(0, (n + 1) * k, k, dtype=torch.int64,
+                            device=series[0].device)
+        return S(series[0].name, DataType.list(dt), offsets=offs,
+                 children=[child.rename("item")])
+
+    return Expression(ScalarFn(
+        "list", make, nodes,
+        lambda f: DataType.list(f[0].dtype)))
+
+
+def struct(*exprs) -> Expression:
+    nodes = [_to_node(e) for e in exprs]
+
+    def make(*series: Series) -> Series:
+        n = max(len(s) for s in series)
+        series = [s.broadcast(n) if len(s) == 1 else s for s in series]
+        from ..series import Series as S
+        dt = DataType.struct({s.name: s.dtype for s in series})
+        return S("struct", dt, children=list(series), length=n)
+
+    def ret(fields):
+        return DataType.struct({f.name: f.dtype for f in fields})
+
+    return Expression(ScalarFn("struct", make, nodes, ret))
+
+
+def resolve_exprs(exprs: Sequence[Union[Expression, str]]) -> List[ExprNode]:
+    out = []
+    for e in exprs:
+        if isinstance(e, str):
+            out.append(ColumnRef(e))
+        elif isinstance(e, Expression):
+            out.append(e._node)
+        elif isinstance(e, ExprNode):
+            out.append(e)
+        else:
+            out.append(Literal(e))
+    return out

@@ -1,0 +1,115 @@
+"""IO layer: in-memory ingestion + file readers/writers (ref capability:
+/root/reference/src/daft-io, daft-parquet, daft-csv, daft-json, daft-scan,
+daft-writers and daft/io/*.py)."""
+from __future__ import annotations
+
+import glob as _glob
+import os
+from typing import Dict, List, Optional, Union
+
+from ..context import get_context
+from ..dataframe import DataFrame
+from ..logical.builder import LogicalPlanBuilder
+from ..recordbatch import RecordBatch
+from ..schema import DataType, Field, Schema
+
+
+def _default_device(device=None):
+    if device is not None:
+        return device
+    return get_context().default_device_str()
+
+
+def from_pydict(data: Dict[str, list], device=None) -> DataFrame:
+    ctx = get_context()
+    rb = RecordBatch.from_pydict(data, device=_default_device(device))
+    key = ctx.cache.new_key()
+    ctx.cache.put(key, [rb])
+    return DataFrame(LogicalPlanBuilder.from_in_memory(
+        rb.schema, key, len(rb), rb.size_bytes()))
+
+
+def from_recordbatches(batches: List[RecordBatch]) -> DataFrame:
+    ctx = get_context()
+    key = ctx.cache.new_key()
+    ctx.cache.put(key, batches)
+    rows = sum(len(b) for b in batches)
+    size = sum(b.size_bytes() for b in batches)
+    return DataFrame(LogicalPlanBuilder.from_in_memory(
+        batches[0].schema, key, rows, size))
+
+
+def from_arrow(table, device=None) -> DataFrame:
+    import pyarrow as pa
+    if isinstance(table, (pa.RecordBatch,)):
+        table = pa.Table.from_batches([table])
+    rb = RecordBatch.from_arrow(table, device=_default_device(device))
+    return from_recordbatches([rb])
+
+
+def from_pandas(df, device=None) -> DataFrame:
+    import pyarrow as pa
+    return from_arrow(pa.Table.from_pandas(df), device=device)
+
+
+def _expand_paths(path: Union[str, List[str]]) -> List[str]:
+    paths = [path] if isinstance(path, str) else list(path)
+    out: List[str] = []
+    for p in paths:
+        if any(ch in p for ch in "*?["):
+            matches = sorted(_glob.glob(p, recursive=True))
+            out.extend(matches)
+        elif os.path.isdir(p):
+            for root, _dirs, files in os.walk(p):
+                for f in sorted(files):
+                    if not f.startswith((".", "_")):
+                        out.append(os.path.join(root, f))
+        else:
+            out.append(p)
+    if not out:
+        raise FileNotFoundError(f"no files match {path}")
+    return out
+
+
+def read_parquet(path, columns: Optional[List[str]] = None,
+                 io_config=None, **kwargs) -> DataFrame:
+    from . import readers
+    paths = _expand_paths(path)
+    schema = readers.infer_schema(paths[0], "parquet")
+    b = LogicalPlanBuilder.from_scan(schema, paths, "parquet",
+                                     storage_options=io_config)
+    df = DataFrame(b)
+    if columns:
+        df = df.select(*columns)
+    return df
+
+
+def read_csv(path, has_headers: bool = True, delimiter: str = ",",
+             schema: Optional[Dict[str, DataType]] = None,
+             io_config=None, **kwargs) -> DataFrame:
+    from . import readers
+    paths = _expand_paths(path)
+    read_options = {"has_headers": has_headers, "delimiter": delimiter}
+    if schema is not None:
+        sch = Schema.from_dict(schema)
+    else:
+        sch = readers.infer_schema(paths[0], "csv", read_options)
+    return DataFrame(LogicalPlanBuilder.from_scan(
+        sch, paths, "csv", storage_options=io_config,
+        read_options=read_options))
+
+
+def read_json(path, io_config=None, **kwargs) -> DataFrame:
+    from . import readers
+    paths = _expand_paths(path)
+    sch = readers.infer_schema(paths[0], "json")
+    return DataFrame(LogicalPlanBuilder.from_scan(
+        sch, paths, "json", storage_options=io_config))
+
+
+def from_glob_path(path: str) -> DataFrame:
+    """List files matching a glob into a DataFrame (path, size, num_rows)."""
+    paths = _expand_paths(path)
+    sizes = [os.path.getsize(p) if os.path.exists(p) else None for p in paths]
+    return from_pydict({"path": paths, "size": sizes,
+                        "num_rows": [None] * len(paths)})

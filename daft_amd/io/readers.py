@@ -1,0 +1,95 @@
+"""File readers: host decode (pyarrow) -> H2D to HBM.  Ref:
+/root/reference/src/daft-parquet/src/read.rs (read_parquet_bulk :342),
+daft-csv, daft-json.  Row-group streaming + column/limit pushdown here;
+GPU-side page decode is a later-round upgrade."""
+from __future__ import annotations
+
+import json as _json
+from typing import Iterator, List, Optional
+
+from ..recordbatch import RecordBatch
+from ..schema import DataType, Field, Schema
+from .. import arrow_interop
+
+
+def infer_schema(path: str, file_format: str,
+                 read_options: Optional[dict] = None) -> Schema:
+    import pyarrow as pa
+    if file_format == "parquet":
+        import pyarrow.parquet as pq
+        a_schema = pq.read_schema(path)
+    elif file_format == "csv":
+        import pyarrow.csv as pacsv
+        ro = read_options or {}
+        read_opts = pacsv.ReadOptions(
+            autogenerate_column_names=not ro.get("has_headers", True))
+        parse_opts = pacsv.ParseOptions(delimiter=ro.get("delimiter", ","))
+        with pacsv.open_csv(path, read_options=read_opts,
+                            parse_options=parse_opts) as reader:
+            a_schema = reader.schema
+    elif file_format == "json":
+        import pyarrow.json as pajson
+        tbl = pajson.read_json(path)
+        a_schema = tbl.schema
+    else:
+        raise ValueError(f"unknown format {file_format}")
+    fields = [Field(f.name, arrow_interop.dtype_from_arrow(f.type))
+              for f in a_schema]
+    return Schema(fields)
+
+
+def read_file(path: str, file_format: str, columns: Optional[List[str]],
+              predicate, limit: Optional[int], storage_options: dict,
+              read_options: dict, device) -> Iterator[RecordBatch]:
+    if file_format == "parquet":
+        yield from _read_parquet(path, columns, limit, device)
+    elif file_format == "csv":
+        yield from _read_csv(path, columns, read_options, device)
+    elif file_format == "json":
+        yield from _read_json(path, columns, device)
+    else:
+        raise ValueError(f"unknown format {file_format}")
+
+
+def _read_parquet(path, columns, limit, device) -> Iterator[RecordBatch]:
+    import pyarrow.parquet as pq
+    f = pq.ParquetFile(path)
+    remaining = limit
+    for rg in range(f.num_row_groups):
+        if remaining is not None and remaining <= 0:
+            return
+        tbl = f.read_row_group(rg, columns=columns)
+        if remaining is not None and tbl.num_rows > remaining:
+            tbl = tbl.slice(0, remaining)
+        rb = RecordBatch.from_arrow(tbl, device=device)
+        if remaining is not None:
+            remaining -= len(rb)
+        yield rb
+
+
+def _read_csv(path, columns, read_options, device) -> Iterator[RecordBatch]:
+    import pyarrow as pa
+    import pyarrow.csv as pacsv
+    ro = read_options or {}
+    read_opts = pacsv.ReadOptions(
+        autogenerate_column_names=not ro.get("has_headers", True),
+        block_size=64 * 1024 * 1024)
+    parse_opts = pacsv.ParseOptions(delimiter=ro.get("delimiter", ","))
+    convert = pacsv.ConvertOptions(include_columns=columns) if columns \
+        else None
+    with pacsv.open_csv(path, read_options=read_opts,
+                        parse_options=parse_opts,
+                        convert_options=convert) as reader:
+        for chunk in reader:
+            if chunk.num_rows == 0:
+                continue
+            tbl = pa.Table.from_batches([chunk])
+            yield RecordBatch.from_arrow(tbl, device=device)
+
+
+def _read_json(path, columns, device) -> Iterator[RecordBatch]:
+    import pyarrow.json as pajson
+    tbl = pajson.read_json(path)
+    if columns:
+        tbl = tbl.select(columns)
+    yield RecordBatch.from_arrow(tbl, device=device)

@@ -1,0 +1,259 @@
+"""Hash groupby-aggregate execution (ref:
+/root/reference/src/daft-local-execution/src/sinks/grouped_aggregate.rs and
+daft-recordbatch/src/ops/agg.rs).  The hash-table build runs as a HIP kernel
+(csrc/groupby.hip) on GPU; per-group reductions are atomic HIP kernels.
+
+Also implements agg-expression decomposition: an agg output expression is an
+arbitrary tree over Agg nodes (e.g. ``sum(a*b) / sum(c) + 1``); we extract the
+distinct Agg nodes, compute them per group, then evaluate the residual
+expression over the per-group columns."""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from ..expressions.expressions import (Agg, AggKind, Alias, ColumnRef,
+                                       ExprNode)
+from ..kernels import rowops
+from ..recordbatch import RecordBatch
+from ..schema import DataType, Field, Schema, TypeKind
+from ..series import Series, full_null
+
+
+def decompose_agg_exprs(exprs: Sequence[ExprNode]
+                        ) -> Tuple[List[Tuple[str, Agg]], List[ExprNode]]:
+    """Extract distinct Agg nodes; return (named agg list, residual exprs).
+
+    Residual exprs reference the aggs by generated column names."""
+    agg_map: Dict[str, Tuple[str, Agg]] = {}
+    residuals: List[ExprNode] = []
+
+    def rewrite(e: ExprNode) -> ExprNode:
+        if isinstance(e, Agg):
+            key = repr(e)
+            if key not in agg_map:
+                agg_map[key] = (f"__agg_{len(agg_map)}", e)
+            return ColumnRef(agg_map[key][0])
+        ch = e.children()
+        if not ch:
+            return e
+        return e.with_children([rewrite(c) for c in ch])
+
+    for e in exprs:
+        out_name = e.out_name()
+        r = rewrite(e)
+        if isinstance(r, ColumnRef) and r.name.startswith("__agg_"):
+            r = Alias(r, out_name)
+        residuals.append(r)
+    return list(agg_map.values()), residuals
+
+
+def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
+                num_groups: int, name: str, agg: Agg) -> Series:
+    """Compute one aggregation over groups; group_ids None = single group."""
+    n = len(batch)
+    dev = batch.device
+    if group_ids is None:
+        group_ids = torch.zeros(n, dtype=torch.int64, device=dev)
+    kind = agg.kind
+
+    if kind == AggKind.COUNT_ALL or (kind == AggKind.COUNT and
+                                     agg.child is None):
+        data, _ = rowops.grouped_agg(group_ids, num_groups,
+                                     _ones_series(n, dev), "count")
+        return Series(name, DataType.uint64(), data=data.view(torch.uint64))
+
+    values = agg.child.evaluate(batch)
+    if len(values) == 1 and n > 1:
+        values = values.broadcast(n)
+
+    if kind == AggKind.COUNT:
+        data, _ = rowops.grouped_agg(group_ids, num_groups, values,
+                                     "count_valid")
+        return Series(name, DataType.uint64(), data=data.view(torch.uint64))
+
+    if kind in (AggKind.SUM, AggKind.MIN, AggKind.MAX):
+        if values.dtype.kind in (TypeKind.STRING, TypeKind.BINARY) and \
+                kind in (AggKind.MIN, AggKind.MAX):
+            return _string_minmax(group_ids, num_groups, values, kind, name)
+        data, cnt = rowops.grouped_agg(group_ids, num_groups, values,
+                                       kind)
+        out_dt = agg.to_field(batch.schema).dtype
+        validity = cnt > 0 if cnt is not None else None
+        if kind == AggKind.SUM and values.dtype.is_integer():
+            out = data.to(torch.int64)
+            return Series(name, out_dt, data=out.view(out_dt.to_torch())
+                          if out_dt.to_torch() != torch.int64 else out,
+                          validity=validity)
+        return Series(name, out_dt, data=data.to(out_dt.to_torch()),
+                      validity=validity)
+
+    if kind == AggKind.MEAN:
+        s, cnt = rowops.grouped_agg(group_ids, num_groups, values, "sum")
+        out = s.to(torch.float64) / cnt.clamp(min=1).to(torch.float64)
+        return Series(name, DataType.float64(), data=out, validity=cnt > 0)
+
+    if kind in (AggKind.STDDEV, AggKind.VARIANCE):
+        s, cnt = rowops.grouped_agg(group_ids, num_groups, values, "sum")
+        s2, _ = rowops.grouped_agg(group_ids, num_groups, values, "sum_sq")
+        c = cnt.clamp(min=1).to(torch.float64)
+        mean = s.to(torch.float64) / c
+        var = (s2.to(torch.float64) / c - mean * mean).clamp(min=0.0)
+        out = torch.sqrt(var) if kind == AggKind.STDDEV else var
+        return Series(name, DataType.float64(), data=out, validity=cnt > 0)
+
+    if kind == AggKind.SKEW:
+        s, cnt = rowops.grouped_agg(group_ids, num_groups, values, "sum")
+        s2, _ = rowops.grouped_agg(group_ids, num_groups, values, "sum_sq")
+        v3 = Series(values.name, DataType.float64(),
+                    data=values.data.to(torch.float64) ** 3,
+                    validity=values.validity)
+        s3, _ = rowops.grouped_agg(group_ids, num_groups, v3, "sum")
+        c = cnt.clamp(min=1).to(torch.float64)
+        m = s.to(torch.float64) / c
+        m2 = s2.to(torch.float64) / c - m * m
+        m3 = s3.to(torch.float64) / c - 3 * m * m2 - m ** 3
+        out = m3 / torch.pow(m2.clamp(min=1e-300), 1.5)
+        return Series(name, DataType.float64(), data=out, validity=cnt > 0)
+
+    if kind == AggKind.ANY_VALUE:
+        first_idx = _first_valid_index(group_ids, num_groups, values)
+        return values.take(first_idx).rename(name)
+
+    if kind in (AggKind.COUNT_DISTINCT, AggKind.APPROX_COUNT_DISTINCT):
+        # exact two-level groupby: distinct (group, value) pairs, then count.
+        # (HLL sketch path lands with the multimodal round.)
+        gseries = Series("__gid", DataType.int64(), data=group_ids)
+        sub_gids, sub_reps = rowops.groupby([gseries, values])
+        outer = group_ids[sub_reps]
+        valid = values.validity[sub_reps] if values.validity is not None \
+            else None
+        ones = torch.ones(len(sub_reps), dtype=torch.int64,
+                          device=group_ids.device)
+        if valid is not None:
+            ones = ones * valid.to(torch.int64)
+        out = torch.zeros(num_groups, dtype=torch.int64,
+                          device=group_ids.device)
+        out.scatter_add_(0, outer, ones)
+        return Series(name, DataType.uint64(), data=out.view(torch.uint64))
+
+    if kind in (AggKind.BOOL_AND, AggKind.BOOL_OR):
+        iv = Series(values.name, DataType.int64(),
+                    data=values.data.to(torch.int64),
+                    validity=values.validity)
+        op = "min" if kind == AggKind.BOOL_AND else "max"
+        data, cnt = rowops.grouped_agg(group_ids, num_groups, iv, op)
+        return Series(name, DataType.bool(), data=data != 0,
+                      validity=cnt > 0)
+
+    if kind in (AggKind.LIST, AggKind.CONCAT):
+        # sort rows by group id (stable) then slice by counts
+        perm = rowops._stable_sort_perm_by(group_ids)
+        sorted_vals = values.take(perm)
+        counts = torch.bincount(group_ids, minlength=num_groups)
+        offs = torch.zeros(num_groups + 1, dtype=torch.int64,
+                           device=group_ids.device)
+        torch.cumsum(counts, 0, out=offs[1:])
+        if kind == AggKind.CONCAT:
+            assert values.dtype.is_list(), "agg_concat requires list input"
+            # flatten one level: gather child ranges
+            inner = sorted_vals
+            child = inner.children[0]
+            lens = inner.offsets[1:] - inner.offsets[:-1]
+            gcounts = torch.zeros(num_groups, dtype=torch.int64,
+                                  device=group_ids.device)
+            gcounts.scatter_add_(0, group_ids[perm], lens)
+            goffs = torch.zeros(num_groups + 1, dtype=torch.int64,
+                                device=group_ids.device)
+            torch.cumsum(gcounts, 0, out=goffs[1:])
+            return Series(name, values.dtype, offsets=goffs,
+                          children=[child])
+        return Series(name, DataType.list(values.dtype), offsets=offs,
+                      children=[sorted_vals.rename("item")])
+
+    if kind == AggKind.APPROX_PERCENTILE:
+        # per-group exact percentile via sort (round-1; DDSketch later)
+        q = float(agg.param)
+        perm = rowops.argsort_multi(
+            [Series("g", DataType.int64(), data=group_ids), values],
+            [False, False], [False, False])
+        svals = values.take(perm).cast(DataType.float64())
+        sg = group_ids[perm]
+        counts = torch.bincount(group_ids, minlength=num_groups)
+        offs = torch.zeros(num_groups + 1, dtype=torch.int64,
+                           device=group_ids.device)
+        torch.cumsum(counts, 0, out=offs[1:])
+        pos = offs[:-1] + ((counts - 1).to(torch.float64) * q).to(torch.int64)
+        pos = torch.minimum(pos, (offs[1:] - 1).clamp(min=0))
+        out = svals.take(pos)
+        return out.rename(name).with_validity(counts > 0)
+
+    raise ValueError(f"unsupported aggregation {kind}")
+
+
+def _ones_series(n: int, dev) -> Series:
+    return Series("ones", DataType.int64(),
+                  data=torch.ones(n, dtype=torch.int64, device=dev))
+
+
+def _first_valid_index(group_ids: torch.Tensor, num_groups: int,
+                       values: Series) -> torch.Tensor:
+    dev = group_ids.device
+    n = len(values)
+    idx = torch.arange(n, dtype=torch.int64, device=dev)
+    if values.validity is not None:
+        idx = torch.where(values.validity, idx,
+                          torch.full_like(idx, n))
+    out = torch.full((num_groups,), n, dtype=torch.int64, device=dev)
+    out.scatter_reduce_(0, group_ids, idx, reduce="amin")
+    return torch.where(out == n, torch.full_like(out, -1), out)
+
+
+def _string_minmax(group_ids, num_groups, values: Series, kind, name):
+    # argsort once, take first per group (min) or last (max)
+    desc = kind == AggKind.MAX
+    perm = rowops.argsort_multi([values], [desc], [False])
+    # rank of each row in sorted order
+    n = len(values)
+    dev = values.device
+    rank = torch.empty(n, dtype=torch.int64, device=dev)
+    rank[perm] = torch.arange(n, dtype=torch.int64, device=dev)
+    if values.validity is not None:
+        rank = torch.where(values.validity, rank, torch.full_like(rank, n))
+    best = torch.full((num_groups,), n, dtype=torch.int64, device=dev)
+    best.scatter_reduce_(0, group_ids, rank, reduce="amin")
+    inv = torch.full((n + 1,), -1, dtype=torch.int64, device=dev)
+    inv[:n] = perm
+    sel = inv[best.clamp(max=n)]
+    sel = torch.where(best == n, torch.full_like(sel, -1), sel)
+    return values.take(sel).rename(name)
+
+
+def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
+                  aggs: List[ExprNode]) -> RecordBatch:
+    """One-shot (grouped or global) aggregation of a materialized batch."""
+    named_aggs, residuals = decompose_agg_exprs(aggs)
+    n = len(batch)
+    dev = batch.device
+
+    if groupby:
+        key_series = [e.evaluate(batch) for e in groupby]
+        key_series = [s.broadcast(n) if len(s) == 1 else s for s in key_series]
+        gids, reps = rowops.groupby(key_series)
+        num_groups = int(reps.shape[0])
+        key_cols = [s.take(reps) for s in key_series]
+    else:
+        gids = torch.zeros(n, dtype=torch.int64, device=dev)
+        num_groups = 1
+        key_cols = []
+
+    agg_cols = [compute_agg(batch, gids, num_groups, cname, a)
+                for cname, a in named_aggs]
+    inter = RecordBatch(key_cols + agg_cols,
+                        num_rows=num_groups)
+    out_cols = list(key_cols)
+    for r in residuals:
+        out_cols.append(r.evaluate(inter))
+    return RecordBatch(out_cols, num_rows=num_groups)

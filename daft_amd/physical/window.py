@@ -1,0 +1,214 @@
+"""Window function execution (ref:
+/root/reference/src/daft-local-execution/src/streaming_sink/
+window_partition_*.rs and daft-recordbatch/src/ops/window_states/).
+
+Supported: per-partition aggregates (sum/min/max/mean/count/...), running
+aggregates under an order-by, row_number / rank / dense_rank, lag / lead.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from ..expressions.expressions import Agg, AggKind, ExprNode
+from ..kernels import rowops
+from ..recordbatch import RecordBatch
+from ..schema import DataType, Field, Schema
+from ..series import Series
+from . import agg as agg_mod
+
+
+class WindowFn(ExprNode):
+    """A window function bound to a Window spec.  kind: 'agg' wraps an Agg
+    node; 'row_number'/'rank'/'dense_rank' are rank functions; 'lag'/'lead'
+    shift within partitions."""
+
+    def __init__(self, kind: str, inner: Optional[ExprNode], spec,
+                 offset: int = 1, default=None):
+        self.kind = kind
+        self.inner = inner
+        self.spec = spec
+        self.offset = offset
+        self.default = default
+
+    def children(self):
+        return [self.inner] if self.inner is not None else []
+
+    def with_children(self, ch):
+        return WindowFn(self.kind, ch[0] if ch else None, self.spec,
+                        self.offset, self.default)
+
+    def out_name(self):
+        if self.inner is not None:
+            return self.inner.out_name()
+        return self.kind
+
+    def to_field(self, schema):
+        return window_out_field(self, self.out_name(), schema)
+
+    def evaluate(self, batch):
+        raise RuntimeError("window expressions must run under a Window node")
+
+    def __repr__(self):
+        return f"{self.kind}({self.inner!r}).over(...)"
+
+
+def window_out_field(e: ExprNode, name: str, schema: Schema) -> Field:
+    if isinstance(e, WindowFn):
+        if e.kind == "agg":
+            f = e.inner.to_field(schema)
+            return Field(name, f.dtype)
+        if e.kind in ("row_number", "rank", "dense_rank"):
+            return Field(name, DataType.uint64())
+        if e.kind in ("lag", "lead"):
+            f = e.inner.to_field(schema)
+            return Field(name, f.dtype)
+    return Field(name, e.to_field(schema).dtype)
+
+
+def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
+               partition_by: List[ExprNode], order_by: List[ExprNode],
+               descending: List[bool], names: List[str]) -> RecordBatch:
+    n = len(batch)
+    dev = batch.device
+    if n == 0:
+        fields = batch.schema.fields()
+        from ..series import empty_series
+        cols = list(batch.columns)
+        for e, nm in zip(window_exprs, names):
+            f = window_out_field(e, nm, batch.schema)
+            cols.append(empty_series(f.name, f.dtype, dev))
+        return RecordBatch(cols, num_rows=0)
+
+    if partition_by:
+        pkeys = [e.evaluate(batch) for e in partition_by]
+        gids, reps = rowops.groupby(pkeys)
+        num_groups = int(reps.shape[0])
+    else:
+        gids = torch.zeros(n, dtype=torch.int64, device=dev)
+        num_groups = 1
+
+    sorted_pos = None
+    if order_by:
+        okeys = [e.evaluate(batch) for e in order_by]
+        gseries = Series("__g", DataType.int64(), data=gids)
+        perm = rowops.argsort_multi([gseries] + okeys,
+                                    [False] + list(descending),
+                                    [False] * (1 + len(okeys)))
+        # rank of each row within its sorted partition
+        sorted_gids = gids[perm]
+        part_start = _partition_starts(sorted_gids, num_groups)
+        pos_in_sorted = torch.empty(n, dtype=torch.int64, device=dev)
+        pos_in_sorted[perm] = torch.arange(n, dtype=torch.int64, device=dev)
+        sorted_pos = (perm, pos_in_sorted, part_start)
+
+    out_cols = list(batch.columns)
+    for e, nm in zip(window_exprs, names):
+        assert isinstance(e, WindowFn)
+        if e.kind == "agg":
+            a: Agg = e.inner
+            per_group = agg_mod.compute_agg(batch, gids, num_groups, nm, a)
+            out_cols.append(per_group.take(gids).rename(nm))
+        elif e.kind in ("row_number", "rank", "dense_rank"):
+            assert sorted_pos is not None, f"{e.kind} requires order_by"
+            perm, pos, part_start = sorted_pos
+            rn = pos - part_start[gids]
+            if e.kind == "row_number":
+                vals = rn + 1
+            else:
+                okeys = [k.evaluate(batch) for k in order_by]
+                same_as_prev = _same_as_prev(okeys, gids, perm, pos)
+                vals = _rank_values(rn, same_as_prev, gids, perm,
+                                    dense=e.kind == "dense_rank")
+            out_cols.append(Series(nm, DataType.uint64(),
+                                   data=vals.view(torch.uint64)))
+        elif e.kind in ("lag", "lead"):
+            assert sorted_pos is not None, f"{e.kind} requires order_by"
+            perm, pos, part_start = sorted_pos
+            vals = e.inner.evaluate(batch)
+            off = e.offset if e.kind == "lag" else -e.offset
+            src_sorted_pos = pos - off
+            # stay within partition bounds
+            inv = torch.empty(n, dtype=torch.int64, device=dev)
+            inv = perm  # inv[sorted_pos] = original row
+            src_pos_clamped = src_sorted_pos.clamp(min=0, max=n - 1)
+            src_row = inv[src_pos_clamped]
+            same_part = gids[src_row] == gids
+            in_bounds = (src_sorted_pos >= 0) & (src_sorted_pos < n) & same_part
+            src_idx = torch.where(in_bounds, src_row,
+                                  torch.full_like(src_row, -1))
+            shifted = vals.take(src_idx).rename(nm)
+            if e.default is not None:
+                fill = Series.from_pylist(nm, [e.default],
+                                          device=dev).broadcast(n)
+                shifted = shifted.fill_null(fill)
+            out_cols.append(shifted)
+        else:
+            raise ValueError(f"unknown window fn {e.kind}")
+    return RecordBatch(out_cols, num_rows=n)
+
+
+def _partition_starts(sorted_gids: torch.Tensor,
+                      num_groups: int) -> torch.Tensor:
+    n = sorted_gids.shape[0]
+    dev = sorted_gids.device
+    idx = torch.arange(n, dtype=torch.int64, device=dev)
+    starts = torch.full((num_groups,), n, dtype=torch.int64, device=dev)
+    starts.scatter_reduce_(0, sorted_gids, idx, reduce="amin")
+    return starts
+
+
+def _same_as_prev(okeys: List[Series], gids: torch.Tensor,
+                  perm: torch.Tensor, pos: torch.Tensor) -> torch.Tensor:
+    """For each row: do its order keys equal the previous row's (sorted order,
+    same partition)?  Returned in original row order."""
+    n = perm.shape[0]
+    dev = perm.device
+    prev_sorted = (pos - 1).clamp(min=0)
+    prev_row = perm[prev_sorted]
+    same = torch.ones(n, dtype=torch.bool, device=dev)
+    for k in okeys:
+        eq = k.compare(k.take(prev_row), "eq")
+        e = eq.data.clone()
+        if eq.validity is not None:
+            both_null = ~(k.validity if k.validity is not None else
+                          torch.ones(n, dtype=torch.bool, device=dev))
+            e = torch.where(eq.validity, e, both_null)
+        same &= e
+    same &= gids[prev_row] == gids
+    same &= pos > 0
+    return same
+
+
+def _rank_values(rn: torch.Tensor, same_as_prev: torch.Tensor,
+                 gids: torch.Tensor, perm: torch.Tensor,
+                 dense: bool) -> torch.Tensor:
+    """rank: 1 + count of strictly-smaller rows; dense_rank: 1 + distinct
+    smaller keys.  Computed via a cumulative pass in sorted order (host loop
+    avoided: cumsum of "new key" indicator per partition)."""
+    n = rn.shape[0]
+    dev = rn.device
+    sap = same_as_prev[perm]       # sorted-order "same key as previous row"
+    newkey = (~sap).to(torch.int64)  # 1 at every key start (incl. partition starts)
+    g_sorted = gids[perm]
+    starts = _partition_starts(g_sorted, int(g_sorted.max().item()) + 1
+                               if n else 1)
+    if dense:
+        # dense rank = per-partition cumsum of newkey
+        csum = torch.cumsum(newkey, 0)
+        base = csum[starts[g_sorted]] - 1  # csum at partition start is its 1
+        vals_sorted = csum - base
+    else:
+        # rank = (position of this key's first row within partition) + 1.
+        # Global key-start positions are strictly increasing, so a global
+        # cummax of (pos+1 at key starts, else 0) never crosses partitions
+        # (newkey is always 1 at partition starts).
+        pos_sorted = torch.arange(n, dtype=torch.int64, device=dev)
+        key_start = torch.where(newkey.bool(), pos_sorted + 1,
+                                torch.zeros_like(pos_sorted))
+        key_start_pos = torch.cummax(key_start, 0).values - 1
+        vals_sorted = key_start_pos - starts[g_sorted] + 1
+    out = torch.empty(n, dtype=torch.int64, device=dev)
+    out[perm] = vals_sorted
+    return out

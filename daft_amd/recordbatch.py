@@ -1,0 +1,224 @@
+"""RecordBatch: a batch of equal-length Series (ref:
+/root/reference/src/daft-recordbatch/src/lib.rs:68-72) and MicroPartition,
+the unit of data movement between operators (ref:
+/root/reference/src/daft-micropartition/src/micropartition.rs:35-53).
+
+On GPU every column buffer lives in HBM3E; the vectorized ops here
+(filter/take/sort/agg/join/partition) dispatch to the HIP kernel layer.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Tuple, Union
+
+import torch
+
+from .schema import DataType, Field, Schema, TypeKind
+from .series import Series, empty_series, full_null
+from . import kernels
+from .kernels import rowops
+
+
+class RecordBatch:
+    __slots__ = ("schema", "columns", "_num_rows")
+
+    def __init__(self, columns: List[Series], num_rows: Optional[int] = None):
+        self.columns = columns
+        self.schema = Schema([c.field() for c in columns])
+        if num_rows is None:
+            num_rows = len(columns[0]) if columns else 0
+        for c in columns:
+            assert len(c) == num_rows, \
+                f"column {c.name} length {len(c)} != {num_rows}"
+        self._num_rows = num_rows
+
+    # ------------------------------------------------------------------
+    @staticmethod
+    def from_pydict(data: Dict[str, list], device="cpu",
+                    schema: Optional[Schema] = None) -> "RecordBatch":
+        cols = []
+        for name, vals in data.items():
+            dt = schema[name].dtype if schema is not None and name in schema else None
+            if isinstance(vals, torch.Tensor):
+                cols.append(Series.from_torch(name, vals.to(device)))
+            else:
+                cols.append(Series.from_pylist(name, vals, dt, device=device))
+        return RecordBatch(cols)
+
+    @staticmethod
+    def empty(schema: Schema, device="cpu") -> "RecordBatch":
+        return RecordBatch([empty_series(f.name, f.dtype, device)
+                            for f in schema], num_rows=0)
+
+    @staticmethod
+    def from_arrow(table, device="cpu") -> "RecordBatch":
+        import pyarrow as pa
+        if isinstance(table, pa.RecordBatch):
+            table = pa.Table.from_batches([table])
+        cols = [Series.from_arrow(name, table.column(name))
+                for name in table.column_names]
+        rb = RecordBatch(cols, num_rows=table.num_rows)
+        return rb.to(device) if str(device) != "cpu" else rb
+
+    # ------------------------------------------------------------------
+    def __len__(self) -> int:
+        return self._num_rows
+
+    @property
+    def num_rows(self) -> int:
+        return self._num_rows
+
+    @property
+    def device(self) -> torch.device:
+        return self.columns[0].device if self.columns else torch.device("cpu")
+
+    def column(self, name: str) -> Series:
+        return self.columns[self.schema.index_of(name)]
+
+    def column_names(self) -> List[str]:
+        return self.schema.names()
+
+    def to(self, device) -> "RecordBatch":
+        return RecordBatch([c.to(device) for c in self.columns],
+                           self._num_rows)
+
+    def cpu(self) -> "RecordBatch":
+        return self.to("cpu")
+
+    def size_bytes(self) -> int:
+        total = 0
+        for c in self.columns:
+            for t in (c.data, c.validity, c.offsets):
+                if t is not None:
+                    total += t.numel() * t.element_size()
+            for ch in c.children:
+                total += RecordBatch([ch]).size_bytes()
+        return total
+
+    # ------------------------------------------------------------------
+    def select_columns(self, names: Sequence[str]) -> "RecordBatch":
+        return RecordBatch([self.column(n) for n in names], self._num_rows)
+
+    def with_columns(self, series: List[Series]) -> "RecordBatch":
+        cols = list(self.columns)
+        idx = {c.name: i for i, c in enumerate(cols)}
+        for s in series:
+            if s.name in idx:
+                cols[idx[s.name]] = s
+            else:
+                idx[s.name] = len(cols)
+                cols.append(s)
+        return RecordBatch(cols, self._num_rows)
+
+    def rename(self, mapping: Dict[str, str]) -> "RecordBatch":
+        return RecordBatch([c.rename(mapping.get(c.name, c.name))
+                            for c in self.columns], self._num_rows)
+
+    # ------------------------------------------------------------------
+    def take(self, indices: torch.Tensor) -> "RecordBatch":
+        return RecordBatch([c.take(indices) for c in self.columns],
+                           int(indices.shape[0]))
+
+    def filter(self, mask: Series) -> "RecordBatch":
+        idx = kernels.compact_indices(mask)
+        return self.take(idx)
+
+    def slice(self, start: int, end: int) -> "RecordBatch":
+        end = min(end, self._num_rows)
+        start = min(start, end)
+        return RecordBatch([c.slice(start, end) for c in self.columns],
+                           end - start)
+
+    def head(self, n: int) -> "RecordBatch":
+        return self.slice(0, n)
+
+    @staticmethod
+    def concat(batches: List["RecordBatch"]) -> "RecordBatch":
+        assert batches
+        if len(batches) == 1:
+            return batches[0]
+        names = batches[0].column_names()
+        cols = [Series.concat([b.column(n) for b in batches]) for n in names]
+        return RecordBatch(cols, sum(len(b) for b in batches))
+
+    # ------------------------------------------------------------------
+    def argsort(self, by: Sequence[str], descending: Sequence[bool],
+                nulls_first: Sequence[bool]) -> torch.Tensor:
+        keys = [self.column(n) for n in by]
+        return rowops.argsort_multi(keys, descending, nulls_first)
+
+    def sort(self, by: Sequence[str], descending: Sequence[bool],
+             nulls_first: Sequence[bool]) -> "RecordBatch":
+        return self.take(self.argsort(by, descending, nulls_first))
+
+    def hash_rows(self, columns: Optional[Sequence[str]] = None,
+                  seed: int = 0) -> torch.Tensor:
+        cols = [self.column(n) for n in (columns or self.column_names())]
+        return rowops.hash_columns(cols, seed)
+
+    def partition_by_hash(self, columns: Sequence[str],
+                          num_partitions: int) -> List["RecordBatch"]:
+        keys = [self.column(n) for n in columns]
+        perm, counts = rowops.partition_by_hash(keys, num_partitions)
+        reordered = self.take(perm)
+        out = []
+        start = 0
+        counts_l = counts.tolist()
+        for c in counts_l:
+            out.append(reordered.slice(start, start + c))
+            start += c
+        return out
+
+    def partition_by_range(self, keys: Sequence[str],
+                           boundaries: "RecordBatch",
+                           descending: Sequence[bool]) -> List["RecordBatch"]:
+        """Range partition: boundaries has num_partitions-1 sorted rows."""
+        part = _range_partition_ids(self, keys, boundaries, descending)
+        perm, counts = rowops.partition_by_value(part, len(boundaries) + 1)
+        reordered = self.take(perm)
+        out, start = [], 0
+        for c in counts.tolist():
+            out.append(reordered.slice(start, start + c))
+            start += c
+        return out
+
+    # ------------------------------------------------------------------
+    def to_pydict(self) -> Dict[str, list]:
+        return {c.name: c.to_pylist() for c in self.columns}
+
+    def to_arrow(self):
+        import pyarrow as pa
+        arrays = [c.to_arrow() for c in self.columns]
+        return pa.table(dict(zip(self.column_names(), arrays)))
+
+    def to_pandas(self):
+        return self.to_arrow().to_pandas()
+
+    def __repr__(self) -> str:
+        return (f"RecordBatch(rows={self._num_rows}, dev={self.device}, "
+                f"schema={self.schema!r})")
+
+
+def _range_partition_ids(rb: RecordBatch, keys: Sequence[str],
+                         boundaries: RecordBatch,
+                         descending: Sequence[bool]) -> torch.Tensor:
+    """For each row, count how many boundary rows sort strictly before it."""
+    n = len(rb)
+    dev = rb.device
+    part = torch.zeros(n, dtype=torch.int64, device=dev)
+    for b in range(len(boundaries)):
+        # row > boundary_b (lexicographically, honoring per-key direction)
+        gt = None  # strictly greater so far
+        eq = None  # equal so far
+        for ki, kname in enumerate(keys):
+            col = rb.column(kname)
+            bval = boundaries.column(kname).slice(b, b + 1).broadcast(n)
+            op_gt = "lt" if descending[ki] else "gt"
+            g = col.compare(bval, op_gt).data
+            e = col.compare(bval, "eq").data
+            if gt is None:
+                gt, eq = g, e
+            else:
+                gt = gt | (eq & g)
+                eq = eq & e
+        part += gt.to(torch.int64)
+    return part

@@ -1,0 +1,155 @@
+"""UDF system (ref: /root/reference/daft/udf/ — legacy @udf, new
+@daft.func/@daft.cls, executed by intermediate_ops/udf.rs).
+
+Round-1: in-process row-wise and batch-wise python UDFs with retry/on_error;
+stateful class UDFs via @cls/@method; subprocess isolation is a later-round
+item (the seam is run_udf_node)."""
+from __future__ import annotations
+
+import functools
+from typing import Any, Callable, List, Optional
+
+from .expressions.expressions import Expression, PyUDF, _to_node
+from .schema import DataType
+from .series import Series
+
+
+def func(fn: Optional[Callable] = None, *, return_dtype: Optional[DataType] = None,
+         batched: bool = False, max_retries: int = 0, on_error: str = "raise",
+         use_process: bool = False, gpus: int = 0,
+         max_concurrency: Optional[int] = None):
+    """@daft_amd.func — row-wise (or batched) scalar UDF."""
+    def wrap(f):
+        rd = return_dtype
+        if rd is None:
+            import typing
+            hints = typing.get_type_hints(f)
+            rd = _dtype_from_hint(hints.get("return"))
+
+        @functools.wraps(f)
+        def make_expr(*args):
+            nodes = [_to_node(a) for a in args]
+            return Expression(PyUDF(f.__name__, f, nodes, rd, batched,
+                                    max_retries, on_error, use_process,
+                                    max_concurrency, gpus))
+        make_expr.__daft_udf__ = True
+        return make_expr
+    if fn is not None:
+        return wrap(fn)
+    return wrap
+
+
+def udf(*, return_dtype: DataType, batch_size: Optional[int] = None,
+        num_cpus=None, num_gpus=None, max_retries: int = 0,
+        on_error: str = "raise", concurrency: Optional[int] = None):
+    """Legacy @udf decorator: fn receives Series arguments and returns a
+    Series/list (ref: daft/udf/legacy.py)."""
+    def wrap(f):
+        @functools.wraps(f)
+        def make_expr(*args):
+            nodes = [_to_node(a) for a in args]
+            return Expression(PyUDF(f.__name__, f, nodes, return_dtype,
+                                    batched=True, max_retries=max_retries,
+                                    on_error=on_error,
+                                    gpus=int(num_gpus or 0),
+                                    concurrency=concurrency))
+        make_expr.__daft_udf__ = True
+        return make_expr
+    return wrap
+
+
+def cls(klass=None, *, gpus: int = 0, max_concurrency: Optional[int] = None):
+    """@daft_amd.cls — stateful UDF class; instances are constructed lazily
+    once per worker process and reused across batches."""
+    def wrap(kls):
+        kls.__daft_cls__ = True
+        kls.__daft_gpus__ = gpus
+        return kls
+    if klass is not None:
+        return wrap(klass)
+    return wrap
+
+
+def method(fn: Optional[Callable] = None, *,
+           return_dtype: Optional[DataType] = None, batched: bool = False):
+    """@daft_amd.method — marks a method of a @cls as a UDF entrypoint."""
+    def wrap(f):
+        rd = return_dtype
+        if rd is None:
+            import typing
+            hints = typing.get_type_hints(f)
+            rd = _dtype_from_hint(hints.get("return"))
+
+        @functools.wraps(f)
+        def make_expr(self, *args):
+            inst_holder = {"inst": self}
+
+            def call(*vals):
+                return f(inst_holder["inst"], *vals)
+            call.__name__ = f.__name__
+            nodes = [_to_node(a) for a in args]
+            return Expression(PyUDF(f.__name__, call, nodes, rd, batched))
+        make_expr.__daft_udf__ = True
+        return make_expr
+    if fn is not None:
+        return wrap(fn)
+    return wrap
+
+
+def udaf(*args, **kwargs):
+    raise NotImplementedError("user-defined aggregations land next round")
+
+
+def _dtype_from_hint(hint) -> DataType:
+    import datetime
+    m = {int: DataType.int64(), float: DataType.float64(),
+         str: DataType.string(), bool: DataType.bool(),
+         bytes: DataType.binary(), datetime.date: DataType.date(),
+         datetime.datetime: DataType.timestamp("us")}
+    if hint in m:
+        return m[hint]
+    if hint is None:
+        raise TypeError("UDF needs return_dtype= or a return type hint")
+    return DataType.python()
+
+
+def run_udf_node(node: PyUDF, batch) -> Series:
+    """Execute a PyUDF over a RecordBatch (the UDF operator hot loop;
+    ref: intermediate_ops/udf.rs:233-273)."""
+    arg_series = [a.evaluate(batch) for a in node.args]
+    n = len(batch)
+    arg_series = [s.broadcast(n) if len(s) == 1 and n != 1 else s
+                  for s in arg_series]
+
+    attempts = node.max_retries + 1
+    last_err: Optional[Exception] = None
+    for _ in range(attempts):
+        try:
+            if node.batched:
+                out = node.fn(*arg_series)
+                if isinstance(out, Series):
+                    return out.rename(node.name).cast(node.return_dtype) \
+                        if out.dtype != node.return_dtype else \
+                        out.rename(node.name)
+                if hasattr(out, "__array__") and not isinstance(out, list):
+                    import numpy as np
+                    return Series.from_numpy(node.name, np.asarray(out)) \
+                        .to(batch.device)
+                import torch
+                if isinstance(out, torch.Tensor):
+                    return Series.from_torch(node.name, out.to(batch.device))
+                return Series.from_pylist(node.name, list(out),
+                                          node.return_dtype,
+                                          device=batch.device)
+            cols = [s.cpu().to_pylist() for s in arg_series]
+            out_vals = []
+            for row in zip(*cols) if cols else [()] * n:
+                out_vals.append(node.fn(*row))
+            return Series.from_pylist(node.name, out_vals, node.return_dtype,
+                                      device=batch.device)
+        except Exception as e:  # retry path (ref: python_udf/retry.rs)
+            last_err = e
+    if node.on_error == "null":
+        from .series import full_null
+        return full_null(node.name, node.return_dtype, n, batch.device)
+    raise last_err
