@@ -1,0 +1,66 @@
+// Host-side launcher declarations for the daft_amd HIP kernels.
+#pragma once
+
+#include <torch/extension.h>
+#include <vector>
+
+using torch::Tensor;
+using OptTensor = c10::optional<Tensor>;
+
+// descriptor packing (kernels.hip)
+Tensor pack_descs(const std::vector<int64_t>& tags,
+                  const std::vector<Tensor>& datas,
+                  const std::vector<OptTensor>& offsets,
+                  const std::vector<OptTensor>& validities);
+
+// hashing
+Tensor hash_rows(const std::vector<int64_t>& tags,
+                 const std::vector<Tensor>& datas,
+                 const std::vector<OptTensor>& offsets,
+                 const std::vector<OptTensor>& validities, int64_t n,
+                 int64_t seed);
+
+// selection
+Tensor compact_indices(Tensor mask);
+std::vector<Tensor> take_string(Tensor offsets, Tensor bytes, Tensor idx);
+Tensor merge_strings(Tensor mask, Tensor t_off, Tensor t_bytes, Tensor f_off,
+                     Tensor f_bytes, Tensor out_off);
+
+// groupby
+std::vector<Tensor> groupby(Tensor hashes, const std::vector<int64_t>& tags,
+                            const std::vector<Tensor>& datas,
+                            const std::vector<OptTensor>& offsets,
+                            const std::vector<OptTensor>& validities);
+std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
+                                Tensor values, Tensor valid,
+                                const std::string& op);
+
+// join
+std::vector<Tensor> join_build(Tensor hashes);
+std::vector<Tensor> join_probe(
+    Tensor table, Tensor next, Tensor probe_hashes, Tensor build_hashes,
+    const std::vector<int64_t>& tags_l, const std::vector<Tensor>& data_l,
+    const std::vector<OptTensor>& off_l, const std::vector<OptTensor>& val_l,
+    const std::vector<int64_t>& tags_r, const std::vector<Tensor>& data_r,
+    const std::vector<OptTensor>& off_r, const std::vector<OptTensor>& val_r,
+    int64_t mode);
+
+// sort
+Tensor radix_argsort(Tensor keys);
+Tensor string_chunk_key(Tensor offsets, Tensor bytes, int64_t chunk);
+
+// partition
+Tensor u64_mod(Tensor hashes, int64_t n_partitions);
+
+// strings (strings.hip)
+Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
+Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,
+                bool anchored_prefix, bool anchored_suffix);
+Tensor str_case(Tensor offsets, Tensor bytes, int64_t mode);
+std::vector<Tensor> str_substr(Tensor offsets, Tensor bytes, int64_t start,
+                               int64_t length);
+std::vector<Tensor> str_concat(const std::vector<Tensor>& offsets,
+                               const std::vector<Tensor>& bytes);
+Tensor str_char_length(Tensor offsets, Tensor bytes);
+Tensor string_compare(Tensor a_off, Tensor a_bytes, Tensor b_off,
+                      Tensor b_bytes);

@@ -127,7 +127,7 @@ def from_arrow_array(name: str, arr) -> Series:
                              count=start + nbytes)[start:] if bufs[2] is not None \
             else np.zeros(0, np.uint8)
         return Series(name, dt, data=torch.from_numpy(data.copy()),
-                      offsets=torch.from_numpy(np.ascontiguousarray(off)),
+                      offsets=torch.from_numpy(np.ascontiguousarray(off).copy()),
                       validity=validity)
     if k == TypeKind.LIST:
         if not pa.types.is_large_list(arr.type):

@@ -51,6 +51,7 @@ def native_required():
 # ---------------------------------------------------------------------------
 # tags shared with csrc/row_ops.h
 _TAG_W1, _TAG_W2, _TAG_W4, _TAG_W8, _TAG_STR = 0, 1, 2, 3, 4
+_TAG_F32, _TAG_F64 = 5, 6
 
 
 def _col_desc(s: Series):
@@ -61,6 +62,10 @@ def _col_desc(s: Series):
     data = s.data
     if data is None:
         raise TypeError(f"column {s.name}: {s.dtype!r} not usable as row key")
+    if data.dtype == torch.float32:
+        return (_TAG_F32, data, None, s.validity)
+    if data.dtype == torch.float64:
+        return (_TAG_F64, data, None, s.validity)
     width = data.element_size()
     tag = {1: _TAG_W1, 2: _TAG_W2, 4: _TAG_W4, 8: _TAG_W8}[width]
     return (tag, data, None, s.validity)

@@ -215,6 +215,13 @@ def join(left_keys: Sequence[Series], right_keys: Sequence[Series],
     is meaningful (right_idx is empty).
     Build side is the RIGHT side (callers put the smaller input right).
     """
+    from ..schema import supertype
+    lk2, rk2 = [], []
+    for l, r in zip(left_keys, right_keys):
+        st = supertype(l.dtype, r.dtype)
+        lk2.append(l.cast(st) if l.dtype != st else l)
+        rk2.append(r.cast(st) if r.dtype != st else r)
+    left_keys, right_keys = lk2, rk2
     if _is_gpu(left_keys[0]):
         return _gpu_join(left_keys, right_keys, how)
     return _cpu_join(left_keys, right_keys, how)
