@@ -161,8 +161,17 @@ def take(s: Series, indices: torch.Tensor) -> Series:
                       length=n_out)
     # fixed width: torch gather (hipified index_select is memory-bound optimal
     # for contiguous gathers; a fused HIP gather handles multi-column takes
-    # at the recordbatch layer)
-    data = s.data[safe_idx] if n_out else s.data[:0]
+    # at the recordbatch layer).  Unsigned ints gather through a signed view
+    # (no unsigned index_cuda kernels in torch-rocm).
+    d = s.data
+    _signed = {torch.uint16: torch.int16, torch.uint32: torch.int32,
+               torch.uint64: torch.int64}
+    uview = _signed.get(d.dtype)
+    if uview is not None:
+        d = d.view(uview)
+    data = d[safe_idx] if n_out else d[:0]
+    if uview is not None:
+        data = data.view(s.data.dtype)
     return Series(s.name, s.dtype, data=data, validity=validity)
 
 
