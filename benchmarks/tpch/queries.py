@@ -1,0 +1,411 @@
+"""TPC-H Q1-Q22 as daft_amd DataFrame programs, written from the TPC-H
+specification SQL (capability parity with the reference's DataFrame answers,
+benchmarking/tpch/answers.py — implemented independently against the spec).
+
+Each qN takes a dict of table-name -> DataFrame and returns a lazy DataFrame.
+"""
+from __future__ import annotations
+
+import datetime as dt
+
+from daft_amd import col, lit
+from daft_amd.expressions.expressions import Expression
+
+
+def _d(y, m, d_):
+    return dt.date(y, m, d_)
+
+
+def q1(T):
+    li = T["lineitem"]
+    disc_price = col("l_extendedprice") * (1 - col("l_discount"))
+    charge = disc_price * (1 + col("l_tax"))
+    return (li.where(col("l_shipdate") <= _d(1998, 9, 2))
+            .groupby("l_returnflag", "l_linestatus")
+            .agg(col("l_quantity").sum().alias("sum_qty"),
+                 col("l_extendedprice").sum().alias("sum_base_price"),
+                 disc_price.sum().alias("sum_disc_price"),
+                 charge.sum().alias("sum_charge"),
+                 col("l_quantity").mean().alias("avg_qty"),
+                 col("l_extendedprice").mean().alias("avg_price"),
+                 col("l_discount").mean().alias("avg_disc"),
+                 col("l_quantity").count().alias("count_order"))
+            .sort(["l_returnflag", "l_linestatus"]))
+
+
+def q2(T):
+    part = T["part"].where((col("p_size") == 15) &
+                           col("p_type").str.endswith("BRASS"))
+    region = T["region"].where(col("r_name") == "EUROPE")
+    nation = T["nation"].join(region, left_on="n_regionkey",
+                              right_on="r_regionkey")
+    supp = T["supplier"].join(nation, left_on="s_nationkey",
+                              right_on="n_nationkey")
+    ps = (T["partsupp"]
+          .join(part, left_on="ps_partkey", right_on="p_partkey")
+          .join(supp, left_on="ps_suppkey", right_on="s_suppkey"))
+    mins = (ps.groupby("ps_partkey")
+            .agg(col("ps_supplycost").min().alias("min_cost")))
+    out = ps.join(mins, on="ps_partkey") \
+        .where(col("ps_supplycost") == col("min_cost"))
+    return (out.select("s_acctbal", "s_name", "n_name",
+                       col("ps_partkey").alias("p_partkey"), "p_mfgr",
+                       "s_address", "s_phone", "s_comment")
+            .sort(["s_acctbal", "n_name", "s_name", "p_partkey"],
+                  desc=[True, False, False, False])
+            .limit(100))
+
+
+def q3(T):
+    cust = T["customer"].where(col("c_mktsegment") == "BUILDING")
+    orders = T["orders"].where(col("o_orderdate") < _d(1995, 3, 15))
+    li = T["lineitem"].where(col("l_shipdate") > _d(1995, 3, 15))
+    j = (li.join(orders, left_on="l_orderkey", right_on="o_orderkey")
+         .join(cust, left_on="o_custkey", right_on="c_custkey"))
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    return (j.groupby("l_orderkey", "o_orderdate", "o_shippriority")
+            .agg(rev.sum().alias("revenue"))
+            .select(col("l_orderkey").alias("o_orderkey"), "revenue",
+                    "o_orderdate", "o_shippriority")
+            .sort(["revenue", "o_orderdate"], desc=[True, False])
+            .limit(10))
+
+
+def q4(T):
+    orders = T["orders"].where(
+        (col("o_orderdate") >= _d(1993, 7, 1)) &
+        (col("o_orderdate") < _d(1993, 10, 1)))
+    late = T["lineitem"].where(col("l_commitdate") < col("l_receiptdate"))
+    j = orders.join(late, left_on="o_orderkey", right_on="l_orderkey",
+                    how="semi")
+    return (j.groupby("o_orderpriority")
+            .agg(col("o_orderpriority").count().alias("order_count"))
+            .sort("o_orderpriority"))
+
+
+def q5(T):
+    region = T["region"].where(col("r_name") == "ASIA")
+    nation = T["nation"].join(region, left_on="n_regionkey",
+                              right_on="r_regionkey")
+    orders = T["orders"].where(
+        (col("o_orderdate") >= _d(1994, 1, 1)) &
+        (col("o_orderdate") < _d(1995, 1, 1)))
+    j = (orders.join(T["customer"], left_on="o_custkey",
+                     right_on="c_custkey")
+         .join(T["lineitem"], left_on="o_orderkey", right_on="l_orderkey")
+         .join(T["supplier"],
+               left_on=["l_suppkey", "c_nationkey"],
+               right_on=["s_suppkey", "s_nationkey"])
+         .join(nation, left_on="c_nationkey", right_on="n_nationkey"))
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    return (j.groupby("n_name").agg(rev.sum().alias("revenue"))
+            .sort("revenue", desc=True))
+
+
+def q6(T):
+    li = T["lineitem"].where(
+        (col("l_shipdate") >= _d(1994, 1, 1)) &
+        (col("l_shipdate") < _d(1995, 1, 1)) &
+        (col("l_discount") >= 0.05) & (col("l_discount") <= 0.07) &
+        (col("l_quantity") < 24))
+    return li.agg((col("l_extendedprice") * col("l_discount")).sum()
+                  .alias("revenue"))
+
+
+def q7(T):
+    n1 = T["nation"].select(col("n_nationkey").alias("n1_key"),
+                            col("n_name").alias("supp_nation"))
+    n2 = T["nation"].select(col("n_nationkey").alias("n2_key"),
+                            col("n_name").alias("cust_nation"))
+    li = T["lineitem"].where(
+        (col("l_shipdate") >= _d(1995, 1, 1)) &
+        (col("l_shipdate") <= _d(1996, 12, 31)))
+    j = (li.join(T["supplier"], left_on="l_suppkey", right_on="s_suppkey")
+         .join(T["orders"], left_on="l_orderkey", right_on="o_orderkey")
+         .join(T["customer"], left_on="o_custkey", right_on="c_custkey")
+         .join(n1, left_on="s_nationkey", right_on="n1_key")
+         .join(n2, left_on="c_nationkey", right_on="n2_key")
+         .where(((col("supp_nation") == "FRANCE") &
+                 (col("cust_nation") == "GERMANY")) |
+                ((col("supp_nation") == "GERMANY") &
+                 (col("cust_nation") == "FRANCE"))))
+    vol = col("l_extendedprice") * (1 - col("l_discount"))
+    return (j.with_column("l_year", col("l_shipdate").dt.year())
+            .groupby("supp_nation", "cust_nation", "l_year")
+            .agg(vol.sum().alias("revenue"))
+            .sort(["supp_nation", "cust_nation", "l_year"]))
+
+
+def q8(T):
+    region = T["region"].where(col("r_name") == "AMERICA")
+    n1 = T["nation"].join(region, left_on="n_regionkey",
+                          right_on="r_regionkey") \
+        .select(col("n_nationkey").alias("n1_key"))
+    n2 = T["nation"].select(col("n_nationkey").alias("n2_key"),
+                            col("n_name").alias("supp_nation"))
+    part = T["part"].where(col("p_type") == "ECONOMY ANODIZED STEEL")
+    orders = T["orders"].where(
+        (col("o_orderdate") >= _d(1995, 1, 1)) &
+        (col("o_orderdate") <= _d(1996, 12, 31)))
+    j = (T["lineitem"]
+         .join(part, left_on="l_partkey", right_on="p_partkey")
+         .join(orders, left_on="l_orderkey", right_on="o_orderkey")
+         .join(T["customer"], left_on="o_custkey", right_on="c_custkey")
+         .join(n1, left_on="c_nationkey", right_on="n1_key")
+         .join(T["supplier"], left_on="l_suppkey", right_on="s_suppkey")
+         .join(n2, left_on="s_nationkey", right_on="n2_key"))
+    vol = col("l_extendedprice") * (1 - col("l_discount"))
+    j = j.with_column("o_year", col("o_orderdate").dt.year()) \
+         .with_column("volume", vol) \
+         .with_column("brazil_vol",
+                      (col("supp_nation") == "BRAZIL")
+                      .if_else(vol, lit(0.0)))
+    return (j.groupby("o_year")
+            .agg(col("brazil_vol").sum().alias("num"),
+                 col("volume").sum().alias("den"))
+            .select(col("o_year"),
+                    (col("num") / col("den")).alias("mkt_share"))
+            .sort("o_year"))
+
+
+def q9(T):
+    part = T["part"].where(col("p_name").str.contains("green"))
+    j = (T["lineitem"]
+         .join(part, left_on="l_partkey", right_on="p_partkey")
+         .join(T["supplier"], left_on="l_suppkey", right_on="s_suppkey")
+         .join(T["partsupp"],
+               left_on=["l_partkey", "l_suppkey"],
+               right_on=["ps_partkey", "ps_suppkey"])
+         .join(T["orders"], left_on="l_orderkey", right_on="o_orderkey")
+         .join(T["nation"], left_on="s_nationkey", right_on="n_nationkey"))
+    profit = (col("l_extendedprice") * (1 - col("l_discount")) -
+              col("ps_supplycost") * col("l_quantity"))
+    return (j.with_column("o_year", col("o_orderdate").dt.year())
+            .groupby(col("n_name").alias("nation"), "o_year")
+            .agg(profit.sum().alias("sum_profit"))
+            .sort(["nation", "o_year"], desc=[False, True]))
+
+
+def q10(T):
+    orders = T["orders"].where(
+        (col("o_orderdate") >= _d(1993, 10, 1)) &
+        (col("o_orderdate") < _d(1994, 1, 1)))
+    li = T["lineitem"].where(col("l_returnflag") == "R")
+    j = (li.join(orders, left_on="l_orderkey", right_on="o_orderkey")
+         .join(T["customer"], left_on="o_custkey", right_on="c_custkey")
+         .join(T["nation"], left_on="c_nationkey", right_on="n_nationkey"))
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    return (j.groupby(col("o_custkey").alias("c_custkey"), "c_name",
+                      "c_acctbal", "c_phone", "n_name", "c_address",
+                      "c_comment")
+            .agg(rev.sum().alias("revenue"))
+            .select("c_custkey", "c_name", "revenue", "c_acctbal", "n_name",
+                    "c_address", "c_phone", "c_comment")
+            .sort("revenue", desc=True)
+            .limit(20))
+
+
+def q11(T, sf: float = 1.0):
+    nation = T["nation"].where(col("n_name") == "GERMANY")
+    j = (T["partsupp"]
+         .join(T["supplier"], left_on="ps_suppkey", right_on="s_suppkey")
+         .join(nation, left_on="s_nationkey", right_on="n_nationkey"))
+    value = col("ps_supplycost") * col("ps_availqty")
+    grouped = j.groupby("ps_partkey").agg(value.sum().alias("value"))
+    total = grouped.agg(col("value").sum().alias("t")).to_pydict()["t"][0]
+    threshold = (total or 0.0) * 0.0001 / sf
+    return (grouped.where(col("value") > threshold)
+            .sort("value", desc=True))
+
+
+def q12(T):
+    li = T["lineitem"].where(
+        col("l_shipmode").is_in(["MAIL", "SHIP"]) &
+        (col("l_commitdate") < col("l_receiptdate")) &
+        (col("l_shipdate") < col("l_commitdate")) &
+        (col("l_receiptdate") >= _d(1994, 1, 1)) &
+        (col("l_receiptdate") < _d(1995, 1, 1)))
+    j = li.join(T["orders"], left_on="l_orderkey", right_on="o_orderkey")
+    high = col("o_orderpriority").is_in(["1-URGENT", "2-HIGH"])
+    return (j.groupby("l_shipmode")
+            .agg(high.if_else(lit(1), lit(0)).sum().alias("high_line_count"),
+                 high.if_else(lit(0), lit(1)).sum().alias("low_line_count"))
+            .sort("l_shipmode"))
+
+
+def q13(T):
+    orders = T["orders"].where(
+        ~col("o_comment").str.like("%special%requests%"))
+    j = T["customer"].join(orders, left_on="c_custkey", right_on="o_custkey",
+                           how="left")
+    counts = (j.groupby("c_custkey")
+              .agg(col("o_orderkey").count().alias("c_count")))
+    return (counts.groupby("c_count")
+            .agg(col("c_count").count().alias("custdist"))
+            .sort(["custdist", "c_count"], desc=[True, True]))
+
+
+def q14(T):
+    li = T["lineitem"].where(
+        (col("l_shipdate") >= _d(1995, 9, 1)) &
+        (col("l_shipdate") < _d(1995, 10, 1)))
+    j = li.join(T["part"], left_on="l_partkey", right_on="p_partkey")
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    promo = col("p_type").str.startswith("PROMO")
+    return j.agg(
+        (promo.if_else(rev, lit(0.0)).sum() * 100.0 / rev.sum())
+        .alias("promo_revenue"))
+
+
+def q15(T):
+    li = T["lineitem"].where(
+        (col("l_shipdate") >= _d(1996, 1, 1)) &
+        (col("l_shipdate") < _d(1996, 4, 1)))
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    revenue = (li.groupby(col("l_suppkey").alias("supplier_no"))
+               .agg(rev.sum().alias("total_revenue"))).collect()
+    top = revenue.agg(col("total_revenue").max().alias("m")) \
+        .to_pydict()["m"][0]
+    return (T["supplier"]
+            .join(revenue.where(col("total_revenue") >= (top or 0.0) - 1e-9),
+                  left_on="s_suppkey", right_on="supplier_no")
+            .select("s_suppkey", "s_name", "s_address", "s_phone",
+                    "total_revenue")
+            .sort("s_suppkey"))
+
+
+def q16(T):
+    part = T["part"].where(
+        (col("p_brand") != "Brand#45") &
+        ~col("p_type").str.startswith("MEDIUM POLISHED") &
+        col("p_size").is_in([49, 14, 23, 45, 19, 3, 36, 9]))
+    bad_supp = T["supplier"].where(
+        col("s_comment").str.like("%Customer%Complaints%"))
+    ps = (T["partsupp"]
+          .join(part, left_on="ps_partkey", right_on="p_partkey")
+          .join(bad_supp, left_on="ps_suppkey", right_on="s_suppkey",
+                how="anti"))
+    return (ps.groupby("p_brand", "p_type", "p_size")
+            .agg(col("ps_suppkey").count_distinct().alias("supplier_cnt"))
+            .sort(["supplier_cnt", "p_brand", "p_type", "p_size"],
+                  desc=[True, False, False, False]))
+
+
+def q17(T):
+    part = T["part"].where((col("p_brand") == "Brand#23") &
+                           (col("p_container") == "MED BOX"))
+    li = T["lineitem"].join(part, left_on="l_partkey", right_on="p_partkey")
+    avgs = (li.groupby("l_partkey")
+            .agg((col("l_quantity").mean() * 0.2).alias("qty_limit")))
+    j = li.join(avgs, on="l_partkey") \
+        .where(col("l_quantity") < col("qty_limit"))
+    return j.agg((col("l_extendedprice").sum() / 7.0).alias("avg_yearly"))
+
+
+def q18(T):
+    big = (T["lineitem"].groupby("l_orderkey")
+           .agg(col("l_quantity").sum().alias("sum_qty"))
+           .where(col("sum_qty") > 300))
+    j = (T["orders"].join(big, left_on="o_orderkey", right_on="l_orderkey")
+         .join(T["customer"], left_on="o_custkey", right_on="c_custkey"))
+    return (j.select("c_name", "c_custkey", "o_orderkey", "o_orderdate",
+                     "o_totalprice", col("sum_qty"))
+            .sort(["o_totalprice", "o_orderdate"], desc=[True, False])
+            .limit(100))
+
+
+def q19(T):
+    j = T["lineitem"].join(T["part"], left_on="l_partkey",
+                           right_on="p_partkey")
+    sm = (col("p_brand") == "Brand#12") & \
+        col("p_container").is_in(["SM CASE", "SM BOX", "SM PACK", "SM PKG"]) & \
+        (col("l_quantity") >= 1) & (col("l_quantity") <= 11) & \
+        (col("p_size") >= 1) & (col("p_size") <= 5)
+    med = (col("p_brand") == "Brand#23") & \
+        col("p_container").is_in(["MED BAG", "MED BOX", "MED PKG", "MED PACK"]) & \
+        (col("l_quantity") >= 10) & (col("l_quantity") <= 20) & \
+        (col("p_size") >= 1) & (col("p_size") <= 10)
+    lg = (col("p_brand") == "Brand#34") & \
+        col("p_container").is_in(["LG CASE", "LG BOX", "LG PACK", "LG PKG"]) & \
+        (col("l_quantity") >= 20) & (col("l_quantity") <= 30) & \
+        (col("p_size") >= 1) & (col("p_size") <= 15)
+    common = col("l_shipmode").is_in(["AIR", "AIR REG"]) & \
+        (col("l_shipinstruct") == "DELIVER IN PERSON")
+    rev = col("l_extendedprice") * (1 - col("l_discount"))
+    return j.where(common & (sm | med | lg)) \
+        .agg(rev.sum().alias("revenue"))
+
+
+def q20(T):
+    part = T["part"].where(col("p_name").str.startswith("forest"))
+    li94 = T["lineitem"].where(
+        (col("l_shipdate") >= _d(1994, 1, 1)) &
+        (col("l_shipdate") < _d(1995, 1, 1)))
+    qty = (li94.groupby("l_partkey", "l_suppkey")
+           .agg((col("l_quantity").sum() * 0.5).alias("half_qty")))
+    ps = (T["partsupp"]
+          .join(part, left_on="ps_partkey", right_on="p_partkey", how="semi")
+          .join(qty, left_on=["ps_partkey", "ps_suppkey"],
+                right_on=["l_partkey", "l_suppkey"])
+          .where(col("ps_availqty") > col("half_qty")))
+    nation = T["nation"].where(col("n_name") == "CANADA")
+    supp = T["supplier"].join(nation, left_on="s_nationkey",
+                              right_on="n_nationkey")
+    return (supp.join(ps, left_on="s_suppkey", right_on="ps_suppkey",
+                      how="semi")
+            .select("s_name", "s_address")
+            .sort("s_name"))
+
+
+def q21(T):
+    orders_f = T["orders"].where(col("o_orderstatus") == "F") \
+        .select("o_orderkey")
+    li = T["lineitem"].join(orders_f, left_on="l_orderkey",
+                            right_on="o_orderkey", how="semi")
+    per_order = (li.groupby("l_orderkey")
+                 .agg(col("l_suppkey").count_distinct().alias("n_supp")))
+    late = li.where(col("l_receiptdate") > col("l_commitdate"))
+    late_per_order = (late.groupby("l_orderkey")
+                      .agg(col("l_suppkey").count_distinct()
+                           .alias("n_late_supp")))
+    qualifying = (per_order.join(late_per_order, on="l_orderkey")
+                  .where((col("n_supp") > 1) & (col("n_late_supp") == 1))
+                  .select("l_orderkey"))
+    nation = T["nation"].where(col("n_name") == "SAUDI ARABIA")
+    supp = T["supplier"].join(nation, left_on="s_nationkey",
+                              right_on="n_nationkey")
+    j = (late.join(qualifying, on="l_orderkey", how="semi")
+         .join(supp, left_on="l_suppkey", right_on="s_suppkey"))
+    return (j.groupby("s_name")
+            .agg(col("s_name").count().alias("numwait"))
+            .sort(["numwait", "s_name"], desc=[True, False])
+            .limit(100))
+
+
+def q22(T):
+    codes = ["13", "31", "23", "29", "30", "18", "17"]
+    cust = T["customer"].with_column(
+        "cntrycode", col("c_phone").str.substr(0, 2)) \
+        .where(col("cntrycode").is_in(codes))
+    avg_bal = (cust.where(col("c_acctbal") > 0.0)
+               .agg(col("c_acctbal").mean().alias("a"))
+               .to_pydict()["a"][0]) or 0.0
+    eligible = (cust.where(col("c_acctbal") > avg_bal)
+                .join(T["orders"], left_on="c_custkey", right_on="o_custkey",
+                      how="anti"))
+    return (eligible.groupby("cntrycode")
+            .agg(col("c_acctbal").count().alias("numcust"),
+                 col("c_acctbal").sum().alias("totacctbal"))
+            .sort("cntrycode"))
+
+
+ALL = [q1, q2, q3, q4, q5, q6, q7, q8, q9, q10, q11, q12, q13, q14, q15, q16,
+       q17, q18, q19, q20, q21, q22]
+
+
+def run_query(i: int, tables, sf: float = 1.0):
+    """Run query i (1-based); returns the collected DataFrame."""
+    q = ALL[i - 1]
+    if q is q11:
+        return q(tables, sf=sf).collect()
+    return q(tables).collect()

@@ -419,6 +419,7 @@ class Join(LogicalPlan):
     def _compute_schema(self):
         ls, rs = self.children[0].schema, self.children[1].schema
         if self.how in ("semi", "anti"):
+            self._right_cols = []
             return ls
         fields = ls.fields()
         # join keys with identical names merge (daft semantics: right key cols
