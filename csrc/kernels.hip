@@ -177,13 +177,24 @@ __global__ void gather_bytes_kernel(const int64_t* src_off,
                                     const int64_t* idx,
                                     const int64_t* out_off, int64_t nrows,
                                     int64_t total, uint8_t* out) {
+  // One binary search per 256-byte block (LDS-shared), then threads walk
+  // offsets linearly — offsets are monotone so the walk is short and the
+  // reads coalesce within the block.
+  __shared__ int64_t block_row;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < total;
-       j += stride) {
-    int64_t row = upper_bound_row(out_off, nrows, j);
-    int64_t within = j - out_off[row];
-    int64_t src_row = idx[row];
-    out[j] = src_bytes[src_off[src_row] + within];
+  for (int64_t j0 = (int64_t)blockIdx.x * blockDim.x; j0 < total;
+       j0 += stride) {
+    if (threadIdx.x == 0) block_row = upper_bound_row(out_off, nrows, j0);
+    __syncthreads();
+    int64_t j = j0 + threadIdx.x;
+    if (j < total) {
+      int64_t row = block_row;
+      while (out_off[row + 1] <= j) ++row;
+      int64_t within = j - out_off[row];
+      int64_t src_row = idx[row];
+      out[j] = src_bytes[src_off[src_row] + within];
+    }
+    __syncthreads();
   }
 }
 
@@ -214,13 +225,21 @@ __global__ void merge_strings_kernel(const bool* m, const int64_t* t_off,
                                      const uint8_t* f_bytes,
                                      const int64_t* out_off, int64_t nrows,
                                      int64_t total, uint8_t* out) {
+  __shared__ int64_t block_row;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < total;
-       j += stride) {
-    int64_t row = upper_bound_row(out_off, nrows, j);
-    int64_t within = j - out_off[row];
-    out[j] = m[row] ? t_bytes[t_off[row] + within]
-                    : f_bytes[f_off[row] + within];
+  for (int64_t j0 = (int64_t)blockIdx.x * blockDim.x; j0 < total;
+       j0 += stride) {
+    if (threadIdx.x == 0) block_row = upper_bound_row(out_off, nrows, j0);
+    __syncthreads();
+    int64_t j = j0 + threadIdx.x;
+    if (j < total) {
+      int64_t row = block_row;
+      while (out_off[row + 1] <= j) ++row;
+      int64_t within = j - out_off[row];
+      out[j] = m[row] ? t_bytes[t_off[row] + within]
+                      : f_bytes[f_off[row] + within];
+    }
+    __syncthreads();
   }
 }
 
@@ -258,12 +277,19 @@ __global__ void groupby_insert_kernel(const uint64_t* hashes,
     uint64_t h = hashes[i];
     uint64_t slot = h & mask;
     while (true) {
-      long long prev = atomicCAS((unsigned long long*)&table[slot],
-                                 (unsigned long long)(-1ll),
-                                 (unsigned long long)i);
-      if (prev == -1ll) {  // claimed: i is the representative
-        row_slot[i] = (int64_t)slot;
-        break;
+      // Plain read first: slots transition -1 -> owner exactly once, so a
+      // non-(-1) read is final; a stale -1 is corrected by the device-scope
+      // CAS below.  This avoids serializing 10^8 CAS ops on a handful of
+      // slots for low-cardinality groupbys (the Q1 shape).
+      long long prev = table[slot];
+      if (prev == -1ll) {
+        prev = atomicCAS((unsigned long long*)&table[slot],
+                         (unsigned long long)(-1ll),
+                         (unsigned long long)i);
+        if (prev == -1ll) {  // claimed: i is the representative
+          row_slot[i] = (int64_t)slot;
+          break;
+        }
       }
       if (hashes[prev] == h &&
           row_eq(cols, cols, ncols, prev, i, /*null_eq=*/true)) {
