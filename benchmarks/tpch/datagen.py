@@ -1,29 +1,34 @@
-"""TPC-H dbgen-equivalent synthetic data generator, device-native.
+"""TPC-H dbgen-equivalent synthetic data generator, device-native and
+SHARD-INVARIANT.
 
 Generates all 8 tables directly in HBM with torch ops (no host round-trip),
 following the TPC-H spec's schema, cardinalities, value distributions and
-column correlations that the 22 queries depend on (date arithmetic between
-o_orderdate / l_shipdate / l_commitdate / l_receiptdate, returnflag/linestatus
-vs the 1995-06-17 current date, phone country codes, comment token patterns,
-p_name/p_type vocabularies, retailprice formula, customers without orders).
+column correlations the 22 queries depend on (date arithmetic between
+o_orderdate / l_shipdate / l_commitdate / l_receiptdate, returnflag /
+linestatus vs the 1995-06-17 current date, phone country codes, comment token
+patterns, p_name/p_type vocabularies, the retailprice formula, customers
+without orders).
 
-Supports sharded generation (rank/world) so each GPU materializes only its
-row range with globally consistent keys — cross-shard joins then exercise the
-RCCL exchange paths.
+Every random value is COUNTER-BASED: a splitmix64 hash of (row key, column
+salt) computed with wrapping int64 torch ops.  Rank r of world w generates
+exactly the rows of its shard and the union over ranks is bit-identical to a
+single-rank generation — so distributed correctness tests can compare against
+single-process runs, and scaling benchmarks see the same dataset at every N.
 
 This replaces the reference's external dbgen pipeline
 (benchmarking/tpch/data_generation.py): there is no network on the bench
-boxes, and generating in HBM is itself the fastest possible "scan".
+boxes, and generating straight into HBM is itself the fastest possible scan.
 """
 from __future__ import annotations
 
 import datetime as dt
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 
 from daft_amd import DataType, Series
 from daft_amd.recordbatch import RecordBatch
+from daft_amd.schema import Schema as _Schema
 
 EPOCH = dt.date(1970, 1, 1)
 STARTDATE = (dt.date(1992, 1, 1) - EPOCH).days
@@ -71,12 +76,61 @@ _WORDS = (
     "sleep haggle nag wake cajole detect integrate maintain lose use boost"
 ).split()
 
+_NATION_SCHEMA = _Schema.from_dict({
+    "n_nationkey": DataType.int64(), "n_name": DataType.string(),
+    "n_regionkey": DataType.int64(), "n_comment": DataType.string()})
+_REGION_SCHEMA = _Schema.from_dict({
+    "r_regionkey": DataType.int64(), "r_name": DataType.string(),
+    "r_comment": DataType.string()})
+
+
+# ---------------------------------------------------------------------------
+# counter-based randomness (wrapping int64 torch ops; shard-invariant)
+# ---------------------------------------------------------------------------
+
+def _i64(c: int) -> int:
+    return c - (1 << 64) if c >= (1 << 63) else c
+
+
+_C1 = _i64(0x9E3779B97F4A7C15)
+_C2 = _i64(0xBF58476D1CE4E5B9)
+_C3 = _i64(0x94D049BB133111EB)
+
+
+def _lshr(x: torch.Tensor, s: int) -> torch.Tensor:
+    return (x >> s) & ((1 << (64 - s)) - 1)
+
+
+def _mix(key: torch.Tensor, salt: int) -> torch.Tensor:
+    salt_c = _i64((salt * 0xD1B54A32D192ED03 + 0x2545F4914F6CDD1D)
+                  & ((1 << 64) - 1))
+    x = key * _C3 + salt_c
+    x = x + _C1
+    x = (x ^ _lshr(x, 30)) * _C2
+    x = (x ^ _lshr(x, 27)) * _C3
+    return x ^ _lshr(x, 31)
+
+
+def _u53(key: torch.Tensor, salt: int) -> torch.Tensor:
+    return _lshr(_mix(key, salt), 11)
+
+
+def _randint(key: torch.Tensor, salt: int, lo: int, hi: int) -> torch.Tensor:
+    """Uniform int64 in [lo, hi)."""
+    return lo + torch.remainder(_u53(key, salt), hi - lo)
+
+
+def _rand(key: torch.Tensor, salt: int) -> torch.Tensor:
+    """Uniform float64 in [0, 1)."""
+    return _u53(key, salt).to(torch.float64) * (1.0 / (1 << 53))
+
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
 
 def _comment_vocab(n_vocab: int, special_fraction: float, seed: int,
                    kind: str) -> list:
-    """Pre-generated comment strings; `special_fraction` of them contain
-    'special ... requests' (the Q13 pattern) / 'Customer ... Complaints'
-    (the Q16 pattern)."""
     import random
     rng = random.Random(seed)
     out = []
@@ -97,26 +151,6 @@ def _comment_vocab(n_vocab: int, special_fraction: float, seed: int,
     return out
 
 
-class _Rng:
-    """Deterministic per-column RNG on the target device."""
-
-    def __init__(self, device, seed: int):
-        self.device = torch.device(device)
-        self.seed = seed
-        gen_dev = self.device if self.device.type == "cuda" else "cpu"
-        self.gen = torch.Generator(device=gen_dev)
-        self.gen.manual_seed(seed)
-
-    def randint(self, lo: int, hi: int, n: int, dtype=torch.int64):
-        return torch.randint(lo, hi, (n,), generator=self.gen,
-                             device=self.gen.device, dtype=dtype) \
-            .to(self.device)
-
-    def rand(self, n: int):
-        return torch.rand(n, generator=self.gen, device=self.gen.device,
-                          dtype=torch.float64).to(self.device)
-
-
 def _vocab_series(name: str, words: list, codes: torch.Tensor) -> Series:
     vocab = Series.from_pylist(name, words, DataType.string(),
                                device=codes.device)
@@ -128,8 +162,7 @@ def _money(x: torch.Tensor) -> torch.Tensor:
 
 
 def _retailprice(partkey: torch.Tensor) -> torch.Tensor:
-    # TPC-H spec: (90000 + ((partkey/10) mod 20001) + 100*(partkey mod 1000))/100
-    pk = partkey.to(torch.float64)
+    # TPC-H: (90000 + ((partkey/10) mod 20001) + 100*(partkey mod 1000))/100
     return (90000.0
             + torch.remainder(torch.div(partkey, 10, rounding_mode="floor"),
                               20001).to(torch.float64)
@@ -138,50 +171,9 @@ def _retailprice(partkey: torch.Tensor) -> torch.Tensor:
 
 def _shard(total: int, rank: int, world: int) -> Tuple[int, int]:
     per = (total + world - 1) // world
-    lo = rank * per
+    lo = min(total, rank * per)
     hi = min(total, lo + per)
-    return lo, max(lo, hi)
-
-
-def gen_nation_region(device) -> Dict[str, RecordBatch]:
-    nation = RecordBatch.from_pydict({
-        "n_nationkey": list(range(25)),
-        "n_name": [n for n, _ in NATIONS],
-        "n_regionkey": [r for _, r in NATIONS],
-        "n_comment": [f"nation {n.lower()} commentary" for n, _ in NATIONS],
-    }, device=device)
-    region = RecordBatch.from_pydict({
-        "r_regionkey": list(range(5)),
-        "r_name": REGIONS,
-        "r_comment": [f"region {r.lower()}" for r in REGIONS],
-    }, device=device)
-    return {"nation": nation, "region": region}
-
-
-def gen_supplier(sf: float, device, rank=0, world=1) -> RecordBatch:
-    total = int(10_000 * sf)
-    lo, hi = _shard(total, rank, world)
-    n = hi - lo
-    r = _Rng(device, 101 + rank)
-    skey = torch.arange(lo + 1, hi + 1, device=device)
-    comments = _comment_vocab(512, 0.002, 7, "supplier")
-    return RecordBatch([
-        Series("s_suppkey", DataType.int64(), data=skey),
-        _supplier_name(skey),
-        _vocab_series("s_address", [f"addr{i}" for i in range(256)],
-                      r.randint(0, 256, n)),
-        Series("s_nationkey", DataType.int64(), data=r.randint(0, 25, n)),
-        _phone(r.randint(0, 25, n), r, "s_phone"),
-        Series("s_acctbal", DataType.float64(),
-               data=_money(r.rand(n) * 10999.98 - 999.99)),
-        _vocab_series("s_comment", comments, r.randint(0, len(comments), n)),
-    ], num_rows=n)
-
-
-def _supplier_name(skey: torch.Tensor) -> Series:
-    # 'Supplier#' + zero-padded key: build via small vocab of padded ints is
-    # impractical; keys are distinct -> format on device via digit gather
-    return _format_keyed("s_name", "Supplier#", skey)
+    return lo, hi
 
 
 def _format_keyed(name: str, prefix: str, keys: torch.Tensor) -> Series:
@@ -199,21 +191,19 @@ def _format_keyed(name: str, prefix: str, keys: torch.Tensor) -> Series:
     return strk.concat_str([pre] + digits).rename(name)
 
 
-def _phone(nationkey: torch.Tensor, r: _Rng, name: str) -> Series:
+def _phone(key: torch.Tensor, nationkey: torch.Tensor, salt: int,
+           name: str) -> Series:
     """'CC-NNN-NNN-NNNN' with country code 10+nationkey (Q22)."""
     dev = nationkey.device
+    from daft_amd.kernels import strings as strk
     n = nationkey.shape[0]
     cc = _vocab_series("cc", [str(10 + i) for i in range(25)], nationkey)
-    parts = [cc]
-    from daft_amd.kernels import strings as strk
     dash = Series.from_pylist("-", ["-"], DataType.string(),
                               device=dev).broadcast(n)
-    for ln in (3, 3, 4):
+    parts = [cc]
+    for si, ln in enumerate((3, 3, 4)):
         lo = 10 ** (ln - 1)
-        seg = r.randint(lo, 10 ** ln, n)
-        segs = _vocab_series("seg", [str(i) for i in range(10)],
-                             torch.zeros(1, dtype=torch.int64, device=dev))
-        # build numeric segment via digit concat
+        seg = _randint(key, salt + si, lo, 10 ** ln)
         digs = []
         for d in range(ln):
             div = 10 ** (ln - 1 - d)
@@ -226,70 +216,114 @@ def _phone(nationkey: torch.Tensor, r: _Rng, name: str) -> Series:
     return strk.concat_str(parts).rename(name)
 
 
+# ---------------------------------------------------------------------------
+# tables
+# ---------------------------------------------------------------------------
+
+def gen_nation_region(device, rank=0, world=1) -> Dict[str, RecordBatch]:
+    nlo, nhi = _shard(25, rank, world)
+    nation = RecordBatch.from_pydict({
+        "n_nationkey": list(range(nlo, nhi)),
+        "n_name": [NATIONS[i][0] for i in range(nlo, nhi)],
+        "n_regionkey": [NATIONS[i][1] for i in range(nlo, nhi)],
+        "n_comment": [f"nation {NATIONS[i][0].lower()} commentary"
+                      for i in range(nlo, nhi)],
+    }, device=device, schema=_NATION_SCHEMA)
+    rlo, rhi = _shard(5, rank, world)
+    region = RecordBatch.from_pydict({
+        "r_regionkey": list(range(rlo, rhi)),
+        "r_name": REGIONS[rlo:rhi],
+        "r_comment": [f"region {r.lower()}" for r in REGIONS[rlo:rhi]],
+    }, device=device, schema=_REGION_SCHEMA)
+    return {"nation": nation, "region": region}
+
+
+def gen_supplier(sf: float, device, rank=0, world=1) -> RecordBatch:
+    total = int(10_000 * sf)
+    lo, hi = _shard(total, rank, world)
+    n = hi - lo
+    skey = torch.arange(lo + 1, hi + 1, device=device)
+    comments = _comment_vocab(512, 0.002, 7, "supplier")
+    return RecordBatch([
+        Series("s_suppkey", DataType.int64(), data=skey),
+        _format_keyed("s_name", "Supplier#", skey),
+        _vocab_series("s_address", [f"addr{i}" for i in range(256)],
+                      _randint(skey, 101, 0, 256)),
+        Series("s_nationkey", DataType.int64(),
+               data=_randint(skey, 102, 0, 25)),
+        _phone(skey, _randint(skey, 102, 0, 25), 103, "s_phone"),
+        Series("s_acctbal", DataType.float64(),
+               data=_money(_rand(skey, 104) * 10999.98 - 999.99)),
+        _vocab_series("s_comment", comments,
+                      _randint(skey, 105, 0, len(comments))),
+    ], num_rows=n)
+
+
 def gen_part(sf: float, device, rank=0, world=1) -> RecordBatch:
     total = int(200_000 * sf)
     lo, hi = _shard(total, rank, world)
     n = hi - lo
-    r = _Rng(device, 202 + rank)
     pkey = torch.arange(lo + 1, hi + 1, device=device)
     from daft_amd.kernels import strings as strk
     sp = Series.from_pylist(" ", [" "], DataType.string(),
                             device=device).broadcast(n)
-    name_words = [_vocab_series("w", COLORS, r.randint(0, len(COLORS), n))
-                  for _ in range(5)]
+    name_words = [_vocab_series("w", COLORS,
+                                _randint(pkey, 201 + i, 0, len(COLORS)))
+                  for i in range(5)]
     p_name = strk.concat_str([name_words[0], sp, name_words[1], sp,
                               name_words[2], sp, name_words[3], sp,
                               name_words[4]]).rename("p_name")
-    m = r.randint(1, 6, n)
-    brand_n = r.randint(1, 6, n)
+    m = _randint(pkey, 206, 1, 6)
+    brand_n = _randint(pkey, 207, 1, 6)
     p_mfgr = _vocab_series("p_mfgr",
                            [f"Manufacturer#{i}" for i in range(1, 6)], m - 1)
     p_brand = _vocab_series(
         "p_brand", [f"Brand#{i}{j}" for i in range(1, 6) for j in range(1, 6)],
         (m - 1) * 5 + (brand_n - 1))
-    t1 = r.randint(0, 6, n)
-    t2 = r.randint(0, 5, n)
-    t3 = r.randint(0, 5, n)
     p_type = _vocab_series(
         "p_type", [f"{a} {b} {c}" for a in TYPE1 for b in TYPE2 for c in TYPE3],
-        t1 * 25 + t2 * 5 + t3)
-    c1 = r.randint(0, 5, n)
-    c2 = r.randint(0, 8, n)
+        _randint(pkey, 208, 0, 6) * 25 + _randint(pkey, 209, 0, 5) * 5 +
+        _randint(pkey, 210, 0, 5))
     p_container = _vocab_series(
         "p_container", [f"{a} {b}" for a in CONT1 for b in CONT2],
-        c1 * 8 + c2)
+        _randint(pkey, 211, 0, 5) * 8 + _randint(pkey, 212, 0, 8))
     return RecordBatch([
         Series("p_partkey", DataType.int64(), data=pkey),
         p_name, p_mfgr, p_brand, p_type,
-        Series("p_size", DataType.int64(), data=r.randint(1, 51, n)),
+        Series("p_size", DataType.int64(), data=_randint(pkey, 213, 1, 51)),
         p_container,
         Series("p_retailprice", DataType.float64(),
                data=_money(_retailprice(pkey))),
-        _vocab_series("p_comment", _WORDS, r.randint(0, len(_WORDS), n))
-        .rename("p_comment"),
+        _vocab_series("p_comment", _WORDS,
+                      _randint(pkey, 214, 0, len(_WORDS))),
     ], num_rows=n)
+
+
+def _ps_suppkey(pkey: torch.Tensor, j: torch.Tensor, S: int) -> torch.Tensor:
+    # TPC-H spec supplier spread for partsupp
+    return torch.remainder(
+        pkey + j * (S // 4 + torch.div(pkey - 1, S, rounding_mode="floor")),
+        S) + 1
 
 
 def gen_partsupp(sf: float, device, rank=0, world=1) -> RecordBatch:
     total_parts = int(200_000 * sf)
-    n_supp = int(10_000 * sf)
+    n_supp = max(int(10_000 * sf), 1)
     lo, hi = _shard(total_parts, rank, world)
     n = (hi - lo) * 4
-    r = _Rng(device, 303 + rank)
     pkey = torch.arange(lo + 1, hi + 1, device=device).repeat_interleave(4)
     j = torch.arange(n, device=device) % 4
-    # spec supplier spread: (partkey + j*(S/4 + (partkey-1)/S)) mod S + 1
-    S = max(n_supp, 1)
-    skey = torch.remainder(
-        pkey + j * (S // 4 + torch.div(pkey - 1, S, rounding_mode="floor")),
-        S) + 1
+    pskey = pkey * 4 + j  # row key
+    skey = _ps_suppkey(pkey, j, n_supp)
     return RecordBatch([
         Series("ps_partkey", DataType.int64(), data=pkey),
         Series("ps_suppkey", DataType.int64(), data=skey),
-        Series("ps_availqty", DataType.int64(), data=r.randint(1, 10_000, n)),
+        Series("ps_availqty", DataType.int64(),
+               data=_randint(pskey, 301, 1, 10_000)),
         Series("ps_supplycost", DataType.float64(),
-               data=_money(r.rand(n) * 999.0 + 1.0)),
-        _vocab_series("ps_comment", _WORDS, r.randint(0, len(_WORDS), n)),
+               data=_money(_rand(pskey, 302) * 999.0 + 1.0)),
+        _vocab_series("ps_comment", _WORDS,
+                      _randint(pskey, 303, 0, len(_WORDS))),
     ], num_rows=n)
 
 
@@ -297,78 +331,69 @@ def gen_customer(sf: float, device, rank=0, world=1) -> RecordBatch:
     total = int(150_000 * sf)
     lo, hi = _shard(total, rank, world)
     n = hi - lo
-    r = _Rng(device, 404 + rank)
     ckey = torch.arange(lo + 1, hi + 1, device=device)
-    nk = r.randint(0, 25, n)
+    nk = _randint(ckey, 401, 0, 25)
     return RecordBatch([
         Series("c_custkey", DataType.int64(), data=ckey),
         _format_keyed("c_name", "Customer#", ckey),
         _vocab_series("c_address", [f"addr{i}" for i in range(256)],
-                      r.randint(0, 256, n)),
+                      _randint(ckey, 402, 0, 256)),
         Series("c_nationkey", DataType.int64(), data=nk),
-        _phone(nk, r, "c_phone"),
+        _phone(ckey, nk, 403, "c_phone"),
         Series("c_acctbal", DataType.float64(),
-               data=_money(r.rand(n) * 10999.98 - 999.99)),
-        _vocab_series("c_mktsegment", SEGMENTS, r.randint(0, 5, n)),
+               data=_money(_rand(ckey, 406) * 10999.98 - 999.99)),
+        _vocab_series("c_mktsegment", SEGMENTS, _randint(ckey, 407, 0, 5)),
         _vocab_series("c_comment", _comment_vocab(256, 0.0, 11, "none"),
-                      r.randint(0, 256, n)),
+                      _randint(ckey, 408, 0, 256)),
     ], num_rows=n)
 
 
 def gen_orders_lineitem(sf: float, device, rank=0, world=1
                         ) -> Tuple[RecordBatch, RecordBatch]:
     total_orders = int(1_500_000 * sf)
-    n_cust = int(150_000 * sf)
+    n_cust = max(int(150_000 * sf), 1)
     lo, hi = _shard(total_orders, rank, world)
     n = hi - lo
-    r = _Rng(device, 505 + rank)
 
     okey = torch.arange(lo + 1, hi + 1, device=device)
-    # spec: only custkeys not divisible by 3 place orders (Q22 needs
-    # customers without orders)
-    raw = r.randint(0, max(n_cust * 2 // 3, 1), n)
-    # map dense index -> custkeys not divisible by 3 (1,2,4,5,7,8,...)
-    ckey = raw + torch.div(raw, 2, rounding_mode="floor") + 1
-    ckey = torch.clamp(ckey, max=max(n_cust, 1))
-    odate = r.randint(STARTDATE, ENDDATE_ORDER + 1, n, dtype=torch.int64)
+    # spec: only custkeys not divisible by 3 place orders (Q22 relies on
+    # customers without orders); map a dense index to 1,2,4,5,7,8,...
+    raw = _randint(okey, 501, 0, max(n_cust * 2 // 3, 1))
+    ckey = torch.clamp(raw + torch.div(raw, 2, rounding_mode="floor") + 1,
+                       max=n_cust)
+    odate = _randint(okey, 502, STARTDATE, ENDDATE_ORDER + 1)
 
-    nlines = r.randint(1, 8, n)
+    nlines = _randint(okey, 503, 1, 8)
     total_lines = int(nlines.sum().item())
     order_row = torch.repeat_interleave(
         torch.arange(n, device=device), nlines)
-    lr = _Rng(device, 606 + rank)
     m = total_lines
     l_okey = okey[order_row]
     l_odate = odate[order_row]
-    n_part = int(200_000 * sf)
-    n_supp = int(10_000 * sf)
-    l_pkey = lr.randint(1, max(n_part, 1) + 1, m)
-    # suppkey must be one of the 4 partsupp suppliers for this part (Q9/Q20
-    # join partsupp on both keys)
-    j = lr.randint(0, 4, m)
-    S = max(n_supp, 1)
-    l_skey = torch.remainder(
-        l_pkey + j * (S // 4 + torch.div(l_pkey - 1, S, rounding_mode="floor")),
-        S) + 1
-    qty = lr.randint(1, 51, m).to(torch.float64)
-    extprice = _money(qty * _retailprice(l_pkey))
-    disc = torch.round(lr.rand(m) * 10) / 100.0          # 0.00..0.10
-    tax = torch.round(lr.rand(m) * 8) / 100.0            # 0.00..0.08
-    shipdate = l_odate + lr.randint(1, 122, m)
-    commitdate = l_odate + lr.randint(30, 91, m)
-    receiptdate = shipdate + lr.randint(1, 31, m)
-    shipped = receiptdate <= CURRENTDATE
-    rf_code = torch.where(
-        shipped,
-        lr.randint(0, 2, m),                  # 'R' or 'A'
-        torch.full((m,), 2, dtype=torch.int64, device=device))  # 'N'
-    linestatus_code = (shipdate > CURRENTDATE).to(torch.int64)  # 0='F',1='O'
 
-    # line numbers within order: position - first position of that order
-    first_pos = torch.zeros(n, dtype=torch.int64, device=device)
-    torch.cumsum(nlines, 0, out=first_pos)
-    first_pos = first_pos - nlines
+    # line numbers within order
+    first_pos = torch.cumsum(nlines, 0) - nlines
     linenumber = torch.arange(m, device=device) - first_pos[order_row] + 1
+    lkey = l_okey * 8 + linenumber  # shard-invariant row key
+
+    n_part = max(int(200_000 * sf), 1)
+    n_supp = max(int(10_000 * sf), 1)
+    l_pkey = _randint(lkey, 601, 1, n_part + 1)
+    # suppkey must be one of the part's 4 partsupp suppliers (Q9/Q20)
+    j = _randint(lkey, 602, 0, 4)
+    l_skey = _ps_suppkey(l_pkey, j, n_supp)
+    qty = _randint(lkey, 603, 1, 51).to(torch.float64)
+    extprice = _money(qty * _retailprice(l_pkey))
+    disc = _randint(lkey, 604, 0, 11).to(torch.float64) / 100.0
+    tax = _randint(lkey, 605, 0, 9).to(torch.float64) / 100.0
+    shipdate = l_odate + _randint(lkey, 606, 1, 122)
+    commitdate = l_odate + _randint(lkey, 607, 30, 91)
+    receiptdate = shipdate + _randint(lkey, 608, 1, 31)
+    shipped = receiptdate <= CURRENTDATE
+    rf_code = torch.where(shipped, _randint(lkey, 609, 0, 2),
+                          torch.full((m,), 2, dtype=torch.int64,
+                                     device=device))
+    linestatus_code = (shipdate > CURRENTDATE).to(torch.int64)
 
     lineitem = RecordBatch([
         Series("l_orderkey", DataType.int64(), data=l_okey),
@@ -381,16 +406,16 @@ def gen_orders_lineitem(sf: float, device, rank=0, world=1
         Series("l_tax", DataType.float64(), data=tax),
         _vocab_series("l_returnflag", ["R", "A", "N"], rf_code),
         _vocab_series("l_linestatus", ["F", "O"], linestatus_code),
-        Series("l_shipdate", DataType.date(),
-               data=shipdate.to(torch.int32)),
+        Series("l_shipdate", DataType.date(), data=shipdate.to(torch.int32)),
         Series("l_commitdate", DataType.date(),
                data=commitdate.to(torch.int32)),
         Series("l_receiptdate", DataType.date(),
                data=receiptdate.to(torch.int32)),
-        _vocab_series("l_shipinstruct", INSTRUCTS, lr.randint(0, 4, m)),
-        _vocab_series("l_shipmode", SHIPMODES, lr.randint(0, 7, m)),
+        _vocab_series("l_shipinstruct", INSTRUCTS,
+                      _randint(lkey, 610, 0, 4)),
+        _vocab_series("l_shipmode", SHIPMODES, _randint(lkey, 611, 0, 7)),
         _vocab_series("l_comment", _comment_vocab(256, 0.0, 13, "none"),
-                      lr.randint(0, 256, m)),
+                      _randint(lkey, 612, 0, 256)),
     ], num_rows=m)
 
     # order status from line statuses; totalprice from line charges
@@ -413,11 +438,14 @@ def gen_orders_lineitem(sf: float, device, rank=0, world=1
         _vocab_series("o_orderstatus", ["F", "O", "P"], status_code),
         Series("o_totalprice", DataType.float64(), data=_money(totalprice)),
         Series("o_orderdate", DataType.date(), data=odate.to(torch.int32)),
-        _vocab_series("o_orderpriority", PRIORITIES, r.randint(0, 5, n)),
-        _format_keyed("o_clerk", "Clerk#", r.randint(1, max(int(1000 * sf), 2), n)),
+        _vocab_series("o_orderpriority", PRIORITIES,
+                      _randint(okey, 504, 0, 5)),
+        _format_keyed("o_clerk", "Clerk#",
+                      _randint(okey, 505, 1, max(int(1000 * sf), 2))),
         Series("o_shippriority", DataType.int64(),
                data=torch.zeros(n, dtype=torch.int64, device=device)),
-        _vocab_series("o_comment", ocomments, r.randint(0, len(ocomments), n)),
+        _vocab_series("o_comment", ocomments,
+                      _randint(okey, 506, 0, len(ocomments))),
     ], num_rows=n)
     return orders, lineitem
 
@@ -425,7 +453,7 @@ def gen_orders_lineitem(sf: float, device, rank=0, world=1
 def generate(sf: float, device="cpu", rank: int = 0,
              world: int = 1) -> Dict[str, RecordBatch]:
     """Generate all TPC-H tables (this rank's shard) at scale factor `sf`."""
-    out = gen_nation_region(device)
+    out = gen_nation_region(device, rank, world)
     out["supplier"] = gen_supplier(sf, device, rank, world)
     out["part"] = gen_part(sf, device, rank, world)
     out["partsupp"] = gen_partsupp(sf, device, rank, world)

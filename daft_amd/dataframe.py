@@ -304,12 +304,20 @@ class DataFrame:
         df = self.collect()
         parts = df._result
         if not parts:
-            return {f.name: [] for f in self.schema}
-        out: Dict[str, list] = {f.name: [] for f in parts[0].schema}
-        for p in parts:
-            d = p.to_pydict()
-            for k, v in d.items():
-                out[k].extend(v)
+            out: Dict[str, list] = {f.name: [] for f in self.schema}
+        else:
+            out = {f.name: [] for f in parts[0].schema}
+            for p in parts:
+                d = p.to_pydict()
+                for k, v in d.items():
+                    out[k].extend(v)
+        # under SPMD execution the result is sharded across ranks: gather so
+        # every rank observes the full result (scalar subqueries rely on it)
+        from .distributed import comm
+        if comm.is_dist():
+            from .distributed.runner import DistributedRunner
+            if isinstance(get_context().runner(), DistributedRunner):
+                out = comm.gather_pydict(out)
         return out
 
     def to_pylist(self) -> List[Dict[str, Any]]:

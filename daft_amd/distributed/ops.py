@@ -1,0 +1,142 @@
+"""Physical exchange operators (RCCL over xGMI; gloo on the CPU test tier)."""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from ..expressions.expressions import ExprNode
+from ..kernels import rowops
+from ..physical.ops import BatchIter, PhysicalOp
+from ..recordbatch import RecordBatch, _range_partition_ids
+from ..schema import DataType, Schema, TypeKind
+from ..series import Series
+from . import comm
+
+
+def _normalize_key(s: Series) -> Series:
+    """Cast keys to width-stable dtypes so both sides of a co-partition hash
+    identically (int->int64, float->float64, bool->int64)."""
+    dt = s.dtype
+    if dt.is_integer() or dt.is_boolean() or dt.is_temporal():
+        return s.cast(DataType.int64()) if s.data is not None and \
+            s.data.dtype != torch.int64 else s
+    if dt.kind == TypeKind.FLOAT32:
+        return s.cast(DataType.float64())
+    if dt.is_decimal():
+        return s.cast(DataType.float64())
+    return s
+
+
+class ExchangeByKeyOp(PhysicalOp):
+    def __init__(self, child: PhysicalOp, keys: List[ExprNode]):
+        super().__init__([child], child.schema, "ExchangeByKey[RCCL a2a]")
+        self.keys = keys
+
+    def execute(self, ectx) -> BatchIter:
+        batch = self._materialize_child(ectx)
+        w = comm.world()
+        if w == 1:
+            yield batch
+            return
+        key_series = [_normalize_key(e.evaluate(batch)) for e in self.keys]
+        if len(batch) == 0:
+            parts = [batch] * w
+        else:
+            perm, counts = rowops.partition_by_hash(key_series, w)
+            reordered = batch.take(perm)
+            parts, start = [], 0
+            for c in counts.tolist():
+                parts.append(reordered.slice(start, start + c))
+                start += c
+        yield comm.exchange_batches(parts)
+
+
+class GatherToRank0Op(PhysicalOp):
+    def __init__(self, child: PhysicalOp):
+        super().__init__([child], child.schema, "GatherToRank0")
+
+    def execute(self, ectx) -> BatchIter:
+        batch = self._materialize_child(ectx)
+        w = comm.world()
+        if w == 1:
+            yield batch
+            return
+        empty = batch.slice(0, 0)
+        parts = [batch if p == 0 else empty for p in range(w)]
+        yield comm.exchange_batches(parts)
+
+
+class ReplicateAllOp(PhysicalOp):
+    def __init__(self, child: PhysicalOp):
+        super().__init__([child], child.schema, "ReplicateAll[allgather]")
+
+    def execute(self, ectx) -> BatchIter:
+        batch = self._materialize_child(ectx)
+        yield comm.allgather_batch(batch, ectx.device)
+
+
+class RangeExchangeOp(PhysicalOp):
+    """Sample keys -> allgather -> boundaries -> range partition -> a2a
+    (ref: daft-distributed pipeline_node/sort.rs:84-130 sample phase)."""
+
+    SAMPLES_PER_RANK = 256
+
+    def __init__(self, child: PhysicalOp, by: List[ExprNode],
+                 descending: List[bool], nulls_first: List[bool]):
+        super().__init__([child], child.schema, "RangeExchange")
+        self.by = by
+        self.descending = descending
+        self.nulls_first = nulls_first
+
+    def execute(self, ectx) -> BatchIter:
+        batch = self._materialize_child(ectx)
+        w = comm.world()
+        if w == 1:
+            yield batch
+            return
+        n = len(batch)
+        key_cols = [e.evaluate(batch).rename(f"__k{i}")
+                    for i, e in enumerate(self.by)]
+        keys_rb = RecordBatch(key_cols, num_rows=n)
+        # sample
+        k = min(n, self.SAMPLES_PER_RANK)
+        if k > 0:
+            step = max(1, n // k)
+            idx = torch.arange(0, n, step, dtype=torch.int64,
+                               device=batch.device)[:k]
+            sample = keys_rb.take(idx)
+        else:
+            sample = keys_rb.slice(0, 0)
+        all_samples = comm.allgather_batch(sample, ectx.device)
+        m = len(all_samples)
+        if m == 0:
+            yield comm.exchange_batches(
+                [batch if p == comm.rank() else batch.slice(0, 0)
+                 for p in range(w)])
+            return
+        names = [c.name for c in key_cols]
+        sorted_samples = all_samples.sort(names, self.descending,
+                                          self.nulls_first)
+        bidx = torch.tensor([min(m - 1, ((i + 1) * m) // w)
+                             for i in range(w - 1)], dtype=torch.int64,
+                            device=batch.device)
+        boundaries = sorted_samples.take(bidx)
+        part = _range_partition_ids(keys_rb, names, boundaries,
+                                    self.descending)
+        perm, counts = rowops.partition_by_value(part, w)
+        reordered = batch.take(perm)
+        parts, start = [], 0
+        for c in counts.tolist():
+            parts.append(reordered.slice(start, start + c))
+            start += c
+        yield comm.exchange_batches(parts)
+
+
+class Rank0OnlyOp(PhysicalOp):
+    def __init__(self, child: PhysicalOp):
+        super().__init__([child], child.schema, "Rank0Only")
+
+    def execute(self, ectx) -> BatchIter:
+        batch = self._materialize_child(ectx)
+        yield batch if comm.rank() == 0 else batch.slice(0, 0)

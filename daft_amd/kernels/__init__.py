@@ -19,6 +19,7 @@ from ..series import Series, full_null
 
 _native = None
 _native_err: Optional[str] = None
+_SIGNFLIP64 = -0x8000000000000000  # order-preserving u64 -> i64 bit flip
 
 
 def load_native():
@@ -405,7 +406,21 @@ def compare_op(l: Series, r: Series, op: str) -> Series:
         return _string_compare(l, r, op, validity)
     st = supertype(l.dtype, r.dtype)
     tdt = st.to_torch()
-    a, b = l.data.to(tdt), r.data.to(tdt)
+
+    def conv(t: torch.Tensor) -> torch.Tensor:
+        # unsigned wide ints lack cuda/cpu comparison kernels in torch:
+        # compare through an order-preserving signed view
+        if tdt == torch.uint64:
+            v = t.view(torch.int64) if t.dtype == torch.uint64 \
+                else t.to(torch.int64)
+            return v ^ _SIGNFLIP64
+        if tdt in (torch.uint16, torch.uint32):
+            src = {torch.uint16: torch.int16,
+                   torch.uint32: torch.int32}.get(t.dtype)
+            return (t.view(src) if src is not None else t).to(torch.int64) & \
+                ((1 << (16 if tdt == torch.uint16 else 32)) - 1)
+        return t.to(tdt)
+    a, b = conv(l.data), conv(r.data)
     out = _COMPARE_TORCH[op](a, b)
     return Series(l.name, DataType.bool(), data=out, validity=validity)
 

@@ -9,6 +9,20 @@ from . import ops
 def translate(plan: lp.LogicalPlan) -> ops.PhysicalOp:
     ch = [translate(c) for c in plan.children]
 
+    from ..distributed import plan_nodes as dn
+    from ..distributed import ops as dops
+    if isinstance(plan, dn.ExchangeByKey):
+        return dops.ExchangeByKeyOp(ch[0], plan.keys)
+    if isinstance(plan, dn.GatherToRank0):
+        return dops.GatherToRank0Op(ch[0])
+    if isinstance(plan, dn.ReplicateAll):
+        return dops.ReplicateAllOp(ch[0])
+    if isinstance(plan, dn.Rank0Only):
+        return dops.Rank0OnlyOp(ch[0])
+    if isinstance(plan, dn.RangeExchange):
+        return dops.RangeExchangeOp(ch[0], plan.by, plan.descending,
+                                    plan.nulls_first)
+
     if isinstance(plan, lp.Source):
         return ops.InMemorySourceOp(plan.schema, plan.cache_key)
     if isinstance(plan, lp.ScanSource):
@@ -57,7 +71,9 @@ def translate(plan: lp.LogicalPlan) -> ops.PhysicalOp:
         return ops.SampleOp(ch[0], plan.fraction, plan.with_replacement,
                             plan.seed)
     if isinstance(plan, lp.MonotonicallyIncreasingId):
-        return ops.MonotonicIdOp(ch[0], plan.column_name, plan.schema)
+        from ..distributed import comm
+        return ops.MonotonicIdOp(ch[0], plan.column_name, plan.schema,
+                                 partition_id=comm.rank())
     if isinstance(plan, lp.IntoBatches):
         return ops.IntoBatchesOp(ch[0], plan.batch_size)
     if isinstance(plan, lp.Window):

@@ -1,0 +1,199 @@
+"""Multi-process SPMD tests: gloo backend, world_size=2, CPU — validates the
+distributed planner + exchange paths that run over RCCL on GPU boxes
+(ref test pattern: the reference's LocalSwordfishWorker fake cluster,
+src/daft-distributed/src/scheduling/local_worker.rs)."""
+import multiprocessing as mp
+import os
+import pickle
+import socket
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.distributed
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _run_worker(rank, world, port, fn_name, conn):
+    try:
+        import torch.distributed as dist
+        dist.init_process_group(
+            backend="gloo", rank=rank, world_size=world,
+            init_method=f"tcp://127.0.0.1:{port}")
+        from daft_amd.context import get_context
+        from daft_amd.distributed.runner import DistributedRunner
+        ctx = get_context()
+        ctx.set_runner(DistributedRunner(ctx))
+        fn = globals()[fn_name]
+        out = fn(rank, world)
+        conn.send(("ok", out))
+        dist.destroy_process_group()
+    except Exception as e:
+        import traceback
+        conn.send(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+def _spawn(fn_name, world=2):
+    ctx = mp.get_context("spawn")
+    port = _free_port()
+    procs, conns = [], []
+    for r in range(world):
+        parent, child = ctx.Pipe()
+        p = ctx.Process(target=_run_worker,
+                        args=(r, world, port, fn_name, child))
+        p.start()
+        procs.append(p)
+        conns.append(parent)
+    results = []
+    for p, c in zip(procs, conns):
+        status, payload = c.recv() if c.poll(180) else ("err", "timeout")
+        p.join(timeout=30)
+        if status == "err":
+            for q in procs:
+                if q.is_alive():
+                    q.terminate()
+            raise AssertionError(payload)
+        results.append(payload)
+    return results
+
+
+# ---------------------------------------------------------------------------
+# worker bodies (run under SPMD; each asserts and returns a summary)
+# ---------------------------------------------------------------------------
+
+def _shard_df(data: dict, rank: int, world: int):
+    import daft_amd as daft
+    n = len(next(iter(data.values())))
+    per = (n + world - 1) // world
+    lo, hi = rank * per, min(n, (rank + 1) * per)
+    return daft.from_pydict({k: v[lo:hi] for k, v in data.items()})
+
+
+def _core_checks(rank, world):
+    import daft_amd as daft
+    from daft_amd import col
+
+    data = {
+        "k": [1, 2, 3, 4, 1, 2, 3, 4, 5, 6],
+        "g": ["a", "b", "a", "b", "a", "b", "a", "b", "a", "b"],
+        "v": [1.0, 2.0, 3.0, 4.0, 5.0, 6.0, 7.0, 8.0, 9.0, 10.0],
+    }
+    df = _shard_df(data, rank, world)
+
+    # grouped agg (partial/final two-phase)
+    out = df.groupby("g").agg(
+        col("v").sum().alias("s"), col("v").count().alias("c"),
+        col("v").mean().alias("m"), col("v").max().alias("mx"),
+    ).sort("g").to_pydict()
+    assert out["g"] == ["a", "b"], out
+    assert out["s"] == [25.0, 30.0]
+    assert out["c"] == [5, 5]
+    assert out["m"] == [5.0, 6.0]
+    assert out["mx"] == [9.0, 10.0]
+
+    # ungrouped agg
+    tot = df.agg(col("v").sum().alias("s"),
+                 col("v").mean().alias("m")).to_pydict()
+    assert tot["s"] == [55.0] and tot["m"] == [5.5]
+
+    # count_distinct (row-exchange fallback)
+    nd = df.groupby("g").agg(col("k").count_distinct().alias("nd")) \
+        .sort("g").to_pydict()
+    assert nd["nd"] == [3, 3], nd
+
+    # join across shards
+    dim = _shard_df({"k": [1, 2, 3, 4, 5, 6],
+                     "name": ["one", "two", "three", "four", "five", "six"]},
+                    rank, world)
+    j = df.join(dim, on="k").groupby("name") \
+        .agg(col("v").sum().alias("s")).sort("name").to_pydict()
+    assert j["name"] == ["five", "four", "one", "six", "three", "two"]
+    assert j["s"] == [9.0, 12.0, 6.0, 10.0, 10.0, 8.0]
+
+    # global sort
+    s = df.sort("v", desc=True).to_pydict()
+    assert s["v"] == sorted(data["v"], reverse=True)
+
+    # topn + limit
+    t = df.sort("v", desc=True).limit(3).to_pydict()
+    assert t["v"] == [10.0, 9.0, 8.0]
+
+    # distinct
+    d = df.select("g").distinct().sort("g").to_pydict()
+    assert d["g"] == ["a", "b"]
+
+    # count_rows
+    assert df.count_rows() == 10
+
+    # semi/anti joins
+    small = _shard_df({"k": [1, 6]}, rank, world)
+    semi = df.join(small, on="k", how="semi").sort("v").to_pydict()
+    assert semi["k"] == [1, 1, 6]
+    anti = df.join(small, on="k", how="anti").count_rows()
+    assert anti == 7
+
+    # monotonically increasing id: unique across ranks
+    ids = df.add_monotonically_increasing_id("id").to_pydict()["id"]
+    assert len(set(ids)) == 10
+
+    # window over partitions
+    from daft_amd.window import Window
+    from daft_amd.functions import row_number
+    w = Window().partition_by("g").order_by("v")
+    wout = df.with_window_columns({"rn": row_number().over(w)}) \
+        .sort(["g", "v"]).to_pydict()
+    assert wout["rn"] == [1, 2, 3, 4, 5] * 2
+    return "ok"
+
+
+def _tpch_checks(rank, world):
+    from benchmarks.tpch import datagen, queries
+    sf = 0.01
+    T = datagen.dataframes(sf, device="cpu", rank=rank, world=world)
+    results = {}
+    for qi in (1, 3, 4, 5, 6, 10, 12, 13, 14, 18, 19, 22):
+        out = queries.run_query(qi, T, sf=sf).to_pydict()
+        results[qi] = out
+    return pickle.dumps(results)
+
+
+def test_distributed_core_ops():
+    res = _spawn("_core_checks", world=2)
+    assert res == ["ok", "ok"]
+
+
+def test_distributed_tpch_matches_single():
+    res = _spawn("_tpch_checks", world=2)
+    dist_results = pickle.loads(res[0])
+    # both ranks must agree (results gathered on every rank)
+    assert pickle.loads(res[1]).keys() == dist_results.keys()
+
+    import math
+
+    def norm_rows(d):
+        rows = list(zip(*d.values()))
+        key = lambda r: tuple(repr(x) for x in r
+                              if not isinstance(x, float))
+        return sorted(rows, key=key)
+
+    from benchmarks.tpch import datagen, queries
+    T = datagen.dataframes(0.01, device="cpu")
+    for qi, got in dist_results.items():
+        want = queries.run_query(qi, T, sf=0.01).to_pydict()
+        assert list(got.keys()) == list(want.keys()), f"q{qi} columns"
+        g_rows, w_rows = norm_rows(got), norm_rows(want)
+        assert len(g_rows) == len(w_rows), f"q{qi} row count"
+        for gr, wr in zip(g_rows, w_rows):
+            for gx, wx in zip(gr, wr):
+                if isinstance(wx, float):
+                    assert math.isclose(gx, wx, rel_tol=1e-9, abs_tol=1e-6), \
+                        f"q{qi}: {gx} != {wx}"
+                else:
+                    assert gx == wx, f"q{qi}: {gx!r} != {wx!r}"
