@@ -1,0 +1,122 @@
+"""SQL frontend tests: basic SELECT coverage + all supported TPC-H SQL
+queries vs the DataFrame implementations (ref pattern: tests/sql/)."""
+import math
+
+import pytest
+
+import daft_amd as daft
+from daft_amd import col
+from daft_amd.sql import sql
+
+from benchmarks.tpch import datagen, queries, queries_sql
+
+SF = 0.01
+
+
+@pytest.fixture(scope="module")
+def T():
+    return datagen.dataframes(SF, device="cpu")
+
+
+def test_sql_basic_select():
+    df = daft.from_pydict({"a": [1, 2, 3], "b": ["x", "y", "z"]})
+    out = sql("select a + 1 as a1, b from df where a >= 2 order by a1 desc")
+    assert out.to_pydict() == {"a1": [4, 3], "b": ["z", "y"]}
+
+
+def test_sql_aggregate():
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1.0, 2.0, 3.0]})
+    out = sql("select g, sum(v) as s, count(*) as n from df "
+              "group by g order by g")
+    assert out.to_pydict() == {"g": ["a", "b"], "s": [3.0, 3.0], "n": [2, 1]}
+
+
+def test_sql_join():
+    a = daft.from_pydict({"k": [1, 2, 3], "v": [10, 20, 30]})
+    b = daft.from_pydict({"k": [2, 3, 4], "w": [200, 300, 400]})
+    out = sql("select a.k, v, w from a join b on a.k = b.k order by a.k")
+    assert out.to_pydict() == {"k": [2, 3], "v": [20, 30], "w": [200, 300]}
+
+
+def test_sql_comma_join_extraction():
+    a = daft.from_pydict({"ak": [1, 2], "v": [10, 20]})
+    b = daft.from_pydict({"bk": [1, 2], "w": [100, 200]})
+    out = sql("select v, w from a, b where ak = bk and w > 100")
+    assert out.to_pydict() == {"v": [20], "w": [200]}
+
+
+def test_sql_case_when():
+    df = daft.from_pydict({"x": [1, 5, 10]})
+    out = sql("select case when x < 3 then 'lo' when x < 7 then 'mid' "
+              "else 'hi' end as c from df")
+    assert out.to_pydict()["c"] == ["lo", "mid", "hi"]
+
+
+def test_sql_having():
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1, 2, 3]})
+    out = sql("select g, sum(v) as s from df group by g having sum(v) > 2 "
+              "order by g")
+    assert out.to_pydict() == {"g": ["a", "b"], "s": [3, 3]}
+
+
+def test_sql_in_subquery():
+    a = daft.from_pydict({"k": [1, 2, 3, 4]})
+    b = daft.from_pydict({"k": [2, 4]})
+    out = sql("select k from a where k in (select k from b) order by k")
+    assert out.to_pydict()["k"] == [2, 4]
+    out = sql("select k from a where k not in (select k from b) order by k")
+    assert out.to_pydict()["k"] == [1, 3]
+
+
+def test_sql_exists_correlated():
+    a = daft.from_pydict({"k": [1, 2, 3]})
+    b = daft.from_pydict({"fk": [2, 3, 3]})
+    out = sql("select k from a where exists "
+              "(select * from b where b.fk = a.k) order by k")
+    assert out.to_pydict()["k"] == [2, 3]
+    out = sql("select k from a where not exists "
+              "(select * from b where b.fk = a.k) order by k")
+    assert out.to_pydict()["k"] == [1]
+
+
+def test_sql_scalar_subquery():
+    a = daft.from_pydict({"v": [1, 2, 3, 4]})
+    out = sql("select v from a where v > (select avg(v) from a) order by v")
+    assert out.to_pydict()["v"] == [3, 4]
+
+
+def test_sql_cte():
+    df = daft.from_pydict({"v": [1, 2, 3]})
+    out = sql("with t as (select v * 2 as w from df) "
+              "select w from t where w > 2 order by w")
+    assert out.to_pydict()["w"] == [4, 6]
+
+
+def test_sql_expr():
+    from daft_amd.sql import sql_expr
+    df = daft.from_pydict({"a": [1, 2, 3]})
+    out = df.where(sql_expr("a >= 2")).to_pydict()
+    assert out["a"] == [2, 3]
+
+
+def _norm_rows(d):
+    rows = list(zip(*d.values()))
+    key = lambda r: tuple(repr(x) for x in r if not isinstance(x, float))
+    return sorted(rows, key=key)
+
+
+@pytest.mark.parametrize("qi", queries_sql.SUPPORTED)
+def test_tpch_sql_matches_dataframe(qi, T):
+    got = queries_sql.run_sql_query(qi, T, sf=SF).to_pydict()
+    want = queries.run_query(qi, T, sf=SF).to_pydict()
+    assert len(got) == len(want), (list(got), list(want))
+    g_rows, w_rows = _norm_rows(got), _norm_rows(want)
+    assert len(g_rows) == len(w_rows), f"rows {len(g_rows)} vs {len(w_rows)}"
+    for gr, wr in zip(g_rows, w_rows):
+        for gx, wx in zip(gr, wr):
+            if isinstance(wx, float):
+                assert gx is not None and (
+                    math.isclose(gx, wx, rel_tol=1e-9, abs_tol=1e-6) or
+                    (math.isnan(gx) and math.isnan(wx))), (gx, wx)
+            else:
+                assert gx == wx, (gx, wx)
