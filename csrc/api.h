@@ -52,6 +52,15 @@ Tensor string_chunk_key(Tensor offsets, Tensor bytes, int64_t chunk);
 // partition
 Tensor u64_mod(Tensor hashes, int64_t n_partitions);
 
+// multimodal (multimodal.hip)
+Tensor minhash(Tensor offsets, Tensor bytes, int64_t num_hashes,
+               int64_t ngram_size, Tensor perm_a, Tensor perm_b);
+Tensor hll_update(Tensor hashes, Tensor gids, Tensor valid,
+                  int64_t num_groups);
+Tensor image_resize(Tensor src_bytes, Tensor src_off, Tensor src_h,
+                    Tensor src_w, int64_t channels, int64_t out_h,
+                    int64_t out_w);
+
 // strings (strings.hip)
 Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
 Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,

@@ -17,6 +17,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("string_chunk_key", &string_chunk_key,
         "big-endian 8-byte chunk keys for string sorting");
   m.def("u64_mod", &u64_mod, "unsigned modulo for hash partitioning");
+  m.def("minhash", &minhash, "word-ngram MinHash signatures (wave/row)");
+  m.def("hll_update", &hll_update, "HyperLogLog register updates");
+  m.def("image_resize", &image_resize, "bilinear uint8 HWC resize");
   m.def("str_find", &str_find, "substring find/prefix/suffix");
   m.def("str_like", &str_like, "ordered multi-substring LIKE");
   m.def("str_case", &str_case, "ASCII upper/lower");

@@ -703,6 +703,14 @@ class Expression:
         from ..functions.image import ImageNamespace
         return ImageNamespace(self._node)
 
+    @property
+    def url(self) -> "UrlNamespace":
+        return UrlNamespace(self._node)
+
+    @property
+    def binary(self) -> "BinaryNamespace":
+        return BinaryNamespace(self._node)
+
     def hash(self, seed: int = 0) -> "Expression":
         from .. import kernels
 
@@ -1242,6 +1250,37 @@ def _float_is_inf(s: Series) -> Series:
 def _float_fill_nan(s: Series, value: float) -> Series:
     return Series(s.name, s.dtype, data=torch.nan_to_num(s.data, nan=value),
                   validity=s.validity)
+
+
+class UrlNamespace(_Namespace):
+    def download(self, on_error: str = "raise",
+                 max_connections: int = 32) -> Expression:
+        from ..functions.url import url_download_series
+        return self._fn("url_download", url_download_series,
+                        DataType.binary(), on_error, max_connections)
+
+    def upload(self, location: str, paths) -> Expression:
+        from ..functions.url import url_upload_series
+        return Expression(ScalarFn(
+            "url_upload", url_upload_series,
+            [self._node, _to_node(paths)], DataType.string(), (location,)))
+
+
+class BinaryNamespace(_Namespace):
+    def length(self) -> Expression:
+        from ..kernels import strings as k
+        return self._fn("binary_length", k.length_bytes, DataType.uint64())
+
+    def concat(self, other) -> Expression:
+        from ..kernels import strings as k
+        return Expression(ScalarFn(
+            "binary_concat", lambda a, b: k.concat_str([a, b]),
+            [self._node, _to_node(other)], DataType.binary()))
+
+    def slice(self, start: int, length: Optional[int] = None) -> Expression:
+        from ..kernels import strings as k
+        return self._fn("binary_slice", k.substr, DataType.binary(), start,
+                        length)
 
 
 class EmbeddingNamespace(_Namespace):

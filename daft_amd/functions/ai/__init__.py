@@ -1,0 +1,72 @@
+"""AI expression functions (ref: /root/reference/daft/functions/ai/
+__init__.py — embed_text :72, embed_image :157, classify_text :250)."""
+from __future__ import annotations
+
+from typing import Optional, Sequence
+
+import torch
+
+from ...expressions.expressions import Expression, ScalarFn, _to_node
+from ...schema import DataType
+from ...series import Series
+
+
+def embed_text(expr, provider: str = "hash", dimensions: int = 256,
+               **kwargs) -> Expression:
+    from ...ai import load_provider
+
+    def impl(s: Series) -> Series:
+        p = load_provider(provider, dimensions=dimensions, **kwargs) \
+            if provider == "hash" else load_provider(provider, **kwargs)
+        texts = s.cpu().to_pylist()
+        emb = p.embed_text(texts)
+        return Series.from_torch(s.name, emb.to(s.device),
+                                 DataType.embedding(DataType.float32(),
+                                                    emb.shape[1]),
+                                 validity=s.validity)
+
+    return Expression(ScalarFn(
+        "embed_text", impl, [_to_node(expr)],
+        DataType.embedding(DataType.float32(), dimensions)))
+
+
+def embed_image(expr, provider: str = "torch", dimensions: int = 512,
+                **kwargs) -> Expression:
+    """expr must be a FixedShapeTensor (image.to_tensor()) column."""
+    from ...ai import load_provider
+
+    def impl(s: Series) -> Series:
+        p = load_provider(provider, dimensions=dimensions, **kwargs)
+        n = len(s)
+        shape = s.dtype.shape
+        x = s.children[0].data.reshape((n,) + tuple(shape))
+        emb = p.embed_image(x)
+        return Series.from_torch(s.name, emb.to(s.device),
+                                 DataType.embedding(DataType.float32(),
+                                                    emb.shape[1]),
+                                 validity=s.validity)
+
+    return Expression(ScalarFn(
+        "embed_image", impl, [_to_node(expr)],
+        DataType.embedding(DataType.float32(), dimensions)))
+
+
+def classify_text(expr, labels: Sequence[str],
+                  provider: str = "hash", **kwargs) -> Expression:
+    """Nearest-label classification over provider text embeddings."""
+    from ...ai import load_provider
+
+    def impl(s: Series) -> Series:
+        p = load_provider(provider, **kwargs)
+        texts = s.cpu().to_pylist()
+        emb = p.embed_text(texts)
+        lab = p.embed_text(list(labels))
+        sim = emb @ lab.T
+        idx = sim.argmax(dim=1).tolist()
+        out = [labels[i] if t is not None else None
+               for i, t in zip(idx, texts)]
+        res = Series.from_pylist(s.name, out, DataType.string())
+        return res.to(s.device) if s.is_gpu() else res
+
+    return Expression(ScalarFn("classify_text", impl, [_to_node(expr)],
+                               DataType.string()))

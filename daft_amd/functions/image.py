@@ -1,0 +1,206 @@
+"""Image kernels (ref: /root/reference/src/daft-image/src/series.rs —
+decode :73, encode :99, resize :123, crop :157, to_mode :186, to_tensor
+:257).  Decode/encode run on host (PIL); resize is a HIP bilinear kernel
+over variable-size device-resident images."""
+from __future__ import annotations
+
+import io
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..expressions.expressions import Expression, ScalarFn, _Namespace
+from ..schema import DataType, TypeKind
+from ..series import Series
+from ..kernels import _is_gpu, native_required
+
+_MODE_CHANNELS = {"L": 1, "LA": 2, "RGB": 3, "RGBA": 4}
+_MODE_CODE = {"L": 1, "LA": 2, "RGB": 3, "RGBA": 4}
+
+
+def _image_struct(name, datas, heights, widths, channels, mode_codes,
+                  validity, device) -> Series:
+    """Build an Image(mode=None) struct series from per-row raw buffers."""
+    offs = np.zeros(len(datas) + 1, dtype=np.int64)
+    for i, d in enumerate(datas):
+        offs[i + 1] = offs[i] + len(d)
+    blob = b"".join(datas)
+    data_child = Series(
+        "data", DataType.binary(),
+        data=torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+        if blob else torch.zeros(0, dtype=torch.uint8),
+        offsets=torch.from_numpy(offs))
+    ch = Series("channel", DataType.uint16(),
+                data=torch.tensor(channels, dtype=torch.int16)
+                .view(torch.uint16))
+    h = Series("height", DataType.uint32(),
+               data=torch.tensor(heights, dtype=torch.int32)
+               .view(torch.uint32))
+    w = Series("width", DataType.uint32(),
+               data=torch.tensor(widths, dtype=torch.int32)
+               .view(torch.uint32))
+    m = Series("mode", DataType.uint8(),
+               data=torch.tensor(mode_codes, dtype=torch.uint8))
+    s = Series(name, DataType.image(), children=[data_child, ch, h, w, m],
+               validity=validity, length=len(datas))
+    return s.to(device) if str(device) != "cpu" else s
+
+
+def decode_series(s: Series, mode: str = "RGB",
+                  on_error: str = "raise") -> Series:
+    """binary (encoded JPEG/PNG/...) -> Image struct (host decode)."""
+    from PIL import Image as PILImage
+    vals = s.cpu().to_pylist()
+    datas, hs, ws, cs, ms, valid = [], [], [], [], [], []
+    for v in vals:
+        if v is None:
+            datas.append(b"")
+            hs.append(0)
+            ws.append(0)
+            cs.append(0)
+            ms.append(0)
+            valid.append(False)
+            continue
+        try:
+            img = PILImage.open(io.BytesIO(v)).convert(mode)
+            arr = np.asarray(img, dtype=np.uint8)
+            if arr.ndim == 2:
+                arr = arr[:, :, None]
+            datas.append(arr.tobytes())
+            hs.append(arr.shape[0])
+            ws.append(arr.shape[1])
+            cs.append(arr.shape[2])
+            ms.append(_MODE_CODE.get(mode, 3))
+            valid.append(True)
+        except Exception:
+            if on_error == "raise":
+                raise
+            datas.append(b"")
+            hs.append(0)
+            ws.append(0)
+            cs.append(0)
+            ms.append(0)
+            valid.append(False)
+    validity = torch.tensor(valid, dtype=torch.bool) \
+        if not all(valid) else None
+    return _image_struct(s.name, datas, hs, ws, cs, ms, validity, s.device)
+
+
+def encode_series(s: Series, fmt: str = "PNG") -> Series:
+    """Image struct -> encoded binary (host)."""
+    from PIL import Image as PILImage
+    cpu = s.cpu()
+    data, ch, h, w, _m = cpu.children
+    offs = data.offsets.numpy()
+    blob = data.data.numpy()
+    out = []
+    for i in range(len(cpu)):
+        if cpu.validity is not None and not bool(cpu.validity[i]):
+            out.append(None)
+            continue
+        hh = int(h.data.view(torch.int32)[i])
+        ww = int(w.data.view(torch.int32)[i])
+        cc = int(ch.data.view(torch.int16)[i])
+        arr = blob[offs[i]:offs[i + 1]].reshape(hh, ww, cc)
+        mode = {1: "L", 2: "LA", 3: "RGB", 4: "RGBA"}[cc]
+        buf = io.BytesIO()
+        PILImage.fromarray(arr.squeeze() if cc == 1 else arr, mode).save(
+            buf, format=fmt)
+        out.append(buf.getvalue())
+    res = Series.from_pylist(s.name, out, DataType.binary())
+    return res.to(s.device) if s.is_gpu() else res
+
+
+def resize_series(s: Series, h: int, w: int) -> Series:
+    """Image struct -> FixedShapeImage (HIP bilinear kernel on GPU)."""
+    assert s.dtype.kind == TypeKind.IMAGE, f"resize expects Image, got {s.dtype!r}"
+    data, ch, hh, ww, _m = s.children
+    channels = 3  # fixed-shape output mode RGB (round-1)
+    n = len(s)
+    out_dt = DataType.fixed_shape_image("RGB", h, w)
+    if _is_gpu(s):
+        src_h = hh.data.view(torch.int32)
+        src_w = ww.data.view(torch.int32)
+        flat = native_required().image_resize(
+            data.data, data.offsets[:-1].contiguous(), src_h.contiguous(),
+            src_w.contiguous(), channels, h, w)
+        child = Series("item", DataType.uint8(), data=flat)
+        return Series(s.name, out_dt, children=[child],
+                      validity=s.validity, length=n)
+    # CPU fallback: torch bilinear per row
+    offs = data.offsets
+    out = torch.zeros(n * h * w * channels, dtype=torch.uint8)
+    blob = data.data
+    for i in range(n):
+        ih = int(hh.data.view(torch.int32)[i])
+        iw = int(ww.data.view(torch.int32)[i])
+        if ih == 0 or iw == 0:
+            continue
+        arr = blob[offs[i]:offs[i + 1]].reshape(ih, iw, channels)
+        t = arr.permute(2, 0, 1).unsqueeze(0).to(torch.float32)
+        r = torch.nn.functional.interpolate(
+            t, size=(h, w), mode="bilinear", align_corners=False)
+        r8 = r.clamp(0, 255).round().to(torch.uint8).squeeze(0) \
+            .permute(1, 2, 0).contiguous()
+        out[i * h * w * channels:(i + 1) * h * w * channels] = r8.reshape(-1)
+    child = Series("item", DataType.uint8(), data=out)
+    return Series(s.name, out_dt, children=[child], validity=s.validity,
+                  length=n)
+
+
+def to_tensor_series(s: Series, dtype=None) -> Series:
+    """FixedShapeImage -> FixedShapeTensor float32 in [0,1], CHW."""
+    assert s.dtype.kind == TypeKind.FIXED_SHAPE_IMAGE
+    hgt, wdt = s.dtype.shape
+    c = _MODE_CHANNELS.get(s.dtype.image_mode or "RGB", 3)
+    n = len(s)
+    hwc = s.children[0].data.reshape(n, hgt, wdt, c)
+    chw = hwc.permute(0, 3, 1, 2).to(torch.float32) / 255.0
+    child = Series("item", DataType.float32(),
+                   data=chw.reshape(-1).contiguous())
+    return Series(s.name, DataType.fixed_shape_tensor(
+        DataType.float32(), (c, hgt, wdt)), children=[child],
+        validity=s.validity, length=n)
+
+
+def crop_series(s: Series, x: int, y: int, w: int, h: int) -> Series:
+    """FixedShapeImage crop (device slice)."""
+    assert s.dtype.kind == TypeKind.FIXED_SHAPE_IMAGE
+    H, W = s.dtype.shape
+    c = _MODE_CHANNELS.get(s.dtype.image_mode or "RGB", 3)
+    n = len(s)
+    img = s.children[0].data.reshape(n, H, W, c)
+    out = img[:, y:y + h, x:x + w, :].contiguous()
+    child = Series("item", DataType.uint8(), data=out.reshape(-1))
+    return Series(s.name, DataType.fixed_shape_image(
+        s.dtype.image_mode or "RGB", h, w), children=[child],
+        validity=s.validity, length=n)
+
+
+class ImageNamespace(_Namespace):
+    def decode(self, mode: str = "RGB", on_error: str = "raise"):
+        return self._fn("image_decode", decode_series, DataType.image(),
+                        mode, on_error)
+
+    def encode(self, image_format: str = "PNG"):
+        return self._fn("image_encode", encode_series, DataType.binary(),
+                        image_format)
+
+    def resize(self, h: int, w: int):
+        return self._fn("image_resize", resize_series,
+                        DataType.fixed_shape_image("RGB", h, w), h, w)
+
+    def to_tensor(self):
+        def ret(fields):
+            dt = fields[0].dtype
+            c = _MODE_CHANNELS.get(dt.image_mode or "RGB", 3)
+            return DataType.fixed_shape_tensor(DataType.float32(),
+                                               (c,) + tuple(dt.shape))
+        return self._fn("image_to_tensor", to_tensor_series, ret)
+
+    def crop(self, x: int, y: int, w: int, h: int):
+        def ret(fields):
+            dt = fields[0].dtype
+            return DataType.fixed_shape_image(dt.image_mode or "RGB", h, w)
+        return self._fn("image_crop", crop_series, ret, x, y, w, h)

@@ -306,6 +306,40 @@ def cast(s: Series, dtype: DataType) -> Series:
         child = s.children[0].cast(dtype.inner)
         return Series(s.name, dtype, children=[child], validity=s.validity,
                       length=len(s))
+    if k == TypeKind.LIST and nk in (TypeKind.FIXED_SIZE_LIST,
+                                     TypeKind.EMBEDDING,
+                                     TypeKind.FIXED_SHAPE_TENSOR):
+        if nk == TypeKind.FIXED_SHAPE_TENSOR:
+            size = 1
+            for d in dtype.shape:
+                size *= d
+            inner = dtype.inner
+        else:
+            size, inner = dtype.size, dtype.inner
+        lens = s.offsets[1:] - s.offsets[:-1]
+        ok = lens == size
+        if s.validity is not None:
+            ok = ok | ~s.validity
+        if not bool(ok.all().item()):
+            raise ValueError(
+                f"cannot cast ragged list to {dtype!r}: lengths differ")
+        # gather child rows densely (null rows fill with nulls)
+        idx = (s.offsets[:-1].unsqueeze(1) +
+               torch.arange(size, device=s.device)).reshape(-1)
+        if s.validity is not None:
+            idx = torch.where(
+                s.validity.repeat_interleave(size), idx,
+                torch.full_like(idx, -1))
+        child = s.children[0].take(idx).cast(inner)
+        return Series(s.name, dtype, children=[child], validity=s.validity,
+                      length=len(s))
+    if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING) and \
+            nk == TypeKind.LIST:
+        offs = torch.arange(0, (len(s) + 1) * s.dtype.size, s.dtype.size,
+                            dtype=torch.int64, device=s.device)
+        child = s.children[0].cast(dtype.inner)
+        return Series(s.name, dtype, offsets=offs, children=[child],
+                      validity=s.validity)
     raise TypeError(f"unsupported cast {s.dtype!r} -> {dtype!r}")
 
 

@@ -122,9 +122,27 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         first_idx = _first_valid_index(group_ids, num_groups, values)
         return values.take(first_idx).rename(name)
 
+    if kind == AggKind.APPROX_COUNT_DISTINCT and values.is_gpu():
+        # HyperLogLog: 2^14 dense registers per group, atomicMax HIP kernel
+        # (ref: hyperloglog/src/lib.rs:19-48 + daft-core hll_sketch/merge)
+        from ..kernels import native_required
+        h = rowops.hash_columns([values])
+        vmask = values.validity if values.validity is not None else \
+            torch.ones(len(values), dtype=torch.bool, device=dev)
+        regs = native_required().hll_update(h, group_ids, vmask, num_groups)
+        regs = regs.view(num_groups, 16384)
+        m = 16384.0
+        alpha = 0.7213 / (1.0 + 1.079 / m)
+        rf = regs.to(torch.float64)
+        est = alpha * m * m / torch.pow(2.0, -rf).sum(dim=1)
+        zeros = (regs == 0).sum(dim=1).to(torch.float64)
+        small = (est < 2.5 * m) & (zeros > 0)
+        lin = m * torch.log(m / zeros.clamp(min=1.0))
+        out = torch.where(small, lin, est).round().to(torch.int64)
+        return Series(name, DataType.uint64(), data=out.view(torch.uint64))
+
     if kind in (AggKind.COUNT_DISTINCT, AggKind.APPROX_COUNT_DISTINCT):
-        # exact two-level groupby: distinct (group, value) pairs, then count.
-        # (HLL sketch path lands with the multimodal round.)
+        # exact two-level groupby: distinct (group, value) pairs, then count
         gseries = Series("__gid", DataType.int64(), data=group_ids)
         sub_gids, sub_reps = rowops.groupby([gseries, values])
         outer = group_ids[sub_reps]

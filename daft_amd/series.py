@@ -217,6 +217,16 @@ class Series:
             child = s.children[0].to_pylist()
             n = self.dtype.size
             return wrap([child[i * n:(i + 1) * n] for i in range(len(s))])
+        if k == TypeKind.FIXED_SHAPE_IMAGE:
+            ch = {"L": 1, "LA": 2, "RGB": 3, "RGBA": 4}.get(
+                self.dtype.image_mode or "RGB", 3)
+            n = self.dtype.shape[0] * self.dtype.shape[1] * ch
+            child = s.children[0].to_pylist()
+            return wrap([child[i * n:(i + 1) * n] for i in range(len(s))])
+        if k == TypeKind.IMAGE:
+            cols = [c.to_pylist() for c in s.children]
+            names = ["data", "channel", "height", "width", "mode"]
+            return wrap([dict(zip(names, vals)) for vals in zip(*cols)])
         if k == TypeKind.FIXED_SHAPE_TENSOR:
             n = 1
             for d in self.dtype.shape:

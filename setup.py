@@ -20,6 +20,7 @@ setup(
                 "csrc/kernels.hip",
                 "csrc/sort.hip",
                 "csrc/strings.hip",
+                "csrc/multimodal.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

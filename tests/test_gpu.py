@@ -279,3 +279,62 @@ def test_pipeline_q1_shape():
     assert got["cnt"] == want["cnt"]
     for k in ("sum_qty", "disc_price", "charge", "avg_qty"):
         assert got[k] == pytest.approx(want[k], rel=1e-9)
+
+
+def test_minhash_gpu_matches_cpu():
+    docs = ["the quick brown fox jumps over the lazy dog",
+            "pack my box with five dozen liquor jugs",
+            "a b", "single", ""] * 200
+    g = daft.from_pydict({"t": docs}, device="cuda:0") \
+        .select(col("t").minhash(64, ngram_size=2).alias("mh")).to_pydict()
+    c = daft.from_pydict({"t": docs}, device="cpu") \
+        .select(col("t").minhash(64, ngram_size=2).alias("mh")).to_pydict()
+    assert g == c
+
+
+def test_hll_approx_count_distinct_gpu():
+    torch.manual_seed(20)
+    true_n = 5000
+    v = torch.randint(0, true_n, (500_000,))
+    est = daft.from_pydict({"v": v}, device="cuda:0") \
+        .agg(col("v").approx_count_distinct().alias("n")).to_pydict()["n"][0]
+    true_d = len(torch.unique(v))
+    assert abs(est - true_d) / true_d < 0.05, (est, true_d)
+
+
+def test_image_resize_gpu_close_to_cpu():
+    import numpy as np
+    import io
+    from PIL import Image as PILImage
+    rng = np.random.RandomState(0)
+    imgs = []
+    for _ in range(8):
+        h, w = rng.randint(8, 64), rng.randint(8, 64)
+        arr = rng.randint(0, 256, (h, w, 3), dtype="uint8")
+        buf = io.BytesIO()
+        PILImage.fromarray(arr, "RGB").save(buf, format="PNG")
+        imgs.append(buf.getvalue())
+    q = lambda dev: daft.from_pydict({"b": imgs}, device=dev) \
+        .select(col("b").image.decode().image.resize(16, 16).alias("r")) \
+        .to_pydict()["r"]
+    g, c = q("cuda:0"), q("cpu")
+    diffs = [abs(int(a) - int(b)) for ga, ca in zip(g, c)
+             for a, b in zip(ga, ca)]
+    assert sum(d <= 1 for d in diffs) / len(diffs) > 0.99, max(diffs)
+
+
+def test_embed_image_gpu():
+    from daft_amd.functions.ai import embed_image
+    import numpy as np
+    import io
+    from PIL import Image as PILImage
+    arr = np.full((32, 32, 3), 128, dtype="uint8")
+    buf = io.BytesIO()
+    PILImage.fromarray(arr, "RGB").save(buf, format="PNG")
+    df = daft.from_pydict({"b": [buf.getvalue()] * 4}, device="cuda:0")
+    out = (df.with_column("t", col("b").image.decode()
+                          .image.resize(32, 32).image.to_tensor())
+           .select(embed_image(col("t"), provider="torch",
+                               dimensions=64).alias("e"))
+           .to_pydict()["e"])
+    assert len(out[0]) == 64
