@@ -152,9 +152,11 @@ def _comment_vocab(n_vocab: int, special_fraction: float, seed: int,
 
 
 def _vocab_series(name: str, words: list, codes: torch.Tensor) -> Series:
+    """Dictionary-encoded string column (codes + vocab) — the GPU-native
+    layout: 4 B/row instead of offsets+bytes."""
     vocab = Series.from_pylist(name, words, DataType.string(),
                                device=codes.device)
-    return vocab.take(codes)
+    return Series.make_dict(name, vocab, codes.to(torch.int32))
 
 
 def _money(x: torch.Tensor) -> torch.Tensor:

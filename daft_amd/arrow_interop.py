@@ -101,7 +101,22 @@ def from_arrow_array(name: str, arr) -> Series:
     if isinstance(arr, pa.ChunkedArray):
         arr = arr.combine_chunks()
     if pa.types.is_dictionary(arr.type):
-        arr = arr.dictionary_decode()
+        # preserve dictionary encoding (codes + vocab)
+        import numpy as _np
+        vt = dtype_from_arrow(arr.type.value_type)
+        if vt.kind in (TypeKind.STRING, TypeKind.BINARY):
+            vocab = from_arrow_array("vocab", arr.dictionary)
+            codes = torch.from_numpy(
+                _np.ascontiguousarray(
+                    arr.indices.to_numpy(zero_copy_only=False)
+                    .astype(_np.int32)))
+            validity = None
+            if arr.null_count:
+                validity = torch.from_numpy(
+                    arr.is_valid().to_numpy(zero_copy_only=False))
+            from .series import Series as _S
+            return _S.make_dict(name, vocab, codes, validity)
+        arr = arr.cast(arr.type.value_type)
     dt = dtype_from_arrow(arr.type)
     k = dt.kind
 
@@ -180,6 +195,12 @@ def from_arrow_array(name: str, arr) -> Series:
 
 def to_arrow_array(s: Series):
     pa = _pa()
+    if s.is_dict():
+        vocab = to_arrow_array(s.children[0])
+        idx = s.data.numpy()
+        mask = None if s.validity is None else ~s.validity.numpy()
+        indices = pa.array(idx, mask=mask)
+        return pa.DictionaryArray.from_arrays(indices, vocab)
     dt = s.dtype
     k = dt.kind
     atype = dtype_to_arrow(dt) if k != TypeKind.PYTHON else None

@@ -335,7 +335,14 @@ class DataFrame:
         return pa.concat_tables(tables)
 
     def to_pandas(self):
-        return self.to_arrow().to_pandas()
+        import pandas as pd
+        df = self.collect()
+        parts = [p for p in df._result if len(p)]
+        if not parts:
+            return RecordBatch.empty(self.schema).to_pandas()
+        frames = [p.to_pandas() for p in parts]
+        return pd.concat(frames, ignore_index=True) if len(frames) > 1 \
+            else frames[0]
 
     def to_torch_map_dataset(self):
         d = self.to_pydict()

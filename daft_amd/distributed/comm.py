@@ -215,17 +215,27 @@ def _blob2t(blob) -> Optional[torch.Tensor]:
     return torch.from_numpy(arr.copy())
 
 
+def _skeleton(s: Series):
+    return (s.name, s.dtype, [_skeleton(c) for c in s.children])
+
+
+def _proto_from_skeleton(sk) -> Series:
+    name, dtype, children = sk
+    return Series(name, dtype, children=[_proto_from_skeleton(c)
+                                         for c in children], length=0)
+
+
 def _pickle_batch(rb: RecordBatch) -> bytes:
     cpu = rb.cpu()
     blobs = [_t2blob(t) for t in flatten_batch(cpu)]
-    return pickle.dumps((cpu.schema, blobs, len(cpu)))
+    skel = [_skeleton(c) for c in cpu.columns]
+    return pickle.dumps((skel, blobs, len(cpu)))
 
 
 def _unpickle_batch(blob: bytes) -> RecordBatch:
-    schema, blobs, n = pickle.loads(blob)
+    skel, blobs, n = pickle.loads(blob)
     bufs = [_blob2t(b) for b in blobs]
-    from ..series import empty_series
-    proto = RecordBatch([empty_series(f.name, f.dtype) for f in schema],
+    proto = RecordBatch([_proto_from_skeleton(sk) for sk in skel],
                         num_rows=0)
     return rebuild_batch(proto, bufs)
 

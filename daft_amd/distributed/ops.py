@@ -25,6 +25,9 @@ def _normalize_key(s: Series) -> Series:
         return s.cast(DataType.float64())
     if dt.is_decimal():
         return s.cast(DataType.float64())
+    if s.is_dict():
+        # vocab order may differ across ranks/tables: hash real bytes
+        return s.dict_decode()
     return s
 
 

@@ -58,6 +58,9 @@ _TAG_F32, _TAG_F64 = 5, 6
 def _col_desc(s: Series):
     """(tag, data_tensor, offsets_or_none, validity_or_none) for row ops."""
     k = s.dtype.kind
+    if s.is_dict():
+        # codes are equality-preserving within one column's vocab
+        return (_TAG_W4, s.data, None, s.validity)
     if k in (TypeKind.STRING, TypeKind.BINARY):
         return (_TAG_STR, s.data, s.offsets, s.validity)
     data = s.data
@@ -124,6 +127,9 @@ def take(s: Series, indices: torch.Tensor) -> Series:
         objs = [None if i < 0 else s.pyobjs[i] for i in idx_cpu]
         return Series(s.name, s.dtype, pyobjs=objs, validity=validity,
                       length=n_out)
+    if s.is_dict():
+        codes = s.data[safe_idx] if n_out else s.data[:0]
+        return Series.make_dict(s.name, s.children[0], codes, validity)
     if k in (TypeKind.STRING, TypeKind.BINARY):
         if _is_gpu(s):
             new_off, new_bytes = native_required().take_string(
@@ -232,6 +238,24 @@ def concat(series: List[Series]) -> Series:
             objs.extend(s.pyobjs)
         return Series(s0.name, dtype, pyobjs=objs, validity=validity,
                       length=total)
+    if any(s.is_dict() for s in series):
+        if all(s.is_dict() for s in series) and \
+                all(_same_vocab(s.children[0], series[0].children[0])
+                    for s in series[1:]):
+            codes = torch.cat([s.data for s in series])
+            return Series.make_dict(s0.name, series[0].children[0], codes,
+                                    validity)
+        series = [s.dict_decode() for s in series]
+        data = torch.cat([s.data for s in series])
+        new_off = torch.zeros(total + 1, dtype=torch.int64, device=dev)
+        pos, base = 0, 0
+        for s in series:
+            n = len(s)
+            new_off[pos + 1: pos + n + 1] = s.offsets[1:] + base
+            base += int(s.offsets[-1].item())
+            pos += n
+        return Series(s0.name, dtype, data=data, offsets=new_off,
+                      validity=validity)
     if k in (TypeKind.STRING, TypeKind.BINARY, TypeKind.LIST):
         new_bytes_parts = []
         new_off = torch.zeros(total + 1, dtype=torch.int64, device=dev)
@@ -267,6 +291,10 @@ def concat(series: List[Series]) -> Series:
 def cast(s: Series, dtype: DataType) -> Series:
     if s.dtype == dtype:
         return s
+    if s.is_dict():
+        s = s.dict_decode()
+        if s.dtype == dtype:
+            return s
     k, nk = s.dtype.kind, dtype.kind
     if nk == TypeKind.NULL:
         return full_null(s.name, dtype, len(s), s.device)
@@ -459,7 +487,54 @@ def compare_op(l: Series, r: Series, op: str) -> Series:
     return Series(l.name, DataType.bool(), data=out, validity=validity)
 
 
+def _same_vocab(a: Series, b: Series) -> bool:
+    if a is b:
+        return True
+    if len(a) != len(b):
+        return False
+    return bool(torch.equal(a.offsets, b.offsets) and
+                torch.equal(a.data, b.data))
+
+
 def _string_compare(l: Series, r: Series, op: str, validity) -> Series:
+    # dictionary fast paths: evaluate against the vocab, compare codes
+    if l.is_dict() or r.is_dict():
+        if l.is_dict() and r.is_dict() and op in ("eq", "ne") and \
+                _same_vocab(l.children[0], r.children[0]):
+            m = l.data == r.data if op == "eq" else l.data != r.data
+            return Series(l.name, DataType.bool(), data=m, validity=validity)
+        if l.is_dict() and not r.is_dict() and op in ("eq", "ne"):
+            # r is (usually) a broadcast literal: match codes against the
+            # vocab entries equal to each distinct r value; fall back to
+            # decode when r is non-constant
+            rv = r.to_pylist() if len(r) <= 1 else None
+            if rv is None and len(r) > 0:
+                first = r.slice(0, 1).to_pylist()[0]
+                rv = [first]
+                # cheap constant check on offsets pattern
+                lens = r.offsets[1:] - r.offsets[:-1]
+                if not bool((lens == lens[0]).all().item()):
+                    rv = None
+            if rv is not None:
+                target = rv[0] if rv else None
+                vocab = l.children[0].to_pylist()
+                code = vocab.index(target) if target in vocab else -1
+                # verify constant by comparing r against broadcast target
+                if len(r) > 1:
+                    from ..series import lit_series
+                    tl = lit_series("t", target, l.dtype,
+                                    device=r.device).broadcast(len(r))
+                    same = _string_compare(r, tl, "eq", None)
+                    if not bool(same.data.all().item()):
+                        return _string_compare(l.dict_decode(), r, op,
+                                               validity)
+                m = l.data == code
+                if op == "ne":
+                    m = ~m
+                return Series(l.name, DataType.bool(), data=m,
+                              validity=validity)
+        return _string_compare(l.dict_decode(), r.dict_decode(), op,
+                               validity)
     if _is_gpu(l):
         nat = native_required()
         cmp = nat.string_compare(l.offsets, l.data, r.offsets, r.data)

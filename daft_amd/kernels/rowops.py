@@ -63,6 +63,8 @@ def _cpu_hash_columns(series: Sequence[Series], seed: int) -> torch.Tensor:
     acc = np.full(n, np.uint64(seed) + np.uint64(0x8445D61A4E774912),
                   dtype=np.uint64)
     for s in series:
+        if s.is_dict():
+            s = s.dict_decode()
         k = s.dtype.kind
         if k in (TypeKind.STRING, TypeKind.BINARY):
             off = s.offsets.numpy()
@@ -218,6 +220,8 @@ def join(left_keys: Sequence[Series], right_keys: Sequence[Series],
     from ..schema import supertype
     lk2, rk2 = [], []
     for l, r in zip(left_keys, right_keys):
+        if l.is_dict() or r.is_dict():
+            l, r = _align_dict_keys(l, r)
         st = supertype(l.dtype, r.dtype)
         lk2.append(l.cast(st) if l.dtype != st else l)
         rk2.append(r.cast(st) if r.dtype != st else r)
@@ -225,6 +229,27 @@ def join(left_keys: Sequence[Series], right_keys: Sequence[Series],
     if _is_gpu(left_keys[0]):
         return _gpu_join(left_keys, right_keys, how)
     return _cpu_join(left_keys, right_keys, how)
+
+
+def _align_dict_keys(l: Series, r: Series):
+    """Re-encode the right dict column into the left vocab so code equality
+    matches string equality (vocab entries are distinct by construction)."""
+    from . import _same_vocab
+    if l.is_dict() and r.is_dict():
+        lv, rv = l.children[0], r.children[0]
+        if _same_vocab(lv, rv):
+            return l, r
+        lmap = {v: i for i, v in enumerate(lv.to_pylist())}
+        rvals = rv.to_pylist()
+        remap = torch.tensor([lmap.get(v, -1) for v in rvals],
+                             dtype=torch.int32, device=r.device)
+        new_codes = remap[r.data.to(torch.int64)]
+        validity = (new_codes >= 0)
+        if r.validity is not None:
+            validity = validity & r.validity
+        return l, Series.make_dict(r.name, lv, new_codes.clamp(min=0),
+                                   validity)
+    return l.dict_decode(), r.dict_decode()
 
 
 def _gpu_join(lk, rk, how):
@@ -352,7 +377,20 @@ def argsort_multi(keys: Sequence[Series], descending: Sequence[bool],
     perm = torch.arange(n, dtype=torch.int64, device=dev)
     # LSD over keys: sort by last key first
     for s, desc, nf in list(zip(keys, descending, nulls_first))[::-1]:
-        if s.dtype.kind in (TypeKind.STRING, TypeKind.BINARY):
+        if s.is_dict():
+            # order-preserving code ranks: argsort the (tiny) vocab once
+            vocab = s.children[0]
+            vperm = argsort_multi([vocab], [False], [False])
+            rank = torch.empty(len(vocab), dtype=torch.int64,
+                               device=s.device)
+            rank[vperm] = torch.arange(len(vocab), dtype=torch.int64,
+                                       device=s.device)
+            k = rank[s.data.to(torch.int64)]
+            if desc:
+                k = ~k
+            kg = k[perm]
+            perm = perm[_stable_sort_perm_by(kg)]
+        elif s.dtype.kind in (TypeKind.STRING, TypeKind.BINARY):
             nchunks = max(1, (_max_strlen(s) + 7) // 8)
             for c in range(nchunks - 1, -1, -1):
                 k = _string_chunk_key(s, c)

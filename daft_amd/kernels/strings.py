@@ -28,9 +28,30 @@ def _bool_out(s: Series, data: torch.Tensor) -> Series:
     return Series(s.name, DataType.bool(), data=data, validity=s.validity)
 
 
+def _dict_pred(s: Series, fn) -> Optional[Series]:
+    """Dictionary fast path for predicates: evaluate on the vocab, gather the
+    boolean by codes."""
+    if not s.is_dict():
+        return None
+    vres = fn(s.children[0])
+    out = vres.data[s.data.to(torch.int64)]
+    return Series(s.name, DataType.bool(), data=out, validity=s.validity)
+
+
+def _dict_map(s: Series, fn) -> Optional[Series]:
+    """Dictionary fast path for string->string maps: transform the vocab."""
+    if not s.is_dict():
+        return None
+    vres = fn(s.children[0])
+    return Series.make_dict(s.name, vres, s.data, s.validity)
+
+
 # --- predicates -----------------------------------------------------------
 
 def contains(s: Series, pat: str) -> Series:
+    d = _dict_pred(s, lambda v: contains(v, pat))
+    if d is not None:
+        return d
     if _is_gpu(s):
         p = _bytes_tensor(pat.encode(), s.device)
         return _bool_out(s, native_required().str_find(
@@ -42,6 +63,9 @@ def contains(s: Series, pat: str) -> Series:
 
 
 def startswith(s: Series, pat: str) -> Series:
+    d = _dict_pred(s, lambda v: startswith(v, pat))
+    if d is not None:
+        return d
     if _is_gpu(s):
         p = _bytes_tensor(pat.encode(), s.device)
         return _bool_out(s, native_required().str_find(
@@ -53,6 +77,9 @@ def startswith(s: Series, pat: str) -> Series:
 
 
 def endswith(s: Series, pat: str) -> Series:
+    d = _dict_pred(s, lambda v: endswith(v, pat))
+    if d is not None:
+        return d
     if _is_gpu(s):
         p = _bytes_tensor(pat.encode(), s.device)
         return _bool_out(s, native_required().str_find(
@@ -69,6 +96,9 @@ def like(s: Series, pattern: str, case_insensitive: bool = False) -> Series:
     Patterns without `_` (every TPC-H LIKE) run as a single ordered
     multi-substring HIP kernel on GPU; general patterns fall back to a host
     regex pass."""
+    d = _dict_pred(s, lambda v: like(v, pattern, case_insensitive))
+    if d is not None:
+        return d
     if "_" not in pattern and not case_insensitive:
         parts = pattern.split("%")
         if len(parts) == 1:
@@ -135,6 +165,9 @@ def _like_regex(s: Series, pattern: str, ci: bool) -> Series:
 
 
 def regexp_match(s: Series, pattern: str) -> Series:
+    d = _dict_pred(s, lambda v: regexp_match(v, pattern))
+    if d is not None:
+        return d
     rx = re.compile(pattern)
     vals = _pylist_str(s.cpu())
     out = torch.tensor([False if v is None else rx.search(v) is not None
@@ -146,6 +179,11 @@ def regexp_match(s: Series, pattern: str) -> Series:
 
 def length(s: Series) -> Series:
     """Length in UTF-8 characters."""
+    if s.is_dict():
+        vlen = length(s.children[0]).data.view(torch.int64)
+        out = vlen[s.data.to(torch.int64)]
+        return Series(s.name, DataType.uint64(), data=out.view(torch.uint64),
+                      validity=s.validity)
     if _is_gpu(s):
         out = native_required().str_char_length(s.offsets, s.data)
         return Series(s.name, DataType.uint64(),
@@ -158,6 +196,11 @@ def length(s: Series) -> Series:
 
 
 def length_bytes(s: Series) -> Series:
+    if s.is_dict():
+        vlen = length_bytes(s.children[0]).data.view(torch.int64)
+        out = vlen[s.data.to(torch.int64)]
+        return Series(s.name, DataType.uint64(), data=out.view(torch.uint64),
+                      validity=s.validity)
     lens = (s.offsets[1:] - s.offsets[:-1])
     return Series(s.name, DataType.uint64(), data=lens.view(torch.uint64),
                   validity=s.validity)
@@ -165,6 +208,9 @@ def length_bytes(s: Series) -> Series:
 
 def substr(s: Series, start: int, length: Optional[int]) -> Series:
     """Byte-offset substring (ASCII-correct; round-1 simplification)."""
+    d = _dict_map(s, lambda v: substr(v, start, length))
+    if d is not None:
+        return d
     if _is_gpu(s):
         new_off, new_bytes = native_required().str_substr(
             s.offsets, s.data, start, -1 if length is None else length)
@@ -180,6 +226,7 @@ def substr(s: Series, start: int, length: Optional[int]) -> Series:
 
 
 def _map_python(s: Series, f) -> Series:
+    s = s.dict_decode()
     vals = _pylist_str(s.cpu())
     out = [None if v is None else f(v) for v in vals]
     res = Series.from_pylist(s.name, out, DataType.string())
@@ -187,6 +234,9 @@ def _map_python(s: Series, f) -> Series:
 
 
 def lower(s: Series) -> Series:
+    d = _dict_map(s, lambda v: lower(v))
+    if d is not None:
+        return d
     if _is_gpu(s):
         out = native_required().str_case(s.offsets, s.data, 0)
         return Series(s.name, s.dtype, data=out, offsets=s.offsets,
@@ -195,6 +245,9 @@ def lower(s: Series) -> Series:
 
 
 def upper(s: Series) -> Series:
+    d = _dict_map(s, lambda v: upper(v))
+    if d is not None:
+        return d
     if _is_gpu(s):
         out = native_required().str_case(s.offsets, s.data, 1)
         return Series(s.name, s.dtype, data=out, offsets=s.offsets,
@@ -203,27 +256,43 @@ def upper(s: Series) -> Series:
 
 
 def lstrip(s: Series) -> Series:
+    d = _dict_map(s, lambda v: lstrip(v))
+    if d is not None:
+        return d
     return _map_python(s, str.lstrip)
 
 
 def rstrip(s: Series) -> Series:
+    d = _dict_map(s, lambda v: rstrip(v))
+    if d is not None:
+        return d
     return _map_python(s, str.rstrip)
 
 
 def strip(s: Series) -> Series:
+    d = _dict_map(s, lambda v: strip(v))
+    if d is not None:
+        return d
     return _map_python(s, str.strip)
 
 
 def reverse(s: Series) -> Series:
+    d = _dict_map(s, lambda v: reverse(v))
+    if d is not None:
+        return d
     return _map_python(s, lambda v: v[::-1])
 
 
 def capitalize(s: Series) -> Series:
+    d = _dict_map(s, lambda v: capitalize(v))
+    if d is not None:
+        return d
     return _map_python(s, str.capitalize)
 
 
 def concat_str(parts: List[Series]) -> Series:
     """Row-wise string concatenation."""
+    parts = [p.dict_decode() for p in parts]
     n = max(len(p) for p in parts)
     parts = [p.broadcast(n) if len(p) == 1 else p for p in parts]
     if _is_gpu(parts[0]):
@@ -262,6 +331,7 @@ def right(s: Series, n: int) -> Series:
 
 
 def find(s: Series, pat: str) -> Series:
+    s = s.dict_decode()
     if _is_gpu(s):
         p = _bytes_tensor(pat.encode(), s.device)
         out = native_required().str_find(s.offsets, s.data, p, 0)

@@ -191,7 +191,8 @@ class RecordBatch:
         return pa.table(dict(zip(self.column_names(), arrays)))
 
     def to_pandas(self):
-        return self.to_arrow().to_pandas()
+        cols = [c.dict_decode() if c.is_dict() else c for c in self.columns]
+        return RecordBatch(cols, self._num_rows).to_arrow().to_pandas()
 
     def __repr__(self) -> str:
         return (f"RecordBatch(rows={self._num_rows}, dev={self.device}, "

@@ -117,6 +117,37 @@ class Series:
             return 0
         return int((~self.validity).sum().item())
 
+    # ------------------------------------------------------------------
+    # dictionary-encoded strings: data = int32 codes, children = [vocab].
+    # The GPU-native representation for low-cardinality string columns —
+    # takes/filters gather 4-byte codes, predicates evaluate on the vocab,
+    # groupby keys compare codes (Arrow dictionary array equivalent).
+    # ------------------------------------------------------------------
+    def is_dict(self) -> bool:
+        return (self.dtype.kind in (TypeKind.STRING, TypeKind.BINARY)
+                and self.offsets is None and bool(self.children))
+
+    @staticmethod
+    def make_dict(name: str, vocab: "Series", codes: torch.Tensor,
+                  validity: Optional[torch.Tensor] = None) -> "Series":
+        assert vocab.dtype.kind in (TypeKind.STRING, TypeKind.BINARY)
+        assert not vocab.is_dict()
+        return Series(name, vocab.dtype, data=codes.to(torch.int32),
+                      children=[vocab], validity=validity,
+                      length=int(codes.shape[0]))
+
+    def dict_decode(self) -> "Series":
+        """Materialize a dictionary-encoded column to plain offsets+bytes."""
+        if not self.is_dict():
+            return self
+        out = self.children[0].take(self.data.to(torch.int64))
+        out = out.rename(self.name)
+        if self.validity is not None:
+            v = self.validity if out.validity is None \
+                else (out.validity & self.validity)
+            out = out.with_validity(v)
+        return out
+
     def to(self, device) -> "Series":
         def mv(t):
             return None if t is None else t.to(device)
@@ -197,6 +228,13 @@ class Series:
 
         if k == TypeKind.PYTHON:
             return list(s.pyobjs)
+        if s.is_dict():
+            vocab = s.children[0].to_pylist()
+            codes = s.data.to(torch.int64).numpy()
+            if valid is None:
+                return [vocab[c] for c in codes]
+            return [vocab[c] if ok else None
+                    for c, ok in zip(codes, valid)]
         if k in (TypeKind.STRING, TypeKind.BINARY):
             off = s.offsets.numpy()
             buf = s.data.numpy().tobytes() if len(s.data) else b""
