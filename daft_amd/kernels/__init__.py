@@ -460,6 +460,13 @@ def _null_validity(l: Series, r: Series):
 
 
 def compare_op(l: Series, r: Series, op: str) -> Series:
+    # dict vs scalar-literal fast path BEFORE broadcasting (avoids
+    # materializing a full-length string column for the literal)
+    if op in ("eq", "ne"):
+        if l.is_dict() and len(r) == 1 and not r.is_dict():
+            return _dict_eq_literal(l, r, op)
+        if r.is_dict() and len(l) == 1 and not l.is_dict():
+            return _dict_eq_literal(r, l, op)
     l, r = _align(l, r)
     validity = _null_validity(l, r)
     lk = l.dtype.kind
@@ -496,6 +503,19 @@ def _same_vocab(a: Series, b: Series) -> bool:
                 torch.equal(a.data, b.data))
 
 
+def _dict_eq_literal(d: Series, lit: Series, op: str) -> Series:
+    """dict column vs single-row literal: code comparison, no broadcast."""
+    target = lit.to_pylist()[0]
+    vocab = d.children[0].to_pylist()
+    code = vocab.index(target) if target in vocab else -1
+    m = d.data == code if op == "eq" else d.data != code
+    validity = None if d.validity is None else d.validity.clone()
+    if target is None:
+        m = torch.zeros(len(d), dtype=torch.bool, device=d.device)
+        validity = torch.zeros(len(d), dtype=torch.bool, device=d.device)
+    return Series(d.name, DataType.bool(), data=m, validity=validity)
+
+
 def _string_compare(l: Series, r: Series, op: str, validity) -> Series:
     # dictionary fast paths: evaluate against the vocab, compare codes
     if l.is_dict() or r.is_dict():
@@ -503,36 +523,6 @@ def _string_compare(l: Series, r: Series, op: str, validity) -> Series:
                 _same_vocab(l.children[0], r.children[0]):
             m = l.data == r.data if op == "eq" else l.data != r.data
             return Series(l.name, DataType.bool(), data=m, validity=validity)
-        if l.is_dict() and not r.is_dict() and op in ("eq", "ne"):
-            # r is (usually) a broadcast literal: match codes against the
-            # vocab entries equal to each distinct r value; fall back to
-            # decode when r is non-constant
-            rv = r.to_pylist() if len(r) <= 1 else None
-            if rv is None and len(r) > 0:
-                first = r.slice(0, 1).to_pylist()[0]
-                rv = [first]
-                # cheap constant check on offsets pattern
-                lens = r.offsets[1:] - r.offsets[:-1]
-                if not bool((lens == lens[0]).all().item()):
-                    rv = None
-            if rv is not None:
-                target = rv[0] if rv else None
-                vocab = l.children[0].to_pylist()
-                code = vocab.index(target) if target in vocab else -1
-                # verify constant by comparing r against broadcast target
-                if len(r) > 1:
-                    from ..series import lit_series
-                    tl = lit_series("t", target, l.dtype,
-                                    device=r.device).broadcast(len(r))
-                    same = _string_compare(r, tl, "eq", None)
-                    if not bool(same.data.all().item()):
-                        return _string_compare(l.dict_decode(), r, op,
-                                               validity)
-                m = l.data == code
-                if op == "ne":
-                    m = ~m
-                return Series(l.name, DataType.bool(), data=m,
-                              validity=validity)
         return _string_compare(l.dict_decode(), r.dict_decode(), op,
                                validity)
     if _is_gpu(l):
@@ -657,6 +647,12 @@ def _cpu_merge_strings(m, t: Series, f: Series, new_off) -> torch.Tensor:
 
 
 def is_in(s: Series, values: Series) -> Series:
+    if s.is_dict():
+        # evaluate membership on the vocab, gather by code
+        vmask = is_in(s.children[0], values)
+        out = vmask.data[s.data.to(torch.int64)]
+        return Series(s.name, DataType.bool(), data=out,
+                      validity=s.validity)
     out = None
     for i in range(len(values)):
         v = values.slice(i, i + 1)
