@@ -1,0 +1,82 @@
+"""Multimodal pipeline benchmark (BASELINE.json config 4): images ->
+decode -> resize (HIP bilinear) -> to_tensor -> embed (MFMA matmul via
+torch-rocm) on device-resident columns.
+
+There is no network on the bench boxes, so "url_download" is replaced by
+in-memory encoded PNG bytes (the download itself is host IO the reference
+also pays); the measured pipeline is decode -> resize -> embed.
+
+  python benchmarks/bench_multimodal.py --images 100000 --batch 8192
+"""
+from __future__ import annotations
+
+import argparse
+import io
+import json
+import time
+
+import numpy as np
+import torch
+
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
+def synth_pngs(n_distinct: int, size_lo=64, size_hi=256, seed=0):
+    from PIL import Image
+    rng = np.random.RandomState(seed)
+    out = []
+    for _ in range(n_distinct):
+        h = int(rng.randint(size_lo, size_hi))
+        w = int(rng.randint(size_lo, size_hi))
+        arr = rng.randint(0, 256, (h, w, 3), dtype="uint8")
+        buf = io.BytesIO()
+        Image.fromarray(arr, "RGB").save(buf, format="PNG")
+        out.append(buf.getvalue())
+    return out
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--images", type=int, default=100_000)
+    ap.add_argument("--distinct", type=int, default=256)
+    ap.add_argument("--batch", type=int, default=16_384)
+    ap.add_argument("--dim", type=int, default=512)
+    ap.add_argument("--size", type=int, default=224)
+    args = ap.parse_args()
+
+    import daft_amd as daft
+    from daft_amd import col
+    from daft_amd.functions.ai import embed_image
+
+    on_gpu = torch.cuda.is_available()
+    device = "cuda:0" if on_gpu else "cpu"
+    pngs = synth_pngs(args.distinct)
+    codes = np.random.RandomState(1).randint(0, args.distinct, args.images)
+    urls = [pngs[c] for c in codes]
+
+    df = daft.from_pydict({"data": urls}, device="cpu").into_batches(
+        args.batch)
+    t0 = time.perf_counter()
+    out = (df
+           .with_column("img", col("data").image.decode())
+           .with_column("small", col("img").image.resize(args.size,
+                                                         args.size))
+           .with_column("t", col("small").image.to_tensor())
+           .select(embed_image(col("t"), provider="torch",
+                               dimensions=args.dim).alias("emb"))
+           .count_rows())
+    if on_gpu:
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(json.dumps({
+        "metric": "images_per_s", "value": round(args.images / dt, 1),
+        "unit": "img/s", "images": args.images, "seconds": round(dt, 2),
+        "device": device, "pipeline": "decode->resize(HIP)->embed(MFMA)",
+    }))
+    assert out == args.images
+
+
+if __name__ == "__main__":
+    main()
