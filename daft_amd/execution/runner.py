@@ -40,7 +40,10 @@ class NativeRunner:
                            [l.strip("* ") for l in phys.explain_lines()])
                 device = device or ctx.device()
                 ectx = ExecContext(ctx, device, query_id)
-                yield from phys.execute(ectx)
+                yield from phys.execute_tracked(ectx)
+                for key, (rows, batches, secs) in ectx.stats.items():
+                    ctx.notify("on_operator_end", query_id, key,
+                               ectx.op_names.get(key, "?"), -1, rows, secs)
                 ctx.notify("on_exec_end", query_id)
         except Exception as e:
             err = str(e)

@@ -33,6 +33,10 @@ class ExecContext:
         self.ctx = ctx
         self.device = device
         self.query_id = query_id
+        # per-operator runtime stats (ref: daft-local-execution
+        # src/runtime_stats/): node id -> [rows_out, batches, seconds]
+        self.stats: Dict[int, list] = {}
+        self.op_names: Dict[int, str] = {}
 
 
 class PhysicalOp:
@@ -47,11 +51,40 @@ class PhysicalOp:
     def execute(self, ectx: ExecContext) -> BatchIter:
         raise NotImplementedError(type(self))
 
+    def execute_tracked(self, ectx: ExecContext) -> BatchIter:
+        """execute() wrapped with per-operator runtime stats and roctx
+        ranges (ref: runtime_stats/ + OTLP spans in common/tracing; GPU
+        ranges show up in rocprof timelines via roctx)."""
+        import time as _time
+        key = id(self)
+        st = ectx.stats.setdefault(key, [0, 0, 0.0])
+        ectx.op_names[key] = self.op_name
+        use_roctx = str(ectx.device).startswith("cuda")
+        gen = self.execute(ectx)
+        while True:
+            t0 = _time.perf_counter()
+            if use_roctx:
+                import torch as _torch
+                _torch.cuda.nvtx.range_push(self.op_name)
+            try:
+                batch = next(gen)
+            except StopIteration:
+                st[2] += _time.perf_counter() - t0
+                return
+            finally:
+                if use_roctx:
+                    import torch as _torch
+                    _torch.cuda.nvtx.range_pop()
+            st[0] += len(batch)
+            st[1] += 1
+            st[2] += _time.perf_counter() - t0
+            yield batch
+
     def _child_iters(self, ectx) -> List[BatchIter]:
-        return [c.execute(ectx) for c in self.children]
+        return [c.execute_tracked(ectx) for c in self.children]
 
     def _materialize_child(self, ectx, i: int = 0) -> RecordBatch:
-        batches = list(self.children[i].execute(ectx))
+        batches = list(self.children[i].execute_tracked(ectx))
         if not batches:
             return RecordBatch.empty(self.children[i].schema,
                                      device=ectx.device)
@@ -120,7 +153,7 @@ class ProjectOp(PhysicalOp):
         self.exprs = exprs
 
     def execute(self, ectx) -> BatchIter:
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             cols = []
             n = len(rb)
             for e in self.exprs:
@@ -137,7 +170,7 @@ class FilterOp(PhysicalOp):
         self.predicate = predicate
 
     def execute(self, ectx) -> BatchIter:
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             mask = self.predicate.evaluate(rb)
             if len(mask) == 1 and len(rb) != 1:
                 mask = mask.broadcast(len(rb))
@@ -157,7 +190,7 @@ class LimitOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         to_skip = self.offset
         to_take = self.limit
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             if to_take <= 0:
                 return
             if to_skip:
@@ -182,7 +215,7 @@ class ExplodeOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         explode_names = [e.to_field(self.children[0].schema).name
                          for e in self.exprs]
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             n = len(rb)
             cols = {c.name: c for c in rb.columns}
             for e, nm in zip(self.exprs, explode_names):
@@ -239,7 +272,7 @@ class UnpivotOp(PhysicalOp):
         self.value_name = value_name
 
     def execute(self, ectx) -> BatchIter:
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             n = len(rb)
             k = len(self.values)
             dev = rb.device
@@ -275,7 +308,7 @@ class SampleOp(PhysicalOp):
         gen = torch.Generator(device="cpu")
         if self.seed is not None:
             gen.manual_seed(self.seed)
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             n = len(rb)
             if self.with_replacement:
                 k = int(round(n * self.fraction))
@@ -302,7 +335,7 @@ class MonotonicIdOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         base = self.partition_id << 36
         count = 0
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             n = len(rb)
             ids = torch.arange(base + count, base + count + n,
                                dtype=torch.int64, device=rb.device)
@@ -320,7 +353,7 @@ class IntoBatchesOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         pending: List[RecordBatch] = []
         pending_rows = 0
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             pending.append(rb)
             pending_rows += len(rb)
             while pending_rows >= self.batch_size:
@@ -406,7 +439,7 @@ class TopNOp(PhysicalOp):
         # per-batch prune to limit+offset, then final sort+slice
         k = self.limit + self.offset
         pruned: List[RecordBatch] = []
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             if len(rb) > k:
                 keys = [e.evaluate(rb) for e in self.by]
                 perm = rowops.argsort_multi(keys, self.descending,
@@ -484,7 +517,7 @@ class ConcatOp(PhysicalOp):
 
     def execute(self, ectx) -> BatchIter:
         for c in self.children:
-            yield from c.execute(ectx)
+            yield from c.execute_tracked(ectx)
 
 
 class RepartitionOp(PhysicalOp):
@@ -588,7 +621,7 @@ class UDFProjectOp(PhysicalOp):
         self.passthrough = passthrough
 
     def execute(self, ectx) -> BatchIter:
-        for rb in self.children[0].execute(ectx):
+        for rb in self.children[0].execute_tracked(ectx):
             cols = [e.evaluate(rb).rename(e.to_field(rb.schema).name)
                     for e in self.passthrough]
             s = self.udf_expr.evaluate(rb)
@@ -615,7 +648,8 @@ class WriteOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         from ..io import writers
         paths = writers.write_batches(
-            self.children[0].execute(ectx), self.file_format, self.root_dir,
+            self.children[0].execute_tracked(ectx), self.file_format,
+            self.root_dir,
             self.write_mode, self.partition_cols, self.options, ectx)
         yield RecordBatch([Series.from_pylist("path", paths,
                                               DataType.string())],

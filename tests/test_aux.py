@@ -1,0 +1,221 @@
+"""Aux subsystem tests: subscribers/events, checkpoint store, IO round trips,
+UDFs, config (ref pattern: tests/test_subscribers.py, tests/io/)."""
+import os
+
+import pytest
+
+import daft_amd as daft
+from daft_amd import DataType, col
+from daft_amd.context import Subscriber, get_context
+
+
+class _Capture(Subscriber):
+    def __init__(self):
+        self.events = []
+
+    def on_query_start(self, query_id, explain):
+        self.events.append(("start", query_id))
+
+    def on_query_end(self, query_id, seconds, error):
+        self.events.append(("end", query_id, error))
+
+    def on_operator_end(self, query_id, node_id, name, rows_in, rows_out,
+                        seconds):
+        self.events.append(("op", name, rows_out))
+
+
+def test_subscriber_events():
+    sub = _Capture()
+    daft.attach_subscriber(sub)
+    try:
+        daft.from_pydict({"a": [1, 2, 3]}).where(col("a") > 1).collect()
+    finally:
+        daft.detach_subscriber(sub)
+    kinds = [e[0] for e in sub.events]
+    assert "start" in kinds and "end" in kinds and "op" in kinds
+    op_events = {e[1]: e[2] for e in sub.events if e[0] == "op"}
+    assert op_events.get("Filter") == 2
+
+
+def test_event_log_subscriber(tmp_path):
+    from daft_amd.subscribers import EventLogSubscriber
+    sub = EventLogSubscriber(str(tmp_path))
+    daft.attach_subscriber(sub)
+    try:
+        daft.from_pydict({"a": [1]}).collect()
+    finally:
+        daft.detach_subscriber(sub)
+    import json
+    lines = [json.loads(l) for l in open(sub.path)]
+    assert any(l["event"] == "query_end" for l in lines)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    from daft_amd.checkpoint import CheckpointConfig, LocalCheckpointStore
+    store = LocalCheckpointStore(str(tmp_path / "ck"))
+    cfg = CheckpointConfig(store, on="id")
+    df1 = daft.from_pydict({"id": [1, 2, 3], "v": [10, 20, 30]})
+    out1 = cfg.filter_processed(df1)
+    assert out1.count_rows() == 3
+    cfg.commit(out1)
+    df2 = daft.from_pydict({"id": [2, 3, 4, 5], "v": [0, 0, 40, 50]})
+    out2 = cfg.filter_processed(df2)
+    assert sorted(out2.to_pydict()["id"]) == [4, 5]
+
+
+def test_parquet_roundtrip(tmp_path):
+    df = daft.from_pydict({
+        "a": [1, 2, 3], "b": ["x", None, "z"], "c": [1.5, 2.5, None],
+    })
+    df.write_parquet(str(tmp_path / "out"))
+    back = daft.read_parquet(str(tmp_path / "out") + "/*.parquet") \
+        .sort("a").to_pydict()
+    assert back == {"a": [1, 2, 3], "b": ["x", None, "z"],
+                    "c": [1.5, 2.5, None]}
+
+
+def test_parquet_pushdown(tmp_path):
+    daft.from_pydict({"a": list(range(100)), "b": list(range(100))}) \
+        .write_parquet(str(tmp_path / "p"))
+    df = daft.read_parquet(str(tmp_path / "p") + "/*.parquet")
+    out = df.select("a").where(col("a") < 5).to_pydict()
+    assert out == {"a": [0, 1, 2, 3, 4]}
+    # limit pushdown
+    assert daft.read_parquet(str(tmp_path / "p") + "/*.parquet") \
+        .limit(3).count_rows() == 3
+
+
+def test_csv_roundtrip(tmp_path):
+    df = daft.from_pydict({"a": [1, 2], "s": ["x", "y"]})
+    df.write_csv(str(tmp_path / "c"))
+    back = daft.read_csv(str(tmp_path / "c") + "/*.csv").sort("a").to_pydict()
+    assert back["a"] == [1, 2] and back["s"] == ["x", "y"]
+
+
+def test_json_write(tmp_path):
+    df = daft.from_pydict({"a": [1, 2]})
+    paths = df.write_json(str(tmp_path / "j")).to_pydict()["path"]
+    import json
+    rows = [json.loads(l) for l in open(paths[0])]
+    assert rows == [{"a": 1}, {"a": 2}]
+
+
+def test_partitioned_write(tmp_path):
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1, 2, 3]})
+    paths = df.write_parquet(str(tmp_path / "pw"),
+                             partition_cols=[col("g")]).to_pydict()["path"]
+    assert any("g=a" in p for p in paths)
+    assert any("g=b" in p for p in paths)
+
+
+def test_udf_row_wise():
+    @daft.func
+    def add1(x: int) -> int:
+        return x + 1
+
+    df = daft.from_pydict({"a": [1, 2, 3]})
+    assert df.select(add1(col("a"))).to_pydict()["add1"] == [2, 3, 4]
+
+
+def test_udf_batched():
+    @daft.func(return_dtype=DataType.float64(), batched=True)
+    def double(s):
+        import torch
+        from daft_amd.series import Series
+        return Series("d", DataType.float64(), data=s.data * 2.0,
+                      validity=s.validity)
+
+    df = daft.from_pydict({"a": [1.0, 2.0]})
+    assert df.select(double(col("a"))).to_pydict()["double"] == [2.0, 4.0]
+
+
+def test_udf_retry_and_null():
+    calls = {"n": 0}
+
+    @daft.func(return_dtype=DataType.int64(), max_retries=2)
+    def flaky(x: int) -> int:
+        calls["n"] += 1
+        if calls["n"] < 3:
+            raise RuntimeError("boom")
+        return x
+
+    df = daft.from_pydict({"a": [7]})
+    assert df.select(flaky(col("a"))).to_pydict()["flaky"] == [7]
+
+    @daft.func(return_dtype=DataType.int64(), on_error="null")
+    def broken(x: int) -> int:
+        raise RuntimeError("always")
+
+    out = df.select(broken(col("a"))).to_pydict()["broken"]
+    assert out == [None]
+
+
+def test_stateful_cls_udf():
+    @daft.cls
+    class Scaler:
+        def __init__(self, k):
+            self.k = k
+
+        @daft.method(return_dtype=DataType.int64())
+        def scale(self, x):
+            return x * self.k
+
+    sc = Scaler(10)
+    df = daft.from_pydict({"a": [1, 2]})
+    assert df.select(sc.scale(col("a"))).to_pydict()["scale"] == [10, 20]
+
+
+def test_execution_config_ctx():
+    from daft_amd import execution_config_ctx
+    ctx = get_context()
+    before = ctx.execution_config.morsel_size_rows
+    with execution_config_ctx(morsel_size_rows=123):
+        assert ctx.execution_config.morsel_size_rows == 123
+    assert ctx.execution_config.morsel_size_rows == before
+
+
+def test_cli_schema(tmp_path, capsys):
+    daft.from_pydict({"a": [1]}).write_parquet(str(tmp_path / "t"))
+    from daft_amd.cli import main
+    import glob
+    path = glob.glob(str(tmp_path / "t") + "/*.parquet")[0]
+    main(["schema", path])
+    out = capsys.readouterr().out
+    assert "a" in out
+
+
+def test_window_functions():
+    from daft_amd.functions import dense_rank, rank, row_number
+    from daft_amd.window import Window
+    df = daft.from_pydict({"g": ["a", "a", "a", "b"],
+                           "v": [3.0, 1.0, 3.0, 5.0]})
+    w = Window().partition_by("g").order_by("v")
+    out = df.with_window_columns({
+        "rn": row_number().over(w),
+        "rk": rank().over(w),
+        "dr": dense_rank().over(w),
+        "sv": col("v").sum().over(w),
+    }).sort(["g", "v", "rn"]).to_pydict()
+    assert out["rn"] == [1, 2, 3, 1]
+    assert out["rk"] == [1, 2, 2, 1]
+    assert out["dr"] == [1, 2, 2, 1]
+    assert out["sv"] == [7.0, 7.0, 7.0, 5.0]
+
+
+def test_lag_lead():
+    df = daft.from_pydict({"g": ["a", "a", "a"], "v": [1, 2, 3]})
+    from daft_amd.window import Window
+    w = Window().partition_by("g").order_by("v")
+    out = df.with_window_columns({
+        "prev": col("v").lag(1).over(w),
+        "next": col("v").lead(1).over(w),
+    }).sort("v").to_pydict()
+    assert out["prev"] == [None, 1, 2]
+    assert out["next"] == [2, 3, None]
+
+
+def test_explain(capsys):
+    df = daft.from_pydict({"a": [1]}).where(col("a") > 0).select(col("a"))
+    df.explain(show_all=True)
+    out = capsys.readouterr().out
+    assert "Filter" in out and "Optimized" in out
