@@ -24,8 +24,13 @@ from .context import (  # noqa: F401
 )
 from .io import (  # noqa: F401
     from_pydict, from_arrow, from_pandas, from_glob_path,
-    read_parquet, read_csv, read_json,
+    read_parquet, read_csv, read_json, read_text,
 )
+from .catalog import Catalog, Identifier, MemoryCatalog, Session, \
+    current_session  # noqa: F401
+from .window import Window  # noqa: F401
+from .checkpoint import (CheckpointConfig, CheckpointStore,  # noqa: F401
+                         LocalCheckpointStore, MemoryCheckpointStore)
 from .sql import sql  # noqa: F401
 from .udf import func, udf, cls, method  # noqa: F401
 from .functions import coalesce  # noqa: F401

@@ -113,3 +113,18 @@ def from_glob_path(path: str) -> DataFrame:
     sizes = [os.path.getsize(p) if os.path.exists(p) else None for p in paths]
     return from_pydict({"path": paths, "size": sizes,
                         "num_rows": [None] * len(paths)})
+
+
+def read_text(path, line_column: str = "text") -> DataFrame:
+    """Read text files line-by-line (ref capability: daft-text read.rs)."""
+    paths = _expand_paths(path)
+    lines: List[str] = []
+    for p in paths:
+        with open(p, "r", errors="replace") as f:
+            lines.extend(l.rstrip("\n") for l in f)
+    return from_pydict({line_column: lines})
+
+
+def read_jsonl(path) -> DataFrame:
+    """Read newline-delimited JSON (alias of read_json for local files)."""
+    return read_json(path)
