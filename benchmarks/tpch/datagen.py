@@ -178,44 +178,55 @@ def _shard(total: int, rank: int, world: int) -> Tuple[int, int]:
     return lo, hi
 
 
+def _fixed_width_string(name: str, mat: torch.Tensor) -> Series:
+    """[n, L] uint8 byte matrix -> string Series (single reshape, no gather)."""
+    n, L = mat.shape
+    offs = torch.arange(0, (n + 1) * L, L, dtype=torch.int64,
+                        device=mat.device)
+    return Series(name, DataType.string(), data=mat.reshape(-1).contiguous(),
+                  offsets=offs)
+
+
 def _format_keyed(name: str, prefix: str, keys: torch.Tensor) -> Series:
-    """prefix + 9-digit zero-padded key, built on device."""
+    """prefix + 9-digit zero-padded key, built on device as a fixed-width
+    byte matrix (no per-row gathers)."""
     dev = keys.device
     n = keys.shape[0]
-    pre = Series.from_pylist("p", [prefix], DataType.string(), device=dev) \
-        .broadcast(n)
-    digits = []
+    pb = prefix.encode()
+    L = len(pb) + 9
+    mat = torch.empty(n, L, dtype=torch.uint8, device=dev)
+    for i, b in enumerate(pb):
+        mat[:, i] = b
     for d in range(9):
         div = 10 ** (8 - d)
         dig = torch.remainder(torch.div(keys, div, rounding_mode="floor"), 10)
-        digits.append(_vocab_series("d", [str(i) for i in range(10)], dig))
-    from daft_amd.kernels import strings as strk
-    return strk.concat_str([pre] + digits).rename(name)
+        mat[:, len(pb) + d] = (48 + dig).to(torch.uint8)
+    return _fixed_width_string(name, mat)
 
 
 def _phone(key: torch.Tensor, nationkey: torch.Tensor, salt: int,
            name: str) -> Series:
-    """'CC-NNN-NNN-NNNN' with country code 10+nationkey (Q22)."""
+    """'CC-NNN-NNN-NNNN' with country code 10+nationkey (Q22); fixed-width
+    byte matrix, built with pure tensor ops."""
     dev = nationkey.device
-    from daft_amd.kernels import strings as strk
     n = nationkey.shape[0]
-    cc = _vocab_series("cc", [str(10 + i) for i in range(25)], nationkey)
-    dash = Series.from_pylist("-", ["-"], DataType.string(),
-                              device=dev).broadcast(n)
-    parts = [cc]
+    L = 15
+    mat = torch.empty(n, L, dtype=torch.uint8, device=dev)
+    cc = 10 + nationkey
+    mat[:, 0] = (48 + torch.div(cc, 10, rounding_mode="floor")).to(torch.uint8)
+    mat[:, 1] = (48 + torch.remainder(cc, 10)).to(torch.uint8)
+    pos = 2
     for si, ln in enumerate((3, 3, 4)):
-        lo = 10 ** (ln - 1)
-        seg = _randint(key, salt + si, lo, 10 ** ln)
-        digs = []
+        mat[:, pos] = ord("-")
+        pos += 1
+        seg = _randint(key, salt + si, 10 ** (ln - 1), 10 ** ln)
         for d in range(ln):
             div = 10 ** (ln - 1 - d)
-            digs.append(_vocab_series(
-                "d", [str(i) for i in range(10)],
-                torch.remainder(torch.div(seg, div, rounding_mode="floor"),
-                                10)))
-        parts.append(dash)
-        parts.extend(digs)
-    return strk.concat_str(parts).rename(name)
+            mat[:, pos] = (48 + torch.remainder(
+                torch.div(seg, div, rounding_mode="floor"), 10)) \
+                .to(torch.uint8)
+            pos += 1
+    return _fixed_width_string(name, mat)
 
 
 # ---------------------------------------------------------------------------
