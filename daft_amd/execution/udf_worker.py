@@ -1,0 +1,90 @@
+"""Out-of-process UDF execution (ref: /root/reference/daft/execution/
+{udf.py,udf_worker.py} + intermediate_ops/udf.rs:351-406 — GIL isolation /
+GPU pinning for stateful or heavy python UDFs).
+
+A worker subprocess is started lazily per function; row batches travel as
+cloudpickle payloads over pipes.  GPU columns hop through host memory (the
+worker does not share the parent's device context)."""
+from __future__ import annotations
+
+import atexit
+import multiprocessing as mp
+import threading
+from typing import Any, Dict, List
+
+
+def _worker_main(conn, fn_blob: bytes):
+    import cloudpickle
+    try:
+        fn = cloudpickle.loads(fn_blob)
+    except Exception as e:  # pragma: no cover
+        conn.send(("init_error", repr(e)))
+        return
+    conn.send(("ready", None))
+    while True:
+        try:
+            msg = conn.recv()
+        except EOFError:
+            return
+        if msg[0] == "stop":
+            return
+        _, rows = msg
+        try:
+            out = [fn(*r) for r in rows]
+            conn.send(("ok", out))
+        except Exception as e:
+            conn.send(("error", repr(e)))
+
+
+class _Worker:
+    def __init__(self, fn):
+        import cloudpickle
+        ctx = mp.get_context("spawn")
+        self._parent, child = ctx.Pipe()
+        self._proc = ctx.Process(
+            target=_worker_main, args=(child, cloudpickle.dumps(fn)),
+            daemon=True)
+        self._proc.start()
+        status, payload = self._parent.recv()
+        if status != "ready":
+            raise RuntimeError(f"UDF worker failed to start: {payload}")
+        self._lock = threading.Lock()
+
+    def call_rows(self, rows: List[tuple]) -> list:
+        with self._lock:
+            self._parent.send(("rows", rows))
+            status, payload = self._parent.recv()
+        if status == "error":
+            raise RuntimeError(f"UDF worker error: {payload}")
+        return payload
+
+    def stop(self):
+        try:
+            self._parent.send(("stop", None))
+        except Exception:
+            pass
+        self._proc.join(timeout=5)
+        if self._proc.is_alive():
+            self._proc.terminate()
+
+
+_workers: Dict[int, _Worker] = {}
+_workers_lock = threading.Lock()
+
+
+def get_worker(fn) -> _Worker:
+    key = id(fn)
+    with _workers_lock:
+        w = _workers.get(key)
+        if w is None or not w._proc.is_alive():
+            w = _Worker(fn)
+            _workers[key] = w
+        return w
+
+
+@atexit.register
+def _shutdown():
+    with _workers_lock:
+        for w in _workers.values():
+            w.stop()
+        _workers.clear()

@@ -125,6 +125,16 @@ def run_udf_node(node: PyUDF, batch) -> Series:
     last_err: Optional[Exception] = None
     for _ in range(attempts):
         try:
+            if node.use_process and not node.batched:
+                # subprocess isolation (ref: udf.rs:351-406 + udf_worker.py)
+                from .execution.udf_worker import get_worker
+                worker = get_worker(node.fn)
+                cols = [s.cpu().to_pylist() for s in arg_series]
+                rows = list(zip(*cols)) if cols else [()] * n
+                out_vals = worker.call_rows(rows)
+                return Series.from_pylist(node.name, out_vals,
+                                          node.return_dtype,
+                                          device=batch.device)
             if node.batched:
                 out = node.fn(*arg_series)
                 if isinstance(out, Series):

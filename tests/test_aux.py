@@ -219,3 +219,33 @@ def test_explain(capsys):
     df.explain(show_all=True)
     out = capsys.readouterr().out
     assert "Filter" in out and "Optimized" in out
+
+
+def test_udf_use_process():
+    @daft.func(return_dtype=DataType.int64(), use_process=True)
+    def triple(x: int) -> int:
+        import os
+        return x * 3
+
+    df = daft.from_pydict({"a": [1, 2, 3]})
+    assert df.select(triple(col("a"))).to_pydict()["triple"] == [3, 6, 9]
+
+
+def test_pivot_and_unpivot_roundtrip():
+    df = daft.from_pydict({"id": [1, 2], "x": [10, 20], "y": [30, 40]})
+    long = df.unpivot(["id"])
+    wide = long.pivot("id", col("variable"), col("value"), "sum",
+                      names=["x", "y"]).sort("id").to_pydict()
+    assert wide == {"id": [1, 2], "x": [10, 20], "y": [30, 40]}
+
+
+def test_from_arrow_dictionary_roundtrip():
+    import pyarrow as pa
+    arr = pa.array(["a", "b", "a", None]).dictionary_encode()
+    t = pa.table({"d": arr})
+    df = daft.from_arrow(t)
+    assert df.to_pydict() == {"d": ["a", "b", "a", None]}
+    back = df.collect()._result[0].column("d")
+    assert back.is_dict()
+    out = df.to_arrow()
+    assert pa.types.is_dictionary(out.column("d").type)
