@@ -270,9 +270,7 @@ Tensor merge_strings(Tensor mask, Tensor t_off, Tensor t_bytes, Tensor f_off,
 __global__ void groupby_insert_kernel(const uint64_t* hashes,
                                       const ColDesc* cols, int ncols,
                                       int64_t n, int64_t* table,
-                                      uint64_t mask, int64_t* row_slot,
-                                      int32_t* slot_gid, int64_t* reps,
-                                      int32_t* counter) {
+                                      uint64_t mask, int64_t* row_slot) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -289,9 +287,6 @@ __global__ void groupby_insert_kernel(const uint64_t* hashes,
                          (unsigned long long)(-1ll),
                          (unsigned long long)i);
         if (prev == -1ll) {  // claimed: i is the representative
-          int32_t gid = atomicAdd(counter, 1);
-          slot_gid[slot] = gid;
-          reps[gid] = i;
           row_slot[i] = (int64_t)slot;
           break;
         }
@@ -302,6 +297,22 @@ __global__ void groupby_insert_kernel(const uint64_t* hashes,
         break;
       }
       slot = (slot + 1) & mask;
+    }
+  }
+}
+
+__global__ void groupby_assign_ids_kernel(const int64_t* table,
+                                          const int64_t* row_slot, int64_t n,
+                                          int32_t* slot_gid, int64_t* reps,
+                                          int32_t* counter) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t slot = row_slot[i];
+    if (table[slot] == i) {
+      int32_t gid = atomicAdd(counter, 1);
+      slot_gid[slot] = gid;
+      reps[gid] = i;
     }
   }
 }
@@ -335,16 +346,19 @@ std::vector<Tensor> groupby(Tensor hashes, const std::vector<int64_t>& tags,
   int64_t cap = table_capacity(n);
   auto table = torch::full({cap}, -1, opts64);
   auto row_slot = torch::empty({n}, opts64);
-  auto slot_gid = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
-  auto reps_full = torch::empty({n}, opts64);
-  auto counter = torch::zeros({1}, torch::dtype(torch::kInt32).device(dev));
   int block = 256;
   hipLaunchKernelGGL(groupby_insert_kernel, dim3(grid_1d(n, block)),
                      dim3(block), 0, cur_stream(),
                      (const uint64_t*)hashes.data_ptr<int64_t>(),
                      (const ColDesc*)descs.data_ptr(), (int)tags.size(), n,
                      table.data_ptr<int64_t>(), (uint64_t)(cap - 1),
-                     row_slot.data_ptr<int64_t>(),
+                     row_slot.data_ptr<int64_t>());
+  auto slot_gid = torch::empty({cap}, torch::dtype(torch::kInt32).device(dev));
+  auto reps_full = torch::empty({n}, opts64);
+  auto counter = torch::zeros({1}, torch::dtype(torch::kInt32).device(dev));
+  hipLaunchKernelGGL(groupby_assign_ids_kernel, dim3(grid_1d(n, block)),
+                     dim3(block), 0, cur_stream(), table.data_ptr<int64_t>(),
+                     row_slot.data_ptr<int64_t>(), n,
                      slot_gid.data_ptr<int32_t>(),
                      reps_full.data_ptr<int64_t>(),
                      counter.data_ptr<int32_t>());
