@@ -107,6 +107,23 @@ def exchange_batches(parts: List[RecordBatch]) -> RecordBatch:
     if backend() != "nccl":
         recv = _object_a2a(parts)
         return RecordBatch.concat(recv)
+    # The buffer layout must agree across ranks (dictionary-encoded columns
+    # add a vocab child).  Structures only diverge through data-dependent
+    # decode fallbacks; agree on a canonical structure first (tiny
+    # control-plane collective), decoding dicts everywhere on mismatch.
+    def skel(rb: RecordBatch):
+        def srec(s: Series):
+            return (repr(s.dtype), s.is_dict(),
+                    tuple(srec(c) for c in s.children))
+        return tuple(srec(c) for c in rb.columns)
+
+    my_skel = skel(parts[0])
+    skels: List[object] = [None] * w
+    dist.all_gather_object(skels, my_skel)
+    if any(sk != my_skel for sk in skels):
+        parts = [RecordBatch(
+            [c.dict_decode() if c.is_dict() else c for c in p.columns],
+            len(p)) for p in parts]
     proto = parts[0]
     flat = [flatten_batch(p) for p in parts]
     nbuf = len(flat[0])
@@ -130,25 +147,29 @@ def exchange_batches(parts: List[RecordBatch]) -> RecordBatch:
             out_bufs.append([None] * w)
             continue
         dtype = anyt.dtype if anyt is not None else torch.uint8
+        # RCCL has no unsigned-wide/bool dtypes: run the collective through
+        # a bit-identical signed/uint8 view
+        wire = {torch.bool: torch.uint8, torch.uint16: torch.int16,
+                torch.uint32: torch.int32,
+                torch.uint64: torch.int64}.get(dtype, dtype)
         send_parts = []
         in_splits = []
         for p in range(w):
             t = flat[p][b]
             if t is None:
                 t = torch.zeros(0, dtype=dtype, device=dev)
-            send_parts.append(t.contiguous().view(-1))
+            t = t.contiguous().view(-1)
+            if wire != dtype:
+                t = t.view(wire)
+            send_parts.append(t)
             in_splits.append(t.numel())
         sendbuf = torch.cat(send_parts) if send_parts else \
-            torch.zeros(0, dtype=dtype, device=dev)
+            torch.zeros(0, dtype=wire, device=dev)
         out_splits = [max(0, int(col_recv[p].item())) for p in range(w)]
-        recvbuf = torch.empty(sum(out_splits), dtype=dtype, device=dev)
-        # RCCL all-to-all over xGMI (bool tensors go as uint8)
-        if dtype == torch.bool:
-            dist.all_to_all_single(recvbuf.view(torch.uint8),
-                                   sendbuf.view(torch.uint8),
-                                   out_splits, in_splits)
-        else:
-            dist.all_to_all_single(recvbuf, sendbuf, out_splits, in_splits)
+        recvbuf = torch.empty(sum(out_splits), dtype=wire, device=dev)
+        dist.all_to_all_single(recvbuf, sendbuf, out_splits, in_splits)
+        if wire != dtype:
+            recvbuf = recvbuf.view(dtype)
         pieces = []
         off = 0
         for p in range(w):
