@@ -40,6 +40,7 @@ def main() -> None:
         dist = tdist
         backend = "nccl" if on_gpu else "gloo"
         if on_gpu:
+            local_rank = local_rank % torch.cuda.device_count()
             torch.cuda.set_device(local_rank)
         tdist.init_process_group(backend=backend, rank=rank,
                                  world_size=world)
