@@ -106,13 +106,16 @@ def compact_indices(mask: Series) -> torch.Tensor:
     return torch.nonzero(m, as_tuple=False).reshape(-1).to(torch.int64)
 
 
-def take(s: Series, indices: torch.Tensor) -> Series:
-    """Gather rows; index -1 yields null."""
+def take(s: Series, indices: torch.Tensor,
+         has_neg: Optional[bool] = None) -> Series:
+    """Gather rows; index -1 yields null.  `has_neg` lets multi-column
+    callers hoist the device sync for the negative-index check."""
     k = s.dtype.kind
     n_out = int(indices.shape[0])
     dev = s.device
     indices = indices.to(dev)
-    has_neg = bool((indices < 0).any().item()) if n_out else False
+    if has_neg is None:
+        has_neg = bool((indices < 0).any().item()) if n_out else False
     safe_idx = indices.clamp(min=0) if has_neg else indices
 
     validity = None
@@ -163,7 +166,7 @@ def take(s: Series, indices: torch.Tensor) -> Series:
         return Series(s.name, s.dtype, children=[child], validity=validity,
                       length=n_out)
     if k == TypeKind.STRUCT:
-        children = [c.take(safe_idx) for c in s.children]
+        children = [c.take(safe_idx, has_neg=False) for c in s.children]
         return Series(s.name, s.dtype, children=children, validity=validity,
                       length=n_out)
     # fixed width: torch gather (hipified index_select is memory-bound optimal
@@ -393,13 +396,28 @@ _COMPARE_TORCH = {"eq": torch.eq, "ne": torch.ne, "lt": torch.lt,
 
 
 def _align(l: Series, r: Series):
+    """Length-align operands.  Length-1 fixed-width operands stay length-1
+    (torch elementwise ops broadcast [1] against [n] — materializing the
+    literal into a full column costs a pointless HBM round trip)."""
     if len(l) == len(r):
         return l, r
     if len(l) == 1:
+        if l.dtype.is_fixed_width() and l.data is not None:
+            return l, r
         return l.broadcast(len(r)), r
     if len(r) == 1:
+        if r.dtype.is_fixed_width() and r.data is not None:
+            return l, r
         return l, r.broadcast(len(l))
     raise ValueError(f"length mismatch {len(l)} vs {len(r)}")
+
+
+
+def _expand_validity(validity, n: int):
+    """Broadcast a length-1 validity to the output length (null literal)."""
+    if validity is not None and validity.numel() == 1 and n != 1:
+        return validity.expand(n).contiguous()
+    return validity
 
 
 def binary_op(l: Series, r: Series, op: str) -> Series:
@@ -429,7 +447,8 @@ def binary_op(l: Series, r: Series, op: str) -> Series:
         out = torch.pow(a, b)
     else:
         raise ValueError(f"unknown binary op {op}")
-    return Series(l.name, out_dt, data=out, validity=validity)
+    return Series(l.name, out_dt, data=out,
+                  validity=_expand_validity(validity, int(out.shape[0])))
 
 
 def _temporal_binary(l: Series, r: Series, op: str, validity):
@@ -437,16 +456,21 @@ def _temporal_binary(l: Series, r: Series, op: str, validity):
     if op == "sub" and lk == rk == TypeKind.DATE:
         out = (l.data.to(torch.int64) - r.data.to(torch.int64))
         return Series(l.name, DataType.duration("us"),
-                      data=out * 86_400_000_000, validity=validity)
+                      data=out * 86_400_000_000,
+                      validity=_expand_validity(validity, int(out.shape[0])))
     if lk == TypeKind.DATE and r.dtype.is_integer():
         out = l.data + r.data.to(torch.int32) * (1 if op == "add" else -1)
-        return Series(l.name, l.dtype, data=out, validity=validity)
+        return Series(l.name, l.dtype, data=out,
+                      validity=_expand_validity(validity, int(out.shape[0])))
     if lk == TypeKind.TIMESTAMP and rk == TypeKind.DURATION:
         out = l.data + r.data * (1 if op == "add" else -1)
-        return Series(l.name, l.dtype, data=out, validity=validity)
+        return Series(l.name, l.dtype, data=out,
+                      validity=_expand_validity(validity, int(out.shape[0])))
     if op == "sub" and lk == rk == TypeKind.TIMESTAMP:
+        out = l.data - r.data
         return Series(l.name, DataType.duration(l.dtype.timeunit),
-                      data=l.data - r.data, validity=validity)
+                      data=out,
+                      validity=_expand_validity(validity, int(out.shape[0])))
     raise TypeError(f"temporal op {op} on {l.dtype!r}, {r.dtype!r}")
 
 
@@ -491,7 +515,8 @@ def compare_op(l: Series, r: Series, op: str) -> Series:
         return t.to(tdt)
     a, b = conv(l.data), conv(r.data)
     out = _COMPARE_TORCH[op](a, b)
-    return Series(l.name, DataType.bool(), data=out, validity=validity)
+    return Series(l.name, DataType.bool(), data=out,
+                  validity=_expand_validity(validity, int(out.shape[0])))
 
 
 def _same_vocab(a: Series, b: Series) -> bool:
@@ -563,7 +588,8 @@ def logical_op(l: Series, r: Series, op: str) -> Series:
         raise ValueError(op)
     if bool(validity.all().item()):
         validity = None
-    return Series(l.name, DataType.bool(), data=out, validity=validity)
+    return Series(l.name, DataType.bool(), data=out,
+                  validity=_expand_validity(validity, int(out.shape[0])))
 
 
 def logical_not(s: Series) -> Series:

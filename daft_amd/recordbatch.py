@@ -115,7 +115,11 @@ class RecordBatch:
 
     # ------------------------------------------------------------------
     def take(self, indices: torch.Tensor) -> "RecordBatch":
-        return RecordBatch([c.take(indices) for c in self.columns],
+        # hoist the negative-index host sync across columns
+        has_neg = bool((indices < 0).any().item()) if indices.numel() else \
+            False
+        return RecordBatch([c.take(indices, has_neg=has_neg)
+                            for c in self.columns],
                            int(indices.shape[0]))
 
     def filter(self, mask: Series) -> "RecordBatch":

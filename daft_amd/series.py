@@ -318,10 +318,11 @@ class Series:
     # ------------------------------------------------------------------
     # selection ops (HIP kernels on GPU — ref: daft-core array/ops/{filter,take,concat}.rs)
     # ------------------------------------------------------------------
-    def take(self, indices: torch.Tensor) -> "Series":
+    def take(self, indices: torch.Tensor,
+             has_neg: Optional[bool] = None) -> "Series":
         """Gather rows by index; index -1 produces null."""
         from . import kernels
-        return kernels.take(self, indices)
+        return kernels.take(self, indices, has_neg=has_neg)
 
     def filter(self, mask: "Series") -> "Series":
         from . import kernels
