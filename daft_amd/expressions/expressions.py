@@ -97,8 +97,11 @@ class Literal(ExprNode):
         return Field("literal", self.dtype)
 
     def evaluate(self, batch) -> Series:
-        s = lit_series("literal", self.value, self.dtype, device=batch.device)
-        return s.broadcast(len(batch)) if len(batch) != 1 else s
+        # stays length-1: elementwise consumers broadcast without
+        # materializing a full column (operators that need full length
+        # broadcast explicitly)
+        return lit_series("literal", self.value, self.dtype,
+                          device=batch.device)
 
     def __repr__(self):
         return f"lit({self.value!r})"

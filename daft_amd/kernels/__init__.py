@@ -572,6 +572,10 @@ def logical_op(l: Series, r: Series, op: str) -> Series:
     l, r = _align(l, r)
     a = l.data.to(torch.bool)
     b = r.data.to(torch.bool)
+    if l.validity is None and r.validity is None:
+        # non-null fast path: plain boolean kernel, no validity algebra
+        out = a & b if op == "and" else (a | b if op == "or" else a ^ b)
+        return Series(l.name, DataType.bool(), data=out)
     av = l.validity if l.validity is not None else torch.ones_like(a)
     bv = r.validity if r.validity is not None else torch.ones_like(b)
     if op == "and":
