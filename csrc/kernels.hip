@@ -539,7 +539,8 @@ std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
                       torch::dtype(torch::kInt64).device(dev));
   }
   if (n > 0) {
-    const bool* vp = valid.defined() ? valid.data_ptr<bool>() : nullptr;
+    const bool* vp = (valid.defined() && valid.numel() > 0)
+                         ? valid.data_ptr<bool>() : nullptr;
     const int64_t* g = group_ids.data_ptr<int64_t>();
     if (is_f64) {
       const double* v = values.data_ptr<double>();

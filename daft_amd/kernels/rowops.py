@@ -178,8 +178,11 @@ def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
 
     if _is_gpu(values) and op in ("sum", "min", "max"):
         vmask = validity if validity is not None else \
-            torch.ones(len(values), dtype=torch.bool, device=dev)
+            torch.empty(0, dtype=torch.bool, device=dev)
         out, cnt = native_required().grouped_agg(gid, num_groups, d, vmask, op)
+        if validity is None and cnt is not None:
+            # kernel counted every row (no validity): cnt is still exact
+            pass
         return out, cnt
 
     # CPU fallback via scatter_reduce

@@ -159,6 +159,54 @@ def _split_conjunctions(e: ExprNode) -> List[ExprNode]:
     return [e]
 
 
+def _split_disjunctions(e: ExprNode) -> List[ExprNode]:
+    if isinstance(e, BinaryOp) and e.op == "or":
+        return _split_disjunctions(e.left) + _split_disjunctions(e.right)
+    return [e]
+
+
+def _derive_or_implications(preds: List[ExprNode]) -> List[ExprNode]:
+    """For `(a=x and ...) or (a=y and ...)`, every branch pins column `a`,
+    so `a IN (x, y)` is implied — derive it as an extra conjunct that CAN be
+    pushed below joins the OR itself cannot cross (the TPC-H Q7/Q19 shape).
+    """
+    derived: List[ExprNode] = []
+    for p in preds:
+        branches = _split_disjunctions(p)
+        if len(branches) < 2:
+            continue
+        # column -> set of pinned literal values per branch
+        per_branch = []
+        for b in branches:
+            pins = {}
+            for cj in _split_conjunctions(b):
+                if isinstance(cj, BinaryOp) and cj.op == "eq":
+                    cr, litv = None, None
+                    if isinstance(cj.left, ColumnRef) and                             isinstance(cj.right, Literal):
+                        cr, litv = cj.left, cj.right
+                    elif isinstance(cj.right, ColumnRef) and                             isinstance(cj.left, Literal):
+                        cr, litv = cj.right, cj.left
+                    if cr is not None and isinstance(
+                            litv.value, (int, float, str, bool)):
+                        pins.setdefault(cr.name, set()).add(litv.value)
+                elif isinstance(cj, IsIn):
+                    base = cj.child
+                    if isinstance(base, ColumnRef) and all(
+                            isinstance(v, (int, float, str, bool))
+                            for v in cj.values):
+                        pins.setdefault(base.name, set()).update(cj.values)
+            per_branch.append(pins)
+        common = set(per_branch[0])
+        for pb in per_branch[1:]:
+            common &= set(pb)
+        for name in common:
+            vals = set()
+            for pb in per_branch:
+                vals |= pb[name]
+            derived.append(IsIn(ColumnRef(name), sorted(vals, key=repr)))
+    return derived
+
+
 def _conjoin(es: List[ExprNode]) -> ExprNode:
     out = es[0]
     for e in es[1:]:
@@ -186,6 +234,13 @@ def push_down_filter(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
         return None
     child = plan.children[0]
     preds = _split_conjunctions(plan.predicate)
+    if not getattr(plan, "_or_derived", False):
+        extra = _derive_or_implications(preds)
+        if extra:
+            new = lp.Filter(child, _conjoin(preds + extra))
+            new._or_derived = True
+            return new
+        plan._or_derived = True
 
     if isinstance(child, lp.Filter):
         # merge adjacent filters
