@@ -34,6 +34,7 @@ std::vector<Tensor> groupby(Tensor hashes, const std::vector<int64_t>& tags,
 std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
                                 Tensor values, Tensor valid,
                                 const std::string& op);
+Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid);
 
 // join
 std::vector<Tensor> join_build(Tensor hashes);

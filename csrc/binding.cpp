@@ -11,6 +11,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("merge_strings", &merge_strings, "if_else merge of string columns");
   m.def("groupby", &groupby, "hash groupby -> (group_ids, rep_idx)");
   m.def("grouped_agg", &grouped_agg, "per-group sum/min/max + valid counts");
+  m.def("grouped_count", &grouped_count, "per-group row counts (LDS staged)");
   m.def("join_build", &join_build, "bucket-chain hash join build");
   m.def("join_probe", &join_probe, "hash join probe -> (lidx, ridx, matched)");
   m.def("radix_argsort", &radix_argsort, "stable LSD radix argsort of u64");

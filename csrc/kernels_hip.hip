@@ -581,6 +581,59 @@ std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
   return {out, cnt};
 }
 
+__global__ void grouped_count_lds_kernel(const int64_t* gids,
+                                         const bool* valid, int64_t n,
+                                         int64_t num_groups, int64_t* cnt) {
+  extern __shared__ uint32_t lcnt[];
+  for (int64_t g = threadIdx.x; g < num_groups; g += blockDim.x) lcnt[g] = 0;
+  __syncthreads();
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (valid && !valid[i]) continue;
+    atomicAdd(&lcnt[gids[i]], 1u);
+  }
+  __syncthreads();
+  for (int64_t g = threadIdx.x; g < num_groups; g += blockDim.x)
+    if (lcnt[g])
+      atomicAdd((unsigned long long*)&cnt[g], (unsigned long long)lcnt[g]);
+}
+
+__global__ void grouped_count_global_kernel(const int64_t* gids,
+                                            const bool* valid, int64_t n,
+                                            int64_t* cnt) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (valid && !valid[i]) continue;
+    atomicAdd((unsigned long long*)&cnt[gids[i]], 1ull);
+  }
+}
+
+Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid) {
+  auto dev = group_ids.device();
+  int64_t n = group_ids.numel();
+  auto cnt = torch::zeros({num_groups},
+                          torch::dtype(torch::kInt64).device(dev));
+  if (n == 0) return cnt;
+  const bool* vp = (valid.defined() && valid.numel() > 0)
+                       ? valid.data_ptr<bool>() : nullptr;
+  int block = 256;
+  int64_t lds = num_groups * sizeof(uint32_t);
+  if (num_groups <= 8192 && lds <= 48 * 1024) {
+    hipLaunchKernelGGL(grouped_count_lds_kernel,
+                       dim3(grid_1d(n, block, 8)), dim3(block), lds,
+                       cur_stream(), group_ids.data_ptr<int64_t>(), vp, n,
+                       num_groups, cnt.data_ptr<int64_t>());
+  } else {
+    hipLaunchKernelGGL(grouped_count_global_kernel,
+                       dim3(grid_1d(n, block)), dim3(block), 0, cur_stream(),
+                       group_ids.data_ptr<int64_t>(), vp, n,
+                       cnt.data_ptr<int64_t>());
+  }
+  return cnt;
+}
+
 // ---------------------------------------------------------------------------
 // hash join: bucket-chain build + verified probe
 // ---------------------------------------------------------------------------

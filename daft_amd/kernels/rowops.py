@@ -149,19 +149,21 @@ def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
     """
     dev = values.device
     gid = group_ids
-    if op == "count":
-        out = torch.zeros(num_groups, dtype=torch.int64, device=dev)
-        out.scatter_add_(0, gid, torch.ones_like(gid))
-        return out, None
-    data = values.data
-    validity = values.validity
-    if op == "count_valid":
+    if op in ("count", "count_valid"):
+        validity = values.validity if op == "count_valid" else None
+        if _is_gpu(values):
+            vmask = validity if validity is not None else \
+                torch.empty(0, dtype=torch.bool, device=dev)
+            out = native_required().grouped_count(gid, num_groups, vmask)
+            return out, None
         ones = torch.ones(len(values), dtype=torch.int64, device=dev)
         if validity is not None:
             ones = ones * validity.to(torch.int64)
         out = torch.zeros(num_groups, dtype=torch.int64, device=dev)
         out.scatter_add_(0, gid, ones)
         return out, None
+    data = values.data
+    validity = values.validity
 
     if values.dtype.is_floating() or values.dtype.is_decimal():
         wdt = torch.float64
