@@ -122,6 +122,9 @@ def groupby(keys: Sequence[Series]) -> Tuple[torch.Tensor, torch.Tensor]:
     values represent group g (for gathering output key columns).
     """
     if _is_gpu(keys[0]):
+        dense = _dense_range_groupby(keys)
+        if dense is not None:
+            return dense
         hashes = hash_columns(keys)
         tags, datas, offs, vals = _descs(keys)
         return native_required().groupby(hashes, tags, datas, offs, vals)
@@ -138,6 +141,49 @@ def groupby(keys: Sequence[Series]) -> Tuple[torch.Tensor, torch.Tensor]:
         gids[i] = g
     return (torch.from_numpy(gids),
             torch.tensor(reps, dtype=torch.int64))
+
+
+_DENSE_RANGE_LIMIT = 1 << 28  # map memory cap: 256M entries (2 GB int64)
+
+
+def _dense_range_groupby(keys):
+    """Dense-range fast path: when every key is a no-null integer (or dict
+    code / date / bool) and the product of value ranges is small, pack keys
+    into one index and densify with two scatter passes — no hash table.
+    The common analytics shape (dense surrogate keys, dict codes)."""
+    packed = None
+    rng_prod = 1
+    n = len(keys[0])
+    if n == 0:
+        return None
+    for s in keys:
+        if s.validity is not None:
+            return None
+        if s.is_dict():
+            k = s.data.to(torch.int64)
+        elif s.data is not None and s.dtype.is_fixed_width() and \
+                s.data.dtype != torch.uint64 and (
+                s.dtype.is_integer() or s.dtype.is_boolean() or
+                s.dtype.kind in (TypeKind.DATE,)):
+            k = s.data.to(torch.int64)
+        else:
+            return None
+        mn = int(k.min().item())
+        mx = int(k.max().item())
+        rng = mx - mn + 1
+        if rng <= 0 or rng_prod * rng > _DENSE_RANGE_LIMIT:
+            return None
+        packed = (k - mn) if packed is None else packed * rng + (k - mn)
+        rng_prod *= rng
+    dev = packed.device
+    idx = torch.arange(n, dtype=torch.int64, device=dev)
+    first = torch.full((rng_prod,), n, dtype=torch.int64, device=dev)
+    first.scatter_reduce_(0, packed, idx, reduce="amin")
+    present = first < n
+    ids_map = torch.cumsum(present.to(torch.int64), 0) - 1
+    gids = ids_map[packed]
+    reps = first[present]
+    return gids, reps
 
 
 def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
