@@ -338,3 +338,33 @@ def test_embed_image_gpu():
                                dimensions=64).alias("e"))
            .to_pydict()["e"])
     assert len(out[0]) == 64
+
+
+def test_dense_range_groupby_matches_hash():
+    torch.manual_seed(33)
+    n = 400_000
+    # dense int keys (dense-range fast path) vs string keys (hash path)
+    k = torch.randint(0, 50_000, (n,))
+    v = torch.rand(n, dtype=torch.float64)
+    got = daft.from_pydict({"k": k, "v": v}, device="cuda:0") \
+        .groupby("k").agg(col("v").sum().alias("s"),
+                          col("v").count().alias("c")) \
+        .sort("k").to_pydict()
+    want = daft.from_pydict({"k": k, "v": v}, device="cpu") \
+        .groupby("k").agg(col("v").sum().alias("s"),
+                          col("v").count().alias("c")) \
+        .sort("k").to_pydict()
+    assert got["k"] == want["k"] and got["c"] == want["c"]
+    assert got["s"] == pytest.approx(want["s"], rel=1e-9)
+
+
+def test_groupby_wide_range_falls_back_to_hash():
+    # range too large for the dense path: must still be correct
+    torch.manual_seed(34)
+    k = torch.randint(-2**60, 2**60, (50_000,))
+    k = torch.cat([k, k])  # ensure duplicates
+    got = daft.from_pydict({"k": k}, device="cuda:0") \
+        .groupby("k").agg(col("k").count().alias("c")).to_pydict()
+    assert sorted(got["c"]) == sorted(
+        daft.from_pydict({"k": k}, device="cpu")
+        .groupby("k").agg(col("k").count().alias("c")).to_pydict()["c"])
