@@ -35,6 +35,7 @@ std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
                                 Tensor values, Tensor valid,
                                 const std::string& op);
 Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid);
+Tensor dense_first_index(Tensor packed, int64_t rng, int64_t sentinel);
 
 // join
 std::vector<Tensor> join_build(Tensor hashes);

@@ -12,6 +12,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("groupby", &groupby, "hash groupby -> (group_ids, rep_idx)");
   m.def("grouped_agg", &grouped_agg, "per-group sum/min/max + valid counts");
   m.def("grouped_count", &grouped_count, "per-group row counts (LDS staged)");
+  m.def("dense_first_index", &dense_first_index,
+        "first-occurrence index per packed dense key");
   m.def("join_build", &join_build, "bucket-chain hash join build");
   m.def("join_probe", &join_probe, "hash join probe -> (lidx, ridx, matched)");
   m.def("radix_argsort", &radix_argsort, "stable LSD radix argsort of u64");

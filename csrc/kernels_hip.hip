@@ -634,6 +634,58 @@ Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid) {
   return cnt;
 }
 
+// dense-range groupby support: first-occurrence index per packed key
+// (LDS-staged atomicMin for small ranges — global atomics on a handful of
+// addresses would serialize)
+__global__ void dense_first_lds_kernel(const int64_t* packed, int64_t n,
+                                       int64_t rng, int64_t* first) {
+  extern __shared__ uint32_t lmin[];
+  for (int64_t g = threadIdx.x; g < rng; g += blockDim.x)
+    lmin[g] = 0xFFFFFFFFu;
+  __syncthreads();
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    atomicMin(&lmin[packed[i]], (uint32_t)i);
+  __syncthreads();
+  for (int64_t g = threadIdx.x; g < rng; g += blockDim.x)
+    if (lmin[g] != 0xFFFFFFFFu)
+      atomicMin((unsigned long long*)&first[g],
+                (unsigned long long)lmin[g]);
+}
+
+__global__ void dense_first_global_kernel(const int64_t* packed, int64_t n,
+                                          int64_t* first) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    atomicMin((unsigned long long*)&first[packed[i]],
+              (unsigned long long)i);
+}
+
+Tensor dense_first_index(Tensor packed, int64_t rng, int64_t sentinel) {
+  auto dev = packed.device();
+  int64_t n = packed.numel();
+  auto first = torch::full({rng}, sentinel,
+                           torch::dtype(torch::kInt64).device(dev));
+  if (n == 0) return first;
+  TORCH_CHECK(n < (int64_t)0xFFFFFFFF, "dense_first_index: n must fit u32");
+  int block = 256;
+  int64_t lds = rng * sizeof(uint32_t);
+  if (rng <= 8192 && lds <= 48 * 1024) {
+    hipLaunchKernelGGL(dense_first_lds_kernel, dim3(grid_1d(n, block, 8)),
+                       dim3(block), lds, cur_stream(),
+                       packed.data_ptr<int64_t>(), n, rng,
+                       first.data_ptr<int64_t>());
+  } else {
+    hipLaunchKernelGGL(dense_first_global_kernel,
+                       dim3(grid_1d(n, block)), dim3(block), 0, cur_stream(),
+                       packed.data_ptr<int64_t>(), n,
+                       first.data_ptr<int64_t>());
+  }
+  return first;
+}
+
 // ---------------------------------------------------------------------------
 // hash join: bucket-chain build + verified probe
 // ---------------------------------------------------------------------------

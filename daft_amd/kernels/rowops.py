@@ -175,10 +175,7 @@ def _dense_range_groupby(keys):
             return None
         packed = (k - mn) if packed is None else packed * rng + (k - mn)
         rng_prod *= rng
-    dev = packed.device
-    idx = torch.arange(n, dtype=torch.int64, device=dev)
-    first = torch.full((rng_prod,), n, dtype=torch.int64, device=dev)
-    first.scatter_reduce_(0, packed, idx, reduce="amin")
+    first = native_required().dense_first_index(packed, rng_prod, n)
     present = first < n
     ids_map = torch.cumsum(present.to(torch.int64), 0) - 1
     gids = ids_map[packed]
