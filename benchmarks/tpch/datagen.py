@@ -477,8 +477,24 @@ def generate(sf: float, device="cpu", rank: int = 0,
     return out
 
 
+_PARTITIONING = {
+    # rank shards are key ranges; equal keys are colocated per family
+    "orders": ("tpch_orderkey", ["o_orderkey"]),
+    "lineitem": ("tpch_orderkey", ["l_orderkey"]),
+    "part": ("tpch_partkey", ["p_partkey"]),
+    "partsupp": ("tpch_partkey", ["ps_partkey"]),
+    "supplier": ("tpch_suppkey", ["s_suppkey"]),
+    "customer": ("tpch_custkey", ["c_custkey"]),
+}
+
+
 def dataframes(sf: float, device="cpu", rank: int = 0, world: int = 1):
-    """Tables as daft_amd DataFrames."""
+    """Tables as daft_amd DataFrames (with declared co-partitioning when
+    sharded across ranks)."""
     from daft_amd.io import from_recordbatches
     tables = generate(sf, device, rank, world)
-    return {name: from_recordbatches([rb]) for name, rb in tables.items()}
+    out = {}
+    for name, rb in tables.items():
+        part = _PARTITIONING.get(name) if world > 1 else None
+        out[name] = from_recordbatches([rb], partitioning=part)
+    return out

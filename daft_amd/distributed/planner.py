@@ -20,10 +20,11 @@ from . import plan_nodes as dn
 
 SHARDED = ("sharded",)
 RANK0 = ("rank0",)
+_RCCL = "__rccl_hash"  # token of planner-inserted hash exchanges
 
 
-def _hash_state(names: List[str]):
-    return ("hash", tuple(sorted(names))) if names else SHARDED
+def _hash_state(names: List[str], token: str = _RCCL):
+    return ("hash", tuple(sorted(names)), token) if names else SHARDED
 
 
 def _key_names(exprs: List[ExprNode], schema) -> Optional[List[str]]:
@@ -46,6 +47,9 @@ def distribute(plan: lp.LogicalPlan, world: int,
 def _rewrite(node: lp.LogicalPlan, world: int,
              rank: int) -> Tuple[lp.LogicalPlan, tuple]:
     if isinstance(node, lp.Source):
+        if node.partitioning:
+            token, keys = node.partitioning
+            return node, _hash_state(list(keys), token)
         return node, SHARDED
     if isinstance(node, lp.ScanSource):
         # split files across ranks
@@ -211,11 +215,20 @@ def _rewrite_join(node: lp.Join, states):
         return node.with_children([left, rep]), states[0]
 
     lst, rst = states
-    lc = left
-    rc = right
-    if not (lst[0] == "hash" and lnames and set(lst[1]) == set(lnames)):
-        lc = dn.ExchangeByKey(left, lkeys)
-    if not (rst[0] == "hash" and rnames and set(rst[1]) == set(rnames)):
-        rc = dn.ExchangeByKey(right, rkeys)
-    out_state = _hash_state(lnames or [])
+    l_ok = lst[0] == "hash" and lnames and set(lst[1]) == set(lnames)
+    r_ok = rst[0] == "hash" and rnames and set(rst[1]) == set(rnames)
+    same_dist = l_ok and r_ok and lst[2] == rst[2]
+    lc, rc = left, right
+    if not same_dist:
+        # both sides must land under the SAME distribution; reuse one side's
+        # planner hash partitioning when present
+        if l_ok and lst[2] == _RCCL:
+            rc = dn.ExchangeByKey(right, rkeys)
+        elif r_ok and rst[2] == _RCCL:
+            lc = dn.ExchangeByKey(left, lkeys)
+        else:
+            lc = dn.ExchangeByKey(left, lkeys)
+            rc = dn.ExchangeByKey(right, rkeys)
+    out_state = ("hash", tuple(sorted(lnames)), lst[2] if same_dist
+                 else _RCCL) if lnames else SHARDED
     return node.with_children([lc, rc]), out_state

@@ -29,14 +29,17 @@ def from_pydict(data: Dict[str, list], device=None) -> DataFrame:
         rb.schema, key, len(rb), rb.size_bytes()))
 
 
-def from_recordbatches(batches: List[RecordBatch]) -> DataFrame:
+def from_recordbatches(batches: List[RecordBatch],
+                       partitioning=None) -> DataFrame:
+    """`partitioning=(token, [key_cols])` declares rank-colocation of equal
+    keys under a named distribution (skips distributed exchanges)."""
     ctx = get_context()
     key = ctx.cache.new_key()
     ctx.cache.put(key, batches)
     rows = sum(len(b) for b in batches)
     size = sum(b.size_bytes() for b in batches)
     return DataFrame(LogicalPlanBuilder.from_in_memory(
-        batches[0].schema, key, rows, size))
+        batches[0].schema, key, rows, size, partitioning))
 
 
 def from_arrow(table, device=None) -> DataFrame:

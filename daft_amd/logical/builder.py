@@ -17,9 +17,11 @@ class LogicalPlanBuilder:
     # ------------------------------------------------------------------
     @staticmethod
     def from_in_memory(schema: Schema, cache_key: str, num_rows: int,
-                       size_bytes: int = 0) -> "LogicalPlanBuilder":
+                       size_bytes: int = 0,
+                       partitioning=None) -> "LogicalPlanBuilder":
         return LogicalPlanBuilder(
-            lp.Source(schema, cache_key, num_rows, size_bytes))
+            lp.Source(schema, cache_key, num_rows, size_bytes,
+                      partitioning))
 
     @staticmethod
     def from_scan(schema: Schema, paths: List[str], file_format: str,

@@ -60,15 +60,21 @@ class LogicalPlan:
 
 
 class Source(LogicalPlan):
-    """In-memory source: holds partition refs (MicroPartition cache keys)."""
+    """In-memory source: holds partition refs (MicroPartition cache keys).
+
+    `partitioning`: optional (token, key_names) declaring that rows with
+    equal key values are colocated on one rank under distribution `token`
+    (ref capability: partition specs on scans/tables; lets the distributed
+    planner skip exchanges for co-sharded tables)."""
 
     def __init__(self, schema: Schema, cache_key: str, num_rows: int,
-                 size_bytes: int = 0):
+                 size_bytes: int = 0, partitioning=None):
         super().__init__([])
         self._schema = schema
         self.cache_key = cache_key
         self.num_rows = num_rows
         self.size_bytes = size_bytes
+        self.partitioning = partitioning
 
     def _compute_schema(self):
         return self._schema

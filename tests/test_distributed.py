@@ -158,7 +158,7 @@ def _tpch_checks(rank, world):
     sf = 0.01
     T = datagen.dataframes(sf, device="cpu", rank=rank, world=world)
     results = {}
-    for qi in (1, 3, 4, 5, 6, 10, 12, 13, 14, 18, 19, 22):
+    for qi in range(1, 23):
         out = queries.run_query(qi, T, sf=sf).to_pydict()
         results[qi] = out
     return pickle.dumps(results)
