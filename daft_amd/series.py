@@ -498,8 +498,9 @@ def _from_pylist_typed(name: str, values: list, dtype: DataType) -> Series:
             bufs.append(b)
             pos += len(b)
             offs[i + 1] = pos
-        data = torch.frombuffer(bytearray(b"".join(bufs)), dtype=torch.uint8) \
-            if bufs else torch.zeros(0, dtype=torch.uint8)
+        blob = b"".join(bufs)
+        data = torch.frombuffer(bytearray(blob), dtype=torch.uint8) \
+            if blob else torch.zeros(0, dtype=torch.uint8)
         return Series(name, dtype, data=data, validity=validity,
                       offsets=torch.from_numpy(offs))
     if k == TypeKind.DATE:

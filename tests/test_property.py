@@ -1,0 +1,115 @@
+"""Property-based tests (hypothesis) for sort/groupby/join invariants
+(ref pattern: tests/property_based_testing/ in the reference)."""
+import math
+
+import hypothesis.strategies as st
+import pytest
+from hypothesis import given, settings
+
+import daft_amd as daft
+from daft_amd import col
+
+_scalars = st.one_of(
+    st.none(),
+    st.integers(min_value=-2**40, max_value=2**40),
+    st.floats(allow_nan=False, allow_infinity=False, width=64),
+    st.text(alphabet="abcxyz ", max_size=12),
+)
+
+
+@st.composite
+def _column(draw, dtype=None):
+    n = draw(st.integers(min_value=0, max_value=40))
+    if dtype == "int":
+        elem = st.one_of(st.none(), st.integers(-1000, 1000))
+    elif dtype == "str":
+        elem = st.one_of(st.none(), st.text(alphabet="abcd", max_size=6))
+    else:
+        elem = st.one_of(st.none(), st.integers(-1000, 1000))
+    return draw(st.lists(elem, min_size=n, max_size=n))
+
+
+@given(vals=_column())
+@settings(max_examples=40, deadline=None)
+def test_sort_is_ordered_and_permutation(vals):
+    df = daft.from_pydict({"v": vals})
+    out = df.sort("v").to_pydict()["v"]
+    non_null = [v for v in out if v is not None]
+    assert non_null == sorted(non_null)
+    assert out[len(non_null):] == [None] * (len(out) - len(non_null))
+    assert sorted(map(repr, out)) == sorted(map(repr, vals))
+
+
+@given(vals=_column(dtype="str"))
+@settings(max_examples=30, deadline=None)
+def test_sort_desc_reverses(vals):
+    df = daft.from_pydict({"v": vals})
+    asc = [v for v in df.sort("v").to_pydict()["v"] if v is not None]
+    desc = [v for v in df.sort("v", desc=True).to_pydict()["v"]
+            if v is not None]
+    assert desc == list(reversed(asc))
+
+
+@given(keys=_column(dtype="int"))
+@settings(max_examples=40, deadline=None)
+def test_groupby_count_partitions_rows(keys):
+    df = daft.from_pydict({"k": keys})
+    out = df.groupby("k").agg(col("k").count("all").alias("n")).to_pydict()
+    assert sum(out["n"]) == len(keys)
+    # number of groups == number of distinct keys (nulls form one group)
+    distinct = {repr(k) for k in keys}
+    assert len(out["n"]) == len(distinct)
+
+
+@given(keys=_column(dtype="int"))
+@settings(max_examples=30, deadline=None)
+def test_groupby_sum_matches_python(keys):
+    vals = list(range(len(keys)))
+    df = daft.from_pydict({"k": keys, "v": vals})
+    out = df.groupby("k").agg(col("v").sum().alias("s")).to_pydict()
+    expect = {}
+    for k, v in zip(keys, vals):
+        expect[repr(k)] = expect.get(repr(k), 0) + v
+    got = dict(zip(map(repr, out["k"]), out["s"]))
+    assert got == expect
+
+
+@given(lk=_column(dtype="int"), rk=_column(dtype="int"))
+@settings(max_examples=30, deadline=None)
+def test_join_matches_nested_loop(lk, rk):
+    l = daft.from_pydict({"k": lk, "li": list(range(len(lk)))})
+    r = daft.from_pydict({"k": rk, "ri": list(range(len(rk)))})
+    out = l.join(r, on="k", how="inner").to_pydict()
+    got = sorted(zip(out["li"], out["ri"]))
+    want = sorted((i, j) for i, a in enumerate(lk)
+                  for j, b in enumerate(rk)
+                  if a is not None and a == b)
+    assert got == want
+
+
+@given(lk=_column(dtype="str"), rk=_column(dtype="str"))
+@settings(max_examples=20, deadline=None)
+def test_semi_anti_partition(lk, rk):
+    l = daft.from_pydict({"k": lk})
+    r = daft.from_pydict({"k": rk})
+    semi = l.join(r, on="k", how="semi").count_rows()
+    anti = l.join(r, on="k", how="anti").count_rows()
+    assert semi + anti == len(lk)
+
+
+@given(vals=_column())
+@settings(max_examples=30, deadline=None)
+def test_filter_concat_roundtrip(vals):
+    df = daft.from_pydict({"v": vals})
+    a = df.where(col("v").is_null())
+    b = df.where(col("v").not_null())
+    total = a.count_rows() + b.count_rows()
+    assert total == len(vals)
+
+
+@given(vals=st.lists(st.integers(-50, 50), max_size=30))
+@settings(max_examples=30, deadline=None)
+def test_distinct_is_set(vals):
+    df = daft.from_pydict({"v": vals})
+    out = df.distinct().to_pydict()["v"]
+    assert sorted(out) == sorted(set(vals))
