@@ -249,3 +249,22 @@ def test_from_arrow_dictionary_roundtrip():
     assert back.is_dict()
     out = df.to_arrow()
     assert pa.types.is_dictionary(out.column("d").type)
+
+
+def test_dashboard_api():
+    from daft_amd import dashboard
+    from fastapi.testclient import TestClient
+    state = dashboard.DashboardState()
+    sub = dashboard.DashboardSubscriber(state)
+    daft.attach_subscriber(sub)
+    try:
+        daft.from_pydict({"a": [1, 2]}).where(col("a") > 1).collect()
+    finally:
+        daft.detach_subscriber(sub)
+    client = TestClient(dashboard.make_app(state))
+    qs = client.get("/api/queries").json()
+    assert len(qs) == 1 and qs[0]["status"] == "done"
+    detail = client.get(f"/api/queries/{qs[0]['id']}").json()
+    assert any(op["name"] == "Filter" for op in detail["operators"])
+    html = client.get("/").text
+    assert "daft_amd queries" in html
