@@ -63,22 +63,57 @@ class Subscriber:
 
 
 class PartitionCache:
-    """In-memory partition-set cache keyed by cache_key (ref:
-    common/partitioning PartitionSetCache)."""
+    """In-memory partition-set cache keyed by cache_key, with LRU access
+    order and HBM->host spill (ref: common/partitioning PartitionSetCache +
+    resource_manager.rs memory permits)."""
 
     def __init__(self):
         self._lock = threading.Lock()
         self._parts: Dict[str, list] = {}
+        self._order: List[str] = []  # LRU: oldest first
+
+    def _touch(self, key: str) -> None:
+        if key in self._order:
+            self._order.remove(key)
+        self._order.append(key)
 
     def put(self, key: str, partitions: list) -> None:
         with self._lock:
             self._parts[key] = partitions
+            self._touch(key)
 
     def get(self, key: str) -> list:
         with self._lock:
             if key not in self._parts:
                 raise KeyError(f"partition set {key} not in cache")
+            self._touch(key)
             return self._parts[key]
+
+    def total_bytes(self) -> int:
+        with self._lock:
+            return sum(p.size_bytes() for parts in self._parts.values()
+                       for p in parts)
+
+    def spill_lru(self, device) -> int:
+        """Move the least-recently-used device-resident partition set to
+        host; returns bytes freed (0 if nothing to spill).  Only GPU
+        residents spill (host->host is not a spill)."""
+        if not str(device).startswith("cuda"):
+            return 0
+        with self._lock:
+            for key in list(self._order):
+                parts = self._parts.get(key, [])
+                on_dev = [p for p in parts
+                          if str(p.device) == str(device)]
+                if not on_dev:
+                    continue
+                freed = sum(p.size_bytes() for p in on_dev)
+                self._parts[key] = [p.cpu() if str(p.device) == str(device)
+                                    else p for p in parts]
+                self._order.remove(key)
+                self._order.insert(0, key)
+                return freed
+            return 0
 
     def contains(self, key: str) -> bool:
         with self._lock:

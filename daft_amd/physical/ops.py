@@ -37,6 +37,8 @@ class ExecContext:
         # src/runtime_stats/): node id -> [rows_out, batches, seconds]
         self.stats: Dict[int, list] = {}
         self.op_names: Dict[int, str] = {}
+        from ..execution.memory import MemoryManager
+        self.memory = MemoryManager(ctx)
 
 
 class PhysicalOp:
@@ -90,6 +92,9 @@ class PhysicalOp:
                                      device=ectx.device)
         if len(batches) == 1:
             return batches[0]
+        # blocking materialization: admit the concat copy under the HBM
+        # watermark (spills cached partition sets if needed)
+        ectx.memory.admit(sum(b.size_bytes() for b in batches), ectx.device)
         return RecordBatch.concat(batches)
 
     def explain_lines(self, indent=0) -> List[str]:
