@@ -368,3 +368,33 @@ def test_groupby_wide_range_falls_back_to_hash():
     assert sorted(got["c"]) == sorted(
         daft.from_pydict({"k": k}, device="cpu")
         .groupby("k").agg(col("k").count().alias("c")).to_pydict()["c"])
+
+
+def test_parquet_roundtrip_gpu(tmp_path):
+    df = daft.from_pydict({"a": [1, 2, 3], "s": ["x", None, "zz"],
+                           "f": [1.5, 2.5, None]}, device="cuda:0")
+    df.write_parquet(str(tmp_path / "o"))
+    back = daft.read_parquet(str(tmp_path / "o") + "/*.parquet") \
+        .sort("a").to_pydict()
+    assert back == {"a": [1, 2, 3], "s": ["x", None, "zz"],
+                    "f": [1.5, 2.5, None]}
+
+
+def test_cache_spill_and_reload_gpu():
+    from daft_amd.context import get_context
+    ctx = get_context()
+    df = daft.from_pydict({"a": list(range(100_000))},
+                          device="cuda:0").collect()
+    freed = ctx.cache.spill_lru("cuda:0")
+    assert freed > 0
+    # spilled partitions transparently reload on the next query
+    out = df.where(col("a") < 3).to_pydict()
+    assert out == {"a": [0, 1, 2]}
+
+
+def test_sql_on_gpu():
+    from daft_amd.sql import sql
+    t = daft.from_pydict({"g": ["a", "b", "a"], "v": [1.0, 2.0, 3.0]},
+                         device="cuda:0")
+    out = sql("select g, sum(v) as s from t group by g order by g")
+    assert out.to_pydict() == {"g": ["a", "b"], "s": [4.0, 2.0]}
