@@ -36,6 +36,11 @@ std::vector<Tensor> grouped_agg(Tensor group_ids, int64_t num_groups,
                                 const std::string& op);
 Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid);
 Tensor dense_first_index(Tensor packed, int64_t rng, int64_t sentinel);
+Tensor count_distinct_pairs(Tensor hashes, const std::vector<int64_t>& tags,
+                            const std::vector<Tensor>& datas,
+                            const std::vector<OptTensor>& offsets,
+                            const std::vector<OptTensor>& validities,
+                            Tensor gids, int64_t num_groups);
 
 // join
 std::vector<Tensor> join_build(Tensor hashes);

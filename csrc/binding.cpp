@@ -14,6 +14,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grouped_count", &grouped_count, "per-group row counts (LDS staged)");
   m.def("dense_first_index", &dense_first_index,
         "first-occurrence index per packed dense key");
+  m.def("count_distinct_pairs", &count_distinct_pairs,
+        "one-pass per-group distinct-value counts");
   m.def("join_build", &join_build, "bucket-chain hash join build");
   m.def("join_probe", &join_probe, "hash join probe -> (lidx, ridx, matched)");
   m.def("radix_argsort", &radix_argsort, "stable LSD radix argsort of u64");

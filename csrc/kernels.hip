@@ -633,6 +633,62 @@ Tensor grouped_count(Tensor group_ids, int64_t num_groups, Tensor valid) {
   return cnt;
 }
 
+// one-pass count-distinct: insert (group, value) pairs into an
+// open-addressing table; the CAS winner (first sighting of a distinct pair)
+// increments its group's counter.  No dense ids / rep gathers needed.
+__global__ void count_distinct_kernel(const uint64_t* hashes,
+                                      const ColDesc* cols, int ncols,
+                                      const int64_t* gids, int64_t n,
+                                      int64_t* table, uint64_t mask,
+                                      int64_t* cnt) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (row_has_null(cols, ncols, i)) continue;  // nulls don't count
+    uint64_t h = hashes[i];
+    uint64_t slot = h & mask;
+    while (true) {
+      long long prev = table[slot];
+      if (prev == -1ll) {
+        prev = atomicCAS((unsigned long long*)&table[slot],
+                         (unsigned long long)(-1ll), (unsigned long long)i);
+        if (prev == -1ll) {
+          atomicAdd((unsigned long long*)&cnt[gids[i]], 1ull);
+          break;
+        }
+      }
+      if (hashes[prev] == h && row_eq(cols, cols, ncols, prev, i, true))
+        break;
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
+Tensor count_distinct_pairs(Tensor hashes, const std::vector<int64_t>& tags,
+                            const std::vector<Tensor>& datas,
+                            const std::vector<OptTensor>& offsets,
+                            const std::vector<OptTensor>& validities,
+                            Tensor gids, int64_t num_groups) {
+  auto dev = hashes.device();
+  int64_t n = hashes.numel();
+  auto cnt = torch::zeros({num_groups},
+                          torch::dtype(torch::kInt64).device(dev));
+  if (n == 0) return cnt;
+  auto descs = pack_descs(tags, datas, offsets, validities);
+  int64_t cap = table_capacity(n);
+  auto table = torch::full({cap}, -1,
+                           torch::dtype(torch::kInt64).device(dev));
+  int block = 256;
+  hipLaunchKernelGGL(count_distinct_kernel, dim3(grid_1d(n, block)),
+                     dim3(block), 0, cur_stream(),
+                     (const uint64_t*)hashes.data_ptr<int64_t>(),
+                     (const ColDesc*)descs.data_ptr(), (int)tags.size(),
+                     gids.data_ptr<int64_t>(), n,
+                     table.data_ptr<int64_t>(), (uint64_t)(cap - 1),
+                     cnt.data_ptr<int64_t>());
+  return cnt;
+}
+
 // dense-range groupby support: first-occurrence index per packed key
 // (LDS-staged atomicMin for small ranges — global atomics on a handful of
 // addresses would serialize)

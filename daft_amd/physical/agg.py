@@ -141,6 +141,17 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         out = torch.where(small, lin, est).round().to(torch.int64)
         return Series(name, DataType.uint64(), data=out.view(torch.uint64))
 
+    if kind == AggKind.COUNT_DISTINCT and values.is_gpu():
+        # one-pass HIP kernel: table insert of (group, value); CAS winners
+        # bump the group counter (ref capability: dedup + count pattern)
+        from ..kernels import _descs, native_required
+        gseries = Series("__gid", DataType.int64(), data=group_ids)
+        h = rowops.hash_columns([gseries, values])
+        tags, datas, offs, vals = _descs([gseries, values])
+        cnt = native_required().count_distinct_pairs(
+            h, tags, datas, offs, vals, group_ids, num_groups)
+        return Series(name, DataType.uint64(), data=cnt.view(torch.uint64))
+
     if kind in (AggKind.COUNT_DISTINCT, AggKind.APPROX_COUNT_DISTINCT):
         # exact two-level groupby: distinct (group, value) pairs, then count
         gseries = Series("__gid", DataType.int64(), data=group_ids)
