@@ -491,14 +491,7 @@ class JoinOp(PhysicalOp):
         # blocking probe (streamed probe is the distributed layer's job)
         left = self._materialize_child(ectx, 0)
         lkeys = [e.evaluate(left) for e in self.left_on]
-        if self.how == "inner" and len(right) > 2 * len(left):
-            # adaptive build-side choice: build the smaller (left) table and
-            # probe with the right rows (ref: join side sizing in
-            # daft-distributed translate_join.rs)
-            ridx2, lidx2 = rowops.join(rkeys, lkeys, "inner")
-            lidx, ridx = lidx2, ridx2
-        else:
-            lidx, ridx = rowops.join(lkeys, rkeys, self.how)
+        lidx, ridx = rowops.join(lkeys, rkeys, self.how)
         if self.how in ("semi", "anti"):
             yield left.take(lidx)
             return
