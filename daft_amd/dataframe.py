@@ -160,6 +160,23 @@ class DataFrame:
             other._builder, list(left_on), list(right_on), how, suffix,
             prefix))
 
+    def join_asof(self, other: "DataFrame", left_on: str, right_on: str,
+                  by: Optional[Sequence[str]] = None,
+                  left_by: Optional[Sequence[str]] = None,
+                  right_by: Optional[Sequence[str]] = None,
+                  strategy: str = "backward",
+                  suffix: str = "_right") -> "DataFrame":
+        """Nearest-key (as-of) join: for each left row take the closest
+        right row by `on` (backward = latest <=, forward = earliest >=),
+        optionally within matching `by` keys."""
+        from .logical.plan import AsofJoin
+        lb = list(left_by or by or [])
+        rb = list(right_by or by or [])
+        node = AsofJoin(self._builder.plan, other._builder.plan, left_on,
+                        right_on, lb, rb, strategy, suffix)
+        from .logical.builder import LogicalPlanBuilder
+        return self._wrap(LogicalPlanBuilder(node))
+
     def repartition(self, num_partitions: Optional[int],
                     *by: ColumnInput) -> "DataFrame":
         scheme = "hash" if by else "random"

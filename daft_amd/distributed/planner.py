@@ -96,6 +96,17 @@ def _rewrite(node: lp.LogicalPlan, world: int,
 
     if isinstance(node, lp.Join):
         return _rewrite_join(node, states)
+    if isinstance(node, lp.AsofJoin):
+        left, right = node.children
+        if node.left_by:
+            lc = dn.ExchangeByKey(left, [ColumnRef(c)
+                                         for c in node.left_by])
+            rc = dn.ExchangeByKey(right, [ColumnRef(c)
+                                          for c in node.right_by])
+            return node.with_children([lc, rc]), \
+                _hash_state(list(node.left_by))
+        rep = dn.ReplicateAll(right)
+        return node.with_children([left, rep]), states[0]
 
     if isinstance(node, lp.Sort):
         ex = dn.RangeExchange(node.children[0], node.by, node.descending,

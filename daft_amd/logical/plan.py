@@ -467,6 +467,54 @@ class Join(LogicalPlan):
         return (f"Join({self.how}, on={self.left_on!r}=={self.right_on!r})")
 
 
+class AsofJoin(LogicalPlan):
+    """Nearest-key join (ref: LogicalPlan::AsofJoin + join/asof_join.rs)."""
+
+    def __init__(self, left: LogicalPlan, right: LogicalPlan, left_on: str,
+                 right_on: str, left_by: List[str], right_by: List[str],
+                 strategy: str = "backward", suffix: str = "_right"):
+        super().__init__([left, right])
+        self.left_on = left_on
+        self.right_on = right_on
+        self.left_by = left_by
+        self.right_by = right_by
+        self.strategy = strategy
+        self.suffix = suffix
+
+    def _compute_schema(self):
+        ls, rs = self.children[0].schema, self.children[1].schema
+        fields = ls.fields()
+        taken = set(ls.names())
+        skip = set(self.right_by) | {self.right_on}
+        self._right_cols = []
+        for f in rs:
+            if f.name in skip and (f.name in set(self.left_by) or
+                                   f.name == self.left_on):
+                continue
+            out = f.name if f.name not in taken else f.name + self.suffix
+            i = 1
+            while out in taken:
+                out = f"{f.name}{self.suffix}{i}"
+                i += 1
+            taken.add(out)
+            fields.append(Field(out, f.dtype))
+            self._right_cols.append((f.name, out))
+        return Schema(fields)
+
+    def right_passthrough(self):
+        _ = self.schema
+        return self._right_cols
+
+    def with_children(self, children):
+        return AsofJoin(children[0], children[1], self.left_on,
+                        self.right_on, self.left_by, self.right_by,
+                        self.strategy, self.suffix)
+
+    def describe(self):
+        return (f"AsofJoin({self.left_on}~{self.right_on}, "
+                f"by={self.left_by}, {self.strategy})")
+
+
 class Sample(LogicalPlan):
     def __init__(self, child: LogicalPlan, fraction: float,
                  with_replacement: bool = False, seed: Optional[int] = None):

@@ -516,6 +516,26 @@ class JoinOp(PhysicalOp):
         return RecordBatch(cols, num_rows=int(lidx.shape[0]))
 
 
+class AsofJoinOp(PhysicalOp):
+    def __init__(self, left: PhysicalOp, right: PhysicalOp, left_on, right_on,
+                 left_by, right_by, strategy, schema, right_cols):
+        super().__init__([left, right], schema, f"AsofJoin({strategy})")
+        self.left_on = left_on
+        self.right_on = right_on
+        self.left_by = left_by
+        self.right_by = right_by
+        self.strategy = strategy
+        self.right_cols = right_cols
+
+    def execute(self, ectx) -> BatchIter:
+        from .asof import run_asof_join
+        left = self._materialize_child(ectx, 0)
+        right = self._materialize_child(ectx, 1)
+        yield run_asof_join(left, right, self.left_on, self.right_on,
+                            self.left_by, self.right_by, self.strategy,
+                            self.right_cols)
+
+
 class ConcatOp(PhysicalOp):
     def __init__(self, children: List[PhysicalOp], schema: Schema):
         super().__init__(children, schema, "Concat")

@@ -63,6 +63,10 @@ def translate(plan: lp.LogicalPlan) -> ops.PhysicalOp:
                            plan.schema)
     if isinstance(plan, lp.Concat):
         return ops.ConcatOp(ch, plan.schema)
+    if isinstance(plan, lp.AsofJoin):
+        return ops.AsofJoinOp(ch[0], ch[1], plan.left_on, plan.right_on,
+                              plan.left_by, plan.right_by, plan.strategy,
+                              plan.schema, plan.right_passthrough())
     if isinstance(plan, lp.Join):
         return ops.JoinOp(ch[0], ch[1], plan.left_on, plan.right_on,
                           plan.how, plan.schema, plan.right_passthrough()

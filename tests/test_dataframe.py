@@ -272,3 +272,22 @@ def test_schema_and_getitem(df):
 def test_topn_rewrite(df):
     out = df.sort("a", desc=True).limit(2).to_pydict()
     assert out["a"] == [6, 5]
+
+
+def test_join_asof_backward():
+    trades = daft.from_pydict({
+        "t": [3, 7, 10], "sym": ["a", "a", "b"], "px": [1.0, 2.0, 3.0]})
+    quotes = daft.from_pydict({
+        "t": [1, 5, 8, 9], "sym": ["a", "a", "a", "b"],
+        "bid": [10.0, 20.0, 30.0, 40.0]})
+    out = trades.join_asof(quotes, left_on="t", right_on="t",
+                           by=["sym"]).sort("t").to_pydict()
+    assert out["bid"] == [10.0, 20.0, 40.0]
+
+
+def test_join_asof_forward_no_by():
+    l = daft.from_pydict({"t": [2, 6], "v": [1, 2]})
+    r = daft.from_pydict({"t": [4, 5], "w": [40, 50]})
+    out = l.join_asof(r, left_on="t", right_on="t",
+                      strategy="forward").sort("t").to_pydict()
+    assert out["w"] == [40, None]
