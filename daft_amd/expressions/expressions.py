@@ -1136,6 +1136,30 @@ class ListNamespace(_Namespace):
     def join(self, sep: str):
         return self._fn("list_join", _list_join, DataType.string(), sep)
 
+    def distinct(self):
+        return self._fn("list_distinct", _list_distinct,
+                        lambda f: f[0].dtype)
+
+    def value_counts(self):
+        def ret(fields):
+            inner = fields[0].dtype.inner
+            return DataType.list(DataType.struct(
+                {"value": inner, "count": DataType.uint64()}))
+        return self._fn("list_value_counts", _list_value_counts, ret)
+
+    def contains(self, value):
+        return self._fn("list_contains", _list_contains, DataType.bool(),
+                        value)
+
+    def chunk(self, size: int):
+        def ret(fields):
+            return DataType.list(DataType.list(fields[0].dtype.inner))
+        return self._fn("list_chunk", _list_chunk, ret, size)
+
+    def slice(self, start: int, end: Optional[int] = None):
+        return self._fn("list_slice", _list_slice,
+                        lambda f: f[0].dtype, start, end)
+
 
 def _list_length(s: Series) -> Series:
     if s.dtype.kind == TypeKind.LIST:
@@ -1207,6 +1231,64 @@ def _list_join(s: Series, sep: str) -> Series:
                                            for x in v) for v in vals]
     res = Series.from_pylist(s.name, out, DataType.string())
     return res.to(s.device) if s.is_gpu() else res
+
+
+def _list_pylist_map(s: Series, f, out_dtype) -> Series:
+    vals = s.cpu().to_pylist()
+    out = [None if v is None else f(v) for v in vals]
+    res = Series.from_pylist(s.name, out, out_dtype)
+    return res.to(s.device) if s.is_gpu() else res
+
+
+def _list_distinct(s: Series) -> Series:
+    from ..schema import DataType as DT
+
+    def dedup(v):
+        seen, out = set(), []
+        for x in v:
+            k = repr(x)
+            if k not in seen:
+                seen.add(k)
+                out.append(x)
+        return out
+    return _list_pylist_map(s, dedup, DT.list(s.dtype.inner)
+                            if s.dtype.kind == TypeKind.LIST else s.dtype)
+
+
+def _list_value_counts(s: Series) -> Series:
+    from collections import Counter
+    from ..schema import DataType as DT
+    inner = s.dtype.inner
+
+    def vc(v):
+        c = Counter(x for x in v if x is not None)
+        return [{"value": k, "count": n} for k, n in c.items()]
+    out_dt = DT.list(DT.struct({"value": inner, "count": DT.uint64()}))
+    return _list_pylist_map(s, vc, out_dt)
+
+
+def _list_contains(s: Series, value) -> Series:
+    vals = s.cpu().to_pylist()
+    out = [None if v is None else (value in v) for v in vals]
+    res = Series.from_pylist(s.name, out, DataType.bool())
+    return res.to(s.device) if s.is_gpu() else res
+
+
+def _list_chunk(s: Series, size: int) -> Series:
+    from ..schema import DataType as DT
+
+    def ch(v):
+        return [v[i:i + size] for i in range(0, len(v), size)]
+    return _list_pylist_map(s, ch, DT.list(DT.list(s.dtype.inner)))
+
+
+def _list_slice(s: Series, start: int, end) -> Series:
+    from ..schema import DataType as DT
+
+    def sl(v):
+        return v[start:end]
+    dt = s.dtype if s.dtype.kind == TypeKind.LIST else DT.list(s.dtype.inner)
+    return _list_pylist_map(s, sl, dt)
 
 
 class StructNamespace(_Namespace):

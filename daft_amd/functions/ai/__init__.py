@@ -70,3 +70,38 @@ def classify_text(expr, labels: Sequence[str],
 
     return Expression(ScalarFn("classify_text", impl, [_to_node(expr)],
                                DataType.string()))
+
+
+def prompt(expr, template: Optional[str] = None, provider: str = "echo",
+           model_path: Optional[str] = None, **kwargs) -> Expression:
+    """LLM prompting over a text column (ref: daft.functions.ai.prompt
+    :430).  Providers: "echo" (deterministic offline template expansion —
+    the only option without local weights) or "transformers" with a local
+    checkpoint path."""
+
+    def impl(s: Series) -> Series:
+        texts = s.cpu().to_pylist()
+        if provider == "transformers":
+            from transformers import AutoModelForCausalLM, AutoTokenizer
+            import torch as _t
+            tok = AutoTokenizer.from_pretrained(model_path)
+            mdl = AutoModelForCausalLM.from_pretrained(model_path)
+            outs = []
+            for t in texts:
+                if t is None:
+                    outs.append(None)
+                    continue
+                text = template.format(input=t) if template else t
+                enc = tok(text, return_tensors="pt")
+                with _t.no_grad():
+                    gen = mdl.generate(**enc, max_new_tokens=64)
+                outs.append(tok.decode(gen[0], skip_special_tokens=True))
+        else:
+            outs = [None if t is None else
+                    (template.format(input=t) if template else t)
+                    for t in texts]
+        res = Series.from_pylist(s.name, outs, DataType.string())
+        return res.to(s.device) if s.is_gpu() else res
+
+    return Expression(ScalarFn("prompt", impl, [_to_node(expr)],
+                               DataType.string()))

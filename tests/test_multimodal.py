@@ -144,3 +144,27 @@ def test_cosine_distance():
         .to_pydict()["d"]
     assert out[0] == pytest.approx(0.0, abs=1e-6)
     assert out[1] == pytest.approx(1.0, abs=1e-6)
+
+
+def test_list_extras():
+    df = daft.from_pydict({"l": [[1, 2, 2, 3], [5], None]})
+    out = df.select(
+        col("l").list.distinct().alias("d"),
+        col("l").list.contains(2).alias("c"),
+        col("l").list.chunk(2).alias("ch"),
+        col("l").list.slice(1, 3).alias("sl"),
+    ).to_pydict()
+    assert out["d"] == [[1, 2, 3], [5], None]
+    assert out["c"] == [True, False, None]
+    assert out["ch"] == [[[1, 2], [2, 3]], [[5]], None]
+    assert out["sl"] == [[2, 2], [], None]
+    vc = df.select(col("l").list.value_counts().alias("v")).to_pydict()["v"]
+    assert {d["value"]: d["count"] for d in vc[0]} == {1: 1, 2: 2, 3: 1}
+
+
+def test_prompt_echo():
+    from daft_amd.functions.ai import prompt
+    df = daft.from_pydict({"t": ["world", None]})
+    out = df.select(prompt(col("t"), template="hello {input}").alias("p")) \
+        .to_pydict()["p"]
+    assert out == ["hello world", None]
