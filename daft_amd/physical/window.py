@@ -108,6 +108,15 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
         assert isinstance(e, WindowFn)
         if e.kind == "agg":
             a: Agg = e.inner
+            if sorted_pos is not None and a.kind in (
+                    AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL,
+                    AggKind.MEAN):
+                # SQL RANGE-default running aggregate: with an ORDER BY the
+                # frame is unbounded-preceding .. current row (peers share)
+                # (ref: window_partition_and_order_by.rs incremental states)
+                out_cols.append(_running_agg(batch, a, nm, gids, order_by,
+                                             sorted_pos))
+                continue
             per_group = agg_mod.compute_agg(batch, gids, num_groups, nm, a)
             out_cols.append(per_group.take(gids).rename(nm))
         elif e.kind in ("row_number", "rank", "dense_rank"):
@@ -147,6 +156,74 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
         else:
             raise ValueError(f"unknown window fn {e.kind}")
     return RecordBatch(out_cols, num_rows=n)
+
+
+def _running_agg(batch: RecordBatch, a: Agg, name: str, gids: torch.Tensor,
+                 order_by, sorted_pos) -> Series:
+    """Running sum/count/mean per partition in sorted order, with SQL RANGE
+    peer sharing (rows with equal order keys take the frame value at the
+    last peer)."""
+    perm, pos, part_start = sorted_pos
+    n = gids.shape[0]
+    dev = gids.device
+    if a.kind == AggKind.COUNT_ALL or a.child is None:
+        vdata = torch.ones(n, dtype=torch.float64, device=dev)
+        vvalid = None
+    else:
+        values = a.child.evaluate(batch)
+        if len(values) == 1 and n > 1:
+            values = values.broadcast(n)
+        if a.kind == AggKind.COUNT:
+            vdata = torch.ones(n, dtype=torch.float64, device=dev)
+            vvalid = values.validity
+        else:
+            vdata = values.data.to(torch.float64)
+            vvalid = values.validity
+    if vvalid is not None:
+        vdata = torch.where(vvalid, vdata, torch.zeros_like(vdata))
+
+    v_sorted = vdata[perm]
+    csum = torch.cumsum(v_sorted, 0)
+    g_sorted = gids[perm]
+    base_idx = part_start[g_sorted]
+    base = csum[base_idx] - v_sorted[base_idx]
+    running_sorted = csum - base
+
+    # peer sharing: propagate the value at each peer-group's LAST row
+    # backward to all peers — take the running value at the last position
+    # with the same (partition, order-keys)
+    okeys = [k.evaluate(batch) for k in order_by]
+    sap = _same_as_prev(okeys, gids, perm, pos)[perm]  # sorted order
+    # last index of each peer run: positions where the NEXT row starts a new
+    # peer group; compute via reversed cummax of position at run ends
+    idx = torch.arange(n, dtype=torch.int64, device=dev)
+    next_new = torch.ones(n, dtype=torch.bool, device=dev)
+    next_new[:-1] = ~sap[1:]
+    run_end = torch.where(next_new, idx, torch.full_like(idx, -1))
+    # backward cummax fills each row with its run's end position
+    run_end = torch.flip(torch.cummax(torch.flip(run_end, [0]), 0).values,
+                         [0])
+    shared_sorted = running_sorted[run_end]
+
+    out_sorted = shared_sorted
+    if a.kind == AggKind.MEAN:
+        ones = torch.ones(n, dtype=torch.float64, device=dev)
+        if vvalid is not None:
+            ones = torch.where(vvalid, ones, torch.zeros_like(ones))
+        c_sorted = ones[perm]
+        ccsum = torch.cumsum(c_sorted, 0)
+        cbase = ccsum[base_idx] - c_sorted[base_idx]
+        cnt_run = (ccsum - cbase)[run_end]
+        out_sorted = shared_sorted / cnt_run.clamp(min=1.0)
+
+    out = torch.empty(n, dtype=torch.float64, device=dev)
+    out[perm] = out_sorted
+    if a.kind in (AggKind.COUNT, AggKind.COUNT_ALL):
+        return Series(name, DataType.uint64(),
+                      data=out.to(torch.int64).view(torch.uint64))
+    from ..schema import Field
+    out_dt = a.to_field(batch.schema).dtype
+    return Series(name, out_dt, data=out.to(out_dt.to_torch()))
 
 
 def _partition_starts(sorted_gids: torch.Tensor,

@@ -199,7 +199,9 @@ def test_window_functions():
     assert out["rn"] == [1, 2, 3, 1]
     assert out["rk"] == [1, 2, 2, 1]
     assert out["dr"] == [1, 2, 2, 1]
-    assert out["sv"] == [7.0, 7.0, 7.0, 5.0]
+    # SQL RANGE-default: sum over (partition, order) is a RUNNING sum with
+    # peer sharing
+    assert out["sv"] == [1.0, 7.0, 7.0, 5.0]
 
 
 def test_lag_lead():
@@ -293,3 +295,38 @@ def test_memory_limit_admission():
         mm.admit(10_000, "cpu")
     out = daft.from_pydict({"a": [1, 2]}).where(col("a") > 0).to_pydict()
     assert out == {"a": [1, 2]}
+
+
+def test_running_window_aggs():
+    df = daft.from_pydict({"g": ["a"] * 4 + ["b"] * 2,
+                           "t": [1, 2, 3, 4, 1, 2],
+                           "v": [10.0, 20.0, 30.0, 40.0, 5.0, 7.0]})
+    from daft_amd.window import Window
+    w = Window().partition_by("g").order_by("t")
+    out = df.with_window_columns({
+        "rs": col("v").sum().over(w),
+        "rc": col("v").count().over(w),
+        "rm": col("v").mean().over(w),
+    }).sort(["g", "t"]).to_pydict()
+    assert out["rs"] == [10.0, 30.0, 60.0, 100.0, 5.0, 12.0]
+    assert out["rc"] == [1, 2, 3, 4, 1, 2]
+    assert out["rm"] == [10.0, 15.0, 20.0, 25.0, 5.0, 6.0]
+
+
+def test_running_window_ties_share_frame():
+    df = daft.from_pydict({"t": [1, 1, 2], "v": [10.0, 20.0, 5.0]})
+    from daft_amd.window import Window
+    w = Window().order_by("t")
+    out = df.with_window_columns({"rs": col("v").sum().over(w)}) \
+        .sort(["t", "v"]).to_pydict()
+    # t=1 rows are peers: both see the full 30.0
+    assert out["rs"] == [30.0, 30.0, 35.0]
+
+
+def test_whole_partition_agg_without_order():
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1.0, 2.0, 5.0]})
+    from daft_amd.window import Window
+    w = Window().partition_by("g")
+    out = df.with_window_columns({"s": col("v").sum().over(w)}) \
+        .sort(["g", "v"]).to_pydict()
+    assert out["s"] == [3.0, 3.0, 5.0]
