@@ -199,9 +199,9 @@ def _running_agg(batch: RecordBatch, a: Agg, name: str, gids: torch.Tensor,
     idx = torch.arange(n, dtype=torch.int64, device=dev)
     next_new = torch.ones(n, dtype=torch.bool, device=dev)
     next_new[:-1] = ~sap[1:]
-    run_end = torch.where(next_new, idx, torch.full_like(idx, -1))
-    # backward cummax fills each row with its run's end position
-    run_end = torch.flip(torch.cummax(torch.flip(run_end, [0]), 0).values,
+    # nearest run-end at-or-after each row: suffix-min over end positions
+    run_end = torch.where(next_new, idx, torch.full_like(idx, n))
+    run_end = torch.flip(torch.cummin(torch.flip(run_end, [0]), 0).values,
                          [0])
     shared_sorted = running_sorted[run_end]
 
