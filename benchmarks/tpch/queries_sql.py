@@ -278,9 +278,25 @@ where s_suppkey in (
   and s_nationkey = n_nationkey and n_name = 'CANADA'
 order by s_name
 """,
-    # Q21/Q22 use correlation shapes (cross-row inequality EXISTS; scalar avg
-    # over a filtered copy of the outer table) that the round-1 decorrelator
-    # does not cover; they run via the DataFrame API (queries.py).
+    22: """
+select cntrycode, count(*) as numcust, sum(c_acctbal) as totacctbal
+from (select substring(c_phone from 1 for 2) as cntrycode, c_acctbal,
+             c_custkey
+      from customer
+      where substring(c_phone from 1 for 2) in
+            ('13', '31', '23', '29', '30', '18', '17')
+        and c_acctbal > (
+            select avg(c_acctbal) from customer
+            where c_acctbal > 0.00
+              and substring(c_phone from 1 for 2) in
+                  ('13', '31', '23', '29', '30', '18', '17'))
+        and not exists (select * from orders
+                        where o_custkey = c_custkey)) custsale
+group by cntrycode
+order by cntrycode
+""",
+    # Q21's correlation shape (cross-row inequality EXISTS) exceeds the
+    # round-1 decorrelator; it runs via the DataFrame API (queries.py).
 }
 
 

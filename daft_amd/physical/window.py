@@ -108,6 +108,13 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
         assert isinstance(e, WindowFn)
         if e.kind == "agg":
             a: Agg = e.inner
+            frame = getattr(e.spec, "frame", None) if e.spec else None
+            if frame is not None and sorted_pos is not None and a.kind in (
+                    AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL,
+                    AggKind.MEAN):
+                out_cols.append(_rows_frame_agg(batch, a, nm, gids,
+                                                sorted_pos, frame))
+                continue
             if sorted_pos is not None and a.kind in (
                     AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL,
                     AggKind.MEAN):
@@ -224,6 +231,84 @@ def _running_agg(batch: RecordBatch, a: Agg, name: str, gids: torch.Tensor,
     from ..schema import Field
     out_dt = a.to_field(batch.schema).dtype
     return Series(name, out_dt, data=out.to(out_dt.to_torch()))
+
+
+def _rows_frame_agg(batch: RecordBatch, a: Agg, name: str,
+                    gids: torch.Tensor, sorted_pos, frame) -> Series:
+    """ROWS BETWEEN start AND end frames for sum/count/mean via per-
+    partition prefix sums (ref: window_partition_and_dynamic_frame.rs)."""
+    perm, pos, part_start = sorted_pos
+    n = gids.shape[0]
+    dev = gids.device
+    if a.kind == AggKind.COUNT_ALL or a.child is None:
+        vdata = torch.ones(n, dtype=torch.float64, device=dev)
+        vvalid = None
+    else:
+        values = a.child.evaluate(batch)
+        if len(values) == 1 and n > 1:
+            values = values.broadcast(n)
+        vdata = torch.ones(n, dtype=torch.float64, device=dev) \
+            if a.kind == AggKind.COUNT else values.data.to(torch.float64)
+        vvalid = values.validity
+    if vvalid is not None:
+        vdata = torch.where(vvalid, vdata, torch.zeros_like(vdata))
+
+    v_sorted = vdata[perm]
+    prefix = torch.cumsum(v_sorted, 0)
+    g_sorted = gids[perm]
+    counts = torch.bincount(g_sorted, minlength=int(part_start.shape[0]))
+    ps = part_start[g_sorted]                     # partition start (sorted)
+    pe = ps + counts[g_sorted] - 1                # partition end (inclusive)
+    idx = torch.arange(n, dtype=torch.int64, device=dev)
+
+    def bound(spec, default):
+        from ..window import Window as W
+        if spec == W.unbounded_preceding:
+            return ps
+        if spec == W.unbounded_following:
+            return pe
+        if spec == W.current_row:
+            return idx
+        return idx + int(spec)
+    start, end = frame
+    lo = torch.maximum(bound(start, ps), ps)
+    hi = torch.minimum(bound(end, pe), pe)
+    empty = lo > hi
+    lo_c = lo.clamp(0, n - 1)
+    hi_c = hi.clamp(0, n - 1)
+    upper = prefix[hi_c]
+    lower = torch.where(lo_c > 0, prefix[(lo_c - 1).clamp(min=0)],
+                        torch.zeros_like(upper))
+    lower = torch.where(lo_c == 0, torch.zeros_like(lower), lower)
+    out_sorted = torch.where(empty, torch.zeros_like(upper), upper - lower)
+    if a.kind == AggKind.MEAN:
+        if vvalid is not None:
+            ones = torch.where(vvalid, torch.ones(n, dtype=torch.float64,
+                                                  device=dev),
+                               torch.zeros(n, dtype=torch.float64,
+                                           device=dev))[perm]
+            cpre = torch.cumsum(ones, 0)
+            cupper = cpre[hi_c]
+            clower = torch.where(lo_c == 0, torch.zeros_like(cupper),
+                                 cpre[(lo_c - 1).clamp(min=0)])
+            nrows = (cupper - clower).clamp(min=1.0)
+        else:
+            nrows = (hi - lo + 1).clamp(min=1).to(torch.float64)
+        out_sorted = out_sorted / nrows
+    out = torch.empty(n, dtype=torch.float64, device=dev)
+    out[perm] = out_sorted
+    validity = None
+    if bool(empty.any().item()):
+        v = torch.empty(n, dtype=torch.bool, device=dev)
+        v[perm] = ~empty
+        validity = v
+    if a.kind in (AggKind.COUNT, AggKind.COUNT_ALL):
+        return Series(name, DataType.uint64(),
+                      data=out.to(torch.int64).view(torch.uint64),
+                      validity=validity)
+    out_dt = a.to_field(batch.schema).dtype
+    return Series(name, out_dt, data=out.to(out_dt.to_torch()),
+                  validity=validity)
 
 
 def _partition_starts(sorted_gids: torch.Tensor,

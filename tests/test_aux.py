@@ -330,3 +330,23 @@ def test_whole_partition_agg_without_order():
     out = df.with_window_columns({"s": col("v").sum().over(w)}) \
         .sort(["g", "v"]).to_pydict()
     assert out["s"] == [3.0, 3.0, 5.0]
+
+
+def test_rows_between_frames():
+    df = daft.from_pydict({"g": ["a"] * 5, "t": [1, 2, 3, 4, 5],
+                           "v": [1.0, 2.0, 3.0, 4.0, 5.0]})
+    from daft_amd.window import Window
+    W = Window
+    w = Window().partition_by("g").order_by("t").rows_between(-1, 1)
+    out = df.with_window_columns({"s": col("v").sum().over(w)}) \
+        .sort("t").to_pydict()
+    assert out["s"] == [3.0, 6.0, 9.0, 12.0, 9.0]
+    w2 = Window().partition_by("g").order_by("t") \
+        .rows_between(W.unbounded_preceding, W.current_row)
+    out2 = df.with_window_columns({"s": col("v").sum().over(w2)}) \
+        .sort("t").to_pydict()
+    assert out2["s"] == [1.0, 3.0, 6.0, 10.0, 15.0]
+    w3 = Window().partition_by("g").order_by("t").rows_between(-1, -1)
+    out3 = df.with_window_columns({"m": col("v").mean().over(w3)}) \
+        .sort("t").to_pydict()
+    assert out3["m"] == [None, 1.0, 2.0, 3.0, 4.0]
