@@ -136,6 +136,12 @@ class ScanOp(PhysicalOp):
     def execute(self, ectx) -> BatchIter:
         from ..io import readers
         remaining = self.limit
+        if remaining is None and len(self.paths) > 1:
+            # no limit to push down: overlap host decode across files
+            yield from readers.read_files_prefetch(
+                self.paths, self.file_format, self.columns,
+                self.storage_options, self.read_options, ectx.device)
+            return
         for path in self.paths:
             if remaining is not None and remaining <= 0:
                 return
