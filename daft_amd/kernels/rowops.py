@@ -176,7 +176,7 @@ def _cpu_groupby_vectorized(keys: Sequence[Series]):
     # ids come out in first-appearance order, matching the dict fallback
     import pyarrow as pa
     enc = pa.array(packed, type=pa.int64()).dictionary_encode()
-    gids = torch.from_numpy(np.asarray(enc.indices).astype(np.int64))
+    gids = torch.from_numpy(np.asarray(enc.indices).astype(np.int64, copy=False))
     card = len(enc.dictionary)
     reps = torch.full((card,), n, dtype=torch.int64)
     reps.scatter_reduce_(0, gids, torch.arange(n, dtype=torch.int64),
@@ -408,7 +408,67 @@ def _gpu_join(lk, rk, how):
     return lidx, ridx
 
 
+def _cpu_join_vectorized(lk, rk, how):
+    """numpy sort-merge join on packed per-row codes (shared encodings via
+    a concatenated unique pass).  None when a key type is unsupported."""
+    from . import concat as _concat
+    nl, nr = len(lk[0]), len(rk[0])
+    packed = None
+    total = 1
+    lvalid = np.ones(nl, dtype=bool)
+    rvalid = np.ones(nr, dtype=bool)
+    for l, r in zip(lk, rk):
+        enc = _cpu_group_codes(_concat([l, r]))
+        if enc is None:
+            return None
+        codes, card = enc
+        if total * card >= (1 << 62):
+            return None
+        packed = codes if packed is None else packed * card + codes
+        total *= card
+        if l.validity is not None:
+            lvalid &= l.validity.numpy()
+        if r.validity is not None:
+            rvalid &= r.validity.numpy()
+    lcodes, rcodes = packed[:nl].copy(), packed[nl:].copy()
+    # null keys never match (SQL semantics): sentinel below any real code
+    lcodes[~lvalid] = -2
+    rcodes[~rvalid] = -1
+    rs = np.argsort(rcodes, kind="stable")
+    rsorted = rcodes[rs]
+    lo = np.searchsorted(rsorted, lcodes, "left")
+    hi = np.searchsorted(rsorted, lcodes, "right")
+    counts = hi - lo
+    if how == "semi":
+        return (torch.from_numpy(np.flatnonzero(counts > 0)),
+                torch.zeros(0, dtype=torch.int64))
+    if how == "anti":
+        return (torch.from_numpy(np.flatnonzero(counts == 0)),
+                torch.zeros(0, dtype=torch.int64))
+    tot = int(counts.sum())
+    li = np.repeat(np.arange(nl, dtype=np.int64), counts)
+    start = np.cumsum(counts) - counts
+    pos = np.arange(tot, dtype=np.int64) - np.repeat(start, counts) \
+        + np.repeat(lo, counts)
+    ri = rs[pos] if tot else np.zeros(0, dtype=np.int64)
+    if how in ("left", "outer"):
+        miss = np.flatnonzero(counts == 0)
+        li = np.concatenate([li, miss])
+        ri = np.concatenate([ri, np.full(len(miss), -1, dtype=np.int64)])
+    if how in ("right", "outer"):
+        matched = np.zeros(nr, dtype=bool)
+        matched[ri[ri >= 0]] = True
+        miss = np.flatnonzero(~matched)
+        li = np.concatenate([li, np.full(len(miss), -1, dtype=np.int64)])
+        ri = np.concatenate([ri, miss])
+    return (torch.from_numpy(li.astype(np.int64, copy=False)),
+            torch.from_numpy(ri.astype(np.int64, copy=False)))
+
+
 def _cpu_join(lk, rk, how):
+    vec = _cpu_join_vectorized(lk, rk, how)
+    if vec is not None:
+        return vec
     lrows = _cpu_key_rows(lk)
     rrows = _cpu_key_rows(rk)
     table = {}
