@@ -748,11 +748,13 @@ def _apply_subquery_conjunct(df, cj, binder: Binder, lookup, ctes):
 
 
 def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
-                      ctes):
+                      ctes, allow_neq: bool = False):
     """Find equality conjuncts in sub.where referencing exactly one outer
     column and one inner column.  Returns (outer_cols_ast, inner_cols_ast,
     residual_where) or None if uncorrelated.  Raises on other correlation
-    shapes."""
+    shapes.  With allow_neq, `inner <> outer` conjuncts are collected too
+    and a 4-tuple (..., neq_pairs) is returned (the Q21 shape;
+    decorrelated via per-group min/max in _plan_exists)."""
     if sub.where is None:
         return None
     inner_tables = {t.alias or t.name for t in sub.from_tables}
@@ -792,7 +794,7 @@ def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
         return False
 
     conjs = _split_conj(sub.where)
-    outer_cols, inner_cols, residual = [], [], []
+    outer_cols, inner_cols, residual, neq_pairs = [], [], [], []
     correlated = False
     for cj in conjs:
         refs: List[P.Col] = []
@@ -802,20 +804,26 @@ def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
             residual.append(cj)
             continue
         correlated = True
-        ok = (isinstance(cj, P.BinOp) and cj.op == "eq" and
+        ok = (isinstance(cj, P.BinOp) and
+              (cj.op == "eq" or (allow_neq and cj.op == "ne")) and
               isinstance(cj.left, P.Col) and isinstance(cj.right, P.Col))
         if not ok:
             raise SQLPlanError(
                 f"unsupported correlated predicate: {cj!r} (only equality "
                 "correlation is decorrelated)")
         if is_outer_ref(cj.left):
-            outer_cols.append(cj.left)
-            inner_cols.append(cj.right)
+            pair = (cj.left, cj.right)
         else:
-            outer_cols.append(cj.right)
-            inner_cols.append(cj.left)
+            pair = (cj.right, cj.left)
+        if cj.op == "eq":
+            outer_cols.append(pair[0])
+            inner_cols.append(pair[1])
+        else:
+            neq_pairs.append(pair)
     if not correlated:
         return None
+    if allow_neq:
+        return outer_cols, inner_cols, residual, neq_pairs
     return outer_cols, inner_cols, residual
 
 
@@ -829,14 +837,17 @@ def _rebuild_where(conjs: List[Any]):
 
 
 def _plan_exists(df, sub: P.SelectStmt, negated: bool, binder, lookup, ctes):
-    corr = _correlation_info(sub, binder, lookup, ctes)
+    corr = _correlation_info(sub, binder, lookup, ctes, allow_neq=True)
     if corr is None:
         # uncorrelated EXISTS: keep/drop everything
         sub_df = _plan_select(sub, lookup, outer=None, ctes=ctes)
         n = sub_df.limit(1).count_rows()
         keep = (n > 0) != negated
         return df if keep else df.where(lit(False))
-    outer_cols, inner_cols, residual = corr
+    outer_cols, inner_cols, residual, neq_pairs = corr
+    if neq_pairs:
+        return _plan_exists_neq(df, sub, negated, binder, lookup, ctes,
+                                outer_cols, inner_cols, residual, neq_pairs)
     inner_stmt = P.SelectStmt(
         items=[P.SelectItem(c, f"__ex{i}") for i, c in enumerate(inner_cols)],
         from_tables=sub.from_tables, joins=sub.joins,
@@ -846,6 +857,52 @@ def _plan_exists(df, sub: P.SelectStmt, negated: bool, binder, lookup, ctes):
     right = [col(f"__ex{i}") for i in range(len(inner_cols))]
     how = "anti" if negated else "semi"
     return df.join(sub_df, left_on=left, right_on=right, how=how)
+
+
+def _plan_exists_neq(df, sub, negated, binder, lookup, ctes,
+                     outer_cols, inner_cols, residual, neq_pairs):
+    """EXISTS with one `inner.b <> outer.b` correlated conjunct beside the
+    equality correlation (TPC-H Q21):
+
+        EXISTS(SELECT .. WHERE in.a = out.a AND in.b <> out.b AND resid)
+          <=>  the b-set of group a (after resid) is non-empty and not
+               exactly {out.b}  <=>  min(b) != out.b OR max(b) != out.b
+
+    so the subquery collapses to GROUP BY a AGG min(b), max(b) + one left
+    join (no per-row re-execution).  Ref behavior:
+    /root/reference/src/daft-sql (EXISTS planning) + duckdb-style
+    decorrelation of anti-dependent predicates."""
+    if len(neq_pairs) != 1:
+        raise SQLPlanError("EXISTS with more than one <> correlated "
+                           "conjunct is not supported")
+    if not inner_cols:
+        raise SQLPlanError("EXISTS with <> correlation requires at least "
+                           "one equality correlation conjunct")
+    outer_b_ast, inner_b_ast = neq_pairs[0]
+    items = [P.SelectItem(c, f"__ex{i}") for i, c in enumerate(inner_cols)]
+    items.append(P.SelectItem(inner_b_ast, "__exb"))
+    inner_stmt = P.SelectStmt(
+        items=items, from_tables=sub.from_tables, joins=sub.joins,
+        where=_rebuild_where(residual))
+    sub_df = _plan_select(inner_stmt, lookup, outer=None, ctes=ctes)
+    keys = [f"__ex{i}" for i in range(len(inner_cols))]
+    agg_df = sub_df.groupby(*keys).agg(
+        col("__exb").min().alias("__exmn"),
+        col("__exb").max().alias("__exmx"))
+    orig = df.column_names()
+    left = [expr_to_daft(c, binder) for c in outer_cols]
+    joined = df.join(agg_df, left_on=left,
+                     right_on=[col(k) for k in keys], how="left")
+    ob = expr_to_daft(outer_b_ast, binder)
+    mn, mx = col("__exmn"), col("__exmx")
+    if negated:
+        # NOT EXISTS: group empty, or its b-set is exactly {out.b}
+        pred = mn.is_null() | ob.is_null() | \
+            ((mn == ob) & (mx == ob))
+    else:
+        pred = mn.is_null().if_else(lit(False),
+                                    (mn != ob) | (mx != ob))
+    return joined.where(pred).select(*orig)
 
 
 def _plan_correlated_scalar(df, sub: P.SelectStmt, corr, other_ast, op,
