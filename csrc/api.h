@@ -60,6 +60,7 @@ Tensor string_chunk_key(Tensor offsets, Tensor bytes, int64_t chunk);
 Tensor u64_mod(Tensor hashes, int64_t n_partitions);
 
 // multimodal (multimodal.hip)
+Tensor simhash(Tensor offsets, Tensor bytes, int64_t ngram_size);
 Tensor minhash(Tensor offsets, Tensor bytes, int64_t num_hashes,
                int64_t ngram_size, Tensor perm_a, Tensor perm_b);
 Tensor hll_update(Tensor hashes, Tensor gids, Tensor valid,

@@ -213,3 +213,61 @@ Tensor image_resize(Tensor src_bytes, Tensor src_off, Tensor src_h,
   }
   return out;
 }
+
+// ---------------------------------------------------------------------------
+// SimHash: 64-bit near-duplicate fingerprint over byte n-grams (capability
+// of /root/reference/src/daft-functions/src/simhash.rs:15-42: per-bit
+// majority vote over ngram hashes).  MI355X-native layout: one 64-lane wave
+// per row, lane j hashes ngram j (strided); the per-bit vote is 64 wave
+// ballots per 64-ngram chunk, and lane i keeps only bit i's popcount —
+// no LDS, no cross-lane shuffles beyond ballot.
+// ---------------------------------------------------------------------------
+__global__ void simhash_kernel(const int64_t* offs, const uint8_t* bytes,
+                               int64_t n, int ngram, uint64_t* out) {
+  int wave_in_block = threadIdx.x / WAVE;
+  int lane = threadIdx.x & (WAVE - 1);
+  int waves_per_block = blockDim.x / WAVE;
+  int64_t row0 = (int64_t)blockIdx.x * waves_per_block + wave_in_block;
+  int64_t row_stride = (int64_t)gridDim.x * waves_per_block;
+  for (int64_t row = row0; row < n; row += row_stride) {
+    int64_t beg = offs[row], end = offs[row + 1];
+    int64_t len = end - beg;
+    int64_t m = len - ngram + 1;  // number of ngrams
+    if (m <= 0) {
+      if (lane == 0) out[row] = 0;
+      continue;
+    }
+    int64_t my_cnt = 0;  // lane i: how many ngrams have bit i set
+    for (int64_t base = 0; base < m; base += WAVE) {
+      int64_t j = base + lane;
+      uint64_t h = 0;
+      bool act = j < m;
+      if (act) h = hash_bytes_dev(bytes + beg + j, ngram);
+      #pragma unroll
+      for (int i = 0; i < 64; ++i) {
+        uint64_t mask = __ballot(act && ((h >> i) & 1));
+        if (lane == i) my_cnt += __popcll(mask);
+      }
+    }
+    // bit i of the fingerprint: weight_i = 2*cnt_i - m > 0
+    uint64_t fp = __ballot(2 * my_cnt > m);
+    if (lane == 0) out[row] = fp;
+  }
+}
+
+Tensor simhash(Tensor offsets, Tensor bytes, int64_t ngram_size) {
+  auto dev = offsets.device();
+  int64_t n = offsets.numel() - 1;
+  auto out = torch::empty({n}, torch::dtype(torch::kInt64).device(dev));
+  if (n > 0) {
+    int block = 256;
+    int waves_per_block = block / WAVE;
+    int grid = (int)std::min<int64_t>((n + waves_per_block - 1) /
+                                      waves_per_block, kMaxBlocks);
+    const uint8_t* bp = bytes.numel() ? bytes.data_ptr<uint8_t>() : nullptr;
+    hipLaunchKernelGGL(simhash_kernel, dim3(grid), dim3(block), 0,
+                       mm_stream(), offsets.data_ptr<int64_t>(), bp, n,
+                       (int)ngram_size, (uint64_t*)out.data_ptr<int64_t>());
+  }
+  return out;
+}

@@ -731,6 +731,12 @@ class Expression:
             DataType.fixed_size_list(DataType.uint32(), num_hashes),
             (num_hashes, ngram_size, seed)))
 
+    def simhash(self, ngram_size: int = 4) -> "Expression":
+        from ..functions.minhash import simhash_series
+        return Expression(ScalarFn(
+            "simhash", simhash_series, [self._node],
+            DataType.uint64(), (ngram_size,)))
+
     # misc
     def abs(self):
         return Expression(ScalarFn(

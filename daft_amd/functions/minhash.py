@@ -87,3 +87,38 @@ def _mulmod61_np(a: np.ndarray, h: np.uint64) -> np.ndarray:
     for i, av in enumerate(a):
         out[i] = (int(av) * hv) % MERSENNE61
     return out
+
+
+def simhash_series(s: Series, ngram_size: int) -> Series:
+    """64-bit SimHash fingerprint over byte n-grams (capability of
+    /root/reference/src/daft-functions/src/simhash.rs:15-42: per-bit
+    majority vote over ngram hashes; our hash is csrc/common.h
+    hash_bytes_dev).  Rows shorter than ngram_size fingerprint to 0."""
+    n = len(s)
+    if _is_gpu(s):
+        out = native_required().simhash(s.offsets, s.data, ngram_size)
+        return Series(s.name, DataType.uint64(),
+                      data=out.view(torch.uint64), validity=s.validity)
+    vals = s.to_pylist()
+    out = np.zeros(n, dtype=np.uint64)
+    for i, v in enumerate(vals):
+        if v is None:
+            continue
+        b = v.encode() if isinstance(v, str) else bytes(v)
+        m = len(b) - ngram_size + 1
+        if m <= 0:
+            continue
+        w = np.zeros(64, dtype=np.int64)
+        for j in range(m):
+            h = _hash_bytes_py(b[j:j + ngram_size])
+            bits = (np.uint64(h) >> np.arange(64, dtype=np.uint64)) \
+                & np.uint64(1)
+            w += np.where(bits.astype(bool), 1, -1)
+        fp = 0
+        for i2 in range(64):
+            if w[i2] > 0:
+                fp |= 1 << i2
+        out[i] = np.uint64(fp)
+    return Series(s.name, DataType.uint64(),
+                  data=torch.from_numpy(out.view(np.int64))
+                  .view(torch.uint64), validity=s.validity)

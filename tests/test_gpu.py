@@ -292,6 +292,17 @@ def test_minhash_gpu_matches_cpu():
     assert g == c
 
 
+def test_simhash_gpu_matches_cpu():
+    docs = ["the quick brown fox jumps over the lazy dog",
+            "the quick brown fox jumps over the lazy cat",
+            "x" * 200, "ab", "", None] * 100
+    g = daft.from_pydict({"t": docs}, device="cuda:0") \
+        .select(col("t").simhash(4).alias("h")).to_pydict()
+    c = daft.from_pydict({"t": docs}, device="cpu") \
+        .select(col("t").simhash(4).alias("h")).to_pydict()
+    assert g == c
+
+
 def test_hll_approx_count_distinct_gpu():
     torch.manual_seed(20)
     true_n = 5000

@@ -168,3 +168,19 @@ def test_prompt_echo():
     out = df.select(prompt(col("t"), template="hello {input}").alias("p")) \
         .to_pydict()["p"]
     assert out == ["hello world", None]
+
+
+def test_simhash():
+    docs = ["the quick brown fox jumps over the lazy dog",
+            "the quick brown fox jumps over the lazy cat",
+            "entirely different words and other content here", None, "ab"]
+    df = daft.from_pydict({"t": docs})
+    out = df.select(col("t").simhash(4).alias("h")).to_pydict()["h"]
+    assert out[3] is None
+    assert out[4] == 0  # shorter than ngram
+    assert out == df.select(col("t").simhash(4).alias("h")) \
+        .to_pydict()["h"]  # deterministic
+
+    def ham(a, b):
+        return bin(a ^ b).count("1")
+    assert ham(out[0], out[1]) < ham(out[0], out[2])
