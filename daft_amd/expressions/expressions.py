@@ -399,6 +399,10 @@ class AggKind:
     APPROX_PERCENTILE = "approx_percentile"
     BOOL_AND = "bool_and"
     BOOL_OR = "bool_or"
+    # internal: DDSketch partial/final pair for distributed
+    # approx_percentile (physical/sketch.py)
+    SKETCH = "__sketch"
+    SKETCH_FINAL = "__sketch_final"
 
 
 _NUMERIC_AGGS = {AggKind.SUM, AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE,
@@ -454,6 +458,11 @@ class Agg(ExprNode):
             return Field(f.name, DataType.list(f.dtype))
         if k in (AggKind.BOOL_AND, AggKind.BOOL_OR):
             return Field(f.name, DataType.bool())
+        if k == AggKind.SKETCH:
+            from ..physical.sketch import SKETCH_DTYPE
+            return Field(f.name, SKETCH_DTYPE)
+        if k == AggKind.SKETCH_FINAL:
+            return Field(f.name, DataType.float64())
         raise ValueError(f"unknown agg kind {k}")
 
     def evaluate(self, batch) -> Series:

@@ -202,8 +202,20 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         return Series(name, DataType.list(values.dtype), offsets=offs,
                       children=[sorted_vals.rename("item")])
 
+    if kind == AggKind.SKETCH:
+        from . import sketch
+        return sketch.grouped_sketch(values.cast(DataType.float64()),
+                                     group_ids, num_groups, name)
+
+    if kind == AggKind.SKETCH_FINAL:
+        from . import sketch
+        return sketch.grouped_sketch_final(values, group_ids, num_groups,
+                                           float(agg.param), name)
+
     if kind == AggKind.APPROX_PERCENTILE:
-        # per-group exact percentile via sort (round-1; DDSketch later)
+        # per-group exact percentile via sort (single-node; the
+        # distributed planner splits into SKETCH/SKETCH_FINAL — a
+        # DDSketch build + mergeable bucket exchange, physical/sketch.py)
         q = float(agg.param)
         perm = rowops.argsort_multi(
             [Series("g", DataType.int64(), data=group_ids), values],

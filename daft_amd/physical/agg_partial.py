@@ -17,7 +17,7 @@ from .agg import decompose_agg_exprs
 _DECOMPOSABLE = {
     AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL, AggKind.MIN, AggKind.MAX,
     AggKind.MEAN, AggKind.ANY_VALUE, AggKind.BOOL_AND, AggKind.BOOL_OR,
-    AggKind.STDDEV, AggKind.VARIANCE,
+    AggKind.STDDEV, AggKind.VARIANCE, AggKind.APPROX_PERCENTILE,
 }
 
 
@@ -66,6 +66,14 @@ def split_partial_final(aggs: List[ExprNode]
                     "sqrt", _sqrt_series, [var], _float64_dt()), name))
             else:
                 final_named.append(Alias(var, name))
+        elif k == AggKind.APPROX_PERCENTILE:
+            # DDSketch build per shard; merged buckets + quantile extract
+            # after the exchange (physical/sketch.py)
+            partials.append(Alias(Agg(AggKind.SKETCH, a.child),
+                                  f"{name}__sk"))
+            final_named.append(Alias(
+                Agg(AggKind.SKETCH_FINAL, ColumnRef(f"{name}__sk"),
+                    a.param), name))
         else:  # pragma: no cover
             return None
     finals = final_named + [r for r in residuals]

@@ -350,3 +350,42 @@ def test_rows_between_frames():
     out3 = df.with_window_columns({"m": col("v").mean().over(w3)}) \
         .sort("t").to_pydict()
     assert out3["m"] == [None, 1.0, 2.0, 3.0, 4.0]
+
+
+def test_ddsketch_accuracy_and_merge():
+    """physical/sketch.py: relative error within alpha; merge = exact
+    concatenation semantics (DDSketch property)."""
+    import numpy as np
+    import torch
+    from daft_amd.physical import sketch
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+
+    rng = np.random.default_rng(3)
+    vals = np.concatenate([
+        rng.lognormal(3, 2, 20000),          # positives over decades
+        -rng.lognormal(1, 1, 5000),          # negatives
+        np.zeros(100),
+    ])
+    rng.shuffle(vals)
+    s = Series("x", DataType.float64(), data=torch.from_numpy(vals))
+    gids = torch.zeros(len(vals), dtype=torch.int64)
+    sk = sketch.grouped_sketch(s, gids, 1, "sk")
+
+    for q in (0.01, 0.25, 0.5, 0.9, 0.99):
+        got = float(sketch.grouped_sketch_final(
+            sk, torch.zeros(1, dtype=torch.int64), 1, q, "p").data[0])
+        want = float(np.quantile(vals, q))
+        denom = max(abs(want), 1e-9)
+        assert abs(got - want) / denom < 3 * sketch.ALPHA, (q, got, want)
+
+    # split in two, sketch each half, merge: same buckets as one pass
+    half = len(vals) // 2
+    s1 = Series("x", DataType.float64(), data=torch.from_numpy(vals[:half]))
+    s2 = Series("x", DataType.float64(), data=torch.from_numpy(vals[half:]))
+    sk1 = sketch.grouped_sketch(s1, torch.zeros(half, dtype=torch.int64),
+                                1, "sk")
+    sk2 = sketch.grouped_sketch(
+        s2, torch.zeros(len(vals) - half, dtype=torch.int64), 1, "sk")
+    merged = sk1.children[0].data + sk2.children[0].data
+    assert torch.equal(merged, sk.children[0].data)

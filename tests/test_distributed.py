@@ -98,6 +98,12 @@ def _core_checks(rank, world):
     assert out["m"] == [5.0, 6.0]
     assert out["mx"] == [9.0, 10.0]
 
+    # approx_percentile: DDSketch partial/final across the exchange
+    ap = df.groupby("g").agg(
+        col("v").approx_percentile(0.5).alias("p")).sort("g").to_pydict()
+    assert abs(ap["p"][0] - 5.0) / 5.0 < 0.05, ap
+    assert abs(ap["p"][1] - 6.0) / 6.0 < 0.05, ap
+
     # ungrouped agg
     tot = df.agg(col("v").sum().alias("s"),
                  col("v").mean().alias("m")).to_pydict()
