@@ -361,10 +361,59 @@ class DataFrame:
         return pd.concat(frames, ignore_index=True) if len(frames) > 1 \
             else frames[0]
 
-    def to_torch_map_dataset(self):
-        d = self.to_pydict()
+    def _shard(self, strategy: str, world_size: int, rank: int) -> "DataFrame":
+        """File-granularity sharding of the source scan for multi-worker
+        dataloading (capability of the reference's DataFrame._shard,
+        /root/reference/daft/dataframe/dataframe.py:3714: rewrites the
+        descendant scan to a round-robin subset of its files)."""
+        if strategy != "file":
+            raise ValueError("Only file-based sharding is supported")
+        if world_size <= 0:
+            raise ValueError("world_size must be > 0")
+        if not (0 <= rank < world_size):
+            raise ValueError("rank must be in [0, world_size)")
+        from .logical.plan import ScanSource
+        import copy as _copy
+        found = [0]
+
+        def rewrite(node):
+            if isinstance(node, ScanSource):
+                found[0] += 1
+                new = _copy.copy(node)
+                new.paths = node.paths[rank::world_size]
+                return new
+            if not node.children:
+                return node
+            return node.with_children([rewrite(c) for c in node.children])
+
+        new_plan = rewrite(self._builder.plan)
+        if found[0] != 1:
+            raise ValueError(
+                f"sharding requires exactly one file scan source, found "
+                f"{found[0]}")
+        from .logical.builder import LogicalPlanBuilder
+        return DataFrame(LogicalPlanBuilder(new_plan))
+
+    def to_torch_map_dataset(self, shard_strategy: Optional[str] = None,
+                             world_size: Optional[int] = None,
+                             rank: Optional[int] = None):
+        df = self
+        if shard_strategy is not None:
+            df = self._shard(shard_strategy, world_size, rank)
+        d = df.to_pydict()
         from .utils.torch_data import DictDataset
         return DictDataset(d)
+
+    def to_torch_iter_dataset(self, shard_strategy: Optional[str] = None,
+                              world_size: Optional[int] = None,
+                              rank: Optional[int] = None):
+        """Streaming torch IterableDataset over this DataFrame's rows
+        (ref: reference daft/dataframe/to_torch.py DaftTorchIterableDataset)."""
+        df = self
+        if shard_strategy is not None:
+            df = self._shard(shard_strategy, world_size, rank)
+        from .utils.torch_data import IterDataset
+        return IterDataset(df)
 
     def to_recordbatch(self) -> RecordBatch:
         df = self.collect()

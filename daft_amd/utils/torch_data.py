@@ -15,3 +15,23 @@ class DictDataset(torch.utils.data.Dataset):
 
     def __getitem__(self, i):
         return {k: self.data[k][i] for k in self.names}
+
+
+class IterDataset(torch.utils.data.IterableDataset):
+    """Streams rows batch-by-batch from a lazy DataFrame (ref capability:
+    daft/dataframe/to_torch.py DaftTorchIterableDataset).  Re-executes the
+    plan on each iteration; honors torch DataLoader worker sharding by
+    round-robin over worker id."""
+
+    def __init__(self, df):
+        self.df = df
+
+    def __iter__(self):
+        info = torch.utils.data.get_worker_info()
+        it = self.df.iter_rows()
+        if info is None or info.num_workers <= 1:
+            yield from it
+            return
+        for i, row in enumerate(it):
+            if i % info.num_workers == info.id:
+                yield row

@@ -389,3 +389,26 @@ def test_ddsketch_accuracy_and_merge():
         s2, torch.zeros(len(vals) - half, dtype=torch.int64), 1, "sk")
     merged = sk1.children[0].data + sk2.children[0].data
     assert torch.equal(merged, sk.children[0].data)
+
+
+def test_shard_and_torch_iter_dataset(tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    import daft_amd as daft
+    paths = []
+    for i in range(5):
+        p = str(tmp_path / f"f{i}.parquet")
+        pq.write_table(pa.table({"a": [i * 10 + j for j in range(3)]}), p)
+        paths.append(p)
+    df = daft.read_parquet(paths)
+    s0 = df._shard("file", 2, 0).to_pydict()["a"]
+    s1 = df._shard("file", 2, 1).to_pydict()["a"]
+    assert sorted(s0 + s1) == sorted(df.to_pydict()["a"])
+    assert not set(s0) & set(s1)
+    rows = list(df.to_torch_iter_dataset(shard_strategy="file",
+                                         world_size=2, rank=0))
+    assert [r["a"] for r in rows] == s0
+    with pytest.raises(ValueError):
+        df._shard("row", 2, 0)
+    with pytest.raises(ValueError):
+        daft.from_pydict({"a": [1]})._shard("file", 2, 0)
