@@ -24,7 +24,7 @@ from .context import (  # noqa: F401
 )
 from .io import (  # noqa: F401
     from_pydict, from_arrow, from_pandas, from_glob_path,
-    read_parquet, read_csv, read_json, read_text,
+    read_parquet, read_csv, read_json, read_text, read_warc,
 )
 from .catalog import Catalog, Identifier, MemoryCatalog, Session, \
     current_session  # noqa: F401

@@ -87,6 +87,32 @@ def read_parquet(path, columns: Optional[List[str]] = None,
     return df
 
 
+def read_warc(path, io_config=None, file_path_column: Optional[str] = None,
+              **kwargs) -> DataFrame:
+    """Read WARC / gzipped-WARC web-archive files (capability of the
+    reference's daft.read_warc, /root/reference/daft/io/_warc.py:24:
+    fixed schema of mandatory WARC headers, raw record content, and the
+    remaining headers as a JSON string column)."""
+    from . import readers
+    paths = _expand_paths(path)
+    b = LogicalPlanBuilder.from_scan(readers.warc_schema(), paths, "warc",
+                                     storage_options=io_config)
+    df = DataFrame(b)
+    if file_path_column:
+        # single-scan case: tag each row with its source path
+        import daft_amd as _d
+        parts = []
+        for p in paths:
+            sub = DataFrame(LogicalPlanBuilder.from_scan(
+                readers.warc_schema(), [p], "warc",
+                storage_options=io_config))
+            parts.append(sub.with_column(file_path_column, _d.lit(p)))
+        df = parts[0]
+        for extra in parts[1:]:
+            df = df.concat(extra)
+    return df
+
+
 def read_csv(path, has_headers: bool = True, delimiter: str = ",",
              schema: Optional[Dict[str, DataType]] = None,
              io_config=None, **kwargs) -> DataFrame:

@@ -412,3 +412,45 @@ def test_shard_and_torch_iter_dataset(tmp_path):
         df._shard("row", 2, 0)
     with pytest.raises(ValueError):
         daft.from_pydict({"a": [1]})._shard("file", 2, 0)
+
+
+def test_read_warc(tmp_path):
+    import gzip
+    import json
+    import daft_amd as daft
+    from daft_amd import col
+
+    def rec(rid, rtype, uri, date, payload, extra=None):
+        h = [b"WARC/1.0",
+             f"WARC-Record-ID: <urn:uuid:{rid}>".encode(),
+             f"WARC-Type: {rtype}".encode()]
+        if uri:
+            h.append(f"WARC-Target-URI: {uri}".encode())
+        h.append(f"WARC-Date: {date}".encode())
+        h.append(f"Content-Length: {len(payload)}".encode())
+        if extra:
+            h.extend(f"{k}: {v}".encode() for k, v in extra.items())
+        return b"\r\n".join(h) + b"\r\n\r\n" + payload + b"\r\n\r\n"
+
+    blob = (rec("aaa-111", "response", "http://x.com/",
+                "2024-01-02T03:04:05Z", b"<html>hi</html>",
+                {"Content-Type": "application/http"}) +
+            rec("bbb-222", "request", None, "2024-01-02T03:04:06Z",
+                b"GET /"))
+    p1 = str(tmp_path / "a.warc")
+    open(p1, "wb").write(blob)
+    p2 = str(tmp_path / "b.warc.gz")
+    open(p2, "wb").write(gzip.compress(blob))
+
+    out = daft.read_warc(p1).to_pydict()
+    assert out["WARC-Record-ID"] == ["aaa-111", "bbb-222"]
+    assert out["warc_content"] == [b"<html>hi</html>", b"GET /"]
+    assert out["WARC-Target-URI"] == ["http://x.com/", None]
+    assert json.loads(out["warc_headers"][0])["Content-Type"] == \
+        "application/http"
+    assert daft.read_warc(p2).to_pydict()["WARC-Record-ID"] == \
+        out["WARC-Record-ID"]
+    assert daft.read_warc([p1, p2]) \
+        .where(col("WARC-Type") == "response").count_rows() == 2
+    fp = daft.read_warc([p1, p2], file_path_column="src").to_pydict()["src"]
+    assert fp == [p1, p1, p2, p2]
