@@ -45,3 +45,5 @@ def from_pylist(rows, device=None):
                 keys.append(k)
     data = {k: [r.get(k) for r in rows] for k in keys}
     return from_pydict(data, device=device)
+
+from . import datasets  # noqa: E402

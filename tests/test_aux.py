@@ -454,3 +454,45 @@ def test_read_warc(tmp_path):
         .where(col("WARC-Type") == "response").count_rows() == 2
     fp = daft.read_warc([p1, p2], file_path_column="src").to_pydict()["src"]
     assert fp == [p1, p1, p2, p2]
+
+
+def test_datasets_common_crawl_local_mirror(tmp_path):
+    import gzip as _gz
+    import daft_amd as daft
+    crawl = "CC-MAIN-2025-33"
+    seg_dir = tmp_path / "crawl-data" / crawl / "segments" / "123" / "warc"
+    seg_dir.mkdir(parents=True)
+    rec = (b"WARC/1.0\r\nWARC-Record-ID: <urn:uuid:x-1>\r\n"
+           b"WARC-Type: response\r\nWARC-Date: 2025-01-01T00:00:00Z\r\n"
+           b"Content-Length: 2\r\n\r\nhi\r\n\r\n")
+    rel = f"crawl-data/{crawl}/segments/123/warc/part-0.warc"
+    (tmp_path / rel).write_bytes(rec)
+    man = tmp_path / "crawl-data" / crawl / "warc.paths.gz"
+    man.write_bytes(_gz.compress((rel + "\n").encode()))
+    df = daft.datasets.common_crawl(crawl, data_root=str(tmp_path))
+    out = df.to_pydict()
+    assert out["warc_content"] == [b"hi"]
+    with pytest.raises(FileNotFoundError):
+        daft.datasets.common_crawl(crawl, segment="999",
+                                   data_root=str(tmp_path))
+
+
+def test_datasets_lerobot(tmp_path):
+    import json
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    import daft_amd as daft
+    (tmp_path / "meta").mkdir()
+    (tmp_path / "meta" / "info.json").write_text(json.dumps({
+        "data_path": "data/chunk-{episode_chunk:03d}/"
+                     "episode_{episode_index:06d}.parquet"}))
+    d = tmp_path / "data" / "chunk-000"
+    d.mkdir(parents=True)
+    for e in range(3):
+        pq.write_table(pa.table({"obs": [e * 1.0, e + 0.5],
+                                 "episode_index": [e, e]}),
+                       str(d / f"episode_{e:06d}.parquet"))
+    df = daft.datasets.lerobot.load(str(tmp_path))
+    assert df.count_rows() == 6
+    df2 = daft.datasets.lerobot.load(str(tmp_path), episodes=[1])
+    assert df2.to_pydict()["episode_index"] == [1, 1]
