@@ -68,16 +68,46 @@ class _Worker:
             self._proc.terminate()
 
 
-_workers: Dict[int, _Worker] = {}
+class _Pool:
+    """Actor pool: `size` worker subprocesses for one UDF; a batch's rows
+    are split across workers and dispatched concurrently (capability of
+    the reference's actor-pool UDFs / max_concurrency,
+    daft/udf/__init__.py cls(max_concurrency=...) + udf.rs:351-406)."""
+
+    def __init__(self, fn, size: int):
+        self.workers = [_Worker(fn) for _ in range(max(1, size))]
+
+    def call_rows(self, rows: List[tuple]) -> list:
+        import concurrent.futures as fut
+        k = len(self.workers)
+        if k == 1 or len(rows) < 2 * k:
+            return self.workers[0].call_rows(rows)
+        per = (len(rows) + k - 1) // k
+        chunks = [rows[i * per:(i + 1) * per] for i in range(k)]
+        with fut.ThreadPoolExecutor(max_workers=k) as ex:
+            outs = list(ex.map(
+                lambda wc: wc[0].call_rows(wc[1]) if wc[1] else [],
+                zip(self.workers, chunks)))
+        return [v for o in outs for v in o]
+
+    def stop(self):
+        for w in self.workers:
+            w.stop()
+
+    def alive(self) -> bool:
+        return all(w._proc.is_alive() for w in self.workers)
+
+
+_workers: Dict[Any, _Pool] = {}
 _workers_lock = threading.Lock()
 
 
-def get_worker(fn) -> _Worker:
-    key = id(fn)
+def get_worker(fn, concurrency: int = 1) -> _Pool:
+    key = (id(fn), int(concurrency or 1))
     with _workers_lock:
         w = _workers.get(key)
-        if w is None or not w._proc.is_alive():
-            w = _Worker(fn)
+        if w is None or not w.alive():
+            w = _Pool(fn, int(concurrency or 1))
             _workers[key] = w
         return w
 

@@ -128,7 +128,7 @@ def run_udf_node(node: PyUDF, batch) -> Series:
             if node.use_process and not node.batched:
                 # subprocess isolation (ref: udf.rs:351-406 + udf_worker.py)
                 from .execution.udf_worker import get_worker
-                worker = get_worker(node.fn)
+                worker = get_worker(node.fn, node.concurrency or 1)
                 cols = [s.cpu().to_pylist() for s in arg_series]
                 rows = list(zip(*cols)) if cols else [()] * n
                 out_vals = worker.call_rows(rows)

@@ -496,3 +496,17 @@ def test_datasets_lerobot(tmp_path):
     assert df.count_rows() == 6
     df2 = daft.datasets.lerobot.load(str(tmp_path), episodes=[1])
     assert df2.to_pydict()["episode_index"] == [1, 1]
+
+
+def test_udf_actor_pool_concurrency():
+    @daft.func(return_dtype=DataType.int64(), use_process=True,
+               max_concurrency=2)
+    def whoami(x: int) -> int:
+        import os
+        return os.getpid()
+
+    df = daft.from_pydict({"x": list(range(64))})
+    pids = set(df.select(whoami(col("x")).alias("p")).to_pydict()["p"])
+    assert len(pids) == 2, pids
+    import os
+    assert os.getpid() not in pids
