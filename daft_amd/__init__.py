@@ -47,3 +47,4 @@ def from_pylist(rows, device=None):
     return from_pydict(data, device=device)
 
 from . import datasets  # noqa: E402
+from .ext import load_extension, ext_function  # noqa: E402

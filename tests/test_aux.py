@@ -510,3 +510,27 @@ def test_udf_actor_pool_concurrency():
     assert len(pids) == 2, pids
     import os
     assert os.getpid() not in pids
+
+
+def test_native_extension_plugin(tmp_path):
+    """daft_amd/ext: build the example plugin with g++, load it, call its
+    functions through expressions."""
+    import subprocess
+    import daft_amd as daft
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(daft.__file__)))
+    src = os.path.join(repo, "examples", "ext_plugin", "example_plugin.cpp")
+    so = str(tmp_path / "example_plugin.so")
+    subprocess.run(
+        ["g++", "-O2", "-shared", "-fPIC",
+         "-I" + os.path.join(repo, "daft_amd", "ext"), src, "-o", so],
+        check=True, capture_output=True)
+    names = daft.load_extension(so)
+    assert set(names) >= {"ext_add1", "ext_hypot"}
+    df = daft.from_pydict({"a": [1, 2, None], "x": [3.0, 5.0, 8.0],
+                           "y": [4.0, 12.0, 15.0]})
+    out = df.select(
+        daft.ext_function("ext_add1", col("a")).alias("a1"),
+        daft.ext_function("ext_hypot", col("x"), col("y")).alias("h"),
+    ).to_pydict()
+    assert out["a1"] == [2, 3, None]
+    assert out["h"] == [5.0, 13.0, 17.0]
