@@ -27,6 +27,10 @@ class ExecutionConfig:
     scan_tasks_max_size_bytes: int = 384 * 1024 * 1024
     broadcast_join_size_bytes_threshold: int = 64 * 1024 * 1024  # sized for HBM3E, vs 10 MiB on CPU
     morsel_size_rows: int = 4_000_000          # GPU-scale morsels (128 ≪ CUs need work)
+    # out-of-core: host-resident source partitions larger than this are
+    # sliced and streamed through HBM morsel-by-morsel (H2D overlap);
+    # blocking aggregates fold each morsel into partial states
+    stream_morsel_rows: int = 1 << 26
     target_batch_rows: int = 4_000_000
     pre_shuffle_merge_threshold: int = 1 << 30
     shuffle_algorithm: str = "rccl_all_to_all"

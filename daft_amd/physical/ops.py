@@ -110,8 +110,18 @@ class InMemorySourceOp(PhysicalOp):
         self.cache_key = cache_key
 
     def execute(self, ectx) -> BatchIter:
+        morsel = getattr(ectx.ctx.execution_config, "stream_morsel_rows",
+                         1 << 26)
         for part in ectx.ctx.cache.get(self.cache_key):
             if part.device != ectx.device:
+                if str(part.device) == "cpu" and \
+                        str(ectx.device).startswith("cuda") and \
+                        len(part) > morsel:
+                    # out-of-core: host partition larger than a morsel —
+                    # stream slices through HBM instead of moving it whole
+                    for lo in range(0, len(part), morsel):
+                        yield part.slice(lo, lo + morsel).to(ectx.device)
+                    continue
                 part = part.to(ectx.device)
             yield part
 
@@ -392,8 +402,40 @@ class AggregateOp(PhysicalOp):
         self.aggs = aggs
 
     def execute(self, ectx) -> BatchIter:
-        batch = self._materialize_child(ectx)
-        yield agg_mod.run_aggregate(batch, self.groupby, self.aggs)
+        from itertools import chain as _chain
+        it = self.children[0].execute_tracked(ectx)
+        first = next(it, None)
+        second = next(it, None) if first is not None else None
+        if second is None:
+            batch = first if first is not None else RecordBatch.empty(
+                self.children[0].schema, device=ectx.device)
+            yield agg_mod.run_aggregate(batch, self.groupby, self.aggs)
+            return
+        # multiple input batches: fold each into per-group partial states
+        # (same partial/final decomposition the distributed planner uses)
+        # instead of concatenating the whole input — this is what lets
+        # larger-than-HBM inputs stream through (out-of-core aggregation)
+        from .agg_partial import split_partial_final
+        split = split_partial_final(self.aggs)
+        if split is None:
+            batches = [first, second] + list(it)
+            ectx.memory.admit(sum(b.size_bytes() for b in batches),
+                              ectx.device)
+            yield agg_mod.run_aggregate(RecordBatch.concat(batches),
+                                        self.groupby, self.aggs)
+            return
+        partials, final_named, residuals = split
+        from ..expressions.expressions import ColumnRef
+        cschema = self.children[0].schema
+        gnames = [e.to_field(cschema).name for e in self.groupby]
+        parts = [agg_mod.run_aggregate(b, self.groupby, partials)
+                 for b in _chain([first, second], it)]
+        merged = RecordBatch.concat(parts)
+        final = agg_mod.run_aggregate(
+            merged, [ColumnRef(n) for n in gnames], final_named)
+        cols = [final.column(n) for n in gnames]
+        cols += [r.evaluate(final) for r in residuals]
+        yield RecordBatch(cols, num_rows=len(final))
 
 
 class DistinctOp(PhysicalOp):
@@ -494,13 +536,24 @@ class JoinOp(PhysicalOp):
             yield self._emit(left, right, lidx, ridx)
             return
 
-        # blocking probe (streamed probe is the distributed layer's job)
+        if self.how in ("inner", "left", "semi", "anti"):
+            # streamed probe: the build side is resident, each probe-side
+            # batch joins and emits independently (out-of-core probes)
+            for left in self.children[0].execute_tracked(ectx):
+                if len(left) == 0:
+                    continue
+                lkeys = [e.evaluate(left) for e in self.left_on]
+                lidx, ridx = rowops.join(lkeys, rkeys, self.how)
+                if self.how in ("semi", "anti"):
+                    yield left.take(lidx)
+                else:
+                    yield self._emit(left, right, lidx, ridx)
+            return
+        # right/outer joins track unmatched build rows across the whole
+        # probe side: materialize
         left = self._materialize_child(ectx, 0)
         lkeys = [e.evaluate(left) for e in self.left_on]
         lidx, ridx = rowops.join(lkeys, rkeys, self.how)
-        if self.how in ("semi", "anti"):
-            yield left.take(lidx)
-            return
         yield self._emit(left, right, lidx, ridx)
 
     def _emit(self, left: RecordBatch, right: RecordBatch,

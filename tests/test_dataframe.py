@@ -291,3 +291,59 @@ def test_join_asof_forward_no_by():
     out = l.join_asof(r, left_on="t", right_on="t",
                       strategy="forward").sort("t").to_pydict()
     assert out["w"] == [40, None]
+
+
+def test_streamed_aggregate_and_join_multibatch(tmp_path):
+    """Multiple input batches take the partial/final streaming path in
+    AggregateOp and the per-batch probe path in JoinOp; results must match
+    the single-batch plan."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    paths = []
+    for i in range(4):
+        p = str(tmp_path / f"f{i}.parquet")
+        pq.write_table(pa.table({
+            "g": [f"k{j % 5}" for j in range(50)],
+            "v": [float(i * 50 + j) for j in range(50)],
+            "k": [(i * 50 + j) % 17 for j in range(50)],
+        }), p)
+        paths.append(p)
+    multi = daft.read_parquet(paths)
+    single = daft.from_pydict(multi.to_pydict())
+
+    for df_pair in [(multi, single)]:
+        m, s = df_pair
+        am = m.groupby("g").agg(
+            col("v").sum().alias("s"), col("v").mean().alias("mu"),
+            col("v").count().alias("c"), col("v").min().alias("lo"),
+            col("v").stddev().alias("sd")).sort("g").to_pydict()
+        as_ = s.groupby("g").agg(
+            col("v").sum().alias("s"), col("v").mean().alias("mu"),
+            col("v").count().alias("c"), col("v").min().alias("lo"),
+            col("v").stddev().alias("sd")).sort("g").to_pydict()
+        assert am["s"] == as_["s"] and am["c"] == as_["c"]
+        assert am["mu"] == pytest.approx(as_["mu"])
+        assert am["sd"] == pytest.approx(as_["sd"])
+
+    # ungrouped
+    assert multi.agg(col("v").sum().alias("s")).to_pydict() == \
+        single.agg(col("v").sum().alias("s")).to_pydict()
+
+    # non-decomposable agg falls back to materialize
+    lm = multi.groupby("g").agg(col("v").agg_list().alias("l")) \
+        .sort("g").to_pydict()
+    ls = single.groupby("g").agg(col("v").agg_list().alias("l")) \
+        .sort("g").to_pydict()
+    assert [sorted(x) for x in lm["l"]] == [sorted(x) for x in ls["l"]]
+
+    # streamed probe join
+    dim = daft.from_pydict({"k": list(range(17)),
+                            "name": [f"n{i}" for i in range(17)]})
+    jm = multi.join(dim, on="k").sort("v").to_pydict()
+    js = single.join(dim, on="k").sort("v").to_pydict()
+    assert jm == js
+    sm = multi.join(dim.where(col("k") < 5), on="k", how="semi") \
+        .sort("v").to_pydict()
+    ss = single.join(dim.where(col("k") < 5), on="k", how="semi") \
+        .sort("v").to_pydict()
+    assert sm == ss

@@ -409,3 +409,37 @@ def test_sql_on_gpu():
                          device="cuda:0")
     out = sql("select g, sum(v) as s from t group by g order by g")
     assert out.to_pydict() == {"g": ["a", "b"], "s": [4.0, 2.0]}
+
+
+def test_out_of_core_host_streaming():
+    """Host-resident source + tiny stream_morsel_rows: the source slices
+    morsels through HBM, aggregates fold partials, joins probe per-morsel;
+    results must equal the device-resident plan."""
+    from daft_amd.context import get_context
+    cfg = get_context().execution_config
+    old = cfg.stream_morsel_rows
+    try:
+        cfg.stream_morsel_rows = 10_000
+        n = 100_000
+        data = {
+            "g": [f"k{i % 7}" for i in range(n)],
+            "v": [float(i % 1000) for i in range(n)],
+            "k": [i % 53 for i in range(n)],
+        }
+        host = daft.from_pydict(data, device="cpu")
+        dev = daft.from_pydict(data, device="cuda:0")
+        dim = daft.from_pydict({"k": list(range(53)),
+                                "w": [i * 2.0 for i in range(53)]},
+                               device="cuda:0")
+        q = lambda df: (df.join(dim, on="k")
+                        .groupby("g")
+                        .agg((col("v") * col("w")).sum().alias("s"),
+                             col("v").count().alias("c"))
+                        .sort("g").to_pydict())
+        out_host = q(host)
+        out_dev = q(dev)
+        assert out_host["g"] == out_dev["g"]
+        assert out_host["c"] == out_dev["c"]
+        assert out_host["s"] == pytest.approx(out_dev["s"], rel=1e-12)
+    finally:
+        cfg.stream_morsel_rows = old
