@@ -180,7 +180,7 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
     if kind in (AggKind.LIST, AggKind.CONCAT):
         # sort rows by group id (stable) then slice by counts
         perm = rowops._stable_sort_perm_by(group_ids)
-        sorted_vals = values.take(perm)
+        sorted_vals = values.take(perm, has_neg=False)
         counts = torch.bincount(group_ids, minlength=num_groups)
         offs = torch.zeros(num_groups + 1, dtype=torch.int64,
                            device=group_ids.device)
@@ -220,7 +220,7 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         perm = rowops.argsort_multi(
             [Series("g", DataType.int64(), data=group_ids), values],
             [False, False], [False, False])
-        svals = values.take(perm).cast(DataType.float64())
+        svals = values.take(perm, has_neg=False).cast(DataType.float64())
         sg = group_ids[perm]
         counts = torch.bincount(group_ids, minlength=num_groups)
         offs = torch.zeros(num_groups + 1, dtype=torch.int64,
@@ -228,7 +228,7 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         torch.cumsum(counts, 0, out=offs[1:])
         pos = offs[:-1] + ((counts - 1).to(torch.float64) * q).to(torch.int64)
         pos = torch.minimum(pos, (offs[1:] - 1).clamp(min=0))
-        out = svals.take(pos)
+        out = svals.take(pos, has_neg=False)
         return out.rename(name).with_validity(counts > 0)
 
     raise ValueError(f"unsupported aggregation {kind}")
@@ -284,7 +284,7 @@ def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
         key_series = [s.broadcast(n) if len(s) == 1 else s for s in key_series]
         gids, reps = rowops.groupby(key_series)
         num_groups = int(reps.shape[0])
-        key_cols = [s.take(reps) for s in key_series]
+        key_cols = [s.take(reps, has_neg=False) for s in key_series]
     else:
         gids = torch.zeros(n, dtype=torch.int64, device=dev)
         num_groups = 1

@@ -453,7 +453,7 @@ class DistinctOp(PhysicalOp):
         else:
             keys = list(batch.columns)
         _gids, reps = rowops.groupby(keys)
-        yield batch.take(reps)
+        yield batch.take(reps, has_neg=False)
 
 
 class SortOp(PhysicalOp):
@@ -474,7 +474,7 @@ class SortOp(PhysicalOp):
             return
         keys = [e.evaluate(batch) for e in self.by]
         perm = rowops.argsort_multi(keys, self.descending, self.nulls_first)
-        yield batch.take(perm)
+        yield batch.take(perm, has_neg=False)
 
 
 class TopNOp(PhysicalOp):
@@ -558,9 +558,15 @@ class JoinOp(PhysicalOp):
 
     def _emit(self, left: RecordBatch, right: RecordBatch,
               lidx: torch.Tensor, ridx: torch.Tensor) -> RecordBatch:
-        cols = [c.take(lidx) for c in left.columns]
+        # index signs are statically known per join type: passing the
+        # has_neg hint skips a (idx < 0).any().item() device sync (and the
+        # clamp kernel) per gathered column
+        l_neg = self.how in ("right", "outer")
+        r_neg = self.how in ("left", "outer")
+        cols = [c.take(lidx, has_neg=l_neg) for c in left.columns]
         for src, out in self.right_cols:
-            cols.append(right.column(src).take(ridx).rename(out))
+            cols.append(right.column(src).take(ridx, has_neg=r_neg)
+                        .rename(out))
         # outer/right joins: fill left-side join keys from the right keys
         if self.how in ("right", "outer") and len(self.left_on):
             has_null_left = bool((lidx < 0).any().item()) if lidx.numel() else False
@@ -624,7 +630,7 @@ class RepartitionOp(PhysicalOp):
         if self.scheme == "hash":
             keys = [e.evaluate(batch) for e in self.by]
             perm, counts = rowops.partition_by_hash(keys, n_parts)
-            reordered = batch.take(perm)
+            reordered = batch.take(perm, has_neg=False)
             start = 0
             for c in counts.tolist():
                 if c:
@@ -667,7 +673,7 @@ class PivotOp(PhysicalOp):
         num_groups = int(reps.shape[0])
         pv = inter.column(pname).cpu().to_pylist()
         vals = inter.column("__pv")
-        out_cols = [k.take(reps) for k in keys]
+        out_cols = [k.take(reps, has_neg=False) for k in keys]
         gid_cpu = gids.cpu().tolist()
         for name in self.names:
             sel = torch.full((num_groups,), -1, dtype=torch.int64,

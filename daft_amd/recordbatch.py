@@ -114,10 +114,12 @@ class RecordBatch:
                             for c in self.columns], self._num_rows)
 
     # ------------------------------------------------------------------
-    def take(self, indices: torch.Tensor) -> "RecordBatch":
+    def take(self, indices: torch.Tensor,
+             has_neg: "bool | None" = None) -> "RecordBatch":
         # hoist the negative-index host sync across columns
-        has_neg = bool((indices < 0).any().item()) if indices.numel() else \
-            False
+        if has_neg is None:
+            has_neg = bool((indices < 0).any().item()) if indices.numel() \
+                else False
         return RecordBatch([c.take(indices, has_neg=has_neg)
                             for c in self.columns],
                            int(indices.shape[0]))
