@@ -68,16 +68,23 @@ class Source(LogicalPlan):
     planner skip exchanges for co-sharded tables)."""
 
     def __init__(self, schema: Schema, cache_key: str, num_rows: int,
-                 size_bytes: int = 0, partitioning=None):
+                 size_bytes: int = 0, partitioning=None, columns=None):
         super().__init__([])
-        self._schema = schema
+        self._full_schema = schema
         self.cache_key = cache_key
         self.num_rows = num_rows
         self.size_bytes = size_bytes
         self.partitioning = partitioning
+        # projection pushed INTO the source: cached partitions narrow to
+        # these columns before any H2D morsel transfer (out-of-core scans
+        # only move what the query reads)
+        self.columns = columns
 
     def _compute_schema(self):
-        return self._schema
+        if self.columns is not None:
+            return Schema([f for f in self._full_schema
+                           if f.name in set(self.columns)])
+        return self._full_schema
 
     def with_children(self, children):
         assert not children
@@ -90,7 +97,8 @@ class Source(LogicalPlan):
         return f"Source(rows={self.num_rows})"
 
     def semantic_id(self) -> str:
-        return f"Source({self.cache_key})"
+        cols = ",".join(self.columns) if self.columns is not None else "*"
+        return f"Source({self.cache_key}|{cols})"
 
 
 class ScanSource(LogicalPlan):

@@ -420,8 +420,13 @@ def _prune(node: lp.LogicalPlan, needed: Set[str]) -> Optional[lp.LogicalPlan]:
                              node.pushdown_filter, node.pushdown_limit,
                              node.read_options)
     if isinstance(node, lp.Source):
-        # narrow in-memory sources with an explicit Project
-        return lp.Project(node, [ColumnRef(n) for n in keep])
+        # narrow INSIDE the source: columns are dropped before partitions
+        # move to the device (matters for host-resident / streamed tables)
+        cols = [f.name for f in node._full_schema if f.name in keep]
+        if node.columns is not None and list(node.columns) == cols:
+            return None
+        return lp.Source(node._full_schema, node.cache_key, node.num_rows,
+                         node.size_bytes, node.partitioning, cols)
     if isinstance(node, lp.Project):
         new_exprs = [e for e in node.exprs
                      if e.to_field(node.children[0].schema).name in keep]

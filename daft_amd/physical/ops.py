@@ -105,14 +105,18 @@ class PhysicalOp:
 
 
 class InMemorySourceOp(PhysicalOp):
-    def __init__(self, schema: Schema, cache_key: str):
+    def __init__(self, schema: Schema, cache_key: str, columns=None):
         super().__init__([], schema, "InMemorySource")
         self.cache_key = cache_key
+        self.columns = columns
 
     def execute(self, ectx) -> BatchIter:
         morsel = getattr(ectx.ctx.execution_config, "stream_morsel_rows",
                          1 << 26)
         for part in ectx.ctx.cache.get(self.cache_key):
+            if self.columns is not None and \
+                    len(self.columns) < len(part.columns):
+                part = part.select_columns(self.columns)
             if part.device != ectx.device:
                 if str(part.device) == "cpu" and \
                         str(ectx.device).startswith("cuda") and \

@@ -24,7 +24,8 @@ def translate(plan: lp.LogicalPlan) -> ops.PhysicalOp:
                                     plan.nulls_first)
 
     if isinstance(plan, lp.Source):
-        return ops.InMemorySourceOp(plan.schema, plan.cache_key)
+        return ops.InMemorySourceOp(plan.schema, plan.cache_key,
+                                     plan.columns)
     if isinstance(plan, lp.ScanSource):
         return ops.ScanOp(plan.schema, plan.paths, plan.file_format,
                           plan.storage_options, plan.read_options,
