@@ -330,11 +330,61 @@ class Series:
         return self.take(idx)
 
     def slice(self, start: int, end: int) -> "Series":
+        """Contiguous row slice — tensor VIEWS, not gathers (hot for
+        out-of-core morsel streaming: slicing a host partition must not
+        copy the table)."""
         n = len(self)
         start = max(0, min(start, n))
         end = max(start, min(end, n))
-        idx = torch.arange(start, end, dtype=torch.int64, device=self.device)
-        return self.take(idx)
+        if start == 0 and end == n:
+            return self
+        k = self.dtype.kind
+        validity = self.validity[start:end] \
+            if self.validity is not None else None
+        if k == TypeKind.PYTHON:
+            return Series(self.name, self.dtype,
+                          pyobjs=self.pyobjs[start:end], validity=validity,
+                          length=end - start)
+        if self.is_dict():
+            return Series.make_dict(self.name, self.children[0],
+                                    self.data[start:end], validity)
+        if k in (TypeKind.STRING, TypeKind.BINARY):
+            offs = self.offsets[start:end + 1]
+            lo = int(offs[0])
+            hi = int(offs[-1])
+            return Series(self.name, self.dtype,
+                          data=self.data[lo:hi], offsets=offs - lo,
+                          validity=validity)
+        if k == TypeKind.LIST:
+            offs = self.offsets[start:end + 1]
+            lo = int(offs[0])
+            hi = int(offs[-1])
+            return Series(self.name, self.dtype, offsets=offs - lo,
+                          children=[self.children[0].slice(lo, hi)],
+                          validity=validity)
+        if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING,
+                 TypeKind.FIXED_SHAPE_TENSOR, TypeKind.FIXED_SHAPE_IMAGE):
+            if k in (TypeKind.FIXED_SHAPE_TENSOR, TypeKind.FIXED_SHAPE_IMAGE):
+                sz = 1
+                for d in self.dtype.shape:
+                    sz *= d
+                if k == TypeKind.FIXED_SHAPE_IMAGE:
+                    sz *= {"L": 1, "LA": 2, "RGB": 3, "RGBA": 4}.get(
+                        self.dtype.image_mode or "RGB", 3)
+            else:
+                sz = self.dtype.size
+            return Series(self.name, self.dtype,
+                          children=[self.children[0].slice(start * sz,
+                                                           end * sz)],
+                          validity=validity, length=end - start)
+        if k in (TypeKind.STRUCT, TypeKind.IMAGE, TypeKind.TENSOR):
+            # row-aligned children: slice each
+            return Series(self.name, self.dtype,
+                          children=[c.slice(start, end)
+                                    for c in self.children],
+                          validity=validity, length=end - start)
+        return Series(self.name, self.dtype, data=self.data[start:end],
+                      validity=validity)
 
     def head(self, n: int) -> "Series":
         return self.slice(0, n)
