@@ -32,7 +32,7 @@ from .window import Window  # noqa: F401
 from .checkpoint import (CheckpointConfig, CheckpointStore,  # noqa: F401
                          LocalCheckpointStore, MemoryCheckpointStore)
 from .sql import sql  # noqa: F401
-from .udf import func, udf, cls, method  # noqa: F401
+from .udf import func, udf, cls, method, udaf  # noqa: F401
 from .functions import coalesce  # noqa: F401
 
 

@@ -403,6 +403,8 @@ class AggKind:
     # approx_percentile (physical/sketch.py)
     SKETCH = "__sketch"
     SKETCH_FINAL = "__sketch_final"
+    # user-defined aggregation (daft_amd.udaf): param = (instance, dtype)
+    PY_UDAF = "__py_udaf"
 
 
 _NUMERIC_AGGS = {AggKind.SUM, AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE,
@@ -463,6 +465,8 @@ class Agg(ExprNode):
             return Field(f.name, SKETCH_DTYPE)
         if k == AggKind.SKETCH_FINAL:
             return Field(f.name, DataType.float64())
+        if k == AggKind.PY_UDAF:
+            return Field(f.name, self.param[1])
         raise ValueError(f"unknown agg kind {k}")
 
     def evaluate(self, batch) -> Series:

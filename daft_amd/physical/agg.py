@@ -212,6 +212,25 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         return sketch.grouped_sketch_final(values, group_ids, num_groups,
                                            float(agg.param), name)
 
+    if kind == AggKind.PY_UDAF:
+        # user-defined aggregation (daft_amd.udaf): sort rows by group,
+        # call aggregate() per group slice, then finalize() (ref:
+        # /root/reference/daft/udf/udaf.py:16-105 aggregate/combine/
+        # finalize pipeline; combine() is exercised when partial states
+        # merge across shards)
+        inst, ret_dt = agg.param[0], agg.param[1]
+        perm = torch.argsort(group_ids, stable=True)
+        svals = values.take(perm, has_neg=False).cpu()
+        counts = torch.bincount(group_ids, minlength=num_groups)
+        offs = torch.zeros(num_groups + 1, dtype=torch.int64)
+        torch.cumsum(counts.cpu(), 0, out=offs[1:])
+        outs = []
+        for g in range(num_groups):
+            lo, hi = int(offs[g]), int(offs[g + 1])
+            state = inst.aggregate(svals.slice(lo, hi))
+            outs.append(inst.finalize(state))
+        return Series.from_pylist(name, outs, ret_dt)
+
     if kind == AggKind.APPROX_PERCENTILE:
         # per-group exact percentile via sort (single-node; the
         # distributed planner splits into SKETCH/SKETCH_FINAL — a

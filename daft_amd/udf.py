@@ -96,8 +96,34 @@ def method(fn: Optional[Callable] = None, *,
     return wrap
 
 
-def udaf(*args, **kwargs):
-    raise NotImplementedError("user-defined aggregations land next round")
+def udaf(cls=None, *, return_dtype: DataType, state=None):
+    """@daft_amd.udaf — user-defined aggregation from a class defining
+    aggregate(values: Series) -> state, combine(states) -> state, and
+    finalize(state) -> value (ref: /root/reference/daft/udf/udaf.py:16:
+    the three-stage aggregate/combine/finalize pipeline; combine must be
+    associative & commutative)."""
+    from .expressions.expressions import Agg, AggKind
+
+    def wrap(klass):
+        for m in ("aggregate", "combine", "finalize"):
+            if not callable(getattr(klass, m, None)):
+                raise ValueError(
+                    f"UDAF class `{klass.__name__}` must define `{m}`")
+
+        class WrappedUDAF:
+            __name__ = klass.__name__
+
+            def __init__(self, *a, **k):
+                self._inst = klass(*a, **k)
+
+            def __call__(self, expr) -> Expression:
+                return Expression(Agg(AggKind.PY_UDAF, _to_node(expr),
+                                      (self._inst, return_dtype, state)))
+        WrappedUDAF.__qualname__ = klass.__qualname__
+        return WrappedUDAF
+    if cls is not None:
+        return wrap(cls)
+    return wrap
 
 
 def _dtype_from_hint(hint) -> DataType:

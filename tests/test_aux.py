@@ -534,3 +534,39 @@ def test_native_extension_plugin(tmp_path):
     ).to_pydict()
     assert out["a1"] == [2, 3, None]
     assert out["h"] == [5.0, 13.0, 17.0]
+
+
+def test_udaf():
+    from daft_amd import udaf
+    from daft_amd.series import Series as S
+
+    @udaf(return_dtype=DataType.float64())
+    class GeoMean:
+        def aggregate(self, values):
+            import math
+            vals = [v for v in values.to_pylist() if v is not None]
+            return (sum(math.log(v) for v in vals), len(vals))
+
+        def combine(self, states):
+            s = c = 0
+            for a, b in states:
+                s += a
+                c += b
+            return (s, c)
+
+        def finalize(self, state):
+            import math
+            s, c = state
+            return math.exp(s / c) if c else None
+
+    df = daft.from_pydict({"g": ["a", "a", "b", "b", "b"],
+                           "v": [2.0, 8.0, 1.0, 1.0, 27.0]})
+    out = df.groupby("g").agg(GeoMean()(col("v")).alias("gm")) \
+        .sort("g").to_pydict()
+    assert out["gm"][0] == pytest.approx(4.0)
+    assert out["gm"][1] == pytest.approx(3.0)
+    # ungrouped
+    tot = df.agg(GeoMean()(col("v")).alias("gm")).to_pydict()["gm"][0]
+    import math
+    assert tot == pytest.approx(math.exp(sum(map(math.log,
+                                                 [2, 8, 1, 1, 27])) / 5))
