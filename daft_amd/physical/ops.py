@@ -568,9 +568,18 @@ class JoinOp(PhysicalOp):
         l_neg = self.how in ("right", "outer")
         r_neg = self.how in ("left", "outer")
         cols = [c.take(lidx, has_neg=l_neg) for c in left.columns]
+        fan_out = len(ridx) >= 4 * max(len(right), 1)
         for src, out in self.right_cols:
-            cols.append(right.column(src).take(ridx, has_neg=r_neg)
-                        .rename(out))
+            c = right.column(src)
+            if fan_out and not c.is_dict() and \
+                    c.dtype.kind in (TypeKind.STRING, TypeKind.BINARY) and \
+                    len(c) <= 8_000_000:
+                # high-fan-out payload strings: dict-encode the (small)
+                # build side once so the probe-scale gather moves int32
+                # codes, not bytes — and downstream groupbys take the
+                # dict fast path
+                c = _dict_encode(c)
+            cols.append(c.take(ridx, has_neg=r_neg).rename(out))
         # outer/right joins: fill left-side join keys from the right keys
         if self.how in ("right", "outer") and len(self.left_on):
             has_null_left = bool((lidx < 0).any().item()) if lidx.numel() else False
@@ -583,6 +592,16 @@ class JoinOp(PhysicalOp):
                     filled = cols[i].fill_null(rcol)
                     cols[i] = filled.rename(lname)
         return RecordBatch(cols, num_rows=int(lidx.shape[0]))
+
+
+def _dict_encode(c: Series) -> Series:
+    """Dedup a string column into a dictionary Series (distinct vocab,
+    int32 codes) — groupby dedup, so safe even with duplicate values."""
+    gids, reps = rowops.groupby([c])
+    vocab = c.take(reps, has_neg=False)
+    vocab = Series(vocab.name, vocab.dtype, data=vocab.data,
+                   offsets=vocab.offsets, children=vocab.children)
+    return Series.make_dict(c.name, vocab, gids.to(torch.int32), c.validity)
 
 
 class AsofJoinOp(PhysicalOp):
