@@ -48,3 +48,4 @@ def from_pylist(rows, device=None):
 
 from . import datasets  # noqa: E402
 from .ext import load_extension, ext_function  # noqa: E402
+from .file import File  # noqa: E402,F401

@@ -80,3 +80,35 @@ def uuid() -> Expression:
     from ..expressions.expressions import ColumnRef
     raise NotImplementedError("uuid() requires a column context; "
                               "use df.add_monotonically_increasing_id")
+
+from .spatial import great_circle_distance  # noqa: E402,F401
+
+
+def file(expr) -> "Expression":
+    """Wrap a path (string) or bytes column as File objects (ref
+    capability: daft.functions.file / daft.File columns)."""
+    from ..expressions.expressions import Expression, ScalarFn, _to_node
+    from ..schema import DataType
+    from ..series import Series
+    from ..file import File
+
+    def run(s):
+        vals = s.cpu().to_pylist()
+        objs = [None if v is None else File(v) for v in vals]
+        return Series(s.name, DataType.python(), pyobjs=objs,
+                      validity=None, length=len(vals))
+    return Expression(ScalarFn("file", run, [_to_node(expr)],
+                               DataType.python()))
+
+
+def file_size(expr) -> "Expression":
+    """Size in bytes of a File column (ref: daft/functions/file_.py:110)."""
+    from ..expressions.expressions import Expression, ScalarFn, _to_node
+    from ..schema import DataType
+    from ..series import Series
+
+    def run(s):
+        out = [None if f is None else f.size() for f in s.pyobjs]
+        return Series.from_pylist(s.name, out, DataType.int64())
+    return Expression(ScalarFn("file_size", run, [_to_node(expr)],
+                               DataType.int64()))

@@ -570,3 +570,34 @@ def test_udaf():
     import math
     assert tot == pytest.approx(math.exp(sum(map(math.log,
                                                  [2, 8, 1, 1, 27])) / 5))
+
+
+def test_great_circle_distance():
+    from daft_amd.functions import great_circle_distance
+    df = daft.from_pydict({"lat1": [48.8566, 91.0], "lon1": [2.3522, 0.0],
+                           "lat2": [51.5074, 0.0], "lon2": [-0.1278, 0.0]})
+    out = df.select(great_circle_distance(
+        col("lat1"), col("lon1"), col("lat2"), col("lon2")).alias("d")) \
+        .to_pydict()["d"]
+    assert abs(out[0] - 343_556) < 2000
+    assert out[1] is None  # invalid latitude
+
+
+def test_file_type(tmp_path):
+    from daft_amd import File
+    from daft_amd.functions import file, file_size
+    p = tmp_path / "x.png"
+    p.write_bytes(b"\x89PNG\r\n\x1a\n" + b"0" * 100)
+    f = File(str(p))
+    assert f.size() == 108 and f.exists() and f.is_image()
+    assert f.mime_type() == "image/png"
+    with f.open() as h:
+        assert h.read(4) == b"\x89PNG"
+    m = File(b"abcdef")
+    assert m.size() == 6 and m.read() == b"abcdef"
+    with m.to_tempfile() as tf:
+        assert open(tf.name, "rb").read() == b"abcdef"
+
+    df = daft.from_pydict({"p": [str(p), None]})
+    out = df.select(file_size(file(col("p"))).alias("sz")).to_pydict()["sz"]
+    assert out == [108, None]
