@@ -44,7 +44,7 @@ def read_file(path: str, file_format: str, columns: Optional[List[str]],
               predicate, limit: Optional[int], storage_options: dict,
               read_options: dict, device) -> Iterator[RecordBatch]:
     if file_format == "parquet":
-        yield from _read_parquet(path, columns, limit, device)
+        yield from _read_parquet(path, columns, limit, device, predicate)
     elif file_format == "csv":
         yield from _read_csv(path, columns, read_options, device)
     elif file_format == "json":
@@ -58,7 +58,8 @@ def read_file(path: str, file_format: str, columns: Optional[List[str]],
 def read_files_prefetch(paths: List[str], file_format: str,
                         columns: Optional[List[str]], storage_options: dict,
                         read_options: dict, device,
-                        window: int = 4) -> Iterator[RecordBatch]:
+                        window: int = 4,
+                        predicate=None) -> Iterator[RecordBatch]:
     """Ordered multi-file scan with background prefetch.
 
     Up to `window` files decode concurrently on threads (pyarrow releases
@@ -70,7 +71,7 @@ def read_files_prefetch(paths: List[str], file_format: str,
     from collections import deque
 
     def load(path):
-        return list(read_file(path, file_format, columns, None, None,
+        return list(read_file(path, file_format, columns, predicate, None,
                               storage_options, read_options, "cpu"))
 
     with fut.ThreadPoolExecutor(max_workers=window) as ex:
@@ -90,13 +91,20 @@ def read_files_prefetch(paths: List[str], file_format: str,
                 yield rb.to(device) if str(device) != "cpu" else rb
 
 
-def _read_parquet(path, columns, limit, device) -> Iterator[RecordBatch]:
+def _read_parquet(path, columns, limit, device,
+                  predicate=None) -> Iterator[RecordBatch]:
     import pyarrow.parquet as pq
     f = pq.ParquetFile(path)
     remaining = limit
+    bounds = _predicate_bounds(predicate)
+    name_to_idx = {c: i for i, c in enumerate(f.schema_arrow.names)} \
+        if bounds else {}
     for rg in range(f.num_row_groups):
         if remaining is not None and remaining <= 0:
             return
+        if bounds and not _rg_may_match(f.metadata.row_group(rg),
+                                        name_to_idx, bounds):
+            continue  # statistics prove no row in this group matches
         tbl = f.read_row_group(rg, columns=columns)
         if remaining is not None and tbl.num_rows > remaining:
             tbl = tbl.slice(0, remaining)
@@ -224,3 +232,55 @@ def _read_warc(path, columns, device,
                 yield flush()
     if cols["warc_content"]:
         yield flush()
+
+
+def _predicate_bounds(predicate):
+    """Extract (column, op, literal) conjuncts usable for row-group stats
+    pruning (capability of the reference's daft-stats TableStatistics
+    min/max pruning + daft-parquet predicate pushdown)."""
+    from ..expressions.expressions import BinaryOp, ColumnRef, Literal
+    out = []
+
+    def walk(e):
+        if isinstance(e, BinaryOp) and e.op == "and":
+            walk(e.left)
+            walk(e.right)
+            return
+        if isinstance(e, BinaryOp) and e.op in ("lt", "le", "gt", "ge",
+                                                "eq"):
+            l, r = e.left, e.right
+            if isinstance(l, ColumnRef) and isinstance(r, Literal):
+                out.append((l.name, e.op, r.value))
+            elif isinstance(r, ColumnRef) and isinstance(l, Literal):
+                flip = {"lt": "gt", "le": "ge", "gt": "lt", "ge": "le",
+                        "eq": "eq"}[e.op]
+                out.append((r.name, flip, l.value))
+    if predicate is not None:
+        walk(predicate)
+    return out
+
+
+def _rg_may_match(meta_rg, name_to_idx, bounds) -> bool:
+    """False only when column chunk statistics PROVE no row matches."""
+    for cname, op, val in bounds:
+        i = name_to_idx.get(cname)
+        if i is None:
+            continue
+        st = meta_rg.column(i).statistics
+        if st is None or not st.has_min_max:
+            continue
+        mn, mx = st.min, st.max
+        try:
+            if op == "lt" and not (mn < val):
+                return False
+            if op == "le" and not (mn <= val):
+                return False
+            if op == "gt" and not (mx > val):
+                return False
+            if op == "ge" and not (mx >= val):
+                return False
+            if op == "eq" and not (mn <= val <= mx):
+                return False
+        except TypeError:
+            continue  # incomparable types: keep the row group
+    return True

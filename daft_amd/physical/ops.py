@@ -154,7 +154,8 @@ class ScanOp(PhysicalOp):
             # no limit to push down: overlap host decode across files
             yield from readers.read_files_prefetch(
                 self.paths, self.file_format, self.columns,
-                self.storage_options, self.read_options, ectx.device)
+                self.storage_options, self.read_options, ectx.device,
+                predicate=self.predicate)
             return
         for path in self.paths:
             if remaining is not None and remaining <= 0:

@@ -601,3 +601,23 @@ def test_file_type(tmp_path):
     df = daft.from_pydict({"p": [str(p), None]})
     out = df.select(file_size(file(col("p"))).alias("sz")).to_pydict()["sz"]
     assert out == [108, None]
+
+
+def test_parquet_rowgroup_stats_pruning(tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    import daft_amd as daft
+    from daft_amd.io import readers
+    p = str(tmp_path / "t.parquet")
+    # 4 row groups of 100 rows, a increasing
+    pq.write_table(pa.table({"a": list(range(400)),
+                             "b": [i * 2.0 for i in range(400)]}),
+                   p, row_group_size=100)
+    df = daft.read_parquet(p).where(col("a") >= 350)
+    out = df.to_pydict()
+    assert out["a"] == list(range(350, 400))
+    # verify groups actually skipped
+    from daft_amd.expressions.expressions import BinaryOp, ColumnRef, Literal
+    pred = BinaryOp("ge", ColumnRef("a"), Literal(350))
+    batches = list(readers._read_parquet(p, None, None, "cpu", pred))
+    assert sum(len(b) for b in batches) == 100  # only the last group read
