@@ -193,6 +193,53 @@ def crop_series(s: Series, x: int, y: int, w: int, h: int) -> Series:
         validity=s.validity, length=n)
 
 
+
+
+_LUMA = (0.299, 0.587, 0.114)  # ITU-R 601-2, matching PIL convert("L")
+
+
+def to_mode_series(s: Series, mode: str) -> Series:
+    """Convert FixedShapeImage between L/RGB/RGBA on the device (ref:
+    /root/reference/src/daft-image/src/series.rs:186 to_mode)."""
+    assert s.dtype.kind == TypeKind.FIXED_SHAPE_IMAGE, \
+        f"to_mode expects FixedShapeImage, got {s.dtype!r}"
+    src_mode = s.dtype.image_mode or "RGB"
+    if mode == src_mode:
+        return s
+    h, w = s.dtype.shape
+    cs = _MODE_CHANNELS[src_mode]
+    cd = _MODE_CHANNELS[mode]
+    n = len(s)
+    img = s.children[0].data.reshape(n, h, w, cs)
+    f = img.to(torch.float32)
+    if src_mode in ("L", "LA"):
+        lum = f[..., 0]
+        alpha = f[..., 1] if src_mode == "LA" else None
+    else:
+        lum = (f[..., 0] * _LUMA[0] + f[..., 1] * _LUMA[1]
+               + f[..., 2] * _LUMA[2])
+        alpha = f[..., 3] if src_mode == "RGBA" else None
+    if alpha is None:
+        alpha = torch.full_like(lum, 255.0)
+    if mode == "L":
+        out = lum.unsqueeze(-1)
+    elif mode == "LA":
+        out = torch.stack([lum, alpha], dim=-1)
+    elif mode in ("RGB", "RGBA"):
+        if src_mode in ("L", "LA"):
+            rgb = lum.unsqueeze(-1).expand(n, h, w, 3)
+        else:
+            rgb = f[..., :3]
+        out = torch.cat([rgb, alpha.unsqueeze(-1)], dim=-1) \
+            if mode == "RGBA" else rgb
+    else:
+        raise ValueError(f"unknown image mode {mode!r}")
+    data = out.round().clamp(0, 255).to(torch.uint8).reshape(-1)
+    child = Series("item", DataType.uint8(), data=data.contiguous())
+    return Series(s.name, DataType.fixed_shape_image(mode, h, w),
+                  children=[child], validity=s.validity, length=n)
+
+
 class ImageNamespace(_Namespace):
     def decode(self, mode: str = "RGB", on_error: str = "raise"):
         return self._fn("image_decode", decode_series, DataType.image(),
@@ -213,6 +260,12 @@ class ImageNamespace(_Namespace):
             return DataType.fixed_shape_tensor(DataType.float32(),
                                                (c,) + tuple(dt.shape))
         return self._fn("image_to_tensor", to_tensor_series, ret)
+
+    def to_mode(self, mode: str):
+        def ret(fields):
+            dt = fields[0].dtype
+            return DataType.fixed_shape_image(mode, *dt.shape)
+        return self._fn("image_to_mode", to_mode_series, ret, mode)
 
     def crop(self, x: int, y: int, w: int, h: int):
         def ret(fields):

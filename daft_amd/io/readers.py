@@ -24,12 +24,13 @@ def infer_schema(path: str, file_format: str,
         read_opts = pacsv.ReadOptions(
             autogenerate_column_names=not ro.get("has_headers", True))
         parse_opts = pacsv.ParseOptions(delimiter=ro.get("delimiter", ","))
-        with pacsv.open_csv(path, read_options=read_opts,
+        with pacsv.open_csv(_open_maybe_compressed(path),
+                            read_options=read_opts,
                             parse_options=parse_opts) as reader:
             a_schema = reader.schema
     elif file_format == "json":
         import pyarrow.json as pajson
-        tbl = pajson.read_json(path)
+        tbl = pajson.read_json(_open_maybe_compressed(path))
         a_schema = tbl.schema
     elif file_format == "warc":
         return warc_schema()
@@ -114,10 +115,22 @@ def _read_parquet(path, columns, limit, device,
         yield rb
 
 
+def _open_maybe_compressed(path):
+    """Transparent .gz/.zst/.bz2 input (capability of the reference's
+    daft-compression CompressionCodec::from_uri)."""
+    import pyarrow as pa
+    for ext, codec in ((".gz", "gzip"), (".zst", "zstd"), (".zstd", "zstd"),
+                       (".bz2", "bz2"), (".lz4", "lz4")):
+        if path.endswith(ext):
+            return pa.CompressedInputStream(pa.OSFile(path, "rb"), codec)
+    return path
+
+
 def _read_csv(path, columns, read_options, device) -> Iterator[RecordBatch]:
     import pyarrow as pa
     import pyarrow.csv as pacsv
     ro = read_options or {}
+    path = _open_maybe_compressed(path)
     read_opts = pacsv.ReadOptions(
         autogenerate_column_names=not ro.get("has_headers", True),
         block_size=64 * 1024 * 1024)
@@ -136,7 +149,7 @@ def _read_csv(path, columns, read_options, device) -> Iterator[RecordBatch]:
 
 def _read_json(path, columns, device) -> Iterator[RecordBatch]:
     import pyarrow.json as pajson
-    tbl = pajson.read_json(path)
+    tbl = pajson.read_json(_open_maybe_compressed(path))
     if columns:
         tbl = tbl.select(columns)
     yield RecordBatch.from_arrow(tbl, device=device)

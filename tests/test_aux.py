@@ -621,3 +621,17 @@ def test_parquet_rowgroup_stats_pruning(tmp_path):
     pred = BinaryOp("ge", ColumnRef("a"), Literal(350))
     batches = list(readers._read_parquet(p, None, None, "cpu", pred))
     assert sum(len(b) for b in batches) == 100  # only the last group read
+
+
+def test_compressed_csv_json(tmp_path):
+    import gzip
+    import daft_amd as daft
+    pc = str(tmp_path / "t.csv.gz")
+    with gzip.open(pc, "wt") as f:
+        f.write("a,b\n1,x\n2,y\n")
+    out = daft.read_csv(pc).to_pydict()
+    assert out == {"a": [1, 2], "b": ["x", "y"]}
+    pj = str(tmp_path / "t.jsonl.gz")
+    with gzip.open(pj, "wt") as f:
+        f.write('{"a": 1}\n{"a": 5}\n')
+    assert daft.read_json(pj).to_pydict() == {"a": [1, 5]}

@@ -184,3 +184,21 @@ def test_simhash():
     def ham(a, b):
         return bin(a ^ b).count("1")
     assert ham(out[0], out[1]) < ham(out[0], out[2])
+
+
+def test_image_to_mode():
+    arr = np.zeros((4, 4, 3), np.uint8)
+    arr[..., 0] = 200
+    arr[..., 1] = 100
+    arr[..., 2] = 50
+    import io as _io
+    from PIL import Image
+    b = _io.BytesIO()
+    Image.fromarray(arr, "RGB").save(b, format="PNG")
+    df = daft.from_pydict({"b": [b.getvalue()]})
+    img = col("b").image.decode().image.resize(4, 4)
+    l = df.select(img.image.to_mode("L").alias("l")).to_pydict()["l"][0]
+    want = np.asarray(Image.fromarray(arr, "RGB").convert("L"))[0, 0]
+    assert abs(int(np.array(l).reshape(4, 4)[0, 0]) - int(want)) <= 1
+    a = df.select(img.image.to_mode("RGBA").alias("a")).to_pydict()["a"][0]
+    assert len(a) == 4 * 4 * 4 and a[3] == 255
