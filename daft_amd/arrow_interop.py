@@ -161,6 +161,18 @@ def from_arrow_array(name: str, arr) -> Series:
         return Series(name, dt, children=children, validity=validity,
                       length=len(arr))
     if k == TypeKind.DECIMAL128:
+        if dt.to_physical().kind == TypeKind.INT64:
+            # exact: arrow decimal128 stores scaled i128 little-endian; for
+            # p <= 18 every value fits the low 64 bits (two's complement)
+            a2 = arr.combine_chunks() if hasattr(arr, "combine_chunks") \
+                else arr
+            buf = a2.buffers()[1]
+            off = a2.offset
+            raw = np.frombuffer(buf, dtype=np.int64,
+                                count=2 * (off + len(a2)))
+            lows = np.ascontiguousarray(raw[2 * off::2][:len(a2)])
+            return Series(name, dt, data=torch.from_numpy(lows.copy()),
+                          validity=validity)
         vals = arr.cast(pa.float64())
         np_vals = vals.to_numpy(zero_copy_only=False)
         np_vals = np.nan_to_num(np_vals) if arr.null_count else np_vals
@@ -233,6 +245,15 @@ def to_arrow_array(s: Series):
         vbuf = pa.array(~mask).buffers()[1] if mask is not None else None
         return pa.Array.from_buffers(atype, len(s), [vbuf], children=children)
     if k == TypeKind.DECIMAL128:
+        if s.data.dtype == torch.int64:
+            lows = s.data.numpy()
+            i128 = np.empty((len(s), 2), dtype=np.int64)
+            i128[:, 0] = lows
+            i128[:, 1] = lows >> 63  # sign extension into the high half
+            vbuf = pa.array(~mask).buffers()[1] if mask is not None \
+                else None
+            return pa.Array.from_buffers(
+                atype, len(s), [vbuf, pa.py_buffer(i128.tobytes())])
         vals = s.data.numpy()
         return pa.array(vals, mask=mask).cast(atype)
     if k == TypeKind.NULL:

@@ -24,6 +24,10 @@ def _normalize_key(s: Series) -> Series:
     if dt.kind == TypeKind.FLOAT32:
         return s.cast(DataType.float64())
     if dt.is_decimal():
+        # p <= 18 decimals are scaled int64: hash the raw ints (exact);
+        # wide decimals stored f64 normalize like floats
+        if s.data is not None and s.data.dtype == torch.int64:
+            return s
         return s.cast(DataType.float64())
     if s.is_dict():
         # vocab order may differ across ranks/tables: hash real bytes

@@ -443,6 +443,12 @@ class Agg(ExprNode):
                  AggKind.APPROX_COUNT_DISTINCT):
             return Field(f.name, DataType.uint64())
         if k == AggKind.SUM:
+            if f.dtype.is_decimal():
+                # widen precision for the running sum (ref: decimal sum
+                # gets p=38; int64 storage caps us at 18)
+                return Field(f.name, DataType.decimal128(
+                    18, f.dtype.scale) if f.dtype.precision <= 18
+                    else DataType.float64())
             if f.dtype.is_integer():
                 return Field(f.name, DataType.int64()
                              if f.dtype.is_signed_integer()

@@ -303,6 +303,41 @@ def cast(s: Series, dtype: DataType) -> Series:
         return full_null(s.name, dtype, len(s), s.device)
     if k == TypeKind.NULL:
         return full_null(s.name, dtype, len(s), s.device)
+    if k == TypeKind.DECIMAL128 and s.data is not None and \
+            s.data.dtype == torch.int64:
+        sc = s.dtype.scale
+        if nk == TypeKind.DECIMAL128:
+            if dtype.to_physical().kind == TypeKind.INT64:
+                d = s.data
+                if dtype.scale > sc:
+                    d = d * (10 ** (dtype.scale - sc))
+                elif dtype.scale < sc:
+                    div = 10 ** (sc - dtype.scale)
+                    d = torch.div(d + torch.sign(d) * (div // 2), div,
+                                  rounding_mode="trunc")
+                return Series(s.name, dtype, data=d, validity=s.validity)
+            return Series(s.name, dtype,
+                          data=s.data.to(torch.float64) / (10 ** sc),
+                          validity=s.validity)
+        if dtype.is_floating():
+            out = (s.data.to(torch.float64) / (10 ** sc)) \
+                .to(dtype.to_torch())
+            return Series(s.name, dtype, data=out, validity=s.validity)
+        if dtype.is_integer():
+            out = torch.div(s.data, 10 ** sc, rounding_mode="trunc") \
+                .to(dtype.to_torch())
+            return Series(s.name, dtype, data=out, validity=s.validity)
+    if nk == TypeKind.DECIMAL128 and \
+            dtype.to_physical().kind == TypeKind.INT64 and \
+            s.data is not None and s.dtype.is_numeric():
+        mul = 10 ** dtype.scale
+        if s.dtype.is_floating() or (s.dtype.is_decimal() and
+                                     s.data.dtype == torch.float64):
+            out = torch.round(s.data.to(torch.float64) * mul) \
+                .to(torch.int64)
+        else:
+            out = s.data.to(torch.int64) * mul
+        return Series(s.name, dtype, data=out, validity=s.validity)
     if dtype.is_fixed_width() and s.dtype.is_fixed_width():
         tdt = dtype.to_torch()
         data = s.data
@@ -321,7 +356,12 @@ def cast(s: Series, dtype: DataType) -> Series:
                                   device=s.device)
     if k == TypeKind.STRING and dtype.is_numeric():
         vals = s.to_pylist()
-        conv = float if (dtype.is_floating() or dtype.is_decimal()) else int
+        if dtype.is_decimal():
+            conv = lambda x: x  # Decimal(str) parse in from_pylist: exact
+        elif dtype.is_floating():
+            conv = float
+        else:
+            conv = int
         out = [None if v is None or v == "" else conv(v) for v in vals]
         return Series.from_pylist(s.name, out, dtype, device=s.device)
     if k == TypeKind.STRING and nk == TypeKind.DATE:
@@ -425,6 +465,10 @@ def binary_op(l: Series, r: Series, op: str) -> Series:
     validity = _null_validity(l, r)
     if l.dtype.is_temporal() or r.dtype.is_temporal():
         return _temporal_binary(l, r, op, validity)
+    if (l.dtype.is_decimal() or r.dtype.is_decimal()):
+        out = _decimal_binary(l, r, op, validity)
+        if out is not None:
+            return out
     out_dt = supertype(l.dtype, r.dtype)
     if op == "div" and not out_dt.is_floating():
         out_dt = DataType.float64()
@@ -449,6 +493,63 @@ def binary_op(l: Series, r: Series, op: str) -> Series:
         raise ValueError(f"unknown binary op {op}")
     return Series(l.name, out_dt, data=out,
                   validity=_expand_validity(validity, int(out.shape[0])))
+
+
+def _dec_phys_int(dt: DataType) -> bool:
+    return dt.is_decimal() and dt.to_physical().kind == TypeKind.INT64
+
+
+def _decimal_binary(l: Series, r: Series, op: str, validity):
+    """Exact decimal arithmetic on scaled int64 (ref semantics:
+    daft-core decimal ops).  add/sub align scales; mul adds scales;
+    div (and anything overflowing p=18) promotes to float64.  Returns
+    None to fall through to the float path."""
+    def to_float(s: Series) -> Series:
+        return cast(s, DataType.float64()) if s.dtype.is_decimal() else s
+
+    if not (_dec_phys_int(l.dtype) or _dec_phys_int(r.dtype)) or \
+            l.dtype.is_floating() or r.dtype.is_floating() or \
+            op in ("div", "pow"):
+        ll, rr = to_float(l), to_float(r)
+        if ll is l and rr is r:
+            return None
+        return binary_op(ll, rr, op)
+
+    def scaled(s: Series):
+        if s.dtype.is_decimal():
+            return s.data.to(torch.int64), s.dtype.scale, s.dtype.precision
+        if len(s) == 1:  # literal: use its true digit count
+            v = abs(int(s.data.item()))
+            return s.data.to(torch.int64), 0, max(1, len(str(v)))
+        return s.data.to(torch.int64), 0, 19
+
+    av, asc, ap = scaled(l)
+    bv, bsc, bp = scaled(r)
+    if op in ("add", "sub"):
+        sc = max(asc, bsc)
+        if asc < sc:
+            av = av * (10 ** (sc - asc))
+        if bsc < sc:
+            bv = bv * (10 ** (sc - bsc))
+        out = av + bv if op == "add" else av - bv
+        p = min(18, max(ap - asc, bp - bsc) + sc + 1)
+        if sc > 18:
+            return binary_op(to_float(l), to_float(r), op)
+        return Series(l.name, DataType.decimal128(p, sc), data=out,
+                      validity=_expand_validity(validity,
+                                                int(out.shape[0])))
+    if op == "mul":
+        sc = asc + bsc
+        p = min(18, ap + bp)
+        if sc > 18 or (ap - asc) + (bp - bsc) + sc > 18:
+            return binary_op(to_float(l), to_float(r), op)
+        out = av * bv
+        return Series(l.name, DataType.decimal128(p, sc), data=out,
+                      validity=_expand_validity(validity,
+                                                int(out.shape[0])))
+    if op in ("floordiv", "mod"):
+        return binary_op(to_float(l), to_float(r), op)
+    return None
 
 
 def _temporal_binary(l: Series, r: Series, op: str, validity):
@@ -497,6 +598,13 @@ def compare_op(l: Series, r: Series, op: str) -> Series:
     if lk in (TypeKind.STRING, TypeKind.BINARY) or \
             r.dtype.kind in (TypeKind.STRING, TypeKind.BINARY):
         return _string_compare(l, r, op, validity)
+    if l.dtype.is_decimal() or r.dtype.is_decimal():
+        # align scales exactly (scaled-int compare), or promote to f64
+        st = supertype(l.dtype, r.dtype)
+        if l.dtype != st:
+            l = cast(l, st)
+        if r.dtype != st:
+            r = cast(r, st)
     st = supertype(l.dtype, r.dtype)
     tdt = st.to_torch()
 

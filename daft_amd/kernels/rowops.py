@@ -58,6 +58,12 @@ def _fnv1a_bytes(b: bytes) -> int:
 _NULL_HASH = np.uint64(0x9E3779B97F4A7C15)
 
 
+
+def _phys_float(dt: DataType) -> bool:
+    """True when the physical storage is floating point (decimals with
+    p <= 18 store scaled int64 and take the exact integer paths)."""
+    return dt.to_physical().kind in (TypeKind.FLOAT32, TypeKind.FLOAT64)
+
 def _cpu_hash_columns(series: Sequence[Series], seed: int) -> torch.Tensor:
     n = len(series[0])
     acc = np.full(n, np.uint64(seed) + np.uint64(0x8445D61A4E774912),
@@ -123,7 +129,7 @@ def _cpu_group_codes(s: Series):
             codes = np.where(valid, codes, card)
             card += 1
         return codes, card
-    if s.dtype.is_floating() or s.dtype.is_decimal():
+    if _phys_float(s.dtype):
         d = s.data.to(torch.float64)
         d = torch.where(d == 0.0, torch.zeros_like(d), d)
         d = torch.where(torch.isnan(d), torch.full_like(d, float("nan")), d)
@@ -188,7 +194,7 @@ def _cpu_key_rows(series: Sequence[Series]) -> list:
     cols = []
     for s in series:
         vals = s.to_pylist()
-        if s.dtype.is_floating() or s.dtype.is_decimal():
+        if _phys_float(s.dtype):
             vals = [None if v is None
                     else (0.0 if v == 0.0 else ("nan" if v != v else v))
                     for v in vals]
@@ -296,7 +302,7 @@ def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
     data = values.data
     validity = values.validity
 
-    if values.dtype.is_floating() or values.dtype.is_decimal():
+    if _phys_float(values.dtype):
         wdt = torch.float64
     elif values.dtype.kind == TypeKind.FLOAT32:
         wdt = torch.float64
@@ -517,7 +523,7 @@ def _order_key_u64(s: Series) -> torch.Tensor:
     k = s.dtype.kind
     if k == TypeKind.BOOL:
         return d.to(torch.int64)
-    if s.dtype.is_floating() or s.dtype.is_decimal():
+    if _phys_float(s.dtype):
         bits = d.to(torch.float64).view(torch.int64)
         neg = bits < 0
         flipped = torch.where(neg, ~bits, bits ^ _SIGN64)

@@ -68,6 +68,12 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
     values = agg.child.evaluate(batch)
     if len(values) == 1 and n > 1:
         values = values.broadcast(n)
+    if values.dtype.is_decimal() and values.data is not None and \
+            values.data.dtype == torch.int64 and kind in (
+            AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE, AggKind.SKEW,
+            AggKind.APPROX_PERCENTILE, AggKind.SKETCH):
+        # moment-style aggs run in f64; SUM/MIN/MAX stay exact scaled-int
+        values = values.cast(DataType.float64())
 
     if kind == AggKind.COUNT:
         data, _ = rowops.grouped_agg(group_ids, num_groups, values,

@@ -198,9 +198,13 @@ class DataType:
         if k in (TypeKind.TIME, TypeKind.TIMESTAMP, TypeKind.DURATION):
             return DataType.int64()
         if k == TypeKind.DECIMAL128:
-            # round-1 storage: float64 (documented deviation; exact i128 in a
-            # later round).  TPC-H monetary aggregates remain within f64 ulp
-            # tolerance at the checked scales.
+            # exact storage: scaled int64 for precision <= 18 (value *
+            # 10^scale as a 64-bit integer — sums/compares/joins are exact;
+            # ref semantics: daft-core Decimal128Array).  Wider decimals
+            # (p > 18) fall back to float64 (documented deviation; full
+            # i128 lands with a dedicated HIP type).
+            if self.precision <= 18:
+                return DataType.int64()
             return DataType.float64()
         if k == TypeKind.EMBEDDING:
             return DataType.fixed_size_list(self.inner, self.size)
@@ -327,7 +331,17 @@ def supertype(a: DataType, b: DataType) -> DataType:
     if b.is_null():
         return a
     if a.is_decimal() or b.is_decimal():
-        # round-1 decimal storage is f64
+        if a.is_decimal() and b.is_decimal():
+            sc = max(a.scale, b.scale)
+            ip = max(a.precision - a.scale, b.precision - b.scale)
+            p = ip + sc + 1
+            if p <= 18:
+                return DataType.decimal128(p, sc)
+            return DataType.float64()
+        dec, other = (a, b) if a.is_decimal() else (b, a)
+        if other.is_integer() and dec.precision <= 18:
+            p = min(18, max(dec.precision, 19 + dec.scale))
+            return DataType.decimal128(p, dec.scale)
         return DataType.float64()
     if a.is_temporal() or b.is_temporal():
         if a.kind == b.kind:

@@ -292,6 +292,10 @@ class Series:
                     v = v * us
                 out.append(_dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=v))
             return wrap(out)
+        if k == TypeKind.DECIMAL128 and s.data.dtype == torch.int64:
+            import decimal as _dec
+            return wrap([_dec.Decimal(int(v)).scaleb(-self.dtype.scale)
+                         for v in s.data.numpy()])
         if k == TypeKind.BOOL:
             return wrap([bool(v) for v in s.data.numpy()])
         vals = s.data
@@ -498,6 +502,11 @@ def _infer_dtype(values: Sequence[Any]) -> DataType:
             return DataType.int64()
         if isinstance(v, float):
             return DataType.float64()
+        if type(v).__name__ == "Decimal":
+            t = v.as_tuple()
+            scl = max(0, -t.exponent) if isinstance(t.exponent, int) else 0
+            digs = max(len(t.digits), scl + 1)
+            return DataType.decimal128(min(digs, 18), scl)
         if isinstance(v, str):
             return DataType.string()
         if isinstance(v, bytes):
@@ -617,6 +626,15 @@ def _from_pylist_typed(name: str, values: list, dtype: DataType) -> Series:
         return Series(name, dtype, children=children, validity=validity,
                       length=n)
     if k == TypeKind.DECIMAL128:
+        import decimal as _dec
+        if dtype.to_physical().kind == TypeKind.INT64:
+            scl = _dec.Decimal(1).scaleb(dtype.scale)
+            ints = [0 if v is None else
+                    int((_dec.Decimal(str(v)) * scl).to_integral_value(
+                        rounding=_dec.ROUND_HALF_EVEN)) for v in values]
+            return Series(name, dtype,
+                          data=torch.tensor(ints, dtype=torch.int64),
+                          validity=validity)
         vals = [0.0 if v is None else float(v) for v in values]
         return Series(name, dtype,
                       data=torch.tensor(vals, dtype=torch.float64),

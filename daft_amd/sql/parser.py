@@ -575,8 +575,12 @@ class Parser:
                 while self.peek().kind == "name":
                     tn += " " + self.next().value.lower()
                 if self.eat_op("("):
+                    args = []
                     while not self.eat_op(")"):
-                        self.next()
+                        tok = self.next()
+                        if tok.value != ",":
+                            args.append(str(tok.value))
+                    tn += "(" + ",".join(args) + ")"
                 self.expect_op(")")
                 return CastExpr(child, tn)
             if t.value == "exists":

@@ -169,6 +169,15 @@ def expr_to_daft(e, binder: Optional[Binder]) -> Expression:
              "numeric": DataType.float64(), "varchar": DataType.string(),
              "text": DataType.string(), "char": DataType.string(),
              "date": DataType.date(), "boolean": DataType.bool()}
+        import re as _re
+        dm = _re.fullmatch(r"(?:decimal|numeric)\s*\((\d+)\s*,\s*(\d+)\)",
+                           tn)
+        if dm:
+            return c.cast(DataType.decimal128(int(dm.group(1)),
+                                              int(dm.group(2))))
+        base = tn.split("(")[0].strip()
+        if tn not in m and base in m:
+            tn = base  # length-parameterized types: varchar(n), char(n)...
         if tn not in m:
             raise SQLPlanError(f"unsupported cast type {tn}")
         return c.cast(m[tn])

@@ -173,3 +173,47 @@ def test_argsort_desc():
     s = Series.from_pylist("a", [3, None, 1, 2])
     perm = s.argsort(descending=True, nulls_first=True)
     assert s.take(perm).to_pylist() == [None, 3, 2, 1]
+
+
+def test_decimal_exact_storage_and_ops(tmp_path):
+    """Decimal128 (p<=18) stores scaled int64: sums/joins/sorts are exact
+    (ref: daft-core Decimal128Array semantics)."""
+    from decimal import Decimal as D
+    import daft_amd as daft
+    from daft_amd import col
+
+    df = daft.from_pydict({"v": [D("0.10")] * 100 + [D("0.05")]})
+    tot = df.agg(col("v").sum().alias("s")).to_pydict()["s"][0]
+    assert tot == D("10.05")  # float64 would drift
+
+    # arithmetic scale rules
+    d2 = daft.from_pydict({"a": [D("1.25")], "b": [D("0.4")]})
+    out = d2.select((col("a") + col("b")).alias("s"),
+                    (col("a") * col("b")).alias("m"),
+                    (col("a") - D("0.05")).alias("d")).to_pydict()
+    assert out["s"][0] == D("1.65")
+    assert out["m"][0] == D("0.500")
+    assert out["d"][0] == D("1.20")
+
+    # join + groupby on decimal keys are exact
+    l = daft.from_pydict({"k": [D("2.50"), D("1.10"), D("2.50")]})
+    r = daft.from_pydict({"k": [D("1.10"), D("2.50")], "w": [1, 2]})
+    j = l.join(r, on="k").groupby("k").agg(col("w").sum().alias("t")) \
+        .sort("k").to_pydict()
+    assert j["t"] == [1, 4]
+
+    # parquet round trip preserves exact values
+    import pyarrow.parquet as pq
+    p = str(tmp_path / "d.parquet")
+    daft.from_pydict({"v": [D("123.45"), None, D("-0.01")]}) \
+        .write_parquet(p)
+    import glob
+    files = glob.glob(p + "/*.parquet") if not p.endswith(".parquet") or \
+        __import__("os").path.isdir(p) else [p]
+    back = daft.read_parquet(files if files else p).to_pydict()["v"]
+    assert back == [D("123.45"), None, D("-0.01")]
+
+    # cast decimal -> string keeps trailing zeros per scale
+    s = daft.from_pydict({"v": [D("1.10")]}).select(
+        col("v").cast(daft.DataType.string()).alias("s")).to_pydict()["s"]
+    assert s == ["1.10"]
