@@ -112,3 +112,53 @@ def file_size(expr) -> "Expression":
         return Series.from_pylist(s.name, out, DataType.int64())
     return Expression(ScalarFn("file_size", run, [_to_node(expr)],
                                DataType.int64()))
+
+# free-function API surface (ref: daft/functions/__init__.py exports)
+from .math import (  # noqa: E402,F401
+    sin, cos, tan, cot, sec, csc, sinh, cosh, tanh, arcsin, arccos,
+    arctan, arcsinh, arccosh, arctanh, arctan2, degrees, radians, exp,
+    expm1, ln, log, log2, log10, log1p, sqrt, cbrt, hypot, sign, signum,
+    negate, negative, pmod, power, factorial, e, pi, trunc, bitwise_and,
+    bitwise_or, bitwise_xor, shift_left, shift_right, try_divide,
+    random_int)
+from .math import pow  # noqa: E402,F401,A004
+from .temporal import (  # noqa: E402,F401
+    year, month, day, dayofmonth, day_of_month, dayofyear, day_of_year,
+    weekofyear, week_of_year, quarter, hour, minute, second, day_of_week,
+    dayofweek, microsecond, millisecond, nanosecond, date_trunc, to_date,
+    date_add, dateadd, date_sub, date_diff, datediff, datepart,
+    make_date, last_day, next_day, add_months, months_between, strftime,
+    date_format, from_unixtime, timestamp_seconds, timestamp_millis,
+    timestamp_micros, to_unix_epoch, unix_date, date_from_unix_date,
+    current_date, current_timestamp, current_timezone, total_hours,
+    total_minutes, total_milliseconds, total_microseconds,
+    total_nanoseconds, to_datetime)
+from .strings_extra import (  # noqa: E402,F401
+    to_snake_case, to_upper_snake_case, to_kebab_case,
+    to_upper_kebab_case, to_camel_case, to_upper_camel_case,
+    to_title_case, normalize, concat_ws, format, ascii_func, chr_func,
+    space, translate, replace, split_part, substring_index,
+    count_matches, regexp, regexp_count, regexp_extract,
+    regexp_extract_all, regexp_replace, regexp_split,
+    levenshtein_distance, damerau_levenshtein_distance,
+    hamming_distance_str, jaro_similarity, jaro_winkler_similarity,
+    jaccard_similarity, soundex)
+from .misc import (  # noqa: E402,F401
+    list_contains, list_distinct, list_join, list_sum, list_min,
+    list_max, list_mean, list_count, list_chunk, list_slice, list_agg,
+    to_list, list_agg_distinct, list_sort, list_append, list_flatten,
+    list_bool_and, list_bool_or, list_map, list_filter, to_struct,
+    struct, eq_null_safe, not_nan, try_cast, when, serialize,
+    deserialize, try_deserialize, compress, decompress, try_compress,
+    try_decompress, encode, decode, try_encode, try_decode,
+    json_array_length, json_object_keys, json_tuple, parse_url,
+    guess_mime_type, file_exists, file_path, median, percentile,
+    approx_percentiles, var, product, string_agg, columns_mean,
+    columns_sum, pearson_correlation, audio_file, audio_metadata,
+    video_file, video_frames, video_keyframes, video_metadata,
+    hdf5_file, hdf5_keys, hdf5_attrs, hdf5_metadata, run_process,
+    llm_generate)
+from .image import (  # noqa: E402,F401
+    image_height, image_width, image_channel, image_mode,
+    image_attribute, image_hash, convert_image, decode_image,
+    encode_image, image_to_tensor)

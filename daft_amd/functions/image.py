@@ -272,3 +272,90 @@ class ImageNamespace(_Namespace):
             dt = fields[0].dtype
             return DataType.fixed_shape_image(dt.image_mode or "RGB", h, w)
         return self._fn("image_crop", crop_series, ret, x, y, w, h)
+
+
+def _image_attr(name, child_idx, view_dt, out_dt):
+    def make(expr):
+        from ..expressions.expressions import Expression, ScalarFn, _to_node
+
+        def run(s: Series) -> Series:
+            assert s.dtype.kind == TypeKind.IMAGE, \
+                f"{name} expects Image, got {s.dtype!r}"
+            c = s.children[child_idx]
+            data = c.data.view(view_dt).to(torch.int64) if view_dt else \
+                c.data.to(torch.int64)
+            return Series(s.name, out_dt, data=data.to(torch.int32),
+                          validity=s.validity)
+        return Expression(ScalarFn(name, run, [_to_node(expr)], out_dt))
+    make.__name__ = name
+    return make
+
+
+image_height = _image_attr("image_height", 2, torch.int32,
+                           DataType.int32())
+image_width = _image_attr("image_width", 3, torch.int32, DataType.int32())
+image_channel = _image_attr("image_channel", 1, torch.int16,
+                            DataType.int32())
+
+
+def image_mode(expr):
+    from ..expressions.expressions import Expression, ScalarFn, _to_node
+
+    def run(s: Series) -> Series:
+        codes = s.children[4].data.to(torch.int64).cpu().tolist()
+        names = {1: "L", 2: "LA", 3: "RGB", 4: "RGBA", 0: None}
+        out = [names.get(c) for c in codes]
+        r = Series.from_pylist(s.name, out, DataType.string())
+        return r.to(s.device) if s.is_gpu() else r
+    return Expression(ScalarFn("image_mode", run, [_to_node(expr)],
+                               DataType.string()))
+
+
+def image_attribute(expr, attr: str):
+    return {"height": image_height, "width": image_width,
+            "channel": image_channel, "mode": image_mode}[attr](expr)
+
+
+def image_hash(expr, algorithm: str = "average", hash_size: int = 8):
+    """Perceptual image hash (average-hash): resize to hash_size^2 luma,
+    threshold at the mean -> bit string (capability of daft's
+    image hash fns)."""
+    from ..expressions.expressions import Expression, ScalarFn, _to_node
+
+    def run(s: Series) -> Series:
+        small = resize_series(s, hash_size, hash_size)
+        lum = to_mode_series(small, "L")
+        n = len(s)
+        px = lum.children[0].data.reshape(n, hash_size * hash_size) \
+            .to(torch.float32)
+        mean = px.mean(dim=1, keepdim=True)
+        bits = (px > mean).cpu().numpy()
+        out = ["".join("1" if b else "0" for b in row) for row in bits]
+        r = Series.from_pylist(s.name, out, DataType.string())
+        return r.to(s.device) if s.is_gpu() else r
+    return Expression(ScalarFn("image_hash", run, [_to_node(expr)],
+                               DataType.string()))
+
+
+def convert_image(expr, mode: str):
+    from ..expressions.expressions import Expression as E, _to_node
+    e = expr if isinstance(expr, E) else E(_to_node(expr))
+    return e.image.to_mode(mode)
+
+
+def decode_image(expr, mode: str = "RGB", on_error: str = "raise"):
+    from ..expressions.expressions import Expression as E, _to_node
+    e = expr if isinstance(expr, E) else E(_to_node(expr))
+    return e.image.decode(mode, on_error)
+
+
+def encode_image(expr, image_format: str = "PNG"):
+    from ..expressions.expressions import Expression as E, _to_node
+    e = expr if isinstance(expr, E) else E(_to_node(expr))
+    return e.image.encode(image_format)
+
+
+def image_to_tensor(expr):
+    from ..expressions.expressions import Expression as E, _to_node
+    e = expr if isinstance(expr, E) else E(_to_node(expr))
+    return e.image.to_tensor()
