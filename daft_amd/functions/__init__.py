@@ -162,3 +162,23 @@ from .image import (  # noqa: E402,F401
     image_height, image_width, image_channel, image_mode,
     image_attribute, image_hash, convert_image, decode_image,
     encode_image, image_to_tensor)
+
+from .aliases import (  # noqa: E402,F401
+    lower, upper, capitalize, strip, lstrip, rstrip, reverse, length,
+    length_bytes, contains, startswith, endswith, like, ilike, find,
+    split, substr, left, right, lpad, rpad, repeat, concat,
+    tokenize_encode, tokenize_decode, is_nan, is_inf, fill_nan, get,
+    chunk, value_counts, explode, date, total_days, total_seconds, time,
+    download, upload, resize, crop, image_file, image_file_metadata,
+    decode_image_file, dot_product, euclidean_distance,
+    cosine_similarity, hamming_distance, map_get, map_keys,
+    to_utc_timestamp, from_utc_timestamp, convert_time_zone,
+    convert_timezone, replace_time_zone, make_timestamp,
+    make_timestamp_ltz, partition_days, partition_months,
+    partition_years, partition_hours, partition_iceberg_bucket,
+    partition_iceberg_truncate, seq, bin, conv, unnest, first_value,
+    last_value, jq, extract_month_uuid7, extract_day_uuid7,
+    extract_hour_uuid7, extract_minute_uuid7, resample)
+from .aliases import slice  # noqa: E402,F401,A004
+from .ai import (  # noqa: E402,F401
+    embed_text, embed_image, classify_text, classify_image, prompt)

