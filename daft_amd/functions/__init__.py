@@ -28,6 +28,21 @@ def dense_rank() -> Expression:
     return Expression(WindowFn("dense_rank", None, None))
 
 
+def w_first_value(expr) -> Expression:
+    """first_value window function (ref: daft/functions/window.py:310)."""
+    from ..physical.window import WindowFn
+    from ..expressions.expressions import _to_node
+    return Expression(WindowFn("first_value", _to_node(expr), None))
+
+
+def w_last_value(expr) -> Expression:
+    """last_value window function (SQL default frame: running last =
+    current row; ref: daft/functions/window.py:371)."""
+    from ..physical.window import WindowFn
+    from ..expressions.expressions import _to_node
+    return Expression(WindowFn("last_value", _to_node(expr), None))
+
+
 def monotonically_increasing_id() -> Expression:
     raise NotImplementedError(
         "use DataFrame.add_monotonically_increasing_id()")

@@ -61,6 +61,8 @@ def window_out_field(e: ExprNode, name: str, schema: Schema) -> Field:
             return Field(name, f.dtype)
         if e.kind in ("row_number", "rank", "dense_rank"):
             return Field(name, DataType.uint64())
+        if e.kind in ("first_value", "last_value"):
+            return Field(name, e.inner.to_field(schema).dtype)
         if e.kind in ("lag", "lead"):
             f = e.inner.to_field(schema)
             return Field(name, f.dtype)
@@ -139,6 +141,18 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
                                     dense=e.kind == "dense_rank")
             out_cols.append(Series(nm, DataType.uint64(),
                                    data=vals.view(torch.uint64)))
+        elif e.kind in ("first_value", "last_value"):
+            assert sorted_pos is not None, f"{e.kind} requires order_by"
+            perm, pos, part_start = sorted_pos
+            vals = e.inner.evaluate(batch)
+            if e.kind == "first_value":
+                # row at each partition's first sorted position
+                src = perm[part_start[gids]]
+            else:
+                # running last value = the current row itself under the
+                # SQL default frame (unbounded preceding .. current row)
+                src = torch.arange(n, dtype=torch.int64, device=dev)
+            out_cols.append(vals.take(src, has_neg=False).rename(nm))
         elif e.kind in ("lag", "lead"):
             assert sorted_pos is not None, f"{e.kind} requires order_by"
             perm, pos, part_start = sorted_pos

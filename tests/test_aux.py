@@ -635,3 +635,16 @@ def test_compressed_csv_json(tmp_path):
     with gzip.open(pj, "wt") as f:
         f.write('{"a": 1}\n{"a": 5}\n')
     assert daft.read_json(pj).to_pydict() == {"a": [1, 5]}
+
+
+def test_window_first_last_value():
+    from daft_amd.functions import w_first_value, w_last_value
+    from daft_amd.window import Window
+    df = daft.from_pydict({"g": ["a", "a", "a", "b", "b"],
+                           "v": [3, 1, 2, 9, 8]})
+    w = Window().partition_by("g").order_by("v")
+    out = df.with_window_columns(
+        {"f": w_first_value(col("v")).over(w),
+         "l": w_last_value(col("v")).over(w)}).sort(["g", "v"]).to_pydict()
+    assert out["f"] == [1, 1, 1, 8, 8]
+    assert out["l"] == [1, 2, 3, 8, 9]

@@ -443,3 +443,22 @@ def test_out_of_core_host_streaming():
         assert out_host["s"] == pytest.approx(out_dev["s"], rel=1e-12)
     finally:
         cfg.stream_morsel_rows = old
+
+
+def test_decimal_exact_gpu():
+    from decimal import Decimal as D
+    vals = [D("0.10")] * 1000 + [D("0.05"), None]
+    g = daft.from_pydict({"v": vals}, device="cuda:0")
+    tot = g.agg(col("v").sum().alias("s")).to_pydict()["s"][0]
+    assert tot == D("100.05")
+    # groupby + join + sort on decimal keys (int64 W8 row-op path)
+    l = daft.from_pydict({"k": [D("2.50"), D("1.10"), D("2.50")] * 100},
+                         device="cuda:0")
+    r = daft.from_pydict({"k": [D("1.10"), D("2.50")], "w": [1, 2]},
+                         device="cuda:0")
+    j = l.join(r, on="k").groupby("k").agg(col("w").sum().alias("t")) \
+        .sort("k").to_pydict()
+    assert j["t"] == [100, 400]
+    srt = daft.from_pydict({"v": [D("3.3"), D("1.1"), D("2.2")]},
+                           device="cuda:0").sort("v").to_pydict()["v"]
+    assert srt == [D("1.1"), D("2.2"), D("3.3")]
