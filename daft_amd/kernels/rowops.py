@@ -368,9 +368,81 @@ def join(left_keys: Sequence[Series], right_keys: Sequence[Series],
         lk2.append(l.cast(st) if l.dtype != st else l)
         rk2.append(r.cast(st) if r.dtype != st else r)
     left_keys, right_keys = lk2, rk2
+    dense = _dense_key_join(left_keys, right_keys, how)
+    if dense is not None:
+        return dense
     if _is_gpu(left_keys[0]):
         return _gpu_join(left_keys, right_keys, how)
     return _cpu_join(left_keys, right_keys, how)
+
+
+_DENSE_JOIN_LIMIT = 1 << 31  # direct table cap: 2G slots (8 GB int32)
+
+
+def _dense_key_join(lk, rk, how):
+    """Direct-address join: when the build side is a single no-null
+    integer key whose values are unique and densely ranged (the PK-join
+    shape — orderkey/partkey/suppkey in TPC-H), replace the hash table
+    with an addressed row array.  Probe is then one gather + a compare:
+    no hashing, no chain walks, no count/fill passes.  Returns None to
+    fall back to the bucket-chain hash join."""
+    if len(lk) != 1 or len(rk) != 1:
+        return None
+    l, r = lk[0], rk[0]
+    for s in (l, r):
+        if s.is_dict() or s.data is None or not (
+                s.dtype.is_integer() or
+                (s.dtype.is_decimal() and s.data.dtype == torch.int64) or
+                s.dtype.kind in (TypeKind.DATE,)):
+            return None
+        if s.data.dtype == torch.uint64:
+            return None
+    if r.validity is not None and bool((~r.validity).any()):
+        return None
+    n_r = len(r)
+    n_l = len(l)
+    if n_r == 0 or n_l == 0:
+        return None
+    dev = r.device
+    rdata = r.data.to(torch.int64)
+    mn = int(rdata.min().item())
+    mx = int(rdata.max().item())
+    rng = mx - mn + 1
+    if rng <= 0 or rng > max(4 * n_r, 1 << 20) or rng > _DENSE_JOIN_LIMIT:
+        return None
+    idx_t = torch.int32 if n_r < (1 << 31) else torch.int64
+    table = torch.full((rng,), -1, dtype=idx_t, device=dev)
+    rows = torch.arange(n_r, dtype=idx_t, device=dev)
+    table[rdata - mn] = rows
+    if not bool((table[rdata - mn] == rows).all()):
+        return None  # duplicate build keys: hash join handles fan-out
+    ldata = l.data.to(torch.int64)
+    in_rng = (ldata >= mn) & (ldata <= mx)
+    if l.validity is not None:
+        in_rng &= l.validity
+    row = table[(ldata - mn).clamp(min=0, max=rng - 1)].to(torch.int64)
+    hit = in_rng & (row >= 0)
+    ar = torch.arange(n_l, dtype=torch.int64, device=dev)
+    if how == "semi":
+        return ar[hit], torch.zeros(0, dtype=torch.int64, device=dev)
+    if how == "anti":
+        return ar[~hit], torch.zeros(0, dtype=torch.int64, device=dev)
+    if how == "inner":
+        lidx = ar[hit]
+        return lidx, row[lidx]
+    if how == "left":
+        return ar, torch.where(hit, row, torch.full_like(row, -1))
+    # right / outer: matched build rows + appended unmatched
+    lidx = ar[hit] if how == "right" else ar
+    ridx = row[lidx] if how == "right" else \
+        torch.where(hit, row, torch.full_like(row, -1))
+    matched = torch.zeros(n_r, dtype=torch.bool, device=dev)
+    matched[row[hit]] = True
+    unmatched = torch.nonzero(~matched).reshape(-1)
+    if unmatched.numel():
+        lidx = torch.cat([lidx, torch.full_like(unmatched, -1)])
+        ridx = torch.cat([ridx, unmatched])
+    return lidx, ridx
 
 
 def _align_dict_keys(l: Series, r: Series):

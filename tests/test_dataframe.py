@@ -347,3 +347,34 @@ def test_streamed_aggregate_and_join_multibatch(tmp_path):
     ss = single.join(dim.where(col("k") < 5), on="k", how="semi") \
         .sort("v").to_pydict()
     assert sm == ss
+
+
+def test_dense_key_join_matches_hash_join():
+    """Direct-address PK-join fast path vs the hash join (rowops)."""
+    import numpy as np
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+    from daft_amd.kernels import rowops
+    rng = np.random.default_rng(11)
+    for how in ("inner", "left", "right", "outer", "semi", "anti"):
+        rk = [int(v) for v in rng.permutation(np.arange(100, 300))[:150]]
+        lvals = [None if rng.random() < 0.1 else int(v)
+                 for v in rng.integers(50, 350, 400)]
+        lk = [Series.from_pylist("k", lvals, DataType.int64())]
+        rkS = [Series.from_pylist("k", rk, DataType.int64())]
+        d = rowops._dense_key_join(lk, rkS, how)
+        assert d is not None, how
+        li1, ri1 = d
+        li2, ri2 = rowops._cpu_join(lk, rkS, how)
+        if how in ("semi", "anti"):
+            assert sorted(li1.tolist()) == sorted(li2.tolist()), how
+        else:
+            assert sorted(zip(li1.tolist(), ri1.tolist())) == \
+                sorted(zip(li2.tolist(), ri2.tolist())), how
+    # duplicate build keys and sparse ranges fall back to the hash join
+    assert rowops._dense_key_join(
+        lk, [Series.from_pylist("k", [1, 2, 2], DataType.int64())],
+        "inner") is None
+    assert rowops._dense_key_join(
+        lk, [Series.from_pylist("k", [1, 10**9], DataType.int64())],
+        "inner") is None
