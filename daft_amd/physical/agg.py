@@ -51,8 +51,11 @@ def decompose_agg_exprs(exprs: Sequence[ExprNode]
 
 
 def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
-                num_groups: int, name: str, agg: Agg) -> Series:
-    """Compute one aggregation over groups; group_ids None = single group."""
+                num_groups: int, name: str, agg: Agg,
+                mask: Optional[torch.Tensor] = None) -> Series:
+    """Compute one aggregation over groups; group_ids None = single group.
+    `mask` (bool[n]) restricts the aggregation to selected rows — the
+    filter-into-aggregate fusion path (no compaction materialized)."""
     n = len(batch)
     dev = batch.device
     if group_ids is None:
@@ -61,13 +64,22 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
 
     if kind == AggKind.COUNT_ALL or (kind == AggKind.COUNT and
                                      agg.child is None):
-        data, _ = rowops.grouped_agg(group_ids, num_groups,
-                                     _ones_series(n, dev), "count")
+        ones = _ones_series(n, dev)
+        if mask is not None:
+            ones = ones.with_validity(mask)
+            data, _ = rowops.grouped_agg(group_ids, num_groups, ones,
+                                         "count_valid")
+        else:
+            data, _ = rowops.grouped_agg(group_ids, num_groups, ones,
+                                         "count")
         return Series(name, DataType.uint64(), data=data.view(torch.uint64))
 
     values = agg.child.evaluate(batch)
     if len(values) == 1 and n > 1:
         values = values.broadcast(n)
+    if mask is not None:
+        v = mask if values.validity is None else (values.validity & mask)
+        values = values.with_validity(v)
     if values.dtype.is_decimal() and values.data is not None and \
             values.data.dtype == torch.int64 and kind in (
             AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE, AggKind.SKEW,
@@ -298,8 +310,11 @@ def _string_minmax(group_ids, num_groups, values: Series, kind, name):
 
 
 def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
-                  aggs: List[ExprNode]) -> RecordBatch:
-    """One-shot (grouped or global) aggregation of a materialized batch."""
+                  aggs: List[ExprNode],
+                  mask: Optional[torch.Tensor] = None) -> RecordBatch:
+    """One-shot (grouped or global) aggregation of a materialized batch.
+    With `mask`, rows where mask is False are excluded (fused filter:
+    the predicate never materializes a compacted copy of the input)."""
     named_aggs, residuals = decompose_agg_exprs(aggs)
     n = len(batch)
     dev = batch.device
@@ -315,11 +330,19 @@ def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
         num_groups = 1
         key_cols = []
 
-    agg_cols = [compute_agg(batch, gids, num_groups, cname, a)
+    agg_cols = [compute_agg(batch, gids, num_groups, cname, a, mask=mask)
                 for cname, a in named_aggs]
     inter = RecordBatch(key_cols + agg_cols,
                         num_rows=num_groups)
     out_cols = list(key_cols)
     for r in residuals:
         out_cols.append(r.evaluate(inter))
-    return RecordBatch(out_cols, num_rows=num_groups)
+    out = RecordBatch(out_cols, num_rows=num_groups)
+    if mask is not None and groupby:
+        # drop groups whose rows were all masked out
+        hit = torch.zeros(num_groups, dtype=torch.int64, device=dev)
+        hit.scatter_add_(0, gids, mask.to(torch.int64))
+        keep = torch.nonzero(hit > 0).reshape(-1)
+        if int(keep.numel()) != num_groups:
+            out = out.take(keep, has_neg=False)
+    return out

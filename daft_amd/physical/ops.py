@@ -406,15 +406,61 @@ class AggregateOp(PhysicalOp):
         self.groupby = groupby
         self.aggs = aggs
 
+    _FUSABLE_AGGS = {
+        AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL, AggKind.MIN,
+        AggKind.MAX, AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE,
+        AggKind.SKEW, AggKind.ANY_VALUE, AggKind.BOOL_AND,
+        AggKind.BOOL_OR, AggKind.COUNT_DISTINCT,
+        AggKind.APPROX_COUNT_DISTINCT, AggKind.SKETCH,
+    }
+
+    def _fusion_pred(self):
+        """Filter-into-aggregate fusion: when the child is a Filter and
+        every aggregation tolerates a row mask, evaluate the predicate
+        as a mask instead of materializing the compacted input (high-
+        selectivity scans stop paying a full-table gather)."""
+        child = self.children[0]
+        if not isinstance(child, FilterOp):
+            return None, child
+        named, _ = agg_mod.decompose_agg_exprs(self.aggs)
+        if not named or any(a.kind not in self._FUSABLE_AGGS
+                            for _, a in named):
+            return None, child
+        return child.predicate, child.children[0]
+
+    @staticmethod
+    def _prep(b, pred):
+        """(batch, mask) for one input batch; low selectivity compacts
+        (cheaper to gather few rows than to stream-mask every column)."""
+        if pred is None or len(b) == 0:
+            return b, None
+        m = pred.evaluate(b)
+        mask = m.data
+        if len(m) == 1 and len(b) != 1:
+            mask = mask.expand(len(b))
+        if m.validity is not None:
+            v = m.validity
+            if v.numel() == 1 and len(b) != 1:
+                v = v.expand(len(b))
+            mask = mask & v
+        nsel = int(mask.sum().item())
+        if nsel * 2 < len(b):
+            idx = torch.nonzero(mask).reshape(-1)
+            return b.take(idx, has_neg=False), None
+        return b, mask.contiguous()
+
     def execute(self, ectx) -> BatchIter:
         from itertools import chain as _chain
-        it = self.children[0].execute_tracked(ectx)
+        pred, child = self._fusion_pred()
+        it = child.execute_tracked(ectx)
         first = next(it, None)
         second = next(it, None) if first is not None else None
         if second is None:
             batch = first if first is not None else RecordBatch.empty(
-                self.children[0].schema, device=ectx.device)
-            yield agg_mod.run_aggregate(batch, self.groupby, self.aggs)
+                child.schema, device=ectx.device)
+            batch, mask = self._prep(batch, pred)
+            yield agg_mod.run_aggregate(batch, self.groupby, self.aggs,
+                                        mask=mask)
             return
         # multiple input batches: fold each into per-group partial states
         # (same partial/final decomposition the distributed planner uses)
@@ -423,7 +469,13 @@ class AggregateOp(PhysicalOp):
         from .agg_partial import split_partial_final
         split = split_partial_final(self.aggs)
         if split is None:
-            batches = [first, second] + list(it)
+            batches = []
+            for b in _chain([first, second], it):
+                b, mask = self._prep(b, pred)
+                if mask is not None:
+                    idx = torch.nonzero(mask).reshape(-1)
+                    b = b.take(idx, has_neg=False)
+                batches.append(b)
             ectx.memory.admit(sum(b.size_bytes() for b in batches),
                               ectx.device)
             yield agg_mod.run_aggregate(RecordBatch.concat(batches),
@@ -431,10 +483,13 @@ class AggregateOp(PhysicalOp):
             return
         partials, final_named, residuals = split
         from ..expressions.expressions import ColumnRef
-        cschema = self.children[0].schema
+        cschema = child.schema
         gnames = [e.to_field(cschema).name for e in self.groupby]
-        parts = [agg_mod.run_aggregate(b, self.groupby, partials)
-                 for b in _chain([first, second], it)]
+        parts = []
+        for b in _chain([first, second], it):
+            b, mask = self._prep(b, pred)
+            parts.append(agg_mod.run_aggregate(b, self.groupby, partials,
+                                               mask=mask))
         merged = RecordBatch.concat(parts)
         final = agg_mod.run_aggregate(
             merged, [ColumnRef(n) for n in gnames], final_named)
