@@ -339,9 +339,14 @@ def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
         out_cols.append(r.evaluate(inter))
     out = RecordBatch(out_cols, num_rows=num_groups)
     if mask is not None and groupby:
-        # drop groups whose rows were all masked out
-        hit = torch.zeros(num_groups, dtype=torch.int64, device=dev)
-        hit.scatter_add_(0, gids, mask.to(torch.int64))
+        # drop groups whose rows were all masked out.  NOT a torch
+        # scatter_add: 600M atomic adds onto a handful of group slots
+        # serialize on L2 (measured ~100 ms on SF100 Q1); the LDS
+        # grouped-count kernel does this in one memory-bound pass.
+        ones = Series("__m", DataType.bool(),
+                      data=torch.ones(n, dtype=torch.bool, device=dev),
+                      validity=mask)
+        hit, _ = rowops.grouped_agg(gids, num_groups, ones, "count_valid")
         keep = torch.nonzero(hit > 0).reshape(-1)
         if int(keep.numel()) != num_groups:
             out = out.take(keep, has_neg=False)
