@@ -445,8 +445,9 @@ class AggregateOp(PhysicalOp):
             mask = mask & v
         nsel = int(mask.sum().item())
         if nsel * 2 < len(b):
-            idx = torch.nonzero(mask).reshape(-1)
-            return b.take(idx, has_neg=False), None
+            from ..kernels import compact_indices
+            ms = Series("__m", DataType.bool(), data=mask.contiguous())
+            return b.take(compact_indices(ms), has_neg=False), None
         return b, mask.contiguous()
 
     def execute(self, ectx) -> BatchIter:
