@@ -462,3 +462,34 @@ def test_decimal_exact_gpu():
     srt = daft.from_pydict({"v": [D("3.3"), D("1.1"), D("2.2")]},
                            device="cuda:0").sort("v").to_pydict()["v"]
     assert srt == [D("1.1"), D("2.2"), D("3.3")]
+
+
+def test_fused_filter_aggregate_gpu_matches_cpu():
+    """High-selectivity filters fuse into the aggregation as a row mask
+    on GPU; results must match the CPU (compacted) plan."""
+    torch.manual_seed(33)
+    n = 300_000
+    data = {
+        "g": [f"k{i % 9}" for i in range(n)],
+        "v": torch.rand(n, dtype=torch.float64),
+        "w": torch.randint(0, 100, (n,)),
+    }
+    q = lambda df: (df.where(col("w") >= 5)     # ~95% selectivity: fused
+                    .groupby("g")
+                    .agg(col("v").sum().alias("s"),
+                         col("v").count().alias("c"),
+                         col("v").min().alias("mn"),
+                         col("v").max().alias("mx"),
+                         col("w").mean().alias("mu"))
+                    .sort("g").to_pydict())
+    got = q(daft.from_pydict(data, device="cuda:0"))
+    want = q(daft.from_pydict(data, device="cpu"))
+    assert got["g"] == want["g"] and got["c"] == want["c"]
+    for k in ("s", "mn", "mx", "mu"):
+        assert got[k] == pytest.approx(want[k], rel=1e-12)
+    # low selectivity takes the compacted path: same results
+    q2 = lambda df: (df.where(col("w") < 5).groupby("g")
+                     .agg(col("v").sum().alias("s")).sort("g").to_pydict())
+    assert q2(daft.from_pydict(data, device="cuda:0"))["s"] == \
+        pytest.approx(q2(daft.from_pydict(data, device="cpu"))["s"],
+                      rel=1e-12)
