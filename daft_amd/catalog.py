@@ -127,8 +127,37 @@ class Session:
         return cat.get_table(".".join(ident.parts[1:]))
 
     def sql(self, query: str):
+        """Run SQL against the session's tables.  Beside SELECT, a small
+        DDL surface is supported (capability of the reference's
+        Session.sql): CREATE [OR REPLACE] [TEMP] TABLE name AS SELECT,
+        DROP TABLE name, SHOW TABLES."""
+        import re
         from .sql.planner import plan_sql
-        return plan_sql(query, lambda n: self.get_table(n))
+        q = query.strip().rstrip(";")
+        m = re.match(
+            r"(?is)^create\s+(or\s+replace\s+)?(temp(?:orary)?\s+)?table"
+            r"\s+([\w.\"]+)\s+as\s+(.*)$", q)
+        if m:
+            replace = m.group(1) is not None
+            name = m.group(3).strip('"')
+            df = plan_sql(m.group(4), lambda n: self.get_table(n))
+            if not replace and name in self._temp.list_tables():
+                raise ValueError(f"table {name!r} already exists "
+                                 f"(use CREATE OR REPLACE)")
+            self.create_temp_table(name, df.collect())
+            return df
+        m = re.match(r"(?is)^drop\s+table\s+(if\s+exists\s+)?([\w.\"]+)$",
+                     q)
+        if m:
+            name = m.group(2).strip('"')
+            if not m.group(1) and name not in self._temp.list_tables():
+                raise KeyError(name)
+            self._temp.drop_table(name)
+            return None
+        if re.match(r"(?is)^show\s+tables$", q):
+            from . import from_pydict
+            return from_pydict({"table": self.list_tables()})
+        return plan_sql(q, lambda n: self.get_table(n))
 
 
 _session: Optional[Session] = None

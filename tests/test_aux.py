@@ -648,3 +648,23 @@ def test_window_first_last_value():
          "l": w_last_value(col("v")).over(w)}).sort(["g", "v"]).to_pydict()
     assert out["f"] == [1, 1, 1, 8, 8]
     assert out["l"] == [1, 2, 3, 8, 9]
+
+
+def test_session_sql_ddl():
+    from daft_amd.catalog import Session
+    s = Session()
+    s.create_temp_table("t", daft.from_pydict({"a": [1, 2, 3]}))
+    s.sql("create temp table big as select a * 10 as b from t where a > 1")
+    out = s.sql("select sum(b) as s from big").to_pydict()
+    assert out["s"] == [50]
+    tables = s.sql("show tables").to_pydict()["table"]
+    assert "big" in tables and "t" in tables
+    with pytest.raises(ValueError):
+        s.sql("create table big as select * from t")
+    s.sql("create or replace table big as select a from t")
+    assert s.sql("select count(*) as c from big").to_pydict()["c"] == [3]
+    s.sql("drop table big")
+    assert "big" not in s.sql("show tables").to_pydict()["table"]
+    with pytest.raises(KeyError):
+        s.sql("drop table nope")
+    s.sql("drop table if exists nope")
