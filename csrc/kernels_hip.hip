@@ -920,3 +920,132 @@ Tensor u64_mod(Tensor hashes, int64_t n_partitions) {
   }
   return out;
 }
+
+// ---------------------------------------------------------------------------
+// fused grouped multi-aggregate: ONE pass over (gids, value columns)
+// computing every sum/min/max/count of an aggregation at once.  The
+// per-agg path reads gids once per aggregate (600M x 8B x n_aggs on a
+// Q1-shaped plan); here gids load once per row and all accumulators live
+// in LDS.  Capability of the reference's per-type agg kernels
+// (daft-core/src/array/ops/{sum,mean,count,compare_agg}.rs) fused the
+// MI355X way.  ops: 0=sum(f64) 1=min 2=max 3=count-only.
+// ---------------------------------------------------------------------------
+struct MAggCol {
+  const double* data;
+  const bool* valid;
+  int64_t op;
+};
+
+__global__ void grouped_multi_agg_kernel(const int64_t* gids, int64_t n,
+                                         int num_groups, int n_aggs,
+                                         const MAggCol* cols, double* out,
+                                         int64_t* cnt) {
+  extern __shared__ char smem[];
+  int slots = num_groups * n_aggs;
+  double* lacc = (double*)smem;
+  uint32_t* lcnt = (uint32_t*)(smem + (size_t)slots * sizeof(double));
+  for (int s = threadIdx.x; s < slots; s += blockDim.x) {
+    int64_t op = cols[s / num_groups].op;
+    if (op == 1)
+      ((uint64_t*)lacc)[s] = ~0ull;       // +inf in order-bits
+    else if (op == 2)
+      ((uint64_t*)lacc)[s] = 0ull;        // -inf in order-bits
+    else
+      lacc[s] = 0.0;
+    lcnt[s] = 0;
+  }
+  __syncthreads();
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int g = (int)gids[i];
+    for (int a = 0; a < n_aggs; ++a) {
+      MAggCol c = cols[a];
+      if (c.valid && !c.valid[i]) continue;
+      int s = a * num_groups + g;
+      atomicAdd(&lcnt[s], 1u);
+      if (c.op == 0) {
+        atomicAdd(&lacc[s], c.data[i]);
+      } else if (c.op == 1) {
+        atomicMin((unsigned long long*)&lacc[s],
+                  (unsigned long long)f64_order_bits(c.data[i]));
+      } else if (c.op == 2) {
+        atomicMax((unsigned long long*)&lacc[s],
+                  (unsigned long long)f64_order_bits(c.data[i]));
+      }
+    }
+  }
+  __syncthreads();
+  for (int s = threadIdx.x; s < slots; s += blockDim.x) {
+    if (lcnt[s] == 0) continue;
+    int64_t op = cols[s / num_groups].op;
+    atomicAdd((unsigned long long*)&cnt[s], (unsigned long long)lcnt[s]);
+    if (op == 0)
+      atomicAdd(&out[s], lacc[s]);
+    else if (op == 1)
+      atomicMin((unsigned long long*)&out[s], ((uint64_t*)lacc)[s]);
+    else if (op == 2)
+      atomicMax((unsigned long long*)&out[s], ((uint64_t*)lacc)[s]);
+  }
+}
+
+__global__ void multi_agg_decode_kernel(const MAggCol* cols, int num_groups,
+                                        int n_aggs, double* out,
+                                        const int64_t* cnt) {
+  int s = blockIdx.x * blockDim.x + threadIdx.x;
+  if (s >= num_groups * n_aggs) return;
+  int64_t op = cols[s / num_groups].op;
+  if (op == 1 || op == 2) {
+    double v = cnt[s] > 0 ? f64_from_order_bits(((uint64_t*)out)[s]) : 0.0;
+    __builtin_memcpy(&out[s], &v, 8);
+  }
+}
+
+std::vector<Tensor> grouped_multi_agg(Tensor gids, int64_t num_groups,
+                                      std::vector<Tensor> datas,
+                                      std::vector<OptTensor> valids,
+                                      std::vector<int64_t> ops) {
+  auto dev = gids.device();
+  int n_aggs = (int)ops.size();
+  int64_t n = gids.numel();
+  int64_t slots = num_groups * n_aggs;
+  TORCH_CHECK(slots <= 2048, "grouped_multi_agg: too many slots");
+  auto host = torch::empty({n_aggs * 3}, torch::dtype(torch::kInt64));
+  int64_t* h = host.data_ptr<int64_t>();
+  for (int i = 0; i < n_aggs; ++i) {
+    h[i * 3 + 0] = datas[i].defined() && datas[i].numel()
+                       ? (int64_t)datas[i].data_ptr<double>() : 0;
+    h[i * 3 + 1] = valids[i].has_value()
+                       ? (int64_t)valids[i]->data_ptr<bool>() : 0;
+    h[i * 3 + 2] = ops[i];
+  }
+  auto descs = host.to(dev);
+  auto out = torch::zeros({slots}, torch::dtype(torch::kFloat64).device(dev));
+  // min/max identities as order bits
+  {
+    auto ob = out.view(torch::kInt64);
+    for (int i = 0; i < n_aggs; ++i) {
+      if (ops[i] == 1)
+        ob.slice(0, i * num_groups, (i + 1) * num_groups).fill_(-1);
+      else if (ops[i] == 2)
+        ob.slice(0, i * num_groups, (i + 1) * num_groups).fill_(0);
+    }
+  }
+  auto cnt = torch::zeros({slots}, torch::dtype(torch::kInt64).device(dev));
+  if (n > 0) {
+    int block = 256;
+    size_t lds = (size_t)slots * (sizeof(double) + sizeof(uint32_t));
+    hipLaunchKernelGGL(grouped_multi_agg_kernel,
+                       dim3(grid_1d(n, block, 8)), dim3(block), lds,
+                       cur_stream(), gids.data_ptr<int64_t>(), n,
+                       (int)num_groups, n_aggs,
+                       (const MAggCol*)descs.data_ptr(),
+                       out.data_ptr<double>(), cnt.data_ptr<int64_t>());
+    hipLaunchKernelGGL(multi_agg_decode_kernel,
+                       dim3((int)((slots + 255) / 256)), dim3(256), 0,
+                       cur_stream(), (const MAggCol*)descs.data_ptr(),
+                       (int)num_groups, n_aggs, out.data_ptr<double>(),
+                       cnt.data_ptr<int64_t>());
+  }
+  return {out, cnt};
+}

@@ -309,6 +309,87 @@ def _string_minmax(group_ids, num_groups, values: Series, kind, name):
     return values.take(sel).rename(name)
 
 
+_MULTI_AGG_KINDS = {AggKind.SUM, AggKind.MIN, AggKind.MAX, AggKind.MEAN,
+                    AggKind.COUNT, AggKind.COUNT_ALL}
+
+
+def _multi_agg(batch, gids, num_groups, named_aggs, mask):
+    """Fused one-pass aggregation (csrc grouped_multi_agg): every
+    sum/min/max/mean/count computed from a single read of (gids, values)
+    with all accumulators in LDS.  Returns None when the shape doesn't
+    fit (few aggs, non-float values, too many slots) — per-agg kernels
+    handle the rest."""
+    from ..kernels import load_native
+    n = len(batch)
+    dev = batch.device
+    if dev.type != "cuda" or len(named_aggs) < 2 or n == 0:
+        return None
+    if num_groups * len(named_aggs) > 2048:
+        return None
+    if any(a.kind not in _MULTI_AGG_KINDS for _, a in named_aggs):
+        return None
+    nat = load_native()
+    if nat is None:
+        return None
+    datas, valids, ops, meta = [], [], [], []
+    value_cache: dict = {}
+    for cname, a in named_aggs:
+        if a.kind == AggKind.COUNT_ALL or (a.kind == AggKind.COUNT and
+                                           a.child is None):
+            datas.append(torch.empty(0, dtype=torch.float64, device=dev))
+            valids.append(mask)
+            ops.append(3)
+            meta.append((cname, a, None))
+            continue
+        key = repr(a.child)
+        values = value_cache.get(key)
+        if values is None:
+            values = a.child.evaluate(batch)
+            if len(values) == 1 and n > 1:
+                values = values.broadcast(n)
+            value_cache[key] = values
+        if values.dtype.kind not in (TypeKind.FLOAT32, TypeKind.FLOAT64):
+            return None
+        d = values.data.to(torch.float64)
+        v = values.validity
+        if mask is not None:
+            v = mask if v is None else (v & mask)
+        if a.kind == AggKind.COUNT:
+            datas.append(torch.empty(0, dtype=torch.float64, device=dev))
+            valids.append(v)
+            ops.append(3)
+        else:
+            datas.append(d.contiguous())
+            valids.append(v.contiguous() if v is not None else None)
+            ops.append({AggKind.SUM: 0, AggKind.MIN: 1, AggKind.MAX: 2,
+                        AggKind.MEAN: 0}[a.kind])
+        meta.append((cname, a, values))
+    out, cnt = nat.grouped_multi_agg(gids, num_groups, datas, valids, ops)
+    out = out.view(len(named_aggs), num_groups)
+    cnt = cnt.view(len(named_aggs), num_groups)
+    cols = []
+    for i, (cname, a, values) in enumerate(meta):
+        k = a.kind
+        if k in (AggKind.COUNT, AggKind.COUNT_ALL):
+            cols.append(Series(cname, DataType.uint64(),
+                               data=cnt[i].contiguous()
+                               .view(torch.uint64)))
+            continue
+        c = cnt[i]
+        validity = c > 0
+        if bool(validity.all()):
+            validity = None
+        if k == AggKind.MEAN:
+            data = out[i] / c.clamp(min=1).to(torch.float64)
+            cols.append(Series(cname, DataType.float64(),
+                               data=data.contiguous(), validity=validity))
+            continue
+        out_dt = a.to_field(batch.schema).dtype
+        data = out[i].contiguous().to(out_dt.to_torch())
+        cols.append(Series(cname, out_dt, data=data, validity=validity))
+    return cols
+
+
 def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
                   aggs: List[ExprNode],
                   mask: Optional[torch.Tensor] = None) -> RecordBatch:
@@ -330,8 +411,11 @@ def run_aggregate(batch: RecordBatch, groupby: List[ExprNode],
         num_groups = 1
         key_cols = []
 
-    agg_cols = [compute_agg(batch, gids, num_groups, cname, a, mask=mask)
-                for cname, a in named_aggs]
+    agg_cols = _multi_agg(batch, gids, num_groups, named_aggs, mask)
+    if agg_cols is None:
+        agg_cols = [compute_agg(batch, gids, num_groups, cname, a,
+                                mask=mask)
+                    for cname, a in named_aggs]
     inter = RecordBatch(key_cols + agg_cols,
                         num_rows=num_groups)
     out_cols = list(key_cols)
