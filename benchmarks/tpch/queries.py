@@ -362,15 +362,23 @@ def q21(T):
         .select("o_orderkey")
     li = T["lineitem"].join(orders_f, left_on="l_orderkey",
                             right_on="o_orderkey", how="semi")
-    per_order = (li.groupby("l_orderkey")
-                 .agg(col("l_suppkey").count_distinct().alias("n_supp")))
-    late = li.where(col("l_receiptdate") > col("l_commitdate"))
-    late_per_order = (late.groupby("l_orderkey")
-                      .agg(col("l_suppkey").count_distinct()
-                           .alias("n_late_supp")))
-    qualifying = (per_order.join(late_per_order, on="l_orderkey")
-                  .where((col("n_supp") > 1) & (col("n_late_supp") == 1))
+    # one groupby computes both the all-rows and the late-rows supplier
+    # spread: >1 distinct suppliers <=> min != max; exactly one late
+    # supplier <=> late-count > 0 and late-min == late-max
+    late_flag = col("l_receiptdate") > col("l_commitdate")
+    lsup = late_flag.if_else(col("l_suppkey"), lit(None))
+    per_order = (li.with_column("l_late_supp", lsup)
+                 .groupby("l_orderkey")
+                 .agg(col("l_suppkey").min().alias("mn"),
+                      col("l_suppkey").max().alias("mx"),
+                      col("l_late_supp").min().alias("lmn"),
+                      col("l_late_supp").max().alias("lmx"),
+                      col("l_late_supp").count().alias("lc")))
+    qualifying = (per_order
+                  .where((col("mn") != col("mx")) & (col("lc") > 0) &
+                         (col("lmn") == col("lmx")))
                   .select("l_orderkey"))
+    late = li.where(late_flag)
     nation = T["nation"].where(col("n_name") == "SAUDI ARABIA")
     supp = T["supplier"].join(nation, left_on="s_nationkey",
                               right_on="n_nationkey")
