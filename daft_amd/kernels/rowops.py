@@ -417,8 +417,10 @@ def _dense_key_join(lk, rk, how):
     table = torch.full((rng,), -1, dtype=idx_t, device=dev)
     rows = torch.arange(n_r, dtype=idx_t, device=dev)
     table[rdata - mn] = rows
-    if not bool((table[rdata - mn] == rows).all()):
+    if how not in ("semi", "anti") and \
+            not bool((table[rdata - mn] == rows).all()):
         return None  # duplicate build keys: hash join handles fan-out
+    # (semi/anti only test existence, so duplicate build keys are fine)
     ldata = l.data.to(torch.int64)
     in_rng = (ldata >= mn) & (ldata <= mx)
     if l.validity is not None:

@@ -371,7 +371,15 @@ def test_dense_key_join_matches_hash_join():
         else:
             assert sorted(zip(li1.tolist(), ri1.tolist())) == \
                 sorted(zip(li2.tolist(), ri2.tolist())), how
-    # duplicate build keys and sparse ranges fall back to the hash join
+    # duplicate build keys: semi/anti stay on the dense path (existence
+    # only); fan-out joins fall back to the hash join
+    dupr = [Series.from_pylist("k", [1, 2, 2, 7], DataType.int64())]
+    for how in ("semi", "anti"):
+        d = rowops._dense_key_join(lk, dupr, how)
+        assert d is not None
+        li1, _ = d
+        li2, _ = rowops._cpu_join(lk, dupr, how)
+        assert sorted(li1.tolist()) == sorted(li2.tolist()), how
     assert rowops._dense_key_join(
         lk, [Series.from_pylist("k", [1, 2, 2], DataType.int64())],
         "inner") is None
