@@ -65,6 +65,10 @@ std::vector<Tensor> grouped_multi_agg(Tensor gids, int64_t num_groups,
                                       std::vector<Tensor> datas,
                                       std::vector<OptTensor> valids,
                                       std::vector<int64_t> ops);
+std::vector<Tensor> grouped_multi_agg_big(Tensor gids, int64_t num_groups,
+                                          std::vector<Tensor> datas,
+                                          std::vector<OptTensor> valids,
+                                          std::vector<int64_t> ops);
 Tensor minhash(Tensor offsets, Tensor bytes, int64_t num_hashes,
                int64_t ngram_size, Tensor perm_a, Tensor perm_b);
 Tensor hll_update(Tensor hashes, Tensor gids, Tensor valid,

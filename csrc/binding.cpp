@@ -24,6 +24,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("u64_mod", &u64_mod, "unsigned modulo for hash partitioning");
   m.def("simhash", &simhash);
   m.def("grouped_multi_agg", &grouped_multi_agg);
+  m.def("grouped_multi_agg_big", &grouped_multi_agg_big);
   m.def("minhash", &minhash, "word-ngram MinHash signatures (wave/row)");
   m.def("hll_update", &hll_update, "HyperLogLog register updates");
   m.def("image_resize", &image_resize, "bilinear uint8 HWC resize");

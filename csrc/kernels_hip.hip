@@ -1049,3 +1049,97 @@ std::vector<Tensor> grouped_multi_agg(Tensor gids, int64_t num_groups,
   }
   return {out, cnt};
 }
+
+// global-memory variant for large group counts (num_groups beyond LDS):
+// still one pass over gids+values for every aggregate — at >>2048 groups
+// atomic contention per slot is negligible, the win is reading gids once
+// instead of once per aggregate.
+__global__ void grouped_multi_agg_global_kernel(const int64_t* gids,
+                                                int64_t n,
+                                                int64_t num_groups,
+                                                int n_aggs,
+                                                const MAggCol* cols,
+                                                double* out, int64_t* cnt) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t g = gids[i];
+    for (int a = 0; a < n_aggs; ++a) {
+      MAggCol c = cols[a];
+      if (c.valid && !c.valid[i]) continue;
+      int64_t s = (int64_t)a * num_groups + g;
+      atomicAdd((unsigned long long*)&cnt[s], 1ull);
+      if (c.op == 0) {
+        atomicAdd(&out[s], c.data[i]);
+      } else if (c.op == 1) {
+        atomicMin((unsigned long long*)&out[s],
+                  (unsigned long long)f64_order_bits(c.data[i]));
+      } else if (c.op == 2) {
+        atomicMax((unsigned long long*)&out[s],
+                  (unsigned long long)f64_order_bits(c.data[i]));
+      }
+    }
+  }
+}
+
+__global__ void multi_agg_decode_big_kernel(const MAggCol* cols,
+                                            int64_t num_groups, int n_aggs,
+                                            double* out,
+                                            const int64_t* cnt) {
+  int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = num_groups * n_aggs;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; s < total; s += stride) {
+    int64_t op = cols[s / num_groups].op;
+    if (op == 1 || op == 2) {
+      double v = cnt[s] > 0 ? f64_from_order_bits(((uint64_t*)out)[s]) : 0.0;
+      __builtin_memcpy(&out[s], &v, 8);
+    }
+  }
+}
+
+std::vector<Tensor> grouped_multi_agg_big(Tensor gids, int64_t num_groups,
+                                          std::vector<Tensor> datas,
+                                          std::vector<OptTensor> valids,
+                                          std::vector<int64_t> ops) {
+  auto dev = gids.device();
+  int n_aggs = (int)ops.size();
+  int64_t n = gids.numel();
+  int64_t slots = num_groups * n_aggs;
+  auto host = torch::empty({n_aggs * 3}, torch::dtype(torch::kInt64));
+  int64_t* h = host.data_ptr<int64_t>();
+  for (int i = 0; i < n_aggs; ++i) {
+    h[i * 3 + 0] = datas[i].defined() && datas[i].numel()
+                       ? (int64_t)datas[i].data_ptr<double>() : 0;
+    h[i * 3 + 1] = valids[i].has_value()
+                       ? (int64_t)valids[i]->data_ptr<bool>() : 0;
+    h[i * 3 + 2] = ops[i];
+  }
+  auto descs = host.to(dev);
+  auto out = torch::zeros({slots}, torch::dtype(torch::kFloat64).device(dev));
+  {
+    auto ob = out.view(torch::kInt64);
+    for (int i = 0; i < n_aggs; ++i) {
+      if (ops[i] == 1)
+        ob.slice(0, i * num_groups, (i + 1) * num_groups).fill_(-1);
+      else if (ops[i] == 2)
+        ob.slice(0, i * num_groups, (i + 1) * num_groups).fill_(0);
+    }
+  }
+  auto cnt = torch::zeros({slots}, torch::dtype(torch::kInt64).device(dev));
+  if (n > 0) {
+    int block = 256;
+    hipLaunchKernelGGL(grouped_multi_agg_global_kernel,
+                       dim3(grid_1d(n, block, 4)), dim3(block), 0,
+                       cur_stream(), gids.data_ptr<int64_t>(), n,
+                       num_groups, n_aggs,
+                       (const MAggCol*)descs.data_ptr(),
+                       out.data_ptr<double>(), cnt.data_ptr<int64_t>());
+    hipLaunchKernelGGL(multi_agg_decode_big_kernel,
+                       dim3(grid_1d(slots, 256)), dim3(256), 0,
+                       cur_stream(), (const MAggCol*)descs.data_ptr(),
+                       num_groups, n_aggs, out.data_ptr<double>(),
+                       cnt.data_ptr<int64_t>());
+  }
+  return {out, cnt};
+}

@@ -324,7 +324,9 @@ def _multi_agg(batch, gids, num_groups, named_aggs, mask):
     dev = batch.device
     if dev.type != "cuda" or len(named_aggs) < 2 or n == 0:
         return None
-    if num_groups * len(named_aggs) > 2048:
+    big = num_groups * len(named_aggs) > 2048
+    # memory guard on the big variant: n_aggs * num_groups * 16B
+    if big and num_groups * len(named_aggs) > (1 << 31):
         return None
     if any(a.kind not in _MULTI_AGG_KINDS for _, a in named_aggs):
         return None
@@ -364,7 +366,8 @@ def _multi_agg(batch, gids, num_groups, named_aggs, mask):
             ops.append({AggKind.SUM: 0, AggKind.MIN: 1, AggKind.MAX: 2,
                         AggKind.MEAN: 0}[a.kind])
         meta.append((cname, a, values))
-    out, cnt = nat.grouped_multi_agg(gids, num_groups, datas, valids, ops)
+    fn = nat.grouped_multi_agg_big if big else nat.grouped_multi_agg
+    out, cnt = fn(gids, num_groups, datas, valids, ops)
     out = out.view(len(named_aggs), num_groups)
     cnt = cnt.view(len(named_aggs), num_groups)
     cols = []
