@@ -38,7 +38,11 @@ class RecordBatch:
         cols = []
         for name, vals in data.items():
             dt = schema[name].dtype if schema is not None and name in schema else None
-            if isinstance(vals, torch.Tensor):
+            if isinstance(vals, Series):
+                v = vals.rename(name)
+                cols.append(v.to(device) if str(v.device) != str(device)
+                            else v)
+            elif isinstance(vals, torch.Tensor):
                 cols.append(Series.from_torch(name, vals.to(device)))
             else:
                 cols.append(Series.from_pylist(name, vals, dt, device=device))

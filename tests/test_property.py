@@ -113,3 +113,54 @@ def test_distinct_is_set(vals):
     df = daft.from_pydict({"v": vals})
     out = df.distinct().to_pydict()["v"]
     assert sorted(out) == sorted(set(vals))
+
+
+@given(st.lists(st.one_of(
+    st.none(),
+    st.decimals(min_value=-10**10, max_value=10**10, places=2,
+                allow_nan=False, allow_infinity=False)),
+    min_size=0, max_size=50))
+@settings(max_examples=40, deadline=None)
+def test_decimal_sum_exact_and_roundtrip(vals):
+    """Decimal columns: to_pylist round-trips exactly and sum matches the
+    python Decimal sum (scaled-int64 storage)."""
+    from decimal import Decimal
+    from daft_amd import DataType
+    df = daft.from_pydict({"v": daft.Series.from_pylist(
+        "v", vals, DataType.decimal128(14, 2))})
+    back = df.to_pydict()["v"]
+    assert back == vals
+    got = df.agg(col("v").sum().alias("s")).to_pydict()["s"][0]
+    nn = [v for v in vals if v is not None]
+    want = sum(nn, Decimal(0)) if nn else None
+    if want is None:
+        assert got is None
+    else:
+        assert got == want
+
+
+@given(st.lists(st.one_of(st.none(), st.integers(-500, 500)),
+                min_size=0, max_size=60),
+       st.lists(st.integers(0, 80), min_size=0, max_size=30))
+@settings(max_examples=40, deadline=None)
+def test_dense_join_property(lvals, rvals):
+    """Join results are identical whether the dense or hash path runs."""
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+    from daft_amd.kernels import rowops
+    rvals = list(dict.fromkeys(rvals))  # unique build keys
+    if not rvals or not lvals:
+        return
+    lk = [Series.from_pylist("k", lvals, DataType.int64())]
+    rk = [Series.from_pylist("k", rvals, DataType.int64())]
+    for how in ("inner", "left", "semi", "anti"):
+        d = rowops._dense_key_join(lk, rk, how)
+        li2, ri2 = rowops._cpu_join(lk, rk, how)
+        if d is None:
+            continue
+        li1, ri1 = d
+        if how in ("semi", "anti"):
+            assert sorted(li1.tolist()) == sorted(li2.tolist())
+        else:
+            assert sorted(zip(li1.tolist(), ri1.tolist())) == \
+                sorted(zip(li2.tolist(), ri2.tolist()))
