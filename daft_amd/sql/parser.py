@@ -20,6 +20,8 @@ KEYWORDS = {
     "all", "distinct", "asc", "desc", "interval", "date", "extract",
     "substring", "for", "with", "count", "sum", "avg", "min", "max",
     "first", "last", "nulls", "semi", "anti", "true", "false", "cast",
+    "over", "partition", "rows", "between", "unbounded", "preceding",
+    "current", "row", "following",
 }
 
 TOKEN_RE = re.compile(r"""
@@ -110,6 +112,14 @@ class FuncCall:
     args: List[Any]
     distinct: bool = False
     star: bool = False
+
+
+@dataclass
+class WindowExpr:
+    func: Any                       # FuncCall underneath
+    partition_by: List[Any]
+    order_by: List[Tuple[Any, bool]]  # (expr, desc)
+    frame: Any = None               # (lo, hi) row offsets or None
 
 
 @dataclass
@@ -620,7 +630,51 @@ class Parser:
             while self.eat_op(","):
                 args.append(self.parse_expr())
         self.expect_op(")")
-        return FuncCall(name, args, distinct=distinct)
+        fn = FuncCall(name, args, distinct=distinct)
+        if self.at_kw("over"):
+            return self._parse_over(fn)
+        return fn
+
+    def _parse_over(self, fn):
+        self.expect_kw("over")
+        self.expect_op("(")
+        parts, order, frame = [], [], None
+        if self.eat_kw("partition"):
+            self.expect_kw("by")
+            parts.append(self.parse_expr())
+            while self.eat_op(","):
+                parts.append(self.parse_expr())
+        if self.eat_kw("order"):
+            self.expect_kw("by")
+            while True:
+                e = self.parse_expr()
+                desc = False
+                if self.eat_kw("desc"):
+                    desc = True
+                elif self.eat_kw("asc"):
+                    pass
+                order.append((e, desc))
+                if not self.eat_op(","):
+                    break
+        if self.eat_kw("rows"):
+            self.expect_kw("between")
+
+            def bound():
+                if self.eat_kw("unbounded"):
+                    side = self.next().value.lower()
+                    return None if side == "preceding" else None
+                if self.eat_kw("current"):
+                    self.expect_kw("row")
+                    return 0
+                n = int(self.next().value)
+                side = self.next().value.lower()
+                return -n if side == "preceding" else n
+            lo = bound()
+            self.expect_kw("and")
+            hi = bound()
+            frame = (lo, hi)
+        self.expect_op(")")
+        return WindowExpr(fn, parts, order, frame)
 
 
 def parse_sql(text: str) -> SelectStmt:

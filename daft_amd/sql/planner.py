@@ -520,15 +520,29 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
         if it.alias and not it.star:
             select_aliases[it.alias] = it.expr
 
+    # window functions: compute as extra columns before projection
+    win_map = {}
+    for i, it in enumerate(stmt.items):
+        if isinstance(it.expr, P.WindowExpr):
+            if has_agg:
+                raise SQLPlanError("window functions with GROUP BY are "
+                                   "not supported yet")
+            wname = it.alias or f"__w{i}"
+            df = df.with_window_columns(
+                {wname: _window_to_daft(it.expr, binder)})
+            win_map[i] = wname
+
     if has_agg:
         df = _plan_aggregate(df, stmt, binder, select_aliases, lookup,
                              ctes)
         out_names = _output_names(stmt, binder)
     else:
         exprs = []
-        for it in stmt.items:
+        for i, it in enumerate(stmt.items):
             if it.star:
                 exprs.extend(col(n) for n in df.column_names())
+            elif i in win_map:
+                exprs.append(col(win_map[i]))
             else:
                 e = expr_to_daft(it.expr, binder)
                 name = it.alias or _default_name(it.expr, binder)
@@ -556,9 +570,52 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
     return df
 
 
+def _window_to_daft(we, binder):
+    """WindowExpr AST -> Expression bound to a Window spec (ref:
+    daft-sql window planning)."""
+    from ..window import Window
+    from .. import functions as F
+    w = Window()
+    if we.partition_by:
+        w = w.partition_by(*[expr_to_daft(p, binder)
+                             for p in we.partition_by])
+    if we.order_by:
+        w = w.order_by(*[expr_to_daft(e, binder) for e, _ in we.order_by],
+                       desc=[d for _, d in we.order_by])
+    if we.frame is not None:
+        lo, hi = we.frame
+        lo = Window.unbounded_preceding if lo is None else lo
+        hi = Window.unbounded_following if hi is None else hi
+        w = w.rows_between(lo, hi)
+    fn = we.func.name
+    args = we.func.args
+    if fn in ("row_number", "rank", "dense_rank"):
+        return getattr(F, fn)().over(w)
+    if fn in ("first_value", "last_value"):
+        e = expr_to_daft(args[0], binder)
+        make = F.w_first_value if fn == "first_value" else F.w_last_value
+        return make(e).over(w)
+    if fn in ("lag", "lead"):
+        e = expr_to_daft(args[0], binder)
+        off = args[1].value if len(args) > 1 else 1
+        dflt = args[2].value if len(args) > 2 else None
+        return getattr(e, fn)(int(off), dflt).over(w)
+    if fn in ("sum", "avg", "min", "max", "count", "mean", "stddev"):
+        if we.func.star:
+            from ..expressions.expressions import Agg, AggKind, Expression
+            return Expression(Agg(AggKind.COUNT_ALL, None)).over(w)
+        e = expr_to_daft(args[0], binder)
+        agg = {"sum": e.sum, "avg": e.mean, "mean": e.mean, "min": e.min,
+               "max": e.max, "count": e.count, "stddev": e.stddev}[fn]()
+        return agg.over(w)
+    raise SQLPlanError(f"unsupported window function {fn!r}")
+
+
 def _has_aggregate(e) -> bool:
     if e is None:
         return False
+    if isinstance(e, P.WindowExpr):
+        return False  # window aggregates are not GROUP BY aggregates
     if isinstance(e, P.FuncCall) and (e.name in _AGG_FUNCS or e.star):
         return True
     for f in getattr(e, "__dataclass_fields__", {}):

@@ -120,3 +120,26 @@ def test_tpch_sql_matches_dataframe(qi, T):
                     (math.isnan(gx) and math.isnan(wx))), (gx, wx)
             else:
                 assert gx == wx, (gx, wx)
+
+
+def test_sql_window_functions():
+    """OVER clause: rank fns, running/partition aggregates, lag,
+    first_value, ROWS BETWEEN frames."""
+    df = daft.from_pydict({"g": ["a", "a", "b", "b", "b"],
+                           "v": [3, 1, 2, 9, 8]})
+    o = daft.sql("""select g, v,
+        row_number() over (partition by g order by v) as rn,
+        sum(v) over (partition by g order by v) as run,
+        sum(v) over (partition by g) as tot,
+        lag(v, 1) over (partition by g order by v) as prev,
+        first_value(v) over (partition by g order by v) as fv
+      from df order by g, v""").to_pydict()
+    assert o["rn"] == [1, 2, 1, 2, 3]
+    assert o["run"] == [1.0, 4.0, 2.0, 10.0, 19.0]
+    assert o["tot"] == [4.0, 4.0, 19.0, 19.0, 19.0]
+    assert o["prev"] == [None, 1, None, 2, 8]
+    assert o["fv"] == [1, 1, 2, 2, 2]
+    o2 = daft.sql("select g, v, sum(v) over (partition by g order by v "
+                  "rows between 1 preceding and current row) as m "
+                  "from df order by g, v").to_pydict()
+    assert o2["m"] == [1.0, 4.0, 2.0, 10.0, 17.0]
