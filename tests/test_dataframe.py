@@ -386,3 +386,40 @@ def test_dense_key_join_matches_hash_join():
     assert rowops._dense_key_join(
         lk, [Series.from_pylist("k", [1, 10**9], DataType.int64())],
         "inner") is None
+
+
+def test_dataframe_parity_methods():
+    """Reference-parity method batch: set ops, agg shortcuts, describe,
+    drop_nan/null, map_groups, write_sink."""
+    df = daft.from_pydict({"a": [1, 2, 2, None],
+                           "b": [1.0, float("nan"), 3.0, 4.0]})
+    assert df.columns == ["a", "b"]
+    assert df.drop_null("a").count_rows() == 3
+    assert df.drop_nan("b").count_rows() == 3
+    d = df.describe().to_pydict()
+    assert d["column"] == ["a", "b"] and d["nulls"] == [1, 0]
+    u = daft.from_pydict({"a": [9], "c": ["x"]})
+    ub = df.union_all_by_name(u).to_pydict()
+    assert set(ub.keys()) == {"a", "b", "c"} and len(ub["a"]) == 5
+    l = daft.from_pydict({"x": [1, 1, 2, 3]})
+    r = daft.from_pydict({"x": [1, 2, 2]})
+    assert sorted(l.intersect_all(r).to_pydict()["x"]) == [1, 2]
+    assert sorted(l.except_all(r).to_pydict()["x"]) == [1, 3]
+    assert df.count_distinct("a").to_pydict()["a"] == [2]
+    g = daft.from_pydict({"k": ["a", "a", "b"], "v": [1, 2, 3]})
+    mg = g.map_groups(lambda s: s.sum("v"), "k").to_pydict()
+    assert sorted(mg["v"]) == [3, 3]
+
+    class Sink:
+        rows = 0
+
+        def write(self, b):
+            Sink.rows += len(b)
+            return len(b)
+
+        def finalize(self, rs):
+            return sum(rs)
+    df.write_sink(Sink())
+    assert Sink.rows == 4
+    with pytest.raises(RuntimeError):
+        df.write_deltalake("x")
