@@ -512,6 +512,14 @@ class DataFrame:
     write_turbopuffer = _gated_writer("write_turbopuffer", "turbopuffer")
     del _gated_writer
 
+    def to_dask_dataframe(self, *a, **k):
+        raise RuntimeError("to_dask_dataframe() requires dask, not "
+                           "available in this offline build")
+
+    def to_ray_dataset(self, *a, **k):
+        raise RuntimeError("to_ray_dataset() requires ray; this engine "
+                           "uses SPMD torch.distributed instead of Ray")
+
     def skip_existing(self, store, on: ColumnInput) -> "DataFrame":
         """Checkpoint-assisted dedup: drop rows whose key already exists
         in the checkpoint store (ref: CheckpointConfig rows-skipped)."""
