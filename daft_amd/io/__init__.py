@@ -136,14 +136,6 @@ def read_json(path, io_config=None, **kwargs) -> DataFrame:
         sch, paths, "json", storage_options=io_config))
 
 
-def from_glob_path(path: str) -> DataFrame:
-    """List files matching a glob into a DataFrame (path, size, num_rows)."""
-    paths = _expand_paths(path)
-    sizes = [os.path.getsize(p) if os.path.exists(p) else None for p in paths]
-    return from_pydict({"path": paths, "size": sizes,
-                        "num_rows": [None] * len(paths)})
-
-
 def read_text(path, line_column: str = "text") -> DataFrame:
     """Read text files line-by-line (ref capability: daft-text read.rs)."""
     paths = _expand_paths(path)
@@ -157,3 +149,78 @@ def read_text(path, line_column: str = "text") -> DataFrame:
 def read_jsonl(path) -> DataFrame:
     """Read newline-delimited JSON (alias of read_json for local files)."""
     return read_json(path)
+
+
+def read_sql(sql: str, conn_factory, partition_col=None, **kwargs) -> DataFrame:
+    """Read from any DB-API connection factory (capability of
+    daft.read_sql; works offline against sqlite3)."""
+    conn = conn_factory() if callable(conn_factory) else conn_factory
+    cur = conn.cursor()
+    cur.execute(sql)
+    names = [d[0] for d in cur.description]
+    rows = cur.fetchall()
+    data = {n: [r[i] for r in rows] for i, n in enumerate(names)}
+    from .. import from_pydict
+    return from_pydict(data)
+
+
+def from_glob_path(path: str, **kwargs) -> DataFrame:
+    """One row per file matching the glob: path, size, num_rows=None
+    (ref: daft.from_glob_path)."""
+    import glob as _g
+    import os as _os
+    paths = sorted(_g.glob(path, recursive=True))
+    return _from_file_rows([p for p in paths if _os.path.isfile(p)])
+
+
+def from_files(paths, **kwargs) -> DataFrame:
+    if isinstance(paths, str):
+        return from_glob_path(paths)
+    return _from_file_rows(list(paths))
+
+
+def _from_file_rows(paths) -> DataFrame:
+    import os as _os
+    from .. import from_pydict
+    return from_pydict({
+        "path": paths,
+        "size": [_os.path.getsize(p) if _os.path.exists(p) else None
+                 for p in paths],
+        "num_rows": [None] * len(paths),
+    })
+
+
+def read_blob(path, **kwargs) -> DataFrame:
+    """Read whole files as binary rows: path + data (ref capability:
+    daft read of raw blobs)."""
+    import glob as _g
+    import os as _os
+    paths = sorted(_g.glob(path, recursive=True)) \
+        if isinstance(path, str) else list(path)
+    paths = [p for p in paths if _os.path.isfile(p)]
+    from .. import from_pydict
+    return from_pydict({
+        "path": paths,
+        "data": [open(p, "rb").read() for p in paths],
+    })
+
+
+def _gated_reader(name, needs):
+    def make(*a, **k):
+        raise RuntimeError(
+            f"{name}() requires {needs}, which is not available in this "
+            f"offline build")
+    make.__name__ = name
+    return make
+
+
+read_deltalake = _gated_reader("read_deltalake", "deltalake")
+read_iceberg = _gated_reader("read_iceberg", "pyiceberg")
+read_lance = _gated_reader("read_lance", "lance")
+read_hudi = _gated_reader("read_hudi", "hudi")
+read_kafka = _gated_reader("read_kafka", "a kafka client")
+read_mcap = _gated_reader("read_mcap", "mcap")
+read_paimon = _gated_reader("read_paimon", "paimon")
+read_huggingface = _gated_reader("read_huggingface",
+                                 "network access to the HF hub")
+read_video_frames = _gated_reader("read_video_frames", "ffmpeg")

@@ -668,3 +668,23 @@ def test_session_sql_ddl():
     with pytest.raises(KeyError):
         s.sql("drop table nope")
     s.sql("drop table if exists nope")
+
+
+def test_read_sql_and_blob(tmp_path):
+    import sqlite3
+    import daft_amd as daft
+    db = str(tmp_path / "t.db")
+    c = sqlite3.connect(db)
+    c.execute("create table t (a int, b text)")
+    c.executemany("insert into t values (?, ?)", [(1, "x"), (2, "y")])
+    c.commit()
+    df = daft.read_sql("select * from t order by a",
+                       lambda: sqlite3.connect(db))
+    assert df.to_pydict() == {"a": [1, 2], "b": ["x", "y"]}
+    (tmp_path / "f1.bin").write_bytes(b"abc")
+    g = daft.from_glob_path(str(tmp_path / "*.bin")).to_pydict()
+    assert g["size"] == [3]
+    b = daft.read_blob(str(tmp_path / "*.bin")).to_pydict()
+    assert b["data"] == [b"abc"]
+    with pytest.raises(RuntimeError):
+        daft.read_deltalake("x")
