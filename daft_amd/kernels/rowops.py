@@ -368,12 +368,62 @@ def join(left_keys: Sequence[Series], right_keys: Sequence[Series],
         lk2.append(l.cast(st) if l.dtype != st else l)
         rk2.append(r.cast(st) if r.dtype != st else r)
     left_keys, right_keys = lk2, rk2
+    packed = _pack_int_keys(left_keys, right_keys)
+    if packed is not None:
+        left_keys, right_keys = packed
     dense = _dense_key_join(left_keys, right_keys, how)
     if dense is not None:
         return dense
     if _is_gpu(left_keys[0]):
         return _gpu_join(left_keys, right_keys, how)
     return _cpu_join(left_keys, right_keys, how)
+
+
+def _pack_int_keys(lk, rk):
+    """Fold a multi-column integer equi-key into ONE synthetic int64
+    column per side ((a-mnA)*rngB + (b-mnB)...): the probe then hashes
+    and compares a single word — and composite PK joins (e.g. partsupp's
+    (partkey, suppkey)) become eligible for the direct-address path.
+    Null semantics preserved: the packed column carries the AND of the
+    per-column validities (null keys never match)."""
+    if len(lk) < 2:
+        return None
+    mins, rngs = [], []
+    for i in range(len(lk)):
+        for s in (lk[i], rk[i]):
+            if s.is_dict() or s.data is None or s.data.dtype == \
+                    torch.uint64 or not (
+                    s.dtype.is_integer() or
+                    (s.dtype.is_decimal() and
+                     s.data.dtype == torch.int64) or
+                    s.dtype.kind == TypeKind.DATE):
+                return None
+            if len(s) == 0:
+                return None
+        lo = min(int(lk[i].data.min().item()),
+                 int(rk[i].data.min().item()))
+        hi = max(int(lk[i].data.max().item()),
+                 int(rk[i].data.max().item()))
+        mins.append(lo)
+        rngs.append(hi - lo + 1)
+    total = 1
+    for r in rngs:
+        total *= r
+        if total >= (1 << 62):
+            return None
+
+    def pack(cols):
+        acc = None
+        validity = None
+        for s, mn, rng in zip(cols, mins, rngs):
+            d = s.data.to(torch.int64) - mn
+            acc = d if acc is None else acc * rng + d
+            if s.validity is not None:
+                validity = s.validity if validity is None \
+                    else (validity & s.validity)
+        return [Series(cols[0].name, DataType.int64(), data=acc,
+                       validity=validity)]
+    return pack(lk), pack(rk)
 
 
 _DENSE_JOIN_LIMIT = 1 << 31  # direct table cap: 2G slots (8 GB int32)

@@ -423,3 +423,29 @@ def test_dataframe_parity_methods():
     assert Sink.rows == 4
     with pytest.raises(RuntimeError):
         df.write_deltalake("x")
+
+
+def test_packed_two_key_join_matches():
+    import numpy as np
+    rng = np.random.default_rng(5)
+    n = 500
+    l = daft.from_pydict({
+        "a": [int(v) for v in rng.integers(0, 30, n)],
+        "b": [int(v) for v in rng.integers(0, 10, n)],
+    })
+    r = daft.from_pydict({
+        "a": [int(v) for v in rng.integers(0, 30, 120)],
+        "b": [int(v) for v in rng.integers(0, 10, 120)],
+        "w": list(range(120)),
+    })
+    got = l.join(r, on=["a", "b"]).sort(["a", "b", "w"]).to_pydict()
+    # reference result via pandas merge
+    import pandas as pd
+    want = pd.merge(l.to_pandas(), r.to_pandas(), on=["a", "b"]) \
+        .sort_values(["a", "b", "w"])
+    assert got["w"] == want["w"].tolist()
+    # with nulls on one side
+    l2 = daft.from_pydict({"a": [1, None, 2], "b": [1, 1, None]})
+    r2 = daft.from_pydict({"a": [1, 2], "b": [1, 3], "w": [10, 20]})
+    out = l2.join(r2, on=["a", "b"]).to_pydict()
+    assert out["w"] == [10]
