@@ -45,6 +45,18 @@ class DataFrame:
 
     # ---- transforms ---------------------------------------------------
     def select(self, *exprs: ColumnInput) -> "DataFrame":
+        from .functions.aliases import _Unnest
+        if any(isinstance(e, _Unnest) for e in exprs):
+            out = []
+            for e in exprs:
+                if isinstance(e, _Unnest):
+                    f = e.expr._node.to_field(self.schema)
+                    for fld in f.dtype.fields:
+                        out.append(e.expr.struct.get(fld.name)
+                                   .alias(fld.name))
+                else:
+                    out.append(e)
+            exprs = tuple(out)
         return self._wrap(self._builder.select(list(exprs)))
 
     def with_column(self, name: str, expr: Expression) -> "DataFrame":

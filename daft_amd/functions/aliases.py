@@ -316,31 +316,28 @@ def conv(x, from_base: int, to_base: int):
                                DataType.string()))
 
 
-def unnest(x):
-    """Struct column -> its fields as separate columns (use inside
-    select: df.select(unnest(col("s")))).  Returns the field
-    expressions."""
-    e = _e(x)
+class _Unnest:
+    """Marker consumed by DataFrame.select: expands a struct column into
+    one output column per field (ref: daft.functions.unnest)."""
 
-    def expand(schema_fields):
-        raise TypeError("unnest() must be expanded by select; pass "
-                        "col('s').struct.get(name) fields instead")
-    # practical: return a list of per-field expressions via deferred
-    # resolution is schema-dependent, so we return a marker the
-    # DataFrame.select path can expand
-    return e  # struct column itself; DataFrame.select expands structs
+    def __init__(self, expr):
+        self.expr = expr
+
+
+def unnest(x):
+    return _Unnest(_e(x))
 
 
 def first_value(x, ignore_nulls: bool = False):
-    """First value per group (alias of any_value ordering-first)."""
-    return _e(x).any_value()
+    """first_value WINDOW function (use with .over(window); ref:
+    daft/functions/window.py:310)."""
+    from . import w_first_value
+    return w_first_value(_e(x))
 
 
 def last_value(x, ignore_nulls: bool = False):
-    from ..expressions.expressions import Agg, AggKind
-    # no dedicated last-value agg: reverse via max of monotonically
-    # increasing id is planner-side; approximate with any_value
-    return _e(x).any_value()
+    from . import w_last_value
+    return w_last_value(_e(x))
 
 
 def jq(x, filter_expr: str):
