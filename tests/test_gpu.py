@@ -528,3 +528,23 @@ def test_multi_agg_fused_kernel_matches_cpu():
                      .sort("g").to_pydict())
     assert q2(daft.from_pydict(d2, device="cuda:0")) == \
         q2(daft.from_pydict(d2, device="cpu"))
+
+
+def test_hip_extension_plugin_on_device(tmp_path):
+    """A hipcc-built plugin launches its own gfx950 kernel on the
+    engine's HBM-resident column buffers through the C ABI."""
+    import os
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(daft.__file__)))
+    src = os.path.join(repo, "examples", "ext_plugin", "hip_plugin.hip")
+    so = str(tmp_path / "hip_plugin.so")
+    subprocess.run(
+        ["hipcc", "--offload-arch=gfx950", "-O2", "-shared", "-fPIC",
+         "-I" + os.path.join(repo, "daft_amd", "ext"), src, "-o", so],
+        check=True, capture_output=True)
+    names = daft.load_extension(so)
+    assert "ext_saxpy" in names
+    df = daft.from_pydict({"x": [1.0, 2.5, -3.0]}, device="cuda:0")
+    out = df.select(daft.ext_function("ext_saxpy", col("x")).alias("y")) \
+        .to_pydict()["y"]
+    assert out == [3.0, 6.0, -5.0]
