@@ -449,3 +449,40 @@ def test_packed_two_key_join_matches():
     r2 = daft.from_pydict({"a": [1, 2], "b": [1, 3], "w": [10, 20]})
     out = l2.join(r2, on=["a", "b"]).to_pydict()
     assert out["w"] == [10]
+
+
+def test_streaming_sinks_multibatch(tmp_path):
+    """Sort / window / asof over multi-batch (multi-file) inputs match
+    the single-batch plan (blocking sinks materialize correctly)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    paths = []
+    for i in range(3):
+        p = str(tmp_path / f"f{i}.parquet")
+        pq.write_table(pa.table({
+            "g": [f"k{j % 4}" for j in range(40)],
+            "v": [float((i * 40 + j) % 23) for j in range(40)],
+            "t": [i * 40 + j for j in range(40)],
+        }), p)
+        paths.append(p)
+    multi = daft.read_parquet(paths)
+    single = daft.from_pydict(multi.to_pydict())
+
+    assert multi.sort(["v", "t"]).to_pydict() == \
+        single.sort(["v", "t"]).to_pydict()
+
+    from daft_amd.window import Window
+    from daft_amd.functions import row_number
+    w = Window().partition_by("g").order_by("t")
+    wm = multi.with_window_columns({"rn": row_number().over(w)}) \
+        .sort("t").to_pydict()
+    ws = single.with_window_columns({"rn": row_number().over(w)}) \
+        .sort("t").to_pydict()
+    assert wm == ws
+
+    right = daft.from_pydict({"t": [10, 50, 90], "mark": ["a", "b", "c"]})
+    am = multi.join_asof(right, left_on="t", right_on="t") \
+        .sort("t").to_pydict()
+    asg = single.join_asof(right, left_on="t", right_on="t") \
+        .sort("t").to_pydict()
+    assert am == asg
