@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import ctypes
 import os
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 

@@ -8,7 +8,7 @@ import math as _m
 import torch
 
 from ..expressions.expressions import Expression, ScalarFn, _to_node
-from ..schema import DataType, TypeKind
+from ..schema import DataType
 from ..series import Series
 
 

@@ -4,7 +4,7 @@ daft/functions modules; see each docstring)."""
 from __future__ import annotations
 
 import json as _json
-from typing import Any, List, Optional
+from typing import Optional
 
 import torch
 

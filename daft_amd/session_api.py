@@ -4,11 +4,10 @@ attach/create/list/drop helpers that operate on the ambient session —
 plus small config/value types: TimeUnit, ImageMode, IOConfig, range)."""
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional
 
-from .catalog import Catalog, Identifier, MemoryCatalog, Session, \
-    current_session
+from .catalog import Catalog, Session, current_session
 
 
 # -- small value/config types ------------------------------------------------
