@@ -21,6 +21,7 @@ KEYWORDS = {
     "substring", "for", "with", "count", "sum", "avg", "min", "max",
     "first", "last", "nulls", "semi", "anti", "true", "false", "cast",
     "over", "partition", "rows", "between", "unbounded", "preceding",
+    "intersect", "except",
     "current", "row", "following",
 }
 
@@ -112,6 +113,17 @@ class FuncCall:
     args: List[Any]
     distinct: bool = False
     star: bool = False
+
+
+@dataclass
+class SetOpStmt:
+    op: str                 # union | intersect | except
+    all: bool
+    left: Any
+    right: Any
+    order_by: Any = None
+    limit: Any = None
+    offset: int = 0
 
 
 @dataclass
@@ -286,6 +298,19 @@ class Parser:
     # -- entry -----------------------------------------------------------
     def parse_statement(self) -> SelectStmt:
         stmt = self.parse_select()
+        while self.at_kw("union", "intersect", "except"):
+            op = self.next().value.lower()
+            is_all = self.eat_kw("all")
+            right = self.parse_select()
+            # ORDER BY / LIMIT written after the set op parse into the
+            # right-hand select; they bind to the combined result
+            order_by, limit, offset = right.order_by, right.limit, \
+                getattr(right, "offset", 0)
+            right.order_by, right.limit = [], None
+            if hasattr(right, "offset"):
+                right.offset = 0
+            stmt = SetOpStmt(op, is_all, stmt, right, order_by, limit,
+                             offset)
         if self.peek().kind != "eof":
             raise SQLParseError(f"trailing tokens at {self.peek()}")
         return stmt

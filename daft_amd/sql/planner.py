@@ -60,7 +60,34 @@ class Binder:
 
 def plan_sql(query: str, lookup_table: Callable[[str], Any]):
     stmt = P.parse_sql(query)
-    return _plan_select(stmt, lookup_table, outer=None)
+    return _plan_stmt(stmt, lookup_table)
+
+
+def _plan_stmt(stmt, lookup):
+    if isinstance(stmt, P.SetOpStmt):
+        left = _plan_stmt(stmt.left, lookup)
+        right = _plan_stmt(stmt.right, lookup)
+        if stmt.op == "union":
+            df = left.concat(right) if stmt.all else left.union(right)
+        elif stmt.op == "intersect":
+            df = left.intersect_all(right) if stmt.all \
+                else left.intersect(right)
+        else:
+            df = left.except_all(right) if stmt.all \
+                else left.except_distinct(right)
+        if stmt.order_by:
+            by = [c.name for c in
+                  (o.expr for o in stmt.order_by)
+                  if hasattr(c, "name")]
+            desc = [o.desc for o in stmt.order_by]
+            if len(by) == len(stmt.order_by):
+                df = df.sort(by, desc=desc)
+        if stmt.limit is not None:
+            if stmt.offset:
+                df = df.offset(stmt.offset)
+            df = df.limit(stmt.limit)
+        return df
+    return _plan_select(stmt, lookup, outer=None)
 
 
 # ---------------------------------------------------------------------------

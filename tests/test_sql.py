@@ -143,3 +143,17 @@ def test_sql_window_functions():
                   "rows between 1 preceding and current row) as m "
                   "from df order by g, v").to_pydict()
     assert o2["m"] == [1.0, 4.0, 2.0, 10.0, 17.0]
+
+
+def test_sql_set_operations():
+    a = daft.from_pydict({"x": [1, 2, 3, 3]})
+    b = daft.from_pydict({"x": [2, 3, 4]})
+    q = lambda s: sorted(daft.sql(s).to_pydict()["x"])
+    assert q("select x from a union select x from b") == [1, 2, 3, 4]
+    assert q("select x from a union all select x from b") == \
+        [1, 2, 2, 3, 3, 3, 4]
+    assert q("select x from a intersect select x from b") == [2, 3]
+    assert q("select x from a except select x from b") == [1]
+    out = daft.sql("select x from a union select x from b "
+                   "order by x desc limit 2").to_pydict()["x"]
+    assert out == [4, 3]
