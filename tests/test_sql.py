@@ -148,12 +148,14 @@ def test_sql_window_functions():
 def test_sql_set_operations():
     a = daft.from_pydict({"x": [1, 2, 3, 3]})
     b = daft.from_pydict({"x": [2, 3, 4]})
-    q = lambda s: sorted(daft.sql(s).to_pydict()["x"])
-    assert q("select x from a union select x from b") == [1, 2, 3, 4]
-    assert q("select x from a union all select x from b") == \
-        [1, 2, 2, 3, 3, 3, 4]
-    assert q("select x from a intersect select x from b") == [2, 3]
-    assert q("select x from a except select x from b") == [1]
+    assert sorted(daft.sql("select x from a union select x from b")
+                  .to_pydict()["x"]) == [1, 2, 3, 4]
+    assert sorted(daft.sql("select x from a union all select x from b")
+                  .to_pydict()["x"]) == [1, 2, 2, 3, 3, 3, 4]
+    assert sorted(daft.sql("select x from a intersect select x from b")
+                  .to_pydict()["x"]) == [2, 3]
+    assert sorted(daft.sql("select x from a except select x from b")
+                  .to_pydict()["x"]) == [1]
     out = daft.sql("select x from a union select x from b "
                    "order by x desc limit 2").to_pydict()["x"]
     assert out == [4, 3]
