@@ -66,8 +66,10 @@ class TorchImageEmbedder:
     def __init__(self, dimensions: int = 512, image_size: int = 224,
                  device: Optional[str] = None, seed: int = 0):
         self.dimensions = dimensions
-        self.device = device or (
-            "cuda:0" if torch.cuda.is_available() else "cpu")
+        if device is None:
+            from ..context import get_context
+            device = str(get_context().device())
+        self.device = device
         dt = torch.bfloat16 if str(self.device).startswith("cuda") \
             else torch.float32
         torch.manual_seed(seed)
@@ -93,8 +95,10 @@ class TorchImageEmbedder:
 class TransformersTextEmbedder:
     def __init__(self, model_path: str, device: Optional[str] = None):
         from transformers import AutoModel, AutoTokenizer
-        self.device = device or (
-            "cuda:0" if torch.cuda.is_available() else "cpu")
+        if device is None:
+            from ..context import get_context
+            device = str(get_context().device())
+        self.device = device
         self.tokenizer = AutoTokenizer.from_pretrained(model_path)
         self.model = AutoModel.from_pretrained(model_path) \
             .to(self.device).eval()
