@@ -38,20 +38,17 @@ def list_agg_distinct(x):
     return _e(x).agg_list().list.distinct()
 
 
-def _list_host(name, fn, ret):
+def _list_host(name, fn, ret_dtype):
+    """Per-row host mapping over a list column with a fixed result
+    dtype."""
     def make(x, *args):
         def run(s: Series, *extra) -> Series:
             vals = s.cpu().to_pylist()
             out = [None if v is None else fn(v, *extra) for v in vals]
-            r = Series.from_pylist(s.name, out, ret(s) if callable(ret)
-                                   else ret)
+            r = Series.from_pylist(s.name, out, ret_dtype)
             return r.to(s.device) if s.is_gpu() else r
-        return Expression(ScalarFn(
-            name, run, [_to_node(x)],
-            (lambda f: ret(None, f)) if callable(ret) and
-            ret.__code__.co_argcount == 2 else
-            (ret if not callable(ret) else (lambda f: f[0].dtype)),
-            tuple(args)))
+        return Expression(ScalarFn(name, run, [_to_node(x)], ret_dtype,
+                                   tuple(args)))
     make.__name__ = name
     return make
 
@@ -174,15 +171,17 @@ def try_cast(x, dtype):
             return kernels.cast(s, dtype)
         except Exception:
             pass
-        vals = s.cpu().to_pylist()
+        # per-value salvage: keep the values that cast cleanly
         out = []
-        for v in vals:
+        for v in s.cpu().to_pylist():
+            if v is None:
+                out.append(None)
+                continue
             try:
-                out.append(None if v is None else v)
-                if v is not None:
-                    Series.from_pylist("t", [v], dtype)
+                Series.from_pylist("t", [v], dtype)
+                out.append(v)
             except Exception:
-                out[-1] = None
+                out.append(None)
         try:
             r = Series.from_pylist(s.name, out, dtype)
         except Exception:
