@@ -179,11 +179,12 @@ class ProjectOp(PhysicalOp):
         self.exprs = exprs
 
     def execute(self, ectx) -> BatchIter:
+        from .cse import evaluate_with_cse
         for rb in self.children[0].execute_tracked(ectx):
             cols = []
             n = len(rb)
-            for e in self.exprs:
-                s = e.evaluate(rb)
+            outs = evaluate_with_cse(self.exprs, rb)
+            for e, s in zip(self.exprs, outs):
                 if len(s) == 1 and n != 1:
                     s = s.broadcast(n)
                 cols.append(s.rename(e.to_field(rb.schema).name))

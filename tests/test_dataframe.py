@@ -486,3 +486,25 @@ def test_streaming_sinks_multibatch(tmp_path):
     asg = single.join_asof(right, left_on="t", right_on="t") \
         .sort("t").to_pydict()
     assert am == asg
+
+
+def test_projection_cse_evaluates_shared_subtree_once():
+    """Pure shared subexpressions in a projection evaluate once per
+    batch (physical/cse.py); results are unchanged."""
+    from daft_amd.expressions import expressions as E
+    calls = {"n": 0}
+    orig = E.BinaryOp.evaluate
+
+    def spy(self, batch):
+        if self.op == "add":
+            calls["n"] += 1
+        return orig(self, batch)
+    E.BinaryOp.evaluate = spy
+    try:
+        df = daft.from_pydict({"a": [1, 2], "b": [10, 20]})
+        out = df.select(((col("a") + col("b")) * 2).alias("x"),
+                        ((col("a") + col("b")) - 1).alias("y")).to_pydict()
+    finally:
+        E.BinaryOp.evaluate = orig
+    assert out == {"x": [22, 44], "y": [10, 21]}
+    assert calls["n"] == 1
