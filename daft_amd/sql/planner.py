@@ -554,7 +554,7 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
             if has_agg:
                 raise SQLPlanError("window functions with GROUP BY are "
                                    "not supported yet")
-            wname = it.alias or f"__w{i}"
+            wname = it.alias or it.expr.func.name
             df = df.with_window_columns(
                 {wname: _window_to_daft(it.expr, binder)})
             win_map[i] = wname
