@@ -281,8 +281,30 @@ def _bind_func(e: P.FuncCall, binder) -> Expression:
         for a in args[1:]:
             out = out.str.concat(a)
         return out
+    if name == "nullif":
+        a, b = args
+        return (a == b).if_else(lit(None), a)
+    if name in ("greatest", "least"):
+        out = args[0]
+        for a in args[1:]:
+            cmp = (a > out) if name == "greatest" else (a < out)
+            out = cmp.if_else(a, out)
+        return out
+    if name in ("if", "iff"):
+        return args[0].if_else(args[1], args[2])
+    if name == "ifnull" or name == "nvl":
+        return args[0].fill_null(args[1])
     if name in simple:
         return simple[name](*args)
+    # fall back to the free-function registry (daft.functions parity
+    # surface: sin/levenshtein_distance/to_snake_case/...)
+    from .. import functions as F
+    fn = getattr(F, name, None)
+    if fn is not None and callable(fn):
+        try:
+            return fn(*args)
+        except TypeError as te:
+            raise SQLPlanError(f"bad arguments for {name}(): {te}")
     raise SQLPlanError(f"unknown function {name}")
 
 

@@ -159,3 +159,22 @@ def test_sql_set_operations():
     out = daft.sql("select x from a union select x from b "
                    "order by x desc limit 2").to_pydict()["x"]
     assert out == [4, 3]
+
+
+def test_sql_scalar_function_registry():
+    """nullif/greatest/least/ifnull plus fallback into the 318-name
+    daft.functions registry (sin, levenshtein_distance, ...)."""
+    import math
+    df = daft.from_pydict({"a": [1.0, 2.0, 0.0], "b": [1.0, 5.0, 7.0]})
+    assert daft.sql("select nullif(a, b) as n from df") \
+        .to_pydict()["n"] == [None, 2, 0]
+    assert daft.sql("select greatest(a, b) as g from df") \
+        .to_pydict()["g"] == [1, 5, 7]
+    assert daft.sql("select least(a, b) as l from df") \
+        .to_pydict()["l"] == [1, 2, 0]
+    assert daft.sql("select ifnull(nullif(a, b), 99) as f from df") \
+        .to_pydict()["f"] == [99, 2, 0]
+    out = daft.sql("select sin(a) as s from df").to_pydict()["s"]
+    assert abs(out[1] - math.sin(2)) < 1e-12
+    assert daft.sql("select levenshtein_distance('kitten', 'sitting') "
+                    "as d from df limit 1").to_pydict()["d"] == [3]
