@@ -55,7 +55,13 @@ def sql(query: str, catalog: Optional[SQLCatalog] = None, **kwargs):
         frame_vars.update(frame.f_globals)
         frame_vars.update(frame.f_locals)
     from .planner import plan_sql
-    return plan_sql(query, lambda n: _lookup(n, catalog, frame_vars))
+    q = query.strip()
+    if q.lower().startswith("explain "):
+        df = plan_sql(q[8:], lambda n: _lookup(n, catalog, frame_vars))
+        text = df._builder.optimize().explain()
+        from ..io import from_pydict
+        return from_pydict({"plan": text.splitlines()})
+    return plan_sql(q, lambda n: _lookup(n, catalog, frame_vars))
 
 
 def sql_expr(text: str):
