@@ -508,3 +508,29 @@ def test_projection_cse_evaluates_shared_subtree_once():
         E.BinaryOp.evaluate = orig
     assert out == {"x": [22, 44], "y": [10, 21]}
     assert calls["n"] == 1
+
+
+@pytest.mark.parametrize("nparts", [1, 2, 5])
+def test_partition_sweep_invariance(nparts):
+    """Results are invariant to the partition count (the reference's
+    conftest partitioning sweep)."""
+    data = {"g": [f"k{i % 4}" for i in range(97)],
+            "v": [float(i % 13) for i in range(97)],
+            "k": [i % 7 for i in range(97)]}
+    base = daft.from_pydict(data)
+    df = base.into_partitions(nparts) if nparts > 1 else base
+
+    want_agg = base.groupby("g").agg(col("v").sum().alias("s")) \
+        .sort("g").to_pydict()
+    assert df.groupby("g").agg(col("v").sum().alias("s")) \
+        .sort("g").to_pydict() == want_agg
+
+    dim = daft.from_pydict({"k": list(range(7)),
+                            "w": [i * 10 for i in range(7)]})
+    want_j = base.join(dim, on="k").sort(["v", "k", "w"]).to_pydict()
+    assert df.join(dim, on="k").sort(["v", "k", "w"]).to_pydict() == want_j
+
+    assert df.sort("v").to_pydict()["v"] == \
+        base.sort("v").to_pydict()["v"]
+    assert df.distinct("g").count_rows() == 4
+    assert df.count_rows() == 97
