@@ -178,3 +178,13 @@ def test_sql_scalar_function_registry():
     assert abs(out[1] - math.sin(2)) < 1e-12
     assert daft.sql("select levenshtein_distance('kitten', 'sitting') "
                     "as d from df limit 1").to_pydict()["d"] == [3]
+
+
+def test_sql_catalog_and_explain():
+    from daft_amd.sql import SQLCatalog
+    cat = SQLCatalog({"t": daft.from_pydict({"a": [1, 2]})})
+    assert daft.sql("select sum(a) as s from t",
+                    catalog=cat).to_pydict()["s"] == [3]
+    plan = daft.sql("explain select a from t where a > 1",
+                    catalog=cat).to_pydict()["plan"]
+    assert any("Filter" in l for l in plan)
