@@ -534,3 +534,18 @@ def test_partition_sweep_invariance(nparts):
         base.sort("v").to_pydict()["v"]
     assert df.distinct("g").count_rows() == 4
     assert df.count_rows() == 97
+
+
+def test_empty_input_sweep_new_surfaces():
+    """Zero-row inputs through the newer surfaces (describe, set ops,
+    union_by_name, SQL set ops, agg_set)."""
+    e = daft.from_pydict({"a": [], "g": []})
+    assert e.describe().to_pydict()["count"] == [0, 0]
+    assert e.drop_null("a").count_rows() == 0
+    assert e.groupby("g").agg(col("a").sum().alias("s")).count_rows() == 0
+    assert e.intersect_all(e).count_rows() == 0
+    assert e.except_all(e).count_rows() == 0
+    u = e.union_all_by_name(daft.from_pydict({"a": [1], "z": ["x"]}))
+    assert u.count_rows() == 1
+    assert daft.sql("select a from e union select a from e") \
+        .to_pydict()["a"] == []
