@@ -548,3 +548,19 @@ def test_hip_extension_plugin_on_device(tmp_path):
     out = df.select(daft.ext_function("ext_saxpy", col("x")).alias("y")) \
         .to_pydict()["y"]
     assert out == [3.0, 6.0, -5.0]
+
+
+def test_sql_windows_and_setops_on_device():
+    """SQL windows/set-ops execute on device-resident tables (the HIP
+    sort/groupby kernels under the window machinery)."""
+    a = daft.from_pydict({"g": ["a", "a", "b", "b", "b"] * 200,
+                          "v": list(range(1000))}, device="cuda:0")
+    o = daft.sql("select g, v, row_number() over (partition by g "
+                 "order by v) as rn, sum(v) over (partition by g) as t "
+                 "from a order by g, v limit 3").to_pydict()
+    assert o["rn"] == [1, 2, 3]
+    b = daft.from_pydict({"v": list(range(500, 1500))}, device="cuda:0")
+    u = daft.sql("select v from a union select v from b").count_rows()
+    assert u == 1500
+    i = daft.sql("select v from a intersect select v from b").count_rows()
+    assert i == 500
