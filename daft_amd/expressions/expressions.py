@@ -104,7 +104,9 @@ class Literal(ExprNode):
                           device=batch.device)
 
     def __repr__(self):
-        return f"lit({self.value!r})"
+        # dtype is part of node identity: CSE/agg-decomposition key subtrees
+        # by repr, and lit(30):int32 must not merge with lit(30):int64
+        return f"lit({self.value!r}:{self.dtype!r})"
 
 
 class Alias(ExprNode):

@@ -111,21 +111,19 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
         if e.kind == "agg":
             a: Agg = e.inner
             frame = getattr(e.spec, "frame", None) if e.spec else None
-            if frame is not None and sorted_pos is not None and a.kind in (
-                    AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL,
-                    AggKind.MEAN):
-                out_cols.append(_rows_frame_agg(batch, a, nm, gids,
-                                                sorted_pos, frame))
-                continue
-            if sorted_pos is not None and a.kind in (
-                    AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL,
-                    AggKind.MEAN):
-                # SQL RANGE-default running aggregate: with an ORDER BY the
-                # frame is unbounded-preceding .. current row (peers share)
+            if sorted_pos is not None:
+                # With an ORDER BY the frame is the explicit ROWS frame or
+                # the SQL RANGE default (unbounded preceding .. current row,
+                # peers share) — never the whole partition.
                 # (ref: window_partition_and_order_by.rs incremental states)
-                out_cols.append(_running_agg(batch, a, nm, gids, order_by,
-                                             sorted_pos))
-                continue
+                if a.kind in _FRAMEABLE:
+                    out_cols.append(_framed_agg(batch, a, nm, gids,
+                                                sorted_pos, order_by, frame))
+                    continue
+                raise NotImplementedError(
+                    f"window aggregate {a.kind!r} with ORDER BY / frame is "
+                    f"not supported (supported: sum/count/mean/min/max/"
+                    f"stddev/variance)")
             per_group = agg_mod.compute_agg(batch, gids, num_groups, nm, a)
             out_cols.append(per_group.take(gids).rename(nm))
         elif e.kind in ("row_number", "rank", "dense_rank"):
@@ -170,159 +168,197 @@ def run_window(batch: RecordBatch, window_exprs: List[ExprNode],
                                   torch.full_like(src_row, -1))
             shifted = vals.take(src_idx).rename(nm)
             if e.default is not None:
+                # SQL applies the default only when the offset falls outside
+                # the partition — genuinely-NULL source values stay NULL.
                 fill = Series.from_pylist(nm, [e.default],
                                           device=dev).broadcast(n)
-                shifted = shifted.fill_null(fill)
+                mask = Series("__m", DataType.bool(), data=in_bounds)
+                shifted = mask.if_else(shifted, fill).rename(nm)
             out_cols.append(shifted)
         else:
             raise ValueError(f"unknown window fn {e.kind}")
     return RecordBatch(out_cols, num_rows=n)
 
 
-def _running_agg(batch: RecordBatch, a: Agg, name: str, gids: torch.Tensor,
-                 order_by, sorted_pos) -> Series:
-    """Running sum/count/mean per partition in sorted order, with SQL RANGE
-    peer sharing (rows with equal order keys take the frame value at the
-    last peer)."""
+_FRAMEABLE = {AggKind.SUM, AggKind.COUNT, AggKind.COUNT_ALL, AggKind.MEAN,
+              AggKind.MIN, AggKind.MAX, AggKind.STDDEV, AggKind.VARIANCE}
+
+
+def _framed_agg(batch: RecordBatch, a: Agg, name: str, gids: torch.Tensor,
+                sorted_pos, order_by, frame) -> Series:
+    """Windowed aggregate over an explicit ROWS frame, or the SQL
+    RANGE-default frame (unbounded preceding .. current row, peers share)
+    when `frame` is None.  sum/count/mean/stddev/variance run on per-
+    partition prefix sums; min/max on a sparse-table range query
+    (ref: window_partition_and_dynamic_frame.rs / window_states/minmax.rs).
+    All positions are in sorted space; results scatter back through perm."""
     perm, pos, part_start = sorted_pos
     n = gids.shape[0]
     dev = gids.device
-    if a.kind == AggKind.COUNT_ALL or a.child is None:
-        vdata = torch.ones(n, dtype=torch.float64, device=dev)
-        vvalid = None
-    else:
-        values = a.child.evaluate(batch)
-        if len(values) == 1 and n > 1:
-            values = values.broadcast(n)
-        if a.kind == AggKind.COUNT:
-            vdata = torch.ones(n, dtype=torch.float64, device=dev)
-            vvalid = values.validity
-        else:
-            vdata = values.data.to(torch.float64)
-            vvalid = values.validity
-    if vvalid is not None:
-        vdata = torch.where(vvalid, vdata, torch.zeros_like(vdata))
-
-    v_sorted = vdata[perm]
-    csum = torch.cumsum(v_sorted, 0)
     g_sorted = gids[perm]
-    base_idx = part_start[g_sorted]
-    base = csum[base_idx] - v_sorted[base_idx]
-    running_sorted = csum - base
-
-    # peer sharing: propagate the value at each peer-group's LAST row
-    # backward to all peers — take the running value at the last position
-    # with the same (partition, order-keys)
-    okeys = [k.evaluate(batch) for k in order_by]
-    sap = _same_as_prev(okeys, gids, perm, pos)[perm]  # sorted order
-    # last index of each peer run: positions where the NEXT row starts a new
-    # peer group; compute via reversed cummax of position at run ends
-    idx = torch.arange(n, dtype=torch.int64, device=dev)
-    next_new = torch.ones(n, dtype=torch.bool, device=dev)
-    next_new[:-1] = ~sap[1:]
-    # nearest run-end at-or-after each row: suffix-min over end positions
-    run_end = torch.where(next_new, idx, torch.full_like(idx, n))
-    run_end = torch.flip(torch.cummin(torch.flip(run_end, [0]), 0).values,
-                         [0])
-    shared_sorted = running_sorted[run_end]
-
-    out_sorted = shared_sorted
-    if a.kind == AggKind.MEAN:
-        ones = torch.ones(n, dtype=torch.float64, device=dev)
-        if vvalid is not None:
-            ones = torch.where(vvalid, ones, torch.zeros_like(ones))
-        c_sorted = ones[perm]
-        ccsum = torch.cumsum(c_sorted, 0)
-        cbase = ccsum[base_idx] - c_sorted[base_idx]
-        cnt_run = (ccsum - cbase)[run_end]
-        out_sorted = shared_sorted / cnt_run.clamp(min=1.0)
-
-    out = torch.empty(n, dtype=torch.float64, device=dev)
-    out[perm] = out_sorted
-    if a.kind in (AggKind.COUNT, AggKind.COUNT_ALL):
-        return Series(name, DataType.uint64(),
-                      data=out.to(torch.int64).view(torch.uint64))
-    from ..schema import Field
-    out_dt = a.to_field(batch.schema).dtype
-    return Series(name, out_dt, data=out.to(out_dt.to_torch()))
-
-
-def _rows_frame_agg(batch: RecordBatch, a: Agg, name: str,
-                    gids: torch.Tensor, sorted_pos, frame) -> Series:
-    """ROWS BETWEEN start AND end frames for sum/count/mean via per-
-    partition prefix sums (ref: window_partition_and_dynamic_frame.rs)."""
-    perm, pos, part_start = sorted_pos
-    n = gids.shape[0]
-    dev = gids.device
-    if a.kind == AggKind.COUNT_ALL or a.child is None:
-        vdata = torch.ones(n, dtype=torch.float64, device=dev)
-        vvalid = None
-    else:
-        values = a.child.evaluate(batch)
-        if len(values) == 1 and n > 1:
-            values = values.broadcast(n)
-        vdata = torch.ones(n, dtype=torch.float64, device=dev) \
-            if a.kind == AggKind.COUNT else values.data.to(torch.float64)
-        vvalid = values.validity
-    if vvalid is not None:
-        vdata = torch.where(vvalid, vdata, torch.zeros_like(vdata))
-
-    v_sorted = vdata[perm]
-    prefix = torch.cumsum(v_sorted, 0)
-    g_sorted = gids[perm]
-    counts = torch.bincount(g_sorted, minlength=int(part_start.shape[0]))
     ps = part_start[g_sorted]                     # partition start (sorted)
+    counts = torch.bincount(g_sorted, minlength=int(part_start.shape[0]))
     pe = ps + counts[g_sorted] - 1                # partition end (inclusive)
     idx = torch.arange(n, dtype=torch.int64, device=dev)
 
-    def bound(spec, default):
+    if frame is not None:
         from ..window import Window as W
-        if spec == W.unbounded_preceding:
-            return ps
-        if spec == W.unbounded_following:
-            return pe
-        if spec == W.current_row:
-            return idx
-        return idx + int(spec)
-    start, end = frame
-    lo = torch.maximum(bound(start, ps), ps)
-    hi = torch.minimum(bound(end, pe), pe)
+
+        def bound(spec):
+            if spec == W.unbounded_preceding:
+                return ps
+            if spec == W.unbounded_following:
+                return pe
+            if spec == W.current_row:
+                return idx
+            return idx + int(spec)
+        start, end = frame
+        lo = torch.maximum(bound(start), ps)
+        hi = torch.minimum(bound(end), pe)
+    else:
+        # RANGE default: frame end is the LAST peer (equal order keys)
+        okeys = [k.evaluate(batch) for k in order_by]
+        sap = _same_as_prev(okeys, gids, perm, pos)[perm]
+        next_new = torch.ones(n, dtype=torch.bool, device=dev)
+        next_new[:-1] = ~sap[1:]
+        run_end = torch.where(next_new, idx, torch.full_like(idx, n))
+        run_end = torch.flip(
+            torch.cummin(torch.flip(run_end, [0]), 0).values, [0])
+        lo, hi = ps, run_end
     empty = lo > hi
-    lo_c = lo.clamp(0, n - 1)
-    hi_c = hi.clamp(0, n - 1)
-    upper = prefix[hi_c]
-    lower = torch.where(lo_c > 0, prefix[(lo_c - 1).clamp(min=0)],
-                        torch.zeros_like(upper))
-    lower = torch.where(lo_c == 0, torch.zeros_like(lower), lower)
-    out_sorted = torch.where(empty, torch.zeros_like(upper), upper - lower)
-    if a.kind == AggKind.MEAN:
-        if vvalid is not None:
-            ones = torch.where(vvalid, torch.ones(n, dtype=torch.float64,
-                                                  device=dev),
-                               torch.zeros(n, dtype=torch.float64,
-                                           device=dev))[perm]
-            cpre = torch.cumsum(ones, 0)
-            cupper = cpre[hi_c]
-            clower = torch.where(lo_c == 0, torch.zeros_like(cupper),
-                                 cpre[(lo_c - 1).clamp(min=0)])
-            nrows = (cupper - clower).clamp(min=1.0)
-        else:
-            nrows = (hi - lo + 1).clamp(min=1).to(torch.float64)
-        out_sorted = out_sorted / nrows
+    lo_c = lo.clamp(0, max(n - 1, 0))
+    hi_c = hi.clamp(0, max(n - 1, 0))
+
+    # values in sorted order
+    if a.kind == AggKind.COUNT_ALL or a.child is None:
+        vdata = torch.ones(n, dtype=torch.float64, device=dev)
+        vvalid = None
+    else:
+        values = a.child.evaluate(batch)
+        if len(values) == 1 and n > 1:
+            values = values.broadcast(n)
+        if values.is_dict():
+            values = values.dict_decode()
+        vdata = values.data
+        vvalid = values.validity
+    valid_sorted = vvalid[perm] if vvalid is not None else None
+
+    def prefix(t):
+        return torch.cumsum(t, 0)
+
+    def rsum(pre):
+        upper = pre[hi_c]
+        lower = torch.where(lo_c > 0, pre[(lo_c - 1).clamp(min=0)],
+                            torch.zeros_like(upper))
+        out = upper - lower
+        return torch.where(empty, torch.zeros_like(out), out)
+
+    ones = torch.ones(n, dtype=torch.float64, device=dev)
+    if valid_sorted is not None:
+        ones = torch.where(valid_sorted, ones, torch.zeros_like(ones))
+    rcnt = rsum(prefix(ones))
+
+    if a.kind in (AggKind.COUNT, AggKind.COUNT_ALL):
+        out_sorted = rcnt
+        validity_sorted = None           # COUNT of an empty frame is 0
+    elif a.kind in (AggKind.MIN, AggKind.MAX):
+        if vdata is None or vdata.dtype in (torch.uint8,):
+            raise NotImplementedError(
+                f"windowed min/max over dtype is not supported")
+        out_sorted = _range_minmax(vdata[perm], valid_sorted, lo_c, hi_c,
+                                   is_max=a.kind == AggKind.MAX)
+        validity_sorted = (rcnt > 0) & ~empty
+        out = torch.empty(n, dtype=out_sorted.dtype, device=dev)
+        out[perm] = out_sorted
+        v = torch.empty(n, dtype=torch.bool, device=dev)
+        v[perm] = validity_sorted
+        out_dt = a.to_field(batch.schema).dtype
+        if out.dtype != out_dt.to_torch():
+            out = out.view(out_dt.to_torch()) if                 out.element_size() == out_dt.to_torch().itemsize and                 not out.dtype.is_floating_point else out.to(out_dt.to_torch())
+        return Series(name, out_dt, data=out,
+                      validity=None if bool(v.all().item()) else v)
+    else:
+        vs = vdata.to(torch.float64)[perm]
+        if valid_sorted is not None:
+            vs = torch.where(valid_sorted, vs, torch.zeros_like(vs))
+        s = rsum(prefix(vs))
+        if a.kind == AggKind.SUM:
+            out_sorted = s
+            validity_sorted = (rcnt > 0) & ~empty
+        elif a.kind == AggKind.MEAN:
+            out_sorted = s / rcnt.clamp(min=1.0)
+            validity_sorted = (rcnt > 0) & ~empty
+        else:  # STDDEV / VARIANCE (population, matching physical/agg.py)
+            s2 = rsum(prefix(vs * vs))
+            c = rcnt.clamp(min=1.0)
+            mean = s / c
+            var = (s2 / c - mean * mean).clamp(min=0.0)
+            out_sorted = torch.sqrt(var) if a.kind == AggKind.STDDEV else var
+            validity_sorted = (rcnt > 0) & ~empty
+
     out = torch.empty(n, dtype=torch.float64, device=dev)
     out[perm] = out_sorted
     validity = None
-    if bool(empty.any().item()):
+    if validity_sorted is not None and not bool(validity_sorted.all().item()):
         v = torch.empty(n, dtype=torch.bool, device=dev)
-        v[perm] = ~empty
+        v[perm] = validity_sorted
         validity = v
     if a.kind in (AggKind.COUNT, AggKind.COUNT_ALL):
         return Series(name, DataType.uint64(),
                       data=out.to(torch.int64).view(torch.uint64),
                       validity=validity)
     out_dt = a.to_field(batch.schema).dtype
+    if a.kind in (AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE):
+        out_dt = DataType.float64()
     return Series(name, out_dt, data=out.to(out_dt.to_torch()),
                   validity=validity)
+
+
+def _range_minmax(v_sorted: torch.Tensor, valid_sorted, lo: torch.Tensor,
+                  hi: torch.Tensor, is_max: bool) -> torch.Tensor:
+    """Range min/max over [lo, hi] (inclusive, sorted space) via an
+    O(n log n) sparse table; nulls excluded through sentinels."""
+    n = v_sorted.shape[0]
+    dt = v_sorted.dtype
+    if dt == torch.bool:
+        v_sorted = v_sorted.to(torch.int8)
+        dt = torch.int8
+    unsigned_view = None
+    if dt in (torch.uint16, torch.uint32, torch.uint64):
+        # compare through the signed view (exact for values < 2^(w-1))
+        unsigned_view = dt
+        signed = {torch.uint16: torch.int16, torch.uint32: torch.int32,
+                  torch.uint64: torch.int64}[dt]
+        v_sorted = v_sorted.view(signed)
+        dt = signed
+    if dt.is_floating_point:
+        sent = float("-inf") if is_max else float("inf")
+    else:
+        ii = torch.iinfo(dt)
+        sent = ii.min if is_max else ii.max
+    v = v_sorted
+    if valid_sorted is not None:
+        v = torch.where(valid_sorted, v, torch.full_like(v, sent))
+    op = torch.maximum if is_max else torch.minimum
+    levels = [v]
+    j = 1
+    while (1 << j) <= n:
+        prev = levels[-1]
+        half = 1 << (j - 1)
+        m = n - (1 << j) + 1
+        cur = torch.full_like(v, sent)
+        cur[:m] = op(prev[:m], prev[half:half + m])
+        levels.append(cur)
+        j += 1
+    st = torch.stack(levels)                       # (J, n)
+    length = (hi - lo + 1).clamp(min=1)
+    jj = torch.floor(torch.log2(length.to(torch.float64))).to(torch.int64)
+    jj = jj.clamp(0, len(levels) - 1)
+    pow2 = torch.bitwise_left_shift(torch.ones_like(jj), jj)
+    res = op(st[jj, lo], st[jj, (hi - pow2 + 1).clamp(min=0)])
+    if unsigned_view is not None:
+        res = res.view(unsigned_view)
+    return res
 
 
 def _partition_starts(sorted_gids: torch.Tensor,
@@ -345,12 +381,16 @@ def _same_as_prev(okeys: List[Series], gids: torch.Tensor,
     prev_row = perm[prev_sorted]
     same = torch.ones(n, dtype=torch.bool, device=dev)
     for k in okeys:
-        eq = k.compare(k.take(prev_row), "eq")
+        prev_k = k.take(prev_row)
+        eq = k.compare(prev_k, "eq")
         e = eq.data.clone()
         if eq.validity is not None:
-            both_null = ~(k.validity if k.validity is not None else
-                          torch.ones(n, dtype=torch.bool, device=dev))
-            e = torch.where(eq.validity, e, both_null)
+            # only null-vs-null counts as a peer: both sides must be null
+            cur_null = ~k.validity if k.validity is not None else \
+                torch.zeros(n, dtype=torch.bool, device=dev)
+            prev_null = ~prev_k.validity if prev_k.validity is not None else \
+                torch.zeros(n, dtype=torch.bool, device=dev)
+            e = torch.where(eq.validity, e, cur_null & prev_null)
         same &= e
     same &= gids[prev_row] == gids
     same &= pos > 0

@@ -297,23 +297,49 @@ class Parser:
 
     # -- entry -----------------------------------------------------------
     def parse_statement(self) -> SelectStmt:
-        stmt = self.parse_select()
-        while self.at_kw("union", "intersect", "except"):
-            op = self.next().value.lower()
-            is_all = self.eat_kw("all")
-            right = self.parse_select()
-            # ORDER BY / LIMIT written after the set op parse into the
-            # right-hand select; they bind to the combined result
-            order_by, limit, offset = right.order_by, right.limit, \
-                getattr(right, "offset", 0)
-            right.order_by, right.limit = [], None
-            if hasattr(right, "offset"):
-                right.offset = 0
-            stmt = SetOpStmt(op, is_all, stmt, right, order_by, limit,
-                             offset)
+        stmt = self._parse_set_expr()
         if self.peek().kind != "eof":
             raise SQLParseError(f"trailing tokens at {self.peek()}")
         return stmt
+
+    def _combine_setop(self, op: str, is_all: bool, left, right):
+        # ORDER BY / LIMIT written after the set op parse into the
+        # right-hand select; they bind to the combined result
+        order_by, limit, offset = right.order_by, right.limit, \
+            getattr(right, "offset", 0)
+        right.order_by, right.limit = [], None
+        if hasattr(right, "offset"):
+            right.offset = 0
+        return SetOpStmt(op, is_all, left, right, order_by, limit, offset)
+
+    def _parse_set_expr(self) -> SelectStmt:
+        # standard SQL precedence: INTERSECT binds tighter than UNION/EXCEPT
+        stmt = self._parse_intersect_expr()
+        while self.at_kw("union", "except"):
+            op = self.next().value.lower()
+            is_all = self.eat_kw("all")
+            right = self._parse_intersect_expr()
+            stmt = self._combine_setop(op, is_all, stmt, right)
+        return stmt
+
+    def _parse_intersect_expr(self) -> SelectStmt:
+        stmt = self._parse_set_operand()
+        while self.at_kw("intersect"):
+            self.next()
+            is_all = self.eat_kw("all")
+            right = self._parse_set_operand()
+            stmt = self._combine_setop("intersect", is_all, stmt, right)
+        return stmt
+
+    def _parse_set_operand(self) -> SelectStmt:
+        # parenthesized query: "(" SELECT/WITH ... ")"
+        if self.at_op("(") and self.peek(1).kind == "kw" and \
+                self.peek(1).value in ("select", "with"):
+            self.next()
+            inner = self._parse_set_expr()
+            self.expect_op(")")
+            return inner
+        return self.parse_select()
 
     def parse_select(self) -> SelectStmt:
         s = SelectStmt()

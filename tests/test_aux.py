@@ -709,3 +709,105 @@ def test_session_api_top_level():
         assert get_context().execution_config.morsel_size_rows == 7
     with pytest.raises(RuntimeError):
         d.set_runner_ray()
+
+
+def test_running_min_max_over_order():
+    # MIN/MAX with ORDER BY must be RUNNING min/max,
+    # not whole-partition
+    df = daft.from_pydict({"g": ["a"] * 4 + ["b"] * 2,
+                           "t": [1, 2, 3, 4, 1, 2],
+                           "v": [30.0, 10.0, 20.0, 5.0, 7.0, 3.0]})
+    from daft_amd.window import Window
+    w = Window().partition_by("g").order_by("t")
+    out = df.with_window_columns({
+        "rmin": col("v").min().over(w),
+        "rmax": col("v").max().over(w),
+    }).sort(["g", "t"]).to_pydict()
+    assert out["rmin"] == [30.0, 10.0, 10.0, 5.0, 7.0, 3.0]
+    assert out["rmax"] == [30.0, 30.0, 30.0, 30.0, 7.0, 7.0]
+
+
+def test_running_min_max_int_and_frames():
+    df = daft.from_pydict({"t": [1, 2, 3, 4, 5],
+                           "v": [5, 1, 4, 2, 3]})
+    from daft_amd.window import Window
+    w = Window().order_by("t").rows_between(-1, 1)
+    out = df.with_window_columns({
+        "fmin": col("v").min().over(w),
+        "fmax": col("v").max().over(w),
+    }).sort("t").to_pydict()
+    assert out["fmin"] == [1, 1, 1, 2, 2]
+    assert out["fmax"] == [5, 5, 4, 4, 3]
+
+
+def test_running_stddev_variance():
+    import statistics
+    df = daft.from_pydict({"t": [1, 2, 3], "v": [1.0, 3.0, 5.0]})
+    from daft_amd.window import Window
+    w = Window().order_by("t")
+    out = df.with_window_columns({
+        "rv": col("v").var().over(w) if hasattr(col("v"), "var")
+        else col("v").stddev().over(w),
+    }).sort("t").to_pydict() if False else None
+    out = df.with_window_columns({
+        "rs": col("v").stddev().over(w),
+    }).sort("t").to_pydict()
+    # population stddev of prefixes: [1], [1,3], [1,3,5]
+    assert abs(out["rs"][0] - 0.0) < 1e-9
+    assert abs(out["rs"][1] - 1.0) < 1e-9
+    assert abs(out["rs"][2] - statistics.pstdev([1.0, 3.0, 5.0])) < 1e-9
+
+
+def test_rank_with_null_order_keys():
+    # a null key after a non-null row is NOT a peer
+    from daft_amd.functions import rank
+    from daft_amd.window import Window
+    df = daft.from_pydict({"t": [1, None, None, 2]})
+    w = Window().order_by("t")  # nulls last by default
+    out = df.with_window_columns({"rk": rank().over(w)}).to_pydict()
+    by_t = {}
+    rows = sorted(zip(out["t"], out["rk"]),
+                  key=lambda r: (r[0] is None, r[0] if r[0] is not None else 0))
+    # t=1 -> rank 1, t=2 -> rank 2, nulls are peers of each other -> rank 3
+    vals = [r[1] for r in rows]
+    assert vals == [1, 2, 3, 3]
+
+
+def test_lag_default_keeps_genuine_nulls():
+    # default fills only out-of-partition offsets
+    df = daft.from_pydict({"t": [1, 2, 3], "v": [10, None, 30]})
+    from daft_amd.window import Window
+    w = Window().order_by("t")
+    out = df.with_window_columns({
+        "prev": col("v").lag(1, default=-1).over(w),
+    }).sort("t").to_pydict()
+    # row t=3's lag is the genuinely-NULL v at t=2: stays NULL
+    assert out["prev"] == [-1, 10, None]
+
+
+def test_intersect_precedence():
+    # INTERSECT binds tighter than UNION
+    import daft_amd as daft2
+    a = daft.from_pydict({"x": [1, 2]})
+    b = daft.from_pydict({"x": [2, 3]})
+    c = daft.from_pydict({"x": [3, 4]})
+    from daft_amd.session_api import Session
+    s = Session()
+    s.create_temp_table("a", a)
+    s.create_temp_table("b", b)
+    s.create_temp_table("c", c)
+    out = s.sql("SELECT x FROM a UNION (SELECT x FROM b INTERSECT SELECT x FROM c)") \
+        .to_pydict()
+    expected = sorted([1, 2, 3])
+    out2 = s.sql("SELECT x FROM a UNION SELECT x FROM b INTERSECT SELECT x FROM c") \
+        .to_pydict()
+    # A UNION (B ∩ C) = {1,2} ∪ {3} = {1,2,3}; the flat-left-assoc reading
+    # (A ∪ B) ∩ C would give {3}
+    assert sorted(out2["x"]) == expected
+
+
+def test_literal_repr_includes_dtype():
+    from daft_amd.expressions.expressions import Literal
+    a = repr(Literal(30, DataType.int32()))
+    b = repr(Literal(30, DataType.int64()))
+    assert a != b
