@@ -40,6 +40,10 @@ class ExecutionConfig:
     csv_target_filesize: int = 512 * 1024 * 1024
     maintain_order: bool = True
     memory_limit_bytes: Optional[int] = None   # DAFT_MEMORY_LIMIT analog
+    # spill tier for RCCL exchanges: when an all-to-all's working set
+    # (send+recv) would exceed this, the exchange runs in row chunks with
+    # received chunks staged to host (None = 50% of free HBM at call time)
+    exchange_hbm_budget_bytes: Optional[int] = None
     enable_gpu: bool = True
 
 

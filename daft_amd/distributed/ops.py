@@ -14,6 +14,32 @@ from ..series import Series
 from . import comm
 
 
+def _exchange_budget(ectx) -> "int | None":
+    """HBM budget for one exchange (send+recv working set); over it, the
+    exchange chunks rows and stages receives to host (spill tier)."""
+    cfg = getattr(ectx.ctx, "execution_config", None)
+    explicit = getattr(cfg, "exchange_hbm_budget_bytes", None) if cfg else None
+    if explicit is not None:
+        return explicit
+    if str(ectx.device).startswith("cuda") and torch.cuda.is_available():
+        free, _total = torch.cuda.mem_get_info(ectx.device)
+        return free // 2
+    return None
+
+
+def _yield_exchanged(out: RecordBatch, ectx):
+    """Yield an exchange result; a spilled (host-staged) result streams back
+    to the device in morsel slices so downstream operators fold partial
+    states instead of re-materializing the whole partition in HBM."""
+    if str(out.device) == "cpu" and str(ectx.device).startswith("cuda"):
+        morsel = getattr(ectx.ctx.execution_config, "stream_morsel_rows",
+                         1 << 26)
+        for lo in range(0, max(len(out), 1), morsel):
+            yield out.slice(lo, lo + morsel).to(ectx.device)
+        return
+    yield out
+
+
 def _normalize_key(s: Series) -> Series:
     """Cast keys to width-stable dtypes so both sides of a co-partition hash
     identically (int->int64, float->float64, bool->int64)."""
@@ -56,7 +82,9 @@ class ExchangeByKeyOp(PhysicalOp):
             for c in counts.tolist():
                 parts.append(reordered.slice(start, start + c))
                 start += c
-        yield comm.exchange_batches(parts)
+        out = comm.exchange_batches(parts,
+                                    hbm_budget=_exchange_budget(ectx))
+        yield from _yield_exchanged(out, ectx)
 
 
 class GatherToRank0Op(PhysicalOp):
@@ -137,7 +165,9 @@ class RangeExchangeOp(PhysicalOp):
         for c in counts.tolist():
             parts.append(reordered.slice(start, start + c))
             start += c
-        yield comm.exchange_batches(parts)
+        out = comm.exchange_batches(parts,
+                                    hbm_budget=_exchange_budget(ectx))
+        yield from _yield_exchanged(out, ectx)
 
 
 class Rank0OnlyOp(PhysicalOp):

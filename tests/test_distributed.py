@@ -245,3 +245,110 @@ def test_distributed_world4_matches_single():
                                         abs_tol=1e-6), f"q{qi}"
                 else:
                     assert gx == wx, f"q{qi}"
+
+
+def _exchange_protocol_checks(rank, world):
+    """Direct exercise of the packed tensor all-to-all: mixed dtypes,
+    strings, nulls, dict columns, empty partitions, skew, and the chunked
+    (spill) path."""
+    import daft_amd as daft
+    from daft_amd.distributed import comm
+    from daft_amd.recordbatch import RecordBatch
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+
+    # each rank builds w parts; part p carries rows tagged (rank, p, i)
+    w = world
+    parts = []
+    for p in range(w):
+        n = 3 + ((rank + p) % 2)          # ragged sizes
+        ints = [rank * 1000 + p * 10 + i for i in range(n)]
+        strs = [f"r{rank}p{p}i{i}" if i % 3 else None for i in range(n)]
+        fls = [float(i) + rank for i in range(n)]
+        rb = daft.from_pydict({"i": ints, "s": strs, "f": fls}).collect()._result[0]
+        parts.append(rb)
+    got = comm.exchange_batches(parts)
+    d = got.to_pydict()
+    # every row this rank received was addressed to it (p == rank)
+    assert all((v % 1000) // 10 == rank for v in d["i"]), d["i"]
+    # one row group from every source rank
+    srcs = sorted(set(v // 1000 for v in d["i"]))
+    assert srcs == list(range(w)), srcs
+    # null pattern preserved
+    for iv, sv in zip(d["i"], d["s"]):
+        i_within = iv % 10
+        if i_within % 3 == 0:
+            assert sv is None
+        else:
+            assert sv == f"r{iv // 1000}p{rank}i{i_within}"
+
+    # empty partitions to everyone but rank 0 (gather pattern)
+    rb = daft.from_pydict({"x": list(range(rank + 1))}).collect()._result[0]
+    empty = rb.slice(0, 0)
+    got2 = comm.exchange_batches(
+        [rb if p == 0 else empty for p in range(w)])
+    if rank == 0:
+        assert len(got2) == sum(r + 1 for r in range(w))
+    else:
+        assert len(got2) == 0
+
+    # skew: everyone sends everything to rank w-1, chunked via tiny budget
+    got3 = comm.exchange_batches(
+        [rb if p == w - 1 else empty for p in range(w)], hbm_budget=64)
+    if rank == w - 1:
+        assert len(got3) == sum(r + 1 for r in range(w))
+        assert sorted(got3.to_pydict()["x"])[:3] == [0, 0, 0]
+    else:
+        assert len(got3) == 0
+
+    # allgather (replicating a2a)
+    got4 = comm.allgather_batch(rb)
+    assert len(got4) == sum(r + 1 for r in range(w))
+    return "ok"
+
+
+def test_exchange_tensor_protocol_world4():
+    assert _spawn("_exchange_protocol_checks", world=4) == ["ok"] * 4
+
+
+def test_exchange_tensor_protocol_world8():
+    assert _spawn("_exchange_protocol_checks", world=8) == ["ok"] * 8
+
+
+def _tpch_pair_checks(rank, world):
+    from benchmarks.tpch import datagen, queries
+    sf = 0.005
+    T = datagen.dataframes(sf, device="cpu", rank=rank, world=world)
+    results = {}
+    for qi in (1, 5, 13, 21):
+        results[qi] = queries.run_query(qi, T, sf=sf).to_pydict()
+    return pickle.dumps(results)
+
+
+def test_distributed_world8_matches_single():
+    """world_size=8 — the exact rank count of the round-end scale run."""
+    outs = _spawn("_tpch_pair_checks", world=8)
+    per_rank = [pickle.loads(o) for o in outs]
+    import math
+
+    def norm_rows(d):
+        rows = list(zip(*d.values()))
+        key = lambda r: tuple(repr(x) for x in r
+                              if not isinstance(x, float))
+        return sorted(rows, key=key)
+
+    from benchmarks.tpch import datagen, queries
+    T = datagen.dataframes(0.005, device="cpu")
+    for qi, got in per_rank[0].items():
+        want = queries.run_query(qi, T, sf=0.005).to_pydict()
+        for r in per_rank[1:]:
+            assert r[qi] == got, f"q{qi} differs across ranks"
+        g_rows, w_rows = norm_rows(got), norm_rows(want)
+        assert len(g_rows) == len(w_rows), f"q{qi} rows"
+        for gr, wr in zip(g_rows, w_rows):
+            for gx, wx in zip(gr, wr):
+                if isinstance(wx, float):
+                    assert math.isclose(gx, wx, rel_tol=1e-9,
+                                        abs_tol=1e-6), f"q{qi}"
+                else:
+                    assert gx == wx, f"q{qi}"
