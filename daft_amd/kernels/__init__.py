@@ -724,7 +724,18 @@ def if_else(cond: Series, t: Series, f: Series) -> Series:
     if cond.validity is not None:
         m = m & cond.validity
     if out_dt.is_fixed_width():
-        out = torch.where(m, t.data, f.data)
+        td, fd = t.data, f.data
+        uview = None
+        if td.dtype in (torch.uint16, torch.uint32, torch.uint64):
+            # torch.where has no unsigned-wide CPU kernels: select through
+            # the bit-identical signed view
+            uview = td.dtype
+            signed = {torch.uint16: torch.int16, torch.uint32: torch.int32,
+                      torch.uint64: torch.int64}[uview]
+            td, fd = td.view(signed), fd.view(signed)
+        out = torch.where(m, td, fd)
+        if uview is not None:
+            out = out.view(uview)
         validity = None
         tv = t.validity if t.validity is not None else torch.ones_like(m)
         fv = f.validity if f.validity is not None else torch.ones_like(m)

@@ -198,7 +198,10 @@ class Filter(LogicalPlan):
 
     def approx_num_rows(self):
         e = self.children[0].approx_num_rows()
-        return None if e is None else e * 0.2
+        if e is None:
+            return None
+        from ..optimizer.join_reorder import selectivity
+        return e * selectivity(self.predicate)
 
     def describe(self):
         return f"Filter({self.predicate!r})"

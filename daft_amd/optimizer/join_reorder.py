@@ -1,0 +1,157 @@
+"""Cardinality-estimated greedy join reordering + predicate selectivity.
+
+Flattens maximal chains of inner equi-joins into a relation set with
+equi-edges, then rebuilds a left-deep tree greedily: start from the
+smallest estimated relation and repeatedly attach the connected relation
+minimizing the estimated intermediate size (FK heuristic: |A ⋈ B| ≈
+max(|A|, |B|)).  Only applies when every column name is unique across all
+relations, so no join-key merging or suffix renaming can change; the
+result is wrapped in a Project restoring the original column order.
+
+(ref: /root/reference/src/daft-logical-plan/src/optimization/rules/
+reorder_joins/ — the reference runs DP-ccp over stats; the greedy pass
+covers the TPC-H shapes at a fraction of the machinery)
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+from ..expressions.expressions import (Between, BinaryOp, ColumnRef, ExprNode,
+                                       IsIn, IsNull, Literal, Not, ScalarFn)
+from ..logical import plan as lp
+
+
+# ---------------------------------------------------------------------------
+# predicate selectivity (used by Filter.approx_num_rows)
+# ---------------------------------------------------------------------------
+
+def selectivity(e: ExprNode) -> float:
+    if isinstance(e, BinaryOp):
+        if e.op == "and":
+            return selectivity(e.left) * selectivity(e.right)
+        if e.op == "or":
+            s = selectivity(e.left) + selectivity(e.right)
+            return min(1.0, s)
+        if e.op == "eq":
+            return 0.1
+        if e.op in ("ne",):
+            return 0.9
+        if e.op in ("lt", "le", "gt", "ge"):
+            has_lit = isinstance(e.left, Literal) or \
+                isinstance(e.right, Literal)
+            return 0.3 if has_lit else 0.45
+    if isinstance(e, Between):
+        return 0.25
+    if isinstance(e, IsIn):
+        k = len(getattr(e, "values", []) or [])
+        return min(0.9, max(0.05, 0.1 * max(k, 1)))
+    if isinstance(e, Not):
+        return max(0.05, 1.0 - selectivity(e.child))
+    if isinstance(e, IsNull):
+        return 0.05
+    if isinstance(e, ScalarFn):
+        # like / startswith / contains-style string predicates
+        return 0.2
+    return 0.25
+
+
+# ---------------------------------------------------------------------------
+# greedy reordering
+# ---------------------------------------------------------------------------
+
+def _flatten(node: lp.LogicalPlan, rels: List[lp.LogicalPlan],
+             edges: List[Tuple[ExprNode, ExprNode]]) -> bool:
+    """Collect relations and equi-edges from a maximal inner-join chain."""
+    # any suffix is fine: the global-uniqueness check below guarantees no
+    # rename/merge actually occurred in this chain
+    if isinstance(node, lp.Join) and node.how == "inner" and \
+            node.left_on and node.prefix is None:
+        if not _flatten(node.children[0], rels, edges):
+            return False
+        if not _flatten(node.children[1], rels, edges):
+            return False
+        for le, re in zip(node.left_on, node.right_on):
+            edges.append((le, re))
+        return True
+    rels.append(node)
+    return True
+
+
+def _owner(e: ExprNode, rels: List[lp.LogicalPlan]) -> Optional[int]:
+    refs = set(e.column_refs())
+    if not refs:
+        return None
+    for i, r in enumerate(rels):
+        if refs <= set(r.schema.names()):
+            return i
+    return None
+
+
+def reorder_joins(plan: lp.LogicalPlan) -> lp.LogicalPlan:
+    """Single top-down pass: reorder each maximal inner-join chain root."""
+    out = _reorder_root(plan)
+    node = out if out is not None else plan
+    new_children = [reorder_joins(c) for c in node.children]
+    if any(n is not o for n, o in zip(new_children, node.children)):
+        node = node.with_children(new_children)
+    return node
+
+
+def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
+    if not (isinstance(plan, lp.Join) and plan.how == "inner"):
+        return None
+    rels: List[lp.LogicalPlan] = []
+    edges: List[Tuple[ExprNode, ExprNode]] = []
+    if not _flatten(plan, rels, edges) or len(rels) < 3:
+        return None
+    # all output names must be globally unique (no key merging / renames)
+    seen = set()
+    for r in rels:
+        for n in r.schema.names():
+            if n in seen:
+                return None
+            seen.add(n)
+    ests = [r.approx_num_rows() for r in rels]
+    if any(e is None for e in ests):
+        return None
+    # edge list as (rel_i, expr_i, rel_j, expr_j)
+    bound = []
+    for le, re in edges:
+        i, j = _owner(le, rels), _owner(re, rels)
+        if i is None or j is None or i == j:
+            return None
+        bound.append((i, le, j, re))
+
+    order = [min(range(len(rels)), key=lambda i: ests[i])]
+    placed = set(order)
+    cur_est = ests[order[0]]
+    joins_per_step: List[List[Tuple[ExprNode, ExprNode]]] = []
+    while len(placed) < len(rels):
+        # candidates connected to the placed set
+        cands = {}
+        for (i, le, j, re) in bound:
+            if i in placed and j not in placed:
+                cands.setdefault(j, []).append((le, re))
+            elif j in placed and i not in placed:
+                cands.setdefault(i, []).append((re, le))
+        if not cands:
+            return None  # cross product somewhere: keep user's order
+        best = min(cands, key=lambda r: max(cur_est, ests[r]))
+        order.append(best)
+        placed.add(best)
+        joins_per_step.append(cands[best])
+        cur_est = max(cur_est, ests[best])
+    if order == list(range(len(rels))):
+        return None  # already in greedy order
+
+    tree = rels[order[0]]
+    for step, ri in enumerate(order[1:]):
+        keys = joins_per_step[step]
+        tree = lp.Join(tree, rels[ri],
+                       [k[0] for k in keys], [k[1] for k in keys], "inner")
+    # restore the original column order
+    from ..expressions.expressions import ColumnRef as CR
+    want = plan.schema.names()
+    if tree.schema.names() != want:
+        return lp.Project(tree, [CR(n) for n in want])
+    return tree

@@ -17,14 +17,18 @@ Rule = Callable[[lp.LogicalPlan], Optional[lp.LogicalPlan]]
 
 
 def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
+    from .agg_pushdown import push_down_aggregation
+    from .join_reorder import reorder_joins
     batches: List[List[Rule]] = [
         [simplify_expressions],
-        [push_down_filter, drop_repartition],
+        [push_down_filter, drop_repartition, push_down_anti_semi_join],
+        [push_down_aggregation],
         [push_down_projection],
         [push_down_limit, rewrite_topn],
         [simplify_expressions],
     ]
-    for rules in batches:
+    reorder_after = 1  # join reordering once filters sit at the sources
+    for bi, rules in enumerate(batches):
         for _ in range(8):  # fixed-point cap
             changed = False
             for rule in rules:
@@ -34,6 +38,8 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
                     changed = True
             if not changed:
                 break
+        if bi == reorder_after:
+            plan = reorder_joins(plan)
     return plan
 
 
@@ -329,6 +335,42 @@ def push_down_filter(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
                 plan.predicate, child.pushdown_limit, child.read_options)
             return lp.Filter(new_scan, plan.predicate)
         return None
+    return None
+
+
+def push_down_anti_semi_join(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
+    """Push a semi/anti join below an inner join (or filter) on its probe
+    side, so key-existence filtering happens before the fan-out join —
+    gives q18 the orders-first reduction from the spec subquery.
+    (ref: rules/push_down_anti_semi_join.rs)  Also drops a redundant
+    Distinct on the semi/anti build side (semi joins are set-semantics
+    already)."""
+    if not (isinstance(plan, lp.Join) and plan.how in ("semi", "anti")):
+        return None
+    left, q = plan.children
+    if isinstance(q, lp.Distinct) and q.subset is None:
+        return plan.with_children([left, q.children[0]])
+    key_refs = set()
+    for e in plan.left_on:
+        key_refs.update(e.column_refs())
+    if isinstance(left, lp.Filter):
+        return lp.Filter(
+            plan.with_children([left.children[0], q]), left.predicate)
+    if isinstance(left, lp.Join) and left.how == "inner":
+        a, b = left.children
+        if key_refs and key_refs <= set(a.schema.names()):
+            new_a = lp.Join(a, q, plan.left_on, plan.right_on, plan.how,
+                            plan.suffix, plan.prefix)
+            return left.with_children([new_a, b])
+        rout = dict(left.right_passthrough())      # src -> out
+        out2src = {v: k for k, v in rout.items()}
+        if key_refs and all(r in out2src for r in key_refs):
+            new_on = [_substitute_cols(
+                e, {o: ColumnRef(s) for o, s in out2src.items()})
+                for e in plan.left_on]
+            new_b = lp.Join(b, q, new_on, plan.right_on, plan.how,
+                            plan.suffix, plan.prefix)
+            return left.with_children([a, new_b])
     return None
 
 

@@ -1,0 +1,157 @@
+"""Round-2 optimizer rules: aggregate pushdown through joins, semi/anti
+pushdown, comma-join connectivity ordering, greedy join reordering, and
+EXISTS-pair fusion (ref: /root/reference/src/daft-logical-plan/src/
+optimization/rules/{push_down_aggregation,push_down_anti_semi_join,
+eliminate_cross_join,reorder_joins}.rs)."""
+import pytest
+
+import daft_amd as daft
+from daft_amd import col, lit
+from daft_amd.logical import plan as lp
+
+
+def _optimized_plan(df):
+    from daft_amd.optimizer.optimizer import optimize
+    return optimize(df._builder.plan)
+
+
+def _find(plan, typ):
+    out = []
+
+    def rec(p):
+        if isinstance(p, typ):
+            out.append(p)
+        for c in p.children:
+            rec(c)
+    rec(plan)
+    return out
+
+
+def test_agg_pushdown_through_left_join():
+    cust = daft.from_pydict({"ck": [1, 2, 3], "seg": ["a", "b", "a"]})
+    orders = daft.from_pydict({"ok": [10, 11, 12, 13],
+                               "ock": [1, 1, 2, 1],
+                               "amt": [5.0, 7.0, 9.0, 11.0]})
+    q = cust.join(orders, left_on="ck", right_on="ock", how="left") \
+        .groupby("seg").agg(col("ok").count().alias("n"),
+                            col("amt").sum().alias("s"),
+                            col("amt").min().alias("mn"))
+    plan = _optimized_plan(q)
+    joins = _find(plan, lp.Join)
+    assert len(joins) == 1
+    # right side of the join is now a pre-aggregate keyed by the join key
+    right = joins[0].children[1]
+    while not isinstance(right, lp.Aggregate):
+        assert right.children, "no pre-aggregate on the join build side"
+        right = right.children[0]
+    # results still correct (customer 3 has no orders)
+    out = q.sort("seg").to_pydict()
+    assert out["seg"] == ["a", "b"]
+    assert out["n"] == [3, 1]
+    assert out["s"] == [23.0, 9.0]
+    assert out["mn"] == [5.0, 9.0]
+
+
+def test_agg_pushdown_count_star_left_join():
+    a = daft.from_pydict({"k": [1, 2], "g": ["x", "y"]})
+    b = daft.from_pydict({"bk": [1, 1, 1]})
+    q = a.join(b, left_on="k", right_on="bk", how="left") \
+        .groupby("g").agg(daft.functions.count().alias("c")) \
+        if hasattr(daft.functions, "count") else None
+    if q is None:
+        pytest.skip("no count() helper")
+    out = q.sort("g").to_pydict()
+    # k=1 matches 3 rows; k=2 has the single null-extended row -> count(*)=1
+    assert out["c"] == [3, 1]
+
+
+def test_semi_join_pushed_below_inner_join():
+    a = daft.from_pydict({"k": [1, 2, 3], "v": ["a", "b", "c"]})
+    b = daft.from_pydict({"k2": [1, 1, 2, 3], "w": [1, 2, 3, 4]})
+    q_keys = daft.from_pydict({"qk": [1, 3]})
+    j = a.join(b, left_on="k", right_on="k2").join(
+        q_keys, left_on="k", right_on="qk", how="semi")
+    plan = _optimized_plan(j)
+    # the semi join must sit BELOW the inner join now
+    top_joins = _find(plan, lp.Join)
+    semi = [x for x in top_joins if x.how == "semi"]
+    inner = [x for x in top_joins if x.how == "inner"]
+    assert semi and inner
+
+    def contains(p, target):
+        if p is target:
+            return True
+        return any(contains(c, target) for c in p.children)
+    assert contains(inner[0], semi[0]), "semi join was not pushed down"
+    out = j.sort(["k", "w"]).to_pydict()
+    assert out["k"] == [1, 1, 3]
+
+
+def test_comma_join_connectivity_no_cross():
+    """FROM a, b, c where edges only connect a-c and c-b must not plan a
+    cross join (the TPC-H q9 shape)."""
+    from daft_amd.session_api import Session
+    s = Session()
+    s.create_temp_table("ta", daft.from_pydict({"x": [1, 2, 3]}))
+    s.create_temp_table("tb", daft.from_pydict({"y": [1, 2]}))
+    s.create_temp_table("tc", daft.from_pydict({"cx": [1, 2, 3],
+                                                "cy": [1, 2, 1]}))
+    df = s.sql("select x, y from ta, tb, tc where x = cx and y = cy")
+    plan = df._builder.plan
+    if plan is not None:
+        assert not _find(plan, lp.Join) or all(
+            j.how != "cross" for j in _find(plan, lp.Join))
+    out = df.sort(["x", "y"]).to_pydict()
+    assert out["x"] == [1, 2, 3]
+    assert out["y"] == [1, 2, 1]
+
+
+def test_join_reorder_smallest_first():
+    big = daft.from_pydict({"bk": list(range(1000)),
+                            "sk2": [i % 10 for i in range(1000)]})
+    mid = daft.from_pydict({"mk": list(range(100))})
+    small = daft.from_pydict({"sk": list(range(10))})
+    # user order: big ⋈ mid ⋈ small
+    j = big.join(mid, left_on="bk", right_on="mk") \
+        .join(small, left_on="sk2", right_on="sk")
+    plan = _optimized_plan(j)
+    joins = _find(plan, lp.Join)
+    assert len(joins) == 2
+    # innermost join should start from the smallest relation (10 rows)
+    innermost = joins[-1]
+    ests = [c.approx_num_rows() for c in innermost.children]
+    assert min(e for e in ests if e is not None) <= 10
+    assert len(j.collect().to_pydict()["bk"]) == 100
+
+
+def test_exists_pair_fusion_single_aggregate():
+    """The q21 EXISTS/NOT-EXISTS pair over the same correlated group plans
+    ONE aggregate and ONE left join."""
+    from benchmarks.tpch import datagen, queries_sql, queries
+    from daft_amd.sql.planner import plan_sql
+    T = datagen.dataframes(0.01, device="cpu")
+    df = plan_sql(queries_sql.sql_for(21, 0.01), lambda n: T[n])
+    from daft_amd.optimizer.optimizer import optimize
+    plan = optimize(df._builder.plan)
+    aggs = _find(plan, lp.Aggregate)
+    # one fused min/max aggregate + the final count aggregate
+    assert len(aggs) == 2, [a.describe() for a in aggs]
+    left_joins = [x for x in _find(plan, lp.Join) if x.how == "left"]
+    assert len(left_joins) == 1
+    # and it still matches the hand-written DataFrame oracle
+    got = queries_sql.run_sql_query(21, T, sf=0.01).to_pydict()
+    want = queries.run_query(21, T, sf=0.01).to_pydict()
+    assert got == want
+
+
+def test_filter_selectivity_estimates():
+    from daft_amd.optimizer.join_reorder import selectivity
+    e_eq = (col("x") == lit(3))._node if hasattr(col("x") == lit(3), "_node") \
+        else (col("x") == lit(3))
+    # Expression wrapper: unwrap via repr-free access
+    from daft_amd.expressions.expressions import BinaryOp, ColumnRef, Literal
+    assert selectivity(BinaryOp("eq", ColumnRef("x"), Literal(3))) == 0.1
+    assert selectivity(BinaryOp("and",
+                                BinaryOp("eq", ColumnRef("x"), Literal(3)),
+                                BinaryOp("eq", ColumnRef("y"), Literal(4)))) \
+        == pytest.approx(0.01)
