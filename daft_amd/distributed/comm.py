@@ -180,6 +180,14 @@ def exchange_batches(parts: List[RecordBatch],
     # merge per-position dtypes across ranks (a validity bitmap may exist
     # on some ranks only)
     nbuf = len(my_dts)
+    if any(len(rd) != nbuf for _, rd in ctrl):
+        # buffer counts disagree even though skeletons matched: the ranks
+        # are in DIFFERENT exchanges — an SPMD plan-divergence bug.  Fail
+        # loudly instead of corrupting data.
+        raise RuntimeError(
+            f"exchange control mismatch (rank {rank()}): buffer counts "
+            f"{[len(rd) for _, rd in ctrl]} — ranks are executing "
+            f"different plans (SPMD divergence)")
     dts: List[Optional[str]] = list(my_dts)
     for _, rd in ctrl:
         for b in range(nbuf):

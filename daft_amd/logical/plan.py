@@ -59,6 +59,11 @@ class LogicalPlan:
         return f"{self.describe()}[{child}]"
 
 
+# global-row-count hints for SPMD planning: cache_key -> total rows across
+# all ranks (populated by the distributed runner before optimization)
+GLOBAL_ROW_HINTS: dict = {}
+
+
 class Source(LogicalPlan):
     """In-memory source: holds partition refs (MicroPartition cache keys).
 
@@ -91,6 +96,12 @@ class Source(LogicalPlan):
         return self
 
     def approx_num_rows(self):
+        # SPMD invariant: under a multi-rank run every rank MUST derive the
+        # same plan, so estimates use the GLOBAL row count (synced by the
+        # distributed runner), never the rank-local shard size
+        hint = GLOBAL_ROW_HINTS.get(self.cache_key)
+        if hint is not None:
+            return float(hint)
         return float(self.num_rows)
 
     def describe(self):
