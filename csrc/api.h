@@ -77,6 +77,17 @@ Tensor image_resize(Tensor src_bytes, Tensor src_off, Tensor src_h,
                     Tensor src_w, int64_t channels, int64_t out_h,
                     int64_t out_w);
 
+// fused expression interpreter (fusedexpr.hip)
+std::vector<Tensor> fused_eval(Tensor prog, Tensor lits,
+                               std::vector<Tensor> cols,
+                               std::vector<OptTensor> valids,
+                               std::vector<int64_t> codes,
+                               std::vector<double> scales,
+                               std::vector<int64_t> out_codes,
+                               std::vector<double> out_scales,
+                               std::vector<int64_t> out_need_valid,
+                               int64_t n);
+
 // strings (strings.hip)
 Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
 Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,

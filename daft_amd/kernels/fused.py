@@ -1,0 +1,338 @@
+"""Fused expression compilation: a projection/predicate list compiles to
+one postfix program executed by ONE HIP kernel (csrc/fusedexpr.hip),
+replacing a launch per expression node.
+
+The reference evaluates whole expression lists per batch
+(/root/reference/src/daft-recordbatch/src/lib.rs:1623 eval_expression_list);
+on MI355X the per-node torch kernels were ~half of remaining GPU time
+(profiles/tpch_sf100_kernel_stats.md), so the list becomes a single
+interpreter launch: each thread walks the program with a register stack
+of (f64 value, valid) pairs.
+
+Compilation is conservative: any node outside the numeric core (strings,
+decimals wider than 18, UDFs, aggs, dict columns, broadcasts) bails out
+and that expression falls back to per-node evaluation.
+"""
+from __future__ import annotations
+
+import datetime as _dt
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..expressions.expressions import (Alias, Between, BinaryOp, Cast,
+                                       ColumnRef, ExprNode, FillNull, IfElse,
+                                       IsIn, IsNull, Literal, Not)
+from ..schema import DataType, TypeKind
+
+OP_COL, OP_LIT = 0, 1
+OP_ADD, OP_SUB, OP_MUL, OP_DIV = 2, 3, 4, 5
+OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE = 6, 7, 8, 9, 10, 11
+OP_AND, OP_OR, OP_NOT, OP_NEG = 12, 13, 14, 15
+OP_ISNULL, OP_NOTNULL, OP_FILLNULL, OP_SELECT = 16, 17, 18, 19
+OP_STORE = 20
+
+_BIN = {"add": OP_ADD, "sub": OP_SUB, "mul": OP_MUL, "div": OP_DIV,
+        "eq": OP_EQ, "ne": OP_NE, "lt": OP_LT, "le": OP_LE,
+        "gt": OP_GT, "ge": OP_GE, "and": OP_AND, "or": OP_OR}
+
+_IN_CODES = {torch.float64: 0, torch.float32: 1, torch.int64: 2,
+             torch.int32: 3, torch.int16: 4, torch.int8: 5,
+             torch.bool: 6, torch.uint8: 6, torch.uint32: 7,
+             torch.uint64: 8}
+_OUT_CODES = {torch.float64: 0, torch.float32: 1, torch.int64: 2,
+              torch.int32: 3, torch.bool: 6}
+
+_MAX_STACK = 12
+_EPOCH = _dt.date(1970, 1, 1)
+
+
+class _Bail(Exception):
+    pass
+
+
+class _Compiler:
+    def __init__(self, batch):
+        self.batch = batch
+        self.n = len(batch)
+        self.ins: List[Tuple[int, int]] = []
+        self.lits: List[float] = []
+        self.cols: List = []          # Series
+        self.col_ix: Dict[str, int] = {}
+        self.depth = 0
+        self.max_depth = 0
+        self.any_valid = False        # any pushed slot may be invalid
+
+    def _push(self, k: int = 1):
+        self.depth += k
+        if self.depth > _MAX_STACK:
+            raise _Bail("stack too deep")
+        self.max_depth = max(self.max_depth, self.depth)
+
+    def _pop(self, k: int = 1):
+        self.depth -= k
+
+    def emit(self, op: int, arg: int = 0):
+        self.ins.append((op, arg))
+
+    def lit(self, v: float) -> int:
+        self.lits.append(float(v))
+        return len(self.lits) - 1
+
+    # -- literal conversion ------------------------------------------------
+    def _lit_value(self, e: Literal) -> float:
+        v = e.value
+        dt = e.dtype
+        if v is None:
+            raise _Bail("null literal")
+        if isinstance(v, bool):
+            return 1.0 if v else 0.0
+        if isinstance(v, _dt.datetime):
+            us = int((v - _dt.datetime(1970, 1, 1)).total_seconds() * 1e6)
+            if dt.kind == TypeKind.TIMESTAMP and dt.timeunit == "ms":
+                return us / 1000.0
+            if dt.kind == TypeKind.TIMESTAMP and dt.timeunit == "ns":
+                return us * 1000.0
+            return float(us)
+        if isinstance(v, _dt.date):
+            return float((v - _EPOCH).days)
+        if isinstance(v, (int, float)):
+            if isinstance(v, int) and abs(v) >= (1 << 53):
+                raise _Bail("literal beyond f64 exact range")
+            if dt is not None and dt.is_decimal():
+                return float(v)        # semantic value; columns descale
+            return float(v)
+        raise _Bail(f"literal {type(v).__name__}")
+
+    # -- column loading ----------------------------------------------------
+    def col(self, name: str):
+        if name in self.col_ix:
+            ix = self.col_ix[name]
+            s = self.cols[ix]
+        else:
+            s = self.batch.column(name)
+            if s.pyobjs is not None or s.data is None or s.is_dict() or \
+                    s.children:
+                raise _Bail(f"column {name} not fusable")
+            if len(s) != self.n:
+                raise _Bail("broadcast column")
+            dt = s.dtype
+            if not (dt.is_numeric() or dt.is_boolean() or dt.is_temporal()
+                    or dt.is_decimal()):
+                raise _Bail(f"dtype {dt}")
+            if dt.is_decimal() and s.data.dtype != torch.int64:
+                raise _Bail("wide decimal")
+            if s.data.dtype not in _IN_CODES:
+                raise _Bail(f"torch dtype {s.data.dtype}")
+            ix = len(self.cols)
+            self.cols.append(s)
+            self.col_ix[name] = ix
+        if s.validity is not None:
+            self.any_valid = True
+        self.emit(OP_COL, ix)
+        self._push()
+
+    # -- tree walk ---------------------------------------------------------
+    def compile(self, e: ExprNode):
+        if isinstance(e, Alias):
+            return self.compile(e.child)
+        if isinstance(e, ColumnRef):
+            return self.col(e.name)
+        if isinstance(e, Literal):
+            self.emit(OP_LIT, self.lit(self._lit_value(e)))
+            self._push()
+            return
+        if isinstance(e, BinaryOp):
+            op = _BIN.get(e.op)
+            if op is None:
+                raise _Bail(f"binop {e.op}")
+            # temporal raw values are unit-dependent (days vs us): both
+            # sides must agree on kind+unit or the f64 compare is wrong
+            try:
+                lt = e.left.to_field(self.batch.schema).dtype
+                rt = e.right.to_field(self.batch.schema).dtype
+            except Exception:
+                raise _Bail("untyped operand")
+            if lt.is_temporal() or rt.is_temporal():
+                if lt.kind != rt.kind or \
+                        getattr(lt, "timeunit", None) != \
+                        getattr(rt, "timeunit", None):
+                    raise _Bail("temporal unit mismatch")
+                if e.op not in ("eq", "ne", "lt", "le", "gt", "ge", "sub"):
+                    raise _Bail("temporal arithmetic")
+            self.compile(e.left)
+            self.compile(e.right)
+            self.emit(op)
+            self._pop()
+            return
+        if isinstance(e, Not):
+            self.compile(e.child)
+            self.emit(OP_NOT)
+            return
+        if isinstance(e, IsNull):
+            self.compile(e.child)
+            self.emit(OP_NOTNULL if getattr(e, "negate", False)
+                      else OP_ISNULL)
+            return
+        if isinstance(e, FillNull):
+            self.compile(e.child)
+            self.compile(e.fill)
+            self.emit(OP_FILLNULL)
+            self._pop()
+            return
+        if isinstance(e, IfElse):
+            self.compile(e.pred)
+            self.compile(e.truthy)
+            self.compile(e.falsy)
+            self.emit(OP_SELECT)
+            self._pop(2)
+            return
+        if isinstance(e, Between):
+            self.compile(e.child)
+            # dup via re-compilation (columns re-load from cache; cheap)
+            self.compile(e.lo)
+            self.emit(OP_GE)
+            self._pop()
+            self.compile(e.child)
+            self.compile(e.hi)
+            self.emit(OP_LE)
+            self._pop()
+            self.emit(OP_AND)
+            self._pop()
+            return
+        if isinstance(e, IsIn):
+            vals = getattr(e, "values", None)
+            if vals is None or len(vals) == 0 or len(vals) > 8:
+                raise _Bail("is_in size")
+            first = True
+            for v in vals:
+                if not isinstance(v, (int, float, bool)) or \
+                        (isinstance(v, int) and abs(v) >= (1 << 53)):
+                    raise _Bail("is_in value")
+                self.compile(e.child)
+                self.emit(OP_LIT, self.lit(float(v)))
+                self._push()
+                self.emit(OP_EQ)
+                self._pop()
+                if not first:
+                    self.emit(OP_OR)
+                    self._pop()
+                first = False
+            return
+        if isinstance(e, Cast):
+            src = e.child.to_field(self.batch.schema).dtype
+            dst = e.dtype
+            ok = (src.is_numeric() or src.is_boolean() or
+                  src.is_temporal()) and \
+                (dst.is_numeric() and not dst.is_decimal())
+            if not ok:
+                raise _Bail(f"cast {src}->{dst}")
+            self.compile(e.child)
+            if dst.is_integer():
+                # the kernel stores through llrint; mid-tree int casts of
+                # non-integral values would diverge from torch truncation
+                if not (src.is_integer() or src.is_boolean() or
+                        src.is_temporal()):
+                    raise _Bail("float->int cast")
+            return
+        raise _Bail(type(e).__name__)
+
+
+class FusedPlan:
+    __slots__ = ("prog", "lits", "cols", "out_meta", "n")
+
+    def __init__(self, prog, lits, cols, out_meta, n):
+        self.prog = prog
+        self.lits = lits
+        self.cols = cols
+        self.out_meta = out_meta
+        self.n = n
+
+
+def try_fuse(exprs: List[ExprNode], batch) -> Optional[List]:
+    """Compile as many of `exprs` as possible into one kernel; returns the
+    full result list (fused + fallback per-node eval) or None when fusion
+    is not worthwhile (fewer than 4 fused instructions)."""
+    if batch.device.type != "cuda" or len(batch) == 0:
+        return None
+    from . import load_native
+    native = load_native()
+    if native is None:
+        return None
+
+    comp = _Compiler(batch)
+    fused_ix: List[int] = []
+    out_meta = []
+    results: List = [None] * len(exprs)
+    for i, e in enumerate(exprs):
+        mark = (len(comp.ins), len(comp.lits), len(comp.cols),
+                dict(comp.col_ix), comp.any_valid)
+        try:
+            f = e.to_field(batch.schema)
+            out_dt = f.dtype
+            tdt = out_dt.to_torch() if out_dt.is_fixed_width() else None
+            if tdt not in _OUT_CODES:
+                raise _Bail(f"out dtype {out_dt}")
+            if out_dt.is_decimal():
+                raise _Bail("decimal out")
+            av_before = comp.any_valid
+            comp.any_valid = False
+            comp.compile(e)
+            produces_null = comp.any_valid or _may_null(e)
+            comp.any_valid = comp.any_valid or av_before
+            comp.emit(OP_STORE, len(out_meta))
+            comp._pop()
+            out_meta.append((i, f.name, out_dt, _OUT_CODES[tdt],
+                             produces_null))
+            fused_ix.append(i)
+        except _Bail:
+            comp.ins = comp.ins[:mark[0]]
+            comp.lits = comp.lits[:mark[1]]
+            comp.cols = comp.cols[:mark[2]]
+            comp.col_ix = mark[3]
+            comp.any_valid = mark[4]
+            comp.depth = 0
+    if not fused_ix or len(comp.ins) < 4:
+        return None
+
+    import itertools
+    prog = torch.tensor(list(itertools.chain.from_iterable(comp.ins)),
+                        dtype=torch.int32)
+    lits = torch.tensor(comp.lits, dtype=torch.float64)
+    datas, valids, codes, scales = [], [], [], []
+    for s in comp.cols:
+        datas.append(s.data)
+        valids.append(s.validity)
+        codes.append(_IN_CODES[s.data.dtype])
+        if s.dtype.is_decimal():
+            scales.append(10.0 ** (-s.dtype.scale))
+        else:
+            scales.append(1.0)
+    out_codes = [m[3] for m in out_meta]
+    out_scales = [1.0] * len(out_meta)
+    out_need_valid = [1 if m[4] else 0 for m in out_meta]
+    n = len(batch)
+    flat = native.fused_eval(prog, lits, datas, valids, codes, scales,
+                             out_codes, out_scales, out_need_valid, n)
+    from ..series import Series
+    for k, (i, name, out_dt, code, pn) in enumerate(out_meta):
+        data = flat[2 * k]
+        valid = flat[2 * k + 1] if pn else None
+        if valid is not None and bool(valid.all().item()):
+            valid = None
+        if out_dt.to_torch() != data.dtype:
+            data = data.view(out_dt.to_torch()) \
+                if data.element_size() == out_dt.to_torch().itemsize \
+                else data.to(out_dt.to_torch())
+        results[i] = Series(name, out_dt, data=data, validity=valid)
+    for i, e in enumerate(exprs):
+        if results[i] is None:
+            results[i] = e.evaluate(batch)
+    return results
+
+
+def _may_null(e: ExprNode) -> bool:
+    """Static: can this expression introduce nulls beyond input validity?"""
+    if isinstance(e, (IsNull,)):
+        return False
+    return any(_may_null(c) for c in e.children())

@@ -334,7 +334,21 @@ def _multi_agg(batch, gids, num_groups, named_aggs, mask):
     if nat is None:
         return None
     datas, valids, ops, meta = [], [], [], []
+    # evaluate all distinct agg inputs in ONE fused kernel (q1: the five
+    # value expressions share column reads and launch once)
     value_cache: dict = {}
+    uniq = []
+    for _cname, a in named_aggs:
+        if a.child is not None and a.kind != AggKind.COUNT_ALL:
+            key = repr(a.child)
+            if key not in value_cache:
+                value_cache[key] = None
+                uniq.append((key, a.child))
+    if uniq:
+        from .cse import evaluate_with_cse
+        for (key, _e), s in zip(uniq, evaluate_with_cse(
+                [e for _k, e in uniq], batch)):
+            value_cache[key] = s
     for cname, a in named_aggs:
         if a.kind == AggKind.COUNT_ALL or (a.kind == AggKind.COUNT and
                                            a.child is None):

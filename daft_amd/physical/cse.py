@@ -57,7 +57,16 @@ def shared_subtrees(exprs: List[ExprNode]) -> set:
 
 def evaluate_with_cse(exprs: List[ExprNode], batch):
     """Evaluate exprs over the batch, computing shared pure subtrees
-    once.  Returns the list of result Series (unnamed)."""
+    once.  Returns the list of result Series (unnamed).
+
+    On device batches the whole list first tries the fused interpreter
+    kernel (kernels/fused.py): everything numeric compiles into ONE
+    launch; only unfusable expressions fall back here."""
+    if batch.device.type == "cuda":
+        from ..kernels.fused import try_fuse
+        fused = try_fuse(exprs, batch)
+        if fused is not None:
+            return fused
     shared = shared_subtrees(exprs)
     if not shared:
         return [e.evaluate(batch) for e in exprs]

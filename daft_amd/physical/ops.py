@@ -197,8 +197,9 @@ class FilterOp(PhysicalOp):
         self.predicate = predicate
 
     def execute(self, ectx) -> BatchIter:
+        from .cse import evaluate_with_cse
         for rb in self.children[0].execute_tracked(ectx):
-            mask = self.predicate.evaluate(rb)
+            mask = evaluate_with_cse([self.predicate], rb)[0]
             if len(mask) == 1 and len(rb) != 1:
                 mask = mask.broadcast(len(rb))
             out = rb.filter(mask)

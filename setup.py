@@ -21,6 +21,7 @@ setup(
                 "csrc/sort.hip",
                 "csrc/strings.hip",
                 "csrc/multimodal.hip",
+                "csrc/fusedexpr.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -155,3 +155,28 @@ def test_filter_selectivity_estimates():
                                 BinaryOp("eq", ColumnRef("x"), Literal(3)),
                                 BinaryOp("eq", ColumnRef("y"), Literal(4)))) \
         == pytest.approx(0.01)
+
+
+def test_fused_expr_compiler_program_shape():
+    """The fused-expression compiler (kernels/fused.py) produces a sane
+    postfix program for a q1-style projection (execution itself is
+    GPU-only; see tests/test_gpu.py)."""
+    from daft_amd.kernels.fused import _Compiler, OP_COL, OP_LIT, OP_MUL, \
+        OP_SUB, OP_STORE, _Bail
+    df = daft.from_pydict({"p": [1.0, 2.0], "d": [0.1, 0.2],
+                           "s": ["x", "y"]}).collect()
+    rb = df._result[0]
+    from daft_amd.expressions.expressions import resolve_exprs
+    (e,) = resolve_exprs([col("p") * (lit(1) - col("d"))])
+    c = _Compiler(rb)
+    c.compile(e)
+    ops = [o for o, _ in c.ins]
+    assert ops == [OP_COL, OP_LIT, OP_COL, OP_SUB, OP_MUL]
+    assert len(c.cols) == 2 and c.depth == 1
+
+    # strings bail
+    (s,) = resolve_exprs([col("s")])
+    c2 = _Compiler(rb)
+    import pytest as _pt
+    with _pt.raises(_Bail):
+        c2.compile(s)
