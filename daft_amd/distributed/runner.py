@@ -43,15 +43,54 @@ def _sync_source_stats(plan) -> None:
     if not sources:
         return
     counts = [s.num_rows for s in sources]
+    # sample-NDV per column for join-cost estimation (lazy computation is
+    # forbidden under SPMD: local samples differ across ranks).  Cached by
+    # cache_key so each table is sampled once per process.
+    from ..optimizer.stats import NDV_HINTS, SAMPLE_ROWS, sample_ndv
+    from ..context import get_context
+    ndvs: List[dict] = []
+    for s in sources:
+        if s.cache_key in getattr(_sync_source_stats, "_seen", set()):
+            ndvs.append({})
+            continue
+        per_col: dict = {}
+        try:
+            parts = get_context().cache.get(s.cache_key)
+            big = max(parts, key=len) if parts else None
+            if big is not None and len(big):
+                for c in big.columns:
+                    if c.pyobjs is None:
+                        per_col[c.name] = sample_ndv(c, n_rows=len(big))
+        except Exception:
+            per_col = {}
+        ndvs.append(per_col)
     import torch.distributed as dist
     gathered: List[list] = [None] * comm.world()  # type: ignore
-    dist.all_gather_object(gathered, counts)
-    if any(len(g) != len(counts) for g in gathered):
+    dist.all_gather_object(gathered, [counts, ndvs])
+    if any(len(g[0]) != len(counts) for g in gathered):
         raise RuntimeError(
             "SPMD plans disagree on source count: "
-            f"{[len(g) for g in gathered]}")
+            f"{[len(g[0]) for g in gathered]}")
+    seen = getattr(_sync_source_stats, "_seen", set())
     for i, s in enumerate(sources):
-        lp.GLOBAL_ROW_HINTS[s.cache_key] = sum(g[i] for g in gathered)
+        lp.GLOBAL_ROW_HINTS[s.cache_key] = sum(g[0][i] for g in gathered)
+        if s.cache_key in seen:
+            continue
+        seen.add(s.cache_key)
+        cols = set()
+        for g in gathered:
+            cols.update(g[1][i].keys())
+        for cname in cols:
+            locs = [g[1][i].get(cname, 0.0) for g in gathered]
+            mx = max(locs)
+            if mx <= SAMPLE_ROWS * 0.1:
+                # category-like everywhere: shards repeat the same values
+                comb = mx
+            else:
+                comb = min(float(lp.GLOBAL_ROW_HINTS[s.cache_key]),
+                           float(sum(locs)))
+            NDV_HINTS[(s.cache_key, cname)] = comb
+    _sync_source_stats._seen = seen
 
 
 class DistributedRunner:

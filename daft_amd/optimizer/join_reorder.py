@@ -87,6 +87,51 @@ def _owner(e: ExprNode, rels: List[lp.LogicalPlan]) -> Optional[int]:
     return None
 
 
+def _ndv_of(rel: lp.LogicalPlan, colname: str,
+            est: Optional[float]) -> Optional[float]:
+    """Distinct-count estimate for a column of a relation subtree."""
+    from .stats import ndv_for_source
+    if isinstance(rel, lp.Source):
+        return ndv_for_source(rel.cache_key, colname, est)
+    if isinstance(rel, lp.Aggregate):
+        for g in rel.groupby:
+            base = g.child if hasattr(g, "child") and \
+                g.__class__.__name__ == "Alias" else g
+            try:
+                out = g.to_field(rel.children[0].schema).name
+            except Exception:
+                return None
+            if out == colname:
+                # grouped output is distinct on its keys
+                return rel.approx_num_rows()
+        return None
+    if isinstance(rel, lp.Project):
+        from ..expressions.expressions import Alias as _A, ColumnRef as _C
+        for e in rel.exprs:
+            try:
+                out = e.to_field(rel.children[0].schema).name
+            except Exception:
+                continue
+            if out == colname:
+                base = e.child if isinstance(e, _A) else e
+                if isinstance(base, _C):
+                    return _ndv_of(rel.children[0], base.name, est)
+                return None
+        return None
+    if isinstance(rel, lp.Join):
+        ls = rel.children[0].schema
+        if colname in set(ls.names()):
+            return _ndv_of(rel.children[0], colname, est)
+        rout = dict(rel.right_passthrough())
+        for src, out in rout.items():
+            if out == colname:
+                return _ndv_of(rel.children[1], src, est)
+        return None
+    if len(rel.children) == 1:
+        return _ndv_of(rel.children[0], colname, est)
+    return None
+
+
 def reorder_joins(plan: lp.LogicalPlan) -> lp.LogicalPlan:
     """Single top-down pass: reorder each maximal inner-join chain root."""
     out = _reorder_root(plan)
@@ -122,25 +167,50 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
             return None
         bound.append((i, le, j, re))
 
+    def edge_ndv(ri: int, e: ExprNode) -> Optional[float]:
+        base = e
+        from ..expressions.expressions import Alias, Cast
+        while isinstance(base, (Alias, Cast)):
+            base = base.child
+        if not isinstance(base, ColumnRef):
+            return None
+        nd = _ndv_of(rels[ri], base.name, ests[ri])
+        if nd is not None and ests[ri] is not None:
+            nd = min(nd, ests[ri])
+        return nd
+
+    def join_size(cur_est: float, ri: int,
+                  edge_list) -> float:
+        # |T ⋈ R| ≈ |T|·|R| / max(ndv); multiple edges take the tightest
+        # key (max ndv) — underestimating the combined NDV only
+        # OVERestimates the join, which is the safe direction.
+        denom = 1.0
+        for (ti, te, re_) in edge_list:
+            nd_t = edge_ndv(ti, te)
+            nd_r = edge_ndv(ri, re_)
+            denom = max(denom, max(nd_t or 1.0, nd_r or 1.0))
+        return cur_est * ests[ri] / denom
+
     order = [min(range(len(rels)), key=lambda i: ests[i])]
     placed = set(order)
     cur_est = ests[order[0]]
     joins_per_step: List[List[Tuple[ExprNode, ExprNode]]] = []
     while len(placed) < len(rels):
-        # candidates connected to the placed set
+        # candidates connected to the placed set; remember the tree-side
+        # owner of each edge for NDV lookup
         cands = {}
         for (i, le, j, re) in bound:
             if i in placed and j not in placed:
-                cands.setdefault(j, []).append((le, re))
+                cands.setdefault(j, []).append((i, le, re))
             elif j in placed and i not in placed:
-                cands.setdefault(i, []).append((re, le))
+                cands.setdefault(i, []).append((j, re, le))
         if not cands:
             return None  # cross product somewhere: keep user's order
-        best = min(cands, key=lambda r: max(cur_est, ests[r]))
+        best = min(cands, key=lambda r: join_size(cur_est, r, cands[r]))
         order.append(best)
         placed.add(best)
-        joins_per_step.append(cands[best])
-        cur_est = max(cur_est, ests[best])
+        joins_per_step.append([(te, re_) for _ti, te, re_ in cands[best]])
+        cur_est = max(1.0, join_size(cur_est, best, cands[best]))
     if order == list(range(len(rels))):
         return None  # already in greedy order
 
