@@ -87,6 +87,47 @@ def _owner(e: ExprNode, rels: List[lp.LogicalPlan]) -> Optional[int]:
     return None
 
 
+def _rel_width(rel: lp.LogicalPlan) -> float:
+    """Estimated bytes per row of a relation's output."""
+    w = 0.0
+    for f in rel.schema:
+        try:
+            w += f.dtype.to_torch().itemsize
+        except TypeError:
+            w += 24.0      # strings / nested: offset + payload estimate
+    return max(w, 1.0)
+
+
+def _tree_cost(node: lp.LogicalPlan):
+    """(total cost, est rows, est width) of an existing join tree using
+    the same rows x width metric as the greedy builder."""
+    if isinstance(node, lp.Join) and node.how == "inner" and node.left_on:
+        lc, le_, lw = _tree_cost(node.children[0])
+        rc, re_, rw = _tree_cost(node.children[1])
+        if None in (lc, le_, rc, re_):
+            return None, None, None
+        denom = 1.0
+        from ..expressions.expressions import Alias as _A, Cast as _Ct, \
+            ColumnRef as _C
+        for a, b in zip(node.left_on, node.right_on):
+            for side, e in ((0, a), (1, b)):
+                base = e
+                while isinstance(base, (_A, _Ct)):
+                    base = base.child
+                if isinstance(base, _C):
+                    sub = node.children[side]
+                    nd = _ndv_of(sub, base.name, sub.approx_num_rows())
+                    if nd is not None:
+                        denom = max(denom, nd)
+        out = le_ * re_ / denom
+        w = lw + rw
+        return lc + rc + out * w, out, w
+    est = node.approx_num_rows()
+    if est is None:
+        return None, None, None
+    return 0.0, est, _rel_width(node)
+
+
 def _ndv_of(rel: lp.LogicalPlan, colname: str,
             est: Optional[float]) -> Optional[float]:
     """Distinct-count estimate for a column of a relation subtree."""
@@ -191,9 +232,13 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
             denom = max(denom, max(nd_t or 1.0, nd_r or 1.0))
         return cur_est * ests[ri] / denom
 
+    widths = [_rel_width(r) for r in rels]
+
     order = [min(range(len(rels)), key=lambda i: ests[i])]
     placed = set(order)
     cur_est = ests[order[0]]
+    cur_width = widths[order[0]]
+    greedy_cost = 0.0
     joins_per_step: List[List[Tuple[ExprNode, ExprNode]]] = []
     while len(placed) < len(rels):
         # candidates connected to the placed set; remember the tree-side
@@ -206,11 +251,24 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
                 cands.setdefault(i, []).append((j, re, le))
         if not cands:
             return None  # cross product somewhere: keep user's order
-        best = min(cands, key=lambda r: join_size(cur_est, r, cands[r]))
+        # cost per step = output rows x output width (bytes): carrying a
+        # wide build side through every later probe is what the row-count
+        # metric missed (q10: customer's strings gathered per join)
+        best = min(cands, key=lambda r: join_size(cur_est, r, cands[r]) *
+                   (cur_width + widths[r]))
         order.append(best)
         placed.add(best)
         joins_per_step.append([(te, re_) for _ti, te, re_ in cands[best]])
         cur_est = max(1.0, join_size(cur_est, best, cands[best]))
+        cur_width += widths[best]
+        greedy_cost += cur_est * cur_width
+
+    # only replace the author's join order when the estimate says the
+    # greedy order is clearly cheaper — hand-tuned DataFrame programs are
+    # usually already good, and estimates are coarse
+    orig_cost, _e, _w = _tree_cost(plan)
+    if orig_cost is not None and greedy_cost >= 0.7 * orig_cost:
+        return None
     if order == list(range(len(rels))):
         return None  # already in greedy order
 
