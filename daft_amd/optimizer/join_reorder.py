@@ -240,6 +240,7 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     cur_width = widths[order[0]]
     greedy_cost = 0.0
     joins_per_step: List[List[Tuple[ExprNode, ExprNode]]] = []
+    step_ests: List[float] = []
     while len(placed) < len(rels):
         # candidates connected to the placed set; remember the tree-side
         # owner of each edge for NDV lookup
@@ -260,6 +261,7 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
         placed.add(best)
         joins_per_step.append([(te, re_) for _ti, te, re_ in cands[best]])
         cur_est = max(1.0, join_size(cur_est, best, cands[best]))
+        step_ests.append(cur_est)
         cur_width += widths[best]
         greedy_cost += cur_est * cur_width
 
@@ -273,10 +275,22 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
         return None  # already in greedy order
 
     tree = rels[order[0]]
+    run_est = ests[order[0]]
     for step, ri in enumerate(order[1:]):
         keys = joins_per_step[step]
-        tree = lp.Join(tree, rels[ri],
-                       [k[0] for k in keys], [k[1] for k in keys], "inner")
+        # physical hash joins build on the RIGHT child (physical/ops.py
+        # JoinOp): put the smaller input there — it also keeps bucket
+        # chains short (build on 15M customers keyed by 25 nationkeys
+        # makes 600k-row chains; build on 25 nations makes chains of 1)
+        if run_est <= ests[ri]:
+            tree = lp.Join(rels[ri], tree,
+                           [k[1] for k in keys], [k[0] for k in keys],
+                           "inner")
+        else:
+            tree = lp.Join(tree, rels[ri],
+                           [k[0] for k in keys], [k[1] for k in keys],
+                           "inner")
+        run_est = step_ests[step]
     # restore the original column order
     from ..expressions.expressions import ColumnRef as CR
     want = plan.schema.names()

@@ -17,16 +17,22 @@ Rule = Callable[[lp.LogicalPlan], Optional[lp.LogicalPlan]]
 
 
 def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
+    import os
     from .agg_pushdown import push_down_aggregation
     from .join_reorder import reorder_joins
+    disabled = set((os.environ.get("DAFT_AMD_DISABLE_RULES") or "")
+                   .split(","))
     batches: List[List[Rule]] = [
         [simplify_expressions],
-        [push_down_filter, drop_repartition, push_down_anti_semi_join],
-        [push_down_aggregation],
+        [push_down_filter, drop_repartition] +
+        ([] if "semipush" in disabled else [push_down_anti_semi_join]),
+        [] if "aggpush" in disabled else [push_down_aggregation],
         [push_down_projection],
         [push_down_limit, rewrite_topn],
         [simplify_expressions],
     ]
+    if "reorder" in disabled:
+        reorder_joins = lambda p: p  # noqa: E731
     reorder_after = 1  # join reordering once filters sit at the sources
     for bi, rules in enumerate(batches):
         for _ in range(8):  # fixed-point cap
