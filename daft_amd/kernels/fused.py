@@ -253,6 +253,9 @@ def try_fuse(exprs: List[ExprNode], batch) -> Optional[List]:
     """Compile as many of `exprs` as possible into one kernel; returns the
     full result list (fused + fallback per-node eval) or None when fusion
     is not worthwhile (fewer than 4 fused instructions)."""
+    import os
+    if os.environ.get("DAFT_AMD_DISABLE_FUSED"):
+        return None
     if batch.device.type != "cuda" or len(batch) == 0:
         return None
     from . import load_native
@@ -265,6 +268,11 @@ def try_fuse(exprs: List[ExprNode], batch) -> Optional[List]:
     out_meta = []
     results: List = [None] * len(exprs)
     for i, e in enumerate(exprs):
+        base = e
+        while isinstance(base, Alias):
+            base = base.child
+        if isinstance(base, (ColumnRef, Literal)):
+            continue      # zero-copy via normal eval; fusing would copy
         mark = (len(comp.ins), len(comp.lits), len(comp.cols),
                 dict(comp.col_ix), comp.any_valid)
         try:

@@ -128,6 +128,16 @@ def push_down_aggregation(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
                                   for a in aggs_found):
         return None
 
+    # only worthwhile when grouping by the join key actually SHRINKS the
+    # right side: pre-aggregating 150M orders by their unique o_orderkey
+    # (q12) is a full extra groupby for zero reduction
+    r_est = right.approx_num_rows()
+    if r_est is not None and r_est > 0 and len(rk_names) == 1:
+        from .join_reorder import _ndv_of
+        nd = _ndv_of(right, rk_names[0], r_est)
+        if nd is not None and nd > 0.5 * r_est:
+            return None
+
     u64 = DataType.uint64()
     pre_aggs: List[ExprNode] = []
     repl: dict = {}
