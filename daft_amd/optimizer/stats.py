@@ -50,8 +50,29 @@ def sample_ndv(series, n_rows: Optional[int] = None) -> float:
     u = int(torch.unique(h).numel())
     ratio = u / max(k, 1)
     if ratio < 0.1:
+        # saturated: the sample has seen (nearly) every value
         return float(u)
-    return float(max(u, ratio * n_rows))
+    # invert the expected-distinct curve u = N·(1 − e^(−k/N)) for the
+    # domain size N (monotone in N: bisection).  Naive ratio-scaling
+    # (u/k × n_rows) overestimates key NDVs by 500×: 65k draws from 1M
+    # distinct values collide only ~2000 times, and the q9 reorder then
+    # priced a 600M-row join at 1M rows (3× regression).
+    import math
+    if k - u < 3:
+        return float(n_rows)      # too few collisions to bound N
+    lo, hi = float(u), float(max(n_rows, u + 1))
+
+    def expected_u(N: float) -> float:
+        return N * (1.0 - math.exp(-k / N))
+    if expected_u(hi) <= u:
+        return hi
+    for _ in range(60):
+        mid = 0.5 * (lo + hi)
+        if expected_u(mid) < u:
+            lo = mid
+        else:
+            hi = mid
+    return float(min(max(0.5 * (lo + hi), u), n_rows))
 
 
 def ndv_for_source(cache_key: str, column: str,

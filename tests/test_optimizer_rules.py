@@ -208,3 +208,28 @@ def test_fused_expr_compiler_program_shape():
     import pytest as _pt
     with _pt.raises(_Bail):
         c2.compile(s)
+
+
+def test_ndv_birthday_estimator():
+    """sample_ndv must recover domain sizes both below and far above the
+    sample size (the q9 regression: 1M-distinct column estimated at 580M
+    by naive ratio scaling)."""
+    import random
+    import torch as _t
+    from daft_amd.optimizer.stats import sample_ndv
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+    random.seed(11)
+    n = 1_000_000
+    for domain, tol in ((25, 1.5), (20_000, 2.0), (300_000, 3.0)):
+        data = _t.randint(0, domain, (n,), dtype=_t.int64)
+        s = Series("x", DataType.int64(), data=data)
+        est = sample_ndv(s, n_rows=n)
+        true_nd = float(len(_t.unique(data)))
+        assert true_nd / tol <= est <= true_nd * tol, \
+            (domain, est, true_nd)
+    # key-like: all distinct
+    s = Series("k", DataType.int64(),
+               data=_t.arange(n, dtype=_t.int64))
+    est = sample_ndv(s, n_rows=n)
+    assert est >= n * 0.5
