@@ -132,7 +132,7 @@ def push_down_aggregation(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     # right side: pre-aggregating 150M orders by their unique o_orderkey
     # (q12) is a full extra groupby for zero reduction
     r_est = right.approx_num_rows()
-    if r_est is not None and r_est > 0 and len(rk_names) == 1:
+    if r_est is not None and r_est > 10_000 and len(rk_names) == 1:
         from .join_reorder import _ndv_of
         nd = _ndv_of(right, rk_names[0], r_est)
         if nd is not None and nd > 0.5 * r_est:
