@@ -67,105 +67,164 @@ DEV_INLINE double fe_load(const FCol& c, int64_t i, bool* valid) {
   return v * c.scale;
 }
 
+template <bool HV>
 __global__ void fused_eval_kernel(const int32_t* __restrict__ prog,
                                   int n_ins,
                                   const double* __restrict__ lits,
+                                  int n_lits,
                                   const FCol* __restrict__ cols,
                                   const FOut* __restrict__ outs,
                                   int64_t n) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    double st[FE_STACK];
-    bool va[FE_STACK];
+  // 4 rows per thread, program walked once per 4-row vector: amortizes
+  // instruction decode/branching 4x and keeps loads coalesced (row =
+  // base + r*blockDim + tid).  Program + literals staged in LDS.
+  constexpr int R = 4;
+  constexpr int VSTACK = 8;
+  extern __shared__ double smem[];
+  double* slits = smem;
+  int32_t* sprog = (int32_t*)(smem + n_lits);
+  for (int t = threadIdx.x; t < n_lits; t += blockDim.x) slits[t] = lits[t];
+  for (int t = threadIdx.x; t < n_ins * 2; t += blockDim.x)
+    sprog[t] = prog[t];
+  __syncthreads();
+
+  const int64_t chunk = (int64_t)blockDim.x * R;
+  const int64_t gstride = (int64_t)gridDim.x * chunk;
+  for (int64_t base = (int64_t)blockIdx.x * chunk; base < n;
+       base += gstride) {
+    double st[VSTACK][R];
+    bool va[VSTACK][R];
+    const int64_t row0 = base + threadIdx.x;
+    const bool full = base + chunk <= n;
     int sp = 0;
     for (int pc = 0; pc < n_ins; ++pc) {
-      int32_t op = prog[pc * 2], arg = prog[pc * 2 + 1];
+      const int32_t op = sprog[pc * 2], arg = sprog[pc * 2 + 1];
       switch (op) {
         case OP_COL: {
-          bool v;
-          st[sp] = fe_load(cols[arg], i, &v);
-          va[sp] = v;
+          const FCol c = cols[arg];
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            const int64_t i = row0 + (int64_t)r * blockDim.x;
+            bool v = true;
+            st[sp][r] = (full || i < n) ? fe_load(c, i, &v) : 0.0;
+            if (HV) va[sp][r] = v;
+          }
           ++sp;
         } break;
-        case OP_LIT:
-          st[sp] = lits[arg];
-          va[sp] = true;
+        case OP_LIT: {
+          const double v = slits[arg];
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            st[sp][r] = v;
+            if (HV) va[sp][r] = true;
+          }
           ++sp;
-          break;
-        case OP_ADD: --sp; st[sp - 1] += st[sp]; va[sp - 1] &= va[sp]; break;
-        case OP_SUB: --sp; st[sp - 1] -= st[sp]; va[sp - 1] &= va[sp]; break;
-        case OP_MUL: --sp; st[sp - 1] *= st[sp]; va[sp - 1] &= va[sp]; break;
-        case OP_DIV: --sp; st[sp - 1] /= st[sp]; va[sp - 1] &= va[sp]; break;
-        case OP_EQ:
-          --sp; st[sp - 1] = st[sp - 1] == st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
-        case OP_NE:
-          --sp; st[sp - 1] = st[sp - 1] != st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
-        case OP_LT:
-          --sp; st[sp - 1] = st[sp - 1] < st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
-        case OP_LE:
-          --sp; st[sp - 1] = st[sp - 1] <= st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
-        case OP_GT:
-          --sp; st[sp - 1] = st[sp - 1] > st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
-        case OP_GE:
-          --sp; st[sp - 1] = st[sp - 1] >= st[sp] ? 1.0 : 0.0;
-          va[sp - 1] &= va[sp]; break;
+        } break;
+#define FE_BIN(expr)                                                      \
+  --sp;                                                                   \
+  _Pragma("unroll") for (int r = 0; r < R; ++r) {                         \
+    const double a = st[sp - 1][r], b = st[sp][r];                        \
+    (void)a; (void)b;                                                     \
+    st[sp - 1][r] = (expr);                                               \
+    if (HV) va[sp - 1][r] = va[sp - 1][r] && va[sp][r];                   \
+  }                                                                       \
+  break;
+        case OP_ADD: FE_BIN(a + b)
+        case OP_SUB: FE_BIN(a - b)
+        case OP_MUL: FE_BIN(a * b)
+        case OP_DIV: FE_BIN(a / b)
+        case OP_EQ: FE_BIN(a == b ? 1.0 : 0.0)
+        case OP_NE: FE_BIN(a != b ? 1.0 : 0.0)
+        case OP_LT: FE_BIN(a < b ? 1.0 : 0.0)
+        case OP_LE: FE_BIN(a <= b ? 1.0 : 0.0)
+        case OP_GT: FE_BIN(a > b ? 1.0 : 0.0)
+        case OP_GE: FE_BIN(a >= b ? 1.0 : 0.0)
+#undef FE_BIN
         case OP_AND: {
           --sp;
-          bool a = st[sp - 1] != 0.0, b = st[sp] != 0.0;
-          bool av = va[sp - 1], bv = va[sp];
-          st[sp - 1] = (a && b) ? 1.0 : 0.0;
-          va[sp - 1] = (av && bv) || (av && !a) || (bv && !b);
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            const bool a = st[sp - 1][r] != 0.0, b = st[sp][r] != 0.0;
+            st[sp - 1][r] = (a && b) ? 1.0 : 0.0;
+            if (HV) {
+              const bool av = va[sp - 1][r], bv = va[sp][r];
+              va[sp - 1][r] = (av && bv) || (av && !a) || (bv && !b);
+            }
+          }
         } break;
         case OP_OR: {
           --sp;
-          bool a = st[sp - 1] != 0.0, b = st[sp] != 0.0;
-          bool av = va[sp - 1], bv = va[sp];
-          st[sp - 1] = (a || b) ? 1.0 : 0.0;
-          va[sp - 1] = (av && bv) || (av && a) || (bv && b);
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            const bool a = st[sp - 1][r] != 0.0, b = st[sp][r] != 0.0;
+            st[sp - 1][r] = (a || b) ? 1.0 : 0.0;
+            if (HV) {
+              const bool av = va[sp - 1][r], bv = va[sp][r];
+              va[sp - 1][r] = (av && bv) || (av && a) || (bv && b);
+            }
+          }
         } break;
         case OP_NOT:
-          st[sp - 1] = st[sp - 1] != 0.0 ? 0.0 : 1.0;
+#pragma unroll
+          for (int r = 0; r < R; ++r)
+            st[sp - 1][r] = st[sp - 1][r] != 0.0 ? 0.0 : 1.0;
           break;
-        case OP_NEG: st[sp - 1] = -st[sp - 1]; break;
+        case OP_NEG:
+#pragma unroll
+          for (int r = 0; r < R; ++r) st[sp - 1][r] = -st[sp - 1][r];
+          break;
         case OP_ISNULL:
-          st[sp - 1] = va[sp - 1] ? 0.0 : 1.0;
-          va[sp - 1] = true;
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            st[sp - 1][r] = (HV && !va[sp - 1][r]) ? 1.0 : 0.0;
+            if (HV) va[sp - 1][r] = true;
+          }
           break;
         case OP_NOTNULL:
-          st[sp - 1] = va[sp - 1] ? 1.0 : 0.0;
-          va[sp - 1] = true;
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            st[sp - 1][r] = (!HV || va[sp - 1][r]) ? 1.0 : 0.0;
+            if (HV) va[sp - 1][r] = true;
+          }
           break;
         case OP_FILLNULL:
           --sp;
-          if (!va[sp - 1]) {
-            st[sp - 1] = st[sp];
-            va[sp - 1] = va[sp];
+          if (HV) {
+#pragma unroll
+            for (int r = 0; r < R; ++r) {
+              if (!va[sp - 1][r]) {
+                st[sp - 1][r] = st[sp][r];
+                va[sp - 1][r] = va[sp][r];
+              }
+            }
           }
           break;
         case OP_SELECT: {
-          // stack: ... cond t f  -> select
           sp -= 2;
-          bool m = (st[sp - 1] != 0.0) && va[sp - 1];
-          st[sp - 1] = m ? st[sp] : st[sp + 1];
-          va[sp - 1] = m ? va[sp] : va[sp + 1];
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            const bool m =
+                (st[sp - 1][r] != 0.0) && (!HV || va[sp - 1][r]);
+            st[sp - 1][r] = m ? st[sp][r] : st[sp + 1][r];
+            if (HV) va[sp - 1][r] = m ? va[sp][r] : va[sp + 1][r];
+          }
         } break;
         case OP_STORE: {
           --sp;
           const FOut o = outs[arg];
-          if (o.valid) o.valid[i] = va[sp];
-          double v = st[sp] / o.scale;
-          switch ((int)o.code) {
-            case 0: ((double*)o.data)[i] = v; break;
-            case 1: ((float*)o.data)[i] = (float)v; break;
-            case 2: ((int64_t*)o.data)[i] = (int64_t)llrint(v); break;
-            case 3: ((int32_t*)o.data)[i] = (int32_t)llrint(v); break;
-            case 6: ((bool*)o.data)[i] = v != 0.0; break;
+#pragma unroll
+          for (int r = 0; r < R; ++r) {
+            const int64_t i = row0 + (int64_t)r * blockDim.x;
+            if (!full && i >= n) continue;
+            if (HV && o.valid) o.valid[i] = va[sp][r];
+            const double v = st[sp][r] / o.scale;
+            switch ((int)o.code) {
+              case 0: ((double*)o.data)[i] = v; break;
+              case 1: ((float*)o.data)[i] = (float)v; break;
+              case 2: ((int64_t*)o.data)[i] = (int64_t)llrint(v); break;
+              case 3: ((int32_t*)o.data)[i] = (int32_t)llrint(v); break;
+              case 6: ((bool*)o.data)[i] = v != 0.0; break;
+            }
           }
         } break;
       }
@@ -226,12 +285,21 @@ std::vector<Tensor> fused_eval(Tensor prog, Tensor lits,
   auto dlits = lits.numel() ? lits.to(dev) : lits;
 
   if (n == 0) return results;
+  bool has_valid = false;
+  for (auto& v : valids)
+    if (v.has_value()) has_valid = true;
+  for (auto nv : out_need_valid)
+    if (nv) has_valid = true;
   int block = 256;
-  hipLaunchKernelGGL(fused_eval_kernel, dim3(grid_1d(n, block)), dim3(block),
-                     0, cur_stream(), (const int32_t*)dprog.data_ptr(),
-                     (int)(dprog.numel() / 2),
-                     dlits.numel() ? (const double*)dlits.data_ptr() : nullptr,
-                     (const FCol*)dcols.data_ptr(),
+  int n_ins = (int)(dprog.numel() / 2);
+  int n_lits = (int)dlits.numel();
+  size_t smem = sizeof(double) * n_lits + sizeof(int32_t) * n_ins * 2;
+  int grid = grid_1d(n, block, /*items_per_thread=*/4);
+  auto kern = has_valid ? fused_eval_kernel<true> : fused_eval_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(block), smem, cur_stream(),
+                     (const int32_t*)dprog.data_ptr(), n_ins,
+                     n_lits ? (const double*)dlits.data_ptr() : nullptr,
+                     n_lits, (const FCol*)dcols.data_ptr(),
                      (const FOut*)douts.data_ptr(), n);
   return results;
 }

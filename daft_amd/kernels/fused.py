@@ -43,7 +43,7 @@ _IN_CODES = {torch.float64: 0, torch.float32: 1, torch.int64: 2,
 _OUT_CODES = {torch.float64: 0, torch.float32: 1, torch.int64: 2,
               torch.int32: 3, torch.bool: 6}
 
-_MAX_STACK = 12
+_MAX_STACK = 8  # must match VSTACK in csrc/fusedexpr.hip
 _EPOCH = _dt.date(1970, 1, 1)
 
 
