@@ -88,6 +88,15 @@ std::vector<Tensor> fused_eval(Tensor prog, Tensor lits,
                                std::vector<int64_t> out_need_valid,
                                int64_t n);
 
+// hipRTC JIT for fused expressions (fusedjit.hip)
+bool fused_jit_available();
+std::vector<Tensor> fused_eval_jit(const std::string& src,
+                                   std::vector<Tensor> cols,
+                                   std::vector<OptTensor> valids,
+                                   std::vector<int64_t> out_codes,
+                                   std::vector<int64_t> out_need_valid,
+                                   int64_t n);
+
 // strings (strings.hip)
 Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
 Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,

@@ -263,6 +263,10 @@ def try_fuse(exprs: List[ExprNode], batch) -> Optional[List]:
     if native is None:
         return None
 
+    jit = try_fuse_jit(exprs, batch)
+    if jit is not None:
+        return jit
+
     comp = _Compiler(batch)
     fused_ix: List[int] = []
     out_meta = []
@@ -344,3 +348,333 @@ def _may_null(e: ExprNode) -> bool:
     if isinstance(e, (IsNull,)):
         return False
     return any(_may_null(c) for c in e.children())
+
+
+# ---------------------------------------------------------------------------
+# hipRTC codegen: straight-line kernel per (expression list, schema,
+# validity pattern), compiled once per process and cached by source
+# ---------------------------------------------------------------------------
+
+_C_LOAD = {0: "((const double*)C{i}d)[i]",
+           1: "(double)((const float*)C{i}d)[i]",
+           2: "(double)((const long long*)C{i}d)[i]",
+           3: "(double)((const int*)C{i}d)[i]",
+           4: "(double)((const short*)C{i}d)[i]",
+           5: "(double)((const signed char*)C{i}d)[i]",
+           6: "(double)((const unsigned char*)C{i}d)[i]",
+           7: "(double)((const unsigned int*)C{i}d)[i]",
+           8: "(double)(long long)((const unsigned long long*)C{i}d)[i]"}
+
+
+class _CodeGen:
+    """Symbolic twin of _Compiler: walks the same node set but emits C
+    statements (value temp, validity temp) instead of opcodes."""
+
+    def __init__(self, comp: "_Compiler"):
+        self.comp = comp
+        self.lines: List[str] = []
+        self.tmp = 0
+
+    def t(self) -> str:
+        self.tmp += 1
+        return f"t{self.tmp}"
+
+    def emit(self, typ: str, name: str, expr: str):
+        self.lines.append(f"      {typ} {name} = {expr};")
+
+    def gen(self, e: ExprNode) -> Tuple[str, Optional[str]]:
+        """Returns (value C expr, validity C expr or None=always valid)."""
+        if isinstance(e, Alias):
+            return self.gen(e.child)
+        if isinstance(e, ColumnRef):
+            ix = self.comp.col_ix[e.name]
+            s = self.comp.cols[ix]
+            code = _IN_CODES[s.data.dtype]
+            v = self.t()
+            load = _C_LOAD[code].format(i=ix)
+            if s.dtype.is_decimal():
+                load = f"({load}) * {10.0 ** (-s.dtype.scale)!r}"
+            self.emit("const double", v, load)
+            if s.validity is None:
+                return v, None
+            vv = self.t()
+            self.emit("const bool", vv, f"C{ix}v[i]")
+            return v, vv
+        if isinstance(e, Literal):
+            return repr(self.comp._lit_value(e)), None
+        if isinstance(e, BinaryOp):
+            a, av = self.gen(e.left)
+            b, bv = self.gen(e.right)
+            out = self.t()
+            cop = {"add": "+", "sub": "-", "mul": "*", "div": "/",
+                   "eq": "==", "ne": "!=", "lt": "<", "le": "<=",
+                   "gt": ">", "ge": ">="}.get(e.op)
+            if cop is not None:
+                if e.op in ("eq", "ne", "lt", "le", "gt", "ge"):
+                    self.emit("const double", out,
+                              f"(({a}) {cop} ({b})) ? 1.0 : 0.0")
+                else:
+                    self.emit("const double", out, f"({a}) {cop} ({b})")
+                return out, self._and_valid(av, bv)
+            if e.op in ("and", "or"):
+                # Kleene three-valued logic (matches kernels.logical_op)
+                ab = self.t()
+                bb = self.t()
+                self.emit("const bool", ab, f"({a}) != 0.0")
+                self.emit("const bool", bb, f"({b}) != 0.0")
+                op = "&&" if e.op == "and" else "||"
+                self.emit("const double", out,
+                          f"(({ab}) {op} ({bb})) ? 1.0 : 0.0")
+                if av is None and bv is None:
+                    return out, None
+                avv = av or "true"
+                bvv = bv or "true"
+                vv = self.t()
+                if e.op == "and":
+                    self.emit("const bool", vv,
+                              f"(({avv}) && ({bvv})) || (({avv}) && "
+                              f"!({ab})) || (({bvv}) && !({bb}))")
+                else:
+                    self.emit("const bool", vv,
+                              f"(({avv}) && ({bvv})) || (({avv}) && "
+                              f"({ab})) || (({bvv}) && ({bb}))")
+                return out, vv
+            raise _Bail(f"binop {e.op}")
+        if isinstance(e, Not):
+            a, av = self.gen(e.child)
+            out = self.t()
+            self.emit("const double", out, f"(({a}) != 0.0) ? 0.0 : 1.0")
+            return out, av
+        if isinstance(e, IsNull):
+            a, av = self.gen(e.child)
+            out = self.t()
+            neg = getattr(e, "negate", False)
+            if av is None:
+                self.emit("const double", out, "0.0" if not neg else "1.0")
+            elif neg:
+                self.emit("const double", out, f"({av}) ? 1.0 : 0.0")
+            else:
+                self.emit("const double", out, f"({av}) ? 0.0 : 1.0")
+            return out, None
+        if isinstance(e, FillNull):
+            a, av = self.gen(e.child)
+            b, bv = self.gen(e.fill)
+            if av is None:
+                return a, None
+            out = self.t()
+            self.emit("const double", out, f"({av}) ? ({a}) : ({b})")
+            if bv is None:
+                return out, None
+            vv = self.t()
+            self.emit("const bool", vv, f"({av}) || ({bv})")
+            return out, vv
+        if isinstance(e, IfElse):
+            c, cv = self.gen(e.pred)
+            t_, tv = self.gen(e.truthy)
+            f_, fv = self.gen(e.falsy)
+            m = self.t()
+            cvv = f" && ({cv})" if cv is not None else ""
+            self.emit("const bool", m, f"(({c}) != 0.0){cvv}")
+            out = self.t()
+            self.emit("const double", out, f"({m}) ? ({t_}) : ({f_})")
+            if tv is None and fv is None:
+                return out, None
+            vv = self.t()
+            self.emit("const bool", vv,
+                      f"({m}) ? ({tv or 'true'}) : ({fv or 'true'})")
+            return out, vv
+        if isinstance(e, Between):
+            a, av = self.gen(e.child)
+            lo, lov = self.gen(e.lo)
+            hi, hiv = self.gen(e.hi)
+            out = self.t()
+            self.emit("const double", out,
+                      f"((({a}) >= ({lo})) && (({a}) <= ({hi}))) "
+                      f"? 1.0 : 0.0")
+            return out, self._and_valid(self._and_valid(av, lov), hiv)
+        if isinstance(e, IsIn):
+            a, av = self.gen(e.child)
+            parts = []
+            for v in e.values:
+                parts.append(f"(({a}) == {float(v)!r})")
+            out = self.t()
+            self.emit("const double", out,
+                      f"({' || '.join(parts)}) ? 1.0 : 0.0")
+            return out, av
+        if isinstance(e, Cast):
+            return self.gen(e.child)
+        raise _Bail(type(e).__name__)
+
+    def _and_valid(self, a: Optional[str], b: Optional[str]) -> Optional[str]:
+        if a is None:
+            return b
+        if b is None:
+            return a
+        vv = self.t()
+        self.emit("const bool", vv, f"({a}) && ({b})")
+        return vv
+
+
+_C_STORE = {0: "((double*)O{k}d)[i] = {v};",
+            1: "((float*)O{k}d)[i] = (float)({v});",
+            2: "((long long*)O{k}d)[i] = (long long)llrint({v});",
+            3: "((int*)O{k}d)[i] = (int)llrint({v});",
+            6: "((bool*)O{k}d)[i] = ({v}) != 0.0;"}
+
+
+def _gen_source(comp: "_Compiler", gens, out_meta) -> str:
+    """Full kernel source: hoisted column pointers, grid-stride row loop,
+    straight-line body."""
+    head = [
+        "extern \"C\" __global__ void fe(const long long* __restrict__ "
+        "cols, const long long* __restrict__ outs, long long n) {",
+    ]
+    for ix, s in enumerate(comp.cols):
+        head.append(f"  const void* C{ix}d = (const void*)cols[{ix} * 2];")
+        if s.validity is not None:
+            head.append(f"  const bool* C{ix}v = "
+                        f"(const bool*)cols[{ix} * 2 + 1];")
+    for k, m in enumerate(out_meta):
+        head.append(f"  void* O{k}d = (void*)outs[{k} * 2];")
+        if m[4]:
+            head.append(f"  bool* O{k}v = (bool*)outs[{k} * 2 + 1];")
+    head.append("  const long long stride = (long long)gridDim.x * "
+                "blockDim.x;")
+    head.append("  for (long long i = (long long)blockIdx.x * blockDim.x + "
+                "threadIdx.x; i < n; i += stride) {")
+    body: List[str] = []
+    for k, (lines, val, vld) in enumerate(gens):
+        body.extend(lines)
+        code = out_meta[k][3]
+        body.append("      " + _C_STORE[code].format(k=k, v=val))
+        if out_meta[k][4]:
+            body.append(f"      O{k}v[i] = {vld if vld else 'true'};")
+    tail = ["  }", "}"]
+    return "\n".join(head + body + tail) + "\n"
+
+
+def try_fuse_jit(exprs: List[ExprNode], batch) -> Optional[List]:
+    """hipRTC path: compile the list to a straight-line kernel.  Returns
+    the full result list or None (caller falls back to the interpreter)."""
+    import os
+    if os.environ.get("DAFT_AMD_DISABLE_JIT"):
+        return None
+    from . import load_native
+    native = load_native()
+    if native is None or not hasattr(native, "fused_eval_jit"):
+        return None
+
+    comp = _Compiler(batch)
+    gens = []
+    out_meta = []
+    results: List = [None] * len(exprs)
+    fused_ops = 0
+    for i, e in enumerate(exprs):
+        base = e
+        while isinstance(base, Alias):
+            base = base.child
+        if isinstance(base, (ColumnRef, Literal)):
+            continue          # zero-copy via normal eval
+        mark_cols = len(comp.cols)
+        mark_ix = dict(comp.col_ix)
+        try:
+            f = e.to_field(batch.schema)
+            out_dt = f.dtype
+            tdt = out_dt.to_torch() if out_dt.is_fixed_width() else None
+            if tdt not in _OUT_CODES or out_dt.is_decimal():
+                raise _Bail("out dtype")
+            # pre-scan columns via the opcode compiler's column loader
+            # rules; then generate C
+            cg = _CodeGen(comp)
+            # register columns through comp.col side effects by generating
+            av_before = comp.any_valid
+            comp.any_valid = False
+            # _CodeGen uses comp.col_ix; make sure every referenced column
+            # is registered first
+            _register_cols(comp, e)
+            val, vld = cg.gen(e)
+            produces_null = vld is not None
+            comp.any_valid = comp.any_valid or av_before
+            gens.append((cg.lines, val, vld))
+            out_meta.append((i, f.name, out_dt, _OUT_CODES[tdt],
+                             produces_null))
+            fused_ops += len(cg.lines)
+        except _Bail:
+            comp.cols = comp.cols[:mark_cols]
+            comp.col_ix = mark_ix
+    if not out_meta or fused_ops < 3:
+        return None
+
+    src = _gen_source(comp, gens, out_meta)
+    datas = [s.data for s in comp.cols]
+    valids = [s.validity for s in comp.cols]
+    out_codes = [m[3] for m in out_meta]
+    out_need_valid = [1 if m[4] else 0 for m in out_meta]
+    n = len(batch)
+    try:
+        flat = native.fused_eval_jit(src, datas, valids, out_codes,
+                                     out_need_valid, n)
+    except RuntimeError:
+        return None        # hiprtc unavailable/failed: interpreter path
+    from ..series import Series
+    for k, (i, name, out_dt, code, pn) in enumerate(out_meta):
+        data = flat[2 * k]
+        valid = flat[2 * k + 1] if pn else None
+        if valid is not None and bool(valid.all().item()):
+            valid = None
+        if out_dt.to_torch() != data.dtype:
+            data = data.to(out_dt.to_torch())
+        results[i] = Series(name, out_dt, data=data, validity=valid)
+    for i, e in enumerate(exprs):
+        if results[i] is None:
+            results[i] = e.evaluate(batch)
+    return results
+
+
+def _register_cols(comp: "_Compiler", e: ExprNode):
+    """Validate + register every column referenced by e (raises _Bail on
+    unfusable columns) without emitting opcodes."""
+    if isinstance(e, ColumnRef):
+        ins_mark = len(comp.ins)
+        d_mark = comp.depth
+        comp.col(e.name)
+        comp.ins = comp.ins[:ins_mark]
+        comp.depth = d_mark
+        return
+    if isinstance(e, Literal):
+        comp._lit_value(e)      # validates the literal kind
+        return
+    if isinstance(e, BinaryOp):
+        if e.op not in _BIN:
+            raise _Bail(f"binop {e.op}")
+        lt = e.left.to_field(comp.batch.schema).dtype
+        rt = e.right.to_field(comp.batch.schema).dtype
+        if lt.is_temporal() or rt.is_temporal():
+            if lt.kind != rt.kind or \
+                    getattr(lt, "timeunit", None) != \
+                    getattr(rt, "timeunit", None):
+                raise _Bail("temporal unit mismatch")
+            if e.op not in ("eq", "ne", "lt", "le", "gt", "ge", "sub"):
+                raise _Bail("temporal arithmetic")
+    if isinstance(e, Cast):
+        src = e.child.to_field(comp.batch.schema).dtype
+        dst = e.dtype
+        ok = (src.is_numeric() or src.is_boolean() or src.is_temporal()) \
+            and (dst.is_numeric() and not dst.is_decimal())
+        if not ok:
+            raise _Bail(f"cast {src}->{dst}")
+        if dst.is_integer() and not (src.is_integer() or src.is_boolean()
+                                     or src.is_temporal()):
+            raise _Bail("float->int cast")
+    if isinstance(e, IsIn):
+        vals = getattr(e, "values", None)
+        if not vals or len(vals) > 8 or any(
+                not isinstance(v, (int, float, bool)) or
+                (isinstance(v, int) and abs(v) >= (1 << 53))
+                for v in vals):
+            raise _Bail("is_in")
+    if not isinstance(e, (Alias, ColumnRef, Literal, BinaryOp, Not, IsNull,
+                          FillNull, IfElse, Between, IsIn, Cast)):
+        raise _Bail(type(e).__name__)
+    for c in e.children():
+        _register_cols(comp, c)

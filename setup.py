@@ -22,7 +22,9 @@ setup(
                 "csrc/strings.hip",
                 "csrc/multimodal.hip",
                 "csrc/fusedexpr.hip",
+                "csrc/fusedjit.hip",
             ],
+            libraries=["hiprtc"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
