@@ -40,7 +40,11 @@ hipFunction_t compile_or_get(const std::string& src) {
   std::lock_guard<std::mutex> lk(g_jit_mu);
   auto& cache = jit_cache();
   auto it = cache.find(src);
-  if (it != cache.end()) return it->second.fn;
+  if (it != cache.end()) {
+    if (it->second.fn == nullptr)
+      throw std::runtime_error("fused jit: compile previously failed");
+    return it->second.fn;
+  }
 
   hiprtcProgram prog;
   if (hiprtcCreateProgram(&prog, src.c_str(), "fe.hip", 0, nullptr,
@@ -54,6 +58,9 @@ hipFunction_t compile_or_get(const std::string& src) {
     std::string log(log_size, '\0');
     hiprtcGetProgramLog(prog, log.data());
     hiprtcDestroyProgram(&prog);
+    // cache the failure: retrying an impossible compile per batch would
+    // cost ~150 ms every call
+    cache.emplace(src, JitEntry{});
     throw std::runtime_error("hiprtc compile failed:\n" + log);
   }
   size_t code_size = 0;

@@ -374,6 +374,8 @@ class _CodeGen:
         self.comp = comp
         self.lines: List[str] = []
         self.tmp = 0
+        self.load_cache: Dict[int, Tuple[str, Optional[str]]] = {}
+        self.expr_cache: Dict[str, Tuple[str, Optional[str]]] = {}
 
     def t(self) -> str:
         self.tmp += 1
@@ -386,8 +388,22 @@ class _CodeGen:
         """Returns (value C expr, validity C expr or None=always valid)."""
         if isinstance(e, Alias):
             return self.gen(e.child)
+        if e.children():
+            key = repr(e)
+            hit = self.expr_cache.get(key)
+            if hit is not None:
+                return hit
+            out = self._gen_inner(e)
+            self.expr_cache[key] = out
+            return out
+        return self._gen_inner(e)
+
+    def _gen_inner(self, e: ExprNode) -> Tuple[str, Optional[str]]:
         if isinstance(e, ColumnRef):
             ix = self.comp.col_ix[e.name]
+            hit = self.load_cache.get(ix)
+            if hit is not None:
+                return hit
             s = self.comp.cols[ix]
             code = _IN_CODES[s.data.dtype]
             v = self.t()
@@ -395,10 +411,11 @@ class _CodeGen:
             if s.dtype.is_decimal():
                 load = f"({load}) * {10.0 ** (-s.dtype.scale)!r}"
             self.emit("const double", v, load)
-            if s.validity is None:
-                return v, None
-            vv = self.t()
-            self.emit("const bool", vv, f"C{ix}v[i]")
+            vv = None
+            if s.validity is not None:
+                vv = self.t()
+                self.emit("const bool", vv, f"C{ix}v[i]")
+            self.load_cache[ix] = (v, vv)
             return v, vv
         if isinstance(e, Literal):
             return repr(self.comp._lit_value(e)), None
@@ -565,44 +582,39 @@ def try_fuse_jit(exprs: List[ExprNode], batch) -> Optional[List]:
         return None
 
     comp = _Compiler(batch)
+    cg = _CodeGen(comp)     # ONE generator: shared temps + cross-expr CSE
     gens = []
     out_meta = []
     results: List = [None] * len(exprs)
-    fused_ops = 0
     for i, e in enumerate(exprs):
         base = e
         while isinstance(base, Alias):
             base = base.child
         if isinstance(base, (ColumnRef, Literal)):
             continue          # zero-copy via normal eval
-        mark_cols = len(comp.cols)
-        mark_ix = dict(comp.col_ix)
+        mark = (len(comp.cols), dict(comp.col_ix), len(cg.lines), cg.tmp,
+                dict(cg.load_cache), dict(cg.expr_cache))
         try:
             f = e.to_field(batch.schema)
             out_dt = f.dtype
             tdt = out_dt.to_torch() if out_dt.is_fixed_width() else None
             if tdt not in _OUT_CODES or out_dt.is_decimal():
                 raise _Bail("out dtype")
-            # pre-scan columns via the opcode compiler's column loader
-            # rules; then generate C
-            cg = _CodeGen(comp)
-            # register columns through comp.col side effects by generating
-            av_before = comp.any_valid
-            comp.any_valid = False
-            # _CodeGen uses comp.col_ix; make sure every referenced column
-            # is registered first
             _register_cols(comp, e)
+            lines_before = len(cg.lines)
             val, vld = cg.gen(e)
             produces_null = vld is not None
-            comp.any_valid = comp.any_valid or av_before
-            gens.append((cg.lines, val, vld))
+            gens.append((cg.lines[lines_before:], val, vld))
             out_meta.append((i, f.name, out_dt, _OUT_CODES[tdt],
                              produces_null))
-            fused_ops += len(cg.lines)
         except _Bail:
-            comp.cols = comp.cols[:mark_cols]
-            comp.col_ix = mark_ix
-    if not out_meta or fused_ops < 3:
+            comp.cols = comp.cols[:mark[0]]
+            comp.col_ix = mark[1]
+            cg.lines = cg.lines[:mark[2]]
+            cg.tmp = mark[3]
+            cg.load_cache = mark[4]
+            cg.expr_cache = mark[5]
+    if not out_meta or len(cg.lines) < 3:
         return None
 
     src = _gen_source(comp, gens, out_meta)
