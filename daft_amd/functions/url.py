@@ -19,11 +19,9 @@ def _fetch_one(u: Optional[str], on_error: str):
     try:
         if u.startswith("file://"):
             u = u[7:]
-        if u.startswith(("http://", "https://")):
-            import requests
-            r = requests.get(u, timeout=30)
-            r.raise_for_status()
-            return r.content
+        if u.startswith(("http://", "https://", "s3://", "s3a://")):
+            from ..io.object_store import get_source
+            return get_source(u).get(u)
         with open(u, "rb") as f:
             return f.read()
     except Exception:
@@ -42,17 +40,28 @@ def url_download_series(s: Series, on_error: str = "raise",
 
 
 def url_upload_series(s: Series, paths: Series, location: str) -> Series:
+    from ..io.object_store import get_source, is_remote
     data = s.cpu().to_pylist()
     names = paths.cpu().to_pylist()
-    os.makedirs(location, exist_ok=True)
+    remote = is_remote(location)
+    if not remote:
+        os.makedirs(location, exist_ok=True)
+        src = None
+    else:
+        src = get_source(location)
     out = []
     for d, n in zip(data, names):
         if d is None:
             out.append(None)
             continue
-        p = os.path.join(location, str(n))
-        with open(p, "wb") as f:
-            f.write(d if isinstance(d, bytes) else str(d).encode())
+        payload = d if isinstance(d, bytes) else str(d).encode()
+        if remote:
+            p = location.rstrip("/") + "/" + str(n)
+            src.put(p, payload)
+        else:
+            p = os.path.join(location, str(n))
+            with open(p, "wb") as f:
+                f.write(payload)
         out.append(p)
     res = Series.from_pylist(s.name, out, DataType.string())
     return res.to(s.device) if s.is_gpu() else res

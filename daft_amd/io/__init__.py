@@ -55,10 +55,22 @@ def from_pandas(df, device=None) -> DataFrame:
     return from_arrow(pa.Table.from_pandas(df), device=device)
 
 
-def _expand_paths(path: Union[str, List[str]]) -> List[str]:
+def _expand_paths(path: Union[str, List[str]],
+                  io_config=None) -> List[str]:
+    from .object_store import get_source, glob_paths, is_remote
     paths = [path] if isinstance(path, str) else list(path)
     out: List[str] = []
     for p in paths:
+        if is_remote(p):
+            if any(ch in p for ch in "*?["):
+                out.extend(glob_paths(p, io_config))
+            elif p.endswith("/"):
+                out.extend(sorted(
+                    f for f, _sz in
+                    get_source(p, io_config).list_prefix(p)))
+            else:
+                out.append(p)
+            continue
         if any(ch in p for ch in "*?["):
             matches = sorted(_glob.glob(p, recursive=True))
             out.extend(matches)
@@ -77,8 +89,9 @@ def _expand_paths(path: Union[str, List[str]]) -> List[str]:
 def read_parquet(path, columns: Optional[List[str]] = None,
                  io_config=None, **kwargs) -> DataFrame:
     from . import readers
-    paths = _expand_paths(path)
-    schema = readers.infer_schema(paths[0], "parquet")
+    paths = _expand_paths(path, io_config)
+    schema = readers.infer_schema(paths[0], "parquet",
+                                  storage_options=io_config)
     b = LogicalPlanBuilder.from_scan(schema, paths, "parquet",
                                      storage_options=io_config)
     df = DataFrame(b)
@@ -122,7 +135,8 @@ def read_csv(path, has_headers: bool = True, delimiter: str = ",",
     if schema is not None:
         sch = Schema.from_dict(schema)
     else:
-        sch = readers.infer_schema(paths[0], "csv", read_options)
+        sch = readers.infer_schema(paths[0], "csv", read_options,
+                                   storage_options=io_config)
     return DataFrame(LogicalPlanBuilder.from_scan(
         sch, paths, "csv", storage_options=io_config,
         read_options=read_options))
@@ -131,7 +145,8 @@ def read_csv(path, has_headers: bool = True, delimiter: str = ",",
 def read_json(path, io_config=None, **kwargs) -> DataFrame:
     from . import readers
     paths = _expand_paths(path)
-    sch = readers.infer_schema(paths[0], "json")
+    sch = readers.infer_schema(paths[0], "json",
+                               storage_options=io_config)
     return DataFrame(LogicalPlanBuilder.from_scan(
         sch, paths, "json", storage_options=io_config))
 

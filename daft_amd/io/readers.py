@@ -12,9 +12,31 @@ from ..schema import DataType, Field, Schema
 from .. import arrow_interop
 
 
+def _io_config_of(storage_options):
+    from .object_store import IOConfig
+    if isinstance(storage_options, IOConfig):
+        return storage_options
+    if isinstance(storage_options, dict):
+        c = storage_options.get("io_config")
+        if isinstance(c, IOConfig):
+            return c
+    return None
+
+
+def _open_input(path: str, storage_options=None):
+    """Local paths pass through; s3://, http(s):// resolve through the
+    object store (ranged/multipart layer in object_store.py)."""
+    from .object_store import get_source, is_remote
+    if isinstance(path, str) and is_remote(path):
+        return get_source(path, _io_config_of(storage_options)).open(path)
+    return path
+
+
 def infer_schema(path: str, file_format: str,
-                 read_options: Optional[dict] = None) -> Schema:
+                 read_options: Optional[dict] = None,
+                 storage_options=None) -> Schema:
     import pyarrow as pa
+    path = _open_input(path, storage_options)
     if file_format == "parquet":
         import pyarrow.parquet as pq
         a_schema = pq.read_schema(path)
@@ -44,6 +66,7 @@ def infer_schema(path: str, file_format: str,
 def read_file(path: str, file_format: str, columns: Optional[List[str]],
               predicate, limit: Optional[int], storage_options: dict,
               read_options: dict, device) -> Iterator[RecordBatch]:
+    path = _open_input(path, storage_options)
     if file_format == "parquet":
         yield from _read_parquet(path, columns, limit, device, predicate)
     elif file_format == "csv":
@@ -119,6 +142,8 @@ def _open_maybe_compressed(path):
     """Transparent .gz/.zst/.bz2 input (capability of the reference's
     daft-compression CompressionCodec::from_uri)."""
     import pyarrow as pa
+    if not isinstance(path, str):
+        return path          # already an open (remote) stream
     for ext, codec in ((".gz", "gzip"), (".zst", "zstd"), (".zstd", "zstd"),
                        (".bz2", "bz2"), (".lz4", "lz4")):
         if path.endswith(ext):

@@ -12,8 +12,10 @@ from ..recordbatch import RecordBatch
 def write_batches(batches: Iterator[RecordBatch], file_format: str,
                   root_dir: str, write_mode: str, partition_cols,
                   options: dict, ectx) -> List[str]:
-    os.makedirs(root_dir, exist_ok=True)
-    if write_mode == "overwrite":
+    from .object_store import is_remote
+    if not is_remote(root_dir):
+        os.makedirs(root_dir, exist_ok=True)
+    if write_mode == "overwrite" and not is_remote(root_dir):
         for f in os.listdir(root_dir):
             p = os.path.join(root_dir, f)
             if os.path.isfile(p) and f.endswith((".parquet", ".csv",
@@ -65,8 +67,39 @@ def _write_partitioned(rb: RecordBatch, file_format, root_dir,
         paths.append(_write_one(part, file_format, sub, options))
 
 
+def _join(dir_: str, name: str) -> str:
+    from .object_store import is_remote
+    if is_remote(dir_):
+        return dir_.rstrip("/") + "/" + name
+    return os.path.join(dir_, name)
+
+
 def _write_one(rb: RecordBatch, file_format: str, dir_: str,
                options: dict) -> str:
+    from .object_store import get_source, is_remote
+    if is_remote(dir_):
+        # encode to a buffer, then one put (multipart for large objects)
+        import io as _io
+        name = uuid.uuid4().hex[:16]
+        tbl = rb.to_arrow()
+        buf = _io.BytesIO()
+        if file_format == "parquet":
+            import pyarrow.parquet as pq
+            pq.write_table(tbl, buf,
+                           compression=options.get("compression", "snappy"),
+                           row_group_size=options.get("row_group_size",
+                                                      1 << 20))
+            path = _join(dir_, f"{name}.parquet")
+        elif file_format == "csv":
+            import pyarrow.csv as pacsv
+            pacsv.write_csv(tbl, buf)
+            path = _join(dir_, f"{name}.csv")
+        else:
+            raise ValueError(
+                f"remote write format {file_format} not supported")
+        get_source(path, options.get("io_config")).put(path,
+                                                       buf.getvalue())
+        return path
     name = uuid.uuid4().hex[:16]
     tbl = rb.to_arrow()
     if file_format == "parquet":

@@ -68,14 +68,9 @@ class UnionMode:
     DENSE = "dense"
 
 
-@dataclass
-class IOConfig:
-    """Storage configuration placeholder (local-filesystem only in this
-    offline build; S3/GCS/Azure fields accepted and ignored)."""
-    s3: Any = None
-    gcs: Any = None
-    azure: Any = None
-    http: Any = None
+# real object-store config (S3-compatible + HTTP backends implemented;
+# GCS/Azure accepted for API compatibility, raise on use)
+from .io.object_store import HTTPConfig, IOConfig, S3Config  # noqa: F401
 
 
 @dataclass

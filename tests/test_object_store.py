@@ -1,0 +1,167 @@
+"""Object-store IO: S3-compatible backend against an in-process mock
+server — ranged reads, multipart upload, listing, globbing, retry with
+backoff, and the full read_parquet/write_parquet round trip (ref:
+/root/reference/src/daft-io/ + tests/integration/io/conftest.py minio
+pattern)."""
+import pytest
+
+import daft_amd as daft
+from daft_amd import col
+from daft_amd.io.object_store import (HTTPSource, IOConfig, ObjectStoreError,
+                                      S3Config, S3Source, get_source,
+                                      glob_paths, set_default_io_config,
+                                      sigv4_headers, _glob_to_regex)
+
+from mock_s3 import MockS3
+
+
+@pytest.fixture()
+def s3():
+    srv = MockS3()
+    cfg = S3Config(endpoint_url=srv.endpoint, key_id="test",
+                   access_key="secret", region_name="us-east-1",
+                   num_tries=4, retry_initial_backoff_ms=1)
+    yield srv, S3Source(cfg), cfg
+    srv.close()
+
+
+def test_put_get_roundtrip(s3):
+    srv, src, _ = s3
+    src.put("s3://bkt/a/b.bin", b"hello world")
+    assert src.get("s3://bkt/a/b.bin") == b"hello world"
+    assert src.get_size("s3://bkt/a/b.bin") == 11
+    # ranged read
+    assert src.get("s3://bkt/a/b.bin", range_=(6, 11)) == b"world"
+
+
+def test_multipart_upload(s3):
+    srv, src, _ = s3
+    big = bytes(range(256)) * (130 * 1024)   # ~33 MB > 2 chunks
+    src.put("s3://bkt/big.bin", big)
+    assert srv.objects["bkt/big.bin"] == big
+    assert len(srv.uploads) == 0             # completed and cleaned up
+
+
+def test_list_and_glob(s3):
+    srv, src, _ = s3
+    for k in ("data/x/1.parquet", "data/x/2.parquet", "data/y/3.parquet",
+              "data/readme.txt"):
+        src.put(f"s3://bkt/{k}", b"z")
+    ls = src.list_prefix("s3://bkt/data/")
+    assert len(ls) == 4
+    got = src.glob("s3://bkt/data/**/*.parquet")
+    assert got == ["s3://bkt/data/x/1.parquet", "s3://bkt/data/x/2.parquet",
+                   "s3://bkt/data/y/3.parquet"]
+    got2 = src.glob("s3://bkt/data/x/*.parquet")
+    assert len(got2) == 2
+
+
+def test_retry_on_transient_faults(s3):
+    srv, src, _ = s3
+    src.put("s3://bkt/r.bin", b"ok")
+    srv.fail_next = 2                        # two 500s, then success
+    before = srv.requests
+    assert src.get("s3://bkt/r.bin") == b"ok"
+    assert srv.requests - before == 3        # 2 failures + 1 success
+
+
+def test_retry_exhaustion_raises(s3):
+    srv, src, _ = s3
+    src.put("s3://bkt/r.bin", b"ok")
+    srv.fail_next = 99
+    with pytest.raises(ObjectStoreError):
+        src.get("s3://bkt/r.bin")
+
+
+def test_not_found_no_retry(s3):
+    srv, src, _ = s3
+    before = srv.requests
+    from daft_amd.io.object_store import NotFoundError
+    with pytest.raises(NotFoundError):
+        src.get("s3://bkt/missing.bin")
+    assert srv.requests - before == 1        # 404s do not retry
+
+
+def test_write_read_parquet_via_s3(s3):
+    srv, _src, cfg = s3
+    io_cfg = IOConfig(s3=cfg)
+    set_default_io_config(io_cfg)
+    try:
+        df = daft.from_pydict({"k": [1, 2, 3, 4],
+                               "v": ["a", "b", "c", "d"],
+                               "x": [1.5, 2.5, None, 4.5]})
+        df.write_parquet("s3://bkt/tbl/")
+        keys = [k for k in srv.objects if k.endswith(".parquet")]
+        assert keys, srv.objects.keys()
+        back = daft.read_parquet("s3://bkt/tbl/*.parquet",
+                                 io_config=io_cfg).sort("k").to_pydict()
+        assert back["k"] == [1, 2, 3, 4]
+        assert back["v"] == ["a", "b", "c", "d"]
+        assert back["x"] == [1.5, 2.5, None, 4.5]
+        # filters still run over the remote scan
+        f = daft.read_parquet("s3://bkt/tbl/*.parquet", io_config=io_cfg) \
+            .where(col("k") > 2).sort("k").to_pydict()
+        assert f["k"] == [3, 4]
+    finally:
+        set_default_io_config(None)
+
+
+def test_url_download_from_s3(s3):
+    srv, src, cfg = s3
+    set_default_io_config(IOConfig(s3=cfg))
+    try:
+        src.put("s3://bkt/f1.bin", b"one")
+        src.put("s3://bkt/f2.bin", b"two")
+        df = daft.from_pydict({"url": ["s3://bkt/f1.bin",
+                                       "s3://bkt/f2.bin"]})
+        out = df.with_column("data",
+                             col("url").url.download()).to_pydict()
+        assert out["data"] == [b"one", b"two"]
+    finally:
+        set_default_io_config(None)
+
+
+def test_url_upload_to_s3(s3):
+    srv, _src, cfg = s3
+    set_default_io_config(IOConfig(s3=cfg))
+    try:
+        df = daft.from_pydict({"name": ["a.bin", "b.bin"],
+                               "data": [b"111", b"222"]})
+        out = df.with_column(
+            "path", daft.functions.url_upload(col("data"), col("name"),
+                                              "s3://bkt/up/")) \
+            .to_pydict() if hasattr(daft.functions, "url_upload") else None
+        if out is None:
+            pytest.skip("no url_upload function")
+        assert srv.objects["bkt/up/a.bin"] == b"111"
+    finally:
+        set_default_io_config(None)
+
+
+def test_glob_regex():
+    rx = _glob_to_regex("s3://b/data/**/*.parquet")
+    assert rx.match("s3://b/data/x/1.parquet")
+    assert rx.match("s3://b/data/x/y/z/1.parquet")
+    assert not rx.match("s3://b/data/x/1.csv")
+    rx2 = _glob_to_regex("s3://b/*.csv")
+    assert rx2.match("s3://b/a.csv")
+    assert not rx2.match("s3://b/sub/a.csv")
+
+
+def test_sigv4_shape():
+    h = sigv4_headers("GET", "https://bkt.s3.us-east-1.amazonaws.com/k",
+                      "us-east-1", "AKID", "SECRET", b"")
+    assert h["Authorization"].startswith("AWS4-HMAC-SHA256 Credential=AKID/")
+    assert "SignedHeaders=host;x-amz-content-sha256;x-amz-date" in \
+        h["Authorization"]
+    assert len(h["x-amz-content-sha256"]) == 64
+
+
+def test_http_source(s3):
+    srv, src, _ = s3
+    src.put("s3://bkt/h.bin", b"http-accessible")
+    h = HTTPSource()
+    url = f"{srv.endpoint}/bkt/h.bin"
+    assert h.get(url) == b"http-accessible"
+    assert h.get_size(url) == 15
+    assert h.get(url, range_=(0, 4)) == b"http"
