@@ -32,10 +32,10 @@ def _yield_exchanged(out: RecordBatch, ectx):
     to the device in morsel slices so downstream operators fold partial
     states instead of re-materializing the whole partition in HBM."""
     if str(out.device) == "cpu" and str(ectx.device).startswith("cuda"):
+        from ..physical.ops import stream_host_batch
         morsel = getattr(ectx.ctx.execution_config, "stream_morsel_rows",
                          1 << 26)
-        for lo in range(0, max(len(out), 1), morsel):
-            yield out.slice(lo, lo + morsel).to(ectx.device)
+        yield from stream_host_batch(out, ectx.device, morsel)
         return
     yield out
 

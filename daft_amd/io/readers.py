@@ -112,7 +112,14 @@ def read_files_prefetch(paths: List[str], file_format: str,
             if p is not None:
                 pending.append(ex.submit(load, p))
             for rb in batches:
-                yield rb.to(device) if str(device) != "cpu" else rb
+                if str(device) == "cpu":
+                    yield rb
+                else:
+                    # pinned-staged async H2D on the copy stream so the
+                    # transfer overlaps downstream compute
+                    from ..physical.ops import stream_host_batch
+                    yield from stream_host_batch(rb, device,
+                                                 max(len(rb), 1))
 
 
 def _read_parquet(path, columns, limit, device,

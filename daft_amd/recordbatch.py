@@ -81,8 +81,12 @@ class RecordBatch:
     def column_names(self) -> List[str]:
         return self.schema.names()
 
-    def to(self, device) -> "RecordBatch":
-        return RecordBatch([c.to(device) for c in self.columns],
+    def to(self, device, non_blocking: bool = False) -> "RecordBatch":
+        return RecordBatch([c.to(device, non_blocking)
+                            for c in self.columns], self._num_rows)
+
+    def pinned(self) -> "RecordBatch":
+        return RecordBatch([c.pinned() for c in self.columns],
                            self._num_rows)
 
     def cpu(self) -> "RecordBatch":

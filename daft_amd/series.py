@@ -148,12 +148,27 @@ class Series:
             out = out.with_validity(v)
         return out
 
-    def to(self, device) -> "Series":
+    def to(self, device, non_blocking: bool = False) -> "Series":
         def mv(t):
-            return None if t is None else t.to(device)
+            return None if t is None else t.to(device,
+                                               non_blocking=non_blocking)
         return Series(self.name, self.dtype, mv(self.data), mv(self.validity),
                       mv(self.offsets),
-                      [c.to(device) for c in self.children], self.pyobjs,
+                      [c.to(device, non_blocking) for c in self.children],
+                      self.pyobjs, self._length)
+
+    def pinned(self) -> "Series":
+        """Copy host buffers into pinned (page-locked) memory so H2D
+        transfers run as async DMA on a copy stream (out-of-core morsel
+        staging; ref role: the reference's scan-task reader prefetch,
+        sources/scan_task_reader.rs)."""
+        def pin(t):
+            if t is None or t.is_cuda or t.is_pinned():
+                return t
+            return t.pin_memory()
+        return Series(self.name, self.dtype, pin(self.data),
+                      pin(self.validity), pin(self.offsets),
+                      [c.pinned() for c in self.children], self.pyobjs,
                       self._length)
 
     # ------------------------------------------------------------------
