@@ -498,3 +498,29 @@ def dataframes(sf: float, device="cpu", rank: int = 0, world: int = 1):
         part = _PARTITIONING.get(name) if world > 1 else None
         out[name] = from_recordbatches([rb], partitioning=part)
     return out
+
+
+def dataframes_host_staged(sf: float, shards: int = 0, gen_device=None,
+                           columns: "dict | None" = None):
+    """SF1000-class out-of-core datagen: every table is generated in
+    `shards` GPU-sized chunks on `gen_device` (the deterministic per-key
+    generators make shard r of world K reproduce exactly rows [lo, hi))
+    and staged to HOST memory, so host RAM — not HBM — bounds the scale.
+    `columns` optionally restricts staged columns per table
+    ({"lineitem": [...]}) to fit host RAM for targeted query sets."""
+    import torch as _t
+    from daft_amd.io import from_recordbatches
+    if gen_device is None:
+        gen_device = "cuda:0" if _t.cuda.is_available() else "cpu"
+    if shards <= 0:
+        shards = max(1, int(sf) // 12)          # ~70M lineitem rows/shard
+    parts: Dict[str, list] = {}
+    for r in range(shards):
+        for name, rb in generate(sf, gen_device, r, shards).items():
+            if columns and name in columns:
+                keep = [c for c in rb.columns if c.name in columns[name]]
+                rb = type(rb)(keep, num_rows=len(rb))
+            parts.setdefault(name, []).append(rb.cpu())
+        if str(gen_device).startswith("cuda"):
+            _t.cuda.empty_cache()
+    return {name: from_recordbatches(ps) for name, ps in parts.items()}

@@ -582,14 +582,106 @@ class SortOp(PhysicalOp):
         self.descending = descending
         self.nulls_first = nulls_first
 
+    def _budget(self, ectx):
+        cfg = getattr(ectx.ctx, "execution_config", None)
+        lim = getattr(cfg, "memory_limit_bytes", None) if cfg else None
+        if lim is not None:
+            return lim
+        if str(ectx.device).startswith("cuda"):
+            import torch as _t
+            if _t.cuda.is_available():
+                free, _tot = _t.cuda.mem_get_info(ectx.device)
+                return free // 3
+        return None
+
     def execute(self, ectx) -> BatchIter:
-        batch = self._materialize_child(ectx)
-        if len(batch) == 0:
-            yield batch
+        budget = self._budget(ectx)
+        runs: List[RecordBatch] = []
+        spilled: List[RecordBatch] = []
+        acc = 0
+        for rb in self.children[0].execute_tracked(ectx):
+            if len(rb) == 0:
+                continue
+            runs.append(rb)
+            acc += rb.size_bytes()
+            if budget is not None and acc > budget:
+                spilled.extend(r.cpu() for r in runs)
+                runs, acc = [], 0
+        if not spilled:
+            if not runs:
+                yield RecordBatch.empty(self.schema, device=ectx.device)
+                return
+            batch = RecordBatch.concat(runs) if len(runs) > 1 else runs[0]
+            keys = [e.evaluate(batch) for e in self.by]
+            perm = rowops.argsort_multi(keys, self.descending,
+                                        self.nulls_first)
+            yield batch.take(perm, has_neg=False)
             return
-        keys = [e.evaluate(batch) for e in self.by]
-        perm = rowops.argsort_multi(keys, self.descending, self.nulls_first)
-        yield batch.take(perm, has_neg=False)
+        spilled.extend(r.cpu() for r in runs)
+        yield from self._external_sort(spilled, ectx, budget)
+
+    def _external_sort(self, spilled: List[RecordBatch], ectx,
+                       budget) -> BatchIter:
+        """Out-of-core sort: inputs larger than the HBM budget spill to
+        host, get range-partitioned by sampled boundaries into host
+        buckets (one device pass), then each bucket sorts in HBM and is
+        emitted in boundary order (ref: sinks/sort.rs buffering + the
+        distributed sample-sort, pipeline_node/sort.rs:84-130 — here the
+        'ranks' are HBM-sized buckets on one GPU)."""
+        from ..recordbatch import _range_partition_ids
+        total = sum(r.size_bytes() for r in spilled)
+        nb = max(2, -(-int(total) // max(int(budget) // 4, 1)))
+        names = [f"__k{i}" for i in range(len(self.by))]
+
+        def keys_of(rb):
+            cols = [e.evaluate(rb).rename(names[i])
+                    for i, e in enumerate(self.by)]
+            return RecordBatch(cols, num_rows=len(rb))
+
+        # sample boundaries across all runs
+        samples = []
+        for run in spilled:
+            k = min(len(run), 1024)
+            if k == 0:
+                continue
+            step = max(1, len(run) // k)
+            idx = torch.arange(0, len(run), step, dtype=torch.int64)[:k]
+            samples.append(keys_of(run.take(idx)))
+        allsamp = RecordBatch.concat(samples)
+        m = len(allsamp)
+        sorted_samp = allsamp.sort(names, self.descending, self.nulls_first)
+        bidx = torch.tensor([min(m - 1, ((i + 1) * m) // nb)
+                             for i in range(nb - 1)], dtype=torch.int64)
+        boundaries = sorted_samp.take(bidx).to(ectx.device)
+
+        morsel = getattr(ectx.ctx.execution_config, "stream_morsel_rows",
+                         1 << 26)
+        buckets: List[List[RecordBatch]] = [[] for _ in range(nb)]
+        for run in spilled:
+            for mo in stream_host_batch(run, ectx.device, morsel):
+                pid = _range_partition_ids(keys_of(mo), names, boundaries,
+                                           self.descending,
+                                           self.nulls_first)
+                perm, counts = rowops.partition_by_value(pid, nb)
+                reordered = mo.take(perm)
+                off = 0
+                for b, c in enumerate(counts.tolist()):
+                    if c:
+                        buckets[b].append(
+                            reordered.slice(off, off + c).cpu())
+                    off += c
+        for b in range(nb):
+            parts = buckets[b]
+            if not parts:
+                continue
+            dev_parts = [p.to(ectx.device) for p in parts]
+            batch = RecordBatch.concat(dev_parts) if len(dev_parts) > 1 \
+                else dev_parts[0]
+            keys = [e.evaluate(batch) for e in self.by]
+            perm = rowops.argsort_multi(keys, self.descending,
+                                        self.nulls_first)
+            yield batch.take(perm, has_neg=False)
+            buckets[b] = []
 
 
 class TopNOp(PhysicalOp):
@@ -604,23 +696,31 @@ class TopNOp(PhysicalOp):
         self.offset = offset
 
     def execute(self, ectx) -> BatchIter:
-        # per-batch prune to limit+offset, then final sort+slice
+        # per-batch prune to limit+offset, folded into a bounded running
+        # top-k accumulator (memory stays O(k), not O(batches*k))
         k = self.limit + self.offset
-        pruned: List[RecordBatch] = []
+
+        def prune(rb: RecordBatch) -> RecordBatch:
+            if len(rb) <= k:
+                return rb
+            keys = [e.evaluate(rb) for e in self.by]
+            perm = rowops.argsort_multi(keys, self.descending,
+                                        self.nulls_first)
+            return rb.take(perm[:k])
+
+        acc: Optional[RecordBatch] = None
         for rb in self.children[0].execute_tracked(ectx):
-            if len(rb) > k:
-                keys = [e.evaluate(rb) for e in self.by]
-                perm = rowops.argsort_multi(keys, self.descending,
-                                            self.nulls_first)
-                rb = rb.take(perm[:k])
-            pruned.append(rb)
-        if not pruned:
+            if len(rb) == 0:
+                continue
+            rb = prune(rb)
+            acc = rb if acc is None else prune(
+                RecordBatch.concat([acc, rb]))
+        if acc is None:
             yield RecordBatch.empty(self.schema, device=ectx.device)
             return
-        batch = RecordBatch.concat(pruned) if len(pruned) > 1 else pruned[0]
-        keys = [e.evaluate(batch) for e in self.by]
+        keys = [e.evaluate(acc) for e in self.by]
         perm = rowops.argsort_multi(keys, self.descending, self.nulls_first)
-        yield batch.take(perm[self.offset:k])
+        yield acc.take(perm[self.offset:k])
 
 
 class JoinOp(PhysicalOp):

@@ -215,10 +215,16 @@ class RecordBatch:
 
 def _range_partition_ids(rb: RecordBatch, keys: Sequence[str],
                          boundaries: RecordBatch,
-                         descending: Sequence[bool]) -> torch.Tensor:
-    """For each row, count how many boundary rows sort strictly before it."""
+                         descending: Sequence[bool],
+                         nulls_first: Optional[Sequence[bool]] = None
+                         ) -> torch.Tensor:
+    """For each row, count how many boundary rows sort strictly before it.
+    Null keys order per nulls_first (default: the argsort convention,
+    nulls first exactly when descending)."""
     n = len(rb)
     dev = rb.device
+    if nulls_first is None:
+        nulls_first = list(descending)
     part = torch.zeros(n, dtype=torch.int64, device=dev)
     for b in range(len(boundaries)):
         # row > boundary_b (lexicographically, honoring per-key direction)
@@ -226,10 +232,36 @@ def _range_partition_ids(rb: RecordBatch, keys: Sequence[str],
         eq = None  # equal so far
         for ki, kname in enumerate(keys):
             col = rb.column(kname)
-            bval = boundaries.column(kname).slice(b, b + 1).broadcast(n)
+            bcol = boundaries.column(kname).slice(b, b + 1)
+            bval = bcol.broadcast(n)
             op_gt = "lt" if descending[ki] else "gt"
-            g = col.compare(bval, op_gt).data
-            e = col.compare(bval, "eq").data
+            g = col.compare(bval, op_gt).data.clone()
+            e = col.compare(bval, "eq").data.clone()
+            row_null = (~col.validity) if col.validity is not None else None
+            b_null = bcol.validity is not None and \
+                not bool(bcol.validity.item())
+            if row_null is not None:
+                if b_null:
+                    # null vs null: equal
+                    g = torch.where(row_null, torch.zeros_like(g), g)
+                    e = torch.where(row_null, torch.ones_like(e), e)
+                else:
+                    # null row vs value: after iff nulls last
+                    val = not nulls_first[ki]
+                    g = torch.where(row_null,
+                                    torch.full_like(g, val), g)
+                    e = torch.where(row_null, torch.zeros_like(e), e)
+                if not b_null:
+                    pass
+            if b_null and row_null is None:
+                # value vs null boundary: after iff nulls first
+                g = torch.full_like(g, nulls_first[ki])
+                e = torch.zeros_like(e)
+            elif b_null and row_null is not None:
+                nonnull = ~row_null
+                g = torch.where(nonnull,
+                                torch.full_like(g, nulls_first[ki]), g)
+                e = torch.where(nonnull, torch.zeros_like(e), e)
             if gt is None:
                 gt, eq = g, e
             else:
