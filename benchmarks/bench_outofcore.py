@@ -9,6 +9,33 @@ table size.
 from __future__ import annotations
 
 import argparse
+
+
+# per-query column whitelists for --prune-columns (host RAM sizing for
+# SF1000; the engine's source-level projection pushdown does the same
+# pruning when reading full tables)
+_QCOLS = {
+    1: {"lineitem": ["l_quantity", "l_extendedprice", "l_discount",
+                     "l_tax", "l_returnflag", "l_linestatus",
+                     "l_shipdate"]},
+    6: {"lineitem": ["l_shipdate", "l_discount", "l_quantity",
+                     "l_extendedprice"]},
+    9: {"lineitem": ["l_partkey", "l_suppkey", "l_orderkey", "l_quantity",
+                     "l_extendedprice", "l_discount"],
+        "part": ["p_partkey", "p_name"],
+        "supplier": ["s_suppkey", "s_nationkey"],
+        "partsupp": ["ps_partkey", "ps_suppkey", "ps_supplycost"],
+        "orders": ["o_orderkey", "o_orderdate"],
+        "nation": ["n_nationkey", "n_name"]},
+}
+
+
+def _columns_for(qs):
+    out = {}
+    for q in qs:
+        for t, cols in _QCOLS.get(q, {}).items():
+            out.setdefault(t, set()).update(cols)
+    return {t: sorted(c) for t, c in out.items()} or None
 import json
 import time
 
@@ -20,6 +47,9 @@ def main():
     ap.add_argument("--sf", type=float, default=100.0)
     ap.add_argument("--queries", type=str, default="1,6")  # "all" = 1..22
     ap.add_argument("--morsel", type=int, default=1 << 26)
+    ap.add_argument("--prune-columns", action="store_true",
+                    help="stage only the columns the selected queries "
+                    "read (fits SF1000 lineitem in host RAM)")
     args = ap.parse_args()
 
     from benchmarks.tpch import datagen
@@ -30,7 +60,16 @@ def main():
     dev = "cuda:0" if torch.cuda.is_available() else "cpu"
 
     t0 = time.time()
-    tables = datagen.dataframes(args.sf, device="cpu")  # host RAM
+    qs_probe = list(range(1, 23)) if args.queries == "all" else \
+        [int(q) for q in args.queries.split(",")]
+    columns = None
+    if args.prune_columns:
+        columns = _columns_for(qs_probe)
+    if torch.cuda.is_available():
+        # GPU-sharded generation staged to host: the SF1000 path
+        tables = datagen.dataframes_host_staged(args.sf, columns=columns)
+    else:
+        tables = datagen.dataframes(args.sf, device="cpu")  # host RAM
     lineitem_rows = tables["lineitem"].count_rows()
     print(f"datagen sf={args.sf} on host: {time.time()-t0:.1f}s, "
           f"lineitem={lineitem_rows:,} rows")
