@@ -520,7 +520,9 @@ def dataframes_host_staged(sf: float, shards: int = 0, gen_device=None,
             if columns and name in columns:
                 keep = [c for c in rb.columns if c.name in columns[name]]
                 rb = type(rb)(keep, num_rows=len(rb))
-            parts.setdefault(name, []).append(rb.cpu())
+            parts.setdefault(name, []).append(
+                rb.cpu_pinned() if str(gen_device).startswith("cuda")
+                else rb.cpu())
         if str(gen_device).startswith("cuda"):
             _t.cuda.empty_cache()
     return {name: from_recordbatches(ps) for name, ps in parts.items()}

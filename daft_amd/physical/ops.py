@@ -120,8 +120,12 @@ def stream_host_batch(part, device, morsel: int):
     compute = _t.cuda.current_stream(device)
     staged = None        # (device_batch, event, pinned_keepalive)
 
+    already_pinned = part.is_pinned()
+
     def stage(lo: int):
-        pin = part.slice(lo, lo + morsel).pinned()
+        sl = part.slice(lo, lo + morsel)
+        # slices of pinned storage stay pinned: skip the staging copy
+        pin = sl if already_pinned else sl.pinned()
         with _t.cuda.stream(copy_stream):
             dev = pin.to(device, non_blocking=True)
             ev = _t.cuda.Event()

@@ -89,6 +89,13 @@ class RecordBatch:
         return RecordBatch([c.pinned() for c in self.columns],
                            self._num_rows)
 
+    def cpu_pinned(self) -> "RecordBatch":
+        return RecordBatch([c.cpu_pinned() for c in self.columns],
+                           self._num_rows)
+
+    def is_pinned(self) -> bool:
+        return all(c.is_pinned() for c in self.columns)
+
     def cpu(self) -> "RecordBatch":
         return self.to("cpu")
 

@@ -157,6 +157,30 @@ class Series:
                       [c.to(device, non_blocking) for c in self.children],
                       self.pyobjs, self._length)
 
+    def cpu_pinned(self) -> "Series":
+        """D2H straight into pinned host memory (one DMA, no pageable
+        staging) — host-staged tables stream back to HBM at full link
+        speed because morsel slices of pinned storage stay pinned."""
+        def mv(t):
+            if t is None:
+                return None
+            if not t.is_cuda:
+                return t.pin_memory() if not t.is_pinned() else t
+            out = torch.empty_like(t, device="cpu", pin_memory=True)
+            out.copy_(t)
+            return out
+        return Series(self.name, self.dtype, mv(self.data),
+                      mv(self.validity), mv(self.offsets),
+                      [c.cpu_pinned() for c in self.children], self.pyobjs,
+                      self._length)
+
+    def is_pinned(self) -> bool:
+        for t in (self.data, self.validity, self.offsets):
+            if t is not None:
+                if t.is_cuda or not t.is_pinned():
+                    return False
+        return all(c.is_pinned() for c in self.children)
+
     def pinned(self) -> "Series":
         """Copy host buffers into pinned (page-locked) memory so H2D
         transfers run as async DMA on a copy stream (out-of-core morsel
