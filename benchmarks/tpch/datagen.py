@@ -520,9 +520,14 @@ def dataframes_host_staged(sf: float, shards: int = 0, gen_device=None,
             if columns and name in columns:
                 keep = [c for c in rb.columns if c.name in columns[name]]
                 rb = type(rb)(keep, num_rows=len(rb))
+            # persistent pinned staging is opt-in: page-locking tens of
+            # GB has crashed boxes on this pool (transient morsel-sized
+            # pinning in stream_host_batch is always on and bounded)
+            import os as _os
+            pin = _os.environ.get("DAFT_AMD_PIN_STAGING") == "1"
             parts.setdefault(name, []).append(
-                rb.cpu_pinned() if str(gen_device).startswith("cuda")
-                else rb.cpu())
+                rb.cpu_pinned() if pin and
+                str(gen_device).startswith("cuda") else rb.cpu())
         if str(gen_device).startswith("cuda"):
             _t.cuda.empty_cache()
     return {name: from_recordbatches(ps) for name, ps in parts.items()}
