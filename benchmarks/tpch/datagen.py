@@ -517,6 +517,8 @@ def dataframes_host_staged(sf: float, shards: int = 0, gen_device=None,
     parts: Dict[str, list] = {}
     for r in range(shards):
         for name, rb in generate(sf, gen_device, r, shards).items():
+            if columns is not None and name not in columns:
+                continue       # query set never touches this table
             if columns and name in columns:
                 keep = [c for c in rb.columns if c.name in columns[name]]
                 rb = type(rb)(keep, num_rows=len(rb))
