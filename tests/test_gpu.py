@@ -639,3 +639,24 @@ def test_fused_expr_in_query_path():
                     assert x == pytest.approx(y, rel=1e-9), f"q{qi}.{k}"
                 else:
                     assert x == y, f"q{qi}.{k}"
+
+
+@pytest.mark.gpu
+def test_out_of_core_hbm_stays_bounded():
+    """Streaming a host-resident table keeps peak HBM far below the
+    table's size (the memory-manager guarantee behind SF1000)."""
+    from benchmarks.tpch import datagen
+    from benchmarks.tpch.queries import run_query
+    from daft_amd.context import get_context
+    get_context().execution_config.stream_morsel_rows = 1 << 24
+    T = datagen.dataframes(5.0, device="cpu")   # ~30M lineitem rows, host
+    li_bytes = 0
+    for p in T["lineitem"].collect()._result:
+        li_bytes += p.size_bytes()
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    out = run_query(6, T, sf=5.0).to_pydict()
+    peak = torch.cuda.max_memory_allocated()
+    assert out["revenue"][0] > 0
+    # morsel streaming must not materialize the table in HBM
+    assert peak < max(li_bytes // 2, 2 << 30), (peak, li_bytes)
