@@ -74,6 +74,35 @@ class LocalCheckpointStore(CheckpointStore):
         os.replace(tmp, path)  # atomic commit (IdempotentCommit analog)
 
 
+class ObjectStoreCheckpointStore(CheckpointStore):
+    """Processed-key sets as JSONL objects under an s3://-style prefix —
+    the reference's object-store checkpoint (daft-checkpoint
+    impls/s3.rs + keys_codec.rs) over daft_amd.io.object_store, with the
+    same atomic per-commit-object idempotent layout."""
+
+    def __init__(self, prefix: str, io_config=None):
+        from .io.object_store import get_source
+        self.prefix = prefix.rstrip("/") + "/"
+        self.src = get_source(self.prefix, io_config)
+
+    def committed_keys(self) -> Set:
+        out: Set = set()
+        for path, _sz in sorted(self.src.list_prefix(self.prefix)):
+            if not path.endswith(".jsonl"):
+                continue
+            for line in self.src.get(path).decode().splitlines():
+                if line.strip():
+                    out.update(json.loads(line))
+        return out
+
+    def commit(self, keys: Iterable) -> None:
+        keys = list(keys)
+        if not keys:
+            return
+        path = f"{self.prefix}keys_{uuid.uuid4().hex[:12]}.jsonl"
+        self.src.put(path, (json.dumps(keys) + "\n").encode())
+
+
 class CheckpointConfig:
     def __init__(self, store: CheckpointStore, on: str):
         self.store = store
