@@ -183,7 +183,7 @@ def test_checkpoint_on_object_store(s3):
     second = ck.filter_processed(df).sort("id").to_pydict()
     assert second["id"] == [3, 4]
     # keys persisted as objects
-    assert any(k.startswith("ckpt/") and k.endswith(".jsonl")
+    assert any(k.startswith("bkt/ckpt/") and k.endswith(".jsonl")
                for k in srv.objects)
     # idempotent double commit
     ck.commit(df)
