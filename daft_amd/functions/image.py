@@ -48,6 +48,64 @@ def _image_struct(name, datas, heights, widths, channels, mode_codes,
     return s.to(device) if str(device) != "cpu" else s
 
 
+def _decode_chunk(args):
+    """Process-pool worker: decode a chunk of encoded images into ONE
+    packed byte buffer + dims (no per-image array pickling)."""
+    blobs, mode, on_error = args
+    import io as _io
+
+    import numpy as _np
+    from PIL import Image as _PIL
+    parts = []
+    dims = []
+    for v in blobs:
+        if v is None:
+            dims.append((0, 0, 0, False))
+            continue
+        try:
+            img = _PIL.open(_io.BytesIO(v)).convert(mode)
+            arr = _np.asarray(img, dtype=_np.uint8)
+            if arr.ndim == 2:
+                arr = arr[:, :, None]
+            parts.append(arr.tobytes())
+            dims.append(arr.shape + (True,))
+        except Exception:
+            if on_error == "raise":
+                raise
+            dims.append((0, 0, 0, False))
+    return b"".join(parts), dims
+
+
+_DECODE_POOL = None
+
+
+def _decode_pool(vals, mode, on_error):
+    """Chunked process-pool decode; returns per-image uint8 arrays."""
+    global _DECODE_POOL
+    import concurrent.futures as fut
+    import multiprocessing as mp
+    workers = min(96, max(8, (os.cpu_count() or 8) // 2))
+    if _DECODE_POOL is None:
+        _DECODE_POOL = fut.ProcessPoolExecutor(
+            max_workers=workers, mp_context=mp.get_context("fork"))
+    chunk = max(64, len(vals) // (workers * 4))
+    chunks = [vals[i:i + chunk] for i in range(0, len(vals), chunk)]
+    outs = list(_DECODE_POOL.map(
+        _decode_chunk, [(c, mode, on_error) for c in chunks]))
+    arrs = []
+    for packed, dims in outs:
+        off = 0
+        buf = np.frombuffer(packed, dtype=np.uint8)
+        for h, w, c, ok in dims:
+            if not ok:
+                arrs.append(None)
+                continue
+            nb = h * w * c
+            arrs.append(buf[off:off + nb].reshape(h, w, c))
+            off += nb
+    return arrs
+
+
 def decode_series(s: Series, mode: str = "RGB",
                   on_error: str = "raise") -> Series:
     """binary (encoded JPEG/PNG/...) -> Image struct.
@@ -72,10 +130,17 @@ def decode_series(s: Series, mode: str = "RGB",
                 raise
             return None
 
-    if len(vals) >= 64:
+    ncpu = os.cpu_count() or 8
+    if len(vals) >= 4096 and ncpu >= 16:
+        # decode-pool path: chunked PROCESS pool sidesteps the GIL-held
+        # numpy/packing phase entirely — on a 256-core host this is the
+        # "pinned decode pool saturating the link" design (each worker
+        # returns one packed buffer, not per-image arrays)
+        arrs = _decode_pool(vals, mode, on_error)
+    elif len(vals) >= 64:
         # Pillow's codecs release the GIL; 3x oversubscription hides the
         # GIL-held numpy-conversion phase (measured +17% vs 1x on 8 cores)
-        workers = min(64, 3 * (os.cpu_count() or 8))
+        workers = min(64, 3 * ncpu)
         with fut.ThreadPoolExecutor(max_workers=workers) as ex:
             arrs = list(ex.map(one, vals))
     else:
