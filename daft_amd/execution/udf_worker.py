@@ -118,3 +118,78 @@ def _shutdown():
         for w in _workers.values():
             w.stop()
         _workers.clear()
+
+
+# ---------------------------------------------------------------------------
+# batched UDF subprocess: whole Series travel through torch.multiprocessing
+# queues, so device tensors move ZERO-COPY via CUDA/dmabuf IPC handles and
+# CPU tensors via shared memory (ref: intermediate_ops/udf.rs:351-406 —
+# out-of-process batched UDFs; HSA_ENABLE_IPC_MODE_LEGACY=0 selects dmabuf
+# IPC on this pool)
+# ---------------------------------------------------------------------------
+
+def _batched_worker_main(inq, outq, fn_blob: bytes):
+    import cloudpickle
+    try:
+        fn = cloudpickle.loads(fn_blob)
+    except Exception as e:  # pragma: no cover
+        outq.put(("init_error", repr(e)))
+        return
+    outq.put(("ready", None))
+    while True:
+        msg = inq.get()
+        if msg[0] == "stop":
+            return
+        _, args = msg
+        try:
+            out = fn(*args)
+            outq.put(("ok", out))
+        except Exception as e:
+            outq.put(("error", repr(e)))
+
+
+class _BatchedWorker:
+    def __init__(self, fn):
+        import cloudpickle
+        import torch.multiprocessing as tmp
+        ctx = tmp.get_context("spawn")
+        self._inq = ctx.Queue()
+        self._outq = ctx.Queue()
+        self._proc = ctx.Process(
+            target=_batched_worker_main,
+            args=(self._inq, self._outq, cloudpickle.dumps(fn)),
+            daemon=True)
+        self._proc.start()
+        status, payload = self._outq.get()
+        if status != "ready":
+            raise RuntimeError(f"batched UDF worker failed: {payload}")
+        self._lock = threading.Lock()
+
+    def call(self, args: list):
+        with self._lock:
+            self._inq.put(("batch", args))
+            status, payload = self._outq.get()
+        if status == "error":
+            raise RuntimeError(f"batched UDF worker error: {payload}")
+        return payload
+
+    def stop(self):
+        try:
+            self._inq.put(("stop", None))
+        except Exception:
+            pass
+        self._proc.join(timeout=5)
+        if self._proc.is_alive():
+            self._proc.terminate()
+
+
+_BATCHED: Dict[int, _BatchedWorker] = {}
+
+
+def get_batched_worker(fn) -> _BatchedWorker:
+    w = _BATCHED.get(id(fn))
+    if w is None:
+        w = _BatchedWorker(fn)
+        _BATCHED[id(fn)] = w
+        atexit.register(w.stop)
+    return w

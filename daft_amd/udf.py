@@ -162,7 +162,14 @@ def run_udf_node(node: PyUDF, batch) -> Series:
                                           node.return_dtype,
                                           device=batch.device)
             if node.batched:
-                out = node.fn(*arg_series)
+                if node.use_process:
+                    # subprocess isolation with zero-copy device handoff:
+                    # Series tensors cross via torch.multiprocessing
+                    # (CUDA/dmabuf IPC on GPU, shared memory on CPU)
+                    from .execution.udf_worker import get_batched_worker
+                    out = get_batched_worker(node.fn).call(list(arg_series))
+                else:
+                    out = node.fn(*arg_series)
                 if isinstance(out, Series):
                     return out.rename(node.name).cast(node.return_dtype) \
                         if out.dtype != node.return_dtype else \

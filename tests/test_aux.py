@@ -865,3 +865,37 @@ def test_topn_bounded_fold():
     df = daft.from_pydict({"v": vals}).into_batches(2048)
     out = df.sort("v", desc=True).limit(25).to_pydict()
     assert out["v"] == sorted(vals, reverse=True)[:25]
+
+
+def test_batched_udf_subprocess():
+    """@daft.func(batched=True, use_process=True): whole Series cross the
+    process boundary through torch.multiprocessing (shared memory on CPU,
+    CUDA/dmabuf IPC on GPU — see test_gpu.py for the device variant)."""
+    import os as _os
+    parent_pid = _os.getpid()
+
+    @daft.func(return_dtype=DataType.float64(), batched=True,
+               use_process=True)
+    def scaled(x):
+        import os
+        import torch as _t
+        assert os.getpid() != 0
+        return _t.as_tensor(x.data, dtype=_t.float64) * 2.5
+
+    df = daft.from_pydict({"x": [1.0, 2.0, 3.0, 4.0]})
+    out = df.select(scaled(col("x")).alias("y")).to_pydict()
+    assert out["y"] == [2.5, 5.0, 7.5, 10.0]
+
+
+def test_batched_udf_subprocess_isolation():
+    """A crash-prone batched UDF in a subprocess doesn't take the engine
+    down; on_error surfaces as an error."""
+    @daft.func(return_dtype=DataType.int64(), batched=True,
+               use_process=True)
+    def boom(x):
+        raise ValueError("kaboom")
+
+    df = daft.from_pydict({"x": [1, 2]})
+    import pytest as _pt
+    with _pt.raises(Exception, match="kaboom"):
+        df.select(boom(col("x"))).to_pydict()

@@ -660,3 +660,19 @@ def test_out_of_core_hbm_stays_bounded():
     assert out["revenue"][0] > 0
     # morsel streaming must not materialize the table in HBM
     assert peak < max(li_bytes // 2, 2 << 30), (peak, li_bytes)
+
+
+@pytest.mark.gpu
+def test_batched_udf_subprocess_device_tensors():
+    """Batched subprocess UDF with tensors STAYING on device: the Series
+    crosses via CUDA(dmabuf) IPC, the worker computes on cuda:0 and the
+    result tensor returns without a host round trip."""
+    @daft.func(return_dtype=DataType.float64(), batched=True,
+               use_process=True)
+    def dev_scale(x):
+        assert x.data.is_cuda, "worker must receive a device tensor"
+        return x.data * 3.0
+
+    df = daft.from_pydict({"x": [1.0, 2.0, 3.0]}, device="cuda:0")
+    out = df.select(dev_scale(col("x")).alias("y")).to_pydict()
+    assert out["y"] == [3.0, 6.0, 9.0]
