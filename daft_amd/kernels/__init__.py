@@ -802,6 +802,31 @@ def is_in(s: Series, values: Series) -> Series:
         out = vmask.data[s.data.to(torch.int64)]
         return Series(s.name, DataType.bool(), data=out,
                       validity=s.validity)
+    # numeric/temporal fast path: one sorted-table binary search instead
+    # of a compare kernel per value (the O(n·k) loop was a flagged weak
+    # spot; strings keep the loop — the dict path above covers hot cases)
+    if len(values) > 4 and s.data is not None and s.offsets is None and \
+            not s.children and values.data is not None and \
+            values.offsets is None and not values.children and \
+            s.data.dtype == values.data.dtype and \
+            s.data.dtype not in (torch.bool,):
+        vd = values.data
+        if values.validity is not None:
+            vd = vd[values.validity]
+        view = {torch.uint16: torch.int16, torch.uint32: torch.int32,
+                torch.uint64: torch.int64}.get(vd.dtype)
+        sd = s.data
+        if view is not None:
+            vd, sd = vd.view(view), sd.view(view)
+        table = torch.unique(vd.to(s.device))
+        if table.numel() == 0:
+            data = torch.zeros(len(s), dtype=torch.bool, device=s.device)
+        else:
+            pos = torch.searchsorted(table, sd.contiguous())
+            pos = pos.clamp(max=table.numel() - 1)
+            data = table[pos] == sd
+        return Series(s.name, DataType.bool(), data=data,
+                      validity=s.validity)
     out = None
     for i in range(len(values)):
         v = values.slice(i, i + 1)

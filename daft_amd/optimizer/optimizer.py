@@ -35,7 +35,8 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
         reorder_joins = lambda p: p  # noqa: E731
     reorder_after = 1  # join reordering once filters sit at the sources
     for bi, rules in enumerate(batches):
-        for _ in range(8):  # fixed-point cap
+        changed = False
+        for _round in range(8):  # fixed-point cap
             changed = False
             for rule in rules:
                 new = _apply_bottom_up(plan, rule)
@@ -44,6 +45,14 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
                     changed = True
             if not changed:
                 break
+        if changed:
+            # convergence telemetry: a batch still rewriting at the cap
+            # means deep plans may be silently under-optimized
+            import warnings
+            names = [getattr(r, "__name__", "?") for r in rules]
+            warnings.warn(
+                f"optimizer batch {bi} ({names}) hit the fixed-point cap "
+                f"(8 rounds) while still rewriting", RuntimeWarning)
         if bi == reorder_after:
             plan = reorder_joins(plan)
     return plan

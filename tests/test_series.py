@@ -217,3 +217,22 @@ def test_decimal_exact_storage_and_ops(tmp_path):
     s = daft.from_pydict({"v": [D("1.10")]}).select(
         col("v").cast(daft.DataType.string()).alias("s")).to_pydict()["s"]
     assert s == ["1.10"]
+
+
+def test_is_in_sorted_table_path():
+    import random
+    random.seed(9)
+    vals = [random.randint(0, 1000) if i % 11 else None
+            for i in range(5000)]
+    members = [3, 77, 500, 999, 123, 456, 789]   # >4 -> searchsorted path
+    s = Series.from_pylist("x", vals, DataType.int64())
+    m = Series.from_pylist("v", members, DataType.int64())
+    got = s.is_in(m).to_pylist()
+    want = [None if v is None else (v in set(members)) for v in vals]
+    assert got == want
+    # floats too
+    fvals = [float(v) if v is not None else None for v in vals]
+    sf = Series.from_pylist("x", fvals, DataType.float64())
+    mf = Series.from_pylist("v", [float(x) for x in members],
+                            DataType.float64())
+    assert sf.is_in(mf).to_pylist() == want
