@@ -50,14 +50,17 @@ def _image_struct(name, datas, heights, widths, channels, mode_codes,
 
 def _decode_chunk(args):
     """Process-pool worker: decode a chunk of encoded images into ONE
-    packed byte buffer + dims (no per-image array pickling)."""
+    shared-memory segment + dims — only (shm name, dims) crosses the
+    pipe, so the parent never unpickles pixel payloads."""
     blobs, mode, on_error = args
     import io as _io
+    from multiprocessing import shared_memory as _shm
 
     import numpy as _np
     from PIL import Image as _PIL
     parts = []
     dims = []
+    total = 0
     for v in blobs:
         if v is None:
             dims.append((0, 0, 0, False))
@@ -67,13 +70,30 @@ def _decode_chunk(args):
             arr = _np.asarray(img, dtype=_np.uint8)
             if arr.ndim == 2:
                 arr = arr[:, :, None]
-            parts.append(arr.tobytes())
+            parts.append(arr)
             dims.append(arr.shape + (True,))
+            total += arr.nbytes
         except Exception:
             if on_error == "raise":
                 raise
             dims.append((0, 0, 0, False))
-    return b"".join(parts), dims
+    seg = _shm.SharedMemory(create=True, size=max(total, 1))
+    off = 0
+    for arr in parts:
+        nb = arr.nbytes
+        _np.frombuffer(seg.buf, dtype=_np.uint8,
+                       count=nb, offset=off)[:] = arr.reshape(-1)
+        off += nb
+    seg_name = seg.name
+    seg.close()
+    # the parent owns the unlink; stop this worker's resource tracker
+    # from double-reporting the segment at shutdown
+    try:
+        from multiprocessing import resource_tracker as _rt
+        _rt.unregister("/" + seg_name, "shared_memory")
+    except Exception:
+        pass
+    return seg_name, total, dims
 
 
 _DECODE_POOL = None
@@ -94,8 +114,20 @@ def _decode_pool_struct(name, vals, mode, on_error, device):
     chunks = [vals[i:i + chunk] for i in range(0, len(vals), chunk)]
     outs = list(_DECODE_POOL.map(
         _decode_chunk, [(c, mode, on_error) for c in chunks]))
-    blob = b"".join(p for p, _d in outs)
-    dims = np.array([d for _p, ds in outs for d in ds], dtype=np.int64)
+    from multiprocessing import shared_memory as _shmmod
+    total = sum(t for _n, t, _d in outs)
+    blob_t = torch.empty(max(total, 0), dtype=torch.uint8)
+    blob_np = blob_t.numpy()
+    off = 0
+    for seg_name, t, _d in outs:
+        seg = _shmmod.SharedMemory(name=seg_name)
+        if t:
+            blob_np[off:off + t] = np.frombuffer(seg.buf, dtype=np.uint8,
+                                                 count=t)
+        off += t
+        seg.close()
+        seg.unlink()
+    dims = np.array([d for _n, _t, ds in outs for d in ds], dtype=np.int64)
     if dims.size == 0:
         dims = dims.reshape(0, 4)
     hs, ws, cs, ok = dims[:, 0], dims[:, 1], dims[:, 2], dims[:, 3]
@@ -104,8 +136,7 @@ def _decode_pool_struct(name, vals, mode, on_error, device):
     np.cumsum(nbytes, out=offs[1:])
     data_child = Series(
         "data", DataType.binary(),
-        data=torch.frombuffer(bytearray(blob), dtype=torch.uint8)
-        if blob else torch.zeros(0, dtype=torch.uint8),
+        data=blob_t,
         offsets=torch.from_numpy(offs))
     ch = Series("channel", DataType.uint16(),
                 data=torch.from_numpy(cs.astype(np.int16)).view(torch.uint16))
