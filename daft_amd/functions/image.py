@@ -49,20 +49,27 @@ def _image_struct(name, datas, heights, widths, channels, mode_codes,
 
 
 def _decode_chunk(args):
-    """Process-pool worker: decode a chunk of encoded images into ONE
-    shared-memory segment + dims — only (shm name, dims) crosses the
-    pipe, so the parent never unpickles pixel payloads."""
-    blobs, mode, on_error = args
+    """Process-pool worker: inputs AND outputs travel via shared memory —
+    only (segment name, offsets, dims) cross the pipe, so neither side
+    ever pickles pixel payloads."""
+    in_name, in_offs, valid_mask, mode, on_error = args
     import io as _io
     from multiprocessing import shared_memory as _shm
 
     import numpy as _np
     from PIL import Image as _PIL
+    src = _shm.SharedMemory(name=in_name)
+    buf = _np.frombuffer(src.buf, dtype=_np.uint8)
     parts = []
     dims = []
     total = 0
-    for v in blobs:
-        if v is None:
+    k = len(in_offs) - 1
+    for i in range(k):
+        if valid_mask is not None and not valid_mask[i]:
+            dims.append((0, 0, 0, False))
+            continue
+        v = bytes(buf[in_offs[i]:in_offs[i + 1]])
+        if len(v) == 0:
             dims.append((0, 0, 0, False))
             continue
         try:
@@ -77,6 +84,8 @@ def _decode_chunk(args):
             if on_error == "raise":
                 raise
             dims.append((0, 0, 0, False))
+    del buf            # release the exported view before closing the mmap
+    src.close()
     seg = _shm.SharedMemory(create=True, size=max(total, 1))
     off = 0
     for arr in parts:
@@ -99,22 +108,43 @@ def _decode_chunk(args):
 _DECODE_POOL = None
 
 
-def _decode_pool_struct(name, vals, mode, on_error, device):
-    """Chunked process-pool decode straight into the Image struct: workers
-    return packed buffers; the parent concatenates buffers and builds
-    offsets vectorized — no per-image Python objects anywhere."""
+def _decode_pool_struct(series, mode, on_error, device):
+    """Chunked process-pool decode straight into the Image struct: the
+    input binary column's contiguous buffer is published ONCE as a
+    shared-memory segment; workers decode their row ranges from it and
+    return their pixels in shared memory — no per-image Python objects
+    and no payload pickling in either direction."""
     global _DECODE_POOL
     import concurrent.futures as fut
     import multiprocessing as mp
+    from multiprocessing import shared_memory as _shmmod
+    name = series.name
+    cpu = series.cpu()
+    data_np = cpu.data.contiguous().numpy() if cpu.data is not None         else np.zeros(0, dtype=np.uint8)
+    offs_np = cpu.offsets.numpy()
+    valid_np = cpu.validity.numpy() if cpu.validity is not None else None
+    n = len(series)
     workers = min(96, max(8, (os.cpu_count() or 8) // 2))
     if _DECODE_POOL is None:
         _DECODE_POOL = fut.ProcessPoolExecutor(
             max_workers=workers, mp_context=mp.get_context("fork"))
-    chunk = max(64, len(vals) // (workers * 4))
-    chunks = [vals[i:i + chunk] for i in range(0, len(vals), chunk)]
-    outs = list(_DECODE_POOL.map(
-        _decode_chunk, [(c, mode, on_error) for c in chunks]))
-    from multiprocessing import shared_memory as _shmmod
+    shm_in = _shmmod.SharedMemory(create=True,
+                                  size=max(int(data_np.nbytes), 1))
+    np.frombuffer(shm_in.buf, dtype=np.uint8,
+                  count=data_np.nbytes)[:] = data_np
+    chunk = max(64, n // (workers * 4))
+    jobs = []
+    for lo in range(0, n, chunk):
+        hi = min(lo + chunk, n)
+        jobs.append((shm_in.name, offs_np[lo:hi + 1].copy(),
+                     valid_np[lo:hi].copy() if valid_np is not None
+                     else None, mode, on_error))
+    try:
+        outs = list(_DECODE_POOL.map(_decode_chunk, jobs))
+    finally:
+        shm_in.close()
+        shm_in.unlink()
+    vals = [None] * n   # only used for length below
     total = sum(t for _n, t, _d in outs)
     blob_t = torch.empty(max(total, 0), dtype=torch.uint8)
     blob_np = blob_t.numpy()
@@ -165,6 +195,14 @@ def decode_series(s: Series, mode: str = "RGB",
     decoded pixels land in HBM for the HIP resize/tensor kernels."""
     import concurrent.futures as fut
     from PIL import Image as PILImage
+
+    ncpu = os.cpu_count() or 8
+    if len(s) >= 4096 and ncpu >= 16 and s.pyobjs is None and \
+            s.offsets is not None and not s.is_dict():
+        # decode-pool path: chunked PROCESS pool with shared-memory
+        # transport in both directions — the "decode pool saturating the
+        # link" design on many-core hosts
+        return _decode_pool_struct(s, mode, on_error, s.device)
     vals = s.cpu().to_pylist()
 
     def one(v):
@@ -181,13 +219,6 @@ def decode_series(s: Series, mode: str = "RGB",
                 raise
             return None
 
-    ncpu = os.cpu_count() or 8
-    if len(vals) >= 4096 and ncpu >= 16:
-        # decode-pool path: chunked PROCESS pool sidesteps the GIL-held
-        # numpy/packing phase entirely — on a 256-core host this is the
-        # "pinned decode pool saturating the link" design (each worker
-        # returns one packed buffer, not per-image arrays)
-        return _decode_pool_struct(s.name, vals, mode, on_error, s.device)
     if len(vals) >= 64:
         # Pillow's codecs release the GIL; 3x oversubscription hides the
         # GIL-held numpy-conversion phase (measured +17% vs 1x on 8 cores)
