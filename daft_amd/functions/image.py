@@ -196,9 +196,22 @@ def decode_series(s: Series, mode: str = "RGB",
     import concurrent.futures as fut
     from PIL import Image as PILImage
 
+    if s.is_dict():
+        # repeated blobs (dict-encoded binary): decode each DISTINCT image
+        # once, then gather the struct by code
+        vocab = decode_series(s.children[0].rename(s.name), mode, on_error)
+        out = vocab.take(s.data.to(torch.int64), has_neg=False)
+        if s.validity is not None:
+            v = out.validity & s.validity if out.validity is not None \
+                else s.validity
+            out = Series(out.name, out.dtype, data=out.data,
+                         validity=v, offsets=out.offsets,
+                         children=out.children, length=len(out))
+        return out
+
     ncpu = os.cpu_count() or 8
     if len(s) >= 4096 and ncpu >= 16 and s.pyobjs is None and \
-            s.offsets is not None and not s.is_dict():
+            s.offsets is not None:
         # decode-pool path: chunked PROCESS pool with shared-memory
         # transport in both directions — the "decode pool saturating the
         # link" design on many-core hosts
