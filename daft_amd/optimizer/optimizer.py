@@ -256,7 +256,11 @@ def push_down_filter(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     child = plan.children[0]
     preds = _split_conjunctions(plan.predicate)
     if not getattr(plan, "_or_derived", False):
-        extra = _derive_or_implications(preds)
+        # rebuilt Filter nodes lose the marker: dedupe derived conjuncts
+        # by repr or the rule re-derives forever (q7/q19 hit the cap)
+        have = {repr(p) for p in preds}
+        extra = [e for e in _derive_or_implications(preds)
+                 if repr(e) not in have]
         if extra:
             new = lp.Filter(child, _conjoin(preds + extra))
             new._or_derived = True
@@ -369,8 +373,21 @@ def push_down_anti_semi_join(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     for e in plan.left_on:
         key_refs.update(e.column_refs())
     if isinstance(left, lp.Filter):
-        return lp.Filter(
-            plan.with_children([left.children[0], q]), left.predicate)
+        # only hop over the filter when an inner join underneath can host
+        # the semi — otherwise this rule and filter pushdown swap the two
+        # nodes forever (observed: batch-1 fixed-point cap on q21)
+        below = left.children[0]
+        while isinstance(below, lp.Filter):
+            below = below.children[0]
+        rout_vals = set(dict(below.right_passthrough()).values()) \
+            if isinstance(below, lp.Join) else set()
+        if isinstance(below, lp.Join) and below.how == "inner" and \
+                key_refs and (
+                key_refs <= set(below.children[0].schema.names()) or
+                key_refs <= rout_vals):
+            return lp.Filter(
+                plan.with_children([left.children[0], q]), left.predicate)
+        return None
     if isinstance(left, lp.Join) and left.how == "inner":
         a, b = left.children
         if key_refs and key_refs <= set(a.schema.names()):
