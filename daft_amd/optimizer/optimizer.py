@@ -22,6 +22,9 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
     from .join_reorder import reorder_joins
     disabled = set((os.environ.get("DAFT_AMD_DISABLE_RULES") or "")
                    .split(","))
+    once = _apply_bottom_up(plan, derive_or_implications_rule)
+    if once is not None:
+        plan = once
     batches: List[List[Rule]] = [
         [simplify_expressions],
         [push_down_filter, drop_repartition] +
@@ -255,17 +258,6 @@ def push_down_filter(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
         return None
     child = plan.children[0]
     preds = _split_conjunctions(plan.predicate)
-    if not getattr(plan, "_or_derived", False):
-        # rebuilt Filter nodes lose the marker: dedupe derived conjuncts
-        # by repr or the rule re-derives forever (q7/q19 hit the cap)
-        have = {repr(p) for p in preds}
-        extra = [e for e in _derive_or_implications(preds)
-                 if repr(e) not in have]
-        if extra:
-            new = lp.Filter(child, _conjoin(preds + extra))
-            new._or_derived = True
-            return new
-        plan._or_derived = True
 
     if isinstance(child, lp.Filter):
         # merge adjacent filters
@@ -354,6 +346,22 @@ def push_down_filter(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
                 plan.predicate, child.pushdown_limit, child.read_options)
             return lp.Filter(new_scan, plan.predicate)
         return None
+    return None
+
+
+def derive_or_implications_rule(plan):
+    """Single-pass rule (NOT in a fixed-point batch: the derived conjuncts
+    get pushed away from this node and would be re-derived forever):
+    `(a=x and ..) or (a=y and ..)` implies `a IN (x, y)`, which CAN cross
+    joins the OR itself cannot (q7/q19)."""
+    if not isinstance(plan, lp.Filter):
+        return None
+    preds = _split_conjunctions(plan.predicate)
+    have = {repr(p) for p in preds}
+    extra = [e for e in _derive_or_implications(preds)
+             if repr(e) not in have]
+    if extra:
+        return lp.Filter(plan.children[0], _conjoin(preds + extra))
     return None
 
 
