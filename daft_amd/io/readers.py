@@ -65,10 +65,12 @@ def infer_schema(path: str, file_format: str,
 
 def read_file(path: str, file_format: str, columns: Optional[List[str]],
               predicate, limit: Optional[int], storage_options: dict,
-              read_options: dict, device) -> Iterator[RecordBatch]:
+              read_options: dict, device,
+              row_groups=None) -> Iterator[RecordBatch]:
     path = _open_input(path, storage_options)
     if file_format == "parquet":
-        yield from _read_parquet(path, columns, limit, device, predicate)
+        yield from _read_parquet(path, columns, limit, device, predicate,
+                                 row_groups=row_groups)
     elif file_format == "csv":
         yield from _read_csv(path, columns, read_options, device)
     elif file_format == "json":
@@ -77,6 +79,56 @@ def read_file(path: str, file_format: str, columns: Optional[List[str]],
         yield from _read_warc(path, columns, device)
     else:
         raise ValueError(f"unknown format {file_format}")
+
+
+def plan_scan_tasks(paths: List[str], file_format: str, storage_options,
+                    min_bytes: int = 96 * 1024 * 1024,
+                    max_bytes: int = 384 * 1024 * 1024):
+    """Split/merge files into scan tasks (ref: daft-scan
+    src/scan_task_iters/ with the 96-384 MB defaults from
+    common/daft-config): a parquet file larger than max_bytes splits into
+    row-group ranges; consecutive small files merge into one task so the
+    prefetch window stays busy.  A task is (paths, rg_ranges) where
+    rg_ranges is None (whole files) or, for a single-path split task, a
+    list of row-group indices."""
+    if file_format != "parquet":
+        return [([p], None) for p in paths]
+    import pyarrow.parquet as pq
+    tasks = []
+    pending_paths: List[str] = []
+    pending_bytes = 0
+    for path in paths:
+        try:
+            src = _open_input(path, storage_options)
+            md = pq.ParquetFile(src).metadata
+            fbytes = sum(md.row_group(i).total_byte_size
+                         for i in range(md.num_row_groups))
+        except Exception:
+            fbytes = None
+        if fbytes is None or fbytes <= max_bytes:
+            pending_paths.append(path)
+            pending_bytes += fbytes or max_bytes
+            if pending_bytes >= min_bytes:
+                tasks.append((pending_paths, None))
+                pending_paths, pending_bytes = [], 0
+            continue
+        if pending_paths:
+            tasks.append((pending_paths, None))
+            pending_paths, pending_bytes = [], 0
+        # split by row groups into ~max_bytes chunks
+        rgs: List[int] = []
+        acc = 0
+        for i in range(md.num_row_groups):
+            rgs.append(i)
+            acc += md.row_group(i).total_byte_size
+            if acc >= max_bytes:
+                tasks.append(([path], list(rgs)))
+                rgs, acc = [], 0
+        if rgs:
+            tasks.append(([path], list(rgs)))
+    if pending_paths:
+        tasks.append((pending_paths, None))
+    return tasks
 
 
 def read_files_prefetch(paths: List[str], file_format: str,
@@ -94,13 +146,20 @@ def read_files_prefetch(paths: List[str], file_format: str,
     import concurrent.futures as fut
     from collections import deque
 
-    def load(path):
-        return list(read_file(path, file_format, columns, predicate, None,
-                              storage_options, read_options, "cpu"))
+    tasks = plan_scan_tasks(paths, file_format, storage_options)
+
+    def load(task):
+        tpaths, rgs = task
+        out = []
+        for pth in tpaths:
+            out.extend(read_file(pth, file_format, columns, predicate,
+                                 None, storage_options, read_options,
+                                 "cpu", row_groups=rgs))
+        return out
 
     with fut.ThreadPoolExecutor(max_workers=window) as ex:
         pending: deque = deque()
-        it = iter(paths)
+        it = iter(tasks)
         for _ in range(window):
             p = next(it, None)
             if p is None:
@@ -123,14 +182,15 @@ def read_files_prefetch(paths: List[str], file_format: str,
 
 
 def _read_parquet(path, columns, limit, device,
-                  predicate=None) -> Iterator[RecordBatch]:
+                  predicate=None, row_groups=None) -> Iterator[RecordBatch]:
     import pyarrow.parquet as pq
     f = pq.ParquetFile(path)
     remaining = limit
     bounds = _predicate_bounds(predicate)
     name_to_idx = {c: i for i, c in enumerate(f.schema_arrow.names)} \
         if bounds else {}
-    for rg in range(f.num_row_groups):
+    for rg in (row_groups if row_groups is not None
+               else range(f.num_row_groups)):
         if remaining is not None and remaining <= 0:
             return
         if bounds and not _rg_may_match(f.metadata.row_group(rg),
