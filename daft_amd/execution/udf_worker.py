@@ -183,13 +183,16 @@ class _BatchedWorker:
             self._proc.terminate()
 
 
-_BATCHED: Dict[int, _BatchedWorker] = {}
+_BATCHED: Dict[int, tuple] = {}
 
 
 def get_batched_worker(fn) -> _BatchedWorker:
-    w = _BATCHED.get(id(fn))
-    if w is None:
-        w = _BatchedWorker(fn)
-        _BATCHED[id(fn)] = w
-        atexit.register(w.stop)
+    # key by id() but keep a strong reference to fn: a collected function
+    # can otherwise reuse the address and hijack another UDF's worker
+    entry = _BATCHED.get(id(fn))
+    if entry is not None and entry[0] is fn:
+        return entry[1]
+    w = _BatchedWorker(fn)
+    _BATCHED[id(fn)] = (fn, w)
+    atexit.register(w.stop)
     return w
