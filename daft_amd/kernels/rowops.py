@@ -698,11 +698,79 @@ def _stable_sort_perm_by(keys: torch.Tensor) -> torch.Tensor:
     return torch.from_numpy(np.argsort(u, kind="stable").astype(np.int64))
 
 
+def _key_bits(s: Series):
+    """(biased int64 key ascending-ordered, bit width) or None when the
+    column can't compose into a packed radix key."""
+    if s.is_dict():
+        vocab = s.children[0]
+        vperm = argsort_multi([vocab], [False], [False])
+        rank = torch.empty(len(vocab), dtype=torch.int64, device=s.device)
+        rank[vperm] = torch.arange(len(vocab), dtype=torch.int64,
+                                   device=s.device)
+        k = rank[s.data.to(torch.int64)]
+        card = max(len(vocab), 1)
+        return k, max(card - 1, 1).bit_length()
+    dt = s.dtype
+    if not (dt.is_integer() or dt.is_boolean() or dt.is_temporal()):
+        return None
+    if s.data is None or s.data.dtype == torch.uint64:
+        return None
+    v = s.data.to(torch.int64)
+    if s.validity is not None:
+        v = torch.where(s.validity, v, torch.zeros_like(v))
+    lo = int(v.min().item()) if v.numel() else 0
+    hi = int(v.max().item()) if v.numel() else 0
+    span = hi - lo
+    if span < 0 or span >= (1 << 62):
+        return None
+    return v - lo, max(span, 1).bit_length()
+
+
+def _try_composed_argsort(keys, descending, nulls_first):
+    """Pack every key (plus a null bit where needed) into ONE u64 and run
+    a single radix sort instead of one full sort per key (SURVEY §2.5
+    key-composition; the per-key LSD loop below is the fallback)."""
+    if len(keys) < 2:
+        return None
+    parts = []
+    total_bits = 0
+    for s, desc, nf in zip(keys, descending, nulls_first):
+        kb = _key_bits(s)
+        if kb is None:
+            return None
+        k, bits = kb
+        if desc:
+            k = ((1 << bits) - 1) - k
+        nbit = 0
+        if s.validity is not None:
+            nbit = 1
+            # null bit dominates the value: 0 sorts first
+            nullv = 0 if nf else 1
+            valuev = 1 - nullv
+            k = k | torch.where(s.validity,
+                                torch.full_like(k, valuev << bits),
+                                torch.full_like(k, nullv << bits))
+            # null rows: clear value bits so ordering among nulls is stable
+            k = torch.where(s.validity, k,
+                            torch.full_like(k, nullv << bits))
+        total_bits += bits + nbit
+        parts.append((k, bits + nbit))
+    if total_bits > 63:
+        return None
+    packed = torch.zeros_like(parts[0][0])
+    for k, bits in parts:                 # first key ends up in the MSBs
+        packed = (packed << bits) | k
+    return _stable_sort_perm_by(packed)
+
+
 def argsort_multi(keys: Sequence[Series], descending: Sequence[bool],
                   nulls_first: Sequence[bool]) -> torch.Tensor:
     """Stable lexicographic argsort over multiple key columns."""
     n = len(keys[0])
     dev = keys[0].device
+    composed = _try_composed_argsort(keys, descending, nulls_first)
+    if composed is not None:
+        return composed
     perm = torch.arange(n, dtype=torch.int64, device=dev)
     # LSD over keys: sort by last key first
     for s, desc, nf in list(zip(keys, descending, nulls_first))[::-1]:

@@ -164,3 +164,39 @@ def test_dense_join_property(lvals, rvals):
         else:
             assert sorted(zip(li1.tolist(), ri1.tolist())) == \
                 sorted(zip(li2.tolist(), ri2.tolist()))
+
+
+def test_composed_multikey_argsort_matches_fallback():
+    """Key-composition packing (single radix pass) must equal the per-key
+    LSD path on random multi-key data incl. nulls and descending."""
+    import random
+    import os
+    import torch
+    from daft_amd.kernels import rowops
+    from daft_amd.series import Series
+    from daft_amd.schema import DataType
+    random.seed(17)
+    n = 20_000
+    a = [random.randint(-5, 5) if i % 9 else None for i in range(n)]
+    b = [random.choice(["x", "y", "z", "w"]) for i in range(n)]
+    c = [random.randint(0, 10**6) for i in range(n)]
+    sa = Series.from_pylist("a", a, DataType.int32())
+    from daft_amd.physical.ops import _dict_encode
+    sb = _dict_encode(Series.from_pylist("b", b, DataType.string()))
+    assert sb.is_dict()
+    sc = Series.from_pylist("c", c, DataType.int64())
+    for desc, nf in (([False, False, False], [False, False, False]),
+                     ([True, False, True], [True, False, False]),
+                     ([False, True, False], [False, False, True])):
+        keys = [sa, sb, sc]
+        comp = rowops._try_composed_argsort(keys, desc, nf)
+        # fallback: force the per-key path
+        perm = None
+        orig = rowops._try_composed_argsort
+        rowops._try_composed_argsort = lambda *args: None
+        try:
+            perm = rowops.argsort_multi(keys, desc, nf)
+        finally:
+            rowops._try_composed_argsort = orig
+        assert comp is not None, "composition must engage here"
+        assert torch.equal(comp, perm), (desc, nf)
