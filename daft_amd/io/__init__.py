@@ -86,14 +86,70 @@ def _expand_paths(path: Union[str, List[str]],
     return out
 
 
+def _parse_hive_partitions(paths: List[str]):
+    """key=value path segments common to every file (ref: daft-scan
+    src/hive.rs).  Returns (per-path {col: str value}, ordered keys) or
+    (None, None) when the layout is not hive-partitioned."""
+    import urllib.parse
+    per = []
+    keysets = []
+    for p in paths:
+        comps = p.replace("\\", "/").split("/")[:-1]
+        d = {}
+        order = []
+        for c in comps:
+            if "=" in c and not c.startswith("="):
+                k, _, v = c.partition("=")
+                if k and k not in d:
+                    d[k] = urllib.parse.unquote(v)
+                    order.append(k)
+        per.append(d)
+        keysets.append(tuple(order))
+    if not per or not keysets[0] or any(ks != keysets[0]
+                                        for ks in keysets):
+        return None, None
+    return per, list(keysets[0])
+
+
+def _hive_typed(values: List[str]):
+    try:
+        return [int(v) for v in values], DataType.int64()
+    except ValueError:
+        pass
+    try:
+        return [float(v) for v in values], DataType.float64()
+    except ValueError:
+        return values, DataType.string()
+
+
 def read_parquet(path, columns: Optional[List[str]] = None,
-                 io_config=None, **kwargs) -> DataFrame:
+                 io_config=None, hive_partitioning: bool = True,
+                 **kwargs) -> DataFrame:
     from . import readers
     paths = _expand_paths(path, io_config)
     schema = readers.infer_schema(paths[0], "parquet",
                                   storage_options=io_config)
+    read_options = None
+    if hive_partitioning:
+        per, keys = _parse_hive_partitions(paths)
+        if keys:
+            keys = [k for k in keys if k not in set(schema.names())]
+        if keys:
+            from ..schema import Field
+            hive_fields = []
+            hive_parts = {}
+            for k in keys:
+                vals, dt = _hive_typed([d[k] for d in per])
+                hive_fields.append(Field(k, dt))
+                for pth, v in zip(paths, vals):
+                    hive_parts.setdefault(pth, {})[k] = v
+            schema = Schema(schema.fields() + hive_fields)                 if hasattr(schema, "fields") else schema
+            read_options = {"hive_parts": hive_parts,
+                            "hive_fields": [(f.name, f.dtype)
+                                            for f in hive_fields]}
     b = LogicalPlanBuilder.from_scan(schema, paths, "parquet",
-                                     storage_options=io_config)
+                                     storage_options=io_config,
+                                     read_options=read_options)
     df = DataFrame(b)
     if columns:
         df = df.select(*columns)

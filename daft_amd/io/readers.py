@@ -67,7 +67,38 @@ def read_file(path: str, file_format: str, columns: Optional[List[str]],
               predicate, limit: Optional[int], storage_options: dict,
               read_options: dict, device,
               row_groups=None) -> Iterator[RecordBatch]:
-    path = _open_input(path, storage_options)
+    # hive-partitioned scan: key=value path columns attach as constants
+    hive_vals = None
+    hive_fields = None
+    if read_options and read_options.get("hive_parts"):
+        hive_vals = read_options["hive_parts"].get(path)
+        hive_fields = read_options.get("hive_fields")
+        if hive_vals is not None and columns is not None:
+            hive_names = {n for n, _dt in hive_fields}
+            file_cols = [c for c in columns if c not in hive_names]
+            hive_fields = [(n, dt) for n, dt in hive_fields
+                           if n in set(columns)]
+            columns = file_cols or None
+    path_open = _open_input(path, storage_options)
+    if hive_vals is not None:
+        from ..series import Series
+        for rb in read_file_inner(path_open, file_format, columns,
+                                  predicate, limit, read_options, device,
+                                  row_groups):
+            cols = list(rb.columns)
+            for n, dt in (hive_fields or []):
+                cols.append(Series.from_pylist(
+                    n, [hive_vals[n]], dt,
+                    device=str(rb.device)).broadcast(len(rb)))
+            yield type(rb)(cols, num_rows=len(rb))
+        return
+    yield from read_file_inner(path_open, file_format, columns, predicate,
+                               limit, read_options, device, row_groups)
+
+
+def read_file_inner(path, file_format, columns, predicate, limit,
+                    read_options, device,
+                    row_groups=None) -> Iterator[RecordBatch]:
     if file_format == "parquet":
         yield from _read_parquet(path, columns, limit, device, predicate,
                                  row_groups=row_groups)

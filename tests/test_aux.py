@@ -929,3 +929,25 @@ def test_scan_task_split_and_merge(tmp_path):
     # reading through the engine still yields every row exactly once
     out = daft.read_parquet(paths).to_pydict()
     assert sorted(out["x"]) == sorted(list(range(100_000)) + [0, 1, 2])
+
+
+def test_hive_partitioned_read(tmp_path):
+    """read_parquet parses key=value directories into typed partition
+    columns (ref: daft-scan src/hive.rs), round-tripping the engine's own
+    hive-partitioned writer."""
+    import os
+    df = daft.from_pydict({"year": [2023, 2023, 2024, 2024],
+                           "region": ["us", "eu", "us", "eu"],
+                           "v": [1.0, 2.0, 3.0, 4.0]})
+    root = str(tmp_path / "t")
+    df.write_parquet(root, partition_cols=["year", "region"])
+    back = daft.read_parquet(root + "/**/*.parquet") \
+        .sort(["year", "region"]).to_pydict()
+    assert back["year"] == [2023, 2023, 2024, 2024]
+    assert back["region"] == ["eu", "us", "eu", "us"]
+    assert sorted(back["v"]) == [1.0, 2.0, 3.0, 4.0]
+    # filters on partition columns work
+    from daft_amd import col as _c
+    f = daft.read_parquet(root + "/**/*.parquet") \
+        .where((_c("year") == 2024) & (_c("region") == "us")).to_pydict()
+    assert f["v"] == [3.0]
