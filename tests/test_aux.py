@@ -951,3 +951,32 @@ def test_hive_partitioned_read(tmp_path):
     f = daft.read_parquet(root + "/**/*.parquet") \
         .where((_c("year") == 2024) & (_c("region") == "us")).to_pydict()
     assert f["v"] == [3.0]
+
+
+def test_otlp_file_span_exporter(tmp_path):
+    """Spans export as OTLP/JSON resourceSpans at query end (ref:
+    common/tracing OTLP wiring, flushed at run.rs:404)."""
+    import json
+    from daft_amd.context import get_context
+    from daft_amd.subscribers.otlp import OTLPFileSpanExporter
+    path = str(tmp_path / "spans.jsonl")
+    sub = OTLPFileSpanExporter(path)
+    ctx = get_context()
+    ctx.attach_subscriber(sub) if hasattr(ctx, "attach_subscriber") else \
+        ctx.subscribers.append(sub)
+    try:
+        df = daft.from_pydict({"x": [1, 2, 3]})
+        df.where(col("x") > 1).select((col("x") * 2).alias("y")).to_pydict()
+    finally:
+        ctx.subscribers.remove(sub)
+    lines = open(path).read().strip().splitlines()
+    assert lines
+    doc = json.loads(lines[-1])
+    spans = doc["resourceSpans"][0]["scopeSpans"][0]["spans"]
+    names = [s["name"] for s in spans]
+    assert "query" in names and "optimize" in names
+    assert any("Filter" in n for n in names)
+    root = next(s for s in spans if s["name"] == "query")
+    assert all(s.get("parentSpanId") == root["spanId"]
+               for s in spans if s is not root)
+    assert root["status"]["code"] == 1
