@@ -174,13 +174,16 @@ def run_udf_node(node: PyUDF, batch) -> Series:
                     return out.rename(node.name).cast(node.return_dtype) \
                         if out.dtype != node.return_dtype else \
                         out.rename(node.name)
+                import torch
+                if isinstance(out, torch.Tensor):
+                    # device tensors stay on device (torch check must come
+                    # before __array__: cuda tensors expose __array__ but
+                    # raise inside it)
+                    return Series.from_torch(node.name, out.to(batch.device))
                 if hasattr(out, "__array__") and not isinstance(out, list):
                     import numpy as np
                     return Series.from_numpy(node.name, np.asarray(out)) \
                         .to(batch.device)
-                import torch
-                if isinstance(out, torch.Tensor):
-                    return Series.from_torch(node.name, out.to(batch.device))
                 return Series.from_pylist(node.name, list(out),
                                           node.return_dtype,
                                           device=batch.device)

@@ -649,7 +649,7 @@ def test_out_of_core_hbm_stays_bounded():
     from benchmarks.tpch.queries import run_query
     from daft_amd.context import get_context
     get_context().execution_config.stream_morsel_rows = 1 << 24
-    T = datagen.dataframes(5.0, device="cpu")   # ~30M lineitem rows, host
+    T = datagen.dataframes(20.0, device="cpu")  # ~120M lineitem rows, host
     li_bytes = 0
     for p in T["lineitem"].collect()._result:
         li_bytes += p.size_bytes()
@@ -660,6 +660,7 @@ def test_out_of_core_hbm_stays_bounded():
     assert out["revenue"][0] > 0
     # morsel streaming must not materialize the table in HBM
     assert peak < max(li_bytes // 2, 2 << 30), (peak, li_bytes)
+    get_context().execution_config.stream_morsel_rows = 1 << 26
 
 
 @pytest.mark.gpu
