@@ -650,9 +650,14 @@ def test_out_of_core_hbm_stays_bounded():
     from daft_amd.context import get_context
     get_context().execution_config.stream_morsel_rows = 1 << 24
     T = datagen.dataframes(20.0, device="cpu")  # ~120M lineitem rows, host
-    li_bytes = 0
-    for p in T["lineitem"].collect()._result:
-        li_bytes += p.size_bytes()
+    # size the HOST partitions from the cache — collect() would run the
+    # identity query on the GPU and materialize the table in HBM
+    from daft_amd.logical import plan as lp
+    node = T["lineitem"]._builder.plan
+    while not isinstance(node, lp.Source):
+        node = node.children[0]
+    li_bytes = sum(p.size_bytes()
+                   for p in get_context().cache.get(node.cache_key))
     torch.cuda.empty_cache()
     torch.cuda.reset_peak_memory_stats()
     out = run_query(6, T, sf=5.0).to_pydict()
