@@ -24,7 +24,7 @@ from .context import (  # noqa: F401
 )
 from .io import (  # noqa: F401
     from_pydict, from_arrow, from_pandas, from_glob_path,
-    read_parquet, read_csv, read_json, read_text, read_warc,
+    read_parquet, read_csv, read_json, read_ipc, read_text, read_warc,
     read_sql, read_blob, from_glob_path, from_files, read_deltalake,
     read_iceberg, read_lance, read_hudi, read_kafka, read_mcap,
     read_paimon, read_huggingface, read_video_frames,

@@ -382,6 +382,11 @@ class DataFrame:
         b = self._builder.write("csv", root_dir, write_mode, partition_cols)
         return self._wrap(b).collect()
 
+    def write_ipc(self, root_dir: str, write_mode: str = "overwrite",
+                  partition_cols=None) -> "DataFrame":
+        b = self._builder.write("ipc", root_dir, write_mode, partition_cols)
+        return self._wrap(b).collect()
+
     def write_json(self, root_dir: str, write_mode: str = "overwrite",
                    partition_cols=None) -> "DataFrame":
         b = self._builder.write("json", root_dir, write_mode, partition_cols)

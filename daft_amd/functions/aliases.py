@@ -341,9 +341,8 @@ def last_value(x, ignore_nulls: bool = False):
 
 
 def jq(x, filter_expr: str):
-    raise RuntimeError("jq() requires the jq library, which is not "
-                       "available in this offline build; use "
-                       "Expression.json.query for simple paths")
+    from .misc import jq as _jq
+    return _jq(_e(x), filter_expr)
 
 
 def _uuid7_part(name, fn):

@@ -590,3 +590,122 @@ hdf5_attrs = _gated("hdf5_attrs", "h5py")
 hdf5_metadata = _gated("hdf5_metadata", "h5py")
 run_process = _gated("run_process", "subprocess execution policy")
 llm_generate = _gated("llm_generate", "a local LLM provider (vllm)")
+
+
+# ---------------------------------------------------------------------------
+# jq-style JSON queries (ref: daft-functions-json — jaq-backed `jq` filter
+# over JSON strings).  Supported subset: identity `.`, field access
+# `.a.b`, optional access `.a?`, array index `.a[0]` (negative ok),
+# array iteration `.a[]`, pipes `f | g`, and `//` defaults.
+# ---------------------------------------------------------------------------
+
+def _jq_compile(filter_expr: str):
+    import re as _re
+
+    def parse_one(src: str):
+        steps = []
+        i = 0
+        src = src.strip()
+        if src == ".":
+            return steps
+        while i < len(src):
+            c = src[i]
+            if c == ".":
+                m = _re.match(r"\.([A-Za-z_][A-Za-z0-9_]*)(\??)",
+                              src[i:])
+                if m:
+                    steps.append(("field", m.group(1),
+                                  m.group(2) == "?"))
+                    i += m.end()
+                    continue
+                if src[i:i + 1] == "." and src[i + 1:i + 2] == "[":
+                    i += 1
+                    continue
+                i += 1
+                continue
+            if c == "[":
+                m = _re.match(r"\[(-?\d*)\]", src[i:])
+                if not m:
+                    raise ValueError(f"jq: bad bracket at {src[i:]}")
+                if m.group(1) == "":
+                    steps.append(("iterate",))
+                else:
+                    steps.append(("index", int(m.group(1))))
+                i += m.end()
+                continue
+            raise ValueError(f"jq filter not supported: {src!r}")
+        return steps
+
+    alts = [a.strip() for a in filter_expr.split("//")]
+    progs = []
+    for alt in alts:
+        progs.append([parse_one(p) for p in alt.split("|")])
+    return progs
+
+def _jq_eval(steps_pipeline, doc):
+    vals = [doc]
+    for steps in steps_pipeline:
+        for step in steps:
+            out = []
+            for v in vals:
+                if step[0] == "field":
+                    if isinstance(v, dict):
+                        out.append(v.get(step[1]))
+                    elif not step[2]:
+                        raise TypeError("jq: cannot index non-object")
+                elif step[0] == "index":
+                    if isinstance(v, list):
+                        try:
+                            out.append(v[step[1]])
+                        except IndexError:
+                            out.append(None)
+                    else:
+                        out.append(None)
+                else:  # iterate
+                    if isinstance(v, list):
+                        out.extend(v)
+                    elif isinstance(v, dict):
+                        out.extend(v.values())
+            vals = out
+    return vals
+
+
+def jq(x, filter_expr: str):
+    """jq-style filter over a JSON string column; multi-value results
+    (array iteration) come back as a JSON array string (ref:
+    daft-functions-json/src — jaq filters)."""
+    progs = _jq_compile(filter_expr)
+
+    def run(s: Series) -> Series:
+        out = []
+        for v in s.cpu().to_pylist():
+            if v is None:
+                out.append(None)
+                continue
+            try:
+                doc = _json.loads(v)
+            except Exception:
+                out.append(None)
+                continue
+            res = None
+            for prog in progs:
+                try:
+                    vals = _jq_eval(prog, doc)
+                except TypeError:
+                    vals = []
+                vals = [x_ for x_ in vals if x_ is not None]
+                if vals:
+                    res = vals
+                    break
+            if not res:
+                out.append(None)
+            elif len(res) == 1:
+                r = res[0]
+                out.append(r if isinstance(r, str) else _json.dumps(r))
+            else:
+                out.append(_json.dumps(res))
+        r = Series.from_pylist(s.name, out, DataType.string())
+        return r.to(s.device) if s.is_gpu() else r
+
+    return Expression(ScalarFn("jq", run, [_to_node(x)],
+                               DataType.string()))

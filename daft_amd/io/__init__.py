@@ -182,6 +182,23 @@ def read_warc(path, io_config=None, file_path_column: Optional[str] = None,
     return df
 
 
+def read_ipc(path, columns: Optional[List[str]] = None,
+             io_config=None, **kwargs) -> DataFrame:
+    """Read Arrow IPC (Feather v2 / .arrow) files (ref: daft-writers
+    src/ipc.rs make_ipc_writer — the reference's shuffle/interchange
+    format)."""
+    from . import readers
+    paths = _expand_paths(path, io_config)
+    schema = readers.infer_schema(paths[0], "ipc",
+                                  storage_options=io_config)
+    b = LogicalPlanBuilder.from_scan(schema, paths, "ipc",
+                                     storage_options=io_config)
+    df = DataFrame(b)
+    if columns:
+        df = df.select(*columns)
+    return df
+
+
 def read_csv(path, has_headers: bool = True, delimiter: str = ",",
              schema: Optional[Dict[str, DataType]] = None,
              io_config=None, **kwargs) -> DataFrame:

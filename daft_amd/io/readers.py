@@ -40,6 +40,9 @@ def infer_schema(path: str, file_format: str,
     if file_format == "parquet":
         import pyarrow.parquet as pq
         a_schema = pq.read_schema(path)
+    elif file_format == "ipc":
+        with pa.ipc.open_file(path) as r:
+            a_schema = r.schema
     elif file_format == "csv":
         import pyarrow.csv as pacsv
         ro = read_options or {}
@@ -102,6 +105,14 @@ def read_file_inner(path, file_format, columns, predicate, limit,
     if file_format == "parquet":
         yield from _read_parquet(path, columns, limit, device, predicate,
                                  row_groups=row_groups)
+    elif file_format == "ipc":
+        import pyarrow as pa
+        with pa.ipc.open_file(path) as r:
+            tbl = r.read_all()
+        if columns:
+            tbl = tbl.select([c for c in columns
+                              if c in tbl.column_names])
+        yield RecordBatch.from_arrow(tbl, device=device)
     elif file_format == "csv":
         yield from _read_csv(path, columns, read_options, device)
     elif file_format == "json":

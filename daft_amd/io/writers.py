@@ -113,6 +113,11 @@ def _write_one(rb: RecordBatch, file_format: str, dir_: str,
         import pyarrow.csv as pacsv
         path = os.path.join(dir_, f"{name}.csv")
         pacsv.write_csv(tbl, path)
+    elif file_format == "ipc":
+        import pyarrow as pa
+        path = os.path.join(dir_, f"{name}.arrow")
+        with pa.ipc.new_file(path, tbl.schema) as w:
+            w.write_table(tbl)
     elif file_format in ("json", "jsonl"):
         path = os.path.join(dir_, f"{name}.jsonl")
         import json

@@ -980,3 +980,17 @@ def test_otlp_file_span_exporter(tmp_path):
     assert all(s.get("parentSpanId") == root["spanId"]
                for s in spans if s is not root)
     assert root["status"]["code"] == 1
+
+
+def test_ipc_write_read_roundtrip(tmp_path):
+    """Arrow IPC writer/reader (ref: daft-writers src/ipc.rs)."""
+    df = daft.from_pydict({"k": [1, 2, 3], "s": ["a", None, "c"],
+                           "f": [1.5, 2.5, 3.5]})
+    root = str(tmp_path / "ipc")
+    df.write_ipc(root)
+    import os
+    assert any(f.endswith(".arrow") for f in os.listdir(root))
+    back = daft.read_ipc(root + "/*.arrow").sort("k").to_pydict()
+    assert back["k"] == [1, 2, 3]
+    assert back["s"] == ["a", None, "c"]
+    assert back["f"] == [1.5, 2.5, 3.5]

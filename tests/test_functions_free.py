@@ -85,3 +85,26 @@ def test_free_function_sweep():
     ens = df.select(F.eq_null_safe(col("x"), col("x")).alias("e")).to_pydict()["e"]
     assert ens == [True, True, True]
     print("free function sweep OK")
+
+
+def test_jq_filters():
+    """jq-style JSON filters (ref: daft-functions-json jaq filters):
+    fields, indices, iteration, pipes, // defaults."""
+    import daft_amd as daft
+    from daft_amd import col
+    from daft_amd.functions import jq
+    df = daft.from_pydict({"j": [
+        '{"u": {"id": 7, "tags": ["a", "b"]}, "n": 1}',
+        '{"u": {"id": 8, "tags": []}}',
+        'not json',
+    ]})
+    out = df.select(
+        jq(col("j"), ".u.id").alias("id"),
+        jq(col("j"), ".u.tags[1]").alias("t1"),
+        jq(col("j"), ".u | .tags[]").alias("tags"),
+        jq(col("j"), ".n // .u.id").alias("d"),
+    ).to_pydict()
+    assert out["id"] == ["7", "8", None]
+    assert out["t1"] == ["b", None, None]
+    assert out["tags"] == ['["a", "b"]', None, None]
+    assert out["d"] == ["1", "8", None]
