@@ -79,17 +79,64 @@ def make_app(state: Optional[DashboardState] = None):
 
     @app.get("/", response_class=HTMLResponse)
     def index():
-        cells = []
-        for q in list(st.queries):
-            secs = "" if q["seconds"] is None else f"{q['seconds']:.3f}"
-            cells.append(f"<tr><td>{q['id']}</td><td>{q['status']}</td>"
-                         f"<td>{secs}</td></tr>")
-        return ("<html><body><h2>daft_amd queries</h2>"
-                "<table border=1><tr><th>id</th><th>status</th>"
-                f"<th>seconds</th></tr>{''.join(cells)}</table>"
-                "</body></html>")
+        return _UI_HTML
 
     return app
+
+
+# self-contained live UI (ref: the reference ships a web app in
+# daft-dashboard; this is the same query/operator browser as one page —
+# polls /api/queries, click-through to per-operator stats and the plan)
+_UI_HTML = """<!doctype html><html><head><title>daft_amd dashboard</title>
+<style>
+ body{font-family:ui-monospace,monospace;margin:1.5rem;background:#111;
+      color:#ddd}
+ h1{font-size:1.2rem} a{color:#7ab8ff;cursor:pointer}
+ table{border-collapse:collapse;margin-top:.6rem;width:100%}
+ th,td{border:1px solid #333;padding:.25rem .6rem;text-align:left;
+       font-size:.85rem}
+ th{background:#1c1c1c} tr:hover{background:#191919}
+ .done{color:#7dd87d}.failed{color:#ff7a7a}.running{color:#ffd37a}
+ #detail{margin-top:1rem;border-top:1px solid #333;padding-top:.8rem}
+ pre{background:#181818;padding:.6rem;overflow-x:auto;font-size:.78rem}
+ .bar{background:#2a6;display:inline-block;height:.6rem}
+</style></head><body>
+<h1>daft_amd — query dashboard</h1>
+<div id="list"></div><div id="detail"></div>
+<script>
+async function refresh(){
+  const qs = await (await fetch('/api/queries')).json();
+  let h = '<table><tr><th>query</th><th>status</th><th>seconds</th>'+
+          '<th>operators</th></tr>';
+  for (const q of qs){
+    h += `<tr><td><a onclick="show('${q.id}')">${q.id}</a></td>`+
+         `<td class="${q.status}">${q.status}</td>`+
+         `<td>${q.seconds==null?'':q.seconds.toFixed(3)}</td>`+
+         `<td>${(q.operators||[]).length}</td></tr>`;
+  }
+  document.getElementById('list').innerHTML = h + '</table>';
+}
+async function show(id){
+  const q = await (await fetch('/api/queries/'+id)).json();
+  const ops = q.operators || [];
+  const tmax = Math.max(1e-9, ...ops.map(o=>o.seconds));
+  let h = `<h1>${q.id} — ${q.status}`+
+          (q.seconds!=null?` (${q.seconds.toFixed(3)}s)`:``)+`</h1>`;
+  if (q.error) h += `<pre class="failed">${q.error}</pre>`;
+  h += '<table><tr><th>operator</th><th>rows out</th><th>seconds</th>'+
+       '<th></th></tr>';
+  for (const o of ops){
+    const w = Math.round(160*o.seconds/tmax);
+    h += `<tr><td>${o.name}</td><td>${o.rows_out}</td>`+
+         `<td>${o.seconds.toFixed(4)}</td>`+
+         `<td><span class="bar" style="width:${w}px"></span></td></tr>`;
+  }
+  h += '</table><h1>plan</h1><pre>'+
+       (q.plan||'').replace(/</g,'&lt;')+'</pre>';
+  document.getElementById('detail').innerHTML = h;
+}
+refresh(); setInterval(refresh, 2000);
+</script></body></html>"""
 
 
 def attach(state: Optional[DashboardState] = None) -> DashboardSubscriber:

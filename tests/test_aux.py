@@ -269,7 +269,7 @@ def test_dashboard_api():
     detail = client.get(f"/api/queries/{qs[0]['id']}").json()
     assert any(op["name"] == "Filter" for op in detail["operators"])
     html = client.get("/").text
-    assert "daft_amd queries" in html
+    assert "daft_amd" in html and "api/queries" in html
 
 
 def test_partition_cache_spill_bookkeeping():
@@ -994,3 +994,27 @@ def test_ipc_write_read_roundtrip(tmp_path):
     assert back["k"] == [1, 2, 3]
     assert back["s"] == ["a", None, "c"]
     assert back["f"] == [1.5, 2.5, 3.5]
+
+
+def test_dashboard_ui_and_api():
+    from fastapi.testclient import TestClient
+    from daft_amd.dashboard import DashboardState, DashboardSubscriber, \
+        make_app
+    st = DashboardState()
+    sub = DashboardSubscriber(st)
+    from daft_amd.context import get_context
+    ctx = get_context()
+    ctx.subscribers.append(sub)
+    try:
+        daft.from_pydict({"x": [1, 2, 3]}).where(col("x") > 1).to_pydict()
+    finally:
+        ctx.subscribers.remove(sub)
+    app = make_app(st)
+    c = TestClient(app)
+    qs = c.get("/api/queries").json()
+    assert qs and qs[0]["status"] == "done"
+    qid = qs[0]["id"]
+    rec = c.get(f"/api/queries/{qid}").json()
+    assert rec["operators"], "per-operator stats missing"
+    html = c.get("/").text
+    assert "daft_amd" in html and "/api/queries" in html
