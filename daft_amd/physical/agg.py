@@ -313,6 +313,55 @@ _MULTI_AGG_KINDS = {AggKind.SUM, AggKind.MIN, AggKind.MAX, AggKind.MEAN,
                     AggKind.COUNT, AggKind.COUNT_ALL}
 
 
+def _gids_sorted(gids: torch.Tensor) -> bool:
+    if gids.numel() < 2:
+        return True
+    return bool((gids[1:] >= gids[:-1]).all().item())
+
+
+def _segmented_multi_agg(gids, num_groups, datas, valids, ops, n, dev):
+    """Sorted-gids multi-aggregate via torch.segment_reduce.  gids are
+    dense AND sorted, so segment k == group k and boundaries come from
+    one vectorized searchsorted.  Output layout matches
+    grouped_multi_agg_big: (flat out[n_aggs*G] f64, cnt[n_aggs*G] i64)."""
+    G = num_groups
+    starts = torch.searchsorted(gids, torch.arange(
+        G, dtype=gids.dtype, device=dev))
+    bounds = torch.cat([starts, torch.tensor([n], dtype=starts.dtype,
+                                             device=dev)])
+    lengths = torch.diff(bounds)
+    outs = []
+    cnts = []
+    for d, v, op in zip(datas, valids, ops):
+        if v is None:
+            cnt = lengths.to(torch.int64)
+        else:
+            cnt = torch.segment_reduce(v.to(torch.float64), "sum",
+                                       lengths=lengths).to(torch.int64)
+        if op == 3:       # count only
+            outs.append(torch.zeros(G, dtype=torch.float64, device=dev))
+            cnts.append(cnt)
+            continue
+        if op == 0:       # sum
+            dd = d if v is None else torch.where(
+                v, d, torch.zeros_like(d))
+            red = "sum"
+        elif op == 1:     # min
+            dd = d if v is None else torch.where(
+                v, d, torch.full_like(d, float("inf")))
+            red = "min"
+        else:             # max
+            dd = d if v is None else torch.where(
+                v, d, torch.full_like(d, float("-inf")))
+            red = "max"
+        outs.append(torch.segment_reduce(dd, red, lengths=lengths,
+                                         initial=0 if red == "sum" else
+                                         (float("inf") if red == "min"
+                                          else float("-inf"))))
+        cnts.append(cnt)
+    return torch.cat(outs), torch.cat(cnts)
+
+
 def _multi_agg(batch, gids, num_groups, named_aggs, mask):
     """Fused one-pass aggregation (csrc grouped_multi_agg): every
     sum/min/max/mean/count computed from a single read of (gids, values)
@@ -380,8 +429,15 @@ def _multi_agg(batch, gids, num_groups, named_aggs, mask):
             ops.append({AggKind.SUM: 0, AggKind.MIN: 1, AggKind.MAX: 2,
                         AggKind.MEAN: 0}[a.kind])
         meta.append((cname, a, values))
-    fn = nat.grouped_multi_agg_big if big else nat.grouped_multi_agg
-    out, cnt = fn(gids, num_groups, datas, valids, ops)
+    if big and num_groups > 1 and _gids_sorted(gids):
+        # clustered keys (lineitem is orderkey-ordered: the q21/q18
+        # per-order aggregations): segmented reduction instead of 1.5B
+        # scattered global atomics
+        out, cnt = _segmented_multi_agg(gids, num_groups, datas, valids,
+                                        ops, n, dev)
+    else:
+        fn = nat.grouped_multi_agg_big if big else nat.grouped_multi_agg
+        out, cnt = fn(gids, num_groups, datas, valids, ops)
     out = out.view(len(named_aggs), num_groups)
     cnt = cnt.view(len(named_aggs), num_groups)
     cols = []
