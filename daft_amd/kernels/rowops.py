@@ -277,6 +277,49 @@ def _dense_range_groupby(keys):
     return gids, reps
 
 
+_SEG_MIN_GROUPS = 1 << 16   # atomics win below this; clustered-key wins above
+
+
+def _segmented_grouped_agg(gid, num_groups, d, validity, op, dev):
+    """Sorted dense gids (orderkey-clustered aggregations, q21/q18):
+    torch.segment_reduce over searchsorted boundaries instead of scattered
+    global atomics.  Returns (out, cnt) or None when not applicable."""
+    n = gid.numel()
+    if num_groups < _SEG_MIN_GROUPS or n < 2:
+        return None
+    if not bool((gid[1:] >= gid[:-1]).all().item()):
+        return None
+    starts = torch.searchsorted(gid, torch.arange(
+        num_groups, dtype=gid.dtype, device=dev))
+    bounds = torch.cat([starts, torch.tensor([n], dtype=starts.dtype,
+                                             device=dev)])
+    lengths = torch.diff(bounds)
+    if validity is None:
+        cnt = lengths.to(torch.int64)
+        dd = d
+    else:
+        cnt = torch.segment_reduce(validity.to(torch.float32), "sum",
+                                   lengths=lengths).to(torch.int64)
+        if op == "sum":
+            dd = torch.where(validity, d, torch.zeros_like(d))
+        else:
+            if d.dtype.is_floating_point:
+                fill = float("inf") if op == "min" else float("-inf")
+            else:
+                fill = (2**63 - 1) if op == "min" else -(2**63)
+            dd = torch.where(validity, d, torch.full_like(d, fill))
+    out = torch.segment_reduce(dd, op if op != "sum" else "sum",
+                               lengths=lengths,
+                               initial=0 if op == "sum" else
+                               ((2**63 - 1) if (op == "min" and
+                                                not d.dtype.is_floating_point)
+                                else (float("inf") if op == "min" else
+                                      (-(2**63) if not
+                                       d.dtype.is_floating_point else
+                                       float("-inf")))))
+    return out, cnt
+
+
 def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
                 op: str) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     """Aggregate values per group.  op in {sum,min,max,count,count_valid,sum_sq}.
@@ -316,6 +359,9 @@ def grouped_agg(group_ids: torch.Tensor, num_groups: int, values: Series,
         op = "sum"
 
     if _is_gpu(values) and op in ("sum", "min", "max"):
+        seg = _segmented_grouped_agg(gid, num_groups, d, validity, op, dev)
+        if seg is not None:
+            return seg
         vmask = validity if validity is not None else \
             torch.empty(0, dtype=torch.bool, device=dev)
         out, cnt = native_required().grouped_agg(gid, num_groups, d, vmask, op)

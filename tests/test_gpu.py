@@ -682,3 +682,36 @@ def test_batched_udf_subprocess_device_tensors():
     df = daft.from_pydict({"x": [1.0, 2.0, 3.0]}, device="cuda:0")
     out = df.select(dev_scale(col("x")).alias("y")).to_pydict()
     assert out["y"] == [3.0, 6.0, 9.0]
+
+
+@pytest.mark.gpu
+def test_segmented_grouped_agg_gpu_matches_hash():
+    """Sorted-gids segmented aggregation on device equals the hash-table
+    kernel path for int and float values with nulls."""
+    import torch as _t
+    from daft_amd.kernels import rowops
+    dev = "cuda:0"
+    n, G = 2_000_000, 500_000
+    g = _t.sort(_t.randint(0, G, (n,), device=dev)).values
+    g[:G] = _t.arange(G, device=dev)
+    g = _t.sort(g).values
+    vi = _t.randint(-10**6, 10**6, (n,), device=dev)
+    vf = _t.rand(n, dtype=_t.float64, device=dev)
+    val = _t.rand(n, device=dev) > 0.1
+    for data, dt in ((vi, DataType.int64()), (vf, DataType.float64())):
+        s = Series("v", dt, data=data, validity=val)
+        for op in ("sum", "min", "max"):
+            seg = rowops._segmented_grouped_agg(g, G, data.to(
+                _t.float64 if dt.is_floating() else _t.int64), val, op, dev)
+            assert seg is not None, "segmented path must engage"
+            out_s, cnt_s = seg
+            orig = rowops._segmented_grouped_agg
+            rowops._segmented_grouped_agg = lambda *a: None
+            try:
+                out_h, cnt_h = rowops.grouped_agg(g, G, s, op)
+            finally:
+                rowops._segmented_grouped_agg = orig
+            assert _t.allclose(out_s.to(_t.float64),
+                               out_h.to(_t.float64)), (dt, op)
+            if cnt_h is not None:
+                assert _t.equal(cnt_s, cnt_h), (dt, op)
