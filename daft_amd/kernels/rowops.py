@@ -289,6 +289,14 @@ def _segmented_grouped_agg(gid, num_groups, d, validity, op, dev):
         return None
     if not bool((gid[1:] >= gid[:-1]).all().item()):
         return None
+    int_in = not d.dtype.is_floating_point
+    if int_in:
+        # segment_reduce has no CUDA Long kernel: run through f64 when
+        # values stay exactly representable, else fall back to the
+        # hash-table kernel
+        if d.numel() and int(d.abs().max().item()) >= (1 << 53):
+            return None
+        d = d.to(torch.float64)
     starts = torch.searchsorted(gid, torch.arange(
         num_groups, dtype=gid.dtype, device=dev))
     bounds = torch.cat([starts, torch.tensor([n], dtype=starts.dtype,
@@ -303,20 +311,14 @@ def _segmented_grouped_agg(gid, num_groups, d, validity, op, dev):
         if op == "sum":
             dd = torch.where(validity, d, torch.zeros_like(d))
         else:
-            if d.dtype.is_floating_point:
-                fill = float("inf") if op == "min" else float("-inf")
-            else:
-                fill = (2**63 - 1) if op == "min" else -(2**63)
+            fill = float("inf") if op == "min" else float("-inf")
             dd = torch.where(validity, d, torch.full_like(d, fill))
-    out = torch.segment_reduce(dd, op if op != "sum" else "sum",
-                               lengths=lengths,
-                               initial=0 if op == "sum" else
-                               ((2**63 - 1) if (op == "min" and
-                                                not d.dtype.is_floating_point)
-                                else (float("inf") if op == "min" else
-                                      (-(2**63) if not
-                                       d.dtype.is_floating_point else
-                                       float("-inf")))))
+    out = torch.segment_reduce(dd, op, lengths=lengths,
+                               initial=0.0 if op == "sum" else
+                               (float("inf") if op == "min"
+                                else float("-inf")))
+    if int_in:
+        out = out.to(torch.int64)
     return out, cnt
 
 
