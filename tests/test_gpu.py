@@ -711,7 +711,8 @@ def test_segmented_grouped_agg_gpu_matches_hash():
                 out_h, cnt_h = rowops.grouped_agg(g, G, s, op)
             finally:
                 rowops._segmented_grouped_agg = orig
-            assert _t.allclose(out_s.to(_t.float64),
-                               out_h.to(_t.float64)), (dt, op)
             if cnt_h is not None:
                 assert _t.equal(cnt_s, cnt_h), (dt, op)
+            live = cnt_s > 0       # empty groups carry sentinel values
+            assert _t.allclose(out_s.to(_t.float64)[live],
+                               out_h.to(_t.float64)[live]), (dt, op)
