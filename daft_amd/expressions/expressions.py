@@ -1187,6 +1187,54 @@ class ListNamespace(_Namespace):
         return self._fn("list_slice", _list_slice,
                         lambda f: f[0].dtype, start, end)
 
+    def sort(self, desc: bool = False):
+        return self._fn("list_sort", _list_sort,
+                        lambda f: f[0].dtype, desc)
+
+    def flatten(self):
+        def ret(fields):
+            return fields[0].dtype.inner
+        return self._fn("list_flatten", _list_flatten, ret)
+
+    def reverse(self):
+        return self._fn("list_reverse", _list_reverse,
+                        lambda f: f[0].dtype)
+
+
+def _list_sort(s: Series, desc: bool) -> Series:
+    from ..schema import DataType as DT
+
+    def srt(v):
+        nn = sorted((x for x in v if x is not None), reverse=desc)
+        nulls = [None] * (len(v) - len(nn))
+        return nn + nulls
+    return _list_pylist_map(s, srt, DT.list(s.dtype.inner)
+                            if s.dtype.kind == TypeKind.LIST else s.dtype)
+
+
+def _list_flatten(s: Series) -> Series:
+    from ..schema import DataType as DT
+
+    def fl(v):
+        out = []
+        for x in v:
+            if isinstance(x, list):
+                out.extend(x)
+            elif x is not None:
+                out.append(x)
+        return out
+    inner = s.dtype.inner
+    if inner is not None and inner.kind == TypeKind.LIST:
+        inner = inner.inner
+    return _list_pylist_map(s, fl, DT.list(inner))
+
+
+def _list_reverse(s: Series) -> Series:
+    from ..schema import DataType as DT
+    return _list_pylist_map(s, lambda v: list(reversed(v)),
+                            DT.list(s.dtype.inner)
+                            if s.dtype.kind == TypeKind.LIST else s.dtype)
+
 
 def _list_length(s: Series) -> Series:
     if s.dtype.kind == TypeKind.LIST:

@@ -108,3 +108,18 @@ def test_jq_filters():
     assert out["t1"] == ["b", None, None]
     assert out["tags"] == ['["a", "b"]', None, None]
     assert out["d"] == ["1", "8", None]
+
+
+def test_list_sort_flatten_reverse():
+    import daft_amd as daft
+    from daft_amd import col
+    a = daft.from_pydict({"l": [[3, 1, None, 2], [5, 4], None]})
+    out = a.select(col("l").list.sort().alias("s"),
+                   col("l").list.sort(desc=True).alias("d"),
+                   col("l").list.reverse().alias("r")).to_pydict()
+    assert out["s"] == [[1, 2, 3, None], [4, 5], None]
+    assert out["d"] == [[3, 2, 1, None], [5, 4], None]
+    assert out["r"] == [[2, None, 1, 3], [4, 5], None]
+    b = daft.from_pydict({"l": [[[1, 2], [3]], [[4]], None]})
+    fo = b.select(col("l").list.flatten().alias("f")).to_pydict()
+    assert fo["f"] == [[1, 2, 3], [4], None]
