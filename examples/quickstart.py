@@ -6,6 +6,10 @@ columns.
 
     python examples/quickstart.py
 """
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(
+    _os.path.abspath(__file__))))
 import daft_amd as daft
 from daft_amd import col
 from daft_amd.functions import format as fmt, when
@@ -56,3 +60,20 @@ back = daft.read_parquet(os.path.join(d, "rides"))
 assert back.count_rows() == df.count_rows()
 print("parquet round-trip:", back.count_rows(), "rows")
 print("done.")
+
+# ---- round-2 additions ------------------------------------------------
+# object storage (S3-compatible; works against minio or AWS):
+#   cfg = daft.IOConfig(s3=daft.io.object_store.S3Config(
+#       endpoint_url="http://minio:9000", key_id=..., access_key=...))
+#   daft.read_parquet("s3://bucket/tbl/**/*.parquet", io_config=cfg)
+#   df.write_parquet("s3://bucket/out/")          # multipart upload
+# hive-partitioned reads:
+#   daft.read_parquet("data/year=*/region=*/*.parquet")
+# checkpoint/resume on an object store:
+#   store = daft.checkpoint.ObjectStoreCheckpointStore("s3://b/ckpt/", cfg)
+# out-of-core: host-resident tables stream through HBM automatically;
+# jq-style JSON filters:
+#   daft.functions.jq(col("j"), ".user.tags[]")
+# observability:
+#   daft_amd.dashboard.serve(8238)                # live query browser
+#   from daft_amd.subscribers.otlp import OTLPFileSpanExporter
