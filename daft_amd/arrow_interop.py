@@ -41,6 +41,10 @@ def dtype_to_arrow(dt: DataType):
         return pa.decimal128(dt.precision, dt.scale)
     if k == TypeKind.LIST:
         return pa.large_list(dtype_to_arrow(dt.inner))
+    if k == TypeKind.MAP:
+        entries = dt.inner.inner
+        return pa.map_(dtype_to_arrow(entries.fields[0].dtype),
+                       dtype_to_arrow(entries.fields[1].dtype))
     if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING):
         return pa.list_(dtype_to_arrow(dt.inner), dt.size)
     if k == TypeKind.FIXED_SHAPE_TENSOR:
@@ -83,6 +87,9 @@ def dtype_from_arrow(t) -> DataType:
         return DataType.duration(t.unit)
     if pa.types.is_time(t):
         return DataType.time("us")
+    if pa.types.is_map(t):
+        return DataType.map(dtype_from_arrow(t.key_type),
+                            dtype_from_arrow(t.item_type))
     if pa.types.is_list(t) or pa.types.is_large_list(t):
         return DataType.list(dtype_from_arrow(t.value_type))
     if pa.types.is_fixed_size_list(t):
@@ -144,6 +151,14 @@ def from_arrow_array(name: str, arr) -> Series:
         return Series(name, dt, data=torch.from_numpy(data.copy()),
                       offsets=torch.from_numpy(np.ascontiguousarray(off).copy()),
                       validity=validity)
+    if k == TypeKind.MAP:
+        off = np.asarray(arr.offsets).astype(np.int64)
+        keys = from_arrow_array("key", arr.keys)
+        items = from_arrow_array("value", arr.items)
+        entries = Series("entries", dt.inner.inner,
+                         children=[keys, items], length=len(keys))
+        return Series(name, dt, offsets=torch.from_numpy(off),
+                      children=[entries], validity=validity)
     if k == TypeKind.LIST:
         if not pa.types.is_large_list(arr.type):
             arr = arr.cast(pa.large_list(arr.type.value_type))
@@ -229,6 +244,20 @@ def to_arrow_array(s: Series):
         if mask is not None:
             vbuf = pa.array(~mask).buffers()[1]
         return pa.Array.from_buffers(atype, len(s), [vbuf, off, data])
+    if k == TypeKind.MAP:
+        entries = s.children[0]
+        keys = to_arrow_array(entries.children[0])
+        items = to_arrow_array(entries.children[1])
+        off32 = pa.array(s.offsets.numpy().astype(np.int32))
+        m = pa.MapArray.from_arrays(off32, keys, items)
+        if mask is not None:
+            import pyarrow.compute as _pc
+            # rebuild with validity via take-on-null trick
+            idx = pa.array([None if bad else i
+                            for i, bad in enumerate(mask)],
+                           type=pa.int64())
+            m = m.take(idx)
+        return m
     if k == TypeKind.LIST:
         child = to_arrow_array(s.children[0])
         off = pa.py_buffer(s.offsets.numpy().tobytes())

@@ -154,24 +154,78 @@ hamming_distance = _emb_pair(
     DataType.int64())
 
 
-# -- map helpers (struct-backed maps) ----------------------------------------
+# -- map helpers (first-class Map + struct-backed fallback) ------------------
+def _map_value_dtype(s: Series):
+    return s.dtype.inner.inner.fields[1].dtype
+
+
 def map_get(x, key):
-    """Struct-backed map access: get field `key` (our map representation
-    is a struct; ref daft Map type)."""
-    return _e(x).struct.get(key)
+    """Map access: value for `key` per row (null when absent).  Works on
+    first-class Map columns (list<struct<key,value>> physical, ref
+    daft-schema Map) and struct-backed maps."""
+    def ret(fields):
+        dt = fields[0].dtype
+        if dt.kind == TypeKind.MAP:
+            return dt.inner.inner.fields[1].dtype
+        if dt.kind == TypeKind.STRUCT:
+            for sub in dt.fields:
+                if sub.name == key:
+                    return sub.dtype
+        return DataType.string()
+
+    def runner(s: Series) -> Series:
+        if s.dtype.kind == TypeKind.MAP:
+            vals = s.cpu().to_pylist()
+            out = [None if v is None else v.get(key) for v in vals]
+            r = Series.from_pylist(s.name, out, _map_value_dtype(s))
+            return r.to(s.device) if s.is_gpu() else r
+        if s.dtype.kind == TypeKind.STRUCT:
+            for i, sub in enumerate(s.dtype.fields):
+                if sub.name == key:
+                    c = s.children[i].rename(s.name)
+                    if s.validity is not None:
+                        v = c.validity & s.validity                             if c.validity is not None else s.validity
+                        c = c.with_validity(v)
+                    return c
+            raise KeyError(key)
+        raise TypeError("map_get expects a Map or struct column")
+    return Expression(ScalarFn("map_get", runner, [_to_node(_e(x))], ret))
 
 
 def map_keys(x):
     def run(s: Series) -> Series:
+        if s.dtype.kind == TypeKind.MAP:
+            vals = s.cpu().to_pylist()
+            out = [None if v is None else list(v.keys()) for v in vals]
+            kd = s.dtype.inner.inner.fields[0].dtype
+            r = Series.from_pylist(s.name, out, DataType.list(kd))
+            return r.to(s.device) if s.is_gpu() else r
         if s.dtype.kind == TypeKind.STRUCT:
             keys = [f.name for f in s.dtype.fields]
             out = [keys] * len(s)
             r = Series.from_pylist(s.name, out,
                                    DataType.list(DataType.string()))
             return r.to(s.device) if s.is_gpu() else r
-        raise TypeError("map_keys expects a struct-backed map")
+        raise TypeError("map_keys expects a Map or struct column")
     return Expression(ScalarFn("map_keys", run, [_to_node(x)],
                                DataType.list(DataType.string())))
+
+
+def map_values(x):
+    def run(s: Series) -> Series:
+        if s.dtype.kind != TypeKind.MAP:
+            raise TypeError("map_values expects a Map column")
+        vals = s.cpu().to_pylist()
+        out = [None if v is None else list(v.values()) for v in vals]
+        vd = s.dtype.inner.inner.fields[1].dtype
+        r = Series.from_pylist(s.name, out, DataType.list(vd))
+        return r.to(s.device) if s.is_gpu() else r
+    def ret(fields):
+        dt = fields[0].dtype
+        if dt.kind == TypeKind.MAP:
+            return DataType.list(dt.inner.inner.fields[1].dtype)
+        return DataType.list(DataType.string())
+    return Expression(ScalarFn("map_values", run, [_to_node(x)], ret))
 
 
 # -- timezone conversions ----------------------------------------------------

@@ -141,7 +141,7 @@ def take(s: Series, indices: torch.Tensor,
             new_off, new_bytes = _cpu_take_string(s.offsets, s.data, safe_idx)
         return Series(s.name, s.dtype, data=new_bytes, offsets=new_off,
                       validity=validity)
-    if k == TypeKind.LIST:
+    if k in (TypeKind.LIST, TypeKind.MAP):
         # gather child ranges
         lens = s.offsets[1:] - s.offsets[:-1]
         sel_lens = lens[safe_idx]
@@ -259,7 +259,8 @@ def concat(series: List[Series]) -> Series:
             pos += n
         return Series(s0.name, dtype, data=data, offsets=new_off,
                       validity=validity)
-    if k in (TypeKind.STRING, TypeKind.BINARY, TypeKind.LIST):
+    if k in (TypeKind.STRING, TypeKind.BINARY, TypeKind.LIST,
+             TypeKind.MAP):
         new_bytes_parts = []
         new_off = torch.zeros(total + 1, dtype=torch.int64, device=dev)
         pos = 0
@@ -269,7 +270,7 @@ def concat(series: List[Series]) -> Series:
             new_off[pos + 1: pos + n + 1] = s.offsets[1:] + base
             base += int(s.offsets[-1].item())
             pos += n
-        if k == TypeKind.LIST:
+        if k in (TypeKind.LIST, TypeKind.MAP):
             child = concat([s.children[0] for s in series])
             return Series(s0.name, dtype, offsets=new_off, children=[child],
                           validity=validity)

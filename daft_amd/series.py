@@ -290,6 +290,12 @@ class Series:
             child = s.children[0].to_pylist()
             off = s.offsets.numpy()
             return wrap([child[off[i]:off[i + 1]] for i in range(len(s))])
+        if k == TypeKind.MAP:
+            child = s.children[0].to_pylist()
+            off = s.offsets.numpy()
+            return wrap([{e["key"]: e["value"]
+                          for e in child[off[i]:off[i + 1]]}
+                         for i in range(len(s))])
         if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING):
             child = s.children[0].to_pylist()
             n = self.dtype.size
@@ -398,7 +404,7 @@ class Series:
             return Series(self.name, self.dtype,
                           data=self.data[lo:hi], offsets=offs - lo,
                           validity=validity)
-        if k == TypeKind.LIST:
+        if k in (TypeKind.LIST, TypeKind.MAP):
             offs = self.offsets[start:end + 1]
             lo = int(offs[0])
             hi = int(offs[-1])
@@ -630,6 +636,24 @@ def _from_pylist_typed(name: str, values: list, dtype: DataType) -> Series:
         child = _from_pylist_typed("item", flat, dtype.inner)
         return Series(name, dtype, offsets=torch.from_numpy(offs),
                       children=[child], validity=validity)
+    if k == TypeKind.MAP:
+        # logical Map over list<struct<key, value>> physical storage
+        # (ref: daft-schema dtype.rs Map -> List(Struct) to_physical)
+        entries = []
+        for v in values:
+            if v is None:
+                entries.append(None)
+            elif isinstance(v, dict):
+                entries.append([{"key": kk, "value": vv}
+                                for kk, vv in v.items()])
+            else:               # already [(k, v)] pairs or entry structs
+                entries.append([e if isinstance(e, dict) and
+                                "key" in e else
+                                {"key": e[0], "value": e[1]}
+                                for e in v])
+        phys = _from_pylist_typed(name, entries, dtype.inner)
+        return Series(name, dtype, offsets=phys.offsets,
+                      children=phys.children, validity=validity)
     if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING):
         sz = dtype.size
         flat = []
@@ -715,6 +739,12 @@ def full_null(name: str, dtype: DataType, length: int, device="cpu") -> Series:
                       validity=validity)
     if k == TypeKind.LIST:
         child = empty_series("item", dtype.inner, device)
+        return Series(name, dtype, offsets=torch.zeros(
+            length + 1, dtype=torch.int64, device=device),
+            children=[child], validity=validity)
+    if k == TypeKind.MAP:
+        # child is the entries struct (Map stores list<struct> physically)
+        child = empty_series("entries", dtype.inner.inner, device)
         return Series(name, dtype, offsets=torch.zeros(
             length + 1, dtype=torch.int64, device=device),
             children=[child], validity=validity)

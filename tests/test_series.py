@@ -236,3 +236,37 @@ def test_is_in_sorted_table_path():
     mf = Series.from_pylist("v", [float(x) for x in members],
                             DataType.float64())
     assert sf.is_in(mf).to_pylist() == want
+
+
+def test_map_first_class():
+    """First-class Map type: construction from dicts, structural ops,
+    map_get/keys/values, and arrow round trip (ref: daft-schema Map ->
+    List(Struct) to_physical; daft-sql map module)."""
+    import torch
+    import daft_amd as daft
+    from daft_amd import col
+    from daft_amd import arrow_interop as ai
+    dt = DataType.map(DataType.string(), DataType.int64())
+    s = Series.from_pylist("m", [{"a": 1}, {"b": 2, "c": 3}, None,
+                                 {"d": 4}], dt)
+    assert s.to_pylist() == [{"a": 1}, {"b": 2, "c": 3}, None, {"d": 4}]
+    assert s.take(torch.tensor([3, 1])).to_pylist() == \
+        [{"d": 4}, {"b": 2, "c": 3}]
+    assert s.slice(1, 3).to_pylist() == [{"b": 2, "c": 3}, None]
+    assert Series.concat([s, s]).to_pylist()[5] == {"b": 2, "c": 3}
+    # arrow round trip through pa.map_
+    arr = ai.to_arrow_array(s)
+    import pyarrow as pa
+    assert pa.types.is_map(arr.type)
+    back = ai.from_arrow_array("m", arr)
+    assert back.to_pylist() == s.to_pylist()
+    # functions
+    from daft_amd.functions import map_get, map_keys
+    from daft_amd.functions.aliases import map_values
+    df = daft.from_pydict({"m": s})
+    out = df.select(map_get(col("m"), "b").alias("b"),
+                    map_keys(col("m")).alias("k"),
+                    map_values(col("m")).alias("v")).to_pydict()
+    assert out["b"] == [None, 2, None, None]
+    assert out["k"] == [["a"], ["b", "c"], None, ["d"]]
+    assert out["v"] == [[1], [2, 3], None, [4]]
