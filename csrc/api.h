@@ -97,6 +97,10 @@ std::vector<Tensor> fused_eval_jit(const std::string& src,
                                    std::vector<int64_t> out_need_valid,
                                    int64_t n);
 
+// BPE tokenize (bpe.hip)
+std::vector<Tensor> bpe_encode(Tensor offsets, Tensor bytes, Tensor byte2id,
+                               Tensor table_keys, Tensor table_vals);
+
 // strings (strings.hip)
 Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
 Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,

@@ -23,6 +23,7 @@ setup(
                 "csrc/multimodal.hip",
                 "csrc/fusedexpr.hip",
                 "csrc/fusedjit.hip",
+                "csrc/bpe.hip",
             ],
             libraries=["hiprtc"],
             extra_compile_args={

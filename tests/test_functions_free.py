@@ -123,3 +123,27 @@ def test_list_sort_flatten_reverse():
     b = daft.from_pydict({"l": [[[1, 2], [3]], [[4]], None]})
     fo = b.select(col("l").list.flatten().alias("f")).to_pydict()
     assert fo["f"] == [[1, 2, 3], [4], None]
+
+
+def test_bpe_tokenizer_json_roundtrip(tmp_path):
+    """tokenizers-JSON BPE loading + encode/decode through the expression
+    API (CPU oracle; the GPU kernel is compared in test_gpu.py)."""
+    import json
+    from daft_amd.functions.tokenize import _bytes_to_unicode
+    b2u = _bytes_to_unicode()
+    vocab = {b2u[b]: b for b in range(256)}
+    h, e = b2u[ord("h")], b2u[ord("e")]
+    vocab[h + e] = 256
+    doc = {"model": {"type": "BPE", "vocab": vocab,
+                     "merges": [f"{h} {e}"]}}
+    p = tmp_path / "tok.json"
+    p.write_text(json.dumps(doc))
+    import daft_amd as daft
+    from daft_amd import col
+    df = daft.from_pydict({"t": ["he", "heh", None]})
+    enc = df.select(col("t").str.tokenize_encode(f"bpe:{p}").alias("ids"))
+    ids = enc.to_pydict()["ids"]
+    assert ids == [[256], [256, vocab[h]], None]
+    dec = enc.select(daft.functions.tokenize_decode(
+        col("ids"), f"bpe:{p}").alias("t")).to_pydict()
+    assert dec["t"] == ["he", "heh", None]

@@ -716,3 +716,41 @@ def test_segmented_grouped_agg_gpu_matches_hash():
             live = cnt_s > 0       # empty groups carry sentinel values
             assert _t.allclose(out_s.to(_t.float64)[live],
                                out_h.to(_t.float64)[live]), (dt, op)
+
+
+@pytest.mark.gpu
+def test_bpe_encode_gpu_matches_oracle():
+    """Wave-per-row BPE kernel vs the python oracle over random strings
+    (synthetic merge table; no tokenizer downloads offline)."""
+    import random
+    from daft_amd.functions.tokenize import (BPETokenizer, bpe_encode_series,
+                                             _bytes_to_unicode)
+    random.seed(23)
+    b2u = _bytes_to_unicode()
+    vocab = {b2u[b]: b for b in range(256)}
+    merges = []
+    nxt = 256
+    # random 2-symbol merges over a small alphabet, chained
+    syms = [b2u[ord(c)] for c in "abcdefgh "]
+    pool = list(syms)
+    for _ in range(60):
+        l, r = random.choice(pool), random.choice(pool)
+        if l + r in vocab:
+            continue
+        vocab[l + r] = nxt
+        merges.append((l, r))
+        pool.append(l + r)
+        nxt += 1
+    tok = BPETokenizer(vocab, merges)
+    texts = ["".join(random.choice("abcdefgh ") for _ in
+                     range(random.randint(0, 300))) for _ in range(500)]
+    texts += ["a" * 5000]           # over the LDS cap: host fallback row
+    texts += [None, ""]
+    s = Series.from_pylist("t", texts, DataType.string()).to("cuda:0")
+    out = bpe_encode_series(s, tok).cpu().to_pylist()
+    for i, t in enumerate(texts):
+        if t is None:
+            assert out[i] is None
+            continue
+        want = tok.encode_py(t.encode("utf-8"))
+        assert out[i] == want, (i, t[:40], out[i][:10], want[:10])
