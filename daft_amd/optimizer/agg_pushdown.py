@@ -177,3 +177,79 @@ def push_down_aggregation(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
             na = Alias(na, want)    # keep the output schema identical
         new_aggs.append(na)
     return lp.Aggregate(new_join, plan.groupby, new_aggs)
+
+
+def push_semi_into_agg(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
+    """When an Aggregate's output inner-joins on its groupby key against a
+    much smaller relation, semi-filter the aggregate's INPUT by that
+    relation first: q17 aggregates per-partkey means over all of lineitem
+    and then joins 0.1% of parts — the filter belongs below the groupby.
+    (ref: the reference's KeyFilteringJoinNode + push_down_aggregation
+    interplay, key_filtering_join.rs:38)"""
+    if not (isinstance(plan, lp.Join) and plan.how == "inner"):
+        return None
+    for side in (0, 1):
+        node = plan.children[side]
+        other = plan.children[1 - side]
+        keys = list(plan.left_on if side == 0 else plan.right_on)
+        other_keys = list(plan.right_on if side == 0 else plan.left_on)
+        o_est = other.approx_num_rows()
+        n_est = node.approx_num_rows()
+        if o_est is None or n_est is None or o_est > 0.3 * n_est:
+            continue
+        # resolve key names through Project renames down to an Aggregate
+        key_names = []
+        ok = True
+        for e in keys:
+            base = e.child if isinstance(e, Alias) else e
+            if not isinstance(base, ColumnRef):
+                ok = False
+                break
+            key_names.append(base.name)
+        if not ok:
+            continue
+        cur = node
+        while isinstance(cur, lp.Project):
+            mapping = {}
+            for pe in cur.exprs:
+                out = pe.to_field(cur.children[0].schema).name
+                pb = pe.child if isinstance(pe, Alias) else pe
+                if isinstance(pb, ColumnRef):
+                    mapping[out] = pb.name
+            if not all(k in mapping for k in key_names):
+                ok = False
+                break
+            key_names = [mapping[k] for k in key_names]
+            cur = cur.children[0]
+        if not ok or not isinstance(cur, lp.Aggregate) or not cur.groupby:
+            continue
+        gb_src = {}
+        for g in cur.groupby:
+            out = g.to_field(cur.children[0].schema).name
+            gb = g.child if isinstance(g, Alias) else g
+            if isinstance(gb, ColumnRef):
+                gb_src[out] = gb.name
+        if not all(k in gb_src for k in key_names):
+            continue
+        src_names = [gb_src[k] for k in key_names]
+        inner = cur.children[0]
+        # already filtered by this relation: stop the fixed-point loop
+        if isinstance(inner, lp.Join) and inner.how == "semi" and \
+                inner.children[1].semantic_id() == other.semantic_id():
+            continue
+        semi = lp.Join(inner, other,
+                       [ColumnRef(s) for s in src_names], other_keys,
+                       "semi")
+        new_agg = lp.Aggregate(semi, cur.groupby, cur.aggs)
+        # rebuild the Project chain above the aggregate
+        rebuilt = new_agg
+        chain = []
+        c = node
+        while isinstance(c, lp.Project):
+            chain.append(c)
+            c = c.children[0]
+        for pr in reversed(chain):
+            rebuilt = lp.Project(rebuilt, pr.exprs)
+        new_children = [rebuilt, other] if side == 0 else [other, rebuilt]
+        return plan.with_children(new_children)
+    return None

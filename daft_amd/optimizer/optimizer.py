@@ -19,6 +19,7 @@ Rule = Callable[[lp.LogicalPlan], Optional[lp.LogicalPlan]]
 def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
     import os
     from .agg_pushdown import push_down_aggregation
+    from .agg_pushdown import push_semi_into_agg as _push_semi_into_agg
     from .join_reorder import reorder_joins
     disabled = set((os.environ.get("DAFT_AMD_DISABLE_RULES") or "")
                    .split(","))
@@ -29,7 +30,8 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
         [simplify_expressions],
         [push_down_filter, drop_repartition] +
         ([] if "semipush" in disabled else [push_down_anti_semi_join]),
-        [] if "aggpush" in disabled else [push_down_aggregation],
+        [] if "aggpush" in disabled else [push_down_aggregation,
+                                          _push_semi_into_agg],
         [push_down_projection],
         [push_down_limit, rewrite_topn],
         [simplify_expressions],
