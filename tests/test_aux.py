@@ -1058,3 +1058,38 @@ def test_segmented_multi_agg_matches_scatter():
     ref_cv = torch.zeros(G, dtype=torch.int64).scatter_add_(
         0, gids, v1.to(torch.int64))
     assert torch.equal(cnt[0], ref_cv)
+
+
+def test_round2_edge_cases():
+    """Empty inputs through the round-2 surfaces: new optimizer rules,
+    external sort, jq, Map, is_in table path, scan tasks, framed windows."""
+    from daft_amd.schema import DataType as DT
+    from daft_amd.series import Series as S
+    e = daft.from_pydict({"k": [], "v": []})
+    big = daft.from_pydict({"k2": [1, 2], "w": [1.0, 2.0]})
+    out = e.join(big, left_on="k", right_on="k2", how="left") \
+        .groupby("v").agg(col("w").sum().alias("s")).to_pydict()
+    assert out["s"] == []
+    from daft_amd.context import get_context
+    cfg = get_context().execution_config
+    old = cfg.memory_limit_bytes
+    cfg.memory_limit_bytes = 1024
+    try:
+        assert e.sort("k").to_pydict()["k"] == []
+    finally:
+        cfg.memory_limit_bytes = old
+    from daft_amd.functions import jq
+    assert daft.from_pydict({"j": []}) \
+        .select(jq(col("j"), ".a")).to_pydict() is not None
+    dt = DT.map(DT.string(), DT.int64())
+    assert S.from_pylist("m", [], dt).to_pylist() == []
+    s2 = S.from_pylist("x", [1, 2, 3], DT.int64())
+    assert s2.is_in(S.from_pylist("v", [], DT.int64())).to_pylist() == \
+        [False] * 3
+    from daft_amd.io.readers import plan_scan_tasks
+    assert plan_scan_tasks([], "parquet", None) == []
+    one = daft.from_pydict({"t": [1], "v": [2.0]})
+    from daft_amd.window import Window
+    w = Window().order_by("t")
+    assert one.with_window_columns(
+        {"m": col("v").min().over(w)}).to_pydict()["m"] == [2.0]
