@@ -297,3 +297,32 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     if tree.schema.names() != want:
         return lp.Project(tree, [CR(n) for n in want])
     return tree
+
+
+def swap_join_builds(plan: lp.LogicalPlan) -> lp.LogicalPlan:
+    """Single pass: any inner equi-join whose LEFT side is estimated much
+    smaller than its right gets its sides swapped (the physical hash join
+    builds on the right child) — q9-SQL had the right join ORDER but
+    built a 600M-row lineitem table because the connectivity planner put
+    the filtered part table on the left.  Wrapped in a schema-restoring
+    Project; only fires when names are globally unique (no rename/merge
+    semantics change).  The 0.5x hysteresis keeps it idempotent."""
+    new_children = [swap_join_builds(c) for c in plan.children]
+    if any(n is not o for n, o in zip(new_children, plan.children)):
+        plan = plan.with_children(new_children)
+    if not (isinstance(plan, lp.Join) and plan.how == "inner" and
+            plan.left_on and plan.prefix is None):
+        return plan
+    l, r = plan.children
+    le, re_ = l.approx_num_rows(), r.approx_num_rows()
+    if le is None or re_ is None or le >= 0.5 * re_:
+        return plan
+    names = l.schema.names() + r.schema.names()
+    if len(set(names)) != len(names):
+        return plan
+    swapped = lp.Join(r, l, plan.right_on, plan.left_on, "inner",
+                      plan.suffix, plan.prefix)
+    want = plan.schema.names()
+    if swapped.schema.names() != want:
+        return lp.Project(swapped, [ColumnRef(n) for n in want])
+    return swapped

@@ -60,6 +60,9 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
                 f"(8 rounds) while still rewriting", RuntimeWarning)
         if bi == reorder_after:
             plan = reorder_joins(plan)
+            if "reorder" not in disabled:
+                from .join_reorder import swap_join_builds
+                plan = swap_join_builds(plan)
     return plan
 
 
