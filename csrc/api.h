@@ -101,6 +101,10 @@ std::vector<Tensor> fused_eval_jit(const std::string& src,
 std::vector<Tensor> bpe_encode(Tensor offsets, Tensor bytes, Tensor byte2id,
                                Tensor table_keys, Tensor table_vals);
 
+// edit distance (editdist.hip)
+Tensor levenshtein(Tensor a_off, Tensor a_bytes, Tensor b_off,
+                   Tensor b_bytes, int64_t max_len);
+
 // strings (strings.hip)
 Tensor str_find(Tensor offsets, Tensor bytes, Tensor pattern, int64_t mode);
 Tensor str_like(Tensor offsets, Tensor bytes, Tensor needles, Tensor lens,

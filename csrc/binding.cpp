@@ -35,6 +35,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("str_concat", &str_concat, "row-wise string concat");
   m.def("str_char_length", &str_char_length, "UTF-8 character length");
   m.def("string_compare", &string_compare, "lexicographic compare");
+  m.def("levenshtein", &levenshtein,
+        "ASCII Levenshtein distance, one thread per row pair");
   m.def("bpe_encode", &bpe_encode,
         "byte-level BPE encode (one wavefront per row, LDS sequence)");
   m.def("fused_jit_available", &fused_jit_available);

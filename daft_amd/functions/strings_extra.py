@@ -262,8 +262,57 @@ def _lev(a: str, b: str) -> int:
     return prev[-1]
 
 
-levenshtein_distance = _host_map2("levenshtein_distance", _lev,
-                                  DataType.int64())
+def _lev_series(x: Series, y: Series) -> Series:
+    """GPU path: thread-per-pair HIP DP kernel over ASCII rows; rows the
+    kernel flags (-1: non-ASCII or longer than 512 bytes) recompute on
+    the host character-level oracle."""
+    import torch as _t
+    from ..kernels import load_native
+    nat = load_native()
+    gpu_ok = (x.is_gpu() and y.is_gpu() and nat is not None and
+              len(x) == len(y) and not x.is_dict() and not y.is_dict()
+              and x.offsets is not None and y.offsets is not None)
+    if gpu_ok:
+        MAXL, CHUNK = 512, 1 << 20
+        outs = []
+        n = len(x)
+        for lo in range(0, n, CHUNK):
+            xs = x.slice(lo, lo + CHUNK)
+            ys = y.slice(lo, lo + CHUNK)
+            outs.append(nat.levenshtein(xs.offsets, xs.data, ys.offsets,
+                                        ys.data, MAXL))
+        d = _t.cat(outs).to(_t.int64)
+        bad = (d < 0).nonzero().reshape(-1)
+        validity = None
+        if x.validity is not None or y.validity is not None:
+            xv = x.validity if x.validity is not None else                 _t.ones(n, dtype=_t.bool, device=x.device)
+            yv = y.validity if y.validity is not None else                 _t.ones(n, dtype=_t.bool, device=y.device)
+            validity = xv & yv
+        res = Series(x.name, DataType.int64(), data=d, validity=validity)
+        if int(bad.numel()):
+            xs = x.take(bad).cpu().to_pylist()
+            ys = y.take(bad).cpu().to_pylist()
+            fix = _t.tensor([0 if (a is None or b is None)
+                             else _lev(a, b) for a, b in zip(xs, ys)],
+                            dtype=_t.int64, device=x.device)
+            d = d.clone()
+            d[bad] = fix
+            res = Series(x.name, DataType.int64(), data=d,
+                         validity=validity)
+        return res
+    xv = x.cpu().to_pylist()
+    yv = y.cpu().to_pylist()
+    n = max(len(xv), len(yv))
+    out = [None if xv[i % len(xv)] is None or yv[i % len(yv)] is None
+           else _lev(xv[i % len(xv)], yv[i % len(yv)]) for i in range(n)]
+    r = Series.from_pylist(x.name, out, DataType.int64())
+    return r.to(x.device) if x.is_gpu() else r
+
+
+def levenshtein_distance(a, b):
+    return Expression(ScalarFn("levenshtein_distance", _lev_series,
+                               [_to_node(a), _to_node(b)],
+                               DataType.int64()))
 
 
 def _dlev(a: str, b: str) -> int:

@@ -24,6 +24,7 @@ setup(
                 "csrc/fusedexpr.hip",
                 "csrc/fusedjit.hip",
                 "csrc/bpe.hip",
+                "csrc/editdist.hip",
             ],
             libraries=["hiprtc"],
             extra_compile_args={

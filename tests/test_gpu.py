@@ -778,3 +778,25 @@ def test_external_sort_gpu_spill_matches_inmemory():
         cfg.memory_limit_bytes = old
     assert got["g"] == want["g"]
     assert got["v"] == want["v"]
+
+
+@pytest.mark.gpu
+def test_levenshtein_gpu_matches_oracle():
+    """Thread-per-pair edit-distance kernel vs the python DP, incl.
+    non-ASCII and >512-byte rows that take the host fallback."""
+    import random
+    from daft_amd.functions.strings_extra import _lev, _lev_series
+    random.seed(31)
+    alpha = "abcdefgh"
+    a = ["".join(random.choice(alpha) for _ in range(random.randint(0, 40)))
+         for _ in range(5000)]
+    b = ["".join(random.choice(alpha) for _ in range(random.randint(0, 40)))
+         for _ in range(5000)]
+    a += ["naïve", "x" * 600, None]
+    b += ["naive", "x" * 599 + "y", "z"]
+    sa = Series.from_pylist("a", a, DataType.string()).to("cuda:0")
+    sb = Series.from_pylist("b", b, DataType.string()).to("cuda:0")
+    out = _lev_series(sa, sb).cpu().to_pylist()
+    for i, (x, y) in enumerate(zip(a, b)):
+        want = None if (x is None or y is None) else _lev(x, y)
+        assert out[i] == want, (i, x, y, out[i], want)
