@@ -19,8 +19,8 @@ def _fetch_one(u: Optional[str], on_error: str):
     try:
         if u.startswith("file://"):
             u = u[7:]
-        if u.startswith(("http://", "https://", "s3://", "s3a://")):
-            from ..io.object_store import get_source
+        from ..io.object_store import get_source, is_remote
+        if is_remote(u):
             return get_source(u).get(u)
         with open(u, "rb") as f:
             return f.read()

@@ -62,9 +62,60 @@ class HTTPConfig:
 
 
 @dataclass
+class GCSConfig:
+    """Google Cloud Storage over its JSON API (ref:
+    daft-io/src/google_cloud.rs + common/io-config/src/gcs.rs).  Auth is
+    a static bearer token (config or GCS_TOKEN / GOOGLE_CLOUD_TOKEN env)
+    or anonymous — the service-account JWT exchange needs RSA signing,
+    which this offline image has no library for."""
+    project_id: Optional[str] = None
+    token: Optional[str] = None
+    anonymous: bool = False
+    endpoint_url: Optional[str] = None   # fake-gcs-server / emulators
+    num_tries: int = 5
+    retry_initial_backoff_ms: int = 100
+    connect_timeout_ms: int = 10_000
+    read_timeout_ms: int = 30_000
+
+    def resolved(self) -> "GCSConfig":
+        c = GCSConfig(**self.__dict__)
+        c.token = c.token or os.environ.get("GCS_TOKEN") or \
+            os.environ.get("GOOGLE_CLOUD_TOKEN")
+        c.endpoint_url = c.endpoint_url or \
+            os.environ.get("STORAGE_EMULATOR_HOST")
+        return c
+
+
+@dataclass
+class AzureConfig:
+    """Azure Blob Storage over the Blob REST API (ref:
+    daft-io/src/azure_blob.rs + common/io-config/src/azure.rs).  Auth is
+    SharedKey (account access key), a SAS token, or anonymous."""
+    storage_account: Optional[str] = None
+    access_key: Optional[str] = None
+    sas_token: Optional[str] = None
+    anonymous: bool = False
+    endpoint_url: Optional[str] = None   # Azurite: http://host:port/acct
+    num_tries: int = 5
+    retry_initial_backoff_ms: int = 100
+    connect_timeout_ms: int = 10_000
+    read_timeout_ms: int = 30_000
+
+    def resolved(self) -> "AzureConfig":
+        c = AzureConfig(**self.__dict__)
+        c.storage_account = c.storage_account or \
+            os.environ.get("AZURE_STORAGE_ACCOUNT")
+        c.access_key = c.access_key or os.environ.get("AZURE_STORAGE_KEY")
+        c.sas_token = c.sas_token or os.environ.get("AZURE_STORAGE_SAS_TOKEN")
+        return c
+
+
+@dataclass
 class IOConfig:
     s3: S3Config = field(default_factory=S3Config)
     http: HTTPConfig = field(default_factory=HTTPConfig)
+    gcs: GCSConfig = field(default_factory=GCSConfig)
+    azure: AzureConfig = field(default_factory=AzureConfig)
 
 
 # ---------------------------------------------------------------------------
@@ -401,6 +452,302 @@ class S3Source(ObjectSource):
         return out
 
 
+class GCSSource(ObjectSource):
+    """GCS JSON API: objects.get (alt=media, ranged), metadata, list with
+    pageToken paging, media/resumable upload.  Works against real GCS
+    (with a bearer token) and fake-gcs-server-style emulators.
+    Ref: daft-io/src/google_cloud.rs."""
+
+    RESUMABLE_CHUNK = 16 * 1024 * 1024
+
+    def __init__(self, config: Optional[GCSConfig] = None):
+        self.cfg = (config or GCSConfig()).resolved()
+        import requests
+        self._sess = requests.Session()
+
+    def _base(self) -> str:
+        ep = self.cfg.endpoint_url or "https://storage.googleapis.com"
+        return ep.rstrip("/")
+
+    def _split(self, path: str) -> Tuple[str, str]:
+        parsed = urllib.parse.urlsplit(path)
+        return parsed.netloc, parsed.path.lstrip("/")
+
+    def _headers(self) -> dict:
+        if self.cfg.anonymous or not self.cfg.token:
+            return {}
+        return {"Authorization": f"Bearer {self.cfg.token}"}
+
+    def _request(self, method: str, url: str, what: str, payload=None,
+                 extra_headers: Optional[dict] = None):
+        def go():
+            headers = self._headers()
+            if extra_headers:
+                headers.update(extra_headers)
+            r = self._sess.request(
+                method, url, data=payload, headers=headers,
+                timeout=(self.cfg.connect_timeout_ms / 1000,
+                         self.cfg.read_timeout_ms / 1000))
+            if r.status_code == 404:
+                raise NotFoundError(url)
+            if r.status_code >= 400:
+                raise ObjectStoreError(
+                    f"GCS {r.status_code} for {what}: {r.text[:200]}")
+            return r
+        return _with_retry(go, self.cfg.num_tries,
+                           self.cfg.retry_initial_backoff_ms, what)
+
+    def _obj_url(self, bucket: str, obj: str, suffix: str = "") -> str:
+        return (f"{self._base()}/storage/v1/b/{bucket}/o/"
+                f"{urllib.parse.quote(obj, safe='')}{suffix}")
+
+    def get(self, path, range_=None):
+        bucket, obj = self._split(path)
+        extra = {}
+        if range_ is not None:
+            extra["Range"] = f"bytes={range_[0]}-{range_[1] - 1}"
+        return self._request("GET", self._obj_url(bucket, obj, "?alt=media"),
+                             f"GET {path}", extra_headers=extra).content
+
+    def get_size(self, path):
+        import json
+        bucket, obj = self._split(path)
+        r = self._request("GET", self._obj_url(bucket, obj), f"STAT {path}")
+        return int(json.loads(r.text)["size"])
+
+    def put(self, path, data: bytes):
+        bucket, obj = self._split(path)
+        base = (f"{self._base()}/upload/storage/v1/b/{bucket}/o"
+                f"?name={urllib.parse.quote(obj, safe='')}")
+        if len(data) <= self.RESUMABLE_CHUNK:
+            self._request("POST", base + "&uploadType=media", f"PUT {path}",
+                          payload=data)
+            return
+        # resumable: one session POST, then sequential Content-Range PUTs
+        r = self._request("POST", base + "&uploadType=resumable",
+                          f"resumable-start {path}")
+        session = r.headers.get("Location") or r.headers.get("location")
+        if not session:
+            raise ObjectStoreError(f"resumable upload: no session URI "
+                                   f"for {path}")
+        total = len(data)
+        for off in range(0, total, self.RESUMABLE_CHUNK):
+            chunk = data[off:off + self.RESUMABLE_CHUNK]
+            hi = off + len(chunk) - 1
+            def go(chunk=chunk, off=off, hi=hi):
+                rr = self._sess.put(
+                    session, data=chunk,
+                    headers={"Content-Range":
+                             f"bytes {off}-{hi}/{total}"},
+                    timeout=(self.cfg.connect_timeout_ms / 1000,
+                             self.cfg.read_timeout_ms / 1000))
+                # 308 = chunk accepted, more expected
+                if rr.status_code not in (200, 201, 308):
+                    raise ObjectStoreError(
+                        f"GCS {rr.status_code} for resumable {path}")
+                return rr
+            _with_retry(go, self.cfg.num_tries,
+                        self.cfg.retry_initial_backoff_ms,
+                        f"resumable-part {path}@{off}")
+
+    def list_prefix(self, path_prefix):
+        import json
+        bucket, prefix = self._split(path_prefix)
+        out = []
+        token = None
+        while True:
+            q = "prefix=" + urllib.parse.quote(prefix, safe="")
+            if token:
+                q += "&pageToken=" + urllib.parse.quote(token, safe="")
+            url = f"{self._base()}/storage/v1/b/{bucket}/o?{q}"
+            doc = json.loads(self._request("GET", url,
+                                           f"LIST {path_prefix}").text)
+            for item in doc.get("items", []):
+                out.append((f"gs://{bucket}/{item['name']}",
+                            int(item.get("size", 0))))
+            token = doc.get("nextPageToken")
+            if not token:
+                break
+        return out
+
+
+def _azure_sharedkey_auth(method: str, url: str, account: str, key_b64: str,
+                          headers: dict, content_length: int) -> str:
+    """SharedKey signature over the 2015+ canonical string
+    (ref: daft-io/src/azure_blob.rs credential handling)."""
+    import base64
+    parsed = urllib.parse.urlsplit(url)
+    # canonicalized x-ms-* headers, lowercase-sorted
+    xms = sorted((k.lower(), v) for k, v in headers.items()
+                 if k.lower().startswith("x-ms-"))
+    canon_headers = "".join(f"{k}:{v}\n" for k, v in xms)
+    # canonicalized resource: /account/path + sorted query params
+    # (Azurite-style endpoints already embed /account in the path)
+    path = parsed.path or "/"
+    if path == f"/{account}" or path.startswith(f"/{account}/"):
+        canon_res = path
+    else:
+        canon_res = f"/{account}{path}"
+    for k, v in sorted(urllib.parse.parse_qsl(parsed.query,
+                                              keep_blank_values=True)):
+        canon_res += f"\n{k.lower()}:{v}"
+    cl = str(content_length) if content_length else ""
+    sts = "\n".join([
+        method,
+        "",                      # Content-Encoding
+        "",                      # Content-Language
+        cl,                      # Content-Length ("" when 0)
+        "",                      # Content-MD5
+        headers.get("Content-Type", ""),
+        "",                      # Date (x-ms-date used instead)
+        "", "", "", "",          # If-Modified/Match/None-Match/Unmodified
+        headers.get("Range", ""),
+        canon_headers + canon_res])
+    sig = hmac.new(base64.b64decode(key_b64), sts.encode("utf-8"),
+                   hashlib.sha256).digest()
+    return f"SharedKey {account}:{base64.b64encode(sig).decode()}"
+
+
+class AzureBlobSource(ObjectSource):
+    """Azure Blob REST API: Get/Put Blob, Put Block + Put Block List for
+    large uploads, List Blobs with marker paging.  az:// and abfs[s]://
+    URIs name container/blob; the account comes from AzureConfig.
+    Ref: daft-io/src/azure_blob.rs."""
+
+    BLOCK_CHUNK = 16 * 1024 * 1024
+    API_VERSION = "2021-08-06"
+
+    def __init__(self, config: Optional[AzureConfig] = None):
+        self.cfg = (config or AzureConfig()).resolved()
+        import requests
+        self._sess = requests.Session()
+
+    def _base(self) -> str:
+        if self.cfg.endpoint_url:
+            return self.cfg.endpoint_url.rstrip("/")
+        acct = self.cfg.storage_account
+        if not acct:
+            raise ObjectStoreError(
+                "AzureConfig.storage_account (or AZURE_STORAGE_ACCOUNT) "
+                "is required for az:// paths")
+        return f"https://{acct}.blob.core.windows.net"
+
+    def _split(self, path: str) -> Tuple[str, str]:
+        parsed = urllib.parse.urlsplit(path)
+        container = parsed.netloc
+        # abfss://container@account.dfs.core.windows.net/blob form
+        if "@" in container:
+            container, host = container.split("@", 1)
+            if self.cfg.storage_account is None:
+                self.cfg.storage_account = host.split(".", 1)[0]
+        return container, parsed.path.lstrip("/")
+
+    def _url(self, container: str, blob: str, query: str = "") -> str:
+        url = f"{self._base()}/{container}"
+        if blob:
+            url += "/" + urllib.parse.quote(blob)
+        if query:
+            sep = "?"
+            if self.cfg.sas_token:
+                url += "?" + self.cfg.sas_token.lstrip("?")
+                sep = "&"
+            url += sep + query
+        elif self.cfg.sas_token:
+            url += "?" + self.cfg.sas_token.lstrip("?")
+        return url
+
+    def _request(self, method: str, url: str, what: str, payload: bytes = b"",
+                 extra_headers: Optional[dict] = None):
+        def go():
+            now = _dt.datetime.now(_dt.timezone.utc)
+            headers = {"x-ms-date": now.strftime("%a, %d %b %Y %H:%M:%S GMT"),
+                       "x-ms-version": self.API_VERSION}
+            if extra_headers:
+                headers.update(extra_headers)
+            if self.cfg.access_key and not self.cfg.anonymous and \
+                    not self.cfg.sas_token:
+                headers["Authorization"] = _azure_sharedkey_auth(
+                    method, url, self.cfg.storage_account or "",
+                    self.cfg.access_key, headers, len(payload))
+            r = self._sess.request(
+                method, url, data=payload if payload else None,
+                headers=headers,
+                timeout=(self.cfg.connect_timeout_ms / 1000,
+                         self.cfg.read_timeout_ms / 1000))
+            if r.status_code == 404:
+                raise NotFoundError(url)
+            if r.status_code >= 400:
+                raise ObjectStoreError(
+                    f"Azure {r.status_code} for {what}: {r.text[:200]}")
+            return r
+        return _with_retry(go, self.cfg.num_tries,
+                           self.cfg.retry_initial_backoff_ms, what)
+
+    def get(self, path, range_=None):
+        container, blob = self._split(path)
+        extra = {}
+        if range_ is not None:
+            extra["Range"] = f"bytes={range_[0]}-{range_[1] - 1}"
+        return self._request("GET", self._url(container, blob),
+                             f"GET {path}", extra_headers=extra).content
+
+    def get_size(self, path):
+        container, blob = self._split(path)
+        r = self._request("HEAD", self._url(container, blob), f"HEAD {path}")
+        return int(r.headers["Content-Length"])
+
+    def put(self, path, data: bytes):
+        import base64
+        container, blob = self._split(path)
+        if len(data) <= self.BLOCK_CHUNK:
+            self._request("PUT", self._url(container, blob), f"PUT {path}",
+                          payload=data,
+                          extra_headers={"x-ms-blob-type": "BlockBlob"})
+            return
+        ids = []
+        for i, off in enumerate(range(0, len(data), self.BLOCK_CHUNK)):
+            bid = base64.b64encode(f"blk{i:08d}".encode()).decode()
+            q = ("comp=block&blockid=" +
+                 urllib.parse.quote(bid, safe=""))
+            self._request("PUT", self._url(container, blob, q),
+                          f"put-block {path}#{i}",
+                          payload=data[off:off + self.BLOCK_CHUNK])
+            ids.append(bid)
+        body = ("<?xml version='1.0' encoding='utf-8'?><BlockList>" +
+                "".join(f"<Latest>{b}</Latest>" for b in ids) +
+                "</BlockList>").encode()
+        self._request("PUT", self._url(container, blob, "comp=blocklist"),
+                      f"put-blocklist {path}", payload=body,
+                      extra_headers={"Content-Type": "application/xml"})
+
+    def list_prefix(self, path_prefix):
+        parsed = urllib.parse.urlsplit(path_prefix)
+        container = parsed.netloc.split("@", 1)[0]
+        prefix = parsed.path.lstrip("/")
+        scheme = parsed.scheme or "az"
+        out = []
+        marker = None
+        while True:
+            q = ("restype=container&comp=list&prefix=" +
+                 urllib.parse.quote(prefix, safe=""))
+            if marker:
+                q += "&marker=" + urllib.parse.quote(marker, safe="")
+            r = self._request("GET", self._url(container, "", q),
+                              f"LIST {path_prefix}")
+            text = r.text
+            for m in re.finditer(
+                    r"<Blob>.*?<Name>([^<]+)</Name>.*?"
+                    r"<Content-Length>(\d+)</Content-Length>.*?</Blob>",
+                    text, re.S):
+                out.append((f"{scheme}://{container}/{m.group(1)}",
+                            int(m.group(2))))
+            mt = re.search(r"<NextMarker>([^<]+)</NextMarker>", text)
+            if mt is None:
+                break
+            marker = mt.group(1)
+        return out
+
+
 # ---------------------------------------------------------------------------
 # dispatch
 # ---------------------------------------------------------------------------
@@ -419,6 +766,10 @@ def get_source(path: str,
     scheme = urllib.parse.urlsplit(path).scheme
     if scheme in ("s3", "s3a"):
         return S3Source(cfg.s3)
+    if scheme in ("gs", "gcs"):
+        return GCSSource(cfg.gcs)
+    if scheme in ("az", "abfs", "abfss", "wasb", "wasbs"):
+        return AzureBlobSource(cfg.azure)
     if scheme in ("http", "https"):
         return HTTPSource(cfg.http)
     if scheme in ("", "file"):
@@ -426,9 +777,12 @@ def get_source(path: str,
     raise ValueError(f"unsupported object-store scheme {scheme!r} in {path}")
 
 
+_REMOTE_SCHEMES = ("s3", "s3a", "gs", "gcs", "az", "abfs", "abfss",
+                   "wasb", "wasbs", "http", "https")
+
+
 def is_remote(path: str) -> bool:
-    return urllib.parse.urlsplit(path).scheme in ("s3", "s3a", "http",
-                                                  "https")
+    return urllib.parse.urlsplit(path).scheme in _REMOTE_SCHEMES
 
 
 def glob_paths(pattern: str,
