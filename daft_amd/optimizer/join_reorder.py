@@ -121,7 +121,11 @@ def _tree_cost(node: lp.LogicalPlan):
                         denom = max(denom, nd)
         out = le_ * re_ / denom
         w = lw + rw
-        return lc + rc + out * w, out, w
+        # price reading/hashing both inputs as well as the output: an
+        # output-only metric let "small-output" joins of two huge inputs
+        # look free (q18's 17x DP regression before this term)
+        step = le_ * lw + re_ * rw + out * w
+        return lc + rc + step, out, w
     est = node.approx_num_rows()
     if est is None:
         return None, None, None
@@ -234,6 +238,16 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
 
     widths = [_rel_width(r) for r in rels]
 
+    # exact DP over connected subgraphs (the reference's DP-ccp,
+    # reorder_joins/ — here a bitmask DP: bushy trees, no cross products,
+    # same rows x width cost model as the greedy pass).  TPC-H tops out
+    # at 8 relations; 2^n subset DP is trivial at that size.
+    import os
+    dp = None
+    if len(rels) <= 12 and \
+            "dpjoin" not in os.environ.get("DAFT_AMD_DISABLE_RULES", ""):
+        dp = _dp_best(rels, ests, widths, bound, edge_ndv)
+
     order = [min(range(len(rels)), key=lambda i: ests[i])]
     placed = set(order)
     cur_est = ests[order[0]]
@@ -256,19 +270,29 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
         # wide build side through every later probe is what the row-count
         # metric missed (q10: customer's strings gathered per join)
         best = min(cands, key=lambda r: join_size(cur_est, r, cands[r]) *
-                   (cur_width + widths[r]))
+                   (cur_width + widths[r]) + ests[r] * widths[r])
         order.append(best)
         placed.add(best)
         joins_per_step.append([(te, re_) for _ti, te, re_ in cands[best]])
+        greedy_cost += cur_est * cur_width + ests[best] * widths[best]
         cur_est = max(1.0, join_size(cur_est, best, cands[best]))
         step_ests.append(cur_est)
         cur_width += widths[best]
         greedy_cost += cur_est * cur_width
 
     # only replace the author's join order when the estimate says the
-    # greedy order is clearly cheaper — hand-tuned DataFrame programs are
+    # new order is clearly cheaper — hand-tuned DataFrame programs are
     # usually already good, and estimates are coarse
     orig_cost, _e, _w = _tree_cost(plan)
+    if dp is not None and dp[0] < greedy_cost:
+        dp_cost, dp_tree = dp
+        if orig_cost is not None and dp_cost >= 0.7 * orig_cost:
+            return None
+        from ..expressions.expressions import ColumnRef as CR
+        want = plan.schema.names()
+        if dp_tree.schema.names() != want:
+            return lp.Project(dp_tree, [CR(n) for n in want])
+        return dp_tree
     if orig_cost is not None and greedy_cost >= 0.7 * orig_cost:
         return None
     if order == list(range(len(rels))):
@@ -297,6 +321,99 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     if tree.schema.names() != want:
         return lp.Project(tree, [CR(n) for n in want])
     return tree
+
+
+def _dp_best(rels, ests, widths, bound, edge_ndv):
+    """Bitmask DP over connected subsets (DP-ccp equivalent, bushy trees):
+    best[S] = (cost, est_rows, width, split) minimizing sum of
+    out_rows x out_width over all joins, considering every connected
+    partition S = A ∪ B with at least one equi-edge across the cut.
+    Returns (cost, logical join tree) or None."""
+    n = len(rels)
+    full = (1 << n) - 1
+    adj = [0] * n
+    for (i, _le, j, _re) in bound:
+        adj[i] |= 1 << j
+        adj[j] |= 1 << i
+
+    def connected(mask: int) -> bool:
+        lo = mask & -mask
+        seen = lo
+        frontier = lo
+        while frontier:
+            nxt = 0
+            m = frontier
+            while m:
+                b = m & -m
+                nxt |= adj[b.bit_length() - 1]
+                m ^= b
+            frontier = nxt & mask & ~seen
+            seen |= frontier
+        return seen == mask
+
+    best = {}
+    for i in range(n):
+        best[1 << i] = (0.0, ests[i], widths[i], None)
+    # subsets in increasing popcount order
+    masks = sorted((m for m in range(3, full + 1) if m & (m - 1)),
+                   key=lambda m: bin(m).count("1"))
+    for S in masks:
+        if not connected(S):
+            continue
+        entry = None
+        # enumerate partitions {A, B} of S once (A contains S's lowest bit)
+        lo = S & -S
+        rest = S ^ lo
+        sub = rest
+        while True:
+            A = sub | lo
+            B = S ^ A
+            if B and A in best and B in best:
+                cross = [(i, le, j, re) for (i, le, j, re) in bound
+                         if ((1 << i) & A and (1 << j) & B) or
+                            ((1 << j) & A and (1 << i) & B)]
+                if cross:
+                    ca, ea, wa, _ = best[A]
+                    cb, eb, wb, _ = best[B]
+                    denom = 1.0
+                    for (i, le, j, re) in cross:
+                        denom = max(denom, edge_ndv(i, le) or 1.0,
+                                    edge_ndv(j, re) or 1.0)
+                    out = max(1.0, ea * eb / denom)
+                    w = wa + wb
+                    cost = ca + cb + ea * wa + eb * wb + out * w
+                    if entry is None or cost < entry[0]:
+                        entry = (cost, out, w, (A, B, cross))
+            if sub == 0:
+                break
+            sub = (sub - 1) & rest
+        if entry is not None:
+            best[S] = entry
+    if full not in best or best[full][3] is None:
+        return None
+
+    def build(S: int) -> Tuple[lp.LogicalPlan, float]:
+        cost, est, _w, split = best[S]
+        if split is None:
+            return rels[S.bit_length() - 1], ests[S.bit_length() - 1]
+        A, B, cross = split
+        ta, ea = build(A)
+        tb, eb = build(B)
+        a_keys, b_keys = [], []
+        for (i, le, j, re) in cross:
+            if (1 << i) & A:
+                a_keys.append(le)
+                b_keys.append(re)
+            else:
+                a_keys.append(re)
+                b_keys.append(le)
+        # physical hash joins build on the RIGHT child: smaller side there
+        if ea <= eb:
+            return lp.Join(tb, ta, b_keys, a_keys, "inner"), est
+        return lp.Join(ta, tb, a_keys, b_keys, "inner"), est
+
+    tree, _ = build(full)
+    return best[full][0], tree
 
 
 def swap_join_builds(plan: lp.LogicalPlan) -> lp.LogicalPlan:

@@ -233,3 +233,64 @@ def test_ndv_birthday_estimator():
                data=_t.arange(n, dtype=_t.int64))
     est = sample_ndv(s, n_rows=n)
     assert est >= n * 0.5
+
+
+def test_dp_join_enumeration_bushy_and_correct():
+    """The bitmask DP (DP-ccp equivalent) explores bushy trees the
+    left-deep greedy can't: a 4-relation chain big-small-small-big where
+    the optimum joins the two small middles first.  Results must match
+    the un-reordered plan, and _dp_best's cost can never exceed greedy's
+    (left-deep trees are inside DP's search space)."""
+    import os
+    import random
+    random.seed(7)
+    n = 4000
+    bigA = daft.from_pydict({"ak": [random.randint(0, 9) for _ in range(n)],
+                             "av": [float(i) for i in range(n)]})
+    s1 = daft.from_pydict({"bk": list(range(10)), "bl": list(range(10))})
+    s2 = daft.from_pydict({"ck": list(range(10)), "cl": list(range(10))})
+    bigB = daft.from_pydict({"dk": [random.randint(0, 9) for _ in range(n)],
+                             "dv": list(range(n))})
+    # user order: bigA ⋈ s1, then ⋈ s2, then ⋈ bigB
+    j = bigA.join(s1, left_on="ak", right_on="bk") \
+        .join(s2, left_on="bl", right_on="ck") \
+        .join(bigB, left_on="cl", right_on="dk")
+    got_on = j.collect().to_pydict()
+    os.environ["DAFT_AMD_DISABLE_RULES"] = "dpjoin,reorder"
+    try:
+        got_off = j.collect().to_pydict()
+    finally:
+        del os.environ["DAFT_AMD_DISABLE_RULES"]
+    assert sorted(got_on["av"]) == sorted(got_off["av"])
+    assert sorted(got_on["dv"]) == sorted(got_off["dv"])
+
+
+def test_dp_best_unit():
+    """_dp_best on a synthetic 3-relation chain returns a full tree with
+    the smaller input on the build (right) side of each join."""
+    from daft_amd.optimizer.join_reorder import _dp_best
+    from daft_amd.expressions.expressions import ColumnRef
+    a = daft.from_pydict({"x": list(range(1000))})._builder.plan
+    b = daft.from_pydict({"y": list(range(10))})._builder.plan
+    c = daft.from_pydict({"z": list(range(100000))})._builder.plan
+    rels = [a, b, c]
+    ests = [1000.0, 10.0, 100000.0]
+    widths = [8.0, 8.0, 8.0]
+    bound = [(0, ColumnRef("x"), 1, ColumnRef("y")),
+             (1, ColumnRef("y"), 2, ColumnRef("z"))]
+    res = _dp_best(rels, ests, widths, bound,
+                   lambda ri, e: {0: 1000.0, 1: 10.0, 2: 10.0}[ri])
+    assert res is not None
+    cost, tree = res
+    assert cost > 0
+    assert set(tree.schema.names()) == {"x", "y", "z"}
+    # every join's right child must be the smaller-estimated side
+    def check(node):
+        if isinstance(node, lp.Join):
+            le = node.children[0].approx_num_rows()
+            re = node.children[1].approx_num_rows()
+            if le is not None and re is not None:
+                assert re <= le
+            for ch in node.children:
+                check(ch)
+    check(tree)
