@@ -20,6 +20,10 @@ from ..expressions.expressions import (Between, BinaryOp, ColumnRef, ExprNode,
                                        IsIn, IsNull, Literal, Not, ScalarFn)
 from ..logical import plan as lp
 
+# rolling telemetry of reorder decisions: (n_rels, dp_cost, greedy_cost,
+# orig_cost, chosen) — read by tests and EXPLAIN debugging
+DECISIONS: List[tuple] = []
+
 
 # ---------------------------------------------------------------------------
 # predicate selectivity (used by Filter.approx_num_rows)
@@ -284,6 +288,12 @@ def _reorder_root(plan: lp.LogicalPlan) -> Optional[lp.LogicalPlan]:
     # new order is clearly cheaper — hand-tuned DataFrame programs are
     # usually already good, and estimates are coarse
     orig_cost, _e, _w = _tree_cost(plan)
+    if len(DECISIONS) > 256:
+        del DECISIONS[:128]
+    DECISIONS.append((len(rels), dp[0] if dp else None, greedy_cost,
+                      orig_cost,
+                      "dp" if dp is not None and dp[0] < greedy_cost
+                      else "greedy"))
     if dp is not None and dp[0] < greedy_cost:
         dp_cost, dp_tree = dp
         if orig_cost is not None and dp_cost >= 0.7 * orig_cost:
