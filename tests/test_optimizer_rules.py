@@ -237,32 +237,39 @@ def test_ndv_birthday_estimator():
 
 def test_dp_join_enumeration_bushy_and_correct():
     """The bitmask DP (DP-ccp equivalent) explores bushy trees the
-    left-deep greedy can't: a 4-relation chain big-small-small-big where
-    the optimum joins the two small middles first.  Results must match
-    the un-reordered plan, and _dp_best's cost can never exceed greedy's
-    (left-deep trees are inside DP's search space)."""
+    left-deep greedy can't: two clusters that each need selective
+    reduction (A⋈B, C⋈D keep 1% of keys) joined on a low-NDV cross key.
+    Any left-deep order must materialize a high-fanout intermediate;
+    the bushy plan reduces both sides first.  DP must be chosen (2x
+    decisive modeled win) and results must match the unoptimized plan."""
     import os
     import random
+    from daft_amd.optimizer import join_reorder as jr
     random.seed(7)
-    n = 4000
-    bigA = daft.from_pydict({"ak": [random.randint(0, 9) for _ in range(n)],
-                             "av": [float(i) for i in range(n)]})
-    s1 = daft.from_pydict({"bk": list(range(10)), "bl": list(range(10))})
-    s2 = daft.from_pydict({"ck": list(range(10)), "cl": list(range(10))})
-    bigB = daft.from_pydict({"dk": [random.randint(0, 9) for _ in range(n)],
-                             "dv": list(range(n))})
-    # user order: bigA ⋈ s1, then ⋈ s2, then ⋈ bigB
-    j = bigA.join(s1, left_on="ak", right_on="bk") \
-        .join(s2, left_on="bl", right_on="ck") \
-        .join(bigB, left_on="cl", right_on="dk")
+    n = 8000
+    A = daft.from_pydict({"akey": [random.randint(0, 99) for _ in range(n)],
+                          "across": [random.randint(0, 9) for _ in range(n)],
+                          "av": [float(i) for i in range(n)]})
+    B = daft.from_pydict({"bkey": [0], "bv": [1.0]})
+    C = daft.from_pydict({"ckey": [random.randint(0, 99) for _ in range(n)],
+                          "ccross": [random.randint(0, 9) for _ in range(n)],
+                          "cv": list(range(n))})
+    D = daft.from_pydict({"dkey": [0], "dv": [2.0]})
+    # author order is the worst left-deep: A ⋈ C (10-NDV fanout) first
+    j = A.join(C, left_on="across", right_on="ccross") \
+        .join(B, left_on="akey", right_on="bkey") \
+        .join(D, left_on="ckey", right_on="dkey")
+    n0 = len(jr.DECISIONS)
     got_on = j.collect().to_pydict()
+    assert any(d[4] == "dp" for d in jr.DECISIONS[n0:]), \
+        "DP should win decisively on the bushy shape"
     os.environ["DAFT_AMD_DISABLE_RULES"] = "dpjoin,reorder"
     try:
         got_off = j.collect().to_pydict()
     finally:
         del os.environ["DAFT_AMD_DISABLE_RULES"]
     assert sorted(got_on["av"]) == sorted(got_off["av"])
-    assert sorted(got_on["dv"]) == sorted(got_off["dv"])
+    assert sorted(got_on["cv"]) == sorted(got_off["cv"])
 
 
 def test_dp_best_unit():
