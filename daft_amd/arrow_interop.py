@@ -188,12 +188,16 @@ def from_arrow_array(name: str, arr) -> Series:
             lows = np.ascontiguousarray(raw[2 * off::2][:len(a2)])
             return Series(name, dt, data=torch.from_numpy(lows.copy()),
                           validity=validity)
-        vals = arr.cast(pa.float64())
-        np_vals = vals.to_numpy(zero_copy_only=False)
-        np_vals = np.nan_to_num(np_vals) if arr.null_count else np_vals
-        return Series(name, dt, data=torch.from_numpy(
-            np.ascontiguousarray(np_vals, dtype=np.float64)),
-            validity=validity)
+        # wide (p > 18): both little-endian limbs -> struct<lo, hi>
+        a2 = arr.combine_chunks() if hasattr(arr, "combine_chunks") else arr
+        buf = a2.buffers()[1]
+        off = a2.offset
+        raw = np.frombuffer(buf, dtype=np.int64, count=2 * (off + len(a2)))
+        lo = np.ascontiguousarray(raw[2 * off::2][:len(a2)])
+        hi = np.ascontiguousarray(raw[2 * off + 1::2][:len(a2)])
+        from .kernels import decimal128 as d128
+        return d128.make(name, dt, torch.from_numpy(lo.copy()),
+                         torch.from_numpy(hi.copy()), validity)
     if k == TypeKind.NULL:
         from .series import full_null
         return full_null(name, DataType.null(), len(arr))
@@ -274,7 +278,7 @@ def to_arrow_array(s: Series):
         vbuf = pa.array(~mask).buffers()[1] if mask is not None else None
         return pa.Array.from_buffers(atype, len(s), [vbuf], children=children)
     if k == TypeKind.DECIMAL128:
-        if s.data.dtype == torch.int64:
+        if s.data is not None and s.data.dtype == torch.int64:
             lows = s.data.numpy()
             i128 = np.empty((len(s), 2), dtype=np.int64)
             i128[:, 0] = lows
@@ -283,8 +287,15 @@ def to_arrow_array(s: Series):
                 else None
             return pa.Array.from_buffers(
                 atype, len(s), [vbuf, pa.py_buffer(i128.tobytes())])
-        vals = s.data.numpy()
-        return pa.array(vals, mask=mask).cast(atype)
+        # wide (p > 18): interleave the limbs into the 16-byte layout
+        from .kernels import decimal128 as d128
+        lo, hi = d128.limbs(s)
+        i128 = np.empty((len(s), 2), dtype=np.int64)
+        i128[:, 0] = lo.numpy()
+        i128[:, 1] = hi.numpy()
+        vbuf = pa.array(~mask).buffers()[1] if mask is not None else None
+        return pa.Array.from_buffers(
+            atype, len(s), [vbuf, pa.py_buffer(i128.tobytes())])
     if k == TypeKind.NULL:
         return pa.nulls(len(s))
     np_arr = s.data.numpy() if s.data.dtype not in (

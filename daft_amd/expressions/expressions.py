@@ -447,10 +447,10 @@ class Agg(ExprNode):
         if k == AggKind.SUM:
             if f.dtype.is_decimal():
                 # widen precision for the running sum (ref: decimal sum
-                # gets p=38; int64 storage caps us at 18)
+                # gets p=38).  p<=18 keeps int64 storage; wider inputs
+                # sum exactly on two limbs (kernels/decimal128.py)
                 return Field(f.name, DataType.decimal128(
-                    18, f.dtype.scale) if f.dtype.precision <= 18
-                    else DataType.float64())
+                    18 if f.dtype.precision <= 18 else 38, f.dtype.scale))
             if f.dtype.is_integer():
                 return Field(f.name, DataType.int64()
                              if f.dtype.is_signed_integer()

@@ -78,6 +78,12 @@ def _col_desc(s: Series):
 def _descs(series: Sequence[Series]):
     tags, datas, offs, vals = [], [], [], []
     for s in series:
+        if s.dtype.is_decimal() and s.children:
+            raise NotImplementedError(
+                "GPU group/join/partition keys of wide decimals (p>18) "
+                "are not supported — cast to a p<=18 decimal, integer or "
+                "string key first (CPU grouping of wide keys works via "
+                "the exact host path)")
         t, d, o, v = _col_desc(s)
         tags.append(t)
         datas.append(d)
@@ -165,7 +171,7 @@ def take(s: Series, indices: torch.Tensor,
         child = s.children[0].take(child_idx)
         return Series(s.name, s.dtype, children=[child], validity=validity,
                       length=n_out)
-    if k == TypeKind.STRUCT:
+    if k == TypeKind.STRUCT or (k == TypeKind.DECIMAL128 and s.children):
         children = [c.take(safe_idx, has_neg=False) for c in s.children]
         return Series(s.name, s.dtype, children=children, validity=validity,
                       length=n_out)
@@ -278,7 +284,8 @@ def concat(series: List[Series]) -> Series:
         return Series(s0.name, dtype, data=data, offsets=new_off,
                       validity=validity)
     if k in (TypeKind.FIXED_SIZE_LIST, TypeKind.EMBEDDING,
-             TypeKind.FIXED_SHAPE_TENSOR, TypeKind.STRUCT):
+             TypeKind.FIXED_SHAPE_TENSOR, TypeKind.STRUCT) or \
+            (k == TypeKind.DECIMAL128 and s0.children):
         nchild = len(s0.children)
         children = [concat([s.children[i] for s in series])
                     for i in range(nchild)]
@@ -304,6 +311,60 @@ def cast(s: Series, dtype: DataType) -> Series:
         return full_null(s.name, dtype, len(s), s.device)
     if k == TypeKind.NULL:
         return full_null(s.name, dtype, len(s), s.device)
+    if k == TypeKind.DECIMAL128 and s.children:
+        # wide (p>18) source: limb arithmetic (kernels/decimal128.py)
+        from . import decimal128 as d128
+        lo, hi = d128.limbs(s)
+        sc = s.dtype.scale
+        if nk == TypeKind.DECIMAL128:
+            if dtype.scale > sc:
+                lo, hi = d128.mul128_pow10(lo, hi, dtype.scale - sc)
+            elif dtype.scale < sc:
+                lo, hi = d128.divround128_pow10(lo, hi, sc - dtype.scale)
+            if dtype.precision > 18:
+                return d128.make(s.name, dtype, lo, hi, s.validity)
+            # narrowing: must fit a scaled int64
+            if bool((hi != (lo >> 63)).any()):
+                raise ValueError(
+                    f"decimal cast overflow: {s.dtype!r} -> {dtype!r}")
+            return Series(s.name, dtype, data=lo, validity=s.validity)
+        if dtype.is_floating():
+            out = (d128.to_float64(lo, hi) / (10.0 ** sc)) \
+                .to(dtype.to_torch())
+            return Series(s.name, dtype, data=out, validity=s.validity)
+        if dtype.is_integer():
+            ilo, _ihi = d128.divround128_pow10(lo, hi, sc,
+                                               round_half=False) \
+                if sc else (lo, hi)
+            tdt = dtype.to_torch()
+            out = ilo if tdt == torch.int64 else (
+                ilo.view(torch.uint64) if tdt == torch.uint64
+                else ilo.to(tdt))
+            return Series(s.name, dtype, data=out, validity=s.validity)
+        # strings / everything else: exact host path below (to_pylist)
+    if nk == TypeKind.DECIMAL128 and dtype.precision > 18 and \
+            s.dtype.is_numeric() and (s.data is not None or s.children):
+        from . import decimal128 as d128
+        if k == TypeKind.DECIMAL128:      # narrow -> wide: sign-extend
+            lo, hi = d128.from_int64(s.data)
+            dsc = dtype.scale - s.dtype.scale
+        elif s.dtype.is_floating():
+            v = s.data.to(torch.float64) * (10.0 ** dtype.scale)
+            fhi = torch.floor(v / float(1 << 64))
+            flo = v - fhi * float(1 << 64)       # in [0, 2^64)
+            half = float(1 << 63)
+            lo = torch.where(flo >= half, (flo - half).to(torch.int64)
+                             + (-(1 << 63)), flo.to(torch.int64))
+            hi = fhi.to(torch.int64)
+            dsc = 0
+        else:
+            lo, hi = d128.from_int64(s.data.to(torch.int64))
+            dsc = dtype.scale
+        if dsc > 0:
+            lo, hi = d128.mul128_pow10(lo, hi, dsc)
+        elif dsc < 0:
+            lo, hi = d128.divround128_pow10(lo, hi, -dsc)
+        return d128.make(s.name, dtype, lo, hi, s.validity)
     if k == TypeKind.DECIMAL128 and s.data is not None and \
             s.data.dtype == torch.int64:
         sc = s.dtype.scale
@@ -508,6 +569,15 @@ def _decimal_binary(l: Series, r: Series, op: str, validity):
     def to_float(s: Series) -> Series:
         return cast(s, DataType.float64()) if s.dtype.is_decimal() else s
 
+    from . import decimal128 as d128
+    any_wide = (l.dtype.is_decimal() and d128.is_wide(l.dtype)) or \
+        (r.dtype.is_decimal() and d128.is_wide(r.dtype))
+    if any_wide:
+        if op in ("add", "sub", "mul") and not l.dtype.is_floating() \
+                and not r.dtype.is_floating():
+            return _decimal_wide_binary(l, r, op, validity)
+        return binary_op(to_float(l), to_float(r), op)   # div etc: f64
+
     if not (_dec_phys_int(l.dtype) or _dec_phys_int(r.dtype)) or \
             l.dtype.is_floating() or r.dtype.is_floating() or \
             op in ("div", "pow"):
@@ -542,7 +612,11 @@ def _decimal_binary(l: Series, r: Series, op: str, validity):
     if op == "mul":
         sc = asc + bsc
         p = min(18, ap + bp)
-        if sc > 18 or (ap - asc) + (bp - bsc) + sc > 18:
+        if (ap - asc) + (bp - bsc) + sc > 18:
+            # result exceeds int64: promote to an exact wide decimal
+            # when it fits p38, else f64
+            if sc <= 38 and ap + bp + 1 <= 38:
+                return _decimal_wide_binary(l, r, op, validity)
             return binary_op(to_float(l), to_float(r), op)
         out = av * bv
         return Series(l.name, DataType.decimal128(p, sc), data=out,
@@ -551,6 +625,47 @@ def _decimal_binary(l: Series, r: Series, op: str, validity):
     if op in ("floordiv", "mod"):
         return binary_op(to_float(l), to_float(r), op)
     return None
+
+
+def _decimal_wide_binary(l: Series, r: Series, op: str, validity):
+    """Exact add/sub/mul when either operand (or the result) is a wide
+    decimal: two-limb carry arithmetic (kernels/decimal128.py), device-
+    portable torch ops."""
+    from . import decimal128 as d128
+
+    def parts(s: Series):
+        if s.dtype.is_decimal():
+            if d128.is_wide(s.dtype):
+                lo, hi = d128.limbs(s)
+            else:
+                lo, hi = d128.from_int64(s.data.to(torch.int64))
+            return lo, hi, s.dtype.scale, s.dtype.precision
+        v = s.data.to(torch.int64)
+        lo, hi = d128.from_int64(v)
+        if len(s) == 1:
+            p = max(1, len(str(abs(int(v.item())))))
+        else:
+            p = 19
+        return lo, hi, 0, p
+
+    alo, ahi, asc, ap = parts(l)
+    blo, bhi, bsc, bp = parts(r)
+    if op in ("add", "sub"):
+        sc = max(asc, bsc)
+        if asc < sc:
+            alo, ahi = d128.mul128_pow10(alo, ahi, sc - asc)
+        if bsc < sc:
+            blo, bhi = d128.mul128_pow10(blo, bhi, sc - bsc)
+        olo, ohi = (d128.add128 if op == "add" else d128.sub128)(
+            alo, ahi, blo, bhi)
+        p = min(38, max(ap - asc, bp - bsc) + sc + 1)
+    else:                                   # mul
+        sc = asc + bsc
+        olo, ohi = d128.mul128(alo, ahi, blo, bhi)
+        p = min(38, ap + bp + 1)
+    dt = DataType.decimal128(max(p, 19), sc)
+    return d128.make(l.name, dt, olo, ohi,
+                     _expand_validity(validity, int(olo.numel())))
 
 
 def _temporal_binary(l: Series, r: Series, op: str, validity):
@@ -606,6 +721,11 @@ def compare_op(l: Series, r: Series, op: str) -> Series:
             l = cast(l, st)
         if r.dtype != st:
             r = cast(r, st)
+        from . import decimal128 as d128
+        if st.is_decimal() and d128.is_wide(st):
+            out = d128.cmp128(*d128.limbs(l), *d128.limbs(r), op)
+            return Series(l.name, DataType.bool(), data=out,
+                          validity=validity)
     st = supertype(l.dtype, r.dtype)
     tdt = st.to_torch()
 
@@ -724,6 +844,18 @@ def if_else(cond: Series, t: Series, f: Series) -> Series:
     m = cond.data.to(torch.bool)
     if cond.validity is not None:
         m = m & cond.validity
+    if out_dt.is_decimal() and t.children:
+        # wide decimal: select each limb
+        from . import decimal128 as d128
+        tlo, thi = d128.limbs(t)
+        flo, fhi = d128.limbs(f)
+        tv = t.validity if t.validity is not None else torch.ones_like(m)
+        fv = f.validity if f.validity is not None else torch.ones_like(m)
+        validity = torch.where(m, tv, fv)
+        if bool(validity.all().item()):
+            validity = None
+        return d128.make(t.name, out_dt, torch.where(m, tlo, flo),
+                         torch.where(m, thi, fhi), validity)
     if out_dt.is_fixed_width():
         td, fd = t.data, f.data
         uview = None

@@ -29,6 +29,12 @@ _SIGN64 = -0x8000000000000000  # 1 << 63 as int64
 
 def hash_columns(series: Sequence[Series], seed: int = 0) -> torch.Tensor:
     """64-bit combined row hash over the given key columns (int64 bit-pattern)."""
+    for s in series:
+        if s.dtype.is_decimal() and s.children:
+            raise NotImplementedError(
+                "group/join/partition keys of wide decimals (p>18) are "
+                "not supported — cast to a p<=18 decimal, integer or "
+                "string key first")
     if _is_gpu(series[0]):
         tags, datas, offs, vals = _descs(series)
         return native_required().hash_rows(tags, datas, offs, vals,
@@ -816,6 +822,26 @@ def argsort_multi(keys: Sequence[Series], descending: Sequence[bool],
     """Stable lexicographic argsort over multiple key columns."""
     n = len(keys[0])
     dev = keys[0].device
+    if any(s.dtype.is_decimal() and s.children for s in keys):
+        # wide decimals: expand into (hi signed, lo unsigned-ordered)
+        # int64 sub-keys — the LSD loop below makes that lexicographic
+        from . import decimal128 as d128
+        ek, ed, en = [], [], []
+        for s, de, nf in zip(keys, descending, nulls_first):
+            if s.dtype.is_decimal() and s.children:
+                lo, hi = d128.limbs(s)
+                ek.append(Series(s.name, DataType.int64(), data=hi,
+                                 validity=s.validity))
+                ek.append(Series(s.name, DataType.int64(),
+                                 data=d128.u_order_key(lo),
+                                 validity=s.validity))
+                ed.extend([de, de])
+                en.extend([nf, nf])
+            else:
+                ek.append(s)
+                ed.append(de)
+                en.append(nf)
+        keys, descending, nulls_first = ek, ed, en
     composed = _try_composed_argsort(keys, descending, nulls_first)
     if composed is not None:
         return composed

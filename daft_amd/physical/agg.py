@@ -80,8 +80,7 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
     if mask is not None:
         v = mask if values.validity is None else (values.validity & mask)
         values = values.with_validity(v)
-    if values.dtype.is_decimal() and values.data is not None and \
-            values.data.dtype == torch.int64 and kind in (
+    if values.dtype.is_decimal() and kind in (
             AggKind.MEAN, AggKind.STDDEV, AggKind.VARIANCE, AggKind.SKEW,
             AggKind.APPROX_PERCENTILE, AggKind.SKETCH):
         # moment-style aggs run in f64; SUM/MIN/MAX stay exact scaled-int
@@ -96,6 +95,10 @@ def compute_agg(batch: RecordBatch, group_ids: Optional[torch.Tensor],
         if values.dtype.kind in (TypeKind.STRING, TypeKind.BINARY) and \
                 kind in (AggKind.MIN, AggKind.MAX):
             return _string_minmax(group_ids, num_groups, values, kind, name)
+        if values.dtype.is_decimal() and values.children:
+            return _wide_decimal_agg(group_ids, num_groups, values, kind,
+                                     name,
+                                     agg.to_field(batch.schema).dtype)
         data, cnt = rowops.grouped_agg(group_ids, num_groups, values,
                                        kind)
         out_dt = agg.to_field(batch.schema).dtype
@@ -287,6 +290,57 @@ def _first_valid_index(group_ids: torch.Tensor, num_groups: int,
     out = torch.full((num_groups,), n, dtype=torch.int64, device=dev)
     out.scatter_reduce_(0, group_ids, idx, reduce="amin")
     return torch.where(out == n, torch.full_like(out, -1), out)
+
+
+def _wide_decimal_agg(group_ids: torch.Tensor, num_groups: int,
+                      values: Series, kind: "AggKind", name: str,
+                      out_dt: DataType) -> Series:
+    """Exact SUM/MIN/MAX for wide (p>18) decimals on two int64 limbs.
+    SUM accumulates 32-bit halves (kernels/decimal128.sum128); MIN/MAX
+    run two passes: extreme of the signed hi limb, then extreme of the
+    unsigned lo limb among rows tying on hi."""
+    from ..kernels import decimal128 as d128
+    lo, hi = d128.limbs(values)
+    validity = values.validity
+    dev = values.device
+    cnt_t = torch.zeros(num_groups, dtype=torch.int64, device=dev)
+    ones = torch.ones_like(lo)
+    if validity is not None:
+        ones = ones * validity.to(torch.int64)
+    cnt_t.scatter_add_(0, group_ids, ones)
+    out_valid = cnt_t > 0
+
+    if kind == AggKind.SUM:
+        slo, shi = lo, hi
+        if validity is not None:
+            z = torch.zeros_like(lo)
+            slo = torch.where(validity, lo, z)
+            shi = torch.where(validity, hi, z)
+
+        def seg(v: torch.Tensor) -> torch.Tensor:
+            out = torch.zeros(num_groups, dtype=torch.int64, device=dev)
+            out.scatter_add_(0, group_ids, v)
+            return out
+
+        olo, ohi = d128.sum128(slo, shi, seg)
+    else:
+        op = "min" if kind == AggKind.MIN else "max"
+        hi_ext, _ = rowops.grouped_agg(
+            group_ids, num_groups,
+            Series("h", DataType.int64(), data=hi, validity=validity), op)
+        hi_ext = hi_ext.to(torch.int64)
+        cand = hi == hi_ext[group_ids]
+        if validity is not None:
+            cand = cand & validity
+        lo_u = d128.u_order_key(lo)
+        lo_ext, _ = rowops.grouped_agg(
+            group_ids, num_groups,
+            Series("l", DataType.int64(), data=lo_u, validity=cand), op)
+        olo = d128.u_order_key(lo_ext.to(torch.int64))   # undo the flip
+        ohi = hi_ext
+    res = d128.make(name, values.dtype, olo, ohi,
+                    None if bool(out_valid.all().item()) else out_valid)
+    return res if res.dtype == out_dt else res.cast(out_dt).rename(name)
 
 
 def _string_minmax(group_ids, num_groups, values: Series, kind, name):

@@ -201,11 +201,13 @@ class DataType:
             # exact storage: scaled int64 for precision <= 18 (value *
             # 10^scale as a 64-bit integer — sums/compares/joins are exact;
             # ref semantics: daft-core Decimal128Array).  Wider decimals
-            # (p > 18) fall back to float64 (documented deviation; full
-            # i128 lands with a dedicated HIP type).
+            # (p > 18) store the scaled i128 as TWO int64 limb children
+            # (kernels/decimal128.py): exact add/sub/mul/compare/sort/sum
+            # via carry arithmetic, identical torch code on CPU and GPU.
             if self.precision <= 18:
                 return DataType.int64()
-            return DataType.float64()
+            return DataType.struct({"lo": DataType.int64(),
+                                    "hi": DataType.int64()})
         if k == TypeKind.EMBEDDING:
             return DataType.fixed_size_list(self.inner, self.size)
         if k == TypeKind.FIXED_SHAPE_TENSOR:
@@ -335,12 +337,12 @@ def supertype(a: DataType, b: DataType) -> DataType:
             sc = max(a.scale, b.scale)
             ip = max(a.precision - a.scale, b.precision - b.scale)
             p = ip + sc + 1
-            if p <= 18:
+            if p <= 38:
                 return DataType.decimal128(p, sc)
             return DataType.float64()
         dec, other = (a, b) if a.is_decimal() else (b, a)
-        if other.is_integer() and dec.precision <= 18:
-            p = min(18, max(dec.precision, 19 + dec.scale))
+        if other.is_integer():
+            p = min(38, max(dec.precision, 19 + dec.scale))
             return DataType.decimal128(p, dec.scale)
         return DataType.float64()
     if a.is_temporal() or b.is_temporal():

@@ -337,10 +337,19 @@ class Series:
                     v = v * us
                 out.append(_dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=v))
             return wrap(out)
-        if k == TypeKind.DECIMAL128 and s.data.dtype == torch.int64:
+        if k == TypeKind.DECIMAL128 and s.data is not None and \
+                s.data.dtype == torch.int64:
             import decimal as _dec
             return wrap([_dec.Decimal(int(v)).scaleb(-self.dtype.scale)
                          for v in s.data.numpy()])
+        if k == TypeKind.DECIMAL128 and s.children:
+            import decimal as _dec
+            from .kernels import decimal128 as d128
+            ints = d128.ints_from_tensors(*d128.limbs(s))
+            with _dec.localcontext() as _ctx:
+                _ctx.prec = 60
+                return wrap([_dec.Decimal(v).scaleb(-self.dtype.scale)
+                             for v in ints])
         if k == TypeKind.BOOL:
             return wrap([bool(v) for v in s.data.numpy()])
         vals = s.data
@@ -426,7 +435,8 @@ class Series:
                           children=[self.children[0].slice(start * sz,
                                                            end * sz)],
                           validity=validity, length=end - start)
-        if k in (TypeKind.STRUCT, TypeKind.IMAGE, TypeKind.TENSOR):
+        if k in (TypeKind.STRUCT, TypeKind.IMAGE, TypeKind.TENSOR) or \
+                (k == TypeKind.DECIMAL128 and self.children):
             # row-aligned children: slice each
             return Series(self.name, self.dtype,
                           children=[c.slice(start, end)
@@ -551,7 +561,7 @@ def _infer_dtype(values: Sequence[Any]) -> DataType:
             t = v.as_tuple()
             scl = max(0, -t.exponent) if isinstance(t.exponent, int) else 0
             digs = max(len(t.digits), scl + 1)
-            return DataType.decimal128(min(digs, 18), scl)
+            return DataType.decimal128(min(digs, 38), scl)
         if isinstance(v, str):
             return DataType.string()
         if isinstance(v, bytes):
@@ -690,14 +700,20 @@ def _from_pylist_typed(name: str, values: list, dtype: DataType) -> Series:
                       length=n)
     if k == TypeKind.DECIMAL128:
         import decimal as _dec
-        if dtype.to_physical().kind == TypeKind.INT64:
+        with _dec.localcontext() as _ctx:
+            _ctx.prec = 60          # default 28 rounds p>28 digit values
             scl = _dec.Decimal(1).scaleb(dtype.scale)
             ints = [0 if v is None else
                     int((_dec.Decimal(str(v)) * scl).to_integral_value(
                         rounding=_dec.ROUND_HALF_EVEN)) for v in values]
+        if dtype.to_physical().kind == TypeKind.INT64:
             return Series(name, dtype,
                           data=torch.tensor(ints, dtype=torch.int64),
                           validity=validity)
+        # wide (p>18): scaled i128 as two int64 limbs
+        from .kernels import decimal128 as d128
+        lo, hi = d128.tensors_from_ints(ints)
+        return d128.make(name, dtype, lo, hi, validity)
         vals = [0.0 if v is None else float(v) for v in values]
         return Series(name, dtype,
                       data=torch.tensor(vals, dtype=torch.float64),
@@ -757,6 +773,13 @@ def full_null(name: str, dtype: DataType, length: int, device="cpu") -> Series:
                     for f in dtype.fields]
         return Series(name, dtype, children=children, validity=validity,
                       length=length)
+    if k == TypeKind.DECIMAL128 and dtype.precision > 18:
+        z = torch.zeros(length, dtype=torch.int64, device=device)
+        return Series(name, dtype,
+                      children=[Series("lo", DataType.int64(), data=z),
+                                Series("hi", DataType.int64(),
+                                       data=z.clone())],
+                      validity=validity, length=length)
     t = torch.zeros(length, dtype=dtype.to_torch(), device=device)
     return Series(name, dtype, data=t, validity=validity)
 
