@@ -45,7 +45,7 @@ class DataFrame:
 
     # ---- transforms ---------------------------------------------------
     def select(self, *exprs: ColumnInput) -> "DataFrame":
-        from .functions.aliases import _Unnest
+        from .functions.aliases import _Explode, _Unnest
         if any(isinstance(e, _Unnest) for e in exprs):
             out = []
             for e in exprs:
@@ -57,7 +57,21 @@ class DataFrame:
                 else:
                     out.append(e)
             exprs = tuple(out)
-        return self._wrap(self._builder.select(list(exprs)))
+        explode_names = []
+        if any(isinstance(e, _Explode) for e in exprs):
+            out = []
+            for e in exprs:
+                if isinstance(e, _Explode):
+                    name = e.name or e.expr._node.out_name()
+                    out.append(e.expr.alias(name))
+                    explode_names.append(name)
+                else:
+                    out.append(e)
+            exprs = tuple(out)
+        df = self._wrap(self._builder.select(list(exprs)))
+        if explode_names:
+            df = df.explode(*explode_names)
+        return df
 
     def with_column(self, name: str, expr: Expression) -> "DataFrame":
         return self._wrap(self._builder.with_columns([expr.alias(name)]))
@@ -556,8 +570,22 @@ class DataFrame:
         ctx = get_context()
         yield from ctx.runner().run_iter(self._builder)
 
-    def iter_rows(self) -> Iterator[Dict[str, Any]]:
+    def iter_rows(self, column_format: str = "python"
+                  ) -> Iterator[Dict[str, Any]]:
+        """Stream rows as dicts.  column_format "python" yields plain
+        python values; "arrow" yields pyarrow scalars (ref:
+        daft/dataframe/dataframe.py iter_rows column_format)."""
+        if column_format not in ("python", "arrow"):
+            raise ValueError(
+                f"Unsupported column_format: {column_format}, supported "
+                "formats are 'python' and 'arrow'")
         for part in self.iter_partitions():
+            if column_format == "arrow":
+                tbl = part.to_arrow()
+                cols = {n: tbl.column(n) for n in tbl.column_names}
+                for i in range(len(tbl)):
+                    yield {n: c[i] for n, c in cols.items()}
+                continue
             d = part.to_pydict()
             names = list(d.keys())
             for i in range(len(part)):

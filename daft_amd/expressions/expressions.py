@@ -735,6 +735,70 @@ class Expression:
     def binary(self) -> "BinaryNamespace":
         return BinaryNamespace(self._node)
 
+    @property
+    def map(self) -> "MapNamespace":
+        return MapNamespace(self._node)
+
+    @property
+    def partitioning(self) -> "PartitioningNamespace":
+        return PartitioningNamespace(self._node)
+
+    # ---- method-form conveniences mirroring the reference Expression
+    # surface; each delegates to the equivalent daft_amd.functions entry
+    # (ref: daft/expressions/expressions.py method list) ----------------
+    def explode(self):
+        from ..functions.aliases import _Explode
+        return _Explode(self)
+
+    def unnest(self):
+        from ..functions.aliases import _Unnest
+        return _Unnest(self)
+
+    def eq_null_safe(self, other) -> "Expression":
+        from ..functions import eq_null_safe as _f
+        return _f(self, other)
+
+    def try_cast(self, dtype) -> "Expression":
+        from ..functions import try_cast as _f
+        return _f(self, dtype)
+
+    def median(self) -> "Expression":
+        from ..functions import median as _f
+        return _f(self)
+
+    def fill_nan(self, value) -> "Expression":
+        from ..functions import fill_nan as _f
+        return _f(self, value)
+
+    def is_nan(self) -> "Expression":
+        return self.float.is_nan()
+
+    def not_nan(self) -> "Expression":
+        return ~self.float.is_nan()
+
+    def is_inf(self) -> "Expression":
+        return self.float.is_inf()
+
+    def shift_left(self, other) -> "Expression":
+        from ..functions import shift_left as _f
+        return _f(self, other)
+
+    def shift_right(self, other) -> "Expression":
+        from ..functions import shift_right as _f
+        return _f(self, other)
+
+    def bitwise_and(self, other) -> "Expression":
+        from ..functions import bitwise_and as _f
+        return _f(self, other)
+
+    def bitwise_or(self, other) -> "Expression":
+        from ..functions import bitwise_or as _f
+        return _f(self, other)
+
+    def bitwise_xor(self, other) -> "Expression":
+        from ..functions import bitwise_xor as _f
+        return _f(self, other)
+
     def hash(self, seed: int = 0) -> "Expression":
         from .. import kernels
 
@@ -976,6 +1040,73 @@ class StringNamespace(_Namespace):
         return self._fn("tokenize_encode", tokenize_encode_series,
                         DataType.list(DataType.int32()), tokenizer)
 
+    def tokenize_decode(self, tokenizer: str = "simple"):
+        from ..functions.tokenize import tokenize_decode_series
+        return self._fn("tokenize_decode", tokenize_decode_series,
+                        DataType.string(), tokenizer)
+
+    # ---- method forms delegating to daft_amd.functions (ref:
+    # ExpressionStringNamespace surface) --------------------------------
+    def _delegate(self, fname, *args, **kwargs):
+        from .. import functions as F
+        return getattr(F, fname)(Expression(self._node), *args, **kwargs)
+
+    def normalize(self, **kwargs):
+        return self._delegate("normalize", **kwargs)
+
+    def count_matches(self, patterns, whole_words: bool = False,
+                      case_sensitive: bool = True):
+        return self._delegate("count_matches", patterns, whole_words,
+                              case_sensitive)
+
+    def replace(self, search: str, replacement: str, regex: bool = False):
+        if regex:
+            return self._delegate("regexp_replace", search, replacement)
+        import re as _re
+        return self._delegate("regexp_replace", _re.escape(search),
+                              replacement)
+
+    def regexp_replace(self, pattern: str, replacement: str):
+        return self._delegate("regexp_replace", pattern, replacement)
+
+    def extract(self, pattern: str, group: int = 0):
+        return self._delegate("regexp_extract", pattern, group)
+
+    def extract_all(self, pattern: str, group: int = 0):
+        return self._delegate("regexp_extract_all", pattern, group)
+
+    def regexp_split(self, pattern: str):
+        return self._delegate("regexp_split", pattern)
+
+    def split_part(self, delim: str, n: int):
+        return self._delegate("split_part", delim, n)
+
+    def substring_index(self, delim: str, n: int):
+        return self._delegate("substring_index", delim, n)
+
+    def translate(self, src: str, dst: str):
+        return self._delegate("translate", src, dst)
+
+    def to_snake_case(self):
+        return self._delegate("to_snake_case")
+
+    def to_camel_case(self):
+        return self._delegate("to_camel_case")
+
+    def to_kebab_case(self):
+        return self._delegate("to_kebab_case")
+
+    def to_title_case(self):
+        return self._delegate("to_title_case")
+
+    def levenshtein(self, other):
+        from ..functions import levenshtein_distance as _f
+        return _f(Expression(self._node), other)
+
+    def jaro_winkler(self, other):
+        from ..functions import jaro_winkler_similarity as _f
+        return _f(Expression(self._node), other)
+
 
 def _str_to_date(s: Series, fmt: str) -> Series:
     vals = s.cpu().to_pylist()
@@ -1014,6 +1145,62 @@ class TemporalNamespace(_Namespace):
     def total_days(self):
         return self._fn("total_days", _dur_total, DataType.int64(),
                         86_400_000_000)
+
+    def total_hours(self):
+        return self._fn("total_hours", _dur_total, DataType.int64(),
+                        3_600_000_000)
+
+    def total_minutes(self):
+        return self._fn("total_minutes", _dur_total, DataType.int64(),
+                        60_000_000)
+
+    def total_milliseconds(self):
+        return self._fn("total_milliseconds", _dur_total,
+                        DataType.int64(), 1_000)
+
+    def total_microseconds(self):
+        return self._fn("total_microseconds", _dur_total,
+                        DataType.int64(), 1)
+
+    def total_nanoseconds(self):
+        return self._fn("total_nanoseconds", _dur_total_ns,
+                        DataType.int64())
+
+    # method forms delegating to daft_amd.functions (ref:
+    # ExpressionDatetimeNamespace)
+    def _delegate(self, fname, *args, **kwargs):
+        from .. import functions as F
+        return getattr(F, fname)(Expression(self._node), *args, **kwargs)
+
+    def strftime(self, fmt: str = "%Y-%m-%d"):
+        return self._delegate("strftime", fmt)
+
+    def to_unix_epoch(self, unit: str = "s"):
+        return self._delegate("to_unix_epoch", unit)
+
+    def unix_date(self):
+        return self._delegate("unix_date")
+
+    def time(self):
+        return self._delegate("time")
+
+    def day_of_month(self):
+        return self._civil("day")
+
+    def millisecond(self):
+        return self._fn("millisecond", _dt_subsecond, DataType.int32(),
+                        1_000, 1_000)
+
+    def microsecond(self):
+        return self._fn("microsecond", _dt_subsecond, DataType.int32(),
+                        1, 1_000_000)
+
+    def nanosecond(self):
+        return self._fn("nanosecond", _dt_subsecond, DataType.int32(),
+                        None, None)
+
+    def date_trunc(self, interval: str):
+        return self.truncate(interval)
 
 
 def _days_to_civil(days: torch.Tensor):
@@ -1118,7 +1305,30 @@ def _dt_truncate(s: Series, interval: str) -> Series:
     elif interval in ("day", "1 day"):
         new_days = days
     else:
-        raise ValueError(f"unsupported truncate interval {interval}")
+        # sub-day intervals ("hour", "15 minutes", "30 seconds", ...):
+        # truncate the raw timestamp to the interval width
+        parts = interval.split()
+        count = 1
+        unit_word = parts[-1].rstrip("s")
+        if len(parts) == 2:
+            try:
+                count = int(parts[0])
+            except ValueError:
+                raise ValueError(
+                    f"unsupported truncate interval {interval}")
+        width_us = {"hour": 3_600_000_000, "minute": 60_000_000,
+                    "second": 1_000_000, "millisecond": 1_000,
+                    "microsecond": 1}.get(unit_word)
+        if width_us is None or s.dtype.kind == TypeKind.DATE:
+            raise ValueError(f"unsupported truncate interval {interval}")
+        unit = s.dtype.timeunit
+        mult = {"s": 1, "ms": 10**3, "us": 10**6, "ns": 10**9}[unit]
+        width = width_us * count * mult // 10**6
+        if width == 0:
+            raise ValueError(f"interval below {unit} resolution: "
+                             f"{interval}")
+        out = torch.div(s.data, width, rounding_mode="floor") * width
+        return Series(s.name, s.dtype, data=out, validity=s.validity)
     if s.dtype.kind == TypeKind.DATE:
         return Series(s.name, s.dtype, data=new_days.to(torch.int32),
                       validity=s.validity)
@@ -1137,9 +1347,60 @@ def _dur_total(s: Series, div_us: int) -> Series:
     return Series(s.name, DataType.int64(), data=out, validity=s.validity)
 
 
+def _dur_total_ns(s: Series) -> Series:
+    assert s.dtype.kind == TypeKind.DURATION
+    mult = {"s": 1_000_000_000, "ms": 1_000_000, "us": 1_000, "ns": 1}[
+        s.dtype.timeunit]
+    return Series(s.name, DataType.int64(), data=s.data * mult,
+                  validity=s.validity)
+
+
+def _dt_subsecond(s: Series, div, mod) -> Series:
+    """Sub-second components of a timestamp (pandas/arrow semantics:
+    millisecond 0..999, microsecond 0..999999, nanosecond 0..999 within
+    the microsecond)."""
+    unit = s.dtype.timeunit
+    if div is None:                     # nanosecond component
+        if unit != "ns":
+            out = torch.zeros_like(s.data, dtype=torch.int64)
+        else:
+            out = ((s.data % 1000) + 1000) % 1000
+        return Series(s.name, DataType.int32(),
+                      data=out.to(torch.int32), validity=s.validity)
+    mult = {"s": 1_000_000, "ms": 1_000, "us": 1}.get(unit)
+    us = s.data * mult if mult is not None else \
+        torch.div(s.data, 1000, rounding_mode="floor")
+    frac = ((us % 1_000_000) + 1_000_000) % 1_000_000
+    out = torch.div(frac, div, rounding_mode="floor") % mod
+    return Series(s.name, DataType.int32(), data=out.to(torch.int32),
+                  validity=s.validity)
+
+
 class ListNamespace(_Namespace):
     def length(self):
         return self._fn("list_length", _list_length, DataType.uint64())
+
+    def _delegate(self, fname, *args, **kwargs):
+        from .. import functions as F
+        return getattr(F, fname)(Expression(self._node), *args, **kwargs)
+
+    def append(self, value):
+        return self._delegate("list_append", value)
+
+    def filter(self, fn):
+        return self._delegate("list_filter", fn)
+
+    def map(self, fn):
+        return self._delegate("list_map", fn)
+
+    def count(self):
+        return self._delegate("list_count")
+
+    def bool_and(self):
+        return self._delegate("list_bool_and")
+
+    def bool_or(self):
+        return self._delegate("list_bool_or")
 
     def get(self, idx: int, default=None):
         return self._fn("list_get", _list_get,
@@ -1383,6 +1644,52 @@ def _struct_get(s: Series, field_name: str) -> Series:
                 child = child.with_validity(v)
             return child.rename(field_name)
     raise KeyError(field_name)
+
+
+class MapNamespace(_Namespace):
+    """col.map.get/keys/values over the first-class Map type (ref:
+    ExpressionMapNamespace, daft/expressions/expressions.py)."""
+
+    def get(self, key):
+        from ..functions import map_get as _f
+        return _f(Expression(self._node), key)
+
+    def keys(self):
+        from ..functions import map_keys as _f
+        return _f(Expression(self._node))
+
+    def values(self):
+        from ..functions import map_values as _f
+        return _f(Expression(self._node))
+
+
+class PartitioningNamespace(_Namespace):
+    """Iceberg-style partition transforms (ref:
+    ExpressionPartitioningNamespace)."""
+
+    def days(self):
+        from ..functions import partition_days as _f
+        return _f(Expression(self._node))
+
+    def hours(self):
+        from ..functions import partition_hours as _f
+        return _f(Expression(self._node))
+
+    def months(self):
+        from ..functions import partition_months as _f
+        return _f(Expression(self._node))
+
+    def years(self):
+        from ..functions import partition_years as _f
+        return _f(Expression(self._node))
+
+    def iceberg_bucket(self, n: int):
+        from ..functions import partition_iceberg_bucket as _f
+        return _f(Expression(self._node), n)
+
+    def iceberg_truncate(self, w):
+        from ..functions import partition_iceberg_truncate as _f
+        return _f(Expression(self._node), w)
 
 
 class FloatNamespace(_Namespace):

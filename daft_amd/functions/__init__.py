@@ -186,7 +186,7 @@ from .aliases import (  # noqa: E402,F401
     chunk, value_counts, explode, date, total_days, total_seconds, time,
     download, upload, resize, crop, image_file, image_file_metadata,
     decode_image_file, dot_product, euclidean_distance,
-    cosine_similarity, hamming_distance, map_get, map_keys,
+    cosine_similarity, hamming_distance, map_get, map_keys, map_values,
     to_utc_timestamp, from_utc_timestamp, convert_time_zone,
     convert_timezone, replace_time_zone, make_timestamp,
     make_timestamp_ltz, partition_days, partition_months,

@@ -58,7 +58,7 @@ def get(x, key, default=None): return _e(x).list.get(key, default)
 def chunk(x, n): return _e(x).list.chunk(n)
 def slice(x, a, b=None): return _e(x).list.slice(a, b)  # noqa: A001
 def value_counts(x): return _e(x).list.value_counts()
-def explode(x): return _e(x).explode()
+def explode(x): return _Explode(_e(x))
 def date(x): return _e(x).dt.date()
 def total_days(x): return _e(x).dt.total_days()
 def total_seconds(x): return _e(x).dt.total_seconds()
@@ -293,11 +293,15 @@ make_timestamp_ltz = make_timestamp
 # -- partitioning transforms (iceberg-style) ---------------------------------
 def partition_days(x): return _e(x).dt.date()
 def partition_months(x):
+    """Months since epoch 1970-01 (iceberg transform; ref:
+    Expression.partition_months)."""
     e = _e(x)
-    return (e.dt.year() * 12 + e.dt.month() - 1).alias("months")
+    return ((e.dt.year() - 1970) * 12 + e.dt.month() - 1).alias("months")
 
 
-def partition_years(x): return _e(x).dt.year()
+def partition_years(x):
+    """Years since epoch 1970 (iceberg transform)."""
+    return (_e(x).dt.year() - 1970).alias("years")
 def partition_hours(x):
     from .temporal import to_unix_epoch
     return (to_unix_epoch(x, "s") // 3600).alias("hours")
@@ -380,6 +384,19 @@ class _Unnest:
 
 def unnest(x):
     return _Unnest(_e(x))
+
+
+class _Explode:
+    """Marker consumed by DataFrame.select: the column's lists expand to
+    one row per element, other columns repeating (ref:
+    daft.functions.explode / Expression.explode)."""
+
+    def __init__(self, expr, name=None):
+        self.expr = expr
+        self.name = name
+
+    def alias(self, name):
+        return _Explode(self.expr, name)
 
 
 def first_value(x, ignore_nulls: bool = False):

@@ -27,7 +27,28 @@ def list_sum(x): return _e(x).list.sum()
 def list_min(x): return _e(x).list.min()
 def list_max(x): return _e(x).list.max()
 def list_mean(x): return _e(x).list.mean()
-def list_count(x): return _e(x).list.length()
+def list_count(x, mode: str = "valid"):
+    """Count list elements: mode 'valid' (default, non-null), 'all', or
+    'null' (ref: daft list_count CountMode)."""
+    if mode == "all":
+        return _e(x).list.length()
+    def run(s):
+        from ..series import Series
+        from ..schema import DataType
+        vals = s.cpu().to_pylist()
+        if mode == "valid":
+            out = [None if v is None else sum(e is not None for e in v)
+                   for v in vals]
+        else:
+            out = [None if v is None else sum(e is None for e in v)
+                   for v in vals]
+        r = Series.from_pylist(s.name, out, DataType.uint64())
+        return r.to(s.device) if s.is_gpu() else r
+    from .aliases import _to_node
+    from ..expressions.expressions import Expression, ScalarFn
+    from ..schema import DataType
+    return Expression(ScalarFn("list_count", run, [_to_node(_e(x))],
+                               DataType.uint64()))
 def list_chunk(x, n): return _e(x).list.chunk(n)
 def list_slice(x, a, b=None): return _e(x).list.slice(a, b)
 def list_agg(x): return _e(x).agg_list()
@@ -178,8 +199,12 @@ def try_cast(x, dtype):
                 out.append(None)
                 continue
             try:
-                Series.from_pylist("t", [v], dtype)
-                out.append(v)
+                one = kernels.cast(
+                    Series.from_pylist("t", [v], s.dtype.cpu_like()
+                                       if hasattr(s.dtype, "cpu_like")
+                                       else s.dtype), dtype)
+                pv = one.to_pylist()[0]
+                out.append(pv)
             except Exception:
                 out.append(None)
         try:

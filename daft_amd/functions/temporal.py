@@ -248,8 +248,13 @@ def to_unix_epoch(x, unit: str = "s"):
 
 def unix_date(x):
     def run(s: Series) -> Series:
-        return Series(s.name, DataType.int64(),
-                      data=s.data.to(torch.int64), validity=s.validity)
+        if s.dtype.kind == TypeKind.DATE:
+            d = s.data.to(torch.int64)
+        else:
+            d = torch.div(_ts_sub_us(s), 86_400_000_000,
+                          rounding_mode="floor")
+        return Series(s.name, DataType.int64(), data=d,
+                      validity=s.validity)
     return Expression(ScalarFn("unix_date", run, [_to_node(x)],
                                DataType.int64()))
 

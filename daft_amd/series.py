@@ -337,6 +337,16 @@ class Series:
                     v = v * us
                 out.append(_dt.datetime(1970, 1, 1) + _dt.timedelta(microseconds=v))
             return wrap(out)
+        if k == TypeKind.DURATION:
+            mult = {"s": 10**6, "ms": 10**3, "us": 1, "ns": 1}[
+                self.dtype.timeunit]
+            vals = s.data.numpy()
+            out = []
+            for v in vals:
+                v = int(v)
+                us = v // 1000 if self.dtype.timeunit == "ns" else v * mult
+                out.append(_dt.timedelta(microseconds=us))
+            return wrap(out)
         if k == TypeKind.DECIMAL128 and s.data is not None and \
                 s.data.dtype == torch.int64:
             import decimal as _dec
@@ -568,6 +578,8 @@ def _infer_dtype(values: Sequence[Any]) -> DataType:
             return DataType.binary()
         if isinstance(v, _dt.datetime):
             return DataType.timestamp("us")
+        if isinstance(v, _dt.timedelta):
+            return DataType.duration("us")
         if isinstance(v, _dt.date):
             return DataType.date()
         if isinstance(v, (list, tuple)):
@@ -632,6 +644,18 @@ def _from_pylist_typed(name: str, values: list, dtype: DataType) -> Series:
             elif isinstance(v, _dt.datetime):
                 out.append(int(v.timestamp() * mult) if v.tzinfo else
                            int((v - _dt.datetime(1970, 1, 1)).total_seconds() * mult))
+            else:
+                out.append(int(v))
+        return Series(name, dtype, data=torch.tensor(out, dtype=torch.int64),
+                      validity=validity)
+    if k == TypeKind.DURATION:
+        mult = {"s": 1, "ms": 10**3, "us": 10**6, "ns": 10**9}[dtype.timeunit]
+        out = []
+        for v in values:
+            if v is None:
+                out.append(0)
+            elif isinstance(v, _dt.timedelta):
+                out.append(int(v.total_seconds() * mult))
             else:
                 out.append(int(v))
         return Series(name, dtype, data=torch.tensor(out, dtype=torch.int64),

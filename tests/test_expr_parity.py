@@ -1,0 +1,188 @@
+"""Method-form Expression surface parity with the reference
+(daft/expressions/expressions.py): explode/unnest in select, map and
+partitioning namespaces, str/dt/list method delegations, bitwise ops,
+null-safe equality, try_cast, median, duration inference/totals,
+column_format='arrow' row iteration."""
+import datetime
+
+import pytest
+
+import daft_amd as daft
+import daft_amd.functions as F
+from daft_amd import col
+from daft_amd.schema import DataType
+
+
+def test_explode_in_select_and_free_fn():
+    df = daft.from_pydict({"k": ["a", "b"], "l": [[1, 2], [3]]})
+    out = df.select(col("k"), col("l").explode().alias("v")).to_pydict()
+    assert out == {"k": ["a", "a", "b"], "v": [1, 2, 3]}
+    out2 = df.select(F.explode(col("l"))).to_pydict()
+    assert out2 == {"l": [1, 2, 3]}
+
+
+def test_unnest_method():
+    df = daft.from_pydict({"s": [{"a": 1, "b": "x"}, {"a": 2, "b": "y"}]})
+    out = df.select(col("s").unnest()).to_pydict()
+    assert out == {"a": [1, 2], "b": ["x", "y"]}
+
+
+def test_map_namespace():
+    import pyarrow as pa
+    t = pa.table({"m": pa.array([[("a", 1), ("b", 2)], [("c", 3)]],
+                                pa.map_(pa.string(), pa.int64()))})
+    df = daft.from_arrow(t)
+    out = df.select(col("m").map.get("a").alias("g"),
+                    col("m").map.keys().alias("k"),
+                    col("m").map.values().alias("v")).to_pydict()
+    assert out["g"] == [1, None]
+    assert out["k"] == [["a", "b"], ["c"]]
+    assert out["v"] == [[1, 2], [3]]
+
+
+def test_partitioning_namespace():
+    ts = [datetime.datetime(2024, 3, 1), datetime.datetime(2020, 1, 15)]
+    df = daft.from_pydict({"t": ts, "x": [100, -7], "s": ["hello", "wo"]})
+    out = df.select(
+        col("t").partitioning.years().alias("y"),
+        col("t").partitioning.months().alias("m"),
+        col("t").partitioning.days().alias("d"),
+        col("x").partitioning.iceberg_truncate(10).alias("tr"),
+        col("t").partitioning.iceberg_bucket(16).alias("b"),
+    ).to_pydict()
+    # iceberg transforms are epoch-relative (ref: partition_years doc)
+    assert out["y"] == [54, 50]
+    assert out["m"] == [(2024 - 1970) * 12 + 2, (2020 - 1970) * 12 + 0]
+    assert out["d"] == [datetime.date(2024, 3, 1),
+                        datetime.date(2020, 1, 15)]
+    assert out["tr"] == [100, -10]
+    assert all(0 <= b < 16 for b in out["b"])
+
+
+def test_str_method_delegations():
+    df = daft.from_pydict({"s": ["Hello World", "a,b,,c"]})
+    out = df.select(
+        col("s").str.to_snake_case().alias("snake"),
+        col("s").str.replace("World", "X").alias("rep"),
+        col("s").str.regexp_replace(r"o+", "0").alias("rre"),
+        col("s").str.extract(r"(\w+) (\w+)", 1).alias("ex"),
+        col("s").str.extract_all(r"[A-Za-z]+").alias("exa"),
+        col("s").str.split_part(",", 2).alias("sp"),
+        col("s").str.translate("lo", "01").alias("tr"),
+        col("s").str.count_matches(["l"]).alias("cm"),
+    ).to_pydict()
+    assert out["snake"][0] == "hello_world"
+    assert out["rep"][0] == "Hello X"
+    assert out["rre"][0] == "Hell0 W0rld"
+    assert out["ex"][0] == "Hello"
+    assert out["exa"][1] == ["a", "b", "c"]
+    assert out["sp"][1] == "b"
+    assert out["tr"][0] == "He001 W1r0d"
+    assert out["cm"][0] == 3
+
+
+def test_dt_method_delegations():
+    t = datetime.datetime(2024, 3, 1, 10, 30, 45, 123456)
+    df = daft.from_pydict({"t": [t]})
+    out = df.select(
+        col("t").dt.strftime("%Y/%m/%d").alias("sf"),
+        col("t").dt.to_unix_epoch().alias("ep"),
+        col("t").dt.to_unix_epoch("ms").alias("epms"),
+        col("t").dt.millisecond().alias("ms"),
+        col("t").dt.microsecond().alias("us"),
+        col("t").dt.nanosecond().alias("ns"),
+        col("t").dt.day_of_month().alias("dom"),
+        col("t").dt.unix_date().alias("ud"),
+        col("t").dt.date_trunc("hour").alias("tr"),
+    ).to_pydict()
+    assert out["sf"] == ["2024/03/01"]
+    epoch = int((t - datetime.datetime(1970, 1, 1)).total_seconds())
+    assert out["ep"] == [epoch]
+    assert out["epms"] == [epoch * 1000 + 123]
+    assert out["ms"] == [123] and out["us"] == [123456]
+    assert out["ns"] == [0]
+    assert out["dom"] == [1]
+    assert out["ud"] == [(t.date() - datetime.date(1970, 1, 1)).days]
+    assert out["tr"] == [datetime.datetime(2024, 3, 1, 10)]
+
+
+def test_duration_inference_totals_roundtrip():
+    d = [datetime.timedelta(hours=26, minutes=5), None,
+         datetime.timedelta(milliseconds=1500)]
+    df = daft.from_pydict({"d": d})
+    assert str(df.schema[0].dtype) == "Duration(us)"
+    out = df.select(
+        col("d").dt.total_hours().alias("h"),
+        col("d").dt.total_minutes().alias("m"),
+        col("d").dt.total_milliseconds().alias("ms"),
+        col("d").dt.total_nanoseconds().alias("ns"),
+    ).to_pydict()
+    assert out["h"] == [26, None, 0]
+    assert out["m"] == [26 * 60 + 5, None, 0]
+    assert out["ms"] == [(26 * 60 + 5) * 60000, None, 1500]
+    assert out["ns"][2] == 1_500_000_000
+    assert df.to_pydict()["d"] == d
+
+
+def test_bitwise_and_shifts():
+    df = daft.from_pydict({"x": [7, 9]})
+    out = df.select(
+        col("x").bitwise_and(5).alias("a"),
+        col("x").bitwise_or(16).alias("o"),
+        col("x").bitwise_xor(3).alias("x2"),
+        col("x").shift_left(2).alias("sl"),
+        col("x").shift_right(1).alias("sr"),
+    ).to_pydict()
+    assert out["a"] == [5, 1]
+    assert out["o"] == [23, 25]
+    assert out["x2"] == [4, 10]
+    assert out["sl"] == [28, 36]
+    assert out["sr"] == [3, 4]
+
+
+def test_null_safe_eq_try_cast_median():
+    df = daft.from_pydict({"x": [1, None, 3], "y": [1, None, 4],
+                           "s": ["2024-01-05", "oops", None]})
+    out = df.select(col("x").eq_null_safe(col("y")).alias("e")).to_pydict()
+    assert out["e"] == [True, True, False]
+    d = df.select(col("s").try_cast(DataType.date()).alias("d")).to_pydict()
+    assert d["d"] == [datetime.date(2024, 1, 5), None, None]
+    m = df.agg(col("x").median().alias("m")).to_pydict()
+    assert 1.0 <= m["m"][0] <= 3.0     # sketch-based (approximate) median
+
+
+def test_float_method_forms():
+    df = daft.from_pydict({"f": [1.5, float("nan"), float("inf")]})
+    out = df.select(col("f").is_nan().alias("n"),
+                    col("f").not_nan().alias("nn"),
+                    col("f").is_inf().alias("i"),
+                    col("f").fill_nan(0.0).alias("fn")).to_pydict()
+    assert out["n"] == [False, True, False]
+    assert out["nn"] == [True, False, True]
+    assert out["i"] == [False, False, True]
+    assert out["fn"][1] == 0.0
+
+
+def test_list_method_delegations():
+    df = daft.from_pydict({"l": [[1, 2, None], [4]],
+                           "b": [[True, False], [True]]})
+    out = df.select(
+        col("l").list.append(9).alias("ap"),
+        col("l").list.count().alias("c"),
+        col("b").list.bool_and().alias("ba"),
+        col("b").list.bool_or().alias("bo"),
+    ).to_pydict()
+    assert out["ap"] == [[1, 2, None, 9], [4, 9]]
+    assert out["c"] == [2, 1]          # valid elements only
+    assert out["ba"] == [False, True]
+    assert out["bo"] == [True, True]
+
+
+def test_iter_rows_arrow_format():
+    import pyarrow as pa
+    df = daft.from_pydict({"a": [1, 2], "s": ["x", None]})
+    rows = list(df.iter_rows(column_format="arrow"))
+    assert isinstance(rows[0]["a"], pa.Scalar)
+    assert rows[0]["a"].as_py() == 1 and rows[1]["s"].as_py() is None
+    with pytest.raises(ValueError):
+        next(df.iter_rows(column_format="nope"))
