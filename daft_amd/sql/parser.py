@@ -211,6 +211,10 @@ class TableRef:
     name: Optional[str] = None           # base table
     subquery: Optional["SelectStmt"] = None
     alias: Optional[str] = None
+    values: Optional[List[List[Any]]] = None   # VALUES rows (expr ASTs)
+    col_names: Optional[List[str]] = None      # AS v(c1, c2) column list
+    fn: Optional[str] = None                   # table function name
+    fn_args: Optional[List[Any]] = None
 
 
 @dataclass
@@ -244,6 +248,7 @@ class SelectStmt:
     where: Optional[Any] = None
     group_by: List[Any] = field(default_factory=list)
     having: Optional[Any] = None
+    qualify: Optional[Any] = None
     order_by: List[OrderItem] = field(default_factory=list)
     limit: Optional[int] = None
     offset: Optional[int] = None
@@ -360,6 +365,7 @@ class Parser:
             s.items.append(self.parse_select_item())
             if not self.eat_op(","):
                 break
+        # SELECT without FROM: constant projection over one dummy row
         if self.eat_kw("from"):
             s.from_tables.append(self.parse_table_ref())
             while True:
@@ -378,12 +384,21 @@ class Parser:
             s.where = self.parse_expr()
         if self.eat_kw("group"):
             self.expect_kw("by")
-            while True:
-                s.group_by.append(self.parse_expr())
-                if not self.eat_op(","):
-                    break
+            if self.at_kw("all") or (self.peek().kind == "name" and
+                                     self.peek().value.lower() == "all"):
+                self.next()
+                s.group_by.append("__GROUP_BY_ALL__")
+            else:
+                while True:
+                    s.group_by.append(self.parse_expr())
+                    if not self.eat_op(","):
+                        break
         if self.eat_kw("having"):
             s.having = self.parse_expr()
+        if self.peek().kind == "name" and \
+                self.peek().value.lower() == "qualify":
+            self.next()
+            s.qualify = self.parse_expr()
         if self.eat_kw("order"):
             self.expect_kw("by")
             while True:
@@ -435,20 +450,72 @@ class Parser:
 
     def parse_table_ref(self) -> TableRef:
         if self.eat_op("("):
-            sub = self.parse_select()
+            if self._at_name("values"):
+                tr = self._parse_values()
+            else:
+                tr = TableRef(subquery=self.parse_select())
             self.expect_op(")")
-            alias = None
-            self.eat_kw("as")
-            if self.peek().kind == "name":
-                alias = self.next().value
-            return TableRef(subquery=sub, alias=alias)
+            self._parse_table_alias(tr)
+            return tr
+        if self._at_name("values"):
+            tr = self._parse_values()
+            self._parse_table_alias(tr)
+            return tr
         name = self.next().value
-        alias = None
+        if self.at_op("("):
+            # table function: read_parquet('path'), read_csv(...), ...
+            self.next()
+            args = []
+            if not self.at_op(")"):
+                while True:
+                    args.append(self.parse_expr())
+                    if not self.eat_op(","):
+                        break
+            self.expect_op(")")
+            tr = TableRef(fn=name, fn_args=args)
+            self._parse_table_alias(tr)
+            return tr
+        tr = TableRef(name=name)
+        self._parse_table_alias(tr)
+        return tr
+
+    def _at_name(self, word: str) -> bool:
+        t = self.peek()
+        return (t.kind == "name" and t.value.lower() == word) or \
+            (t.kind == "kw" and t.value == word)
+
+    def _parse_values(self) -> TableRef:
+        self.next()                      # VALUES
+        rows = []
+        while True:
+            self.expect_op("(")
+            row = []
+            while True:
+                row.append(self.parse_expr())
+                if not self.eat_op(","):
+                    break
+            self.expect_op(")")
+            rows.append(row)
+            if not self.eat_op(","):
+                break
+        return TableRef(values=rows)
+
+    def _parse_table_alias(self, tr: TableRef):
         if self.eat_kw("as"):
-            alias = self.next().value
-        elif self.peek().kind == "name":
-            alias = self.next().value
-        return TableRef(name=name, alias=alias)
+            tr.alias = self.next().value
+        elif self.peek().kind == "name" and \
+                self.peek().value.lower() not in ("values", "qualify"):
+            tr.alias = self.next().value
+        else:
+            return
+        if self.eat_op("("):
+            names = []
+            while True:
+                names.append(self.next().value)
+                if not self.eat_op(","):
+                    break
+            self.expect_op(")")
+            tr.col_names = names
 
     # -- expressions (precedence climbing) --------------------------------
     def parse_expr(self) -> Any:

@@ -393,6 +393,48 @@ def _col_refs(e, out: List[P.Col]):
 # FROM planning with equi-join extraction
 # ---------------------------------------------------------------------------
 
+def _plan_values(t: P.TableRef):
+    """(VALUES (..), (..)) AS v(c1, c2): evaluate each row's constant
+    expressions over a one-row dummy frame (ref: daft-sql VALUES)."""
+    import daft_amd as daft
+    dummy = daft.from_pydict({"__one__": [0]})
+    binder = Binder({"__v__": {}})
+    ncols = len(t.values[0])
+    cols = {}
+    for ci in range(ncols):
+        vals = []
+        for row in t.values:
+            if len(row) != ncols:
+                raise SQLPlanError("VALUES rows have differing arity")
+            e = expr_to_daft(row[ci], binder)
+            vals.append(dummy.select(e.alias("x")).to_pydict()["x"][0])
+        cols[f"column{ci + 1}"] = vals
+    return daft.from_pydict(cols)
+
+
+def _plan_table_fn(t: P.TableRef):
+    """FROM-position table functions: read_parquet/read_csv/read_json/
+    read_ipc('path', ...) (ref: daft-sql table providers)."""
+    import daft_amd as daft
+    fns = {"read_parquet": daft.read_parquet, "read_csv": daft.read_csv,
+           "read_json": daft.read_json,
+           "read_ipc": getattr(daft, "read_ipc", None)}
+    fn = fns.get(t.fn)
+    if fn is None:
+        raise SQLPlanError(f"unknown table function {t.fn!r}")
+    dummy_binder = Binder({"__v__": {}})
+    args = []
+    for a in t.fn_args or []:
+        if isinstance(a, P.Lit):
+            args.append(a.value)
+        else:
+            import daft_amd as daft
+            d = daft.from_pydict({"__one__": [0]})
+            e = expr_to_daft(a, dummy_binder)
+            args.append(d.select(e.alias("x")).to_pydict()["x"][0])
+    return fn(*args)
+
+
 class _FromPlanner:
     def __init__(self, lookup, outer: Optional[Binder]):
         self.lookup = lookup
@@ -406,10 +448,21 @@ class _FromPlanner:
         if t.subquery is not None:
             df = _plan_select(t.subquery, self.lookup, outer=self.outer,
                               ctes=ctes)
+        elif t.values is not None:
+            df = _plan_values(t)
+        elif t.fn is not None:
+            df = _plan_table_fn(t)
         elif t.name in ctes:
             df = ctes[t.name]
         else:
             df = self.lookup(t.name)
+        if t.col_names:
+            old = df.column_names()
+            if len(t.col_names) != len(old):
+                raise SQLPlanError(
+                    f"alias column list has {len(t.col_names)} names for "
+                    f"{len(old)} columns")
+            df = df.with_columns_renamed(dict(zip(old, t.col_names)))
         mapping = {}
         renames = {}
         for c in df.column_names():
@@ -567,8 +620,12 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
 
     fp = _FromPlanner(lookup, outer)
     if not stmt.from_tables:
-        raise SQLPlanError("SELECT without FROM is not supported")
-    fp.add_first(stmt.from_tables[0], ctes)
+        # constant projection: one dummy row (ref: daft-sql SELECT 1+1)
+        import daft_amd as daft
+        fp.df = daft.from_pydict({"__one__": [0]})
+        fp.binder.tables["__dual__"] = {}
+    else:
+        fp.add_first(stmt.from_tables[0], ctes)
 
     where_conjs = _split_conj(stmt.where) if stmt.where is not None else []
     where_conjs = _hoist_common_from_or(where_conjs)
@@ -616,6 +673,13 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
     for cj in subq:
         df = _apply_subquery_conjunct(df, cj, binder, lookup, ctes)
 
+    # GROUP BY ALL: every non-aggregate select item is a group key
+    if any(g == "__GROUP_BY_ALL__" for g in stmt.group_by):
+        import dataclasses
+        gb = [it.expr for it in stmt.items
+              if not it.star and not _has_aggregate(it.expr)]
+        stmt = dataclasses.replace(stmt, group_by=gb)
+
     # aggregate or plain projection
     has_agg = any(_has_aggregate(it.expr) for it in stmt.items
                   if not it.star) or stmt.group_by or \
@@ -655,6 +719,42 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
                 exprs.append(e.alias(name))
         df = df.select(*exprs)
         out_names = df.column_names()
+
+    if stmt.qualify is not None:
+        # QUALIFY filters on window results after projection; inline
+        # window exprs compute as hidden columns first
+        q = stmt.qualify
+        hidden = []
+
+        def _rewrite(e):
+            if isinstance(e, P.WindowExpr):
+                hname = f"__q{len(hidden)}"
+                hidden.append((hname, e))
+                return P.Col(None, hname)
+            import dataclasses as _dc
+            if _dc.is_dataclass(e):
+                kw = {}
+                for f in _dc.fields(e):
+                    v = getattr(e, f.name)
+                    if isinstance(v, list):
+                        kw[f.name] = [_rewrite(x) if _dc.is_dataclass(x)
+                                      or isinstance(x, P.WindowExpr)
+                                      else x for x in v]
+                    elif _dc.is_dataclass(v) or isinstance(v, P.WindowExpr):
+                        kw[f.name] = _rewrite(v)
+                    else:
+                        kw[f.name] = v
+                return type(e)(**kw)
+            return e
+
+        q = _rewrite(q)
+        for hname, we in hidden:
+            df = df.with_window_columns(
+                {hname: _window_to_daft(we, binder)})
+        pred = _bind_order_expr(q, out_names, select_aliases, binder, stmt)
+        df = df.where(pred)
+        if hidden:
+            df = df.select(*[col(n) for n in out_names])
 
     if stmt.distinct:
         df = df.distinct()

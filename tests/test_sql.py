@@ -188,3 +188,48 @@ def test_sql_catalog_and_explain():
     plan = daft.sql("explain select a from t where a > 1",
                     catalog=cat).to_pydict()["plan"]
     assert any("Filter" in l for l in plan)
+
+
+def test_sql_values_clause():
+    out = sql("SELECT x, y FROM (VALUES (1, 'a'), (2, 'b')) AS v(x, y) "
+              "WHERE x > 1").to_pydict()
+    assert out == {"x": [2], "y": ["b"]}
+    out2 = sql("SELECT * FROM (VALUES (1), (2)) v").to_pydict()
+    assert out2 == {"column1": [1, 2]}
+
+
+def test_sql_select_without_from():
+    out = sql("SELECT 1+1 AS x, 'hi' AS s").to_pydict()
+    assert out == {"x": [2], "s": ["hi"]}
+
+
+def test_sql_group_by_all():
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1, 2, 3]})
+    out = sql("SELECT g, sum(v) AS s FROM df GROUP BY ALL ORDER BY g") \
+        .to_pydict()
+    assert out == {"g": ["a", "b"], "s": [3, 3]}
+
+
+def test_sql_qualify():
+    df = daft.from_pydict({"g": ["a", "a", "b"], "v": [1, 2, 3]})
+    out = sql("SELECT g, v, row_number() OVER (PARTITION BY g ORDER BY v "
+              "DESC) AS rn FROM df QUALIFY rn = 1 ORDER BY g").to_pydict()
+    assert out == {"g": ["a", "b"], "v": [2, 3], "rn": [1, 1]}
+    out2 = sql("SELECT g, v FROM df QUALIFY row_number() OVER "
+               "(ORDER BY v DESC) = 1").to_pydict()
+    assert out2 == {"g": ["b"], "v": [3]}
+
+
+def test_sql_table_function_read_parquet(tmp_path):
+    df = daft.from_pydict({"a": [1, 2, 3]})
+    df.write_parquet(str(tmp_path / "t"))
+    out = sql(f"SELECT sum(a) AS s FROM "
+              f"read_parquet('{tmp_path}/t/**/*.parquet')").to_pydict()
+    assert out == {"s": [6]}
+
+
+def test_sql_subquery_column_alias_list():
+    df = daft.from_pydict({"a": [1, 2, 3]})
+    out = sql("SELECT x * 10 AS y FROM (SELECT a FROM df) AS q(x) "
+              "WHERE x >= 2 ORDER BY y").to_pydict()
+    assert out == {"y": [20, 30]}
