@@ -26,9 +26,11 @@ def optimize(plan: lp.LogicalPlan) -> lp.LogicalPlan:
     once = _apply_bottom_up(plan, derive_or_implications_rule)
     if once is not None:
         plan = once
+    from .truth_filter import fold_filter_by_stats
     batches: List[List[Rule]] = [
         [simplify_expressions],
         [push_down_filter, drop_repartition] +
+        ([] if "statsfold" in disabled else [fold_filter_by_stats]) +
         ([] if "semipush" in disabled else [push_down_anti_semi_join]),
         [] if "aggpush" in disabled else [push_down_aggregation,
                                           _push_semi_into_agg],

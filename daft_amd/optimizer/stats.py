@@ -107,3 +107,70 @@ def ndv_for_source(cache_key: str, column: str,
     nd = sample_ndv(s, n_rows=total)
     NDV_HINTS[(cache_key, column)] = nd
     return min(nd, est_rows) if est_rows is not None else nd
+
+
+# ---------------------------------------------------------------------------
+# exact per-source column ranges for TruthValue filter folding (ref:
+# daft-stats/src/column_stats/ ColumnRangeStatistics + TruthValue:
+# a predicate over a column whose [min, max] is known evaluates to
+# definitely-True / definitely-False / Maybe)
+# ---------------------------------------------------------------------------
+
+# (cache_key, column) -> (min, max, has_nulls) — EXACT bounds (not
+# sampled: pruning on approximate bounds would be a correctness bug)
+RANGE_HINTS: Dict[Tuple[str, str], tuple] = {}
+
+
+def range_for_source(cache_key: str, column: str):
+    """Exact (min, max, has_nulls) of a cached source column, or None.
+    Lazily computed on single-rank runs only — under SPMD rank-local
+    ranges could fold a filter on one rank and not another, desyncing
+    the collective schedule (same invariant as ndv_for_source)."""
+    hint = RANGE_HINTS.get((cache_key, column))
+    if hint is not None:
+        return hint
+    try:
+        from ..distributed import comm
+        if comm.is_dist() and comm.world() > 1:
+            return None
+    except Exception:
+        pass
+    try:
+        from ..context import get_context
+        parts = get_context().cache.get(cache_key)
+    except Exception:
+        return None
+    if not parts:
+        return None
+    lo = hi = None
+    has_nulls = False
+    for p in parts:
+        try:
+            s = p.column(column)
+        except Exception:
+            return None
+        if s.pyobjs is not None or s.data is None or s.offsets is not None \
+                or s.children or len(s) == 0:
+            return None          # numeric/temporal fixed-width only
+        if not (s.dtype.is_numeric() or s.dtype.is_temporal() or
+                s.dtype.kind.value in ("date", "bool")):
+            return None
+        d = s.data
+        if s.validity is not None:
+            if bool((~s.validity).any()):
+                has_nulls = True
+            valid = s.validity
+            if not bool(valid.any()):
+                continue
+            d = d[valid]
+        if d.dtype == torch.uint64:
+            d = d.view(torch.int64)
+        pmin = d.min().item()
+        pmax = d.max().item()
+        lo = pmin if lo is None else min(lo, pmin)
+        hi = pmax if hi is None else max(hi, pmax)
+    if lo is None:
+        return None
+    out = (lo, hi, has_nulls)
+    RANGE_HINTS[(cache_key, column)] = out
+    return out

@@ -302,3 +302,47 @@ def test_dp_best_unit():
             for ch in node.children:
                 check(ch)
     check(tree)
+
+
+def test_truthvalue_filter_folding():
+    """Plan-level TruthValue: a filter provably TRUE over the source's
+    exact column range is dropped; a provably FALSE one becomes Limit 0
+    (ref: daft-stats ColumnRangeStatistics + TruthValue)."""
+    df = daft.from_pydict({"x": list(range(10, 110)),
+                           "y": [float(i) for i in range(100)]})
+    df.collect()                 # cache partitions so ranges are exact
+    # always true: x between 10 and 109 inclusive
+    t = df.where((col("x") >= 10) & (col("x") < 200))
+    pt = _optimized_plan(t)
+    assert not _find(pt, lp.Filter), "provably-true filter should fold"
+    assert t.to_pydict()["x"] == list(range(10, 110))
+    # always false
+    f = df.where(col("x") > 1000)
+    pf = _optimized_plan(f)
+    assert not _find(pf, lp.Filter)
+    assert f.count_rows() == 0
+    # maybe: stays a filter, result correct
+    m = df.where(col("x") > 50)
+    assert _find(_optimized_plan(m), lp.Filter)
+    assert m.count_rows() == 59
+
+
+def test_truthvalue_respects_nulls():
+    """A definitely-true comparison must NOT fold when the column has
+    nulls: the filter also drops null rows."""
+    df = daft.from_pydict({"x": [1, 2, None, 4]})
+    df.collect()
+    q = df.where(col("x") >= 0)
+    assert q.count_rows() == 3        # null row filtered out
+    plan = _optimized_plan(q)
+    assert _find(plan, lp.Filter), "nullable column: filter must remain"
+
+
+def test_truthvalue_not_and_between():
+    df = daft.from_pydict({"x": [5, 6, 7]})
+    df.collect()
+    assert df.where(~(col("x") > 100)).count_rows() == 3
+    p = _optimized_plan(df.where(~(col("x") > 100)))
+    assert not _find(p, lp.Filter)
+    assert df.where(col("x").between(0, 100)).count_rows() == 3
+    assert df.where(col("x").between(8, 9)).count_rows() == 0
