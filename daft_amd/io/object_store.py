@@ -748,6 +748,112 @@ class AzureBlobSource(ObjectSource):
         return out
 
 
+class HuggingFaceSource(ObjectSource):
+    """hf://{datasets|models|spaces}/owner/repo[@revision]/path over the
+    Hub's resolve + tree APIs (ref: daft-io/src/huggingface/{mod,path}.rs).
+    Auth: Bearer token from HFConfig-style env HF_TOKEN; endpoint
+    overridable via HF_ENDPOINT (mock servers / mirrors).  Read-only."""
+
+    def __init__(self, config: Optional[HTTPConfig] = None):
+        self.cfg = config or HTTPConfig()
+        import requests
+        self._sess = requests.Session()
+        self.endpoint = (os.environ.get("HF_ENDPOINT") or
+                         "https://huggingface.co").rstrip("/")
+        self.token = os.environ.get("HF_TOKEN")
+
+    def _split(self, path: str):
+        parsed = urllib.parse.urlsplit(path)
+        parts = (parsed.netloc + parsed.path).split("/")
+        if len(parts) < 3:
+            raise ValueError(f"hf:// path needs repo_type/owner/repo: "
+                             f"{path}")
+        repo_type, owner, repo = parts[0], parts[1], parts[2]
+        rev = "main"
+        if "@" in repo:
+            repo, rev = repo.split("@", 1)
+        return repo_type, f"{owner}/{repo}", rev, "/".join(parts[3:])
+
+    def _resolve_url(self, path: str) -> str:
+        repo_type, repo, rev, file = self._split(path)
+        prefix = "" if repo_type == "models" else f"{repo_type}/"
+        return (f"{self.endpoint}/{prefix}{repo}/resolve/"
+                f"{urllib.parse.quote(rev)}/{file}")
+
+    def _headers(self) -> dict:
+        return {"Authorization": f"Bearer {self.token}"} if self.token \
+            else {}
+
+    def get(self, path, range_=None):
+        url = self._resolve_url(path)
+        def go():
+            headers = self._headers()
+            if range_ is not None:
+                headers["Range"] = f"bytes={range_[0]}-{range_[1] - 1}"
+            r = self._sess.get(url, headers=headers,
+                               timeout=self.cfg.read_timeout_ms / 1000)
+            if r.status_code == 404:
+                raise NotFoundError(path)
+            if r.status_code >= 400:
+                raise ObjectStoreError(f"HF {r.status_code} for {path}")
+            return r.content
+        return _with_retry(go, self.cfg.num_tries,
+                           self.cfg.retry_initial_backoff_ms, f"GET {path}")
+
+    def get_size(self, path):
+        url = self._resolve_url(path)
+        def go():
+            r = self._sess.head(url, headers=self._headers(),
+                                allow_redirects=True,
+                                timeout=self.cfg.read_timeout_ms / 1000)
+            if r.status_code == 404:
+                raise NotFoundError(path)
+            n = r.headers.get("X-Linked-Size") or \
+                r.headers.get("Content-Length")
+            if r.status_code >= 400 or n is None:
+                raise ObjectStoreError(f"HF HEAD {r.status_code} {path}")
+            return int(n)
+        return _with_retry(go, self.cfg.num_tries,
+                           self.cfg.retry_initial_backoff_ms,
+                           f"HEAD {path}")
+
+    def put(self, path, data):
+        raise ObjectStoreError("hf:// is read-only here (upload via the "
+                               "huggingface_hub client)")
+
+    def list_prefix(self, path_prefix):
+        import json
+        repo_type, repo, rev, prefix = self._split(path_prefix)
+        api_type = repo_type if repo_type != "models" else "models"
+        base = (f"{self.endpoint}/api/{api_type}/{repo}/tree/"
+                f"{urllib.parse.quote(rev)}")
+        out = []
+
+        def walk(sub: str):
+            url = base + (f"/{sub}" if sub else "") + "?recursive=true"
+            def go():
+                r = self._sess.get(url, headers=self._headers(),
+                                   timeout=self.cfg.read_timeout_ms / 1000)
+                if r.status_code == 404:
+                    raise NotFoundError(path_prefix)
+                if r.status_code >= 400:
+                    raise ObjectStoreError(f"HF tree {r.status_code}")
+                return r
+            r = _with_retry(go, self.cfg.num_tries,
+                            self.cfg.retry_initial_backoff_ms,
+                            f"LIST {path_prefix}")
+            for item in json.loads(r.text):
+                if item.get("type") == "file":
+                    p = item["path"]
+                    at = "" if rev == "main" else f"@{rev}"
+                    out.append((f"hf://{repo_type}/{repo}{at}/{p}",
+                                int(item.get("size", 0))))
+
+        # list from the deepest directory of the static prefix
+        walk(prefix.rsplit("/", 1)[0] if "/" in prefix else "")
+        return [(p, sz) for (p, sz) in out]
+
+
 # ---------------------------------------------------------------------------
 # dispatch
 # ---------------------------------------------------------------------------
@@ -770,6 +876,8 @@ def get_source(path: str,
         return GCSSource(cfg.gcs)
     if scheme in ("az", "abfs", "abfss", "wasb", "wasbs"):
         return AzureBlobSource(cfg.azure)
+    if scheme == "hf":
+        return HuggingFaceSource(cfg.http)
     if scheme in ("http", "https"):
         return HTTPSource(cfg.http)
     if scheme in ("", "file"):
@@ -778,7 +886,7 @@ def get_source(path: str,
 
 
 _REMOTE_SCHEMES = ("s3", "s3a", "gs", "gcs", "az", "abfs", "abfss",
-                   "wasb", "wasbs", "http", "https")
+                   "wasb", "wasbs", "hf", "http", "https")
 
 
 def is_remote(path: str) -> bool:

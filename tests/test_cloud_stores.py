@@ -158,3 +158,93 @@ def test_scheme_dispatch():
     for p in ("gs://b/k", "az://c/k", "abfss://c@a.dfs.core.windows.net/k"):
         assert is_remote(p)
     assert not is_remote("/tmp/x")
+
+
+# ---------------------------------------------------------------- hf://
+
+class _MockHub:
+    """Tiny in-process Hugging Face Hub: resolve + tree API."""
+
+    def __init__(self):
+        import json
+        import threading
+        from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+        self.files = {"data/a.parquet": b"PARQA", "data/b.txt": b"hello",
+                      "readme.md": b"# hi"}
+        srv = self
+
+        class H(BaseHTTPRequestHandler):
+            def log_message(self, *a):
+                pass
+
+            def do_GET(self):
+                import urllib.parse as up
+                path = up.urlsplit(self.path).path
+                if path.startswith("/api/datasets/org/repo/tree/main"):
+                    body = json.dumps([
+                        {"type": "file", "path": p, "size": len(b)}
+                        for p, b in srv.files.items()]).encode()
+                    self.send_response(200)
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                    return
+                pre = "/datasets/org/repo/resolve/main/"
+                if path.startswith(pre) and path[len(pre):] in srv.files:
+                    data = srv.files[path[len(pre):]]
+                    rng = self.headers.get("Range")
+                    code = 200
+                    if rng:
+                        import re as _re
+                        m = _re.match(r"bytes=(\d+)-(\d+)", rng)
+                        data = data[int(m.group(1)):int(m.group(2)) + 1]
+                        code = 206
+                    self.send_response(code)
+                    self.send_header("Content-Length", str(len(data)))
+                    self.end_headers()
+                    self.wfile.write(data)
+                    return
+                self.send_response(404)
+                self.end_headers()
+
+            def do_HEAD(self):
+                import urllib.parse as up
+                path = up.urlsplit(self.path).path
+                pre = "/datasets/org/repo/resolve/main/"
+                if path.startswith(pre) and path[len(pre):] in srv.files:
+                    self.send_response(200)
+                    self.send_header(
+                        "Content-Length",
+                        str(len(srv.files[path[len(pre):]])))
+                    self.end_headers()
+                    return
+                self.send_response(404)
+                self.end_headers()
+
+        self._server = ThreadingHTTPServer(("127.0.0.1", 0), H)
+        self.endpoint = f"http://127.0.0.1:{self._server.server_port}"
+        import threading
+        threading.Thread(target=self._server.serve_forever,
+                         daemon=True).start()
+
+    def close(self):
+        self._server.shutdown()
+
+
+def test_hf_source(monkeypatch):
+    from daft_amd.io.object_store import HuggingFaceSource, NotFoundError
+    hub = _MockHub()
+    try:
+        monkeypatch.setenv("HF_ENDPOINT", hub.endpoint)
+        src = HuggingFaceSource()
+        assert src.get("hf://datasets/org/repo/data/b.txt") == b"hello"
+        assert src.get("hf://datasets/org/repo/data/b.txt",
+                       range_=(1, 4)) == b"ell"
+        assert src.get_size("hf://datasets/org/repo/data/a.parquet") == 5
+        got = src.glob("hf://datasets/org/repo/data/*.parquet")
+        assert got == ["hf://datasets/org/repo/data/a.parquet"]
+        with pytest.raises(NotFoundError):
+            src.get("hf://datasets/org/repo/nope")
+        assert is_remote("hf://datasets/org/repo/x")
+    finally:
+        hub.close()
