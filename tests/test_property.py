@@ -200,3 +200,77 @@ def test_composed_multikey_argsort_matches_fallback():
             rowops._try_composed_argsort = orig
         assert comp is not None, "composition must engage here"
         assert torch.equal(comp, perm), (desc, nf)
+
+
+_i128 = st.integers(min_value=-(10 ** 37) + 1, max_value=10 ** 37 - 1)
+
+
+@given(st.lists(_i128, min_size=1, max_size=60),
+       st.lists(_i128, min_size=1, max_size=60))
+@settings(max_examples=60, deadline=None)
+def test_wide_decimal_limb_arithmetic_matches_python(a, b):
+    """add/sub/compare over random i128-scale magnitudes: the two-limb
+    carry arithmetic must agree with Python's arbitrary-precision ints
+    (kernels/decimal128.py)."""
+    import torch
+    from daft_amd.kernels import decimal128 as d128
+    n = min(len(a), len(b))
+    a, b = a[:n], b[:n]
+    alo, ahi = d128.tensors_from_ints(a)
+    blo, bhi = d128.tensors_from_ints(b)
+    MOD = 1 << 128
+
+    def wrap(v):
+        v %= MOD
+        return v - MOD if v >= (1 << 127) else v
+
+    slo, shi = d128.add128(alo, ahi, blo, bhi)
+    assert d128.ints_from_tensors(slo, shi) == \
+        [wrap(x + y) for x, y in zip(a, b)]
+    dlo, dhi = d128.sub128(alo, ahi, blo, bhi)
+    assert d128.ints_from_tensors(dlo, dhi) == \
+        [wrap(x - y) for x, y in zip(a, b)]
+    mlo, mhi = d128.mul128(alo, ahi, blo, bhi)
+    assert d128.ints_from_tensors(mlo, mhi) == \
+        [wrap(x * y) for x, y in zip(a, b)]
+    for op, fn in (("lt", lambda x, y: x < y), ("le", lambda x, y: x <= y),
+                   ("gt", lambda x, y: x > y), ("eq", lambda x, y: x == y)):
+        got = d128.cmp128(alo, ahi, blo, bhi, op).tolist()
+        assert got == [fn(x, y) for x, y in zip(a, b)], op
+
+
+@given(st.lists(_i128, min_size=1, max_size=50),
+       st.integers(min_value=1, max_value=30))
+@settings(max_examples=40, deadline=None)
+def test_wide_decimal_pow10_roundtrip(vals, k):
+    """x * 10^k followed by truncating / 10^k returns x whenever the
+    product stays inside 128 bits."""
+    from daft_amd.kernels import decimal128 as d128
+    keep = [v for v in vals if abs(v) * 10 ** k < (1 << 126)]
+    if not keep:
+        return
+    lo, hi = d128.tensors_from_ints(keep)
+    mlo, mhi = d128.mul128_pow10(lo, hi, k)
+    assert d128.ints_from_tensors(mlo, mhi) == \
+        [v * 10 ** k for v in keep]
+    dlo, dhi = d128.divround128_pow10(mlo, mhi, k, round_half=False)
+    assert d128.ints_from_tensors(dlo, dhi) == keep
+
+
+@given(st.lists(st.tuples(st.integers(0, 4), _i128), min_size=1,
+                max_size=200))
+@settings(max_examples=40, deadline=None)
+def test_wide_decimal_grouped_sum_property(rows):
+    """Grouped exact SUM over random wide values == Python int sums."""
+    import decimal as pydec
+    pydec.getcontext().prec = 80
+    keys = [k for k, _ in rows]
+    vals = [pydec.Decimal(v).scaleb(-10) for _, v in rows]
+    df = daft.from_pydict({"k": keys, "v": vals})
+    out = df.groupby("k").agg(col("v").sum().alias("s")).sort("k") \
+        .to_pydict()
+    import collections
+    want = collections.defaultdict(pydec.Decimal)
+    for k, v in zip(keys, vals):
+        want[k] += v
+    assert out["s"] == [want[k] for k in out["k"]]
