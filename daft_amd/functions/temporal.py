@@ -89,8 +89,27 @@ def dateadd(unit, n, x):
 def date_sub(x, days): return _e(x) - days
 
 
-def date_diff(end, start):
-    """Whole days between two dates."""
+def date_diff(end, start, start2=None):
+    """Whole days between two dates; the SQL 3-arg form
+    DATEDIFF(unit, start, end) is also accepted (unit day/month/year)."""
+    if start2 is not None:
+        unit, a, b = end, start2, start     # DATEDIFF('d', start, end)
+        if isinstance(unit, Expression):    # SQL binds literals as exprs
+            from ..expressions.expressions import Literal as _L
+            node = unit._node
+            if isinstance(node, _L):
+                unit = node.value
+        unit = str(unit).rstrip("s").lower()
+        if unit in ("day", "d"):
+            return date_diff(a, b)
+        if unit == "month":
+            ey, em = _e(a).dt.year(), _e(a).dt.month()
+            sy, sm = _e(b).dt.year(), _e(b).dt.month()
+            return (ey - sy) * 12 + (em - sm)
+        if unit == "year":
+            return _e(a).dt.year() - _e(b).dt.year()
+        raise ValueError(f"datediff unit {unit!r}")
+
     def run(a: Series, b: Series) -> Series:
         out = a.data.to(torch.int64) - b.data.to(torch.int64)
         v = a.validity

@@ -140,6 +140,16 @@ def expr_to_daft(e, binder: Optional[Binder]) -> Expression:
         raise SQLPlanError("month/year intervals only combine with date "
                            "literals")
     if isinstance(e, P.BinOp):
+        # date column ± INTERVAL 'n month/year' (literal dates folded
+        # above; columns need calendar arithmetic)
+        if e.op in ("add", "sub") and isinstance(e.right, P.IntervalLit) \
+                and e.right.unit in ("month", "year"):
+            from ..functions import add_months
+            iv = e.right
+            months = iv.n * (12 if iv.unit == "year" else 1)
+            if e.op == "sub":
+                months = -months
+            return add_months(expr_to_daft(e.left, binder), months)
         l = expr_to_daft(e.left, binder)
         r = expr_to_daft(e.right, binder)
         ops = {"add": l.__add__, "sub": l.__sub__, "mul": l.__mul__,

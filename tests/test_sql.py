@@ -233,3 +233,25 @@ def test_sql_subquery_column_alias_list():
     out = sql("SELECT x * 10 AS y FROM (SELECT a FROM df) AS q(x) "
               "WHERE x >= 2 ORDER BY y").to_pydict()
     assert out == {"y": [20, 30]}
+
+
+def test_sql_interval_on_date_columns():
+    t = daft.from_pydict({"d": ["2024-01-31", "2024-02-01"]})
+    out = sql("SELECT CAST(d AS DATE) + INTERVAL '1 month' AS m, "
+              "CAST(d AS DATE) - INTERVAL '1 year' AS y, "
+              "CAST(d AS DATE) + INTERVAL '3 day' AS dd FROM t").to_pydict()
+    import datetime as dt
+    assert out["m"] == [dt.date(2024, 2, 29), dt.date(2024, 3, 1)]
+    assert out["y"] == [dt.date(2023, 1, 31), dt.date(2023, 2, 1)]
+    assert out["dd"] == [dt.date(2024, 2, 3), dt.date(2024, 2, 4)]
+
+
+def test_sql_datediff_units():
+    t = daft.from_pydict({"d": ["2024-01-31"]})
+    out = sql("SELECT DATEDIFF('day', CAST(d AS DATE), "
+              "CAST('2024-03-01' AS DATE)) AS dd, "
+              "DATEDIFF('month', CAST(d AS DATE), "
+              "CAST('2024-06-15' AS DATE)) AS dm, "
+              "DATEDIFF('year', CAST(d AS DATE), "
+              "CAST('2026-01-01' AS DATE)) AS dy FROM t").to_pydict()
+    assert out == {"dd": [30], "dm": [5], "dy": [2]}
