@@ -197,3 +197,8 @@ from .aliases import (  # noqa: E402,F401
 from .aliases import slice  # noqa: E402,F401,A004
 from .ai import (  # noqa: E402,F401
     embed_text, embed_image, classify_text, classify_image, prompt)
+from .aliases import (  # noqa: F401  method-form free functions
+    abs, any_value, approx_count_distinct, avg, between, bool_and,
+    bool_or, cast, ceil, clip, count, count_distinct, fill_null, floor,
+    hash, is_in, is_null, lag, lead, max, mean, min, minhash, not_null,
+    over, round, simhash, skew, stddev, sum)

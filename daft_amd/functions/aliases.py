@@ -447,3 +447,39 @@ extract_minute_uuid7 = _uuid7_part("extract_minute_uuid7",
 def resample(x, every: str):
     """Truncate timestamps to a resampling interval (group key helper)."""
     return _e(x).dt.truncate(every)
+
+
+# -- free-function forms of Expression methods (ref: daft/functions
+# exports the method surface as functions too) -------------------------------
+
+def abs(x): return _e(x).abs()                     # noqa: A001
+def any_value(x): return _e(x).any_value()
+def approx_count_distinct(x): return _e(x).approx_count_distinct()
+def avg(x): return _e(x).avg()
+def between(x, lo, hi): return _e(x).between(lo, hi)
+def bool_and(x): return _e(x).bool_and()
+def bool_or(x): return _e(x).bool_or()
+def cast(x, dtype): return _e(x).cast(dtype)
+def ceil(x): return _e(x).ceil()
+def clip(x, lo=None, hi=None): return _e(x).clip(lo, hi)
+def count(x): return _e(x).count()
+def count_distinct(x): return _e(x).count_distinct()
+def fill_null(x, v): return _e(x).fill_null(v)
+def floor(x): return _e(x).floor()
+def hash(x, seed: int = 0): return _e(x).hash(seed)        # noqa: A001
+def is_in(x, values): return _e(x).is_in(values)
+def is_null(x): return _e(x).is_null()
+def lag(x, n: int = 1, default=None): return _e(x).lag(n, default)
+def lead(x, n: int = 1, default=None): return _e(x).lead(n, default)
+def max(x): return _e(x).max()                     # noqa: A001
+def mean(x): return _e(x).mean()
+def min(x): return _e(x).min()                     # noqa: A001
+def minhash(x, num_hashes: int, ngram_size: int = 1, seed: int = 1):
+    return _e(x).minhash(num_hashes, ngram_size, seed)
+def not_null(x): return _e(x).not_null()
+def over(x, window): return _e(x).over(window)
+def round(x, decimals: int = 0): return _e(x).round(decimals)  # noqa: A001
+def simhash(x, ngram_size: int = 4): return _e(x).simhash(ngram_size)
+def skew(x): return _e(x).skew()
+def stddev(x): return _e(x).stddev()
+def sum(x): return _e(x).sum()                     # noqa: A001
