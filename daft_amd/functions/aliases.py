@@ -462,7 +462,12 @@ def bool_or(x): return _e(x).bool_or()
 def cast(x, dtype): return _e(x).cast(dtype)
 def ceil(x): return _e(x).ceil()
 def clip(x, lo=None, hi=None): return _e(x).clip(lo, hi)
-def count(x): return _e(x).count()
+def count(x=None):
+    """count(expr) counts non-null rows of expr; count() is COUNT(*)."""
+    if x is None:
+        from ..expressions.expressions import Agg, AggKind, Expression
+        return Expression(Agg(AggKind.COUNT_ALL, None))
+    return _e(x).count()
 def count_distinct(x): return _e(x).count_distinct()
 def fill_null(x, v): return _e(x).fill_null(v)
 def floor(x): return _e(x).floor()

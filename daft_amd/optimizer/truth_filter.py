@@ -131,7 +131,9 @@ def truth_value(e, ranges: Dict[str, tuple]
             return None, False
         has_nulls = cr[2]
         if not has_nulls:
-            return False, True
+            # is_null -> definitely False; not_null (negate) -> True
+            return (True, True) if getattr(e, "negate", False) \
+                else (False, True)
         return None, False
     return None, False
 

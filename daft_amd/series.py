@@ -568,9 +568,29 @@ def _infer_dtype(values: Sequence[Any]) -> DataType:
         if isinstance(v, float):
             return DataType.float64()
         if type(v).__name__ == "Decimal":
-            t = v.as_tuple()
-            scl = max(0, -t.exponent) if isinstance(t.exponent, int) else 0
-            digs = max(len(t.digits), scl + 1)
+            # precision/scale must cover EVERY value, not just the first
+            # (a leading small value would infer p<=18 and overflow the
+            # scaled-int64 storage on a later wide one)
+            scl = 0
+            digs = 1
+            for w in values:
+                if w is None or type(w).__name__ != "Decimal":
+                    continue
+                t = w.as_tuple()
+                s_ = max(0, -t.exponent) if isinstance(t.exponent, int) \
+                    else 0
+                scl = max(scl, s_)
+                digs = max(digs, len(t.digits) + max(0, s_ - (-t.exponent
+                           if isinstance(t.exponent, int) else 0)))
+            # integer digits + final scale
+            idigs = 0
+            for w in values:
+                if w is None or type(w).__name__ != "Decimal":
+                    continue
+                t = w.as_tuple()
+                exp = t.exponent if isinstance(t.exponent, int) else 0
+                idigs = max(idigs, len(t.digits) + exp)
+            digs = max(digs, idigs + scl, scl + 1)
             return DataType.decimal128(min(digs, 38), scl)
         if isinstance(v, str):
             return DataType.string()
