@@ -702,17 +702,120 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
 
     # window functions: compute as extra columns before projection
     win_map = {}
-    for i, it in enumerate(stmt.items):
-        if isinstance(it.expr, P.WindowExpr):
-            if has_agg:
-                raise SQLPlanError("window functions with GROUP BY are "
-                                   "not supported yet")
-            wname = it.alias or it.expr.func.name
-            df = df.with_window_columns(
-                {wname: _window_to_daft(it.expr, binder)})
-            win_map[i] = wname
+    agg_win = has_agg and any(_has_window(it.expr) for it in stmt.items)
+    if not agg_win:
+        for i, it in enumerate(stmt.items):
+            if isinstance(it.expr, P.WindowExpr):
+                wname = it.alias or it.expr.func.name
+                df = df.with_window_columns(
+                    {wname: _window_to_daft(it.expr, binder)})
+                win_map[i] = wname
+            elif _has_window(it.expr):
+                # window embedded in an expression: hidden window
+                # columns + residual evaluated afterwards
+                wsub = []
+                residual = _extract_window_subtrees(it.expr, wsub)
+                for nm, we in wsub:
+                    df = df.with_window_columns(
+                        {nm: _window_to_daft(we, binder)})
+                binder.tables["__win__"] = {nm: nm for nm, _ in wsub}
+                name = it.alias or _default_name(it.expr, binder)
+                df = df.with_column(name,
+                                    expr_to_daft(residual, binder))
+                del binder.tables["__win__"]
+                win_map[i] = name
 
-    if has_agg:
+    if agg_win:
+        # windows over GROUP BY results (rank() OVER (ORDER BY sum(v))):
+        # aggregate first, then window over the aggregated frame with
+        # aggregate sub-expressions rewritten to their output columns
+        # (hidden extra aggregates when the window uses one the SELECT
+        # list doesn't) — ref: daft-sql window-over-aggregate planning
+        import dataclasses as _dc
+        name_of = []
+        for it in stmt.items:
+            if it.star or _has_window(it.expr):
+                continue
+            name_of.append((it.expr, it.alias or
+                            _default_name(it.expr, binder)))
+        hidden: List[Tuple[Any, str]] = []
+
+        def _rw(node):
+            for ast, nm in name_of:
+                if node == ast:
+                    return P.Col(None, nm)
+            if isinstance(node, P.WindowExpr):
+                # the window FUNC head itself is not a group aggregate —
+                # rewrite only its arguments / partition / order exprs
+                fn = node.func
+                new_fn = P.FuncCall(fn.name, [_rw(a) for a in fn.args],
+                                    fn.distinct, fn.star) \
+                    if isinstance(fn, P.FuncCall) else _rw(fn)
+                return P.WindowExpr(
+                    new_fn, [_rw(x) for x in node.partition_by],
+                    [(_rw(e), d) for e, d in node.order_by], node.frame)
+            if isinstance(node, P.FuncCall) and _has_aggregate(node):
+                for h_ast, h_nm in hidden:
+                    if node == h_ast:
+                        return P.Col(None, h_nm)
+                h_nm = f"__wagg{len(hidden)}"
+                hidden.append((node, h_nm))
+                return P.Col(None, h_nm)
+            if _dc.is_dataclass(node) and not isinstance(node, type):
+                kw = {}
+                for f in _dc.fields(node):
+                    v = getattr(node, f.name)
+                    if isinstance(v, list):
+                        kw[f.name] = [
+                            tuple(_rw(y) if _dc.is_dataclass(y) else y
+                                  for y in x) if isinstance(x, tuple)
+                            else (_rw(x) if _dc.is_dataclass(x) else x)
+                            for x in v]
+                    elif _dc.is_dataclass(v) and not isinstance(v, type):
+                        kw[f.name] = _rw(v)
+                    else:
+                        kw[f.name] = v
+                return type(node)(**kw)
+            return node
+
+        rewritten = {}
+        for i, it in enumerate(stmt.items):
+            if _has_window(it.expr):
+                rewritten[i] = _rw(it.expr)
+        base_items = [it for it in stmt.items
+                      if not _has_window(it.expr)]
+        base_items += [P.SelectItem(ast, nm) for ast, nm in hidden]
+        agg_stmt = _dc.replace(stmt, items=base_items, order_by=[],
+                               limit=None, offset=None, qualify=None,
+                               distinct=False)
+        df = _plan_aggregate(df, agg_stmt, binder, select_aliases,
+                             lookup, ctes)
+        post_binder = Binder(
+            {"__agg__": {c: c for c in df.column_names()}})
+        for i, it in enumerate(stmt.items):
+            if i not in rewritten:
+                continue
+            wname = it.alias or _default_name(it.expr, binder) \
+                if not isinstance(it.expr, P.WindowExpr) \
+                else (it.alias or it.expr.func.name)
+            wsub = []
+            residual = _extract_window_subtrees(rewritten[i], wsub)
+            for nm, we in wsub:
+                df = df.with_window_columns(
+                    {nm: _window_to_daft(we, post_binder)})
+            pb2 = Binder({"__agg__": {c: c for c in df.column_names()}})
+            df = df.with_column(wname, expr_to_daft(residual, pb2))
+            win_map[i] = wname
+        sel = []
+        for i, it in enumerate(stmt.items):
+            if i in win_map:
+                sel.append(col(win_map[i]))
+            else:
+                sel.append(col(it.alias or
+                               _default_name(it.expr, binder)))
+        df = df.select(*sel)
+        out_names = df.column_names()
+    elif has_agg:
         df = _plan_aggregate(df, stmt, binder, select_aliases, lookup,
                              ctes)
         out_names = _output_names(stmt, binder)
@@ -825,6 +928,53 @@ def _window_to_daft(we, binder):
                "max": e.max, "count": e.count, "stddev": e.stddev}[fn]()
         return agg.over(w)
     raise SQLPlanError(f"unsupported window function {fn!r}")
+
+
+
+def _has_window(e) -> bool:
+    if isinstance(e, P.WindowExpr):
+        return True
+    import dataclasses as _dc
+    if _dc.is_dataclass(e) and not isinstance(e, type):
+        for f in _dc.fields(e):
+            v = getattr(e, f.name)
+            if isinstance(v, (list, tuple)):
+                for x in v:
+                    if isinstance(x, tuple):
+                        if any(_has_window(y) for y in x):
+                            return True
+                    elif _has_window(x):
+                        return True
+            elif _has_window(v):
+                return True
+    return False
+
+
+def _extract_window_subtrees(ast, out):
+    """Replace every WindowExpr subtree with Col(__w{i}); append
+    (name, window_ast) to `out`.  Returns the residual AST."""
+    import dataclasses as _dc
+    if isinstance(ast, P.WindowExpr):
+        nm = f"__w{len(out)}"
+        out.append((nm, ast))
+        return P.Col(None, nm)
+    if _dc.is_dataclass(ast) and not isinstance(ast, type):
+        kw = {}
+        for f in _dc.fields(ast):
+            v = getattr(ast, f.name)
+            if isinstance(v, list):
+                kw[f.name] = [
+                    tuple(_extract_window_subtrees(y, out)
+                          if _dc.is_dataclass(y) else y for y in x)
+                    if isinstance(x, tuple) else
+                    (_extract_window_subtrees(x, out)
+                     if _dc.is_dataclass(x) else x) for x in v]
+            elif _dc.is_dataclass(v) and not isinstance(v, type):
+                kw[f.name] = _extract_window_subtrees(v, out)
+            else:
+                kw[f.name] = v
+        return type(ast)(**kw)
+    return ast
 
 
 def _has_aggregate(e) -> bool:

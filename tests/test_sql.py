@@ -255,3 +255,28 @@ def test_sql_datediff_units():
               "DATEDIFF('year', CAST(d AS DATE), "
               "CAST('2026-01-01' AS DATE)) AS dy FROM t").to_pydict()
     assert out == {"dd": [30], "dm": [5], "dy": [2]}
+
+
+def test_sql_window_over_group_by():
+    """Windows over aggregated results: rank by sum, percent-of-total
+    with a window embedded in arithmetic, hidden window aggregates
+    (ref: daft-sql window-over-aggregate)."""
+    t = daft.from_pydict({"g": ["a", "a", "b", "b", "c"],
+                          "v": [1, 2, 3, 4, 10]})
+    out = sql("SELECT g, sum(v) AS s, RANK() OVER (ORDER BY sum(v) DESC)"
+              " AS r FROM t GROUP BY g ORDER BY g").to_pydict()
+    assert out == {"g": ["a", "b", "c"], "s": [3, 7, 10], "r": [3, 2, 1]}
+    out2 = sql("SELECT g, ROW_NUMBER() OVER (ORDER BY max(v)) AS rn "
+               "FROM t GROUP BY g ORDER BY g").to_pydict()
+    assert out2 == {"g": ["a", "b", "c"], "rn": [1, 2, 3]}
+    out3 = sql("SELECT g, sum(v) AS s, sum(v) * 100.0 / "
+               "SUM(sum(v)) OVER () AS pct FROM t GROUP BY g "
+               "ORDER BY g").to_pydict()
+    assert out3["pct"] == [15.0, 35.0, 50.0]
+
+
+def test_sql_window_embedded_in_expression():
+    t = daft.from_pydict({"g": ["a", "a", "b"], "v": [1, 3, 4]})
+    out = sql("SELECT g, v, v * 100.0 / SUM(v) OVER (PARTITION BY g) AS "
+              "pct FROM t ORDER BY g, v").to_pydict()
+    assert out["pct"] == [25.0, 75.0, 100.0]
