@@ -6,6 +6,43 @@ import threading
 from typing import Dict, List, Optional
 
 
+# catalog property bags and errors (ref: daft/catalog/__init__.py:73-76)
+Properties = dict
+
+
+class NotFoundError(Exception):
+    """Raised when a catalog object (table/namespace/function) is
+    missing (ref: daft.catalog.NotFoundError)."""
+
+
+class Function:
+    """A catalog-registered function: calling it with expression
+    arguments produces an Expression (ref: daft/catalog Function ABC)."""
+
+    def __init__(self, identifier, fn=None):
+        self._identifier = identifier if isinstance(identifier, Identifier) \
+            else Identifier(str(identifier))
+        self._fn = fn
+
+    @property
+    def identifier(self):
+        return self._identifier
+
+    @property
+    def name(self):
+        return self._identifier.parts[-1]
+
+    @property
+    def namespace(self):
+        return Identifier(*self._identifier.parts[:-1]) \
+            if len(self._identifier.parts) > 1 else Identifier()
+
+    def __call__(self, *args, **kwargs):
+        if self._fn is None:
+            raise NotFoundError(f"function {self.name} has no binding")
+        return self._fn(*args, **kwargs)
+
+
 class Identifier:
     """Dotted table identifier (ref: daft-catalog Identifier)."""
 

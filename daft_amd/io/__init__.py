@@ -312,3 +312,11 @@ read_paimon = _gated_reader("read_paimon", "paimon")
 read_huggingface = _gated_reader("read_huggingface",
                                  "network access to the HF hub")
 read_video_frames = _gated_reader("read_video_frames", "ffmpeg")
+
+from .object_store import (  # noqa: F401,E402  config/API parity exports
+    AzureConfig, CosConfig, GCSConfig, GooseFSConfig, GravitinoConfig,
+    HTTPConfig, HdfsConfig, HuggingFaceConfig, IOConfig, S3Config,
+    S3Credentials, TosConfig, UnityConfig)
+from .source import (  # noqa: F401,E402
+    DataSource, DataSourceTask, Pushdowns, read_source)
+from .sink import DataSink, WriteResult  # noqa: F401,E402

