@@ -71,5 +71,5 @@ from .session_api import (  # noqa: E402,F401
     get_loaded_extension_paths, with_subscriber, register_viz_hook,
     refresh_logger, planning_config_ctx, runners, get_or_create_runner,
     get_or_infer_runner_type, set_runner_ray, from_dask_dataframe,
-    from_ray_dataset)
+    from_ray_dataset, list_namespaces)
 from .session_api import range  # noqa: E402,F401,A004

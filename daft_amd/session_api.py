@@ -233,6 +233,15 @@ def list_tables(pattern: Optional[str] = None) -> List[str]:
     return out
 
 
+def list_namespaces(pattern: Optional[str] = None) -> List[str]:
+    """Available namespaces in the current session (ref:
+    daft/session.py list_namespaces)."""
+    out = sorted(current_session().options.get("namespaces", set()))
+    if pattern:
+        out = [n for n in out if pattern in n]
+    return out
+
+
 # namespaces (flat in the memory catalog: namespace == name prefix)
 def create_namespace(name: str) -> None:
     current_session().options.setdefault("namespaces", set()).add(name)

@@ -702,6 +702,7 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
 
     # window functions: compute as extra columns before projection
     win_map = {}
+    _wsub_all: List[Tuple[str, Any]] = []
     agg_win = has_agg and any(_has_window(it.expr) for it in stmt.items)
     if not agg_win:
         for i, it in enumerate(stmt.items):
@@ -712,10 +713,12 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
                 win_map[i] = wname
             elif _has_window(it.expr):
                 # window embedded in an expression: hidden window
-                # columns + residual evaluated afterwards
-                wsub = []
+                # columns + residual evaluated afterwards (shared name
+                # counter keeps hidden columns unique across items)
+                wsub = _wsub_all
+                n0 = len(wsub)
                 residual = _extract_window_subtrees(it.expr, wsub)
-                for nm, we in wsub:
+                for nm, we in wsub[n0:]:
                     df = df.with_window_columns(
                         {nm: _window_to_daft(we, binder)})
                 binder.tables["__win__"] = {nm: nm for nm, _ in wsub}
@@ -798,9 +801,10 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
             wname = it.alias or _default_name(it.expr, binder) \
                 if not isinstance(it.expr, P.WindowExpr) \
                 else (it.alias or it.expr.func.name)
-            wsub = []
+            wsub = _wsub_all
+            n0 = len(wsub)
             residual = _extract_window_subtrees(rewritten[i], wsub)
-            for nm, we in wsub:
+            for nm, we in wsub[n0:]:
                 df = df.with_window_columns(
                     {nm: _window_to_daft(we, post_binder)})
             pb2 = Binder({"__agg__": {c: c for c in df.column_names()}})
