@@ -166,6 +166,20 @@ class BinaryOp(ExprNode):
             return Field(lf.name, lf.dtype)
         if lf.dtype.is_string() and self.op == "add":
             return Field(lf.name, DataType.string())
+        if lf.dtype.is_decimal() or rf.dtype.is_decimal():
+            from ..schema import decimal_binary_result
+
+            def _digs(node):
+                if isinstance(node, Literal) and \
+                        isinstance(node.value, int) and \
+                        not isinstance(node.value, bool):
+                    return max(1, len(str(abs(node.value))))
+                return None
+            out = decimal_binary_result(lf.dtype, rf.dtype, self.op,
+                                        _digs(self.left),
+                                        _digs(self.right))
+            if out is not None:
+                return Field(lf.name, out)
         return Field(lf.name, supertype(lf.dtype, rf.dtype))
 
     def evaluate(self, batch) -> Series:

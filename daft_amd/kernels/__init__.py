@@ -596,38 +596,41 @@ def _decimal_binary(l: Series, r: Series, op: str, validity):
 
     av, asc, ap = scaled(l)
     bv, bsc, bp = scaled(r)
+    # result dtype via the SAME rule static typing uses
+    # (schema.decimal_binary_result) — schema and runtime never diverge
+    from ..schema import decimal_binary_result
+
+    def int_digits(s, p):
+        return p if not s.dtype.is_decimal() else None
+    res_dt = decimal_binary_result(l.dtype, r.dtype, op,
+                                   int_digits(l, ap), int_digits(r, bp))
+    if res_dt is None or res_dt.is_floating():
+        return binary_op(to_float(l), to_float(r), op)
     if op in ("add", "sub"):
-        sc = max(asc, bsc)
+        if res_dt.precision > 18:
+            # overflows scaled int64: exact two-limb wide path
+            return _decimal_wide_binary(l, r, op, validity, res_dt)
+        sc = res_dt.scale
         if asc < sc:
             av = av * (10 ** (sc - asc))
         if bsc < sc:
             bv = bv * (10 ** (sc - bsc))
         out = av + bv if op == "add" else av - bv
-        p = min(18, max(ap - asc, bp - bsc) + sc + 1)
-        if sc > 18:
-            return binary_op(to_float(l), to_float(r), op)
-        return Series(l.name, DataType.decimal128(p, sc), data=out,
+        return Series(l.name, res_dt, data=out,
                       validity=_expand_validity(validity,
                                                 int(out.shape[0])))
     if op == "mul":
-        sc = asc + bsc
-        p = min(18, ap + bp)
-        if (ap - asc) + (bp - bsc) + sc > 18:
-            # result exceeds int64: promote to an exact wide decimal
-            # when it fits p38, else f64
-            if sc <= 38 and ap + bp + 1 <= 38:
-                return _decimal_wide_binary(l, r, op, validity)
-            return binary_op(to_float(l), to_float(r), op)
+        if res_dt.precision > 18:
+            return _decimal_wide_binary(l, r, op, validity, res_dt)
         out = av * bv
-        return Series(l.name, DataType.decimal128(p, sc), data=out,
+        return Series(l.name, res_dt, data=out,
                       validity=_expand_validity(validity,
                                                 int(out.shape[0])))
-    if op in ("floordiv", "mod"):
-        return binary_op(to_float(l), to_float(r), op)
     return None
 
 
-def _decimal_wide_binary(l: Series, r: Series, op: str, validity):
+def _decimal_wide_binary(l: Series, r: Series, op: str, validity,
+                         res_dt=None):
     """Exact add/sub/mul when either operand (or the result) is a wide
     decimal: two-limb carry arithmetic (kernels/decimal128.py), device-
     portable torch ops."""
@@ -663,7 +666,8 @@ def _decimal_wide_binary(l: Series, r: Series, op: str, validity):
         sc = asc + bsc
         olo, ohi = d128.mul128(alo, ahi, blo, bhi)
         p = min(38, ap + bp + 1)
-    dt = DataType.decimal128(max(p, 19), sc)
+    dt = res_dt if res_dt is not None else \
+        DataType.decimal128(max(p, 19), sc)
     return d128.make(l.name, dt, olo, ohi,
                      _expand_validity(validity, int(olo.numel())))
 

@@ -324,6 +324,52 @@ def _int_width(k: TypeKind) -> int:
             "uint8": 8, "uint16": 16, "uint32": 32, "uint64": 64}[k.value]
 
 
+def decimal_binary_result(a: "DataType", b: "DataType", op: str,
+                          a_digits=None, b_digits=None):
+    """Result dtype of ARITHMETIC between decimal/integer operands —
+    the single rule both static typing (BinaryOp.to_field) and the
+    runtime kernel (_decimal_binary) follow, so schema and data dtypes
+    never diverge.  Returns None when plain float semantics apply.
+    a_digits/b_digits: known digit count of an integer-literal operand
+    (defaults to int64's 19)."""
+    if not (a.is_decimal() or b.is_decimal()):
+        return None
+    if op in ("div", "pow", "floordiv", "mod"):
+        return DataType.float64()
+    if a.is_floating() or b.is_floating():
+        return DataType.float64()
+    for dt in (a, b):
+        if not (dt.is_decimal() or dt.is_integer() or dt.is_null()):
+            return None
+
+    def parts(dt, digs):
+        if dt.is_decimal():
+            return dt.precision, dt.scale
+        return (digs if digs is not None else 19), 0
+
+    ap, asc = parts(a, a_digits)
+    bp, bsc = parts(b, b_digits)
+    if op in ("add", "sub"):
+        sc = max(asc, bsc)
+        if sc > 18:
+            return DataType.float64()
+        p = max(ap - asc, bp - bsc) + sc + 1
+        if p > 18:
+            return DataType.decimal128(max(19, min(38, p)), sc)
+        return DataType.decimal128(p, sc)
+    if op == "mul":
+        sc = asc + bsc
+        if sc > 18:
+            return DataType.float64()
+        if (ap - asc) + (bp - bsc) + sc > 18:
+            if sc <= 38 and ap + bp + 1 <= 38:
+                return DataType.decimal128(max(19, min(38, ap + bp + 1)),
+                                           sc)
+            return DataType.float64()
+        return DataType.decimal128(min(18, ap + bp), sc)
+    return None
+
+
 def supertype(a: DataType, b: DataType) -> DataType:
     """Least common supertype for binary operations."""
     if a == b:
