@@ -919,9 +919,12 @@ class HuggingFaceSource(ObjectSource):
                     out.append((f"hf://{repo_type}/{repo}{at}/{p}",
                                 int(item.get("size", 0))))
 
-        # list from the deepest directory of the static prefix
+        # list from the deepest directory of the static prefix, then
+        # keep only entries under the full requested prefix
         walk(prefix.rsplit("/", 1)[0] if "/" in prefix else "")
-        return [(p, sz) for (p, sz) in out]
+        at = "" if rev == "main" else f"@{rev}"
+        want = f"hf://{repo_type}/{repo}{at}/{prefix}"
+        return [(p, sz) for (p, sz) in out if p.startswith(want)]
 
 
 # ---------------------------------------------------------------------------
