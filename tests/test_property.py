@@ -274,3 +274,31 @@ def test_wide_decimal_grouped_sum_property(rows):
     for k, v in zip(keys, vals):
         want[k] += v
     assert out["s"] == [want[k] for k in out["k"]]
+
+
+@given(st.lists(st.one_of(st.none(), st.integers(-50, 50)), min_size=1,
+                max_size=80),
+       st.integers(-60, 60), st.integers(-60, 60),
+       st.sampled_from(["lt", "le", "gt", "ge", "eq", "ne"]),
+       st.booleans(), st.booleans())
+@settings(max_examples=80, deadline=None)
+def test_truthvalue_fold_never_changes_results(vals, a, b, op, neg, disj):
+    """The stats-based filter folding must be result-invariant: compare
+    every (possibly negated / disjunctive) range predicate with the rule
+    disabled."""
+    import os
+    df = daft.from_pydict({"x": vals})
+    df.collect()
+    c = col("x")
+    cmp1 = getattr(c, f"__{op}__")(a)
+    pred = ~cmp1 if neg else cmp1
+    if disj:
+        pred = pred | (c <= b)
+    q = df.where(pred)
+    got = sorted(v for v in q.to_pydict()["x"] if v is not None)
+    os.environ["DAFT_AMD_DISABLE_RULES"] = "statsfold"
+    try:
+        want = sorted(v for v in q.to_pydict()["x"] if v is not None)
+    finally:
+        del os.environ["DAFT_AMD_DISABLE_RULES"]
+    assert got == want
