@@ -280,3 +280,24 @@ def test_sql_window_embedded_in_expression():
     out = sql("SELECT g, v, v * 100.0 / SUM(v) OVER (PARTITION BY g) AS "
               "pct FROM t ORDER BY g, v").to_pydict()
     assert out["pct"] == [25.0, 75.0, 100.0]
+
+
+def test_sql_statements():
+    """Non-SELECT statements (ref: daft-sql statement.rs — ShowTables,
+    Use, CreateTable, Describe)."""
+    t = daft.from_pydict({"a": [1, 2]})
+    assert sql("CREATE TABLE tmp_ct AS SELECT a * 2 AS b FROM t") \
+        .to_pydict() == {"b": [2, 4]}
+    assert sql("SELECT * FROM tmp_ct").to_pydict() == {"b": [2, 4]}
+    assert "tmp_ct" in sql("SHOW TABLES").to_pydict()["table"]
+    assert sql("SHOW TABLES LIKE 'tmp%'").to_pydict()["table"] == \
+        ["tmp_ct"]
+    d = sql("DESCRIBE tmp_ct").to_pydict()
+    assert d == {"column_name": ["b"], "type": ["Int64"]}
+    d2 = sql("DESCRIBE SELECT a, a * 1.5 AS f FROM t").to_pydict()
+    assert d2["type"] == ["Int64", "Float64"]
+    sql("DROP TABLE tmp_ct")
+    with pytest.raises(Exception):
+        sql("SELECT * FROM tmp_ct").collect()
+    sql("DROP TABLE IF EXISTS tmp_ct")      # no raise
+    assert sql("USE cat.ns").to_pydict() == {"ok": [True]}
