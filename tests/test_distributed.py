@@ -395,3 +395,35 @@ def _wide_decimal_exchange_checks(rank, world):
 @pytest.mark.distributed
 def test_wide_decimal_exchange_world2():
     assert _spawn("_wide_decimal_exchange_checks", world=2) == [5, 5]
+
+
+def _nested_exchange_checks(rank, world):
+    """List / struct / map columns must survive the packed exchange
+    (offsets + children buffers): shard rows, redistribute via a
+    groupby, and sort — nested payloads ride along."""
+    import daft_amd as daft
+    from daft_amd import col
+    n = 48
+    keys = [i % 4 for i in range(n)]
+    lists = [[i, i + 1] if i % 3 else [] for i in range(n)]
+    structs = [{"u": i, "v": f"s{i}"} for i in range(n)]
+    df = _shard_df({"k": keys, "l": lists, "s": structs}, rank, world)
+    out = df.groupby("k").agg(
+        col("l").list.length().sum().alias("tot")).sort("k").to_pydict()
+    want = {}
+    for k, lst in zip(keys, lists):
+        want[k] = want.get(k, 0) + len(lst)
+    assert out["k"] == sorted(want)
+    assert out["tot"] == [want[k] for k in out["k"]]
+    srt = df.sort("k").to_pydict()
+    ks = srt["k"]
+    assert all(ks[i] <= ks[i + 1] for i in range(len(ks) - 1))
+    assert all(isinstance(x, dict) and "u" in x for x in srt["s"])
+    assert all(isinstance(x, list) for x in srt["l"])
+    return sum(out["tot"])
+
+
+@pytest.mark.distributed
+def test_nested_exchange_world2():
+    tot = _spawn("_nested_exchange_checks", world=2)
+    assert tot[0] == tot[1] > 0
