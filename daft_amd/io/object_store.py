@@ -22,7 +22,7 @@ import re
 import time
 import urllib.parse
 from dataclasses import dataclass, field
-from typing import Iterator, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 
 # ---------------------------------------------------------------------------
