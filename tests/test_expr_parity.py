@@ -186,3 +186,20 @@ def test_iter_rows_arrow_format():
     assert rows[0]["a"].as_py() == 1 and rows[1]["s"].as_py() is None
     with pytest.raises(ValueError):
         next(df.iter_rows(column_format="nope"))
+
+
+def test_str_distance_and_case_methods():
+    df = daft.from_pydict({"a": ["kitten", "Hello World"],
+                           "b": ["sitting", "hello world"]})
+    out = df.select(
+        col("a").str.levenshtein(col("b")).alias("lev"),
+        col("a").str.jaro_winkler(col("b")).alias("jw"),
+        col("a").str.to_camel_case().alias("cc"),
+        col("a").str.to_kebab_case().alias("kc"),
+        col("a").str.to_title_case().alias("tc"),
+    ).to_pydict()
+    assert out["lev"] == [3, 2]
+    assert 0.7 < out["jw"][0] < 0.8
+    assert out["cc"] == ["kitten", "helloWorld"]
+    assert out["kc"] == ["kitten", "hello-world"]
+    assert out["tc"] == ["Kitten", "Hello World"]
