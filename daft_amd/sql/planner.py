@@ -683,6 +683,14 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
     for cj in subq:
         df = _apply_subquery_conjunct(df, cj, binder, lookup, ctes)
 
+    # scalar subqueries in the SELECT list (ref: planner.rs
+    # SQLExpr::Subquery): uncorrelated evaluate to literals; equality-
+    # correlated decorrelate into a grouped LEFT join (missing groups
+    # yield NULL, matching scalar-subquery semantics)
+    if any((not it.star) and _has_subquery(it.expr) for it in stmt.items):
+        df, stmt = _rewrite_select_subqueries(df, stmt, binder, lookup,
+                                              ctes)
+
     # GROUP BY ALL: every non-aggregate select item is a group key
     if any(g == "__GROUP_BY_ALL__" for g in stmt.group_by):
         import dataclasses
@@ -827,7 +835,8 @@ def _plan_select(stmt: P.SelectStmt, lookup, outer: Optional[Binder],
         exprs = []
         for i, it in enumerate(stmt.items):
             if it.star:
-                exprs.extend(col(n) for n in df.column_names())
+                exprs.extend(col(n) for n in df.column_names()
+                             if not n.startswith(("__ssv", "__ssk")))
             elif i in win_map:
                 exprs.append(col(win_map[i]))
             else:
@@ -1107,6 +1116,74 @@ def _plan_aggregate(df, stmt: P.SelectStmt, binder: Binder,
     if having_expr is not None:
         out = out.where(having_expr)
     return out.select(*finals)
+
+
+def _rewrite_select_subqueries(df, stmt, binder, lookup, ctes):
+    import dataclasses as _dc
+    counter = [0]
+    dfbox = [df]
+
+    def walk(node):
+        if isinstance(node, P.SubqueryExpr):
+            corr = _correlation_info(node.query, binder, lookup, ctes)
+            if corr is None:
+                return P.Lit(_eval_scalar_subquery(node.query, binder,
+                                                   lookup, ctes))
+            outer_cols, inner_cols, residual = corr
+            sub = node.query
+            if len(sub.items) != 1 or sub.items[0].star:
+                raise SQLPlanError(
+                    "scalar subquery must select one column")
+            agg_ast = sub.items[0].expr
+            if not _has_aggregate(agg_ast):
+                # a well-formed scalar subquery yields <= 1 row per
+                # outer key; MAX selects that single value (multi-row
+                # subqueries are rejected at runtime by other engines)
+                agg_ast = P.FuncCall("max", [agg_ast])
+            n = counter[0]
+            counter[0] += 1
+            vname = f"__ssv{n}"
+            keys = [f"__ssk{n}_{i}" for i in range(len(inner_cols))]
+            inner_stmt = P.SelectStmt(
+                items=[P.SelectItem(c, k)
+                       for c, k in zip(inner_cols, keys)] +
+                      [P.SelectItem(agg_ast, vname)],
+                from_tables=sub.from_tables, joins=sub.joins,
+                where=_rebuild_where(residual),
+                group_by=list(inner_cols))
+            sub_df = _plan_select(inner_stmt, lookup, outer=None,
+                                  ctes=ctes)
+            left = [expr_to_daft(c, binder) for c in outer_cols]
+            dfbox[0] = dfbox[0].join(sub_df, left_on=left,
+                                     right_on=[col(k) for k in keys],
+                                     how="left")
+            binder.tables.setdefault("__ssv__", {})[vname] = vname
+            return P.Col(None, vname)
+        if _dc.is_dataclass(node) and not isinstance(node, type):
+            kw = {}
+            for f in _dc.fields(node):
+                v = getattr(node, f.name)
+                if isinstance(v, list):
+                    kw[f.name] = [
+                        tuple(walk(y) if _dc.is_dataclass(y) else y
+                              for y in x) if isinstance(x, tuple)
+                        else (walk(x) if _dc.is_dataclass(x) else x)
+                        for x in v]
+                elif _dc.is_dataclass(v) and not isinstance(v, type):
+                    kw[f.name] = walk(v)
+                else:
+                    kw[f.name] = v
+            return type(node)(**kw)
+        return node
+
+    new_items = []
+    for it in stmt.items:
+        if it.star or not _has_subquery(it.expr):
+            new_items.append(it)
+        else:
+            new_items.append(P.SelectItem(walk(it.expr), it.alias))
+    stmt = _dc.replace(stmt, items=new_items)
+    return dfbox[0], stmt
 
 
 def _inline_uncorrelated(e, binder, lookup, ctes):

@@ -301,3 +301,23 @@ def test_sql_statements():
         sql("SELECT * FROM tmp_ct").collect()
     sql("DROP TABLE IF EXISTS tmp_ct")      # no raise
     assert sql("USE cat.ns").to_pydict() == {"ok": [True]}
+
+
+def test_sql_scalar_subquery_in_select():
+    """Scalar subqueries in the SELECT list (ref: planner.rs
+    SQLExpr::Subquery): uncorrelated inline, correlated decorrelate to
+    a grouped LEFT join (missing keys -> NULL)."""
+    t = daft.from_pydict({"k": [1, 2, 3], "v": [10, 20, 30]})
+    u = daft.from_pydict({"k": [1, 2], "w": [5, 7]})
+    out = sql("SELECT k, (SELECT max(w) FROM u) AS mw FROM t "
+              "ORDER BY k").to_pydict()
+    assert out == {"k": [1, 2, 3], "mw": [7, 7, 7]}
+    out2 = sql("SELECT k, (SELECT w FROM u WHERE u.k = t.k) AS w "
+               "FROM t ORDER BY k").to_pydict()
+    assert out2 == {"k": [1, 2, 3], "w": [5, 7, None]}
+    out3 = sql("SELECT k, v + (SELECT sum(w) FROM u WHERE u.k = t.k) "
+               "AS vw FROM t ORDER BY k").to_pydict()
+    assert out3["vw"] == [15, 27, None]
+    out4 = sql("SELECT *, (SELECT max(w) FROM u WHERE u.k = t.k) AS mw "
+               "FROM t ORDER BY k").to_pydict()
+    assert list(out4.keys()) == ["k", "v", "mw"]
