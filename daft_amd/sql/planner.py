@@ -1308,6 +1308,7 @@ def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
 
     conjs = _split_conj(sub.where)
     outer_cols, inner_cols, residual, neq_pairs = [], [], [], []
+    ineq_triples = []
     correlated = False
     for cj in conjs:
         refs: List[P.Col] = []
@@ -1317,8 +1318,10 @@ def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
             residual.append(cj)
             continue
         correlated = True
+        _INEQ = ("lt", "le", "gt", "ge")
         ok = (isinstance(cj, P.BinOp) and
-              (cj.op == "eq" or (allow_neq and cj.op == "ne")) and
+              (cj.op == "eq" or (allow_neq and cj.op == "ne") or
+               (allow_neq and cj.op in _INEQ)) and
               isinstance(cj.left, P.Col) and isinstance(cj.right, P.Col))
         if not ok:
             raise SQLPlanError(
@@ -1326,17 +1329,22 @@ def _correlation_info(sub: P.SelectStmt, outer_binder: Binder, lookup,
                 "correlation is decorrelated)")
         if is_outer_ref(cj.left):
             pair = (cj.left, cj.right)
+            op_outer_left = cj.op
         else:
             pair = (cj.right, cj.left)
+            op_outer_left = {"lt": "gt", "le": "ge", "gt": "lt",
+                             "ge": "le"}.get(cj.op, cj.op)
         if cj.op == "eq":
             outer_cols.append(pair[0])
             inner_cols.append(pair[1])
+        elif cj.op in _INEQ:
+            ineq_triples.append((pair[0], pair[1], op_outer_left))
         else:
             neq_pairs.append(pair)
     if not correlated:
         return None
     if allow_neq:
-        return outer_cols, inner_cols, residual, neq_pairs
+        return outer_cols, inner_cols, residual, neq_pairs, ineq_triples
     return outer_cols, inner_cols, residual
 
 
@@ -1520,7 +1528,14 @@ def _plan_exists(df, sub: P.SelectStmt, negated: bool, binder, lookup, ctes):
         n = sub_df.limit(1).count_rows()
         keep = (n > 0) != negated
         return df if keep else df.where(lit(False))
-    outer_cols, inner_cols, residual, neq_pairs = corr
+    outer_cols, inner_cols, residual, neq_pairs, ineq_triples = corr
+    if ineq_triples:
+        if neq_pairs:
+            raise SQLPlanError("EXISTS with mixed <> and range "
+                               "correlation is not supported")
+        return _plan_exists_ineq(df, sub, negated, binder, lookup, ctes,
+                                 outer_cols, inner_cols, residual,
+                                 ineq_triples)
     if neq_pairs:
         return _plan_exists_neq(df, sub, negated, binder, lookup, ctes,
                                 outer_cols, inner_cols, residual, neq_pairs)
@@ -1533,6 +1548,50 @@ def _plan_exists(df, sub: P.SelectStmt, negated: bool, binder, lookup, ctes):
     right = [col(f"__ex{i}") for i in range(len(inner_cols))]
     how = "anti" if negated else "semi"
     return df.join(sub_df, left_on=left, right_on=right, how=how)
+
+
+def _plan_exists_ineq(df, sub, negated, binder, lookup, ctes,
+                      outer_cols, inner_cols, residual, ineq_triples):
+    """EXISTS with range correlation beside (optional) equality keys:
+
+        EXISTS(SELECT 1 FROM u WHERE u.k = t.k AND u.w < t.v AND resid)
+
+    decorrelates via a row-id semi join: tag outer rows, inner-join on
+    the equality keys, filter the range predicates on the joined frame,
+    and semi/anti join the distinct matching row ids back (duckdb-style
+    general unnesting; ref: daft subquery unnesting rules)."""
+    rid = "__xrid"
+    dfr = df.add_monotonically_increasing_id(rid)
+    keys = [f"__exk{i}" for i in range(len(inner_cols))]
+    ivs = [f"__exv{i}" for i in range(len(ineq_triples))]
+    inner_stmt = P.SelectStmt(
+        items=[P.SelectItem(c, k) for c, k in zip(inner_cols, keys)] +
+              [P.SelectItem(t[1], v)
+               for t, v in zip(ineq_triples, ivs)],
+        from_tables=sub.from_tables, joins=sub.joins,
+        where=_rebuild_where(residual))
+    sub_df = _plan_select(inner_stmt, lookup, outer=None, ctes=ctes)
+    if keys:
+        joined = dfr.join(sub_df,
+                          left_on=[expr_to_daft(c, binder)
+                                   for c in outer_cols],
+                          right_on=[col(k) for k in keys], how="inner")
+    else:
+        # no equality keys: constant-key join (cross product)
+        joined = dfr.with_column("__one", lit(1)).join(
+            sub_df.with_column("__one2", lit(1)),
+            left_on=[col("__one")], right_on=[col("__one2")],
+            how="inner")
+    for (outer_ast, _inner_ast, op), v in zip(ineq_triples, ivs):
+        lhs = expr_to_daft(outer_ast, binder)
+        cmp = getattr(lhs, {"lt": "__lt__", "le": "__le__",
+                            "gt": "__gt__", "ge": "__ge__"}[op])
+        joined = joined.where(cmp(col(v)))
+    matched = joined.select(col(rid)).distinct()
+    how = "anti" if negated else "semi"
+    out = dfr.join(matched, left_on=[col(rid)], right_on=[col(rid)],
+                   how=how)
+    return out.select(*[col(c) for c in df.column_names()])
 
 
 def _plan_exists_neq(df, sub, negated, binder, lookup, ctes,

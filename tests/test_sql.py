@@ -321,3 +321,20 @@ def test_sql_scalar_subquery_in_select():
     out4 = sql("SELECT *, (SELECT max(w) FROM u WHERE u.k = t.k) AS mw "
                "FROM t ORDER BY k").to_pydict()
     assert list(out4.keys()) == ["k", "v", "mw"]
+
+
+def test_sql_exists_range_correlation():
+    """EXISTS with range (non-equality) correlation decorrelates via a
+    row-id semi join (duckdb-style general unnesting)."""
+    t = daft.from_pydict({"k": [1, 2, 3], "v": [10, 20, 30]})
+    u = daft.from_pydict({"k": [1, 2, 2], "w": [5, 25, 19]})
+    out = sql("SELECT k FROM t WHERE EXISTS (SELECT 1 FROM u WHERE "
+              "u.k = t.k AND u.w < t.v) ORDER BY k").to_pydict()
+    assert out == {"k": [1, 2]}
+    out2 = sql("SELECT k FROM t WHERE NOT EXISTS (SELECT 1 FROM u WHERE "
+               "u.k = t.k AND u.w < t.v) ORDER BY k").to_pydict()
+    assert out2 == {"k": [3]}
+    # pure range correlation (no equality keys): constant-key join
+    out3 = sql("SELECT k FROM t WHERE EXISTS (SELECT 1 FROM u WHERE "
+               "u.w >= t.v) ORDER BY k").to_pydict()
+    assert out3 == {"k": [1, 2]}
