@@ -1428,10 +1428,10 @@ def _try_fuse_exists_pair(df, subq_conjs, binder, lookup, ctes):
                                      allow_neq=True)
         except SQLPlanError:
             continue
-        if corr is None or len(corr) != 4 or not corr[3]:
+        if corr is None or len(corr) != 5 or not corr[3]:
             continue
-        outer_cols, inner_cols, residual, neq_pairs = corr
-        if len(neq_pairs) != 1 or not inner_cols:
+        outer_cols, inner_cols, residual, neq_pairs, ineq_triples = corr
+        if len(neq_pairs) != 1 or not inner_cols or ineq_triples:
             continue
         cands.append((cj, sub, negated, outer_cols, inner_cols,
                       residual, neq_pairs[0]))
