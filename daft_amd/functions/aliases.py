@@ -273,8 +273,9 @@ def replace_time_zone(x, tz):
 
 def make_timestamp(y, mo, d, h, mi, s_, tz=None):
     def run(*cols) -> Series:
+        import builtins
         vs = [c.cpu().data.to(torch.int64).numpy() for c in cols]
-        n = max(len(v) for v in vs)
+        n = builtins.max(len(v) for v in vs)
         out = []
         for i in range(n):
             g = [int(v[i % len(v)]) for v in vs]
@@ -354,10 +355,11 @@ def conv(x, from_base: int, to_base: int):
     digits = "0123456789abcdefghijklmnopqrstuvwxyz"
 
     def enc(n: int, base: int) -> str:
+        import builtins
         if n == 0:
             return "0"
         neg = n < 0
-        n = abs(n)
+        n = builtins.abs(n)
         out = ""
         while n:
             out = digits[n % base] + out
