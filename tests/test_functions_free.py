@@ -147,3 +147,24 @@ def test_bpe_tokenizer_json_roundtrip(tmp_path):
     dec = enc.select(daft.functions.tokenize_decode(
         col("ids"), f"bpe:{p}").alias("t")).to_pydict()
     assert dec["t"] == ["he", "heh", None]
+
+
+def test_make_timestamp_and_conv_builtins():
+    """Regression: module-level max/abs free functions must not shadow
+    the builtins used inside make_timestamp/conv."""
+    import daft_amd as daft
+    import daft_amd.functions as F
+    from daft_amd import col
+    import datetime
+    df = daft.from_pydict({"y": [2024], "mo": [3], "d": [1], "h": [10],
+                           "mi": [5], "s": [6], "x": [255, ]})
+    t = df.select(F.make_timestamp(col("y"), col("mo"), col("d"),
+                                   col("h"), col("mi"), col("s"))
+                  .alias("t")).to_pydict()["t"]
+    assert t == [datetime.datetime(2024, 3, 1, 10, 5, 6)]
+    hx = df.select(F.conv(col("x"), 10, 16).alias("h")).to_pydict()["h"]
+    assert hx == ["ff"]
+    # the new free-function forms coexist with python builtins
+    out = df.select(F.abs(col("x") * -1).alias("a"),
+                    F.max(col("x")).alias("m")).to_pydict()
+    assert out["a"] == [255]
