@@ -276,3 +276,18 @@ def _range_partition_ids(rb: RecordBatch, keys: Sequence[str],
                 eq = eq & e
         part += gt.to(torch.int64)
     return part
+
+
+# reference-name alias: the reference's MicroPartition is its batch-of-
+# record-batches unit; here RecordBatch fills both roles
+MicroPartition = RecordBatch
+
+
+def read_parquet_into_pyarrow(path, columns=None, **kwargs):
+    """Thin helper mirroring daft.recordbatch.read_parquet_into_pyarrow."""
+    import pyarrow.parquet as pq
+    return pq.read_table(path, columns=columns)
+
+
+def read_parquet_into_pyarrow_bulk(paths, columns=None, **kwargs):
+    return [read_parquet_into_pyarrow(p, columns=columns) for p in paths]

@@ -99,3 +99,23 @@ class ProgressSubscriber(Subscriber):
         status = "failed" if error else "done"
         print(f"[daft_amd] query {query_id} {status} in {seconds:.2f}s",
               file=self.file, flush=True)
+
+
+import enum
+
+
+class StatType(enum.Enum):
+    """Operator stat kinds surfaced to subscribers (ref:
+    daft/subscribers StatType)."""
+    COUNT = "count"
+    BYTES = "bytes"
+    PERCENT = "percent"
+    FLOAT = "float"
+    DURATION = "duration"
+
+
+def launch(detach: bool = False, port: int = 8238):
+    """Launch the dashboard subscriber (ref: daft.subscribers.launch —
+    starts the dashboard server and attaches its subscriber)."""
+    from ..dashboard import serve
+    return serve(port=port)

@@ -199,3 +199,40 @@ def run_udf_node(node: PyUDF, batch) -> Series:
         from .series import full_null
         return full_null(node.name, node.return_dtype, n, batch.device)
     raise last_err
+
+
+class UDF:
+    """Callable UDF wrapper (ref: daft/udf/legacy.py UDF dataclass):
+    holds the function + options; calling it builds the expression.
+    `daft_amd.udf(...)` decorators normally produce plain callables;
+    this class is the reference-shaped handle for programmatic use."""
+
+    def __init__(self, fn: Callable, return_dtype: DataType,
+                 batch_size: Optional[int] = None, num_gpus=None,
+                 concurrency: Optional[int] = None,
+                 max_retries: int = 0, on_error: str = "raise"):
+        self.fn = fn
+        self.return_dtype = return_dtype
+        self.batch_size = batch_size
+        self.num_gpus = num_gpus
+        self.concurrency = concurrency
+        self.max_retries = max_retries
+        self.on_error = on_error
+        functools.update_wrapper(self, fn)
+
+    def __call__(self, *args):
+        nodes = [_to_node(a) for a in args]
+        return Expression(PyUDF(self.fn.__name__, self.fn, nodes,
+                                self.return_dtype, batched=True,
+                                max_retries=self.max_retries,
+                                on_error=self.on_error,
+                                gpus=int(self.num_gpus or 0),
+                                concurrency=self.concurrency))
+
+    def override_options(self, **kw) -> "UDF":
+        opts = dict(fn=self.fn, return_dtype=self.return_dtype,
+                    batch_size=self.batch_size, num_gpus=self.num_gpus,
+                    concurrency=self.concurrency,
+                    max_retries=self.max_retries, on_error=self.on_error)
+        opts.update(kw)
+        return UDF(**opts)
