@@ -137,3 +137,27 @@ def load_provider(name: str = "hash", **kwargs):
 def register_provider(name: str, provider) -> None:
     with _lock:
         _providers[f"{name}:[]"] = provider
+
+
+# reference-name surface (daft.ai): Embedding is the vector value type
+# (here a fixed-size-list / numpy vector); Provider the plugin base
+from typing import Any, List, Union  # noqa: E402
+
+Embedding = Union[List[float], "np.ndarray"]  # type: ignore[name-defined]
+
+
+class Provider:
+    """Base class for AI providers (ref: daft/ai Provider protocol):
+    subclasses expose get_text_embedder / get_image_embedder /
+    get_text_classifier and register via register_provider()."""
+
+    name: str = "provider"
+
+    def get_text_embedder(self, model: str = "", **kwargs):
+        raise NotImplementedError
+
+    def get_image_embedder(self, model: str = "", **kwargs):
+        raise NotImplementedError
+
+    def get_text_classifier(self, model: str = "", **kwargs):
+        raise NotImplementedError

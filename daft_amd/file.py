@@ -127,3 +127,20 @@ class File:
         tmp.flush()
         tmp.seek(0)
         return tmp
+
+
+class DaftFileIO:
+    """File-opening adapter over the object-store layer (ref:
+    daft/file DaftFileIO): open(path) returns a binary file object for
+    local or remote (s3/gs/az/hf/http) paths."""
+
+    def __init__(self, io_config=None):
+        self.io_config = io_config
+
+    def open(self, path: str, mode: str = "rb"):
+        if "w" in mode:
+            raise NotImplementedError("DaftFileIO is read-only here")
+        from .io.object_store import get_source, is_remote
+        if is_remote(path):
+            return get_source(path, self.io_config).open(path)
+        return open(path, mode)

@@ -320,3 +320,7 @@ from .object_store import (  # noqa: F401,E402  config/API parity exports
 from .source import (  # noqa: F401,E402
     DataSource, DataSourceTask, Pushdowns, read_source)
 from .sink import DataSink, WriteResult  # noqa: F401,E402
+from .vendor import (  # noqa: E402,F401  gated vendor surfaces
+    BigtableDataSink, ClickHouseDataSink, GravitinoCatalog,
+    GravitinoClient, PaimonDataSink, TurbopufferDataSink, UnityCatalog,
+    UnityCatalogClient, UnityCatalogTable, load_gravitino)
